@@ -1,0 +1,217 @@
+"""Benchmark: TPC-H Q1 hot path on MI355X (BASELINE.json configs[1]).
+
+A "step" = one full Q1 pass over the HBM-resident SF100 lineitem columns:
+fused filter+group-by partial-agg kernel (qk_q1_agg) + partial combine +
+host finalize — i.e. the whole reference pipeline of SURVEY.md §3.4 with
+scan IO excluded (inputs resident, `data: synthetic`).
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W] [--sf S]
+Multi-rank: launched by torch.distributed.run, one rank per GPU (weak
+scaling: each rank owns a full SF-sized shard; the only exchange on Q1's
+path is the 384-byte partial-aggregate combine — SURVEY.md §8e / BASELINE
+configs[1] is "no shuffle" — done host-side via gloo).
+
+Prints ONE JSON line from rank 0 (driver contract), including:
+  roofline      — dominant-kernel HBM roofline measured with HIP events on
+                  the launching stream (38 algorithmic B/row)
+  cpu_baseline  — the CPU oracle (numpy restatement, oracle/queries.py)
+                  timed on this box's host cores over a bounded sample
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+SF100_LINEITEM_ROWS = 600_037_902   # dbgen SF100 cardinality
+Q1_BYTES_PER_ROW = 38               # 4*f64 + i32 + 2*u8 read once
+HBM_PEAK_GBPS = 8000.0              # MI355X spec (measured achievable ~6290)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--sf", type=float, default=100.0,
+                   help="scale factor per GPU (rows = SF/100 * 600037902)")
+    p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
+    p.add_argument("--skip-cpu-baseline", action="store_true")
+    return p.parse_args()
+
+
+def gen_device_lineitem(shim, n, rank):
+    """Device-side synthetic lineitem (Q1 columns) for this rank's shard."""
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    cols = {name: DevColumn(dt, n) for name, dt in [
+        ("l_quantity", np.float64), ("l_extendedprice", np.float64),
+        ("l_discount", np.float64), ("l_tax", np.float64),
+        ("l_returnflag", np.uint8), ("l_linestatus", np.uint8),
+        ("l_shipdate", np.int32)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(150_000_000),
+              None, None,
+              cols["l_quantity"].ptr, cols["l_extendedprice"].ptr,
+              cols["l_discount"].ptr, cols["l_tax"].ptr,
+              cols["l_returnflag"].ptr, cols["l_linestatus"].ptr,
+              cols["l_shipdate"].ptr)
+    return cols
+
+
+def cpu_baseline(sample_rows, target_seconds=12.0):
+    """Time the CPU oracle restatement (kind='port') on a bounded sample."""
+    from oracle import tpch_gen as G, queries as OQ
+    sf = sample_rows / 6_000_000
+    li = G.gen_lineitem(sf, seed=42)
+    n = len(li["l_shipdate"])
+    # one calibration pass, then enough passes for ~target_seconds
+    t0 = time.time()
+    OQ.q1_partials(li)
+    per = time.time() - t0
+    passes = max(1, min(16, int(target_seconds / max(per, 1e-3))))
+    t0 = time.time()
+    for _ in range(passes):
+        OQ.q1_partials(li)
+    dt = time.time() - t0
+    return {
+        "value": n * passes / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": "%.1fM-row seeded lineitem x %d passes of the numpy "
+                  "oracle (oracle/queries.py:q1_partials), single-threaded"
+                  % (n / 1e6, passes),
+    }
+
+
+def read_traffic():
+    """Per-launch HBM bytes from the committed rocprofv3 PMC measurement
+    (profiles/traffic_q1.json), or None before one exists."""
+    path = os.path.join(ROOT, "profiles", "traffic_q1.json")
+    if os.path.exists(path):
+        with open(path) as f:
+            d = json.load(f)
+        return d.get("traffic_bytes_per_launch")
+    return None
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    from quokka_amd import shim, ops, queries as DQ
+    shim.init(local_rank)
+
+    n = int(round(args.sf / 100.0 * SF100_LINEITEM_ROWS))
+    n &= ~1  # even row count -> fully vectorized path
+    cols = gen_device_lineitem(shim, n, rank)
+    stream = shim.Stream()
+    timer = shim.Timer()
+    acc = None
+
+    torch = None
+    part_t = None
+    if dist is not None:
+        import torch  # only needed for the gloo combine; slow first import
+        part_t = torch.zeros(48, dtype=torch.float64)
+
+    def step(timed):
+        nonlocal acc
+        from quokka_amd.shim import DevBuffer, c_u64
+        if acc is None:
+            acc = DevBuffer(48 * 8)
+        shim.call("qk_dmemset", acc.ptr, 0, c_u64(48 * 8))
+        if timed:
+            timer.start(stream)
+        DQ.q1_partials_device(cols, stream=stream, acc=acc)
+        if timed:
+            timer.stop(stream)
+        stream.sync()
+        p = ops.q1_read_partials(acc)
+        if dist is not None:
+            part_t[:36] = torch.from_numpy(p.reshape(-1))
+            dist.all_reduce(part_t, op=dist.ReduceOp.SUM)
+            p = part_t[:36].numpy().reshape(6, 6)
+        return DQ.q1_finalize(p), (timer.elapsed_ms() if timed else None)
+
+    for _ in range(args.warmup):
+        step(False)
+    if dist is not None:
+        dist.barrier()
+    stream.sync()
+    t0 = time.time()
+    kernel_ms = []
+    result = None
+    for _ in range(args.steps):
+        result, kms = step(True)
+        kernel_ms.append(kms)
+    stream.sync()
+    elapsed = time.time() - t0
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+        dist.barrier()
+
+    if rank == 0:
+        total_rows = n * world * args.steps
+        avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
+        achieved_gbps = n * Q1_BYTES_PER_ROW / avg_kernel_s / 1e9
+        traffic = read_traffic()
+        out = {
+            "metric": "rows/s",
+            "value": total_rows / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": "TPC-H SF%g Q1, fused filter+group-by, "
+                            "%d lineitem rows/GPU resident in HBM "
+                            "(BASELINE.json configs[1])" % (args.sf, n),
+                "sf_per_gpu": args.sf,
+                "rows_per_gpu": n,
+                "query": "Q1",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbps,
+                "peak": HBM_PEAK_GBPS,
+                "unit": "GB/s",
+                "frac": achieved_gbps / HBM_PEAK_GBPS,
+                "traffic": traffic,
+            },
+            "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
+                             else cpu_baseline(args.cpu_sample_rows)),
+            "q1_result_rows": len(result["count_order"]) if result else 0,
+        }
+        print(json.dumps(out))
+
+    timer.destroy()
+    stream.destroy()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,183 @@
+/* quokka_amd — C ABI of the MI355X-native columnar execution path.
+ *
+ * This is the drop-in boundary beneath the reference's (marsupialtail/quokka)
+ * Executor / partition-function plugin API. The reference is pure Python
+ * (pyquokka/executors/base_executor.py:26-32 `Executor.execute/done`,
+ * pyquokka/core.py:152-195 `partition_fn`) delegating all arithmetic to
+ * polars/duckdb; this library replaces exactly that arithmetic with HIP
+ * kernels for gfx950. The Python classes in quokka_amd/executors.py bind
+ * these entry points via ctypes and present the reference's own interface
+ * (same names / argument meaning / error behaviour); see INTEGRATION.md for
+ * the binding a pyquokka maintainer would add.
+ *
+ * Conventions:
+ *  - every function returns 0 on success, nonzero HIP/ABI error code;
+ *    qk_last_error() returns a static string for the calling thread's last
+ *    failure. No function falls back to CPU: with no usable GPU, calls fail.
+ *  - `stream` is an opaque HIP stream handle from qk_stream_create (NULL =
+ *    default stream). All kernel entry points are asynchronous on `stream`.
+ *  - device pointers are void* from qk_dmalloc; columns are dense device
+ *    arrays (no nulls — the reference's TPC-H hot path has none).
+ *  - dates are int32 days since 1970-01-01 (Arrow date32), strings are
+ *    host-side dictionary codes (u8), matching the staging layer.
+ */
+#ifndef QUOKKA_AMD_H
+#define QUOKKA_AMD_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- context ------------------------------------------------------- */
+int qk_init(int device);                 /* set + warm the HIP device      */
+int qk_device_count(int *out);
+const char *qk_last_error(void);
+const char *qk_build_arch(void);         /* "gfx950"                       */
+
+/* ---- device memory -------------------------------------------------- */
+int qk_dmalloc(uint64_t nbytes, void **dptr);
+int qk_dfree(void *dptr);
+int qk_h2d(void *dst_dev, const void *src_host, uint64_t nbytes);
+int qk_d2h(void *dst_host, const void *src_dev, uint64_t nbytes);
+int qk_dmemset(void *dst_dev, int value, uint64_t nbytes);
+int qk_fill_i64(void *stream, int64_t *dst_dev, int64_t value, uint64_t n);
+
+/* ---- streams & timing ----------------------------------------------- */
+int qk_stream_create(void **stream);
+int qk_stream_destroy(void *stream);
+int qk_stream_sync(void *stream);
+/* HIP-event timer pair on the launching stream (roofline measurement). */
+int qk_timer_create(void **timer);
+int qk_timer_destroy(void *timer);
+int qk_timer_start(void *timer, void *stream);
+int qk_timer_stop(void *timer, void *stream);
+int qk_timer_elapsed_ms(void *timer, float *out_ms);   /* syncs the stop event */
+
+/* ---- synthetic data (bench only; distributions = oracle/tpch_gen.py) */
+/* Fill lineitem columns for rows [row_offset, row_offset+n) with the TPC-H
+ * distributions of oracle/tpch_gen.py (counter-based RNG: value depends only
+ * on (seed, global row)). Any output pointer may be NULL to skip a column.
+ * n_parts/n_suppliers/n_orders scale key ranges (pass tpch_gen values). */
+int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
+                    int64_t n_parts, int64_t n_suppliers, int64_t n_orders,
+                    int64_t *l_orderkey, int64_t *l_suppkey,
+                    double *l_quantity, double *l_extendedprice,
+                    double *l_discount, double *l_tax,
+                    uint8_t *l_returnflag, uint8_t *l_linestatus,
+                    int32_t *l_shipdate);
+
+/* ---- TPC-H Q1: fused filter + group-by partial aggregate ------------- *
+ * Replaces the map-side partial agg the reference folds into partition_fn
+ * (pyquokka/core.py:173-176 + df.py:1354-1394: per-batch DuckDB
+ * "SUM/COUNT group by l_returnflag,l_linestatus") and the probe-side concat
+ * of SQLAggExecutor.execute (sql_executors.py:587-590), in one pass.
+ * out_dev = device f64[6 groups][8]:
+ *   [sum_qty, sum_base_price, sum_disc_price, sum_charge, sum_disc, count,
+ *    pad, pad]; group id = l_returnflag*2 + l_linestatus.
+ * ACCUMULATES into out_dev (caller zeroes once; repeated calls = executor
+ * state accumulation across batches, sql_executors.py:587-590). */
+int qk_q1_agg(void *stream, uint64_t n,
+              const int32_t *l_shipdate, const double *l_quantity,
+              const double *l_extendedprice, const double *l_discount,
+              const double *l_tax, const uint8_t *l_returnflag,
+              const uint8_t *l_linestatus, int32_t cutoff_date,
+              double *out_dev /* f64[48] */);
+
+/* ---- TPC-H Q6: fused filter + sum ------------------------------------ *
+ * tpch_ref.py:171-183 semantics. out_dev = f64[2] {sum_revenue, count};
+ * accumulates like qk_q1_agg. Bounds are the exact f64 constants of the
+ * reference SQL expressions. */
+int qk_q6_agg(void *stream, uint64_t n,
+              const int32_t *l_shipdate, const double *l_quantity,
+              const double *l_extendedprice, const double *l_discount,
+              int32_t date_lo, int32_t date_hi,
+              double disc_lo, double disc_hi, double qty_hi,
+              double *out_dev /* f64[2] */);
+
+/* ---- generic filter: compare + compaction ----------------------------- *
+ * Replaces the predicate filter of partition_fn (core.py:170, polars
+ * DataFrame.filter) for single-column compares; emits the ORDERED indices
+ * of passing rows (row order preserved, as polars filter does).
+ * op: 0 '<', 1 '<=', 2 '>', 3 '>=', 4 '==', 5 '!='.
+ * out_count_dev: device u64, set to number of passing rows (must be zeroed).
+ * out_idx capacity must be >= n. */
+int qk_filter_i32(void *stream, uint64_t n, const int32_t *col, int op,
+                  int32_t value, uint32_t *out_idx, uint64_t *out_count_dev);
+int qk_filter_u8(void *stream, uint64_t n, const uint8_t *col, int op,
+                 uint8_t value, uint32_t *out_idx, uint64_t *out_count_dev);
+
+/* elementwise revenue: out[i] = a[i] * (1 - b[i]) (the per-row product the
+ * reference computes before summing, apps/tpc-h/tpch.py:151) */
+int qk_mul_1md(void *stream, uint64_t n, const double *a, const double *b,
+               double *out);
+
+/* gather: dst[i] = src[idx[i]] (column compaction / join payload gather) */
+int qk_gather_i64(void *stream, uint64_t n_idx, const uint32_t *idx,
+                  const int64_t *src, int64_t *dst);
+int qk_gather_f64(void *stream, uint64_t n_idx, const uint32_t *idx,
+                  const double *src, double *dst);
+int qk_gather_i32(void *stream, uint64_t n_idx, const uint32_t *idx,
+                  const int32_t *src, int32_t *dst);
+int qk_gather_u8(void *stream, uint64_t n_idx, const uint32_t *idx,
+                 const uint8_t *src, uint8_t *dst);
+
+/* ---- hash join build / probe (i64 keys) ------------------------------- *
+ * Replaces BuildProbeJoinExecutor (sql_executors.py:325-377): build side =
+ * stream 1 (vstacked state), probe = stream 0, how in {inner,semi,anti}
+ * (left via inner + host fill). Open-addressing table, linear probing,
+ * capacity a power of two >= 2x build rows. Duplicate build keys chain
+ * through chain_next. slot_keys must be pre-filled with QK_JOIN_EMPTY
+ * (qk_fill_i64), slot_head with -1 bytes (qk_dmemset 0xff).
+ * Build may be called repeatedly (batch accumulation): build_row_offset is
+ * added to local row indices so chains index the concatenated build side. */
+#define QK_JOIN_EMPTY INT64_MIN
+int qk_join_build(void *stream, uint64_t n_build, const int64_t *keys,
+                  uint32_t build_row_offset, int64_t *slot_keys,
+                  int32_t *slot_head, int32_t *chain_next, uint64_t capacity);
+/* mode: 0 = inner (emit probe_idx,build_idx pairs), 1 = semi (probe_idx
+ * only), 2 = anti (probe_idx only). out_cursor_dev (u64, zeroed) returns the
+ * TOTAL match count even when it exceeds out_capacity (no silent
+ * truncation: caller re-runs with larger buffers when cursor > capacity). */
+int qk_join_probe(void *stream, uint64_t n_probe, const int64_t *keys,
+                  const int64_t *slot_keys, const int32_t *slot_head,
+                  const int32_t *chain_next, uint64_t capacity, int mode,
+                  uint32_t *out_probe_idx, uint32_t *out_build_idx,
+                  uint64_t out_capacity, uint64_t *out_cursor_dev);
+
+/* ---- group-by (i64 key) sum ------------------------------------------- *
+ * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for
+ * distributive SUM over an i64 key (the post-rewrite partial form,
+ * sql_utils.py:299-413). Open-addressing accumulate table; slot_keys
+ * pre-filled with QK_JOIN_EMPTY, slot_sums zeroed. Repeated calls
+ * accumulate. nvals value columns share one key lookup (vals/slot_sums are
+ * arrays-of-columns, each `capacity` doubles: slot_sums[c*capacity+slot]). */
+int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
+                       const double *const *vals_dev, int nvals,
+                       int64_t *slot_keys, double *slot_sums,
+                       uint64_t capacity);
+/* Compact occupied slots to out_keys/out_sums (unordered); out_cursor_dev
+ * (u64, zeroed) = number of groups. out capacity must be >= group count. */
+int qk_groupby_extract(void *stream, const int64_t *slot_keys,
+                       const double *slot_sums, int nvals, uint64_t capacity,
+                       int64_t *out_keys, double *out_sums,
+                       uint64_t out_capacity, uint64_t *out_cursor_dev);
+
+/* ---- hash partition (shuffle map side) -------------------------------- *
+ * Replaces partition_key_str (quokka_runtime.py:217-231). Int key semantics
+ * bit-exact with the reference (:222): part = key % nparts (non-negative
+ * keys). hist_dev: u64[nparts], zeroed. */
+int qk_partition_hist(void *stream, uint64_t n, const int64_t *keys,
+                      uint32_t nparts, uint64_t *hist_dev);
+/* Scatter row indices grouped by partition: out_idx[offsets[p] ... ] = rows
+ * of partition p (order within a partition unspecified). cursors_dev must be
+ * initialized to the exclusive prefix sum of hist (u64[nparts]). */
+int qk_partition_scatter(void *stream, uint64_t n, const int64_t *keys,
+                         uint32_t nparts, uint64_t *cursors_dev,
+                         uint32_t *out_idx);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* QUOKKA_AMD_H */

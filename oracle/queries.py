@@ -1,0 +1,178 @@
+"""CPU oracle queries — TEST INFRASTRUCTURE ONLY.
+
+Each function restates, on numpy, the result-defining SQL of the reference's
+own golden oracle /root/reference/apps/tpc-h/tpch_ref.py (DuckDB), over the
+columns produced by oracle.tpch_gen. All float arithmetic is f64, matching
+the reference executors (DuckDB/polars compute these aggregates in double).
+
+Used only by tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg.
+"""
+import numpy as np
+
+from . import tpch_gen as G
+
+N_Q1_GROUPS = 6  # returnflag in {A,N,R} x linestatus in {F,O}
+
+
+def q1_group_id(returnflag_code, linestatus_code):
+    """Group id = rflag*2 + lstatus; code order == lexicographic order, so
+    ascending gid == 'order by l_returnflag, l_linestatus' (tpch_ref.py:34-37)."""
+    return returnflag_code.astype(np.int64) * 2 + linestatus_code.astype(np.int64)
+
+
+def q1_partials(li, cutoff=G.Q1_CUTOFF):
+    """Q1 partial aggregates (the map-side / kernel-side contract).
+
+    Restates tpch_ref.py:16-38 filter+aggregate in its two-phase form
+    (sql_utils.py:299-413 rewrite: avg -> sum+count): returns a
+    (6 groups x 6) f64 array with columns
+    [sum_qty, sum_base_price, sum_disc_price, sum_charge, sum_disc, count].
+    """
+    mask = li["l_shipdate"] <= cutoff
+    gid = q1_group_id(li["l_returnflag"], li["l_linestatus"])[mask]
+    qty = li["l_quantity"][mask]
+    price = li["l_extendedprice"][mask]
+    disc = li["l_discount"][mask]
+    tax = li["l_tax"][mask]
+    disc_price = price * (1.0 - disc)
+    charge = disc_price * (1.0 + tax)
+
+    out = np.zeros((N_Q1_GROUPS, 6), dtype=np.float64)
+    out[:, 0] = np.bincount(gid, weights=qty, minlength=N_Q1_GROUPS)
+    out[:, 1] = np.bincount(gid, weights=price, minlength=N_Q1_GROUPS)
+    out[:, 2] = np.bincount(gid, weights=disc_price, minlength=N_Q1_GROUPS)
+    out[:, 3] = np.bincount(gid, weights=charge, minlength=N_Q1_GROUPS)
+    out[:, 4] = np.bincount(gid, weights=disc, minlength=N_Q1_GROUPS)
+    out[:, 5] = np.bincount(gid, minlength=N_Q1_GROUPS)
+    return out
+
+
+def q1_finalize(partials):
+    """Final aggregate (SQLAggExecutor.done semantics, sql_executors.py:592-599):
+    sum the partials, derive avgs as sum/sum per the rewrite, drop empty
+    groups, order by (l_returnflag, l_linestatus).
+
+    Returns dict of column name -> np array, columns as tpch_ref.py:16-38.
+    """
+    p = partials if partials.ndim == 2 else partials.reshape(-1, 6)
+    nonempty = p[:, 5] > 0
+    gids = np.nonzero(nonempty)[0]
+    p = p[nonempty]
+    cnt = p[:, 5]
+    return {
+        "l_returnflag": np.array([G.RETURNFLAG[g // 2] for g in gids]),
+        "l_linestatus": np.array([G.LINESTATUS[g % 2] for g in gids]),
+        "sum_qty": p[:, 0],
+        "sum_base_price": p[:, 1],
+        "sum_disc_price": p[:, 2],
+        "sum_charge": p[:, 3],
+        "avg_qty": p[:, 0] / cnt,
+        "avg_price": p[:, 1] / cnt,
+        "avg_disc": p[:, 4] / cnt,
+        "count_order": cnt.astype(np.int64),
+    }
+
+
+def q1(li, cutoff=G.Q1_CUTOFF):
+    return q1_finalize(q1_partials(li, cutoff))
+
+
+def q6(li):
+    """tpch_ref.py:171-183. The bounds are the literal fp64 expressions the
+    reference SQL evaluates (0.06 - 0.01, 0.06 + 0.01): both oracle and GPU
+    kernel must use these exact doubles."""
+    lo, hi = 0.06 - 0.01, 0.06 + 0.01
+    m = (
+        (li["l_shipdate"] >= G.Q5_LO)
+        & (li["l_shipdate"] < G.Q5_HI)
+        & (li["l_discount"] >= lo)
+        & (li["l_discount"] <= hi)
+        & (li["l_quantity"] < 24.0)
+    )
+    return {
+        "revenue": np.float64((li["l_extendedprice"][m] * li["l_discount"][m]).sum()),
+        "rows_passed": int(m.sum()),
+    }
+
+
+def q3(li, orders, customer, limit=10):
+    """tpch_ref.py:89-115: customer(BUILDING) x orders(<1995-03-15) x
+    lineitem(ship>1995-03-15), group by (l_orderkey, o_orderdate,
+    o_shippriority), revenue = sum(extendedprice*(1-discount)),
+    order by revenue desc, o_orderdate asc, limit 10.
+
+    Exploits the generator's dense o_orderkey/c_custkey for O(n) semi-join
+    lookups; the general (key-agnostic) join restatement used for kernel
+    parity is oracle.executors.build_probe_join.
+    Returns (full_group_table_dict, top10_dict)."""
+    seg_building = G.MKTSEGMENT.index("BUILDING")
+    cust_ok = np.zeros(int(customer["c_custkey"].max()) + 2, dtype=bool)
+    cust_ok[customer["c_custkey"][customer["c_mktsegment"] == seg_building]] = True
+
+    omask = (orders["o_orderdate"] < G.Q3_DATE) & cust_ok[orders["o_custkey"]]
+    okeys = orders["o_orderkey"][omask]
+    odate = orders["o_orderdate"][omask]
+    oprio = orders["o_shippriority"][omask]
+    nkey = int(orders["o_orderkey"].max()) + 2
+    order_ok = np.zeros(nkey, dtype=bool)
+    order_ok[okeys] = True
+    odate_by_key = np.zeros(nkey, dtype=np.int32)
+    odate_by_key[okeys] = odate
+    oprio_by_key = np.zeros(nkey, dtype=np.int32)
+    oprio_by_key[okeys] = oprio
+
+    lmask = (li["l_shipdate"] > G.Q3_DATE) & order_ok[li["l_orderkey"]]
+    lkey = li["l_orderkey"][lmask]
+    rev = li["l_extendedprice"][lmask] * (1.0 - li["l_discount"][lmask])
+
+    sums = np.bincount(lkey, weights=rev, minlength=nkey)
+    hit = np.zeros(nkey, dtype=bool)
+    hit[lkey] = True
+    gk = np.nonzero(hit)[0]
+    full = {
+        "l_orderkey": gk.astype(np.int64),
+        "o_orderdate": odate_by_key[gk],
+        "o_shippriority": oprio_by_key[gk],
+        "revenue": sums[gk],
+    }
+    # order by revenue desc, o_orderdate asc, limit 10 (ties broken by
+    # orderkey asc for determinism of the fixture; the reference SQL leaves
+    # ties unordered)
+    order = np.lexsort((full["l_orderkey"], full["o_orderdate"], -full["revenue"]))
+    top = order[:limit]
+    top10 = {k: v[top] for k, v in full.items()}
+    return full, top10
+
+
+def q5(li, orders, customer, supplier, nation, region):
+    """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
+    o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate
+    c_nationkey = s_nationkey; group by n_name; order by revenue desc.
+    Returns dict n_name -> revenue (sorted desc)."""
+    asia = G.REGIONS.index("ASIA")
+    nat_in_asia = nation["n_regionkey"] == asia  # indexed by nationkey 0..24
+
+    cust_nat = np.full(int(customer["c_custkey"].max()) + 2, -1, dtype=np.int32)
+    cust_nat[customer["c_custkey"]] = customer["c_nationkey"]
+
+    omask = (orders["o_orderdate"] >= G.Q5_LO) & (orders["o_orderdate"] < G.Q5_HI)
+    okeys = orders["o_orderkey"][omask]
+    ocust = orders["o_custkey"][omask]
+    nkey = int(orders["o_orderkey"].max()) + 2
+    order_cnat = np.full(nkey, -1, dtype=np.int32)
+    ocn = cust_nat[ocust]
+    keep = (ocn >= 0) & nat_in_asia[np.clip(ocn, 0, 24)]
+    order_cnat[okeys[keep]] = ocn[keep]
+
+    supp_nat = np.full(int(supplier["s_suppkey"].max()) + 2, -1, dtype=np.int32)
+    supp_nat[supplier["s_suppkey"]] = supplier["s_nationkey"]
+
+    lcn = order_cnat[li["l_orderkey"]]
+    lsn = supp_nat[li["l_suppkey"]]
+    m = (lcn >= 0) & (lcn == lsn)
+    rev = li["l_extendedprice"][m] * (1.0 - li["l_discount"][m])
+    by_nat = np.bincount(lcn[m], weights=rev, minlength=25)
+    names = [n for n, _ in G.NATIONS]
+    out = [(names[i], by_nat[i]) for i in range(25) if nat_in_asia[i]]
+    out.sort(key=lambda t: -t[1])
+    return out

@@ -1,0 +1,807 @@
+// quokka_amd — MI355X (gfx950, CDNA4) kernels for the Quokka columnar hot
+// path, behind the C ABI of include/quokka_amd.h.
+//
+// Design: every kernel here is HBM-bound integer/byte/f64
+// streaming work (scan+filter, hash build/probe, group-by, partition) — the
+// operations the reference delegates to polars/duckdb.
+// No MFMA (no dense contraction on this path). Rules applied from the CDNA4
+// guide: 64-lane waves, 256-thread blocks, 16B/lane coalesced loads where
+// layout permits, grid-stride with ~2048 blocks, per-wave shuffle reduction
+// then per-block LDS reduction before any global atomic.
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <stdio.h>
+#include <string.h>
+#include "../../include/quokka_amd.h"
+
+#define WAVE 64
+
+static __host__ __device__ inline uint64_t qk_min_u64(uint64_t a, uint64_t b) {
+  return a < b ? a : b;
+}
+#define BLOCK 256
+#define MAX_BLOCKS 2048
+
+static __thread char g_err[512] = "";
+static __thread int g_err_set = 0;
+
+static int qk_fail(const char *where, hipError_t e) {
+  snprintf(g_err, sizeof(g_err), "%s: %s", where, hipGetErrorString(e));
+  g_err_set = 1;
+  return (int)(e == hipSuccess ? hipErrorUnknown : e);
+}
+#define QK_TRY(where, expr)                                                    \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) return qk_fail(where, _e);                           \
+  } while (0)
+
+extern "C" const char *qk_last_error(void) { return g_err_set ? g_err : ""; }
+extern "C" const char *qk_build_arch(void) { return "gfx950"; }
+
+extern "C" int qk_init(int device) {
+  QK_TRY("qk_init.setDevice", hipSetDevice(device));
+  QK_TRY("qk_init.warm", hipFree(nullptr));
+  return 0;
+}
+extern "C" int qk_device_count(int *out) {
+  QK_TRY("qk_device_count", hipGetDeviceCount(out));
+  return 0;
+}
+
+// ---- memory -----------------------------------------------------------
+extern "C" int qk_dmalloc(uint64_t nbytes, void **dptr) {
+  QK_TRY("qk_dmalloc", hipMalloc(dptr, nbytes ? nbytes : 1));
+  return 0;
+}
+extern "C" int qk_dfree(void *dptr) {
+  QK_TRY("qk_dfree", hipFree(dptr));
+  return 0;
+}
+extern "C" int qk_h2d(void *dst, const void *src, uint64_t n) {
+  QK_TRY("qk_h2d", hipMemcpy(dst, src, n, hipMemcpyHostToDevice));
+  return 0;
+}
+extern "C" int qk_d2h(void *dst, const void *src, uint64_t n) {
+  QK_TRY("qk_d2h", hipMemcpy(dst, src, n, hipMemcpyDeviceToHost));
+  return 0;
+}
+extern "C" int qk_dmemset(void *dst, int value, uint64_t n) {
+  QK_TRY("qk_dmemset", hipMemset(dst, value, n));
+  return 0;
+}
+
+__global__ void k_fill_i64(int64_t *dst, int64_t v, uint64_t n) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dst[i] = v;
+}
+extern "C" int qk_fill_i64(void *stream, int64_t *dst, int64_t v, uint64_t n) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_fill_i64, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, dst, v, n);
+  QK_TRY("qk_fill_i64", hipGetLastError());
+  return 0;
+}
+
+// ---- streams & timing -------------------------------------------------
+extern "C" int qk_stream_create(void **stream) {
+  QK_TRY("qk_stream_create", hipStreamCreate((hipStream_t *)stream));
+  return 0;
+}
+extern "C" int qk_stream_destroy(void *stream) {
+  QK_TRY("qk_stream_destroy", hipStreamDestroy((hipStream_t)stream));
+  return 0;
+}
+extern "C" int qk_stream_sync(void *stream) {
+  QK_TRY("qk_stream_sync", hipStreamSynchronize((hipStream_t)stream));
+  return 0;
+}
+
+struct QkTimer {
+  hipEvent_t start, stop;
+};
+extern "C" int qk_timer_create(void **timer) {
+  QkTimer *t = new QkTimer();
+  QK_TRY("qk_timer_create.start", hipEventCreate(&t->start));
+  QK_TRY("qk_timer_create.stop", hipEventCreate(&t->stop));
+  *timer = t;
+  return 0;
+}
+extern "C" int qk_timer_destroy(void *timer) {
+  QkTimer *t = (QkTimer *)timer;
+  hipEventDestroy(t->start);
+  hipEventDestroy(t->stop);
+  delete t;
+  return 0;
+}
+extern "C" int qk_timer_start(void *timer, void *stream) {
+  QK_TRY("qk_timer_start",
+         hipEventRecord(((QkTimer *)timer)->start, (hipStream_t)stream));
+  return 0;
+}
+extern "C" int qk_timer_stop(void *timer, void *stream) {
+  QK_TRY("qk_timer_stop",
+         hipEventRecord(((QkTimer *)timer)->stop, (hipStream_t)stream));
+  return 0;
+}
+extern "C" int qk_timer_elapsed_ms(void *timer, float *out_ms) {
+  QkTimer *t = (QkTimer *)timer;
+  QK_TRY("qk_timer_elapsed.sync", hipEventSynchronize(t->stop));
+  QK_TRY("qk_timer_elapsed", hipEventElapsedTime(out_ms, t->start, t->stop));
+  return 0;
+}
+
+// ---- RNG (counter-based; device twin of oracle/executors.py splitmix64) --
+__device__ __host__ inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ULL;
+  x ^= x >> 30;
+  x *= 0xBF58476D1CE4E5B9ULL;
+  x ^= x >> 27;
+  x *= 0x94D049BB133111EBULL;
+  x ^= x >> 31;
+  return x;
+}
+
+// TPC-H date constants (days since 1970-01-01; = oracle/tpch_gen.py values,
+// pinned by tests/test_abi.py against the Python side)
+#define QK_ORDERDATE_LO 8035   // 1992-01-01
+#define QK_ORDERDATE_HI 10440  // 1998-08-02
+#define QK_RECEIPT_CUTOFF 9298 // 1995-06-17
+
+__global__ void k_gen_lineitem(uint64_t n, uint64_t row_offset, uint64_t seed,
+                               int64_t n_parts, int64_t n_suppliers,
+                               int64_t n_orders, int64_t *l_orderkey,
+                               int64_t *l_suppkey, double *l_quantity,
+                               double *l_extendedprice, double *l_discount,
+                               double *l_tax, uint8_t *l_returnflag,
+                               uint8_t *l_linestatus, int32_t *l_shipdate) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t row = row_offset + i;
+    uint64_t base = splitmix64(seed ^ 0x51A2B3C4D5E6F708ULL) ^
+                    (row * 0x9E3779B97F4A7C15ULL);
+    uint64_t h0 = splitmix64(base + 0), h1 = splitmix64(base + 1),
+             h2 = splitmix64(base + 2), h3 = splitmix64(base + 3),
+             h4 = splitmix64(base + 4), h5 = splitmix64(base + 5),
+             h6 = splitmix64(base + 6), h7 = splitmix64(base + 7),
+             h8 = splitmix64(base + 8);
+    int32_t odate =
+        QK_ORDERDATE_LO + (int32_t)(h0 % (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
+    int32_t ship = odate + 1 + (int32_t)(h1 % 121);
+    int32_t receipt = ship + 1 + (int32_t)(h2 % 30);
+    if (l_shipdate) l_shipdate[i] = ship;
+    if (l_returnflag)
+      l_returnflag[i] =
+          (receipt <= QK_RECEIPT_CUTOFF) ? ((h3 & 1) ? (uint8_t)0 : (uint8_t)2)
+                                         : (uint8_t)1;
+    if (l_linestatus) l_linestatus[i] = ship > QK_RECEIPT_CUTOFF ? 1 : 0;
+    double qty = (double)(1 + (int32_t)(h4 % 50));
+    if (l_quantity) l_quantity[i] = qty;
+    if (l_extendedprice) {
+      int64_t pk = 1 + (int64_t)(h5 % (uint64_t)n_parts);
+      int64_t cents = 90000 + (pk / 10) % 20001 + 100 * (pk % 1000);
+      l_extendedprice[i] = qty * ((double)cents / 100.0);
+    }
+    if (l_discount) l_discount[i] = (double)(h6 % 11) / 100.0;
+    if (l_tax) l_tax[i] = (double)(h7 % 9) / 100.0;
+    if (l_suppkey) l_suppkey[i] = 1 + (int64_t)(h8 % (uint64_t)n_suppliers);
+    if (l_orderkey) l_orderkey[i] = 1 + (int64_t)(row / 4 % (uint64_t)n_orders);
+  }
+}
+extern "C" int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset,
+                               uint64_t seed, int64_t n_parts,
+                               int64_t n_suppliers, int64_t n_orders,
+                               int64_t *l_orderkey, int64_t *l_suppkey,
+                               double *l_quantity, double *l_extendedprice,
+                               double *l_discount, double *l_tax,
+                               uint8_t *l_returnflag, uint8_t *l_linestatus,
+                               int32_t *l_shipdate) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gen_lineitem, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, row_offset, seed, n_parts,
+                     n_suppliers, n_orders, l_orderkey, l_suppkey, l_quantity,
+                     l_extendedprice, l_discount, l_tax, l_returnflag,
+                     l_linestatus, l_shipdate);
+  QK_TRY("qk_gen_lineitem", hipGetLastError());
+  return 0;
+}
+
+// ---- Q1 fused filter + group-by partial aggregate ---------------------
+// 6 groups x 6 accumulators kept in REGISTERS per thread (statically
+// indexed; runtime-indexed per-thread arrays spill to scratch on hipcc —
+// guide §5.4 rule 20), divergent-but-cheap if-chain per row, then wave
+// shuffle-reduce, block LDS reduce, one global f64 atomicAdd per
+// accumulator per block (guide G12).
+
+#define Q1_ACC_DECL(g)                                                         \
+  double a##g##0 = 0, a##g##1 = 0, a##g##2 = 0, a##g##3 = 0, a##g##4 = 0,      \
+         a##g##5 = 0;
+#define Q1_ACC_ADD(g)                                                          \
+  if (gid == g) {                                                              \
+    a##g##0 += qty;                                                            \
+    a##g##1 += price;                                                          \
+    a##g##2 += disc_price;                                                     \
+    a##g##3 += charge;                                                         \
+    a##g##4 += disc;                                                           \
+    a##g##5 += 1.0;                                                            \
+  }
+#define Q1_ACC_ROW()                                                           \
+  do {                                                                         \
+    double disc_price = price * (1.0 - disc);                                  \
+    double charge = disc_price * (1.0 + tax);                                  \
+    Q1_ACC_ADD(0) else Q1_ACC_ADD(1) else Q1_ACC_ADD(2) else Q1_ACC_ADD(       \
+        3) else Q1_ACC_ADD(4) else Q1_ACC_ADD(5)                               \
+  } while (0)
+
+__device__ inline double wave_reduce(double v) {
+  for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_down(v, off);
+  return v;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_q1_agg(
+    uint64_t n, const int32_t *__restrict__ shipdate,
+    const double *__restrict__ quantity, const double *__restrict__ extprice,
+    const double *__restrict__ discount, const double *__restrict__ tax_,
+    const uint8_t *__restrict__ rflag, const uint8_t *__restrict__ lstat,
+    int32_t cutoff, double *__restrict__ out) {
+  Q1_ACC_DECL(0) Q1_ACC_DECL(1) Q1_ACC_DECL(2)
+  Q1_ACC_DECL(3) Q1_ACC_DECL(4) Q1_ACC_DECL(5)
+
+  uint64_t npairs = n / 2;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npairs;
+       p += stride) {
+    uint64_t r = 2 * p;
+    // 16B/lane coalesced vector loads for the f64 columns
+    double2 q2 = *reinterpret_cast<const double2 *>(quantity + r);
+    double2 p2 = *reinterpret_cast<const double2 *>(extprice + r);
+    double2 d2 = *reinterpret_cast<const double2 *>(discount + r);
+    double2 t2 = *reinterpret_cast<const double2 *>(tax_ + r);
+    int2 s2 = *reinterpret_cast<const int2 *>(shipdate + r);
+    uint8_t f0 = rflag[r], f1 = rflag[r + 1];
+    uint8_t l0 = lstat[r], l1 = lstat[r + 1];
+    if (s2.x <= cutoff) {
+      int gid = (int)f0 * 2 + (int)l0;
+      double qty = q2.x, price = p2.x, disc = d2.x, tax = t2.x;
+      Q1_ACC_ROW();
+    }
+    if (s2.y <= cutoff) {
+      int gid = (int)f1 * 2 + (int)l1;
+      double qty = q2.y, price = p2.y, disc = d2.y, tax = t2.y;
+      Q1_ACC_ROW();
+    }
+  }
+  // odd tail row handled by global thread 0
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    uint64_t r = n - 1;
+    if (shipdate[r] <= cutoff) {
+      int gid = (int)rflag[r] * 2 + (int)lstat[r];
+      double qty = quantity[r], price = extprice[r], disc = discount[r],
+             tax = tax_[r];
+      Q1_ACC_ROW();
+    }
+  }
+
+  // wave shuffle-reduce each accumulator, lane 0 stages to LDS
+  __shared__ double lds[BLOCK / WAVE][36];
+  int lane = threadIdx.x & (WAVE - 1);
+  int wid = threadIdx.x / WAVE;
+#define Q1_RED(g, j, var)                                                      \
+  {                                                                            \
+    double s = wave_reduce(var);                                               \
+    if (lane == 0) lds[wid][g * 6 + j] = s;                                    \
+  }
+#define Q1_RED_G(g)                                                            \
+  Q1_RED(g, 0, a##g##0) Q1_RED(g, 1, a##g##1) Q1_RED(g, 2, a##g##2)            \
+  Q1_RED(g, 3, a##g##3) Q1_RED(g, 4, a##g##4) Q1_RED(g, 5, a##g##5)
+  Q1_RED_G(0) Q1_RED_G(1) Q1_RED_G(2) Q1_RED_G(3) Q1_RED_G(4) Q1_RED_G(5)
+  __syncthreads();
+  // threads 0..35 each combine the per-wave partials for one accumulator
+  if (threadIdx.x < 36) {
+    double s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w][threadIdx.x];
+    int g = threadIdx.x / 6, j = threadIdx.x % 6;
+    if (s != 0.0) atomicAdd(&out[g * 8 + j], s);
+  }
+}
+
+extern "C" int qk_q1_agg(void *stream, uint64_t n, const int32_t *shipdate,
+                         const double *quantity, const double *extprice,
+                         const double *discount, const double *tax,
+                         const uint8_t *rflag, const uint8_t *lstat,
+                         int32_t cutoff, double *out) {
+  uint64_t npairs = n / 2;
+  uint32_t blocks =
+      (uint32_t)qk_min_u64(MAX_BLOCKS, (npairs + BLOCK - 1) / BLOCK);
+  if (!blocks) blocks = 1;
+  hipLaunchKernelGGL(k_q1_agg, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, shipdate, quantity, extprice,
+                     discount, tax, rflag, lstat, cutoff, out);
+  QK_TRY("qk_q1_agg", hipGetLastError());
+  return 0;
+}
+
+// ---- Q6 fused filter + sum -------------------------------------------
+__global__ void __launch_bounds__(BLOCK) k_q6_agg(
+    uint64_t n, const int32_t *__restrict__ shipdate,
+    const double *__restrict__ quantity, const double *__restrict__ extprice,
+    const double *__restrict__ discount, int32_t date_lo, int32_t date_hi,
+    double disc_lo, double disc_hi, double qty_hi, double *__restrict__ out) {
+  double rev = 0, cnt = 0;
+  uint64_t npairs = n / 2;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npairs;
+       p += stride) {
+    uint64_t r = 2 * p;
+    double2 q2 = *reinterpret_cast<const double2 *>(quantity + r);
+    double2 p2 = *reinterpret_cast<const double2 *>(extprice + r);
+    double2 d2 = *reinterpret_cast<const double2 *>(discount + r);
+    int2 s2 = *reinterpret_cast<const int2 *>(shipdate + r);
+    if (s2.x >= date_lo && s2.x < date_hi && d2.x >= disc_lo &&
+        d2.x <= disc_hi && q2.x < qty_hi) {
+      rev += p2.x * d2.x;
+      cnt += 1.0;
+    }
+    if (s2.y >= date_lo && s2.y < date_hi && d2.y >= disc_lo &&
+        d2.y <= disc_hi && q2.y < qty_hi) {
+      rev += p2.y * d2.y;
+      cnt += 1.0;
+    }
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    uint64_t r = n - 1;
+    if (shipdate[r] >= date_lo && shipdate[r] < date_hi &&
+        discount[r] >= disc_lo && discount[r] <= disc_hi &&
+        quantity[r] < qty_hi) {
+      rev += extprice[r] * discount[r];
+      cnt += 1.0;
+    }
+  }
+  __shared__ double lds[2][BLOCK / WAVE];
+  int lane = threadIdx.x & (WAVE - 1);
+  int wid = threadIdx.x / WAVE;
+  double s0 = wave_reduce(rev), s1 = wave_reduce(cnt);
+  if (lane == 0) {
+    lds[0][wid] = s0;
+    lds[1][wid] = s1;
+  }
+  __syncthreads();
+  if (threadIdx.x < 2) {
+    double s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[threadIdx.x][w];
+    if (s != 0.0) atomicAdd(&out[threadIdx.x], s);
+  }
+}
+extern "C" int qk_q6_agg(void *stream, uint64_t n, const int32_t *shipdate,
+                         const double *quantity, const double *extprice,
+                         const double *discount, int32_t date_lo,
+                         int32_t date_hi, double disc_lo, double disc_hi,
+                         double qty_hi, double *out) {
+  uint64_t npairs = n / 2;
+  uint32_t blocks =
+      (uint32_t)qk_min_u64(MAX_BLOCKS, (npairs + BLOCK - 1) / BLOCK);
+  if (!blocks) blocks = 1;
+  hipLaunchKernelGGL(k_q6_agg, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, shipdate, quantity, extprice,
+                     discount, date_lo, date_hi, disc_lo, disc_hi, qty_hi, out);
+  QK_TRY("qk_q6_agg", hipGetLastError());
+  return 0;
+}
+
+// ---- generic filter (ordered compaction) ------------------------------
+// Three launches: per-block count over contiguous chunks -> single-block
+// exclusive scan of block counts -> re-evaluate + rank + scatter. Row order
+// is preserved (polars filter semantics, core.py:170).
+
+template <typename T>
+__device__ inline bool cmp_op(T v, int op, T ref) {
+  switch (op) {
+  case 0: return v < ref;
+  case 1: return v <= ref;
+  case 2: return v > ref;
+  case 3: return v >= ref;
+  case 4: return v == ref;
+  default: return v != ref;
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLOCK) k_filter_count(
+    uint64_t n, const T *__restrict__ col, int op, T ref, uint64_t chunk,
+    uint64_t *__restrict__ block_counts) {
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  uint32_t cnt = 0;
+  for (uint64_t r = lo + threadIdx.x; r < hi; r += BLOCK)
+    cnt += cmp_op(col[r], op, ref) ? 1u : 0u;
+  // wave reduce then block reduce
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w];
+    block_counts[blockIdx.x] = s;
+  }
+}
+
+__global__ void k_scan_blocks(uint64_t nblocks, uint64_t *__restrict__ counts,
+                              uint64_t *__restrict__ total) {
+  // single thread serial scan; nblocks <= 4096 so this is microseconds
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    uint64_t acc = 0;
+    for (uint64_t b = 0; b < nblocks; b++) {
+      uint64_t c = counts[b];
+      counts[b] = acc;
+      acc += c;
+    }
+    *total = acc;
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLOCK) k_filter_scatter(
+    uint64_t n, const T *__restrict__ col, int op, T ref, uint64_t chunk,
+    const uint64_t *__restrict__ block_offsets, uint32_t *__restrict__ out_idx) {
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  __shared__ uint64_t base;
+  __shared__ uint32_t wave_tot[BLOCK / WAVE];
+  if (threadIdx.x == 0) base = block_offsets[blockIdx.x];
+  __syncthreads();
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  for (uint64_t r0 = lo; r0 < hi; r0 += BLOCK) {
+    uint64_t r = r0 + threadIdx.x;
+    bool pass = r < hi && cmp_op(col[r], op, ref);
+    uint64_t mask = __ballot(pass);
+    uint32_t rank = __popcll(mask & ((1ULL << lane) - 1));
+    uint32_t wtot = __popcll(mask);
+    if (lane == 0) wave_tot[wid] = wtot;
+    __syncthreads();
+    uint32_t wbase = 0;
+    for (int w = 0; w < wid; w++) wbase += wave_tot[w];
+    if (pass) out_idx[base + wbase + rank] = (uint32_t)r;
+    uint32_t btot = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) btot += wave_tot[w];
+    __syncthreads();
+    if (threadIdx.x == 0) base += btot;
+    __syncthreads();
+  }
+}
+
+template <typename T>
+static int qk_filter_impl(const char *who, void *stream, uint64_t n,
+                          const T *col, int op, T value, uint32_t *out_idx,
+                          uint64_t *out_count_dev) {
+  if (!n) return 0;
+  uint64_t rows_per_block = (n + MAX_BLOCKS - 1) / MAX_BLOCKS;
+  rows_per_block = ((rows_per_block + BLOCK - 1) / BLOCK) * BLOCK;
+  uint32_t blocks = (uint32_t)((n + rows_per_block - 1) / rows_per_block);
+  static __thread uint64_t *scratch = nullptr;  // per-thread block_counts buf
+  static __thread uint64_t scratch_cap = 0;
+  if (scratch_cap < blocks) {
+    if (scratch) hipFree(scratch);
+    QK_TRY(who, hipMalloc(&scratch, (MAX_BLOCKS + 1) * sizeof(uint64_t)));
+    scratch_cap = MAX_BLOCKS + 1;
+  }
+  hipLaunchKernelGGL(k_filter_count<T>, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, col, op, value, rows_per_block,
+                     scratch);
+  hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0, (hipStream_t)stream,
+                     (uint64_t)blocks, scratch, out_count_dev);
+  hipLaunchKernelGGL(k_filter_scatter<T>, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, col, op, value, rows_per_block,
+                     scratch, out_idx);
+  QK_TRY(who, hipGetLastError());
+  return 0;
+}
+extern "C" int qk_filter_i32(void *stream, uint64_t n, const int32_t *col,
+                             int op, int32_t value, uint32_t *out_idx,
+                             uint64_t *out_count_dev) {
+  return qk_filter_impl("qk_filter_i32", stream, n, col, op, value, out_idx,
+                        out_count_dev);
+}
+extern "C" int qk_filter_u8(void *stream, uint64_t n, const uint8_t *col,
+                            int op, uint8_t value, uint32_t *out_idx,
+                            uint64_t *out_count_dev) {
+  return qk_filter_impl("qk_filter_u8", stream, n, col, op, value, out_idx,
+                        out_count_dev);
+}
+
+// ---- elementwise revenue ----------------------------------------------
+__global__ void k_mul_1md(uint64_t n, const double *__restrict__ a,
+                          const double *__restrict__ b,
+                          double *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = a[i] * (1.0 - b[i]);
+}
+extern "C" int qk_mul_1md(void *stream, uint64_t n, const double *a,
+                          const double *b, double *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_mul_1md, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, a, b, out);
+  QK_TRY("qk_mul_1md", hipGetLastError());
+  return 0;
+}
+
+// ---- gathers ----------------------------------------------------------
+template <typename T>
+__global__ void k_gather(uint64_t n, const uint32_t *__restrict__ idx,
+                         const T *__restrict__ src, T *__restrict__ dst) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dst[i] = src[idx[i]];
+}
+template <typename T>
+static int qk_gather_impl(const char *who, void *stream, uint64_t n,
+                          const uint32_t *idx, const T *src, T *dst) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gather<T>, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, idx, src, dst);
+  QK_TRY(who, hipGetLastError());
+  return 0;
+}
+extern "C" int qk_gather_i64(void *s, uint64_t n, const uint32_t *i,
+                             const int64_t *src, int64_t *dst) {
+  return qk_gather_impl("qk_gather_i64", s, n, i, src, dst);
+}
+extern "C" int qk_gather_f64(void *s, uint64_t n, const uint32_t *i,
+                             const double *src, double *dst) {
+  return qk_gather_impl("qk_gather_f64", s, n, i, src, dst);
+}
+extern "C" int qk_gather_i32(void *s, uint64_t n, const uint32_t *i,
+                             const int32_t *src, int32_t *dst) {
+  return qk_gather_impl("qk_gather_i32", s, n, i, src, dst);
+}
+extern "C" int qk_gather_u8(void *s, uint64_t n, const uint32_t *i,
+                            const uint8_t *src, uint8_t *dst) {
+  return qk_gather_impl("qk_gather_u8", s, n, i, src, dst);
+}
+
+// ---- hash join --------------------------------------------------------
+// Open addressing, linear probing, splitmix64(key) & (cap-1). Duplicate
+// build keys chain via chain_next (head swap is a single atomicExch, so
+// build order within a key is unspecified — as is polars', sql_executors
+// :371; parity compares multisets).
+
+__device__ inline uint64_t slot_of(int64_t key, uint64_t cap) {
+  return splitmix64((uint64_t)key) & (cap - 1);
+}
+
+__global__ void __launch_bounds__(BLOCK) k_join_build2(
+    uint64_t n, const int64_t *__restrict__ keys, uint32_t row_offset,
+    int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
+    int32_t *__restrict__ chain_next, uint64_t cap) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    int32_t me = (int32_t)(row_offset + i);
+    int32_t old = atomicExch(&slot_head[s], me);
+    chain_next[me] = old;
+  }
+}
+
+extern "C" int qk_join_build(void *stream, uint64_t n, const int64_t *keys,
+                             uint32_t row_offset, int64_t *slot_keys,
+                             int32_t *slot_head, int32_t *chain_next,
+                             uint64_t cap) {
+  if (!n) return 0;
+  if (cap & (cap - 1)) return qk_fail("qk_join_build.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_join_build2, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, row_offset, slot_keys,
+                     slot_head, chain_next, cap);
+  QK_TRY("qk_join_build", hipGetLastError());
+  return 0;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_join_probe(
+    uint64_t n, const int64_t *__restrict__ keys,
+    const int64_t *__restrict__ slot_keys, const int32_t *__restrict__ slot_head,
+    const int32_t *__restrict__ chain_next, uint64_t cap, int mode,
+    uint32_t *__restrict__ out_probe, uint32_t *__restrict__ out_build,
+    uint64_t out_cap, uint64_t *__restrict__ cursor) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    uint64_t s = slot_of(key, cap);
+    int32_t head = -1;
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) {
+        head = slot_head[s];
+        break;
+      }
+      if (cur == QK_JOIN_EMPTY) break;
+      s = (s + 1) & (cap - 1);
+    }
+    if (mode == 1) {  // semi
+      if (head >= 0) {
+        uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
+        if (pos < out_cap) out_probe[pos] = (uint32_t)i;
+      }
+      continue;
+    }
+    if (mode == 2) {  // anti
+      if (head < 0) {
+        uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
+        if (pos < out_cap) out_probe[pos] = (uint32_t)i;
+      }
+      continue;
+    }
+    if (head < 0) continue;
+    // inner: count chain, claim a contiguous range, emit
+    uint32_t cnt = 0;
+    for (int32_t b = head; b >= 0; b = chain_next[b]) cnt++;
+    uint64_t pos = atomicAdd((unsigned long long *)cursor, (unsigned long long)cnt);
+    for (int32_t b = head; b >= 0; b = chain_next[b]) {
+      if (pos < out_cap) {
+        out_probe[pos] = (uint32_t)i;
+        out_build[pos] = (uint32_t)b;
+      }
+      pos++;
+    }
+  }
+}
+extern "C" int qk_join_probe(void *stream, uint64_t n, const int64_t *keys,
+                             const int64_t *slot_keys, const int32_t *slot_head,
+                             const int32_t *chain_next, uint64_t cap, int mode,
+                             uint32_t *out_probe, uint32_t *out_build,
+                             uint64_t out_cap, uint64_t *cursor) {
+  if (!n) return 0;
+  if (cap & (cap - 1)) return qk_fail("qk_join_probe.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_join_probe, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, slot_keys, slot_head,
+                     chain_next, cap, mode, out_probe, out_build, out_cap,
+                     cursor);
+  QK_TRY("qk_join_probe", hipGetLastError());
+  return 0;
+}
+
+// ---- group-by i64 -> f64 sums ----------------------------------------
+__global__ void __launch_bounds__(BLOCK) k_groupby_sum(
+    uint64_t n, const int64_t *__restrict__ keys, const double *const *vals,
+    int nvals, int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
+    uint64_t cap) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    for (int c = 0; c < nvals; c++)
+      atomicAdd(&slot_sums[(uint64_t)c * cap + s], vals[c][i]);
+  }
+}
+extern "C" int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
+                                  const double *const *vals_dev, int nvals,
+                                  int64_t *slot_keys, double *slot_sums,
+                                  uint64_t cap) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_groupby_i64_sum.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_groupby_sum, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, vals_dev, nvals, slot_keys,
+                     slot_sums, cap);
+  QK_TRY("qk_groupby_i64_sum", hipGetLastError());
+  return 0;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_groupby_extract(
+    const int64_t *__restrict__ slot_keys, const double *__restrict__ slot_sums,
+    int nvals, uint64_t cap, int64_t *__restrict__ out_keys,
+    double *__restrict__ out_sums, uint64_t out_cap,
+    uint64_t *__restrict__ cursor) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += stride) {
+    int64_t key = slot_keys[s];
+    if (key == QK_JOIN_EMPTY) continue;
+    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
+    if (pos < out_cap) {
+      out_keys[pos] = key;
+      for (int c = 0; c < nvals; c++)
+        out_sums[(uint64_t)c * out_cap + pos] = slot_sums[(uint64_t)c * cap + s];
+    }
+  }
+}
+extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
+                                  const double *slot_sums, int nvals,
+                                  uint64_t cap, int64_t *out_keys,
+                                  double *out_sums, uint64_t out_cap,
+                                  uint64_t *cursor) {
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_groupby_extract, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
+                     out_keys, out_sums, out_cap, cursor);
+  QK_TRY("qk_groupby_extract", hipGetLastError());
+  return 0;
+}
+
+// ---- hash partition (int key: part = key % nparts, quokka_runtime:222) --
+__global__ void __launch_bounds__(BLOCK) k_partition_hist(
+    uint64_t n, const int64_t *__restrict__ keys, uint32_t nparts,
+    uint64_t *__restrict__ hist) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint32_t *lh = (uint32_t *)smem;
+  for (uint32_t p = threadIdx.x; p < nparts; p += BLOCK) lh[p] = 0;
+  __syncthreads();
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    atomicAdd(&lh[(uint32_t)(keys[i] % nparts)], 1u);
+  __syncthreads();
+  for (uint32_t p = threadIdx.x; p < nparts; p += BLOCK)
+    if (lh[p]) atomicAdd((unsigned long long *)&hist[p], (unsigned long long)lh[p]);
+}
+extern "C" int qk_partition_hist(void *stream, uint64_t n, const int64_t *keys,
+                                 uint32_t nparts, uint64_t *hist) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_partition_hist, dim3(blocks), dim3(BLOCK),
+                     nparts * sizeof(uint32_t), (hipStream_t)stream, n, keys,
+                     nparts, hist);
+  QK_TRY("qk_partition_hist", hipGetLastError());
+  return 0;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_partition_scatter(
+    uint64_t n, const int64_t *__restrict__ keys, uint32_t nparts,
+    uint64_t *__restrict__ cursors, uint32_t *__restrict__ out_idx) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint32_t p = (uint32_t)(keys[i] % nparts);
+    uint64_t pos = atomicAdd((unsigned long long *)&cursors[p], 1ULL);
+    out_idx[pos] = (uint32_t)i;
+  }
+}
+extern "C" int qk_partition_scatter(void *stream, uint64_t n,
+                                    const int64_t *keys, uint32_t nparts,
+                                    uint64_t *cursors, uint32_t *out_idx) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_partition_scatter, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, nparts, cursors, out_idx);
+  QK_TRY("qk_partition_scatter", hipGetLastError());
+  return 0;
+}

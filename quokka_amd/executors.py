@@ -1,0 +1,384 @@
+"""Drop-in Executor subclasses and partition function for pyquokka.
+
+These mirror the reference's plugin boundary exactly
+(pyquokka/executors/base_executor.py:26-32, call sites core.py:632/:667,
+task.py:131-134): stateful objects whose `execute(batches, stream_id,
+channel_id)` consumes `list[pyarrow.Table]` and whose `done(channel_id)`
+flushes. Stream mapping for joins: stream 1 = build, stream 0 = probe
+(logical.py:459-471). Objects must be picklable at registration time
+(quokka_runtime.py:325) — all GPU state is created lazily on first
+execute, never in __init__.
+
+A pyquokka deployment drops these in at lowering time in place of
+BuildProbeJoinExecutor / SQLAggExecutor."""
+import re
+
+import numpy as np
+
+
+class Executor:
+    """= pyquokka.executors.base_executor.Executor (:26-32)."""
+
+    def __init__(self):
+        raise NotImplementedError
+
+    def execute(self, batches, stream_id, executor_id):
+        raise NotImplementedError
+
+    def done(self, executor_id):
+        raise NotImplementedError
+
+
+def _lazy_gpu():
+    """Import the HIP shim at first use (fails loudly without the .so /
+    a GPU; keeps executors picklable before first execute)."""
+    from . import ops, shim, staging
+    return ops, shim, staging
+
+
+def _to_table(d):
+    import pyarrow as pa
+    return pa.table({k: v for k, v in d.items()})
+
+
+class GPUBuildProbeJoinExecutor(Executor):
+    """GPU replacement for BuildProbeJoinExecutor (sql_executors.py:325-377).
+
+    Same constructor signature and argument meaning; same stream contract
+    (1 = build, vstacked into state; 0 = probe, one result per probe batch);
+    same `how` set {inner, left, semi, anti}; key_to_keep renames the key
+    column like :372-373. Differences documented in DESIGN.md: the build
+    state is a device hash table (the reference's vstack+sort+polars-join is
+    an implementation detail, :358/:369); result row ORDER within a batch is
+    unspecified (as is polars'). i64 keys supported."""
+
+    def __init__(self, on=None, left_on=None, right_on=None, how="inner",
+                 key_to_keep="left"):
+        if on is not None:
+            assert left_on is None and right_on is None
+            self.left_on = on
+            self.right_on = on
+        else:
+            assert left_on is not None and right_on is not None
+            self.left_on = left_on
+            self.right_on = right_on
+        assert how in {"inner", "left", "semi", "anti"}
+        self.how = how
+        self.key_to_keep = key_to_keep
+        self.phase = "build"
+        # lazy GPU state
+        self._table = None
+        self._build_cols = None     # dict name -> list of numpy arrays
+        self._build_names = None
+        self._stream = None
+
+    def __getstate__(self):
+        assert self._table is None, "executor must be pickled before first execute"
+        d = dict(self.__dict__)
+        return d
+
+    def _ensure_stream(self):
+        ops, shim, staging = _lazy_gpu()
+        if self._stream is None:
+            self._stream = ops and None  # stream created lazily below
+        return ops, shim, staging
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        ops, shim, staging = _lazy_gpu()
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if len(batches) == 0:
+            return
+        batch = pa.concat_tables(batches)
+
+        if stream_id == 1:                       # build (state vstack)
+            assert self.phase == "build"
+            if self._build_cols is None:
+                self._build_names = [c for c in batch.column_names]
+                self._build_cols = {c: [] for c in self._build_names}
+            for c in self._build_names:
+                self._build_cols[c].append(
+                    staging.column_to_numpy(batch.column(c)))
+            return
+
+        # probe
+        if self._table is None and self._build_cols is None:
+            if self.how == "anti":
+                return batch
+            return
+        if self.phase == "build":
+            self._finish_build()
+        self.phase = "probe"
+        return self._probe(batch)
+
+    def _finish_build(self):
+        ops, shim, staging = _lazy_gpu()
+        self._host_build = {c: np.concatenate(v)
+                            for c, v in self._build_cols.items()}
+        keys = self._host_build[self.right_on]
+        if keys.dtype != np.int64:
+            raise TypeError("GPU join requires i64 keys, got %s" % keys.dtype)
+        n = len(keys)
+        self._table = ops.JoinTable(max(16, n))
+        if n:
+            kcol = shim.DevColumn.from_numpy(keys)
+            self._table.build(kcol)
+            kcol.free()
+        self._build_payload_dev = {}
+        for c in self._build_names:
+            if c == self.right_on:
+                continue
+            self._build_payload_dev[c] = shim.DevColumn.from_numpy(
+                self._host_build[c])
+        self._build_cols = None
+
+    def _probe(self, batch):
+        ops, shim, staging = _lazy_gpu()
+        probe_keys = staging.column_to_numpy(batch.column(self.left_on))
+        if probe_keys.dtype != np.int64:
+            raise TypeError("GPU join requires i64 keys")
+        kcol = shim.DevColumn.from_numpy(probe_keys)
+        mode = {"inner": 0, "left": 0, "semi": 1, "anti": 2}[self.how]
+        pidx, bidx, nm = self._table.probe(kcol, mode=mode)
+
+        out = {}
+        if self.how in ("semi", "anti"):
+            sel = pidx.to_numpy(nm)
+            for c in batch.column_names:
+                out[c] = staging.column_to_numpy(batch.column(c))[sel]
+        else:
+            sel = pidx.to_numpy(nm)
+            bsel = bidx.to_numpy(nm)
+            for c in batch.column_names:
+                out[c] = staging.column_to_numpy(batch.column(c))[sel]
+            for c, dev in self._build_payload_dev.items():
+                host = self._host_build[c]
+                out[c] = host[bsel]
+            if self.how == "left":
+                # unmatched probe rows appended with null build payload
+                matched = np.zeros(len(probe_keys), dtype=bool)
+                matched[sel] = True
+                un = np.nonzero(~matched)[0]
+                if len(un):
+                    import pyarrow as pa
+                    for c in batch.column_names:
+                        out[c] = np.concatenate(
+                            [out[c], staging.column_to_numpy(batch.column(c))[un]])
+                    tbl = _to_table({k: v for k, v in out.items()
+                                     if k in batch.column_names})
+                    pay = {}
+                    for c in self._build_payload_dev:
+                        pay[c] = pa.chunked_array([
+                            pa.array(out[c]),
+                            pa.nulls(len(un), pa.from_numpy_dtype(out[c].dtype)),
+                        ])
+                    for c, v in pay.items():
+                        tbl = tbl.append_column(c, v)
+                    pidx.free()
+                    if bidx:
+                        bidx.free()
+                    kcol.free()
+                    return tbl
+        if self.key_to_keep == "right" and self.how in ("inner", "left"):
+            out[self.right_on] = out.pop(self.left_on)
+        pidx.free()
+        if bidx:
+            bidx.free()
+        kcol.free()
+        return _to_table(out)
+
+    def done(self, executor_id):
+        pass
+
+
+_SUM_RE = re.compile(r"sum\s*\(\s*([A-Za-z_][A-Za-z0-9_]*)\s*\)", re.I)
+
+
+class GPUAggExecutor(Executor):
+    """GPU replacement for SQLAggExecutor (sql_executors.py:556-599).
+
+    Same constructor contract: groupby_keys (list), orderby_keys (list of
+    (key, 'asc'|'desc')), sql_statement = the FINAL aggregate clause the
+    two-phase rewrite produces (sql_utils.py:379-413) — expressions over
+    SUM(partial_col) terms with aliases, e.g.
+    "sum(e0_agg_0) as revenue" or "sum(e3_agg_0) / sum(e3_agg_1) as avg_x".
+
+    execute() accumulates partial rows into a device group-by hash table
+    (single i64 group key, or multiple keys composite-encoded — DESIGN.md);
+    done() extracts groups, evaluates the final expressions host-side over
+    the per-group sums (tiny), applies order-by, returns a pyarrow Table.
+    """
+
+    def __init__(self, groupby_keys, orderby_keys, sql_statement):
+        assert isinstance(groupby_keys, list)
+        self.groupby_keys = groupby_keys
+        self.orderby_keys = orderby_keys or []
+        self.sql_statement = sql_statement
+        self.sum_cols = []          # partial columns referenced by SUM()
+        self.exprs = []             # (alias, python expr over s['col'])
+        for part in self._split_top(sql_statement):
+            alias = None
+            m = re.search(r"\s+as\s+([A-Za-z_][A-Za-z0-9_]*)\s*$", part,
+                          re.I)
+            if m:
+                alias = m.group(1)
+                part = part[: m.start()]
+            cols = _SUM_RE.findall(part)
+            if not cols:
+                raise ValueError("unsupported aggregate (round 1 supports "
+                                 "expressions over SUM(col)): %r" % part)
+            for c in cols:
+                if c not in self.sum_cols:
+                    self.sum_cols.append(c)
+            expr = _SUM_RE.sub(lambda m: "s[%r]" % m.group(1), part)
+            self.exprs.append((alias or part.strip(), expr))
+        self._gb = None
+        self._key_state = None
+
+    @staticmethod
+    def _split_top(s):
+        parts, depth, cur = [], 0, []
+        for ch in s:
+            if ch == "(":
+                depth += 1
+            elif ch == ")":
+                depth -= 1
+            if ch == "," and depth == 0:
+                parts.append("".join(cur))
+                cur = []
+            else:
+                cur.append(ch)
+        if cur:
+            parts.append("".join(cur))
+        return [p.strip() for p in parts if p.strip()]
+
+    def __getstate__(self):
+        assert self._gb is None, "executor must be pickled before first execute"
+        return dict(self.__dict__)
+
+    def _encode_keys(self, batch):
+        """Multiple group keys -> one i64 composite via accumulated host
+        codebooks (DESIGN.md §Group keys). Single i64 key passes through."""
+        ops, shim, staging = _lazy_gpu()
+        if self._key_state is None:
+            self._key_state = {"dicts": {}, "decode": []}
+        arrs = []
+        for k in self.groupby_keys:
+            col = batch.column(k)
+            sd = self._key_state["dicts"].setdefault(k, staging.StringDict())
+            try:
+                arr = staging.column_to_numpy(col, sd)
+            except TypeError:
+                arr = staging.column_to_numpy(col)
+            arrs.append(arr)
+        if len(arrs) == 1 and arrs[0].dtype == np.int64:
+            self._key_state["mode"] = "passthrough"
+            return arrs[0]
+        # composite: each non-i64 key coded via np codebook, packed base-N
+        self._key_state["mode"] = "composite"
+        packed = np.zeros(len(arrs[0]), dtype=np.int64)
+        widths = []
+        comps = []
+        for k, arr in zip(self.groupby_keys, arrs):
+            cb = self._key_state.setdefault("codebooks", {}).setdefault(k, {})
+            vals, inv = np.unique(arr, return_inverse=True)
+            codes = np.empty(len(vals), dtype=np.int64)
+            for i, v in enumerate(vals):
+                key = v.item() if hasattr(v, "item") else v
+                if key not in cb:
+                    cb[key] = len(cb)
+                codes[i] = cb[key]
+            comps.append(codes[inv])
+        for c in comps:
+            packed = packed * (1 << 21) + c  # 21 bits per key, <= 3 keys
+        self._key_state["ncomps"] = len(comps)
+        return packed
+
+    def _decode_keys(self, packed):
+        ncomps = self._key_state.get("ncomps", 1)
+        outs = []
+        rem = packed.astype(np.int64)
+        for _ in range(ncomps):
+            outs.append(rem % (1 << 21))
+            rem = rem // (1 << 21)
+        outs.reverse()
+        cols = {}
+        for k, codes in zip(self.groupby_keys, outs):
+            cb = self._key_state["codebooks"][k]
+            inv = {v: kk for kk, v in cb.items()}
+            cols[k] = np.array([inv[c] for c in codes.astype(np.int64)])
+        return cols
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        ops, shim, staging = _lazy_gpu()
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if not batches:
+            return
+        batch = pa.concat_tables(batches)
+        if self.groupby_keys:
+            keys = self._encode_keys(batch)
+        else:
+            keys = np.zeros(len(batch), dtype=np.int64)
+        if self._gb is None:
+            self._gb = ops.GroupByI64(
+                expected_groups=max(1024, len(batch)), nvals=len(self.sum_cols))
+        kcol = shim.DevColumn.from_numpy(keys)
+        vcols = [shim.DevColumn.from_numpy(
+            staging.column_to_numpy(batch.column(c)).astype(np.float64))
+            for c in self.sum_cols]
+        self._gb.update(kcol, vcols)
+        kcol.free()
+        for v in vcols:
+            v.free()
+
+    def done(self, executor_id):
+        if self._gb is None:
+            return None
+        keys, sums = self._gb.extract()
+        s = {c: sums[i] for i, c in enumerate(self.sum_cols)}
+        out = {}
+        if self.groupby_keys:
+            if self._key_state and self._key_state.get("mode") == "composite":
+                out.update(self._decode_keys(keys))
+            else:
+                out[self.groupby_keys[0]] = keys
+        for alias, expr in self.exprs:
+            out[alias] = eval(expr, {"s": s})  # noqa: S307 — expr built above
+        if self.orderby_keys:
+            cols = []
+            for item in reversed(self.orderby_keys):
+                k, d = item if isinstance(item, (tuple, list)) else (item, "asc")
+                v = out[k]
+                cols.append(-v if d == "desc" else v)
+            order = np.lexsort(cols)
+            out = {k: np.asarray(v)[order] for k, v in out.items()}
+        self._gb.free()
+        self._gb = None
+        return _to_table(out)
+
+
+def gpu_partition_fn(data, source_channel, num_target_channels, key=None):
+    """GPU hash partitioner mirroring partition_key_str
+    (quokka_runtime.py:217-231): int key -> key % N, bit-exact with :222.
+    `data`: pyarrow Table; returns dict target_channel -> pyarrow Table."""
+    import pyarrow as pa
+    ops, shim, staging = _lazy_gpu()
+    keys = staging.column_to_numpy(data.column(key))
+    if keys.dtype != np.int64:
+        raise TypeError("gpu_partition_fn: int64 keys only in round 1")
+    kcol = shim.DevColumn.from_numpy(keys)
+    offsets, idx = ops.partition_i64(kcol, num_target_channels)
+    sel = idx.to_numpy(len(keys))
+    host_cols = {c: staging.column_to_numpy(data.column(c))
+                 for c in data.column_names}
+    out = {}
+    for p in range(num_target_channels):
+        lo, hi = int(offsets[p]), int(offsets[p + 1])
+        if hi == lo:
+            continue
+        rows = sel[lo:hi]
+        out[p] = pa.table({c: v[rows] for c, v in host_cols.items()})
+    kcol.free()
+    idx.free()
+    return out

@@ -1,0 +1,246 @@
+"""Mid-level device operators over DevColumns.
+
+Each operator maps 1:1 onto a reference delegate (SURVEY.md §2 native-
+accounting table): filter -> polars DataFrame.filter (core.py:170);
+join build/probe -> polars join (sql_executors.py:371); groupby -> DuckDB
+group-by (sql_executors.py:595); partition -> partition_key_str
+(quokka_runtime.py:217-231).
+"""
+import ctypes
+
+import numpy as np
+
+from . import shim
+from .shim import DevBuffer, DevColumn, c_u64, c_u32, c_i64, c_i32, c_vp
+
+# comparison ops for filter kernels
+LT, LE, GT, GE, EQ, NE = range(6)
+
+
+def _count_buf():
+    b = DevBuffer(8)
+    shim.call("qk_dmemset", b.ptr, 0, c_u64(8))
+    return b
+
+
+def _read_u64(buf):
+    out = np.zeros(1, dtype=np.uint64)
+    shim.call("qk_d2h", out.ctypes.data_as(c_vp), buf.ptr, c_u64(8))
+    return int(out[0])
+
+
+def filter_col(col, op, value, stream=None):
+    """Ordered compaction: returns (idx DevColumn u32, count). Row order of
+    passing rows is preserved (polars filter semantics)."""
+    sh = stream.handle if stream else None
+    idx = DevColumn(np.uint32, col.n)
+    cnt = _count_buf()
+    if col.dtype == np.dtype(np.int32):
+        shim.call("qk_filter_i32", sh, c_u64(col.n), col.ptr, op,
+                  c_i32(int(value)), idx.ptr, cnt.ptr)
+    elif col.dtype == np.dtype(np.uint8):
+        shim.call("qk_filter_u8", sh, c_u64(col.n), col.ptr, op,
+                  ctypes.c_uint8(int(value)), idx.ptr, cnt.ptr)
+    else:
+        raise TypeError("filter_col: unsupported dtype %s" % col.dtype)
+    if stream:
+        stream.sync()
+    n = _read_u64(cnt)
+    cnt.free()
+    idx.n = n  # logical length; allocation stays col.n
+    return idx, n
+
+
+def q1_agg(n, shipdate, qty, price, disc, tax, rflag, lstat, cutoff,
+           out_buf, stream=None):
+    """Accumulate Q1 partials into out_buf (DevBuffer of 48 f64, zeroed by
+    caller). See include/quokka_amd.h qk_q1_agg."""
+    sh = stream.handle if stream else None
+    shim.call("qk_q1_agg", sh, c_u64(n), shipdate.ptr, qty.ptr, price.ptr,
+              disc.ptr, tax.ptr, rflag.ptr, lstat.ptr, c_i32(cutoff),
+              out_buf.ptr)
+
+
+def q1_read_partials(out_buf):
+    """d2h the 6x8 accumulator block -> (6,6) float array
+    [sum_qty,sum_base,sum_disc_price,sum_charge,sum_disc,count]."""
+    host = np.zeros(48, dtype=np.float64)
+    shim.call("qk_d2h", host.ctypes.data_as(c_vp), out_buf.ptr, c_u64(48 * 8))
+    return host.reshape(6, 8)[:, :6].copy()
+
+
+def q6_agg(n, shipdate, qty, price, disc, date_lo, date_hi, disc_lo, disc_hi,
+           qty_hi, out_buf, stream=None):
+    sh = stream.handle if stream else None
+    shim.call("qk_q6_agg", sh, c_u64(n), shipdate.ptr, qty.ptr, price.ptr,
+              disc.ptr, c_i32(date_lo), c_i32(date_hi),
+              ctypes.c_double(disc_lo), ctypes.c_double(disc_hi),
+              ctypes.c_double(qty_hi), out_buf.ptr)
+
+
+def _pow2_at_least(n):
+    c = 1
+    while c < n:
+        c <<= 1
+    return c
+
+
+class JoinTable:
+    """Device hash-join state = BuildProbeJoinExecutor's vstacked build side
+    (sql_executors.py:346-374). build() may be called per build batch;
+    probe() per probe batch."""
+
+    def __init__(self, expected_build_rows, stream=None):
+        self.cap = _pow2_at_least(max(16, 2 * int(expected_build_rows)))
+        self.stream = stream
+        self.slot_keys = DevColumn(np.int64, self.cap)
+        self.slot_head = DevColumn(np.int32, self.cap)
+        self.chain_cap = max(16, int(expected_build_rows))
+        self.chain_next = DevColumn(np.int32, self.chain_cap)
+        self.n_build = 0
+        sh = stream.handle if stream else None
+        shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
+                  c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
+        shim.call("qk_dmemset", self.slot_head.ptr, 0xFF,
+                  c_u64(self.cap * 4))
+
+    def build(self, keys_col, n=None):
+        n = keys_col.n if n is None else n
+        if self.n_build + n > self.chain_cap or self.n_build + n > self.cap // 2:
+            raise RuntimeError(
+                "JoinTable overflow: built %d + %d > capacity %d; size the "
+                "table from the build side's total rows"
+                % (self.n_build, n, min(self.chain_cap, self.cap // 2)))
+        sh = self.stream.handle if self.stream else None
+        shim.call("qk_join_build", sh, c_u64(n), keys_col.ptr,
+                  c_u32(self.n_build), self.slot_keys.ptr,
+                  self.slot_head.ptr, self.chain_next.ptr, c_u64(self.cap))
+        self.n_build += n
+
+    def probe(self, keys_col, mode=0, out_factor=1.5, n=None):
+        """mode 0 inner / 1 semi / 2 anti. Returns (probe_idx, build_idx,
+        count) DevColumns (build_idx None for semi/anti). Grows the output
+        and re-probes when the first guess undershoots (counted, never
+        truncated)."""
+        n = keys_col.n if n is None else n
+        sh = self.stream.handle if self.stream else None
+        out_cap = max(16, int(n * out_factor))
+        while True:
+            probe_idx = DevColumn(np.uint32, out_cap)
+            build_idx = DevColumn(np.uint32, out_cap) if mode == 0 else None
+            cur = _count_buf()
+            shim.call("qk_join_probe", sh, c_u64(n), keys_col.ptr,
+                      self.slot_keys.ptr, self.slot_head.ptr,
+                      self.chain_next.ptr, c_u64(self.cap), mode,
+                      probe_idx.ptr, build_idx.ptr if build_idx else None,
+                      c_u64(out_cap), cur.ptr)
+            if self.stream:
+                self.stream.sync()
+            total = _read_u64(cur)
+            cur.free()
+            if total <= out_cap:
+                probe_idx.n = total
+                if build_idx is not None:
+                    build_idx.n = total
+                return probe_idx, build_idx, total
+            probe_idx.free()
+            if build_idx is not None:
+                build_idx.free()
+            out_cap = int(total)
+
+    def free(self):
+        self.slot_keys.free()
+        self.slot_head.free()
+        self.chain_next.free()
+
+
+class GroupByI64:
+    """Device group-by accumulate table over an i64 key with nvals f64 SUM
+    columns (SQLAggExecutor's post-rewrite distributive form,
+    sql_utils.py:299-413). update() per batch, extract() at done()."""
+
+    def __init__(self, expected_groups, nvals, stream=None):
+        self.cap = _pow2_at_least(max(16, 2 * int(expected_groups)))
+        self.nvals = nvals
+        self.stream = stream
+        self.slot_keys = DevColumn(np.int64, self.cap)
+        self.slot_sums = DevColumn(np.float64, self.cap * nvals)
+        sh = stream.handle if stream else None
+        shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
+                  c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
+        shim.call("qk_dmemset", self.slot_sums.ptr, 0,
+                  c_u64(self.cap * nvals * 8))
+
+    def update(self, keys_col, val_cols, n=None):
+        n = keys_col.n if n is None else n
+        if not n:
+            return
+        assert len(val_cols) == self.nvals
+        sh = self.stream.handle if self.stream else None
+        ptrs = np.array([c.ptr.value if hasattr(c.ptr, "value") else c.ptr
+                         for c in val_cols], dtype=np.uint64)
+        dptrs = DevBuffer(ptrs.nbytes)
+        shim.call("qk_h2d", dptrs.ptr, ptrs.ctypes.data_as(c_vp),
+                  c_u64(ptrs.nbytes))
+        shim.call("qk_groupby_i64_sum", sh, c_u64(n), keys_col.ptr,
+                  dptrs.ptr, self.nvals, self.slot_keys.ptr,
+                  self.slot_sums.ptr, c_u64(self.cap))
+        if self.stream:
+            self.stream.sync()
+        dptrs.free()
+
+    def extract(self):
+        """-> (keys np.int64[K], sums np.float64[nvals, K]); unordered."""
+        sh = self.stream.handle if self.stream else None
+        out_cap = self.cap
+        out_keys = DevColumn(np.int64, out_cap)
+        out_sums = DevColumn(np.float64, out_cap * self.nvals)
+        cur = _count_buf()
+        shim.call("qk_groupby_extract", sh, self.slot_keys.ptr,
+                  self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+                  out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
+        if self.stream:
+            self.stream.sync()
+        k = _read_u64(cur)
+        cur.free()
+        keys = out_keys.to_numpy(k)
+        sums = out_sums.to_numpy(out_cap * self.nvals).reshape(
+            self.nvals, out_cap)[:, :k].copy()
+        out_keys.free()
+        out_sums.free()
+        return keys, sums
+
+    def free(self):
+        self.slot_keys.free()
+        self.slot_sums.free()
+
+
+def partition_i64(keys_col, nparts, stream=None, n=None):
+    """Hash partition, int-key semantics key % nparts
+    (quokka_runtime.py:222). Returns (offsets np.uint64[nparts+1],
+    idx DevColumn u32 with rows grouped by partition)."""
+    n = keys_col.n if n is None else n
+    sh = stream.handle if stream else None
+    hist = DevBuffer(nparts * 8)
+    shim.call("qk_dmemset", hist.ptr, 0, c_u64(nparts * 8))
+    shim.call("qk_partition_hist", sh, c_u64(n), keys_col.ptr,
+              c_u32(nparts), hist.ptr)
+    if stream:
+        stream.sync()
+    h = np.zeros(nparts, dtype=np.uint64)
+    shim.call("qk_d2h", h.ctypes.data_as(c_vp), hist.ptr, c_u64(nparts * 8))
+    offsets = np.zeros(nparts + 1, dtype=np.uint64)
+    np.cumsum(h, out=offsets[1:])
+    cursors = DevBuffer(nparts * 8)
+    shim.call("qk_h2d", cursors.ptr,
+              np.ascontiguousarray(offsets[:nparts]).ctypes.data_as(c_vp),
+              c_u64(nparts * 8))
+    idx = DevColumn(np.uint32, max(1, n))
+    shim.call("qk_partition_scatter", sh, c_u64(n), keys_col.ptr,
+              c_u32(nparts), cursors.ptr, idx.ptr)
+    if stream:
+        stream.sync()
+    hist.free()
+    cursors.free()
+    idx.n = n
+    return offsets, idx

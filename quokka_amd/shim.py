@@ -1,0 +1,222 @@
+"""ctypes binding of the quokka_amd C ABI (include/quokka_amd.h).
+
+This is the ONLY place the product path touches the HIP library. If the
+library is missing or was built for the wrong arch, import fails loudly —
+there is no CPU fallback anywhere in quokka_amd (the CPU restatement under
+oracle/ is test infrastructure and is never imported from here).
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_LIB_PATH = os.path.join(os.path.dirname(__file__), "libquokka_amd.so")
+
+if not os.path.exists(_LIB_PATH):
+    raise ImportError(
+        "quokka_amd: HIP extension libquokka_amd.so not found at %s. "
+        "Build it with `make -C quokka_amd/csrc` (hipcc --offload-arch=gfx950). "
+        "quokka_amd has no CPU fallback." % _LIB_PATH
+    )
+
+_lib = ctypes.CDLL(_LIB_PATH)
+
+c_u64 = ctypes.c_uint64
+c_i64 = ctypes.c_int64
+c_i32 = ctypes.c_int32
+c_u32 = ctypes.c_uint32
+c_u8 = ctypes.c_uint8
+c_f64 = ctypes.c_double
+c_vp = ctypes.c_void_p
+
+_lib.qk_last_error.restype = ctypes.c_char_p
+_lib.qk_build_arch.restype = ctypes.c_char_p
+
+_SIGS = {
+    "qk_init": [ctypes.c_int],
+    "qk_device_count": [c_vp],
+    "qk_dmalloc": [c_u64, c_vp],
+    "qk_dfree": [c_vp],
+    "qk_h2d": [c_vp, c_vp, c_u64],
+    "qk_d2h": [c_vp, c_vp, c_u64],
+    "qk_dmemset": [c_vp, ctypes.c_int, c_u64],
+    "qk_fill_i64": [c_vp, c_vp, c_i64, c_u64],
+    "qk_stream_create": [c_vp],
+    "qk_stream_destroy": [c_vp],
+    "qk_stream_sync": [c_vp],
+    "qk_timer_create": [c_vp],
+    "qk_timer_destroy": [c_vp],
+    "qk_timer_start": [c_vp, c_vp],
+    "qk_timer_stop": [c_vp, c_vp],
+    "qk_timer_elapsed_ms": [c_vp, c_vp],
+    "qk_gen_lineitem": [c_vp, c_u64, c_u64, c_u64, c_i64, c_i64, c_i64,
+                        c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
+    "qk_q1_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                  c_i32, c_vp],
+    "qk_q6_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_i32,
+                  c_f64, c_f64, c_f64, c_vp],
+    "qk_filter_i32": [c_vp, c_u64, c_vp, ctypes.c_int, c_i32, c_vp, c_vp],
+    "qk_filter_u8": [c_vp, c_u64, c_vp, ctypes.c_int, c_u8, c_vp, c_vp],
+    "qk_mul_1md": [c_vp, c_u64, c_vp, c_vp, c_vp],
+    "qk_gather_i64": [c_vp, c_u64, c_vp, c_vp, c_vp],
+    "qk_gather_f64": [c_vp, c_u64, c_vp, c_vp, c_vp],
+    "qk_gather_i32": [c_vp, c_u64, c_vp, c_vp, c_vp],
+    "qk_gather_u8": [c_vp, c_u64, c_vp, c_vp, c_vp],
+    "qk_join_build": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp, c_vp, c_u64],
+    "qk_join_probe": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_u64,
+                      ctypes.c_int, c_vp, c_vp, c_u64, c_vp],
+    "qk_groupby_i64_sum": [c_vp, c_u64, c_vp, c_vp, ctypes.c_int, c_vp, c_vp,
+                           c_u64],
+    "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,
+                           c_u64, c_vp],
+    "qk_partition_hist": [c_vp, c_u64, c_vp, c_u32, c_vp],
+    "qk_partition_scatter": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp],
+}
+for name, argtypes in _SIGS.items():
+    fn = getattr(_lib, name)
+    fn.argtypes = argtypes
+    fn.restype = ctypes.c_int
+
+
+class QkError(RuntimeError):
+    pass
+
+
+def _check(rc, name):
+    if rc != 0:
+        raise QkError("%s failed (rc=%d): %s"
+                      % (name, rc, _lib.qk_last_error().decode()))
+
+
+def call(name, *args):
+    _check(getattr(_lib, name)(*args), name)
+
+
+def build_arch():
+    return _lib.qk_build_arch().decode()
+
+
+def init(device=0):
+    call("qk_init", device)
+
+
+def device_count():
+    n = ctypes.c_int(0)
+    rc = _lib.qk_device_count(ctypes.byref(n))
+    if rc == 100:  # hipErrorNoDevice: a GPU-less host has 0 devices
+        return 0
+    _check(rc, "qk_device_count")
+    return n.value
+
+
+JOIN_EMPTY = np.int64(-(2 ** 63))  # QK_JOIN_EMPTY
+
+_DTYPE_GATHER = {
+    np.dtype(np.int64): "qk_gather_i64",
+    np.dtype(np.float64): "qk_gather_f64",
+    np.dtype(np.int32): "qk_gather_i32",
+    np.dtype(np.uint8): "qk_gather_u8",
+    np.dtype(np.uint32): "qk_gather_i32",  # same width
+}
+
+
+class DevBuffer:
+    """Owning device allocation."""
+
+    def __init__(self, nbytes):
+        p = c_vp(0)
+        call("qk_dmalloc", c_u64(max(1, nbytes)), ctypes.byref(p))
+        self.ptr = p
+        self.nbytes = nbytes
+
+    def free(self):
+        if self.ptr is not None and self.ptr.value:
+            call("qk_dfree", self.ptr)
+            self.ptr = None
+
+    def __del__(self):
+        try:
+            self.free()
+        except Exception:
+            pass
+
+
+class DevColumn:
+    """A dense device column of numpy dtype `dtype` with `n` rows."""
+
+    def __init__(self, dtype, n):
+        self.dtype = np.dtype(dtype)
+        self.n = int(n)
+        self.buf = DevBuffer(self.n * self.dtype.itemsize)
+
+    @property
+    def ptr(self):
+        return self.buf.ptr
+
+    @classmethod
+    def from_numpy(cls, arr):
+        arr = np.ascontiguousarray(arr)
+        col = cls(arr.dtype, len(arr))
+        if len(arr):
+            call("qk_h2d", col.ptr, arr.ctypes.data_as(c_vp),
+                 c_u64(arr.nbytes))
+        return col
+
+    def to_numpy(self, n=None):
+        n = self.n if n is None else int(n)
+        out = np.empty(n, dtype=self.dtype)
+        if n:
+            call("qk_d2h", out.ctypes.data_as(c_vp), self.ptr,
+                 c_u64(out.nbytes))
+        return out
+
+    def gather(self, idx_col, n_idx, stream=None):
+        """New column: self[idx] for a device u32 index column."""
+        fn = _DTYPE_GATHER[self.dtype]
+        out = DevColumn(self.dtype, n_idx)
+        if n_idx:
+            call(fn, stream, c_u64(n_idx), idx_col.ptr, self.ptr, out.ptr)
+        return out
+
+    def free(self):
+        self.buf.free()
+
+
+class Stream:
+    def __init__(self):
+        p = c_vp(0)
+        call("qk_stream_create", ctypes.byref(p))
+        self.handle = p
+
+    def sync(self):
+        call("qk_stream_sync", self.handle)
+
+    def destroy(self):
+        if self.handle is not None:
+            call("qk_stream_destroy", self.handle)
+            self.handle = None
+
+
+class Timer:
+    """HIP-event timer pair on a given stream (roofline measurement)."""
+
+    def __init__(self):
+        p = c_vp(0)
+        call("qk_timer_create", ctypes.byref(p))
+        self.handle = p
+
+    def start(self, stream):
+        call("qk_timer_start", self.handle, stream.handle if stream else None)
+
+    def stop(self, stream):
+        call("qk_timer_stop", self.handle, stream.handle if stream else None)
+
+    def elapsed_ms(self):
+        out = ctypes.c_float(0)
+        call("qk_timer_elapsed_ms", self.handle, ctypes.byref(out))
+        return float(out.value)
+
+    def destroy(self):
+        if self.handle is not None:
+            call("qk_timer_destroy", self.handle)
+            self.handle = None
