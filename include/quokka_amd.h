@@ -68,6 +68,17 @@ int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed
                     uint8_t *l_returnflag, uint8_t *l_linestatus,
                     int32_t *l_shipdate);
 
+/* Orders columns for rows [row_offset, row_offset+n): o_orderkey dense
+ * row+1, o_custkey uniform 1..n_customers, o_orderdate uniform spec range,
+ * o_shippriority 0. Consistent with qk_gen_lineitem's l_orderkey
+ * (= 1 + row/4 % n_orders: every order has exactly 4 lines). */
+int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
+                  int64_t n_customers, int64_t *o_orderkey, int64_t *o_custkey,
+                  int32_t *o_orderdate, int32_t *o_shippriority);
+/* Customer columns: c_custkey dense row+1, c_mktsegment uniform u8 0..4. */
+int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
+                    uint64_t seed, int64_t *c_custkey, uint8_t *c_mktsegment);
+
 /* ---- TPC-H Q1: fused filter + group-by partial aggregate ------------- *
  * Replaces the map-side partial agg the reference folds into partition_fn
  * (pyquokka/core.py:173-176 + df.py:1354-1394: per-batch DuckDB

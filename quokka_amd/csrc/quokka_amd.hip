@@ -212,6 +212,63 @@ extern "C" int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset,
   return 0;
 }
 
+__global__ void k_gen_orders(uint64_t n, uint64_t row_offset, uint64_t seed,
+                             int64_t n_customers, int64_t *o_orderkey,
+                             int64_t *o_custkey, int32_t *o_orderdate,
+                             int32_t *o_shippriority) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t row = row_offset + i;
+    uint64_t base = splitmix64(seed ^ 0x0DE50DE50DE50DE5ULL) ^
+                    (row * 0x9E3779B97F4A7C15ULL);
+    uint64_t h0 = splitmix64(base + 0), h1 = splitmix64(base + 1);
+    if (o_orderkey) o_orderkey[i] = (int64_t)row + 1;
+    if (o_custkey)
+      o_custkey[i] = 1 + (int64_t)(h0 % (uint64_t)n_customers);
+    if (o_orderdate)
+      o_orderdate[i] = QK_ORDERDATE_LO +
+          (int32_t)(h1 % (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
+    if (o_shippriority) o_shippriority[i] = 0;
+  }
+}
+extern "C" int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset,
+                             uint64_t seed, int64_t n_customers,
+                             int64_t *o_orderkey, int64_t *o_custkey,
+                             int32_t *o_orderdate, int32_t *o_shippriority) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gen_orders, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, row_offset, seed, n_customers,
+                     o_orderkey, o_custkey, o_orderdate, o_shippriority);
+  QK_TRY("qk_gen_orders", hipGetLastError());
+  return 0;
+}
+
+__global__ void k_gen_customer(uint64_t n, uint64_t row_offset, uint64_t seed,
+                               int64_t *c_custkey, uint8_t *c_mktsegment) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t row = row_offset + i;
+    uint64_t h = splitmix64((seed ^ 0xC0573C0573C0573CULL) +
+                            row * 0x9E3779B97F4A7C15ULL);
+    if (c_custkey) c_custkey[i] = (int64_t)row + 1;
+    if (c_mktsegment) c_mktsegment[i] = (uint8_t)(h % 5);
+  }
+}
+extern "C" int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
+                               uint64_t seed, int64_t *c_custkey,
+                               uint8_t *c_mktsegment) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gen_customer, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, row_offset, seed, c_custkey,
+                     c_mktsegment);
+  QK_TRY("qk_gen_customer", hipGetLastError());
+  return 0;
+}
+
 // ---- Q1 fused filter + group-by partial aggregate ---------------------
 // 6 groups x 6 accumulators kept in REGISTERS per thread (statically
 // indexed; runtime-indexed per-thread arrays spill to scratch on hipcc —
