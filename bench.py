@@ -41,6 +41,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
+    p.add_argument("--query", choices=["q1", "q3"], default="q1")
     p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -90,15 +91,192 @@ def cpu_baseline(sample_rows, target_seconds=12.0):
     }
 
 
+def gen_device_q3_tables(shim, n, rank):
+    """Lineitem (4 cols) + orders + customer shards for the fused Q3."""
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n_ord = max(1, n // 4)
+    n_cust = max(1, n_ord // 10)
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_shipdate", np.int32),
+        ("l_extendedprice", np.float64), ("l_discount", np.float64)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord),
+              li["l_orderkey"].ptr, None, None, li["l_extendedprice"].ptr,
+              li["l_discount"].ptr, None, None, None, li["l_shipdate"].ptr)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_custkey", np.int64),
+        ("o_orderdate", np.int32), ("o_shippriority", np.int32)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
+              c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
+              od["o_orderdate"].ptr, od["o_shippriority"].ptr)
+    cu = {"c_custkey": DevColumn(np.int64, n_cust),
+          "c_mktsegment": DevColumn(np.uint8, n_cust)}
+    shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
+              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr)
+    return li, od, cu
+
+
+def cpu_baseline_q3(sample_rows, target_seconds=12.0):
+    from oracle import tpch_gen as G, queries as OQ
+    sf = sample_rows / 6_000_000
+    d = {"orders": G.gen_orders(sf, 42)}
+    d["lineitem"] = G.gen_lineitem(sf, 42, d["orders"])
+    d["customer"] = G.gen_customer(sf, 42)
+    n = len(d["lineitem"]["l_orderkey"])
+    t0 = time.time()
+    OQ.q3(d["lineitem"], d["orders"], d["customer"])
+    per = time.time() - t0
+    passes = max(1, min(16, int(target_seconds / max(per, 1e-3))))
+    t0 = time.time()
+    for _ in range(passes):
+        OQ.q3(d["lineitem"], d["orders"], d["customer"])
+    dt = time.time() - t0
+    return {
+        "value": n * passes / dt, "unit": "rows/s", "cores": 1,
+        "kind": "port",
+        "sample": "%.1fM-row seeded lineitem (+orders/customer) x %d passes "
+                  "of the numpy oracle (oracle/queries.py:q3), "
+                  "single-threaded" % (n / 1e6, passes),
+    }
+
+
 def read_traffic():
     """Per-launch HBM bytes from the committed rocprofv3 PMC measurement
     (profiles/traffic_q1.json), or None before one exists."""
-    path = os.path.join(ROOT, "profiles", "traffic_q1.json")
+    path = os.path.join(ROOT, "profiles", "traffic_%s.json" % QUERY)
     if os.path.exists(path):
         with open(path) as f:
             d = json.load(f)
         return d.get("traffic_bytes_per_launch")
     return None
+
+
+QUERY = "q1"
+
+
+def main_q3(args, n, world, rank, dist, shim, DQ):
+    """TPC-H Q3 on the fused device path (BASELINE.json configs[2]).
+    A step = rebuild customer+orders tables + fused probe/agg + extract
+    top-10. Multi-rank: lineitem sharded by row range; the build tables are
+    broadcast-replicated (small sides); group tables merged
+    host-side at each step."""
+    import ctypes as ct
+    from quokka_amd import ops
+    from quokka_amd.shim import DevBuffer, c_u64
+
+    li, od, cu = gen_device_q3_tables(shim, n, rank)
+    stream = shim.Stream()
+    fused = DQ.Q3Fused(od, cu, stream)
+    stream.sync()
+    timer = shim.Timer()
+
+    # one-off counts for the algorithmic-byte accounting (DESIGN.md §Q3)
+    mc = ops._count_buf()
+    fused.probe(li, mc)
+    stream.sync()
+    n_match = ops._read_u64(mc)
+    mc.free()
+    fused.reset_sums()
+    idxbuf = shim.DevColumn(np.uint32, n)
+    cntbuf = ops._count_buf()
+    shim.call("qk_filter_i32", stream.handle, c_u64(n),
+              li["l_shipdate"].ptr, 2, ct.c_int32(DQ.Q3_DATE),
+              idxbuf.ptr, cntbuf.ptr)
+    stream.sync()
+    n_pass = ops._read_u64(cntbuf)
+    idxbuf.free(); cntbuf.free()
+
+    def step(timed):
+        fused.rebuild()
+        if timed:
+            timer.start(stream)
+        fused.probe(li)
+        if timed:
+            timer.stop(stream)
+        full, top10 = fused.extract(10)
+        if dist is not None:
+            import torch.distributed as _d
+            gathered = [None] * world
+            _d.all_gather_object(gathered, (full["l_orderkey"],
+                                            full["revenue"]))
+            if rank == 0:
+                import collections
+                acc = collections.defaultdict(float)
+                for ks, vs in gathered:
+                    for k, v in zip(ks.tolist(), vs.tolist()):
+                        acc[k] += v
+                top10 = dict(n_groups=len(acc))
+        return full, top10, (timer.elapsed_ms() if timed else None)
+
+    for _ in range(args.warmup):
+        step(False)
+    if dist is not None:
+        dist.barrier()
+    stream.sync()
+    t0 = time.time()
+    kernel_ms = []
+    full = None
+    for _ in range(args.steps):
+        full, top10, kms = step(True)
+        kernel_ms.append(kms)
+    stream.sync()
+    elapsed = time.time() - t0
+    if dist is not None:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+        dist.barrier()
+
+    if rank == 0:
+        # probe-kernel algorithmic bytes: 12 B/row (orderkey+shipdate) every
+        # row + 12 B bucket read per ship-passing row + 16 B (price+disc)
+        # per matched row
+        alg_bytes = 12 * n + 12 * n_pass + 16 * n_match
+        avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
+        achieved_gbps = alg_bytes / avg_kernel_s / 1e9
+        out = {
+            "metric": "rows/s",
+            "value": n * world * args.steps / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": "TPC-H SF%g Q3 (3-way hash join + group-by), "
+                            "%d lineitem rows/GPU resident in HBM "
+                            "(BASELINE.json configs[2])" % (args.sf, n),
+                "sf_per_gpu": args.sf,
+                "rows_per_gpu": n,
+                "query": "Q3",
+                "n_groups": int(len(full["l_orderkey"])),
+                "orders_build_rows": int(fused.n_build),
+                "lineitem_ship_pass": int(n_pass),
+                "joined_rows": int(n_match),
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbps,
+                "peak": HBM_PEAK_GBPS,
+                "unit": "GB/s",
+                "frac": achieved_gbps / HBM_PEAK_GBPS,
+                "traffic": read_traffic(),
+            },
+            "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
+                             else cpu_baseline_q3(args.cpu_sample_rows)),
+        }
+        print(json.dumps(out))
+    timer.destroy()
+    fused.free()
+    stream.destroy()
+    if dist is not None:
+        dist.destroy_process_group()
 
 
 def main():
@@ -116,8 +294,12 @@ def main():
     from quokka_amd import shim, ops, queries as DQ
     shim.init(local_rank)
 
+    global QUERY
+    QUERY = args.query
     n = int(round(args.sf / 100.0 * SF100_LINEITEM_ROWS))
-    n &= ~1  # even row count -> fully vectorized path
+    n &= ~3  # multiple of 4 -> vectorized Q1 path, 4 lines/order for Q3
+    if args.query == "q3":
+        return main_q3(args, n, world, rank, dist, shim, DQ)
     cols = gen_device_lineitem(shim, n, rank)
     stream = shim.Stream()
     timer = shim.Timer()

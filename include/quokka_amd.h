@@ -157,6 +157,42 @@ int qk_join_probe(void *stream, uint64_t n_probe, const int64_t *keys,
                   uint32_t *out_probe_idx, uint32_t *out_build_idx,
                   uint64_t out_capacity, uint64_t *out_cursor_dev);
 
+/* ---- fused Q3 path ---------------------------------------------------- *
+ * Filter + semi-join + build and filter + probe + group-by aggregate fused
+ * into single passes (the reference folds filters and per-batch partial
+ * aggs into partition_fn the same way, core.py:152-195 + df.py:1354-1394).
+ * Build keys must be UNIQUE (orders/customer primary keys); slot_head
+ * stores the build ROW index; the orders table's slots double as group-by
+ * slots for the probe aggregate. Tables pre-filled like qk_join_build's. */
+int qk_build_u8eq(void *stream, uint64_t n, const int64_t *keys,
+                  const uint8_t *flag, uint8_t flag_val, int64_t *slot_keys,
+                  int32_t *slot_head, uint64_t capacity);
+int qk_q3_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
+                       const int64_t *o_custkey, const int32_t *o_orderdate,
+                       int32_t date_lt, const int64_t *cust_keys,
+                       const int32_t *cust_head, uint64_t cust_cap,
+                       int64_t *slot_keys, int32_t *slot_head,
+                       uint64_t capacity);
+/* Count the rows qk_q3_build_orders would insert (for tight table sizing;
+ * count_dev u64, zeroed). */
+int qk_q3_count_orders(void *stream, uint64_t n, const int64_t *o_custkey,
+                       const int32_t *o_orderdate, int32_t date_lt,
+                       const int64_t *cust_keys, const int32_t *cust_head,
+                       uint64_t cust_cap, uint64_t *count_dev);
+int qk_q3_probe_agg(void *stream, uint64_t n, const int64_t *l_orderkey,
+                    const int32_t *l_shipdate, const double *l_price,
+                    const double *l_disc, int32_t date_gt,
+                    const int64_t *slot_keys, const int32_t *slot_head,
+                    uint64_t capacity, double *slot_sums /* f64[capacity],
+                    zeroed; groups keyed by slot */,
+                    uint64_t *match_count_dev /* nullable, zeroed u64 */);
+/* Emit (orderkey, orders_build_row, revenue) for slots with sum != 0.
+ * cursor (u64, zeroed) = group count (counted even past out_cap). */
+int qk_q3_extract(void *stream, const int64_t *slot_keys,
+                  const int32_t *slot_head, const double *slot_sums,
+                  uint64_t capacity, int64_t *out_keys, int32_t *out_row,
+                  double *out_sums, uint64_t out_cap, uint64_t *cursor_dev);
+
 /* ---- group-by (i64 key) sum ------------------------------------------- *
  * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for
  * distributive SUM over an i64 key (the post-rewrite partial form,

@@ -165,13 +165,20 @@ __global__ void k_gen_lineitem(uint64_t n, uint64_t row_offset, uint64_t seed,
     uint64_t row = row_offset + i;
     uint64_t base = splitmix64(seed ^ 0x51A2B3C4D5E6F708ULL) ^
                     (row * 0x9E3779B97F4A7C15ULL);
-    uint64_t h0 = splitmix64(base + 0), h1 = splitmix64(base + 1),
+    uint64_t h1 = splitmix64(base + 1),
              h2 = splitmix64(base + 2), h3 = splitmix64(base + 3),
              h4 = splitmix64(base + 4), h5 = splitmix64(base + 5),
              h6 = splitmix64(base + 6), h7 = splitmix64(base + 7),
              h8 = splitmix64(base + 8);
-    int32_t odate =
-        QK_ORDERDATE_LO + (int32_t)(h0 % (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
+    // the ORDER's orderdate, derived with k_gen_orders' exact formula for
+    // order row = row/4, so l_shipdate correlates with o_orderdate as in
+    // TPC-H (Q3's date-window join selectivity depends on this)
+    uint64_t order_row = (row / 4) % (uint64_t)n_orders;
+    uint64_t obase = splitmix64(seed ^ 0x0DE50DE50DE50DE5ULL) ^
+                     (order_row * 0x9E3779B97F4A7C15ULL);
+    int32_t odate = QK_ORDERDATE_LO +
+        (int32_t)(splitmix64(obase + 1) %
+                  (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
     int32_t ship = odate + 1 + (int32_t)(h1 % 121);
     int32_t receipt = ship + 1 + (int32_t)(h2 % 30);
     if (l_shipdate) l_shipdate[i] = ship;
@@ -740,6 +747,254 @@ extern "C" int qk_join_probe(void *stream, uint64_t n, const int64_t *keys,
                      chain_next, cap, mode, out_probe, out_build, out_cap,
                      cursor);
   QK_TRY("qk_join_probe", hipGetLastError());
+  return 0;
+}
+
+// ---- fused Q3 path -----------------------------------------------------
+// Build/probe with the surrounding filter and aggregate FUSED into the
+// table pass (the reference folds the filter and per-batch partial agg
+// into partition_fn the same way, core.py:152-195 + df.py:1354-1394).
+// Build keys must be UNIQUE (orders/customer primary keys): slot_head
+// stores the build ROW, and the orders table's slots double as group-by
+// slots for the probe-side aggregate.
+
+// customer: insert keys[i] where flag[i] == flag_val
+__global__ void __launch_bounds__(BLOCK) k_build_u8eq(
+    uint64_t n, const int64_t *__restrict__ keys,
+    const uint8_t *__restrict__ flag, uint8_t flag_val,
+    int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
+    uint64_t cap) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (flag[i] != flag_val) continue;
+    int64_t key = keys[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    slot_head[s] = (int32_t)i;
+  }
+}
+extern "C" int qk_build_u8eq(void *stream, uint64_t n, const int64_t *keys,
+                             const uint8_t *flag, uint8_t flag_val,
+                             int64_t *slot_keys, int32_t *slot_head,
+                             uint64_t cap) {
+  if (!n) return 0;
+  if (cap & (cap - 1)) return qk_fail("qk_build_u8eq.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_build_u8eq, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, flag, flag_val, slot_keys,
+                     slot_head, cap);
+  QK_TRY("qk_build_u8eq", hipGetLastError());
+  return 0;
+}
+
+__device__ inline int32_t probe_unique(const int64_t *__restrict__ slot_keys,
+                                       const int32_t *__restrict__ slot_head,
+                                       uint64_t cap, int64_t key,
+                                       uint64_t *slot_out) {
+  uint64_t s = slot_of(key, cap);
+  for (;;) {
+    int64_t cur = slot_keys[s];
+    if (cur == key) {
+      if (slot_out) *slot_out = s;
+      return slot_head[s];
+    }
+    if (cur == QK_JOIN_EMPTY) return -1;
+    s = (s + 1) & (cap - 1);
+  }
+}
+
+// orders: insert o_orderkey[i] where o_orderdate[i] < date_lt AND
+// o_custkey[i] hits the customer table (fused filter + semi join + build)
+__global__ void __launch_bounds__(BLOCK) k_q3_build_orders(
+    uint64_t n, const int64_t *__restrict__ o_orderkey,
+    const int64_t *__restrict__ o_custkey,
+    const int32_t *__restrict__ o_orderdate, int32_t date_lt,
+    const int64_t *__restrict__ cust_keys,
+    const int32_t *__restrict__ cust_head, uint64_t cust_cap,
+    int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
+    uint64_t cap) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (o_orderdate[i] >= date_lt) continue;
+    if (probe_unique(cust_keys, cust_head, cust_cap, o_custkey[i], nullptr) < 0)
+      continue;
+    int64_t key = o_orderkey[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    slot_head[s] = (int32_t)i;
+  }
+}
+extern "C" int qk_q3_build_orders(void *stream, uint64_t n,
+                                  const int64_t *o_orderkey,
+                                  const int64_t *o_custkey,
+                                  const int32_t *o_orderdate, int32_t date_lt,
+                                  const int64_t *cust_keys,
+                                  const int32_t *cust_head, uint64_t cust_cap,
+                                  int64_t *slot_keys, int32_t *slot_head,
+                                  uint64_t cap) {
+  if (!n) return 0;
+  if ((cap & (cap - 1)) || (cust_cap & (cust_cap - 1)))
+    return qk_fail("qk_q3_build_orders.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_build_orders, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, o_orderkey, o_custkey,
+                     o_orderdate, date_lt, cust_keys, cust_head, cust_cap,
+                     slot_keys, slot_head, cap);
+  QK_TRY("qk_q3_build_orders", hipGetLastError());
+  return 0;
+}
+
+// count the rows k_q3_build_orders would insert (sizes the build table
+// tightly so probes stay cache-resident)
+__global__ void __launch_bounds__(BLOCK) k_q3_count_orders(
+    uint64_t n, const int64_t *__restrict__ o_custkey,
+    const int32_t *__restrict__ o_orderdate, int32_t date_lt,
+    const int64_t *__restrict__ cust_keys,
+    const int32_t *__restrict__ cust_head, uint64_t cust_cap,
+    uint64_t *__restrict__ count) {
+  uint32_t cnt = 0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (o_orderdate[i] >= date_lt) continue;
+    if (probe_unique(cust_keys, cust_head, cust_cap, o_custkey[i], nullptr) >= 0)
+      cnt++;
+  }
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w];
+    if (s) atomicAdd((unsigned long long *)count, (unsigned long long)s);
+  }
+}
+extern "C" int qk_q3_count_orders(void *stream, uint64_t n,
+                                  const int64_t *o_custkey,
+                                  const int32_t *o_orderdate, int32_t date_lt,
+                                  const int64_t *cust_keys,
+                                  const int32_t *cust_head, uint64_t cust_cap,
+                                  uint64_t *count_dev) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_count_orders, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, o_custkey, o_orderdate, date_lt,
+                     cust_keys, cust_head, cust_cap, count_dev);
+  QK_TRY("qk_q3_count_orders", hipGetLastError());
+  return 0;
+}
+
+// lineitem: one fused pass — filter l_shipdate > date_gt, probe orders,
+// revenue = price*(1-disc), atomicAdd into the orders table's slot
+// (slot == group, since orderkey is unique on the build side)
+__global__ void __launch_bounds__(BLOCK) k_q3_probe_agg(
+    uint64_t n, const int64_t *__restrict__ l_orderkey,
+    const int32_t *__restrict__ l_shipdate,
+    const double *__restrict__ l_price, const double *__restrict__ l_disc,
+    int32_t date_gt, const int64_t *__restrict__ slot_keys,
+    const int32_t *__restrict__ slot_head, uint64_t cap,
+    double *__restrict__ slot_sums, uint64_t *__restrict__ match_count) {
+  uint32_t matches = 0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (l_shipdate[i] <= date_gt) continue;
+    uint64_t s;
+    if (probe_unique(slot_keys, slot_head, cap, l_orderkey[i], &s) < 0)
+      continue;
+    matches++;
+    atomicAdd(&slot_sums[s], l_price[i] * (1.0 - l_disc[i]));
+  }
+  if (match_count) {
+    __shared__ uint32_t lds[BLOCK / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      matches += __shfl_down(matches, off);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = matches;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint64_t t = 0;
+      for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+      if (t) atomicAdd((unsigned long long *)match_count,
+                       (unsigned long long)t);
+    }
+  }
+}
+extern "C" int qk_q3_probe_agg(void *stream, uint64_t n,
+                               const int64_t *l_orderkey,
+                               const int32_t *l_shipdate,
+                               const double *l_price, const double *l_disc,
+                               int32_t date_gt, const int64_t *slot_keys,
+                               const int32_t *slot_head, uint64_t cap,
+                               double *slot_sums, uint64_t *match_count) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_q3_probe_agg.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_probe_agg, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, l_orderkey, l_shipdate, l_price,
+                     l_disc, date_gt, slot_keys, slot_head, cap, slot_sums,
+                     match_count);
+  QK_TRY("qk_q3_probe_agg", hipGetLastError());
+  return 0;
+}
+
+// extract groups: slots with a nonzero revenue sum; emits the group key,
+// the orders build ROW (for o_orderdate/o_shippriority attach) and the sum
+__global__ void __launch_bounds__(BLOCK) k_q3_extract(
+    const int64_t *__restrict__ slot_keys,
+    const int32_t *__restrict__ slot_head,
+    const double *__restrict__ slot_sums, uint64_t cap,
+    int64_t *__restrict__ out_keys, int32_t *__restrict__ out_row,
+    double *__restrict__ out_sums, uint64_t out_cap,
+    uint64_t *__restrict__ cursor) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += stride) {
+    if (slot_keys[s] == QK_JOIN_EMPTY || slot_sums[s] == 0.0) continue;
+    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
+    if (pos < out_cap) {
+      out_keys[pos] = slot_keys[s];
+      out_row[pos] = slot_head[s];
+      out_sums[pos] = slot_sums[s];
+    }
+  }
+}
+extern "C" int qk_q3_extract(void *stream, const int64_t *slot_keys,
+                             const int32_t *slot_head,
+                             const double *slot_sums, uint64_t cap,
+                             int64_t *out_keys, int32_t *out_row,
+                             double *out_sums, uint64_t out_cap,
+                             uint64_t *cursor) {
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_extract, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, slot_keys, slot_head, slot_sums,
+                     cap, out_keys, out_row, out_sums, out_cap, cursor);
+  QK_TRY("qk_q3_extract", hipGetLastError());
   return 0;
 }
 

@@ -65,6 +65,18 @@ _SIGS = {
     "qk_join_build": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp, c_vp, c_u64],
     "qk_join_probe": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_u64,
                       ctypes.c_int, c_vp, c_vp, c_u64, c_vp],
+    "qk_gen_orders": [c_vp, c_u64, c_u64, c_u64, c_i64, c_vp, c_vp, c_vp,
+                      c_vp],
+    "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
+    "qk_build_u8eq": [c_vp, c_u64, c_vp, c_vp, c_u8, c_vp, c_vp, c_u64],
+    "qk_q3_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_vp, c_vp,
+                           c_u64, c_vp, c_vp, c_u64],
+    "qk_q3_count_orders": [c_vp, c_u64, c_vp, c_vp, c_i32, c_vp, c_vp,
+                           c_u64, c_vp],
+    "qk_q3_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
+                        c_vp, c_u64, c_vp, c_vp],
+    "qk_q3_extract": [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
+                      c_u64, c_vp],
     "qk_groupby_i64_sum": [c_vp, c_u64, c_vp, c_vp, ctypes.c_int, c_vp, c_vp,
                            c_u64],
     "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,

@@ -375,3 +375,120 @@ def test_gen_lineitem_device_vs_oracle_shape(gpu):
     assert 0.975 < frac < 0.995
     for c in cols.values():
         c.free()
+
+
+# ---------- fused Q3 path ----------------------------------------------
+
+def test_q3_fused_parity(gpu, data):
+    """Fused build/probe/agg kernels == oracle == generic pipeline."""
+    from quokka_amd import queries as DQ, staging
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipdate",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders)
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
+    full, top10 = DQ.q3_fused(lcols, ocols, ccols)
+    wfull, wtop = OQ.q3(li, orders, cust)
+    og = np.argsort(full["l_orderkey"])
+    ow = np.argsort(wfull["l_orderkey"])
+    assert np.array_equal(full["l_orderkey"][og], wfull["l_orderkey"][ow])
+    np.testing.assert_allclose(full["revenue"][og], wfull["revenue"][ow],
+                               rtol=1e-9)
+    assert np.array_equal(full["o_orderdate"][og], wfull["o_orderdate"][ow])
+    assert np.array_equal(top10["l_orderkey"], wtop["l_orderkey"])
+    np.testing.assert_allclose(top10["revenue"], wtop["revenue"], rtol=1e-9)
+    for cs in (lcols, ocols, ccols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q3_fused_rebuild_idempotent(gpu, data):
+    from quokka_amd import queries as DQ, staging
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipdate",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders)
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
+    st = DQ.Q3Fused(ocols, ccols)
+    st.probe(lcols)
+    f1, _ = st.extract()
+    st.rebuild()
+    st.probe(lcols)
+    f2, _ = st.extract()
+    o1, o2 = np.argsort(f1["l_orderkey"]), np.argsort(f2["l_orderkey"])
+    assert np.array_equal(f1["l_orderkey"][o1], f2["l_orderkey"][o2])
+    np.testing.assert_allclose(f1["revenue"][o1], f2["revenue"][o2],
+                               rtol=1e-12)
+    st.free()
+    for cs in (lcols, ocols, ccols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q3_fused_on_device_generated(gpu):
+    """Device-generated lineitem/orders/customer -> fused Q3 == oracle on
+    the d2h copies of the SAME data (validates the gen kernels' join-key
+    and date-correlation structure)."""
+    from quokka_amd import shim, queries as DQ
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n = 400_000
+    n_ord, n_cust = n // 4, n // 40
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_shipdate", np.int32),
+        ("l_extendedprice", np.float64), ("l_discount", np.float64)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(0), c_u64(5),
+              c_i64(20000), c_i64(1000), c_i64(n_ord),
+              li["l_orderkey"].ptr, None, None,
+              li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
+              None, None, li["l_shipdate"].ptr)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_custkey", np.int64),
+        ("o_orderdate", np.int32), ("o_shippriority", np.int32)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(5),
+              c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
+              od["o_orderdate"].ptr, od["o_shippriority"].ptr)
+    cu = {"c_custkey": DevColumn(np.int64, n_cust),
+          "c_mktsegment": DevColumn(np.uint8, n_cust)}
+    shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(5),
+              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr)
+    li_h = {k: v.to_numpy() for k, v in li.items()}
+    od_h = {k: v.to_numpy() for k, v in od.items()}
+    cu_h = {k: v.to_numpy() for k, v in cu.items()}
+    # shipdate must correlate with the order's orderdate (1..121 ahead)
+    omap = od_h["o_orderdate"][(li_h["l_orderkey"] - 1).astype(np.int64)]
+    delta = li_h["l_shipdate"] - omap
+    assert delta.min() >= 1 and delta.max() <= 121
+    full, top10 = DQ.q3_fused(li, od, cu)
+    wfull, wtop = OQ.q3(li_h, od_h, cu_h)
+    og = np.argsort(full["l_orderkey"])
+    ow = np.argsort(wfull["l_orderkey"])
+    assert np.array_equal(full["l_orderkey"][og], wfull["l_orderkey"][ow])
+    np.testing.assert_allclose(full["revenue"][og], wfull["revenue"][ow],
+                               rtol=1e-9)
+    assert np.array_equal(top10["l_orderkey"], wtop["l_orderkey"])
+    for cs in (li, od, cu):
+        for c in cs.values():
+            c.free()
+
+
+def test_q3_fused_edge_empty(gpu):
+    """No BUILDING customers -> empty result; empty lineitem -> empty."""
+    from quokka_amd import queries as DQ, staging
+    od = staging.stage_columns({
+        "o_orderkey": np.arange(1, 101, dtype=np.int64),
+        "o_custkey": np.ones(100, dtype=np.int64),
+        "o_orderdate": np.full(100, 9000, np.int32),
+        "o_shippriority": np.zeros(100, np.int32)})
+    cu = staging.stage_columns({
+        "c_custkey": np.arange(1, 11, dtype=np.int64),
+        "c_mktsegment": np.zeros(10, np.uint8)})  # none == BUILDING(1)
+    li = staging.stage_columns({
+        "l_orderkey": np.arange(1, 101, dtype=np.int64),
+        "l_shipdate": np.full(100, 9500, np.int32),
+        "l_extendedprice": np.full(100, 10.0),
+        "l_discount": np.zeros(100)})
+    full, top = DQ.q3_fused(li, od, cu)
+    assert len(full["l_orderkey"]) == 0
+    for cs in (od, cu, li):
+        for c in cs.values():
+            c.free()
