@@ -183,11 +183,13 @@ class DevColumn:
         return out
 
     def gather(self, idx_col, n_idx, stream=None):
-        """New column: self[idx] for a device u32 index column."""
+        """New column: self[idx] for a device u32 index column. `stream`
+        may be a Stream object, a raw handle, or None."""
         fn = _DTYPE_GATHER[self.dtype]
+        sh = stream.handle if isinstance(stream, Stream) else stream
         out = DevColumn(self.dtype, n_idx)
         if n_idx:
-            call(fn, stream, c_u64(n_idx), idx_col.ptr, self.ptr, out.ptr)
+            call(fn, sh, c_u64(n_idx), idx_col.ptr, self.ptr, out.ptr)
         return out
 
     def free(self):

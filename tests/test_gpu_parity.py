@@ -387,7 +387,10 @@ def test_q3_fused_parity(gpu, data):
                                              "l_extendedprice", "l_discount"])
     ocols = staging.stage_columns(orders)
     ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
-    full, top10 = DQ.q3_fused(lcols, ocols, ccols)
+    from quokka_amd import shim as _shim
+    stream = _shim.Stream()   # exercise the non-default-stream path
+    full, top10 = DQ.q3_fused(lcols, ocols, ccols, stream=stream)
+    stream.destroy()
     wfull, wtop = OQ.q3(li, orders, cust)
     og = np.argsort(full["l_orderkey"])
     ow = np.argsort(wfull["l_orderkey"])
