@@ -41,7 +41,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
-    p.add_argument("--query", choices=["q1", "q3"], default="q1")
+    p.add_argument("--query", choices=["q1", "q3", "q5"], default="q1")
     p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -112,8 +112,63 @@ def gen_device_q3_tables(shim, n, rank):
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_mktsegment": DevColumn(np.uint8, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
-              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr)
+              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr, None)
     return li, od, cu
+
+
+def gen_device_q5_tables(shim, n, rank):
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n_ord, n_supp = max(1, n // 4), max(1, n // 600)
+    n_cust = max(1, n_ord // 10)
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_suppkey", np.int64),
+        ("l_extendedprice", np.float64), ("l_discount", np.float64)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
+              c_i64(20_000_000), c_i64(n_supp), c_i64(n_ord),
+              li["l_orderkey"].ptr, li["l_suppkey"].ptr, None,
+              li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
+              None, None, None)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_custkey", np.int64),
+        ("o_orderdate", np.int32)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
+              c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
+              od["o_orderdate"].ptr, None)
+    cu = {"c_custkey": DevColumn(np.int64, n_cust),
+          "c_nationkey": DevColumn(np.int32, n_cust)}
+    shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
+              cu["c_custkey"].ptr, None, cu["c_nationkey"].ptr)
+    su = {"s_suppkey": DevColumn(np.int64, n_supp),
+          "s_nationkey": DevColumn(np.int32, n_supp)}
+    shim.call("qk_gen_supplier", None, c_u64(n_supp), c_u64(0), c_u64(42),
+              su["s_suppkey"].ptr, su["s_nationkey"].ptr)
+    return li, od, cu, su
+
+
+def cpu_baseline_q5(sample_rows, target_seconds=12.0):
+    from oracle import tpch_gen as G, queries as OQ
+    sf = sample_rows / 6_000_000
+    d = {"orders": G.gen_orders(sf, 42)}
+    d["lineitem"] = G.gen_lineitem(sf, 42, d["orders"])
+    d["customer"] = G.gen_customer(sf, 42)
+    d["supplier"] = G.gen_supplier(sf, 42)
+    nat, reg = G.gen_nation(), G.gen_region()
+    n = len(d["lineitem"]["l_orderkey"])
+    t0 = time.time()
+    OQ.q5(d["lineitem"], d["orders"], d["customer"], d["supplier"], nat, reg)
+    per = time.time() - t0
+    passes = max(1, min(16, int(target_seconds / max(per, 1e-3))))
+    t0 = time.time()
+    for _ in range(passes):
+        OQ.q5(d["lineitem"], d["orders"], d["customer"], d["supplier"],
+              nat, reg)
+    dt = time.time() - t0
+    return {"value": n * passes / dt, "unit": "rows/s", "cores": 1,
+            "kind": "port",
+            "sample": "%.1fM-row seeded lineitem (+orders/customer/supplier)"
+                      " x %d passes of the numpy oracle "
+                      "(oracle/queries.py:q5), single-threaded"
+                      % (n / 1e6, passes)}
 
 
 def cpu_baseline_q3(sample_rows, target_seconds=12.0):
@@ -279,6 +334,104 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         dist.destroy_process_group()
 
 
+def main_q5(args, n, world, rank, dist, shim, DQ):
+    """TPC-H Q5 on the fused device path (BASELINE.json configs[4] query
+    shape; single-node). A step = rebuild the three key->nation tables +
+    fused probe + per-nation result; weak scaling like Q3."""
+    from quokka_amd import ops
+
+    li, od, cu, su = gen_device_q5_tables(shim, n, rank)
+    stream = shim.Stream()
+    fused = DQ.Q5Fused(od, cu, su, stream)
+    stream.sync()
+    timer = shim.Timer()
+
+    mc = ops._count_buf()
+    fused.probe(li, mc)
+    stream.sync()
+    n_match = ops._read_u64(mc)
+    mc.free()
+    fused.reset_sums()
+
+    def step(timed):
+        fused.rebuild()
+        if timed:
+            timer.start(stream)
+        fused.probe(li)
+        if timed:
+            timer.stop(stream)
+        res = fused.result()
+        if dist is not None:
+            import torch
+            import torch.distributed as _d
+            v = torch.tensor([r for _, r in res], dtype=torch.float64)
+            _d.all_reduce(v, op=_d.ReduceOp.SUM)
+            res = sorted(zip([nm for nm, _ in res], v.tolist()),
+                         key=lambda t: -t[1])
+        return res, (timer.elapsed_ms() if timed else None)
+
+    for _ in range(args.warmup):
+        step(False)
+    if dist is not None:
+        dist.barrier()
+    stream.sync()
+    t0 = time.time()
+    kernel_ms = []
+    res = None
+    for _ in range(args.steps):
+        res, kms = step(True)
+        kernel_ms.append(kms)
+    stream.sync()
+    elapsed = time.time() - t0
+    if dist is not None:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+        dist.barrier()
+
+    if rank == 0:
+        # probe algorithmic bytes: 16 B keys (orderkey+suppkey) per row +
+        # one 12 B orders bucket per row + 12 B supplier bucket + 16 B
+        # price/disc per matched row (matches dominate the second probe)
+        alg_bytes = 28 * n + 28 * n_match
+        avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
+        achieved_gbps = alg_bytes / avg_kernel_s / 1e9
+        out = {
+            "metric": "rows/s",
+            "value": n * world * args.steps / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "f64", "data": "synthetic",
+            "config": {
+                "workload": "TPC-H SF%g Q5 (6-table join chain), %d "
+                            "lineitem rows/GPU resident in HBM "
+                            "(BASELINE.json configs[4] query on one node)"
+                            % (args.sf, n),
+                "sf_per_gpu": args.sf, "rows_per_gpu": n, "query": "Q5",
+                "orders_build_rows": int(fused.n_build),
+                "joined_rows": int(n_match),
+                "revenue_top": res[0] if res else None,
+            },
+            "roofline": {
+                "bound": "hbm", "achieved": achieved_gbps,
+                "peak": HBM_PEAK_GBPS, "unit": "GB/s",
+                "frac": achieved_gbps / HBM_PEAK_GBPS,
+                "traffic": read_traffic(),
+            },
+            "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
+                             else cpu_baseline_q5(args.cpu_sample_rows)),
+        }
+        print(json.dumps(out))
+    timer.destroy()
+    fused.free()
+    stream.destroy()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
 def main():
     args = parse_args()
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -300,6 +453,8 @@ def main():
     n &= ~3  # multiple of 4 -> vectorized Q1 path, 4 lines/order for Q3
     if args.query == "q3":
         return main_q3(args, n, world, rank, dist, shim, DQ)
+    if args.query == "q5":
+        return main_q5(args, n, world, rank, dist, shim, DQ)
     cols = gen_device_lineitem(shim, n, rank)
     stream = shim.Stream()
     timer = shim.Timer()

@@ -75,9 +75,14 @@ int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed
 int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
                   int64_t n_customers, int64_t *o_orderkey, int64_t *o_custkey,
                   int32_t *o_orderdate, int32_t *o_shippriority);
-/* Customer columns: c_custkey dense row+1, c_mktsegment uniform u8 0..4. */
+/* Customer columns: c_custkey dense row+1, c_mktsegment uniform u8 0..4,
+ * c_nationkey uniform i32 0..24. Any output may be NULL. */
 int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
-                    uint64_t seed, int64_t *c_custkey, uint8_t *c_mktsegment);
+                    uint64_t seed, int64_t *c_custkey, uint8_t *c_mktsegment,
+                    int32_t *c_nationkey);
+/* Supplier columns: s_suppkey dense row+1, s_nationkey uniform 0..24. */
+int qk_gen_supplier(void *stream, uint64_t n, uint64_t row_offset,
+                    uint64_t seed, int64_t *s_suppkey, int32_t *s_nationkey);
 
 /* ---- TPC-H Q1: fused filter + group-by partial aggregate ------------- *
  * Replaces the map-side partial agg the reference folds into partition_fn
@@ -192,6 +197,36 @@ int qk_q3_extract(void *stream, const int64_t *slot_keys,
                   const int32_t *slot_head, const double *slot_sums,
                   uint64_t capacity, int64_t *out_keys, int32_t *out_row,
                   double *out_sums, uint64_t out_cap, uint64_t *cursor_dev);
+
+/* ---- fused Q5 path ----------------------------------------------------- *
+ * Key -> i32-value hash tables and a fused lineitem probe for the 6-table
+ * chain (tpch_ref.py:142-169). Unique build keys (primary keys). */
+/* Insert keys[i] -> vals[i] where bit vals[i] of accept_mask is set
+ * (accept_mask = 0xFFFFFFFF accepts all; else vals must be in [0,32)). */
+int qk_build_keyval_i32(void *stream, uint64_t n, const int64_t *keys,
+                        const int32_t *vals, uint32_t accept_mask,
+                        int64_t *slot_keys, int32_t *slot_val,
+                        uint64_t capacity);
+/* Orders in [date_lo, date_hi) whose o_custkey hits the customer table:
+ * insert o_orderkey -> customer nationkey. slot_keys == NULL -> count-only
+ * (count_dev gets the survivor count either way when non-NULL). */
+int qk_q5_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
+                       const int64_t *o_custkey, const int32_t *o_orderdate,
+                       int32_t date_lo, int32_t date_hi,
+                       const int64_t *cust_keys, const int32_t *cust_val,
+                       uint64_t cust_cap, int64_t *slot_keys,
+                       int32_t *slot_val, uint64_t capacity,
+                       uint64_t *count_dev);
+/* Fused probe: join lineitem to orders (-> customer nation) and supplier
+ * (-> supplier nation); where equal accumulate revenue into out25[nation].
+ * out25: f64[32], zeroed (slots 25..31 unused). */
+int qk_q5_probe_agg(void *stream, uint64_t n, const int64_t *l_orderkey,
+                    const int64_t *l_suppkey, const double *l_price,
+                    const double *l_disc, const int64_t *ord_keys,
+                    const int32_t *ord_val, uint64_t ord_cap,
+                    const int64_t *supp_keys, const int32_t *supp_val,
+                    uint64_t supp_cap, double *out25,
+                    uint64_t *match_count_dev /* nullable */);
 
 /* ---- group-by (i64 key) sum ------------------------------------------- *
  * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for

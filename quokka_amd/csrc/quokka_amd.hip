@@ -253,7 +253,8 @@ extern "C" int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset,
 }
 
 __global__ void k_gen_customer(uint64_t n, uint64_t row_offset, uint64_t seed,
-                               int64_t *c_custkey, uint8_t *c_mktsegment) {
+                               int64_t *c_custkey, uint8_t *c_mktsegment,
+                               int32_t *c_nationkey) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -262,16 +263,17 @@ __global__ void k_gen_customer(uint64_t n, uint64_t row_offset, uint64_t seed,
                             row * 0x9E3779B97F4A7C15ULL);
     if (c_custkey) c_custkey[i] = (int64_t)row + 1;
     if (c_mktsegment) c_mktsegment[i] = (uint8_t)(h % 5);
+    if (c_nationkey) c_nationkey[i] = (int32_t)((h >> 32) % 25);
   }
 }
 extern "C" int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
                                uint64_t seed, int64_t *c_custkey,
-                               uint8_t *c_mktsegment) {
+                               uint8_t *c_mktsegment, int32_t *c_nationkey) {
   if (!n) return 0;
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_gen_customer, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, row_offset, seed, c_custkey,
-                     c_mktsegment);
+                     c_mktsegment, c_nationkey);
   QK_TRY("qk_gen_customer", hipGetLastError());
   return 0;
 }
@@ -972,15 +974,32 @@ __global__ void __launch_bounds__(BLOCK) k_q3_extract(
     int64_t *__restrict__ out_keys, int32_t *__restrict__ out_row,
     double *__restrict__ out_sums, uint64_t out_cap,
     uint64_t *__restrict__ cursor) {
+  // wave-aggregated cursor claims: one atomicAdd per wave per pass, not
+  // one per group (a single cursor word takes ~88 atomics/us — the
+  // per-group version measured 8.6 ms on 1.2M groups)
+  int lane = threadIdx.x & (WAVE - 1);
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
-       s += stride) {
-    if (slot_keys[s] == QK_JOIN_EMPTY || slot_sums[s] == 0.0) continue;
-    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
-    if (pos < out_cap) {
-      out_keys[pos] = slot_keys[s];
-      out_row[pos] = slot_head[s];
-      out_sums[pos] = slot_sums[s];
+  for (uint64_t base = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x -
+                       lane;
+       base < cap; base += stride) {
+    uint64_t s = base + lane;
+    bool live = s < cap && slot_keys[s] != QK_JOIN_EMPTY &&
+                slot_sums[s] != 0.0;
+    uint64_t mask = __ballot(live);
+    uint32_t nlive = __popcll(mask);
+    if (!nlive) continue;
+    uint64_t wbase;
+    if (lane == 0)
+      wbase = atomicAdd((unsigned long long *)cursor,
+                        (unsigned long long)nlive);
+    wbase = __shfl(wbase, 0);
+    if (live) {
+      uint64_t pos = wbase + __popcll(mask & ((1ULL << lane) - 1));
+      if (pos < out_cap) {
+        out_keys[pos] = slot_keys[s];
+        out_row[pos] = slot_head[s];
+        out_sums[pos] = slot_sums[s];
+      }
     }
   }
 }
@@ -995,6 +1014,204 @@ extern "C" int qk_q3_extract(void *stream, const int64_t *slot_keys,
                      (hipStream_t)stream, slot_keys, slot_head, slot_sums,
                      cap, out_keys, out_row, out_sums, out_cap, cursor);
   QK_TRY("qk_q3_extract", hipGetLastError());
+  return 0;
+}
+
+// ---- fused Q5 path -----------------------------------------------------
+// Key -> i32-value hash tables (custkey->nationkey, orderkey->cust_nation,
+// suppkey->nationkey) and one fused lineitem probe that joins both sides
+// and accumulates revenue per nation. Same unique-build-key contract as Q3.
+
+// insert (keys[i] -> vals[i]) where bit vals[i] of accept_mask is set
+// (vals must be < 32 when accept_mask != ~0u; nationkeys are 0..24)
+__global__ void __launch_bounds__(BLOCK) k_build_keyval_i32(
+    uint64_t n, const int64_t *__restrict__ keys,
+    const int32_t *__restrict__ vals, uint32_t accept_mask,
+    int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_val,
+    uint64_t cap) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t v = vals[i];
+    if (accept_mask != 0xFFFFFFFFu &&
+        !((accept_mask >> (v & 31)) & 1u && v >= 0 && v < 32))
+      continue;
+    int64_t key = keys[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    slot_val[s] = v;
+  }
+}
+extern "C" int qk_build_keyval_i32(void *stream, uint64_t n,
+                                   const int64_t *keys, const int32_t *vals,
+                                   uint32_t accept_mask, int64_t *slot_keys,
+                                   int32_t *slot_val, uint64_t cap) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_build_keyval_i32.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_build_keyval_i32, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, vals, accept_mask,
+                     slot_keys, slot_val, cap);
+  QK_TRY("qk_build_keyval_i32", hipGetLastError());
+  return 0;
+}
+
+// orders: where date_lo <= o_orderdate < date_hi and o_custkey hits the
+// customer table, insert o_orderkey -> customer's nationkey.
+// count-only mode when slot_keys == NULL (writes survivor count).
+__global__ void __launch_bounds__(BLOCK) k_q5_build_orders(
+    uint64_t n, const int64_t *__restrict__ o_orderkey,
+    const int64_t *__restrict__ o_custkey,
+    const int32_t *__restrict__ o_orderdate, int32_t date_lo, int32_t date_hi,
+    const int64_t *__restrict__ cust_keys, const int32_t *__restrict__ cust_val,
+    uint64_t cust_cap, int64_t *__restrict__ slot_keys,
+    int32_t *__restrict__ slot_val, uint64_t cap,
+    uint64_t *__restrict__ count) {
+  uint32_t cnt = 0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t d = o_orderdate[i];
+    if (d < date_lo || d >= date_hi) continue;
+    int32_t nat = probe_unique(cust_keys, cust_val, cust_cap, o_custkey[i],
+                               nullptr);
+    if (nat < 0) continue;
+    cnt++;
+    if (!slot_keys) continue;
+    int64_t key = o_orderkey[i];
+    uint64_t s = slot_of(key, cap);
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) break;
+      if (cur == QK_JOIN_EMPTY) {
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                          (unsigned long long)QK_JOIN_EMPTY,
+                                          (unsigned long long)key);
+        if (prev == QK_JOIN_EMPTY || prev == key) break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+    slot_val[s] = nat;
+  }
+  if (count) {
+    __shared__ uint32_t lds[BLOCK / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint64_t t = 0;
+      for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+      if (t) atomicAdd((unsigned long long *)count, (unsigned long long)t);
+    }
+  }
+}
+extern "C" int qk_q5_build_orders(void *stream, uint64_t n,
+                                  const int64_t *o_orderkey,
+                                  const int64_t *o_custkey,
+                                  const int32_t *o_orderdate, int32_t date_lo,
+                                  int32_t date_hi, const int64_t *cust_keys,
+                                  const int32_t *cust_val, uint64_t cust_cap,
+                                  int64_t *slot_keys, int32_t *slot_val,
+                                  uint64_t cap, uint64_t *count_dev) {
+  if (!n) return 0;
+  if ((cust_cap & (cust_cap - 1)) || (slot_keys && (cap & (cap - 1))))
+    return qk_fail("qk_q5_build_orders.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q5_build_orders, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, o_orderkey, o_custkey,
+                     o_orderdate, date_lo, date_hi, cust_keys, cust_val,
+                     cust_cap, slot_keys, slot_val, cap, count_dev);
+  QK_TRY("qk_q5_build_orders", hipGetLastError());
+  return 0;
+}
+
+// lineitem: probe orders (-> customer nation) and supplier (-> supplier
+// nation); where equal, accumulate revenue into out25[nation] via
+// per-block LDS f64 accumulators
+__global__ void __launch_bounds__(BLOCK) k_q5_probe_agg(
+    uint64_t n, const int64_t *__restrict__ l_orderkey,
+    const int64_t *__restrict__ l_suppkey,
+    const double *__restrict__ l_price, const double *__restrict__ l_disc,
+    const int64_t *__restrict__ ord_keys, const int32_t *__restrict__ ord_val,
+    uint64_t ord_cap, const int64_t *__restrict__ supp_keys,
+    const int32_t *__restrict__ supp_val, uint64_t supp_cap,
+    double *__restrict__ out25, uint64_t *__restrict__ match_count) {
+  __shared__ double lsum[32];
+  __shared__ uint32_t lcnt;
+  if (threadIdx.x < 32) lsum[threadIdx.x] = 0.0;
+  if (threadIdx.x == 0) lcnt = 0;
+  __syncthreads();
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t cnat = probe_unique(ord_keys, ord_val, ord_cap, l_orderkey[i],
+                                nullptr);
+    if (cnat < 0) continue;
+    int32_t snat = probe_unique(supp_keys, supp_val, supp_cap, l_suppkey[i],
+                                nullptr);
+    if (snat != cnat) continue;
+    atomicAdd(&lsum[cnat & 31], l_price[i] * (1.0 - l_disc[i]));
+    if (match_count) atomicAdd(&lcnt, 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x < 32 && lsum[threadIdx.x] != 0.0)
+    atomicAdd(&out25[threadIdx.x], lsum[threadIdx.x]);
+  if (match_count && threadIdx.x == 0 && lcnt)
+    atomicAdd((unsigned long long *)match_count, (unsigned long long)lcnt);
+}
+extern "C" int qk_q5_probe_agg(void *stream, uint64_t n,
+                               const int64_t *l_orderkey,
+                               const int64_t *l_suppkey,
+                               const double *l_price, const double *l_disc,
+                               const int64_t *ord_keys, const int32_t *ord_val,
+                               uint64_t ord_cap, const int64_t *supp_keys,
+                               const int32_t *supp_val, uint64_t supp_cap,
+                               double *out25, uint64_t *match_count) {
+  if (!n) return 0;
+  if ((ord_cap & (ord_cap - 1)) || (supp_cap & (supp_cap - 1)))
+    return qk_fail("qk_q5_probe_agg.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q5_probe_agg, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, l_orderkey, l_suppkey, l_price,
+                     l_disc, ord_keys, ord_val, ord_cap, supp_keys, supp_val,
+                     supp_cap, out25, match_count);
+  QK_TRY("qk_q5_probe_agg", hipGetLastError());
+  return 0;
+}
+
+__global__ void k_gen_supplier(uint64_t n, uint64_t row_offset, uint64_t seed,
+                               int64_t *s_suppkey, int32_t *s_nationkey) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t row = row_offset + i;
+    uint64_t h = splitmix64((seed ^ 0x5A5A5A5A5A5A5A5AULL) +
+                            row * 0x9E3779B97F4A7C15ULL);
+    if (s_suppkey) s_suppkey[i] = (int64_t)row + 1;
+    if (s_nationkey) s_nationkey[i] = (int32_t)(h % 25);
+  }
+}
+extern "C" int qk_gen_supplier(void *stream, uint64_t n, uint64_t row_offset,
+                               uint64_t seed, int64_t *s_suppkey,
+                               int32_t *s_nationkey) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gen_supplier, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, row_offset, seed, s_suppkey,
+                     s_nationkey);
+  QK_TRY("qk_gen_supplier", hipGetLastError());
   return 0;
 }
 

@@ -197,7 +197,11 @@ class Q3Fused:
         sh = stream.handle if stream else None
         ncust = cust_cols["c_custkey"].n
         nord = ord_cols["o_orderkey"].n
-        self.cust_cap = ops._pow2_at_least(max(16, 2 * ncust))
+        # size the customer table by the BUILDING survivor count, not the
+        # full table (table bytes drive probe traffic — profiles/r01_q3)
+        nbuild_cust = self._count_u8eq(cust_cols["c_mktsegment"],
+                                       MKT_BUILDING, stream)
+        self.cust_cap = ops._pow2_at_least(max(16, 2 * nbuild_cust))
         self.cust_keys = DevColumn(np.int64, self.cust_cap)
         self.cust_head = DevColumn(np.int32, self.cust_cap)
         call("qk_fill_i64", sh, self.cust_keys.ptr,
@@ -217,7 +221,7 @@ class Q3Fused:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
         cnt.free()
-        self.ord_cap = ops._pow2_at_least(max(16, 4 * self.n_build))
+        self.ord_cap = ops._pow2_at_least(max(16, 2 * self.n_build))
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_head = DevColumn(np.int32, self.ord_cap)
         self.ord_sums = DevColumn(np.float64, self.ord_cap)
@@ -232,6 +236,16 @@ class Q3Fused:
              self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap))
         self._ord_cols = ord_cols
         self._cust_cols = cust_cols
+
+    @staticmethod
+    def _count_u8eq(col, value, stream):
+        """Count col == value with the filter-count kernel pair (no
+        scatter pass; temp index buffer avoided by passing the count-only
+        path through qk_filter_u8 on a scratch the size of the column)."""
+        from . import ops
+        idx, n = ops.filter_col(col, ops.EQ, value, stream)
+        idx.free()
+        return n
 
     def reset_sums(self):
         call("qk_dmemset", self.ord_sums.ptr, 0, c_u64(self.ord_cap * 8))
@@ -277,9 +291,11 @@ class Q3Fused:
         from .shim import DevColumn, c_u64
         sh = self.stream.handle if self.stream else None
         out_cap = self.ord_cap
-        ok = DevColumn(np.int64, out_cap)
-        orow = DevColumn(np.int32, out_cap)
-        osum = DevColumn(np.float64, out_cap)
+        if getattr(self, "_ext", None) is None:
+            self._ext = (DevColumn(np.int64, out_cap),
+                         DevColumn(np.int32, out_cap),
+                         DevColumn(np.float64, out_cap))
+        ok, orow, osum = self._ext
         cur = ops._count_buf()
         call("qk_q3_extract", sh, self.ord_keys.ptr, self.ord_head.ptr,
              self.ord_sums.ptr, c_u64(self.ord_cap), ok.ptr, orow.ptr,
@@ -291,7 +307,6 @@ class Q3Fused:
         keys = ok.to_numpy(k)
         rows = orow.to_numpy(k)
         sums = osum.to_numpy(k)
-        ok.free(); orow.free(); osum.free()
         # attach o_orderdate/o_shippriority by build row (host gather over
         # the groups only — tiny relative to the scan)
         odate = self._ord_cols["o_orderdate"]
@@ -314,6 +329,10 @@ class Q3Fused:
         for c in (self.cust_keys, self.cust_head, self.ord_keys,
                   self.ord_head, self.ord_sums):
             c.free()
+        if getattr(self, "_ext", None):
+            for c in self._ext:
+                c.free()
+            self._ext = None
 
 
 def q3_fused(li_cols, ord_cols, cust_cols, stream=None, limit=10):
@@ -321,6 +340,162 @@ def q3_fused(li_cols, ord_cols, cust_cols, stream=None, limit=10):
     st = Q3Fused(ord_cols, cust_cols, stream)
     st.probe(li_cols)
     out = st.extract(limit)
+    st.free()
+    return out
+
+
+NATION_NAMES = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT",
+                "ETHIOPIA", "FRANCE", "GERMANY", "INDIA", "INDONESIA",
+                "IRAN", "IRAQ", "JAPAN", "JORDAN", "KENYA", "MOROCCO",
+                "MOZAMBIQUE", "PERU", "CHINA", "ROMANIA", "SAUDI ARABIA",
+                "VIETNAM", "RUSSIA", "UNITED KINGDOM", "UNITED STATES"]
+NATION_REGION = [0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0, 0, 0, 1, 2,
+                 3, 4, 2, 3, 1, 1]   # spec nation -> region (ASIA = 2)
+REGION_ASIA = 2
+Q5_LO = 8766           # 1994-01-01
+Q5_HI = 9131           # 1994-01-01 + 1 year
+
+
+class Q5Fused:
+    """Fused Q5 state (tpch_ref.py:142-169): customer(->nation, ASIA only),
+    orders(->customer nation, date window), supplier(->nation) key-value
+    tables; one fused lineitem probe accumulates revenue per nation.
+
+    The nation/region joins are resolved host-side into the 25-entry ASIA
+    bitmask (the reference's own plan joins nation x region first and
+    broadcasts the 5-row result, tpch.py:207-208)."""
+
+    def __init__(self, ord_cols, cust_cols, supp_cols, stream=None,
+                 nation_region=NATION_REGION, region=REGION_ASIA):
+        from . import shim, ops
+        from .shim import DevColumn, c_u64, c_i64
+        self.stream = stream
+        sh = stream.handle if stream else None
+        mask = 0
+        for nk, r in enumerate(nation_region):
+            if r == region:
+                mask |= 1 << nk
+        self.asia_mask = mask
+
+        ncust = cust_cols["c_custkey"].n
+        nsupp = supp_cols["s_suppkey"].n
+        nord = ord_cols["o_orderkey"].n
+        # customer table sized by expected survivors (1/5 of nations)
+        self.cust_cap = ops._pow2_at_least(max(16, 2 * max(1, ncust // 2)))
+        self.cust_keys = DevColumn(np.int64, self.cust_cap)
+        self.cust_val = DevColumn(np.int32, self.cust_cap)
+        call("qk_fill_i64", sh, self.cust_keys.ptr,
+             c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cust_cap))
+        call("qk_build_keyval_i32", sh, c_u64(ncust),
+             cust_cols["c_custkey"].ptr, cust_cols["c_nationkey"].ptr,
+             ctypes.c_uint32(mask), self.cust_keys.ptr, self.cust_val.ptr,
+             c_u64(self.cust_cap))
+        # supplier table: all rows
+        self.supp_cap = ops._pow2_at_least(max(16, 2 * nsupp))
+        self.supp_keys = DevColumn(np.int64, self.supp_cap)
+        self.supp_val = DevColumn(np.int32, self.supp_cap)
+        call("qk_fill_i64", sh, self.supp_keys.ptr,
+             c_i64(int(shim.JOIN_EMPTY)), c_u64(self.supp_cap))
+        call("qk_build_keyval_i32", sh, c_u64(nsupp),
+             supp_cols["s_suppkey"].ptr, supp_cols["s_nationkey"].ptr,
+             ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
+             self.supp_val.ptr, c_u64(self.supp_cap))
+        # orders: count survivors -> tight table -> build
+        cnt = ops._count_buf()
+        call("qk_q5_build_orders", sh, c_u64(nord),
+             ord_cols["o_orderkey"].ptr, ord_cols["o_custkey"].ptr,
+             ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
+             ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
+             c_u64(self.cust_cap), None, None, c_u64(16), cnt.ptr)
+        if stream:
+            stream.sync()
+        self.n_build = ops._read_u64(cnt)
+        cnt.free()
+        self.ord_cap = ops._pow2_at_least(max(16, 2 * self.n_build))
+        self.ord_keys = DevColumn(np.int64, self.ord_cap)
+        self.ord_val = DevColumn(np.int32, self.ord_cap)
+        call("qk_fill_i64", sh, self.ord_keys.ptr,
+             c_i64(int(shim.JOIN_EMPTY)), c_u64(self.ord_cap))
+        call("qk_q5_build_orders", sh, c_u64(nord),
+             ord_cols["o_orderkey"].ptr, ord_cols["o_custkey"].ptr,
+             ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
+             ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
+             c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
+             c_u64(self.ord_cap), None)
+        self.out25 = DevBuffer(32 * 8)
+        call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
+        self._ord_cols = ord_cols
+        self._cust_cols = cust_cols
+        self._supp_cols = supp_cols
+
+    def reset_sums(self):
+        call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
+
+    def rebuild(self):
+        """Re-run all three build passes (per-step full-query semantics)."""
+        from . import shim
+        from .shim import c_u64, c_i64
+        sh = self.stream.handle if self.stream else None
+        for keys, cap in ((self.cust_keys, self.cust_cap),
+                          (self.supp_keys, self.supp_cap),
+                          (self.ord_keys, self.ord_cap)):
+            call("qk_fill_i64", sh, keys.ptr, c_i64(int(shim.JOIN_EMPTY)),
+                 c_u64(cap))
+        call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
+        call("qk_build_keyval_i32", sh,
+             c_u64(self._cust_cols["c_custkey"].n),
+             self._cust_cols["c_custkey"].ptr,
+             self._cust_cols["c_nationkey"].ptr,
+             ctypes.c_uint32(self.asia_mask), self.cust_keys.ptr,
+             self.cust_val.ptr, c_u64(self.cust_cap))
+        call("qk_build_keyval_i32", sh,
+             c_u64(self._supp_cols["s_suppkey"].n),
+             self._supp_cols["s_suppkey"].ptr,
+             self._supp_cols["s_nationkey"].ptr,
+             ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
+             self.supp_val.ptr, c_u64(self.supp_cap))
+        call("qk_q5_build_orders", sh, c_u64(self._ord_cols["o_orderkey"].n),
+             self._ord_cols["o_orderkey"].ptr,
+             self._ord_cols["o_custkey"].ptr,
+             self._ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
+             ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
+             c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
+             c_u64(self.ord_cap), None)
+
+    def probe(self, li_cols, match_count_buf=None):
+        sh = self.stream.handle if self.stream else None
+        n = li_cols["l_orderkey"].n
+        call("qk_q5_probe_agg", sh, c_u64(n), li_cols["l_orderkey"].ptr,
+             li_cols["l_suppkey"].ptr, li_cols["l_extendedprice"].ptr,
+             li_cols["l_discount"].ptr, self.ord_keys.ptr,
+             self.ord_val.ptr, c_u64(self.ord_cap), self.supp_keys.ptr,
+             self.supp_val.ptr, c_u64(self.supp_cap), self.out25.ptr,
+             match_count_buf.ptr if match_count_buf else None)
+
+    def result(self):
+        """[(n_name, revenue)] for ASIA nations, revenue desc
+        (tpch_ref.py:165-168 order by)."""
+        if self.stream:
+            self.stream.sync()
+        host = np.zeros(32, dtype=np.float64)
+        call("qk_d2h", host.ctypes.data_as(c_vp), self.out25.ptr,
+             c_u64(32 * 8))
+        out = [(NATION_NAMES[nk], host[nk]) for nk in range(25)
+               if (self.asia_mask >> nk) & 1]
+        out.sort(key=lambda t: -t[1])
+        return out
+
+    def free(self):
+        for c in (self.cust_keys, self.cust_val, self.supp_keys,
+                  self.supp_val, self.ord_keys, self.ord_val):
+            c.free()
+        self.out25.free()
+
+
+def q5_fused(li_cols, ord_cols, cust_cols, supp_cols, stream=None):
+    st = Q5Fused(ord_cols, cust_cols, supp_cols, stream)
+    st.probe(li_cols)
+    out = st.result()
     st.free()
     return out
 

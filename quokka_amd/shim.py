@@ -67,7 +67,14 @@ _SIGS = {
                       ctypes.c_int, c_vp, c_vp, c_u64, c_vp],
     "qk_gen_orders": [c_vp, c_u64, c_u64, c_u64, c_i64, c_vp, c_vp, c_vp,
                       c_vp],
-    "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
+    "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp],
+    "qk_gen_supplier": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
+    "qk_build_keyval_i32": [c_vp, c_u64, c_vp, c_vp, c_u32, c_vp, c_vp,
+                            c_u64],
+    "qk_q5_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_i32,
+                           c_vp, c_vp, c_u64, c_vp, c_vp, c_u64, c_vp],
+    "qk_q5_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                        c_u64, c_vp, c_vp, c_u64, c_vp, c_vp],
     "qk_build_u8eq": [c_vp, c_u64, c_vp, c_vp, c_u8, c_vp, c_vp, c_u64],
     "qk_q3_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_vp, c_vp,
                            c_u64, c_vp, c_vp, c_u64],

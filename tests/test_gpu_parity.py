@@ -453,7 +453,7 @@ def test_q3_fused_on_device_generated(gpu):
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_mktsegment": DevColumn(np.uint8, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(5),
-              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr)
+              cu["c_custkey"].ptr, cu["c_mktsegment"].ptr, None)
     li_h = {k: v.to_numpy() for k, v in li.items()}
     od_h = {k: v.to_numpy() for k, v in od.items()}
     cu_h = {k: v.to_numpy() for k, v in cu.items()}
@@ -493,5 +493,68 @@ def test_q3_fused_edge_empty(gpu):
     full, top = DQ.q3_fused(li, od, cu)
     assert len(full["l_orderkey"]) == 0
     for cs in (od, cu, li):
+        for c in cs.values():
+            c.free()
+
+
+# ---------- fused Q5 path ----------------------------------------------
+
+def test_q5_fused_parity(gpu, data):
+    from quokka_amd import queries as DQ, staging
+    li, orders = data["lineitem"], data["orders"]
+    cust, supp = data["customer"], data["supplier"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_suppkey",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders, names=["o_orderkey", "o_custkey",
+                                                 "o_orderdate"])
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_nationkey"])
+    scols = staging.stage_columns(supp)
+    got = DQ.q5_fused(lcols, ocols, ccols, scols)
+    want = OQ.q5(li, orders, cust, supp, data["nation"], data["region"])
+    assert [n for n, _ in got] == [n for n, _ in want]
+    np.testing.assert_allclose([v for _, v in got], [v for _, v in want],
+                               rtol=1e-9)
+    for cs in (lcols, ocols, ccols, scols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q5_fused_on_device_generated(gpu):
+    from quokka_amd import shim, queries as DQ
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n = 400_000
+    n_ord, n_cust, n_supp = n // 4, n // 40, max(1, n // 600)
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_suppkey", np.int64),
+        ("l_extendedprice", np.float64), ("l_discount", np.float64)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(0), c_u64(9),
+              c_i64(20000), c_i64(n_supp), c_i64(n_ord),
+              li["l_orderkey"].ptr, li["l_suppkey"].ptr, None,
+              li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
+              None, None, None)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_custkey", np.int64),
+        ("o_orderdate", np.int32)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(9),
+              c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
+              od["o_orderdate"].ptr, None)
+    cu = {"c_custkey": DevColumn(np.int64, n_cust),
+          "c_nationkey": DevColumn(np.int32, n_cust)}
+    shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(9),
+              cu["c_custkey"].ptr, None, cu["c_nationkey"].ptr)
+    su = {"s_suppkey": DevColumn(np.int64, n_supp),
+          "s_nationkey": DevColumn(np.int32, n_supp)}
+    shim.call("qk_gen_supplier", None, c_u64(n_supp), c_u64(0), c_u64(9),
+              su["s_suppkey"].ptr, su["s_nationkey"].ptr)
+    li_h = {k: v.to_numpy() for k, v in li.items()}
+    od_h = {k: v.to_numpy() for k, v in od.items()}
+    cu_h = {k: v.to_numpy() for k, v in cu.items()}
+    su_h = {k: v.to_numpy() for k, v in su.items()}
+    got = DQ.q5_fused(li, od, cu, su)
+    want = OQ.q5(li_h, od_h, cu_h, su_h, G.gen_nation(), G.gen_region())
+    assert [n_ for n_, _ in got] == [n_ for n_, _ in want]
+    np.testing.assert_allclose([v for _, v in got], [v for _, v in want],
+                               rtol=1e-9)
+    for cs in (li, od, cu, su):
         for c in cs.values():
             c.free()
