@@ -42,6 +42,8 @@ def parse_args():
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
     p.add_argument("--query", choices=["q1", "q3", "q5"], default="q1")
+    p.add_argument("--nt", type=int, default=1,
+                   help="non-temporal probe loads (Q3/Q5)")
     p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -245,7 +247,7 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         fused.rebuild()
         if timed:
             timer.start(stream)
-        fused.probe(li)
+        fused.probe(li, nt=bool(args.nt))
         if timed:
             timer.stop(stream)
         full, top10 = fused.extract(10)
@@ -357,7 +359,7 @@ def main_q5(args, n, world, rank, dist, shim, DQ):
         fused.rebuild()
         if timed:
             timer.start(stream)
-        fused.probe(li)
+        fused.probe(li, nt=bool(args.nt))
         if timed:
             timer.stop(stream)
         res = fused.result()

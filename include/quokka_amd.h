@@ -191,6 +191,15 @@ int qk_q3_probe_agg(void *stream, uint64_t n, const int64_t *l_orderkey,
                     uint64_t capacity, double *slot_sums /* f64[capacity],
                     zeroed; groups keyed by slot */,
                     uint64_t *match_count_dev /* nullable, zeroed u64 */);
+/* Non-temporal-load variant of qk_q3_probe_agg: streams the lineitem
+ * columns without polluting the caches so the build table stays resident
+ * (same results; perf variant, A/B-measured in profiles/). */
+int qk_q3_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
+                       const int32_t *l_shipdate, const double *l_price,
+                       const double *l_disc, int32_t date_gt,
+                       const int64_t *slot_keys, const int32_t *slot_head,
+                       uint64_t capacity, double *slot_sums,
+                       uint64_t *match_count_dev);
 /* Emit (orderkey, orders_build_row, revenue) for slots with sum != 0.
  * cursor (u64, zeroed) = group count (counted even past out_cap). */
 int qk_q3_extract(void *stream, const int64_t *slot_keys,
@@ -228,6 +237,15 @@ int qk_q5_probe_agg(void *stream, uint64_t n, const int64_t *l_orderkey,
                     uint64_t supp_cap, double *out25,
                     uint64_t *match_count_dev /* nullable */);
 
+/* Non-temporal-load variant of qk_q5_probe_agg (see qk_q3_probe_agg_nt). */
+int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
+                       const int64_t *l_suppkey, const double *l_price,
+                       const double *l_disc, const int64_t *ord_keys,
+                       const int32_t *ord_val, uint64_t ord_cap,
+                       const int64_t *supp_keys, const int32_t *supp_val,
+                       uint64_t supp_cap, double *out25,
+                       uint64_t *match_count_dev);
+
 /* ---- group-by (i64 key) sum ------------------------------------------- *
  * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for
  * distributive SUM over an i64 key (the post-rewrite partial form,
@@ -258,6 +276,31 @@ int qk_partition_hist(void *stream, uint64_t n, const int64_t *keys,
 int qk_partition_scatter(void *stream, uint64_t n, const int64_t *keys,
                          uint32_t nparts, uint64_t *cursors_dev,
                          uint32_t *out_idx);
+
+/* ---- RCCL exchange (multi-GPU hash repartition) ----------------------- *
+ * Replaces the reference's shuffle data plane (core.py:276-376 push ->
+ * Arrow Flight do_put/do_get, flight.py) for the join/group-by
+ * repartition: per-rank column buffers, partition-ordered by
+ * `key % world` (quokka_runtime.py:222), exchanged with grouped
+ * ncclSend/ncclRecv over xGMI — direct per-peer sends (7 p2p links/GPU),
+ * not a ring (SURVEY.md §5 backend note). One rank per GPU; the 128-byte
+ * unique id is broadcast host-side (gloo) by the caller. */
+#define QK_UID_BYTES 128
+int qk_comm_unique_id(uint8_t out[QK_UID_BYTES]);
+int qk_comm_init(int rank, int world, const uint8_t uid[QK_UID_BYTES],
+                 void **comm_out);
+int qk_comm_destroy(void *comm);
+/* All-to-all of one column: for each peer p, send send_counts[p] elements
+ * of `elem_size` bytes from send_buf + send_offsets[p]*elem_size and
+ * receive recv_counts[p] into recv_buf + recv_offsets[p]*elem_size.
+ * Counts/offsets are HOST arrays (exchanged by the caller beforehand).
+ * Grouped send/recv, completes on `stream`. */
+int qk_alltoallv(void *stream, void *comm, int world, uint32_t elem_size,
+                 const void *send_buf, const uint64_t *send_offsets,
+                 const uint64_t *send_counts, void *recv_buf,
+                 const uint64_t *recv_offsets, const uint64_t *recv_counts);
+/* All-reduce SUM of an f64 device buffer (tiny partial-agg combine). */
+int qk_allreduce_f64(void *stream, void *comm, double *buf, uint64_t n);
 
 #ifdef __cplusplus
 }

@@ -946,6 +946,52 @@ __global__ void __launch_bounds__(BLOCK) k_q3_probe_agg(
     }
   }
 }
+// nt variant: non-temporal loads on the streamed lineitem columns keep
+// the hash table cache-resident (the streaming scan otherwise evicts it;
+// probe line reads dominate actual traffic — profiles/r01_q3)
+__global__ void __launch_bounds__(BLOCK) k_q3_probe_agg_nt(
+    uint64_t n, const int64_t *__restrict__ l_orderkey,
+    const int32_t *__restrict__ l_shipdate,
+    const double *__restrict__ l_price, const double *__restrict__ l_disc,
+    int32_t date_gt, const int64_t *__restrict__ slot_keys,
+    const int32_t *__restrict__ slot_head, uint64_t cap,
+    double *__restrict__ slot_sums, uint64_t *__restrict__ match_count) {
+  uint32_t matches = 0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (__builtin_nontemporal_load(&l_shipdate[i]) <= date_gt) continue;
+    int64_t key = __builtin_nontemporal_load(&l_orderkey[i]);
+    uint64_t s = slot_of(key, cap);
+    int32_t head = -1;
+    for (;;) {
+      int64_t cur = slot_keys[s];
+      if (cur == key) { head = slot_head[s]; break; }
+      if (cur == QK_JOIN_EMPTY) break;
+      s = (s + 1) & (cap - 1);
+    }
+    if (head < 0) continue;
+    matches++;
+    double price = __builtin_nontemporal_load(&l_price[i]);
+    double disc = __builtin_nontemporal_load(&l_disc[i]);
+    atomicAdd(&slot_sums[s], price * (1.0 - disc));
+  }
+  if (match_count) {
+    __shared__ uint32_t lds[BLOCK / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      matches += __shfl_down(matches, off);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = matches;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint64_t t = 0;
+      for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+      if (t) atomicAdd((unsigned long long *)match_count,
+                       (unsigned long long)t);
+    }
+  }
+}
+
 extern "C" int qk_q3_probe_agg(void *stream, uint64_t n,
                                const int64_t *l_orderkey,
                                const int32_t *l_shipdate,
@@ -964,55 +1010,110 @@ extern "C" int qk_q3_probe_agg(void *stream, uint64_t n,
   QK_TRY("qk_q3_probe_agg", hipGetLastError());
   return 0;
 }
+extern "C" int qk_q3_probe_agg_nt(void *stream, uint64_t n,
+                                  const int64_t *l_orderkey,
+                                  const int32_t *l_shipdate,
+                                  const double *l_price, const double *l_disc,
+                                  int32_t date_gt, const int64_t *slot_keys,
+                                  const int32_t *slot_head, uint64_t cap,
+                                  double *slot_sums, uint64_t *match_count) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_q3_probe_agg_nt.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_probe_agg_nt, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, l_orderkey, l_shipdate, l_price,
+                     l_disc, date_gt, slot_keys, slot_head, cap, slot_sums,
+                     match_count);
+  QK_TRY("qk_q3_probe_agg_nt", hipGetLastError());
+  return 0;
+}
 
 // extract groups: slots with a nonzero revenue sum; emits the group key,
-// the orders build ROW (for o_orderdate/o_shippriority attach) and the sum
-__global__ void __launch_bounds__(BLOCK) k_q3_extract(
-    const int64_t *__restrict__ slot_keys,
-    const int32_t *__restrict__ slot_head,
-    const double *__restrict__ slot_sums, uint64_t cap,
-    int64_t *__restrict__ out_keys, int32_t *__restrict__ out_row,
-    double *__restrict__ out_sums, uint64_t out_cap,
-    uint64_t *__restrict__ cursor) {
-  // wave-aggregated cursor claims: one atomicAdd per wave per pass, not
-  // one per group (a single cursor word takes ~88 atomics/us — the
-  // per-group version measured 8.6 ms on 1.2M groups)
-  int lane = threadIdx.x & (WAVE - 1);
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t base = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x -
-                       lane;
-       base < cap; base += stride) {
-    uint64_t s = base + lane;
-    bool live = s < cap && slot_keys[s] != QK_JOIN_EMPTY &&
-                slot_sums[s] != 0.0;
-    uint64_t mask = __ballot(live);
-    uint32_t nlive = __popcll(mask);
-    if (!nlive) continue;
-    uint64_t wbase;
-    if (lane == 0)
-      wbase = atomicAdd((unsigned long long *)cursor,
-                        (unsigned long long)nlive);
-    wbase = __shfl(wbase, 0);
-    if (live) {
-      uint64_t pos = wbase + __popcll(mask & ((1ULL << lane) - 1));
-      if (pos < out_cap) {
-        out_keys[pos] = slot_keys[s];
-        out_row[pos] = slot_head[s];
-        out_sums[pos] = slot_sums[s];
-      }
-    }
+// the orders build ROW (for o_orderdate/o_shippriority attach) and the sum.
+// Two passes over contiguous slot chunks (per-block count -> single-block
+// scan -> rank+scatter): no shared cursor word (a single cursor measured
+// 8.6 ms on 1.2M groups; wave-aggregated claims still 5.7 ms).
+__global__ void __launch_bounds__(BLOCK) k_extract_count(
+    const int64_t *__restrict__ slot_keys, const double *__restrict__ slot_sums,
+    uint64_t cap, uint64_t chunk, uint64_t *__restrict__ block_counts) {
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(cap, lo + chunk);
+  uint32_t cnt = 0;
+  for (uint64_t st = lo + threadIdx.x; st < hi; st += BLOCK)
+    cnt += (slot_keys[st] != QK_JOIN_EMPTY && slot_sums[st] != 0.0) ? 1u : 0u;
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t t = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+    block_counts[blockIdx.x] = t;
   }
 }
+
+__global__ void __launch_bounds__(BLOCK) k_extract_scatter(
+    const int64_t *__restrict__ slot_keys, const int32_t *__restrict__ slot_head,
+    const double *__restrict__ slot_sums, uint64_t cap, uint64_t chunk,
+    const uint64_t *__restrict__ block_offsets, int64_t *__restrict__ out_keys,
+    int32_t *__restrict__ out_row, double *__restrict__ out_sums,
+    uint64_t out_cap) {
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(cap, lo + chunk);
+  __shared__ uint64_t base;
+  __shared__ uint32_t wave_tot[BLOCK / WAVE];
+  if (threadIdx.x == 0) base = block_offsets[blockIdx.x];
+  __syncthreads();
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  for (uint64_t s0 = lo; s0 < hi; s0 += BLOCK) {
+    uint64_t st = s0 + threadIdx.x;
+    bool live = st < hi && slot_keys[st] != QK_JOIN_EMPTY &&
+                slot_sums[st] != 0.0;
+    uint64_t mask = __ballot(live);
+    uint32_t rank = __popcll(mask & ((1ULL << lane) - 1));
+    if (lane == 0) wave_tot[wid] = __popcll(mask);
+    __syncthreads();
+    uint32_t wbase = 0;
+    for (int w = 0; w < wid; w++) wbase += wave_tot[w];
+    if (live) {
+      uint64_t pos = base + wbase + rank;
+      if (pos < out_cap) {
+        out_keys[pos] = slot_keys[st];
+        out_row[pos] = slot_head[st];
+        out_sums[pos] = slot_sums[st];
+      }
+    }
+    uint32_t btot = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) btot += wave_tot[w];
+    __syncthreads();
+    if (threadIdx.x == 0) base += btot;
+    __syncthreads();
+  }
+}
+
 extern "C" int qk_q3_extract(void *stream, const int64_t *slot_keys,
                              const int32_t *slot_head,
                              const double *slot_sums, uint64_t cap,
                              int64_t *out_keys, int32_t *out_row,
                              double *out_sums, uint64_t out_cap,
                              uint64_t *cursor) {
-  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
-  hipLaunchKernelGGL(k_q3_extract, dim3(blocks), dim3(BLOCK), 0,
+  uint64_t chunk = (cap + MAX_BLOCKS - 1) / MAX_BLOCKS;
+  chunk = ((chunk + BLOCK - 1) / BLOCK) * BLOCK;
+  uint32_t blocks = (uint32_t)((cap + chunk - 1) / chunk);
+  static __thread uint64_t *scratch = nullptr;
+  if (!scratch) QK_TRY("qk_q3_extract", hipMalloc(&scratch,
+                       (MAX_BLOCKS + 1) * sizeof(uint64_t)));
+  hipLaunchKernelGGL(k_extract_count, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, slot_keys, slot_sums, cap, chunk,
+                     scratch);
+  hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0, (hipStream_t)stream,
+                     (uint64_t)blocks, scratch, cursor);
+  hipLaunchKernelGGL(k_extract_scatter, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, slot_keys, slot_head, slot_sums,
-                     cap, out_keys, out_row, out_sums, out_cap, cursor);
+                     cap, chunk, scratch, out_keys, out_row, out_sums,
+                     out_cap);
   QK_TRY("qk_q3_extract", hipGetLastError());
   return 0;
 }
@@ -1171,6 +1272,40 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg(
   if (match_count && threadIdx.x == 0 && lcnt)
     atomicAdd((unsigned long long *)match_count, (unsigned long long)lcnt);
 }
+__global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
+    uint64_t n, const int64_t *__restrict__ l_orderkey,
+    const int64_t *__restrict__ l_suppkey,
+    const double *__restrict__ l_price, const double *__restrict__ l_disc,
+    const int64_t *__restrict__ ord_keys, const int32_t *__restrict__ ord_val,
+    uint64_t ord_cap, const int64_t *__restrict__ supp_keys,
+    const int32_t *__restrict__ supp_val, uint64_t supp_cap,
+    double *__restrict__ out25, uint64_t *__restrict__ match_count) {
+  __shared__ double lsum[32];
+  __shared__ uint32_t lcnt;
+  if (threadIdx.x < 32) lsum[threadIdx.x] = 0.0;
+  if (threadIdx.x == 0) lcnt = 0;
+  __syncthreads();
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t okey = __builtin_nontemporal_load(&l_orderkey[i]);
+    int32_t cnat = probe_unique(ord_keys, ord_val, ord_cap, okey, nullptr);
+    if (cnat < 0) continue;
+    int64_t skey = __builtin_nontemporal_load(&l_suppkey[i]);
+    int32_t snat = probe_unique(supp_keys, supp_val, supp_cap, skey, nullptr);
+    if (snat != cnat) continue;
+    double price = __builtin_nontemporal_load(&l_price[i]);
+    double disc = __builtin_nontemporal_load(&l_disc[i]);
+    atomicAdd(&lsum[cnat & 31], price * (1.0 - disc));
+    if (match_count) atomicAdd(&lcnt, 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x < 32 && lsum[threadIdx.x] != 0.0)
+    atomicAdd(&out25[threadIdx.x], lsum[threadIdx.x]);
+  if (match_count && threadIdx.x == 0 && lcnt)
+    atomicAdd((unsigned long long *)match_count, (unsigned long long)lcnt);
+}
+
 extern "C" int qk_q5_probe_agg(void *stream, uint64_t n,
                                const int64_t *l_orderkey,
                                const int64_t *l_suppkey,
@@ -1188,6 +1323,26 @@ extern "C" int qk_q5_probe_agg(void *stream, uint64_t n,
                      l_disc, ord_keys, ord_val, ord_cap, supp_keys, supp_val,
                      supp_cap, out25, match_count);
   QK_TRY("qk_q5_probe_agg", hipGetLastError());
+  return 0;
+}
+extern "C" int qk_q5_probe_agg_nt(void *stream, uint64_t n,
+                                  const int64_t *l_orderkey,
+                                  const int64_t *l_suppkey,
+                                  const double *l_price, const double *l_disc,
+                                  const int64_t *ord_keys,
+                                  const int32_t *ord_val, uint64_t ord_cap,
+                                  const int64_t *supp_keys,
+                                  const int32_t *supp_val, uint64_t supp_cap,
+                                  double *out25, uint64_t *match_count) {
+  if (!n) return 0;
+  if ((ord_cap & (ord_cap - 1)) || (supp_cap & (supp_cap - 1)))
+    return qk_fail("qk_q5_probe_agg_nt.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q5_probe_agg_nt, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, l_orderkey, l_suppkey, l_price,
+                     l_disc, ord_keys, ord_val, ord_cap, supp_keys, supp_val,
+                     supp_cap, out25, match_count);
+  QK_TRY("qk_q5_probe_agg_nt", hipGetLastError());
   return 0;
 }
 
@@ -1283,6 +1438,72 @@ extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
                      (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
                      out_keys, out_sums, out_cap, cursor);
   QK_TRY("qk_groupby_extract", hipGetLastError());
+  return 0;
+}
+
+// ---- RCCL exchange ----------------------------------------------------
+// Grouped send/recv all-to-allv over xGMI (direct per-peer, not a ring);
+// replaces the reference's Flight-based shuffle (core.py:276-376) for the
+// repartition step. See include/quokka_amd.h.
+#include <rccl/rccl.h>
+
+static int qk_nccl_fail(const char *where, ncclResult_t r) {
+  snprintf(g_err, sizeof(g_err), "%s: %s", where, ncclGetErrorString(r));
+  g_err_set = 1;
+  return 1000 + (int)r;
+}
+#define QK_NCCL(where, expr)                                                   \
+  do {                                                                         \
+    ncclResult_t _r = (expr);                                                  \
+    if (_r != ncclSuccess) return qk_nccl_fail(where, _r);                     \
+  } while (0)
+
+extern "C" int qk_comm_unique_id(uint8_t out[QK_UID_BYTES]) {
+  static_assert(sizeof(ncclUniqueId) == QK_UID_BYTES, "uid size");
+  QK_NCCL("qk_comm_unique_id", ncclGetUniqueId((ncclUniqueId *)out));
+  return 0;
+}
+extern "C" int qk_comm_init(int rank, int world,
+                            const uint8_t uid[QK_UID_BYTES],
+                            void **comm_out) {
+  ncclUniqueId id;
+  memcpy(&id, uid, sizeof(id));
+  ncclComm_t comm;
+  QK_NCCL("qk_comm_init", ncclCommInitRank(&comm, world, id, rank));
+  *comm_out = (void *)comm;
+  return 0;
+}
+extern "C" int qk_comm_destroy(void *comm) {
+  QK_NCCL("qk_comm_destroy", ncclCommDestroy((ncclComm_t)comm));
+  return 0;
+}
+extern "C" int qk_alltoallv(void *stream, void *comm, int world,
+                            uint32_t elem_size, const void *send_buf,
+                            const uint64_t *send_offsets,
+                            const uint64_t *send_counts, void *recv_buf,
+                            const uint64_t *recv_offsets,
+                            const uint64_t *recv_counts) {
+  QK_NCCL("qk_alltoallv.group_start", ncclGroupStart());
+  for (int p = 0; p < world; p++) {
+    if (send_counts[p])
+      QK_NCCL("qk_alltoallv.send",
+              ncclSend((const char *)send_buf + send_offsets[p] * elem_size,
+                       send_counts[p] * elem_size, ncclUint8, p,
+                       (ncclComm_t)comm, (hipStream_t)stream));
+    if (recv_counts[p])
+      QK_NCCL("qk_alltoallv.recv",
+              ncclRecv((char *)recv_buf + recv_offsets[p] * elem_size,
+                       recv_counts[p] * elem_size, ncclUint8, p,
+                       (ncclComm_t)comm, (hipStream_t)stream));
+  }
+  QK_NCCL("qk_alltoallv.group_end", ncclGroupEnd());
+  return 0;
+}
+extern "C" int qk_allreduce_f64(void *stream, void *comm, double *buf,
+                                uint64_t n) {
+  QK_NCCL("qk_allreduce_f64",
+          ncclAllReduce(buf, buf, n, ncclDouble, ncclSum, (ncclComm_t)comm,
+                        (hipStream_t)stream));
   return 0;
 }
 

@@ -221,7 +221,7 @@ class Q3Fused:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
         cnt.free()
-        self.ord_cap = ops._pow2_at_least(max(16, 2 * self.n_build))
+        self.ord_cap = ops._pow2_at_least(max(16, 4 * self.n_build))
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_head = DevColumn(np.int32, self.ord_cap)
         self.ord_sums = DevColumn(np.float64, self.ord_cap)
@@ -275,11 +275,13 @@ class Q3Fused:
              self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
              self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap))
 
-    def probe(self, li_cols, match_count_buf=None):
-        """The fused filter+probe+group-by-aggregate pass (one kernel)."""
+    def probe(self, li_cols, match_count_buf=None, nt=True):
+        """The fused filter+probe+group-by-aggregate pass (one kernel).
+        nt=True streams the lineitem columns with non-temporal loads."""
         sh = self.stream.handle if self.stream else None
         n = li_cols["l_orderkey"].n
-        call("qk_q3_probe_agg", sh, c_u64(n), li_cols["l_orderkey"].ptr,
+        call("qk_q3_probe_agg_nt" if nt else "qk_q3_probe_agg", sh,
+             c_u64(n), li_cols["l_orderkey"].ptr,
              li_cols["l_shipdate"].ptr, li_cols["l_extendedprice"].ptr,
              li_cols["l_discount"].ptr, ctypes.c_int32(Q3_DATE),
              self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
@@ -462,10 +464,11 @@ class Q5Fused:
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
              c_u64(self.ord_cap), None)
 
-    def probe(self, li_cols, match_count_buf=None):
+    def probe(self, li_cols, match_count_buf=None, nt=True):
         sh = self.stream.handle if self.stream else None
         n = li_cols["l_orderkey"].n
-        call("qk_q5_probe_agg", sh, c_u64(n), li_cols["l_orderkey"].ptr,
+        call("qk_q5_probe_agg_nt" if nt else "qk_q5_probe_agg", sh,
+             c_u64(n), li_cols["l_orderkey"].ptr,
              li_cols["l_suppkey"].ptr, li_cols["l_extendedprice"].ptr,
              li_cols["l_discount"].ptr, self.ord_keys.ptr,
              self.ord_val.ptr, c_u64(self.ord_cap), self.supp_keys.ptr,

@@ -82,6 +82,10 @@ _SIGS = {
                            c_u64, c_vp],
     "qk_q3_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
                         c_vp, c_u64, c_vp, c_vp],
+    "qk_q3_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
+                           c_vp, c_u64, c_vp, c_vp],
+    "qk_q5_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                           c_u64, c_vp, c_vp, c_u64, c_vp, c_vp],
     "qk_q3_extract": [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
                       c_u64, c_vp],
     "qk_groupby_i64_sum": [c_vp, c_u64, c_vp, c_vp, ctypes.c_int, c_vp, c_vp,

@@ -413,10 +413,10 @@ def test_q3_fused_rebuild_idempotent(gpu, data):
     ocols = staging.stage_columns(orders)
     ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
     st = DQ.Q3Fused(ocols, ccols)
-    st.probe(lcols)
+    st.probe(lcols, nt=False)     # plain-load variant
     f1, _ = st.extract()
     st.rebuild()
-    st.probe(lcols)
+    st.probe(lcols, nt=True)      # non-temporal variant must agree
     f2, _ = st.extract()
     o1, o2 = np.argsort(f1["l_orderkey"]), np.argsort(f2["l_orderkey"])
     assert np.array_equal(f1["l_orderkey"][o1], f2["l_orderkey"][o2])
