@@ -42,6 +42,9 @@ def parse_args():
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
     p.add_argument("--query", choices=["q1", "q3", "q5"], default="q1")
+    p.add_argument("--exchange", choices=["auto", "none", "rccl"],
+                   default="auto",
+                   help="q3 multi-rank repartition (auto: rccl when world>1)")
     p.add_argument("--nt", type=int, default=1,
                    help="non-temporal probe loads (Q3/Q5)")
     p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
@@ -212,58 +215,97 @@ QUERY = "q1"
 
 
 def main_q3(args, n, world, rank, dist, shim, DQ):
-    """TPC-H Q3 on the fused device path (BASELINE.json configs[2]).
-    A step = rebuild customer+orders tables + fused probe/agg + extract
-    top-10. Multi-rank: lineitem sharded by row range; the build tables are
-    broadcast-replicated (small sides); group tables merged
-    host-side at each step."""
+    """TPC-H Q3 on the fused device path.
+
+    world == 1 (BASELINE.json configs[2]): a step = rebuild customer+orders
+    tables + fused probe/agg + extract top-10, all rows local.
+
+    world > 1 or --exchange rccl (configs[3]): STRONG scaling over the
+    total --sf: each rank owns 1/world of lineitem+orders rows, and every
+    step hash-repartitions both by orderkey % world with RCCL grouped
+    send/recv over xGMI (quokka_amd/exchange.py — the reference's Flight
+    shuffle, core.py:276-376), rebuilds from the received partition,
+    probes, extracts; per-rank groups are disjoint so the global top-10 is
+    merged from per-rank top-10s."""
     import ctypes as ct
-    from quokka_amd import ops
+    from quokka_amd import ops, exchange
     from quokka_amd.shim import DevBuffer, c_u64
 
-    li, od, cu = gen_device_q3_tables(shim, n, rank)
+    use_exchange = args.exchange == "rccl" or (args.exchange == "auto"
+                                               and world > 1)
+    if use_exchange:
+        n_local = (n // world) & ~3
+    else:
+        n_local = n
+    li, od, cu = gen_device_q3_tables(shim, n_local, rank)
     stream = shim.Stream()
-    fused = DQ.Q3Fused(od, cu, stream)
+    comm = exchange.Comm(rank, world, dist) if use_exchange else None
+
+    def do_exchange():
+        """Repartition lineitem + orders by orderkey % world; returns new
+        (li, od) column dicts (received partitions)."""
+        rk, rp = exchange.repartition(
+            comm, li["l_orderkey"],
+            {k: v for k, v in li.items() if k != "l_orderkey"}, stream)
+        li2 = {"l_orderkey": rk, **rp}
+        ok, op = exchange.repartition(
+            comm, od["o_orderkey"],
+            {k: v for k, v in od.items() if k != "o_orderkey"}, stream)
+        od2 = {"o_orderkey": ok, **op}
+        return li2, od2
+
+    if use_exchange:
+        li_x, od_x = do_exchange()
+        fused = DQ.Q3Fused(od_x, cu, stream)
+    else:
+        li_x, od_x = li, od
+        fused = DQ.Q3Fused(od, cu, stream)
     stream.sync()
     timer = shim.Timer()
 
     # one-off counts for the algorithmic-byte accounting (DESIGN.md §Q3)
     mc = ops._count_buf()
-    fused.probe(li, mc)
+    fused.probe(li_x, mc)
     stream.sync()
     n_match = ops._read_u64(mc)
     mc.free()
     fused.reset_sums()
-    idxbuf = shim.DevColumn(np.uint32, n)
+    idxbuf = shim.DevColumn(np.uint32, max(1, li_x["l_shipdate"].n))
     cntbuf = ops._count_buf()
-    shim.call("qk_filter_i32", stream.handle, c_u64(n),
-              li["l_shipdate"].ptr, 2, ct.c_int32(DQ.Q3_DATE),
+    shim.call("qk_filter_i32", stream.handle, c_u64(li_x["l_shipdate"].n),
+              li_x["l_shipdate"].ptr, 2, ct.c_int32(DQ.Q3_DATE),
               idxbuf.ptr, cntbuf.ptr)
     stream.sync()
     n_pass = ops._read_u64(cntbuf)
     idxbuf.free(); cntbuf.free()
 
     def step(timed):
-        fused.rebuild()
+        nonlocal li_x, od_x, fused
+        if use_exchange:
+            for c in list(li_x.values()) + list(od_x.values()):
+                c.free()
+            li_x, od_x = do_exchange()
+            fused.free()
+            fused = DQ.Q3Fused(od_x, cu, stream)
+        else:
+            fused.rebuild()
         if timed:
             timer.start(stream)
-        fused.probe(li, nt=bool(args.nt))
+        fused.probe(li_x, nt=bool(args.nt))
         if timed:
             timer.stop(stream)
-        full, top10 = fused.extract(10)
+        n_groups, top10 = fused.extract_top10(10)
         if dist is not None:
+            # per-rank orderkey partitions are disjoint: merge top-10s
             import torch.distributed as _d
             gathered = [None] * world
-            _d.all_gather_object(gathered, (full["l_orderkey"],
-                                            full["revenue"]))
+            _d.all_gather_object(gathered, top10)
             if rank == 0:
-                import collections
-                acc = collections.defaultdict(float)
-                for ks, vs in gathered:
-                    for k, v in zip(ks.tolist(), vs.tolist()):
-                        acc[k] += v
-                top10 = dict(n_groups=len(acc))
-        return full, top10, (timer.elapsed_ms() if timed else None)
+                cand = {k: np.concatenate([g[k] for g in gathered])
+                        for k in top10}
+                sel = DQ._topk(cand, 10)
+                top10 = {k: v[sel] for k, v in cand.items()}
+        return n_groups, top10, (timer.elapsed_ms() if timed else None)
 
     for _ in range(args.warmup):
         step(False)
@@ -272,9 +314,9 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
     stream.sync()
     t0 = time.time()
     kernel_ms = []
-    full = None
+    n_groups = 0
     for _ in range(args.steps):
-        full, top10, kms = step(True)
+        n_groups, top10, kms = step(True)
         kernel_ms.append(kms)
     stream.sync()
     elapsed = time.time() - t0
@@ -289,30 +331,35 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         # probe-kernel algorithmic bytes: 12 B/row (orderkey+shipdate) every
         # row + 12 B bucket read per ship-passing row + 16 B (price+disc)
         # per matched row
-        alg_bytes = 12 * n + 12 * n_pass + 16 * n_match
+        alg_bytes = 12 * li_x["l_shipdate"].n + 12 * n_pass + 16 * n_match
         avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
         achieved_gbps = alg_bytes / avg_kernel_s / 1e9
+        total_rows = (n_local * world if not use_exchange else n_local * world)
         out = {
             "metric": "rows/s",
-            "value": n * world * args.steps / elapsed,
+            "value": total_rows * args.steps / elapsed,
             "unit": "rows/s",
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1e3,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if use_exchange else "weak",
             "vs_baseline": None,
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "TPC-H SF%g Q3 (3-way hash join + group-by), "
+                "workload": "TPC-H SF%g Q3 (3-way hash join + group-by%s), "
                             "%d lineitem rows/GPU resident in HBM "
-                            "(BASELINE.json configs[2])" % (args.sf, n),
-                "sf_per_gpu": args.sf,
-                "rows_per_gpu": n,
+                            "(BASELINE.json configs[%d])"
+                            % (args.sf,
+                               ", RCCL all-to-all repartition" if use_exchange
+                               else "", n_local, 3 if use_exchange else 2),
+                "sf_per_gpu": args.sf if not use_exchange else args.sf / world,
+                "rows_per_gpu": n_local,
                 "query": "Q3",
-                "n_groups": int(len(full["l_orderkey"])),
+                "exchange": "rccl" if use_exchange else "none",
+                "n_groups": int(n_groups),
                 "orders_build_rows": int(fused.n_build),
                 "lineitem_ship_pass": int(n_pass),
                 "joined_rows": int(n_match),

@@ -829,10 +829,12 @@ __global__ void __launch_bounds__(BLOCK) k_q3_build_orders(
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    if (o_orderdate[i] >= date_lt) continue;
-    if (probe_unique(cust_keys, cust_head, cust_cap, o_custkey[i], nullptr) < 0)
+    // nt loads: stream the orders columns without evicting the tables
+    if (__builtin_nontemporal_load(&o_orderdate[i]) >= date_lt) continue;
+    if (probe_unique(cust_keys, cust_head, cust_cap,
+                     __builtin_nontemporal_load(&o_custkey[i]), nullptr) < 0)
       continue;
-    int64_t key = o_orderkey[i];
+    int64_t key = __builtin_nontemporal_load(&o_orderkey[i]);
     uint64_t s = slot_of(key, cap);
     for (;;) {
       int64_t cur = slot_keys[s];

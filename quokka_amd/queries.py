@@ -327,6 +327,60 @@ class Q3Fused:
         top10 = {c: v[top] for c, v in full.items()}
         return full, top10
 
+    def extract_top10(self, limit=10):
+        """Bench-lean extract: d2h only the revenue column, select the
+        top-k candidates host-side, gather the candidates' key/date/prio
+        on-device. Returns (n_groups, top10 dict)."""
+        from . import ops
+        from .shim import DevColumn, c_u64
+        sh = self.stream.handle if self.stream else None
+        out_cap = self.ord_cap
+        if getattr(self, "_ext", None) is None:
+            self._ext = (DevColumn(np.int64, out_cap),
+                         DevColumn(np.int32, out_cap),
+                         DevColumn(np.float64, out_cap))
+        ok, orow, osum = self._ext
+        cur = ops._count_buf()
+        call("qk_q3_extract", sh, self.ord_keys.ptr, self.ord_head.ptr,
+             self.ord_sums.ptr, c_u64(self.ord_cap), ok.ptr, orow.ptr,
+             osum.ptr, c_u64(out_cap), cur.ptr)
+        if self.stream:
+            self.stream.sync()
+        k = ops._read_u64(cur)
+        cur.free()
+        if k == 0:
+            empty = {c: np.empty(0) for c in
+                     ("l_orderkey", "o_orderdate", "o_shippriority",
+                      "revenue")}
+            return 0, empty
+        rev = osum.to_numpy(k)
+        if k > 4 * limit + 64:
+            cand = np.argpartition(-rev, 2 * limit)[: 2 * limit]
+            cutoff = rev[cand].min()
+            cand = np.nonzero(rev >= cutoff)[0]
+        else:
+            cand = np.arange(k)
+        cidx = DevColumn.from_numpy(cand.astype(np.uint32))
+        ckeys = ok.gather(cidx, len(cand), self.stream)
+        crows = orow.gather(cidx, len(cand), self.stream)
+        rows_h = crows.to_numpy(len(cand)).astype(np.uint32)
+        ridx = DevColumn.from_numpy(rows_h)
+        dcol = self._ord_cols["o_orderdate"].gather(ridx, len(cand),
+                                                    self.stream)
+        pcol = self._ord_cols["o_shippriority"].gather(ridx, len(cand),
+                                                       self.stream)
+        cdict = {
+            "l_orderkey": ckeys.to_numpy(len(cand)),
+            "o_orderdate": dcol.to_numpy(len(cand)),
+            "o_shippriority": pcol.to_numpy(len(cand)),
+            "revenue": rev[cand],
+        }
+        sel = _topk(cdict, limit)
+        top10 = {c: v[sel] for c, v in cdict.items()}
+        for c in (cidx, ckeys, crows, ridx, dcol, pcol):
+            c.free()
+        return int(k), top10
+
     def free(self):
         for c in (self.cust_keys, self.cust_head, self.ord_keys,
                   self.ord_head, self.ord_sums):

@@ -69,3 +69,25 @@ def test_partition_shards_are_disjoint_and_complete():
     uk_ref, sums_ref = E.groupby_sum_i64(keys, vals)
     assert np.array_equal(gk[order], uk_ref)
     np.testing.assert_allclose(gs[order], sums_ref, rtol=1e-12)
+
+
+def test_repartition_count_bookkeeping():
+    """The host-side plan of exchange.repartition (counts/offsets both
+    sides) is consistent: what rank r sends to p equals what p receives
+    from r, and every row lands exactly once (CPU, no GPU/RCCL)."""
+    rng = np.random.default_rng(7)
+    world = 4
+    per_rank_keys = [rng.integers(0, 1000, rng.integers(50, 200)).astype(np.int64)
+                     for _ in range(world)]
+    send = np.zeros((world, world), dtype=np.int64)
+    for r in range(world):
+        parts = E.partition_int(per_rank_keys[r], world)
+        for p in range(world):
+            send[r, p] = (parts == p).sum()
+    recv = send.T
+    for p in range(world):
+        got = int(recv[p].sum())
+        want = sum(int((E.partition_int(k, world) == p).sum())
+                   for k in per_rank_keys)
+        assert got == want
+    assert send.sum() == sum(len(k) for k in per_rank_keys)

@@ -558,3 +558,93 @@ def test_q5_fused_on_device_generated(gpu):
     for cs in (li, od, cu, su):
         for c in cs.values():
             c.free()
+
+
+# ---------- RCCL exchange (world_size == 1 self-exchange) ---------------
+
+def test_exchange_world1_roundtrip(gpu):
+    """Comm init + alltoallv at world 1: the full repartition path runs on
+    one GPU and must return exactly the rows whose key % 1 == 0 (all),
+    partition-ordered == original multiset."""
+    from quokka_amd import exchange, shim
+    rng = np.random.default_rng(41)
+    keys = rng.integers(0, 10_000, 100_000).astype(np.int64)
+    vals = rng.random(100_000)
+    kcol = shim.DevColumn.from_numpy(keys)
+    vcol = shim.DevColumn.from_numpy(vals)
+    comm = exchange.Comm(0, 1)
+    rk, rp = exchange.repartition(comm, kcol, {"v": vcol})
+    got_k = rk.to_numpy(rk.n)
+    got_v = rp["v"].to_numpy(rp["v"].n)
+    assert len(got_k) == len(keys)
+    order_g = np.lexsort((got_v, got_k))
+    order_w = np.lexsort((vals, keys))
+    assert np.array_equal(got_k[order_g], keys[order_w])
+    np.testing.assert_allclose(got_v[order_g], vals[order_w], rtol=0)
+    rk.free(); rp["v"].free(); kcol.free(); vcol.free()
+    comm.destroy()
+
+
+def test_exchange_allreduce_world1(gpu):
+    from quokka_amd import exchange, shim
+    from quokka_amd.shim import DevColumn
+    comm = exchange.Comm(0, 1)
+    arr = np.arange(48, dtype=np.float64)
+    col = DevColumn.from_numpy(arr)
+    comm.allreduce_f64(col.ptr, 48)
+    got = col.to_numpy()
+    np.testing.assert_allclose(got, arr)  # world 1: identity
+    col.free()
+    comm.destroy()
+
+
+def test_extract_top10_matches_full_extract(gpu, data):
+    from quokka_amd import queries as DQ, staging
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipdate",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders)
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
+    st = DQ.Q3Fused(ocols, ccols)
+    st.probe(lcols)
+    full, top_a = st.extract(10)
+    k, top_b = st.extract_top10(10)
+    assert k == len(full["l_orderkey"])
+    assert np.array_equal(top_a["l_orderkey"], top_b["l_orderkey"])
+    np.testing.assert_allclose(top_a["revenue"], top_b["revenue"], rtol=0)
+    assert np.array_equal(top_a["o_orderdate"], top_b["o_orderdate"])
+    st.free()
+    for cs in (lcols, ocols, ccols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q3_after_exchange_world1(gpu, data):
+    """repartition (world 1 self-exchange) -> fused Q3 on the received
+    partition == oracle (the configs[3] path on one GPU)."""
+    from quokka_amd import queries as DQ, staging, exchange
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipdate",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders)
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
+    comm = exchange.Comm(0, 1)
+    rk, rp = exchange.repartition(
+        comm, lcols["l_orderkey"],
+        {k: v for k, v in lcols.items() if k != "l_orderkey"})
+    li_x = {"l_orderkey": rk, **rp}
+    ok, op = exchange.repartition(
+        comm, ocols["o_orderkey"],
+        {k: v for k, v in ocols.items() if k != "o_orderkey"})
+    od_x = {"o_orderkey": ok, **op}
+    full, top10 = DQ.q3_fused(li_x, od_x, ccols)
+    wfull, wtop = OQ.q3(li, orders, cust)
+    og, ow = np.argsort(full["l_orderkey"]), np.argsort(wfull["l_orderkey"])
+    assert np.array_equal(full["l_orderkey"][og], wfull["l_orderkey"][ow])
+    np.testing.assert_allclose(full["revenue"][og], wfull["revenue"][ow],
+                               rtol=1e-9)
+    assert np.array_equal(top10["l_orderkey"], wtop["l_orderkey"])
+    comm.destroy()
+    for cs in (lcols, ocols, ccols, li_x, od_x):
+        for c in cs.values():
+            c.free()
