@@ -172,12 +172,15 @@ int qk_join_probe(void *stream, uint64_t n_probe, const int64_t *keys,
 int qk_build_u8eq(void *stream, uint64_t n, const int64_t *keys,
                   const uint8_t *flag, uint8_t flag_val, int64_t *slot_keys,
                   int32_t *slot_head, uint64_t capacity);
+/* bloom/bloom_mask (nullable/0): optional Bloom prefilter the probe-side
+ * kernels test before the table walk (bits = pow2, mask = bits-1). */
 int qk_q3_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
                        const int64_t *o_custkey, const int32_t *o_orderdate,
                        int32_t date_lt, const int64_t *cust_keys,
                        const int32_t *cust_head, uint64_t cust_cap,
                        int64_t *slot_keys, int32_t *slot_head,
-                       uint64_t capacity);
+                       uint64_t capacity, uint32_t *bloom,
+                       uint64_t bloom_mask);
 /* Count the rows qk_q3_build_orders would insert (for tight table sizing;
  * count_dev u64, zeroed). */
 int qk_q3_count_orders(void *stream, uint64_t n, const int64_t *o_custkey,
@@ -199,7 +202,8 @@ int qk_q3_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
                        const double *l_disc, int32_t date_gt,
                        const int64_t *slot_keys, const int32_t *slot_head,
                        uint64_t capacity, double *slot_sums,
-                       uint64_t *match_count_dev);
+                       uint64_t *match_count_dev, const uint32_t *bloom,
+                       uint64_t bloom_mask);
 /* Emit (orderkey, orders_build_row, revenue) for slots with sum != 0.
  * cursor (u64, zeroed) = group count (counted even past out_cap). */
 int qk_q3_extract(void *stream, const int64_t *slot_keys,
@@ -225,7 +229,8 @@ int qk_q5_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
                        const int64_t *cust_keys, const int32_t *cust_val,
                        uint64_t cust_cap, int64_t *slot_keys,
                        int32_t *slot_val, uint64_t capacity,
-                       uint64_t *count_dev);
+                       uint64_t *count_dev, uint32_t *bloom,
+                       uint64_t bloom_mask);
 /* Fused probe: join lineitem to orders (-> customer nation) and supplier
  * (-> supplier nation); where equal accumulate revenue into out25[nation].
  * out25: f64[32], zeroed (slots 25..31 unused). */
@@ -244,7 +249,8 @@ int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
                        const int32_t *ord_val, uint64_t ord_cap,
                        const int64_t *supp_keys, const int32_t *supp_val,
                        uint64_t supp_cap, double *out25,
-                       uint64_t *match_count_dev);
+                       uint64_t *match_count_dev, const uint32_t *bloom,
+                       uint64_t bloom_mask);
 
 /* ---- group-by (i64 key) sum ------------------------------------------- *
  * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for

@@ -358,6 +358,46 @@ class GPUAggExecutor(Executor):
         return _to_table(out)
 
 
+class GPUTopKExecutor(Executor):
+    """Mirror of ConcatThenSQLExecutor as lowered by DataStream.top_k
+    (sql_executors.py:45-67; datastream.py:1746-1767): concatenate incoming
+    batches, done() returns the top-k rows by `sort_keys`/`descending`.
+
+    Like the reference (one DuckDB 'order by ... limit k' over the
+    concatenated partials in done()), the final selection runs host-side —
+    the inputs at this operator are the already-aggregated partials, tiny
+    relative to the scan the GPU kernels consumed."""
+
+    def __init__(self, sort_keys, k, descending=None):
+        assert isinstance(sort_keys, list) and len(sort_keys) >= 1
+        self.sort_keys = sort_keys
+        self.k = int(k)
+        self.descending = descending or [False] * len(sort_keys)
+        self.state = None
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if not batches:
+            return
+        batch = pa.concat_tables(batches)
+        self.state = batch if self.state is None \
+            else pa.concat_tables([self.state, batch])
+        return None
+
+    def done(self, executor_id):
+        if self.state is None:
+            return None
+        cols = []
+        for key, desc in zip(reversed(self.sort_keys),
+                             reversed(self.descending)):
+            v = np.asarray(self.state.column(key).to_numpy(
+                zero_copy_only=False))
+            cols.append(-v if desc else v)
+        order = np.lexsort(cols)[: self.k]
+        return self.state.take(order)
+
+
 def gpu_partition_fn(data, source_channel, num_target_channels, key=None):
     """GPU hash partitioner mirroring partition_key_str
     (quokka_runtime.py:217-231): int key -> key % N, bit-exact with :222.

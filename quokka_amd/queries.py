@@ -225,6 +225,11 @@ class Q3Fused:
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_head = DevColumn(np.int32, self.ord_cap)
         self.ord_sums = DevColumn(np.float64, self.ord_cap)
+        # Bloom prefilter: ~8 bits/build key, k=2 (skips the table-line
+        # read for ~95% of probe misses)
+        self.bloom_bits = ops._pow2_at_least(max(1 << 16, 8 * self.n_build))
+        self.bloom = DevColumn(np.uint32, self.bloom_bits // 32)
+        call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
         call("qk_fill_i64", sh, self.ord_keys.ptr,
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.ord_cap))
         call("qk_dmemset", self.ord_head.ptr, 0xFF, c_u64(self.ord_cap * 4))
@@ -233,7 +238,8 @@ class Q3Fused:
              ord_cols["o_orderkey"].ptr, ord_cols["o_custkey"].ptr,
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q3_DATE),
              self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
-             self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap))
+             self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
+             self.bloom.ptr, c_u64(self.bloom_bits - 1))
         self._ord_cols = ord_cols
         self._cust_cols = cust_cols
 
@@ -263,6 +269,7 @@ class Q3Fused:
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.ord_cap))
         call("qk_dmemset", self.ord_head.ptr, 0xFF, c_u64(self.ord_cap * 4))
         call("qk_dmemset", self.ord_sums.ptr, 0, c_u64(self.ord_cap * 8))
+        call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
         call("qk_build_u8eq", sh, c_u64(self._cust_cols["c_custkey"].n),
              self._cust_cols["c_custkey"].ptr,
              self._cust_cols["c_mktsegment"].ptr,
@@ -273,20 +280,29 @@ class Q3Fused:
              self._ord_cols["o_custkey"].ptr,
              self._ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q3_DATE),
              self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
-             self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap))
+             self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
+             self.bloom.ptr, c_u64(self.bloom_bits - 1))
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         """The fused filter+probe+group-by-aggregate pass (one kernel).
         nt=True streams the lineitem columns with non-temporal loads."""
         sh = self.stream.handle if self.stream else None
         n = li_cols["l_orderkey"].n
-        call("qk_q3_probe_agg_nt" if nt else "qk_q3_probe_agg", sh,
-             c_u64(n), li_cols["l_orderkey"].ptr,
-             li_cols["l_shipdate"].ptr, li_cols["l_extendedprice"].ptr,
-             li_cols["l_discount"].ptr, ctypes.c_int32(Q3_DATE),
-             self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
-             self.ord_sums.ptr,
-             match_count_buf.ptr if match_count_buf else None)
+        if nt:
+            call("qk_q3_probe_agg_nt", sh, c_u64(n),
+                 li_cols["l_orderkey"].ptr, li_cols["l_shipdate"].ptr,
+                 li_cols["l_extendedprice"].ptr, li_cols["l_discount"].ptr,
+                 ctypes.c_int32(Q3_DATE), self.ord_keys.ptr,
+                 self.ord_head.ptr, c_u64(self.ord_cap), self.ord_sums.ptr,
+                 match_count_buf.ptr if match_count_buf else None,
+                 self.bloom.ptr, c_u64(self.bloom_bits - 1))
+        else:
+            call("qk_q3_probe_agg", sh, c_u64(n),
+                 li_cols["l_orderkey"].ptr, li_cols["l_shipdate"].ptr,
+                 li_cols["l_extendedprice"].ptr, li_cols["l_discount"].ptr,
+                 ctypes.c_int32(Q3_DATE), self.ord_keys.ptr,
+                 self.ord_head.ptr, c_u64(self.ord_cap), self.ord_sums.ptr,
+                 match_count_buf.ptr if match_count_buf else None)
 
     def extract(self, limit=10):
         from . import ops
@@ -383,7 +399,7 @@ class Q3Fused:
 
     def free(self):
         for c in (self.cust_keys, self.cust_head, self.ord_keys,
-                  self.ord_head, self.ord_sums):
+                  self.ord_head, self.ord_sums, self.bloom):
             c.free()
         if getattr(self, "_ext", None):
             for c in self._ext:
@@ -462,14 +478,18 @@ class Q5Fused:
              ord_cols["o_orderkey"].ptr, ord_cols["o_custkey"].ptr,
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
-             c_u64(self.cust_cap), None, None, c_u64(16), cnt.ptr)
+             c_u64(self.cust_cap), None, None, c_u64(16), cnt.ptr, None,
+             c_u64(0))
         if stream:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
         cnt.free()
-        self.ord_cap = ops._pow2_at_least(max(16, 2 * self.n_build))
+        self.ord_cap = ops._pow2_at_least(max(16, 4 * self.n_build))
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_val = DevColumn(np.int32, self.ord_cap)
+        self.bloom_bits = ops._pow2_at_least(max(1 << 16, 8 * self.n_build))
+        self.bloom = DevColumn(np.uint32, self.bloom_bits // 32)
+        call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
         call("qk_fill_i64", sh, self.ord_keys.ptr,
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.ord_cap))
         call("qk_q5_build_orders", sh, c_u64(nord),
@@ -477,7 +497,8 @@ class Q5Fused:
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
-             c_u64(self.ord_cap), None)
+             c_u64(self.ord_cap), None, self.bloom.ptr,
+             c_u64(self.bloom_bits - 1))
         self.out25 = DevBuffer(32 * 8)
         call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
         self._ord_cols = ord_cols
@@ -510,18 +531,30 @@ class Q5Fused:
              self._supp_cols["s_nationkey"].ptr,
              ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
              self.supp_val.ptr, c_u64(self.supp_cap))
+        call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
         call("qk_q5_build_orders", sh, c_u64(self._ord_cols["o_orderkey"].n),
              self._ord_cols["o_orderkey"].ptr,
              self._ord_cols["o_custkey"].ptr,
              self._ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
-             c_u64(self.ord_cap), None)
+             c_u64(self.ord_cap), None, self.bloom.ptr,
+             c_u64(self.bloom_bits - 1))
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         sh = self.stream.handle if self.stream else None
         n = li_cols["l_orderkey"].n
-        call("qk_q5_probe_agg_nt" if nt else "qk_q5_probe_agg", sh,
+        if nt:
+            call("qk_q5_probe_agg_nt", sh, c_u64(n),
+                 li_cols["l_orderkey"].ptr, li_cols["l_suppkey"].ptr,
+                 li_cols["l_extendedprice"].ptr, li_cols["l_discount"].ptr,
+                 self.ord_keys.ptr, self.ord_val.ptr, c_u64(self.ord_cap),
+                 self.supp_keys.ptr, self.supp_val.ptr,
+                 c_u64(self.supp_cap), self.out25.ptr,
+                 match_count_buf.ptr if match_count_buf else None,
+                 self.bloom.ptr, c_u64(self.bloom_bits - 1))
+            return
+        call("qk_q5_probe_agg", sh,
              c_u64(n), li_cols["l_orderkey"].ptr,
              li_cols["l_suppkey"].ptr, li_cols["l_extendedprice"].ptr,
              li_cols["l_discount"].ptr, self.ord_keys.ptr,
@@ -544,7 +577,7 @@ class Q5Fused:
 
     def free(self):
         for c in (self.cust_keys, self.cust_val, self.supp_keys,
-                  self.supp_val, self.ord_keys, self.ord_val):
+                  self.supp_val, self.ord_keys, self.ord_val, self.bloom):
             c.free()
         self.out25.free()
 

@@ -98,3 +98,23 @@ def test_aggexec_sql_parsing():
     s = {"e0_agg_0": np.array([2.0]), "e3_agg_0": np.array([10.0]),
          "e3_agg_1": np.array([4.0])}
     assert eval(a.exprs[1][1], {"s": s})[0] == 2.5
+
+
+def test_topk_executor_cpu():
+    """GPUTopKExecutor (ConcatThenSQLExecutor mirror) is pure host logic
+    over tiny partials -> testable without a GPU."""
+    import numpy as np
+    import pyarrow as pa
+    from quokka_amd import GPUTopKExecutor
+    rng = np.random.default_rng(3)
+    rev = rng.random(1000)
+    date = rng.integers(0, 100, 1000)
+    t = pa.table({"revenue": rev, "o_orderdate": date,
+                  "k": np.arange(1000)})
+    ex = GPUTopKExecutor(["revenue", "o_orderdate"], 10,
+                         descending=[True, False])
+    ex.execute([t.slice(0, 500)], 0, 0)
+    ex.execute([t.slice(500)], 0, 0)
+    out = ex.done(0)
+    order = np.lexsort((date, -rev))[:10]
+    assert out.column("k").to_pylist() == list(order)
