@@ -249,17 +249,18 @@ def test_groupby_parity(gpu):
 
 # ---------- partition ---------------------------------------------------
 
-def test_partition_parity_bitexact(gpu):
+@pytest.mark.parametrize("nparts", [1, 3, 8, 512])
+def test_partition_parity_bitexact(gpu, nparts):
     from quokka_amd import ops, shim
     rng = np.random.default_rng(23)
     keys = rng.integers(0, 1 << 40, 300000).astype(np.int64)
     kcol = shim.DevColumn.from_numpy(keys)
-    offsets, idx = ops.partition_i64(kcol, 8)
+    offsets, idx = ops.partition_i64(kcol, nparts)
     sel = idx.to_numpy(len(keys))
-    want = OE.partition_int(keys, 8)
-    hist = np.bincount(want, minlength=8)
+    want = OE.partition_int(keys, nparts)
+    hist = np.bincount(want, minlength=nparts)
     assert np.array_equal(np.diff(offsets.astype(np.int64)), hist)
-    for p in range(8):
+    for p in range(nparts):
         rows = sel[int(offsets[p]):int(offsets[p + 1])]
         assert np.all(want[rows] == p)            # right bucket
     assert len(np.unique(sel)) == len(keys)       # a permutation
