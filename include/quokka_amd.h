@@ -259,10 +259,15 @@ int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
  * pre-filled with QK_JOIN_EMPTY, slot_sums zeroed. Repeated calls
  * accumulate. nvals value columns share one key lookup (vals/slot_sums are
  * arrays-of-columns, each `capacity` doubles: slot_sums[c*capacity+slot]). */
+/* agg_ops_dev (device i32[nvals], nullable -> all SUM): per value column
+ * 0 = SUM (slot init 0), 1 = MIN (init +inf), 2 = MAX (init -inf) — the
+ * distributive ops the two-phase rewrite emits (sql_utils.py:299-413). */
 int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
-                       const double *const *vals_dev, int nvals,
+                       const double *const *vals_dev,
+                       const int32_t *agg_ops_dev, int nvals,
                        int64_t *slot_keys, double *slot_sums,
                        uint64_t capacity);
+int qk_fill_f64(void *stream, double *dst_dev, double value, uint64_t n);
 /* Compact occupied slots to out_keys/out_sums (unordered); out_cursor_dev
  * (u64, zeroed) = number of groups. out capacity must be >= group count. */
 int qk_groupby_extract(void *stream, const int64_t *slot_keys,

@@ -1404,9 +1404,33 @@ extern "C" int qk_gen_supplier(void *stream, uint64_t n, uint64_t row_offset,
 }
 
 // ---- group-by i64 -> f64 sums ----------------------------------------
+// f64 atomic min/max via CAS on the bit pattern (ordered-compare loop)
+__device__ inline void atomic_min_f64(double *addr, double v) {
+  unsigned long long *a = (unsigned long long *)addr;
+  unsigned long long cur = *a;
+  while (__longlong_as_double((long long)cur) > v) {
+    unsigned long long prev =
+        atomicCAS(a, cur, (unsigned long long)__double_as_longlong(v));
+    if (prev == cur) break;
+    cur = prev;
+  }
+}
+__device__ inline void atomic_max_f64(double *addr, double v) {
+  unsigned long long *a = (unsigned long long *)addr;
+  unsigned long long cur = *a;
+  while (__longlong_as_double((long long)cur) < v) {
+    unsigned long long prev =
+        atomicCAS(a, cur, (unsigned long long)__double_as_longlong(v));
+    if (prev == cur) break;
+    cur = prev;
+  }
+}
+
+// agg_ops[c]: 0 = SUM (slot init 0), 1 = MIN (init +inf), 2 = MAX (-inf)
 __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
     uint64_t n, const int64_t *__restrict__ keys, const double *const *vals,
-    int nvals, int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
+    const int32_t *__restrict__ agg_ops, int nvals,
+    int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
     uint64_t cap) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -1424,12 +1448,22 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
       }
       s = (s + 1) & (cap - 1);
     }
-    for (int c = 0; c < nvals; c++)
-      atomicAdd(&slot_sums[(uint64_t)c * cap + s], vals[c][i]);
+    for (int c = 0; c < nvals; c++) {
+      double *dst = &slot_sums[(uint64_t)c * cap + s];
+      double v = vals[c][i];
+      int op = agg_ops ? agg_ops[c] : 0;
+      if (op == 1)
+        atomic_min_f64(dst, v);
+      else if (op == 2)
+        atomic_max_f64(dst, v);
+      else
+        atomicAdd(dst, v);
+    }
   }
 }
 extern "C" int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
-                                  const double *const *vals_dev, int nvals,
+                                  const double *const *vals_dev,
+                                  const int32_t *agg_ops_dev, int nvals,
                                   int64_t *slot_keys, double *slot_sums,
                                   uint64_t cap) {
   if (!n) return 0;
@@ -1437,9 +1471,24 @@ extern "C" int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
     return qk_fail("qk_groupby_i64_sum.cap_pow2", hipErrorInvalidValue);
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_groupby_sum, dim3(blocks), dim3(BLOCK), 0,
-                     (hipStream_t)stream, n, keys, vals_dev, nvals, slot_keys,
-                     slot_sums, cap);
+                     (hipStream_t)stream, n, keys, vals_dev, agg_ops_dev,
+                     nvals, slot_keys, slot_sums, cap);
   QK_TRY("qk_groupby_i64_sum", hipGetLastError());
+  return 0;
+}
+
+__global__ void k_fill_f64(double *dst, double v, uint64_t n) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dst[i] = v;
+}
+extern "C" int qk_fill_f64(void *stream, double *dst, double v, uint64_t n) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_fill_f64, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, dst, v, n);
+  QK_TRY("qk_fill_f64", hipGetLastError());
   return 0;
 }
 

@@ -77,12 +77,6 @@ class GPUBuildProbeJoinExecutor(Executor):
         d = dict(self.__dict__)
         return d
 
-    def _ensure_stream(self):
-        ops, shim, staging = _lazy_gpu()
-        if self._stream is None:
-            self._stream = ops and None  # stream created lazily below
-        return ops, shim, staging
-
     def execute(self, batches, stream_id, executor_id):
         import pyarrow as pa
         ops, shim, staging = _lazy_gpu()
@@ -191,7 +185,9 @@ class GPUBuildProbeJoinExecutor(Executor):
         pass
 
 
-_SUM_RE = re.compile(r"sum\s*\(\s*([A-Za-z_][A-Za-z0-9_]*)\s*\)", re.I)
+_AGG_RE = re.compile(r"(sum|min|max)\s*\(\s*([A-Za-z_][A-Za-z0-9_]*)\s*\)",
+                     re.I)
+_AGG_OP = {"sum": 0, "min": 1, "max": 2}
 
 
 class GPUAggExecutor(Executor):
@@ -201,7 +197,8 @@ class GPUAggExecutor(Executor):
     (key, 'asc'|'desc')), sql_statement = the FINAL aggregate clause the
     two-phase rewrite produces (sql_utils.py:379-413) — expressions over
     SUM(partial_col) terms with aliases, e.g.
-    "sum(e0_agg_0) as revenue" or "sum(e3_agg_0) / sum(e3_agg_1) as avg_x".
+    "sum(e0_agg_0) as revenue" or "sum(e3_agg_0) / sum(e3_agg_1) as avg_x";
+    MIN(col)/MAX(col) partials are supported alongside SUM.
 
     execute() accumulates partial rows into a device group-by hash table
     (single i64 group key, or multiple keys composite-encoded — DESIGN.md);
@@ -214,7 +211,8 @@ class GPUAggExecutor(Executor):
         self.groupby_keys = groupby_keys
         self.orderby_keys = orderby_keys or []
         self.sql_statement = sql_statement
-        self.sum_cols = []          # partial columns referenced by SUM()
+        self.sum_cols = []          # partial columns referenced by aggs
+        self.agg_ops = []           # per column: 0 SUM / 1 MIN / 2 MAX
         self.exprs = []             # (alias, python expr over s['col'])
         for part in self._split_top(sql_statement):
             alias = None
@@ -223,14 +221,19 @@ class GPUAggExecutor(Executor):
             if m:
                 alias = m.group(1)
                 part = part[: m.start()]
-            cols = _SUM_RE.findall(part)
-            if not cols:
-                raise ValueError("unsupported aggregate (round 1 supports "
-                                 "expressions over SUM(col)): %r" % part)
-            for c in cols:
+            hits = _AGG_RE.findall(part)
+            if not hits:
+                raise ValueError("unsupported aggregate (supported: "
+                                 "expressions over SUM/MIN/MAX(col)): %r"
+                                 % part)
+            for fn, c in hits:
                 if c not in self.sum_cols:
                     self.sum_cols.append(c)
-            expr = _SUM_RE.sub(lambda m: "s[%r]" % m.group(1), part)
+                    self.agg_ops.append(_AGG_OP[fn.lower()])
+                elif self.agg_ops[self.sum_cols.index(c)] != _AGG_OP[fn.lower()]:
+                    raise ValueError("column %r used with two different "
+                                     "aggregate functions" % c)
+            expr = _AGG_RE.sub(lambda m: "s[%r]" % m.group(2), part)
             self.exprs.append((alias or part.strip(), expr))
         self._gb = None
         self._key_state = None
@@ -322,7 +325,8 @@ class GPUAggExecutor(Executor):
             keys = np.zeros(len(batch), dtype=np.int64)
         if self._gb is None:
             self._gb = ops.GroupByI64(
-                expected_groups=max(1024, len(batch)), nvals=len(self.sum_cols))
+                expected_groups=max(1024, len(batch)),
+                nvals=len(self.sum_cols), agg_ops=self.agg_ops)
         kcol = shim.DevColumn.from_numpy(keys)
         vcols = [shim.DevColumn.from_numpy(
             staging.column_to_numpy(batch.column(c)).astype(np.float64))

@@ -159,17 +159,30 @@ class GroupByI64:
     columns (SQLAggExecutor's post-rewrite distributive form,
     sql_utils.py:299-413). update() per batch, extract() at done()."""
 
-    def __init__(self, expected_groups, nvals, stream=None):
+    AGG_INIT = {0: 0.0, 1: float("inf"), 2: float("-inf")}  # SUM/MIN/MAX
+
+    def __init__(self, expected_groups, nvals, stream=None, agg_ops=None):
+        """agg_ops: per value column 0=SUM, 1=MIN, 2=MAX (default all SUM),
+        the distributive ops of the two-phase rewrite (sql_utils.py)."""
         self.cap = _pow2_at_least(max(16, 2 * int(expected_groups)))
         self.nvals = nvals
+        self.agg_ops = list(agg_ops) if agg_ops else [0] * nvals
+        assert len(self.agg_ops) == nvals
         self.stream = stream
         self.slot_keys = DevColumn(np.int64, self.cap)
         self.slot_sums = DevColumn(np.float64, self.cap * nvals)
         sh = stream.handle if stream else None
         shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
                   c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
-        shim.call("qk_dmemset", self.slot_sums.ptr, 0,
-                  c_u64(self.cap * nvals * 8))
+        import ctypes as _ct
+        for c, op in enumerate(self.agg_ops):
+            shim.call("qk_fill_f64", sh,
+                      shim.c_vp(self.slot_sums.ptr.value + c * self.cap * 8),
+                      _ct.c_double(self.AGG_INIT[op]), c_u64(self.cap))
+        self._ops_dev = None
+        if any(self.agg_ops):
+            self._ops_dev = DevColumn.from_numpy(
+                np.asarray(self.agg_ops, dtype=np.int32))
 
     def update(self, keys_col, val_cols, n=None):
         n = keys_col.n if n is None else n
@@ -183,7 +196,8 @@ class GroupByI64:
         shim.call("qk_h2d", dptrs.ptr, ptrs.ctypes.data_as(c_vp),
                   c_u64(ptrs.nbytes))
         shim.call("qk_groupby_i64_sum", sh, c_u64(n), keys_col.ptr,
-                  dptrs.ptr, self.nvals, self.slot_keys.ptr,
+                  dptrs.ptr, self._ops_dev.ptr if self._ops_dev else None,
+                  self.nvals, self.slot_keys.ptr,
                   self.slot_sums.ptr, c_u64(self.cap))
         if self.stream:
             self.stream.sync()
@@ -213,6 +227,8 @@ class GroupByI64:
     def free(self):
         self.slot_keys.free()
         self.slot_sums.free()
+        if self._ops_dev is not None:
+            self._ops_dev.free()
 
 
 def partition_i64(keys_col, nparts, stream=None, n=None):

@@ -649,3 +649,30 @@ def test_q3_after_exchange_world1(gpu, data):
     for cs in (lcols, ocols, ccols, li_x, od_x):
         for c in cs.values():
             c.free()
+
+
+def test_executor_agg_min_max(gpu):
+    """GPUAggExecutor with MIN/MAX partials (the other distributive ops
+    the two-phase rewrite emits, sql_utils.py:299-413)."""
+    import pyarrow as pa
+    from quokka_amd import GPUAggExecutor
+    rng = np.random.default_rng(51)
+    g = rng.integers(0, 40, 4000).astype(np.int64)
+    x = rng.random(4000)
+    y = rng.random(4000)
+    ex = GPUAggExecutor(["g"], [("g", "asc")],
+                        "min(e0_agg_0) as mn, max(e1_agg_0) as mx, "
+                        "sum(e0_agg_0) as sx")
+    t = pa.table({"g": g, "e0_agg_0": x, "e1_agg_0": y})
+    ex.execute([t.slice(0, 2000)], 0, 0)
+    ex.execute([t.slice(2000)], 0, 0)
+    out = ex.done(0)
+    uk = np.unique(g)
+    assert out.column("g").to_pylist() == uk.tolist()
+    want_mn = np.array([x[g == k].min() for k in uk])
+    want_mx = np.array([y[g == k].max() for k in uk])
+    want_sx = np.array([x[g == k].sum() for k in uk])
+    np.testing.assert_allclose(out.column("mn").to_numpy(), want_mn, rtol=0)
+    np.testing.assert_allclose(out.column("mx").to_numpy(), want_mx, rtol=0)
+    np.testing.assert_allclose(out.column("sx").to_numpy(), want_sx,
+                               rtol=1e-9)
