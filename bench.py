@@ -494,7 +494,10 @@ def main():
         dist.init_process_group("gloo", rank=rank, world_size=world)
 
     from quokka_amd import shim, ops, queries as DQ
-    shim.init(local_rank)
+    # one rank per GPU; oversubscribe gracefully when ranks > devices
+    # (lets the torchrun path be smoke-tested on a 1-GPU box)
+    ndev = shim.device_count()
+    shim.init(local_rank % max(1, ndev))
 
     global QUERY
     QUERY = args.query
