@@ -647,21 +647,27 @@ __device__ inline uint64_t slot_of(int64_t key, uint64_t cap) {
   return splitmix64((uint64_t)key) & (cap - 1);
 }
 
-// Bloom prefilter (k=2 bits from one splitmix64): ~95% of probe misses
-// skip the table-line read entirely. bloom_mask = bit count - 1 (pow2).
+// Blocked Bloom prefilter: k=2 bits in ONE u32 word (one 4 B read per
+// probe, one cache line touched) — ~94% of probe misses skip the
+// table-line read. bloom_mask = bit count - 1 (pow2; words = bits/32).
+__device__ inline uint32_t bloom_word_mask(int64_t key, uint64_t bloom_mask,
+                                           uint64_t *word_idx) {
+  uint64_t h = splitmix64((uint64_t)key ^ 0xB10011B10011B100ULL);
+  *word_idx = (h & bloom_mask) >> 5;
+  uint32_t b1 = (uint32_t)(h >> 40) & 31u, b2 = (uint32_t)(h >> 48) & 31u;
+  return (1u << b1) | (1u << b2);
+}
 __device__ inline void bloom_set(uint32_t *bloom, uint64_t bloom_mask,
                                  int64_t key) {
-  uint64_t h = splitmix64((uint64_t)key ^ 0xB10011B10011B100ULL);
-  uint64_t b1 = h & bloom_mask, b2 = (h >> 32) & bloom_mask;
-  atomicOr(&bloom[b1 >> 5], 1u << (b1 & 31));
-  atomicOr(&bloom[b2 >> 5], 1u << (b2 & 31));
+  uint64_t w;
+  uint32_t m = bloom_word_mask(key, bloom_mask, &w);
+  atomicOr(&bloom[w], m);
 }
 __device__ inline bool bloom_test(const uint32_t *bloom, uint64_t bloom_mask,
                                   int64_t key) {
-  uint64_t h = splitmix64((uint64_t)key ^ 0xB10011B10011B100ULL);
-  uint64_t b1 = h & bloom_mask, b2 = (h >> 32) & bloom_mask;
-  if (!((bloom[b1 >> 5] >> (b1 & 31)) & 1u)) return false;
-  return ((bloom[b2 >> 5] >> (b2 & 31)) & 1u) != 0;
+  uint64_t w;
+  uint32_t m = bloom_word_mask(key, bloom_mask, &w);
+  return (bloom[w] & m) == m;
 }
 
 __global__ void __launch_bounds__(BLOCK) k_join_build2(
