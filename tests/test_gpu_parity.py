@@ -660,10 +660,12 @@ def test_executor_agg_min_max(gpu):
     g = rng.integers(0, 40, 4000).astype(np.int64)
     x = rng.random(4000)
     y = rng.random(4000)
+    # distinct partial columns per aggregate, as the reference's rewrite
+    # emits (sql_utils.py:379-413 assigns each partial a unique alias)
     ex = GPUAggExecutor(["g"], [("g", "asc")],
                         "min(e0_agg_0) as mn, max(e1_agg_0) as mx, "
-                        "sum(e0_agg_0) as sx")
-    t = pa.table({"g": g, "e0_agg_0": x, "e1_agg_0": y})
+                        "sum(e2_agg_0) as sx")
+    t = pa.table({"g": g, "e0_agg_0": x, "e1_agg_0": y, "e2_agg_0": x})
     ex.execute([t.slice(0, 2000)], 0, 0)
     ex.execute([t.slice(2000)], 0, 0)
     out = ex.done(0)
