@@ -18,7 +18,6 @@ Prints ONE JSON line from rank 0 (driver contract), including:
                   timed on this box's host cores over a bounded sample
 """
 import argparse
-import ctypes
 import json
 import os
 import sys

@@ -66,11 +66,10 @@ class GPUBuildProbeJoinExecutor(Executor):
         self.how = how
         self.key_to_keep = key_to_keep
         self.phase = "build"
-        # lazy GPU state
+        # lazy GPU state (created on first execute, never in __init__)
         self._table = None
         self._build_cols = None     # dict name -> list of numpy arrays
         self._build_names = None
-        self._stream = None
 
     def __getstate__(self):
         assert self._table is None, "executor must be pickled before first execute"

@@ -1,9 +1,10 @@
-"""Device query pipelines (TPC-H Q1/Q6/Q3) composed from ops.
+"""Device query pipelines (TPC-H Q1/Q3/Q5/Q6) composed from ops.
 
 These mirror how the reference lowers each query:
 Q1 = scan -> fused filter+partial group-by (map side) -> tiny final agg;
-Q3 = filter customer/orders -> hash-join build x2 -> probe lineitem ->
-     group-by orderkey -> top-k on host;
+Q3 = customer/orders filtered builds -> fused lineitem probe+group-by
+     (Q3Fused; a composable q3() pipeline exists for parity);
+Q5 = three key->nation tables -> fused double-probe + per-nation agg;
 Q6 = scan -> fused filter+sum.
 
 Date constants are days-since-1970 (Arrow date32); the literals are the
