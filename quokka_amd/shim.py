@@ -146,18 +146,64 @@ _DTYPE_GATHER = {
 }
 
 
+class _DevPool:
+    """Size-exact free-list for device buffers. Allocation patterns here
+    repeat identically across bench steps (same column sizes every step),
+    so recycling by exact size removes the hipMalloc/hipFree churn that
+    dominated the per-step exchange path (~17 GB realloc/step). Capped;
+    flushed on OOM so a fresh hipMalloc can succeed."""
+
+    def __init__(self, cap_bytes=None):
+        if cap_bytes is None:
+            cap_bytes = int(os.environ.get("QK_POOL_CAP_GB", "24")) << 30
+        self.cap = cap_bytes
+        self.cached = 0
+        self.buckets = {}
+
+    def get(self, nbytes):
+        lst = self.buckets.get(nbytes)
+        if lst:
+            self.cached -= nbytes
+            return lst.pop()
+        return None
+
+    def put(self, nbytes, ptr):
+        if self.cached + nbytes > self.cap:
+            return False
+        self.buckets.setdefault(nbytes, []).append(ptr)
+        self.cached += nbytes
+        return True
+
+    def flush(self):
+        for nb, lst in self.buckets.items():
+            for p in lst:
+                _lib.qk_dfree(p)
+        self.buckets.clear()
+        self.cached = 0
+
+
+_pool = _DevPool()
+
+
 class DevBuffer:
-    """Owning device allocation."""
+    """Owning device allocation (recycled through _pool)."""
 
     def __init__(self, nbytes):
-        p = c_vp(0)
-        call("qk_dmalloc", c_u64(max(1, nbytes)), ctypes.byref(p))
+        nbytes = max(1, nbytes)
+        p = _pool.get(nbytes)
+        if p is None:
+            p = c_vp(0)
+            rc = _lib.qk_dmalloc(c_u64(nbytes), ctypes.byref(p))
+            if rc != 0:          # OOM: flush the pool and retry once
+                _pool.flush()
+                call("qk_dmalloc", c_u64(nbytes), ctypes.byref(p))
         self.ptr = p
         self.nbytes = nbytes
 
     def free(self):
         if self.ptr is not None and self.ptr.value:
-            call("qk_dfree", self.ptr)
+            if not _pool.put(self.nbytes, self.ptr):
+                call("qk_dfree", self.ptr)
             self.ptr = None
 
     def __del__(self):
