@@ -1014,10 +1014,12 @@ __global__ void __launch_bounds__(BLOCK) k_q3_probe_agg_nt(
   for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
        p < npairs; p += stride) {
     uint64_t i = 2 * p;
-    int2 s2 = __builtin_nontemporal_load(
-        reinterpret_cast<const int2 *>(l_shipdate + i));
-    longlong2 k2 = __builtin_nontemporal_load(
-        reinterpret_cast<const longlong2 *>(l_orderkey + i));
+    typedef int v2i __attribute__((ext_vector_type(2)));
+    typedef long long v2l __attribute__((ext_vector_type(2)));
+    v2i s2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2i *>(l_shipdate + i));
+    v2l k2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2l *>(l_orderkey + i));
     bool pass0 = s2.x > date_gt, pass1 = s2.y > date_gt;
     if (bloom) {
       if (pass0) pass0 = bloom_test(bloom, bloom_mask, k2.x);
