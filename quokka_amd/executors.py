@@ -134,19 +134,30 @@ class GPUBuildProbeJoinExecutor(Executor):
         mode = {"inner": 0, "left": 0, "semi": 1, "anti": 2}[self.how]
         pidx, bidx, nm = self._table.probe(kcol, mode=mode)
 
+        def dev_gather(host_arr, idx_col):
+            """stage -> device gather by idx -> d2h (the payload gather is
+            the polars-join emit step, sql_executors.py:371)."""
+            src = shim.DevColumn.from_numpy(host_arr)
+            g = src.gather(idx_col, nm)
+            res = g.to_numpy(nm)
+            src.free()
+            g.free()
+            return res
+
         out = {}
         if self.how in ("semi", "anti"):
-            sel = pidx.to_numpy(nm)
             for c in batch.column_names:
-                out[c] = staging.column_to_numpy(batch.column(c))[sel]
+                out[c] = dev_gather(
+                    staging.column_to_numpy(batch.column(c)), pidx)
         else:
             sel = pidx.to_numpy(nm)
-            bsel = bidx.to_numpy(nm)
             for c in batch.column_names:
-                out[c] = staging.column_to_numpy(batch.column(c))[sel]
+                out[c] = dev_gather(
+                    staging.column_to_numpy(batch.column(c)), pidx)
             for c, dev in self._build_payload_dev.items():
-                host = self._host_build[c]
-                out[c] = host[bsel]
+                g = dev.gather(bidx, nm)
+                out[c] = g.to_numpy(nm)
+                g.free()
             if self.how == "left":
                 # unmatched probe rows appended with null build payload
                 matched = np.zeros(len(probe_keys), dtype=bool)
