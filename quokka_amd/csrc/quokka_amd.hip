@@ -1341,6 +1341,26 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg(
   if (match_count && threadIdx.x == 0 && lcnt)
     atomicAdd((unsigned long long *)match_count, (unsigned long long)lcnt);
 }
+__device__ inline void q5_probe_row(
+    int64_t okey, bool pass, uint64_t i,
+    const int64_t *__restrict__ l_suppkey, const double *__restrict__ l_price,
+    const double *__restrict__ l_disc, const int64_t *__restrict__ ord_keys,
+    const int32_t *__restrict__ ord_val, uint64_t ord_cap,
+    const int64_t *__restrict__ supp_keys, const int32_t *__restrict__ supp_val,
+    uint64_t supp_cap, double *lsum, uint32_t *lcnt, bool count) {
+  if (!pass) return;
+  int32_t cnat = probe_unique(ord_keys, ord_val, ord_cap, okey, nullptr);
+  if (cnat < 0) return;
+  int64_t skey = __builtin_nontemporal_load(&l_suppkey[i]);
+  int32_t snat = probe_unique(supp_keys, supp_val, supp_cap, skey, nullptr);
+  if (snat != cnat) return;
+  double price = __builtin_nontemporal_load(&l_price[i]);
+  double disc = __builtin_nontemporal_load(&l_disc[i]);
+  atomicAdd(&lsum[cnat & 31], price * (1.0 - disc));
+  if (count) atomicAdd(lcnt, 1u);
+}
+
+// 2 rows/thread (16 B/lane key loads, two bloom/table lookups in flight)
 __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
     uint64_t n, const int64_t *__restrict__ l_orderkey,
     const int64_t *__restrict__ l_suppkey,
@@ -1355,20 +1375,31 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
   if (threadIdx.x < 32) lsum[threadIdx.x] = 0.0;
   if (threadIdx.x == 0) lcnt = 0;
   __syncthreads();
+  bool count = match_count != nullptr;
+  uint64_t npairs = n / 2;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int64_t okey = __builtin_nontemporal_load(&l_orderkey[i]);
-    if (bloom && !bloom_test(bloom, bloom_mask, okey)) continue;
-    int32_t cnat = probe_unique(ord_keys, ord_val, ord_cap, okey, nullptr);
-    if (cnat < 0) continue;
-    int64_t skey = __builtin_nontemporal_load(&l_suppkey[i]);
-    int32_t snat = probe_unique(supp_keys, supp_val, supp_cap, skey, nullptr);
-    if (snat != cnat) continue;
-    double price = __builtin_nontemporal_load(&l_price[i]);
-    double disc = __builtin_nontemporal_load(&l_disc[i]);
-    atomicAdd(&lsum[cnat & 31], price * (1.0 - disc));
-    if (match_count) atomicAdd(&lcnt, 1u);
+  for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       p < npairs; p += stride) {
+    uint64_t i = 2 * p;
+    typedef long long v2l __attribute__((ext_vector_type(2)));
+    v2l k2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2l *>(l_orderkey + i));
+    bool pass0 = !bloom || bloom_test(bloom, bloom_mask, k2.x);
+    bool pass1 = !bloom || bloom_test(bloom, bloom_mask, k2.y);
+    q5_probe_row(k2.x, pass0, i, l_suppkey, l_price, l_disc, ord_keys,
+                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
+                 &lcnt, count);
+    q5_probe_row(k2.y, pass1, i + 1, l_suppkey, l_price, l_disc, ord_keys,
+                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
+                 &lcnt, count);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    uint64_t i = n - 1;
+    int64_t okey = l_orderkey[i];
+    bool pass = !bloom || bloom_test(bloom, bloom_mask, okey);
+    q5_probe_row(okey, pass, i, l_suppkey, l_price, l_disc, ord_keys,
+                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
+                 &lcnt, count);
   }
   __syncthreads();
   if (threadIdx.x < 32 && lsum[threadIdx.x] != 0.0)
