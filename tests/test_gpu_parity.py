@@ -678,3 +678,84 @@ def test_executor_agg_min_max(gpu):
     np.testing.assert_allclose(out.column("mx").to_numpy(), want_mx, rtol=0)
     np.testing.assert_allclose(out.column("sx").to_numpy(), want_sx,
                                rtol=1e-9)
+
+
+def test_executor_agg_q1_composite_keys(gpu, data):
+    """Q1's final aggregate through the GENERIC GPUAggExecutor with the
+    reference's two-string group keys and rewritten final-agg SQL
+    (datastream.py:1819-1856 lowering; avg -> sum/sum per
+    sql_utils.py:299-413) — composite key encode/decode path."""
+    import pyarrow as pa
+    from quokka_amd import GPUAggExecutor
+    li = data["lineitem"]
+    # map-side partials per chunk, as the folded batch_funcs produce them
+    n = len(li["l_orderkey"])
+    rows = {"l_returnflag": [], "l_linestatus": [], "e0": [], "e1": [],
+            "e2": [], "e3": [], "cnt": []}
+    for lo in range(0, n, 9973):
+        chunk = {k: v[lo:lo + 9973] for k, v in li.items()}
+        p = Q_partials(chunk)
+        for g in range(6):
+            if p[g, 5] > 0:
+                rows["l_returnflag"].append(G.RETURNFLAG[g // 2])
+                rows["l_linestatus"].append(G.LINESTATUS[g % 2])
+                rows["e0"].append(p[g, 0])
+                rows["e1"].append(p[g, 1])
+                rows["e2"].append(p[g, 2])
+                rows["e3"].append(p[g, 3])
+                rows["cnt"].append(p[g, 5])
+    t = pa.table({k: (np.array(v) if k not in ("l_returnflag",
+                                               "l_linestatus")
+                      else np.array(v, dtype=object).astype(str))
+                  for k, v in rows.items()})
+    ex = GPUAggExecutor(
+        ["l_returnflag", "l_linestatus"],
+        [("l_returnflag", "asc"), ("l_linestatus", "asc")],
+        "sum(e0) as sum_qty, sum(e1) as sum_base_price, "
+        "sum(e2) as sum_disc_price, sum(e3) as sum_charge, "
+        "sum(e0) / sum(cnt) as avg_qty, sum(cnt) as count_order")
+    half = len(t) // 2
+    ex.execute([t.slice(0, half)], 0, 0)
+    ex.execute([t.slice(half)], 0, 0)
+    out = ex.done(0)
+    want = OQ.q1(li)
+    assert out.column("l_returnflag").to_pylist() == \
+        list(want["l_returnflag"])
+    assert out.column("l_linestatus").to_pylist() == \
+        list(want["l_linestatus"])
+    np.testing.assert_allclose(out.column("sum_qty").to_numpy(),
+                               want["sum_qty"], rtol=1e-9)
+    np.testing.assert_allclose(out.column("avg_qty").to_numpy(),
+                               want["avg_qty"], rtol=1e-9)
+    np.testing.assert_allclose(out.column("count_order").to_numpy(),
+                               want["count_order"].astype(float), rtol=0)
+
+
+def Q_partials(chunk):
+    return OQ.q1_partials(chunk)
+
+
+def test_gpu_partition_fn_plugin_api(gpu):
+    """gpu_partition_fn through the registration-time contract
+    (core.py:152-206): pyarrow table in, dict channel -> table out,
+    int-key buckets bit-exact with quokka_runtime.py:222."""
+    import pyarrow as pa
+    from quokka_amd import gpu_partition_fn
+    rng = np.random.default_rng(61)
+    keys = rng.integers(0, 100000, 20000).astype(np.int64)
+    vals = rng.random(20000)
+    t = pa.table({"k": keys, "v": vals})
+    out = gpu_partition_fn(t, source_channel=0, num_target_channels=4,
+                           key="k")
+    want = OE.partition_int(keys, 4)
+    total = 0
+    for ch, tbl in out.items():
+        got_k = np.asarray(tbl.column("k").to_numpy())
+        assert np.all(got_k % 4 == ch)
+        total += len(tbl)
+        # payload rows stay attached to their keys
+        got_v = np.asarray(tbl.column("v").to_numpy())
+        lookup = {(k_, round(v_, 12)) for k_, v_ in zip(keys, vals)}
+        assert all((k_, round(v_, 12)) in lookup
+                   for k_, v_ in zip(got_k[:50], got_v[:50]))
+    assert total == len(keys)
