@@ -171,7 +171,8 @@ int qk_join_probe(void *stream, uint64_t n_probe, const int64_t *keys,
  * slots for the probe aggregate. Tables pre-filled like qk_join_build's. */
 int qk_build_u8eq(void *stream, uint64_t n, const int64_t *keys,
                   const uint8_t *flag, uint8_t flag_val, int64_t *slot_keys,
-                  int32_t *slot_head, uint64_t capacity);
+                  int32_t *slot_head, uint64_t capacity, uint32_t *bloom,
+                  uint64_t bloom_mask);
 /* bloom/bloom_mask (nullable/0): optional Bloom prefilter the probe-side
  * kernels test before the table walk (bits = pow2, mask = bits-1). */
 int qk_q3_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
@@ -180,13 +181,15 @@ int qk_q3_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
                        const int32_t *cust_head, uint64_t cust_cap,
                        int64_t *slot_keys, int32_t *slot_head,
                        uint64_t capacity, uint32_t *bloom,
-                       uint64_t bloom_mask);
+                       uint64_t bloom_mask, const uint32_t *cust_bloom,
+                       uint64_t cust_bloom_mask);
 /* Count the rows qk_q3_build_orders would insert (for tight table sizing;
  * count_dev u64, zeroed). */
 int qk_q3_count_orders(void *stream, uint64_t n, const int64_t *o_custkey,
                        const int32_t *o_orderdate, int32_t date_lt,
                        const int64_t *cust_keys, const int32_t *cust_head,
-                       uint64_t cust_cap, uint64_t *count_dev);
+                       uint64_t cust_cap, uint64_t *count_dev,
+                       const uint32_t *cust_bloom, uint64_t cust_bloom_mask);
 int qk_q3_probe_agg(void *stream, uint64_t n, const int64_t *l_orderkey,
                     const int32_t *l_shipdate, const double *l_price,
                     const double *l_disc, int32_t date_gt,
@@ -219,7 +222,8 @@ int qk_q3_extract(void *stream, const int64_t *slot_keys,
 int qk_build_keyval_i32(void *stream, uint64_t n, const int64_t *keys,
                         const int32_t *vals, uint32_t accept_mask,
                         int64_t *slot_keys, int32_t *slot_val,
-                        uint64_t capacity);
+                        uint64_t capacity, uint32_t *bloom,
+                        uint64_t bloom_mask);
 /* Orders in [date_lo, date_hi) whose o_custkey hits the customer table:
  * insert o_orderkey -> customer nationkey. slot_keys == NULL -> count-only
  * (count_dev gets the survivor count either way when non-NULL). */
@@ -230,7 +234,8 @@ int qk_q5_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
                        uint64_t cust_cap, int64_t *slot_keys,
                        int32_t *slot_val, uint64_t capacity,
                        uint64_t *count_dev, uint32_t *bloom,
-                       uint64_t bloom_mask);
+                       uint64_t bloom_mask, const uint32_t *cust_bloom,
+                       uint64_t cust_bloom_mask);
 /* Fused probe: join lineitem to orders (-> customer nation) and supplier
  * (-> supplier nation); where equal accumulate revenue into out25[nation].
  * out25: f64[32], zeroed (slots 25..31 unused). */

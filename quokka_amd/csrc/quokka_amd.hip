@@ -788,7 +788,7 @@ __global__ void __launch_bounds__(BLOCK) k_build_u8eq(
     uint64_t n, const int64_t *__restrict__ keys,
     const uint8_t *__restrict__ flag, uint8_t flag_val,
     int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
-    uint64_t cap) {
+    uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -807,18 +807,20 @@ __global__ void __launch_bounds__(BLOCK) k_build_u8eq(
       s = (s + 1) & (cap - 1);
     }
     slot_head[s] = (int32_t)i;
+    if (bloom) bloom_set(bloom, bloom_mask, key);
   }
 }
 extern "C" int qk_build_u8eq(void *stream, uint64_t n, const int64_t *keys,
                              const uint8_t *flag, uint8_t flag_val,
                              int64_t *slot_keys, int32_t *slot_head,
-                             uint64_t cap) {
+                             uint64_t cap, uint32_t *bloom,
+                             uint64_t bloom_mask) {
   if (!n) return 0;
   if (cap & (cap - 1)) return qk_fail("qk_build_u8eq.cap_pow2", hipErrorInvalidValue);
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_build_u8eq, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, keys, flag, flag_val, slot_keys,
-                     slot_head, cap);
+                     slot_head, cap, bloom, bloom_mask);
   QK_TRY("qk_build_u8eq", hipGetLastError());
   return 0;
 }
@@ -848,14 +850,16 @@ __global__ void __launch_bounds__(BLOCK) k_q3_build_orders(
     const int64_t *__restrict__ cust_keys,
     const int32_t *__restrict__ cust_head, uint64_t cust_cap,
     int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
-    uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
+    uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask,
+    const uint32_t *__restrict__ cbloom, uint64_t cbloom_mask) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
     // nt loads: stream the orders columns without evicting the tables
     if (__builtin_nontemporal_load(&o_orderdate[i]) >= date_lt) continue;
-    if (probe_unique(cust_keys, cust_head, cust_cap,
-                     __builtin_nontemporal_load(&o_custkey[i]), nullptr) < 0)
+    int64_t ck = __builtin_nontemporal_load(&o_custkey[i]);
+    if (cbloom && !bloom_test(cbloom, cbloom_mask, ck)) continue;
+    if (probe_unique(cust_keys, cust_head, cust_cap, ck, nullptr) < 0)
       continue;
     int64_t key = __builtin_nontemporal_load(&o_orderkey[i]);
     uint64_t s = slot_of(key, cap);
@@ -882,7 +886,9 @@ extern "C" int qk_q3_build_orders(void *stream, uint64_t n,
                                   const int32_t *cust_head, uint64_t cust_cap,
                                   int64_t *slot_keys, int32_t *slot_head,
                                   uint64_t cap, uint32_t *bloom,
-                                  uint64_t bloom_mask) {
+                                  uint64_t bloom_mask,
+                                  const uint32_t *cbloom,
+                                  uint64_t cbloom_mask) {
   if (!n) return 0;
   if ((cap & (cap - 1)) || (cust_cap & (cust_cap - 1)))
     return qk_fail("qk_q3_build_orders.cap_pow2", hipErrorInvalidValue);
@@ -890,7 +896,8 @@ extern "C" int qk_q3_build_orders(void *stream, uint64_t n,
   hipLaunchKernelGGL(k_q3_build_orders, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, o_orderkey, o_custkey,
                      o_orderdate, date_lt, cust_keys, cust_head, cust_cap,
-                     slot_keys, slot_head, cap, bloom, bloom_mask);
+                     slot_keys, slot_head, cap, bloom, bloom_mask, cbloom,
+                     cbloom_mask);
   QK_TRY("qk_q3_build_orders", hipGetLastError());
   return 0;
 }
@@ -902,13 +909,16 @@ __global__ void __launch_bounds__(BLOCK) k_q3_count_orders(
     const int32_t *__restrict__ o_orderdate, int32_t date_lt,
     const int64_t *__restrict__ cust_keys,
     const int32_t *__restrict__ cust_head, uint64_t cust_cap,
-    uint64_t *__restrict__ count) {
+    uint64_t *__restrict__ count, const uint32_t *__restrict__ cbloom,
+    uint64_t cbloom_mask) {
   uint32_t cnt = 0;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    if (o_orderdate[i] >= date_lt) continue;
-    if (probe_unique(cust_keys, cust_head, cust_cap, o_custkey[i], nullptr) >= 0)
+    if (__builtin_nontemporal_load(&o_orderdate[i]) >= date_lt) continue;
+    int64_t ck = __builtin_nontemporal_load(&o_custkey[i]);
+    if (cbloom && !bloom_test(cbloom, cbloom_mask, ck)) continue;
+    if (probe_unique(cust_keys, cust_head, cust_cap, ck, nullptr) >= 0)
       cnt++;
   }
   __shared__ uint32_t lds[BLOCK / WAVE];
@@ -927,12 +937,14 @@ extern "C" int qk_q3_count_orders(void *stream, uint64_t n,
                                   const int32_t *o_orderdate, int32_t date_lt,
                                   const int64_t *cust_keys,
                                   const int32_t *cust_head, uint64_t cust_cap,
-                                  uint64_t *count_dev) {
+                                  uint64_t *count_dev, const uint32_t *cbloom,
+                                  uint64_t cbloom_mask) {
   if (!n) return 0;
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_q3_count_orders, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, o_custkey, o_orderdate, date_lt,
-                     cust_keys, cust_head, cust_cap, count_dev);
+                     cust_keys, cust_head, cust_cap, count_dev, cbloom,
+                     cbloom_mask);
   QK_TRY("qk_q3_count_orders", hipGetLastError());
   return 0;
 }
@@ -1194,7 +1206,7 @@ __global__ void __launch_bounds__(BLOCK) k_build_keyval_i32(
     uint64_t n, const int64_t *__restrict__ keys,
     const int32_t *__restrict__ vals, uint32_t accept_mask,
     int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_val,
-    uint64_t cap) {
+    uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -1216,19 +1228,21 @@ __global__ void __launch_bounds__(BLOCK) k_build_keyval_i32(
       s = (s + 1) & (cap - 1);
     }
     slot_val[s] = v;
+    if (bloom) bloom_set(bloom, bloom_mask, key);
   }
 }
 extern "C" int qk_build_keyval_i32(void *stream, uint64_t n,
                                    const int64_t *keys, const int32_t *vals,
                                    uint32_t accept_mask, int64_t *slot_keys,
-                                   int32_t *slot_val, uint64_t cap) {
+                                   int32_t *slot_val, uint64_t cap,
+                                   uint32_t *bloom, uint64_t bloom_mask) {
   if (!n) return 0;
   if (cap & (cap - 1))
     return qk_fail("qk_build_keyval_i32.cap_pow2", hipErrorInvalidValue);
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_build_keyval_i32, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, keys, vals, accept_mask,
-                     slot_keys, slot_val, cap);
+                     slot_keys, slot_val, cap, bloom, bloom_mask);
   QK_TRY("qk_build_keyval_i32", hipGetLastError());
   return 0;
 }
@@ -1244,15 +1258,17 @@ __global__ void __launch_bounds__(BLOCK) k_q5_build_orders(
     uint64_t cust_cap, int64_t *__restrict__ slot_keys,
     int32_t *__restrict__ slot_val, uint64_t cap,
     uint64_t *__restrict__ count, uint32_t *__restrict__ bloom,
-    uint64_t bloom_mask) {
+    uint64_t bloom_mask, const uint32_t *__restrict__ cbloom,
+    uint64_t cbloom_mask) {
   uint32_t cnt = 0;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    int32_t d = o_orderdate[i];
+    int32_t d = __builtin_nontemporal_load(&o_orderdate[i]);
     if (d < date_lo || d >= date_hi) continue;
-    int32_t nat = probe_unique(cust_keys, cust_val, cust_cap, o_custkey[i],
-                               nullptr);
+    int64_t ck = __builtin_nontemporal_load(&o_custkey[i]);
+    if (cbloom && !bloom_test(cbloom, cbloom_mask, ck)) continue;
+    int32_t nat = probe_unique(cust_keys, cust_val, cust_cap, ck, nullptr);
     if (nat < 0) continue;
     cnt++;
     if (!slot_keys) continue;
@@ -1293,7 +1309,9 @@ extern "C" int qk_q5_build_orders(void *stream, uint64_t n,
                                   const int32_t *cust_val, uint64_t cust_cap,
                                   int64_t *slot_keys, int32_t *slot_val,
                                   uint64_t cap, uint64_t *count_dev,
-                                  uint32_t *bloom, uint64_t bloom_mask) {
+                                  uint32_t *bloom, uint64_t bloom_mask,
+                                  const uint32_t *cbloom,
+                                  uint64_t cbloom_mask) {
   if (!n) return 0;
   if ((cust_cap & (cust_cap - 1)) || (slot_keys && (cap & (cap - 1))))
     return qk_fail("qk_q5_build_orders.cap_pow2", hipErrorInvalidValue);
@@ -1302,7 +1320,7 @@ extern "C" int qk_q5_build_orders(void *stream, uint64_t n,
                      (hipStream_t)stream, n, o_orderkey, o_custkey,
                      o_orderdate, date_lo, date_hi, cust_keys, cust_val,
                      cust_cap, slot_keys, slot_val, cap, count_dev, bloom,
-                     bloom_mask);
+                     bloom_mask, cbloom, cbloom_mask);
   QK_TRY("qk_q5_build_orders", hipGetLastError());
   return 0;
 }

@@ -205,19 +205,24 @@ class Q3Fused:
         self.cust_cap = ops._pow2_at_least(max(16, 2 * nbuild_cust))
         self.cust_keys = DevColumn(np.int64, self.cust_cap)
         self.cust_head = DevColumn(np.int32, self.cust_cap)
+        self.cbloom_bits = ops._pow2_at_least(max(1 << 16, 8 * nbuild_cust))
+        self.cbloom = DevColumn(np.uint32, self.cbloom_bits // 32)
+        call("qk_dmemset", self.cbloom.ptr, 0, c_u64(self.cbloom_bits // 8))
         call("qk_fill_i64", sh, self.cust_keys.ptr,
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cust_cap))
         call("qk_dmemset", self.cust_head.ptr, 0xFF, c_u64(self.cust_cap * 4))
         call("qk_build_u8eq", sh, c_u64(ncust), cust_cols["c_custkey"].ptr,
              cust_cols["c_mktsegment"].ptr, ctypes.c_uint8(MKT_BUILDING),
-             self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap))
+             self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         # tight orders-table sizing: count the fused-predicate survivors
         # first so probes stay cache-resident
         cnt = ops._count_buf()
         call("qk_q3_count_orders", sh, c_u64(nord),
              ord_cols["o_custkey"].ptr, ord_cols["o_orderdate"].ptr,
              ctypes.c_int32(Q3_DATE), self.cust_keys.ptr,
-             self.cust_head.ptr, c_u64(self.cust_cap), cnt.ptr)
+             self.cust_head.ptr, c_u64(self.cust_cap), cnt.ptr,
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         if stream:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
@@ -240,7 +245,8 @@ class Q3Fused:
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q3_DATE),
              self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
              self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
-             self.bloom.ptr, c_u64(self.bloom_bits - 1))
+             self.bloom.ptr, c_u64(self.bloom_bits - 1),
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         self._ord_cols = ord_cols
         self._cust_cols = cust_cols
 
@@ -271,18 +277,21 @@ class Q3Fused:
         call("qk_dmemset", self.ord_head.ptr, 0xFF, c_u64(self.ord_cap * 4))
         call("qk_dmemset", self.ord_sums.ptr, 0, c_u64(self.ord_cap * 8))
         call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
+        call("qk_dmemset", self.cbloom.ptr, 0, c_u64(self.cbloom_bits // 8))
         call("qk_build_u8eq", sh, c_u64(self._cust_cols["c_custkey"].n),
              self._cust_cols["c_custkey"].ptr,
              self._cust_cols["c_mktsegment"].ptr,
              ctypes.c_uint8(MKT_BUILDING), self.cust_keys.ptr,
-             self.cust_head.ptr, c_u64(self.cust_cap))
+             self.cust_head.ptr, c_u64(self.cust_cap),
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         call("qk_q3_build_orders", sh, c_u64(self._ord_cols["o_orderkey"].n),
              self._ord_cols["o_orderkey"].ptr,
              self._ord_cols["o_custkey"].ptr,
              self._ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q3_DATE),
              self.cust_keys.ptr, self.cust_head.ptr, c_u64(self.cust_cap),
              self.ord_keys.ptr, self.ord_head.ptr, c_u64(self.ord_cap),
-             self.bloom.ptr, c_u64(self.bloom_bits - 1))
+             self.bloom.ptr, c_u64(self.bloom_bits - 1),
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         """The fused filter+probe+group-by-aggregate pass (one kernel).
@@ -400,7 +409,7 @@ class Q3Fused:
 
     def free(self):
         for c in (self.cust_keys, self.cust_head, self.ord_keys,
-                  self.ord_head, self.ord_sums, self.bloom):
+                  self.ord_head, self.ord_sums, self.bloom, self.cbloom):
             c.free()
         if getattr(self, "_ext", None):
             for c in self._ext:
@@ -457,12 +466,16 @@ class Q5Fused:
         self.cust_cap = ops._pow2_at_least(max(16, 2 * max(1, ncust // 2)))
         self.cust_keys = DevColumn(np.int64, self.cust_cap)
         self.cust_val = DevColumn(np.int32, self.cust_cap)
+        self.cbloom_bits = ops._pow2_at_least(max(1 << 16, 4 * ncust))
+        self.cbloom = DevColumn(np.uint32, self.cbloom_bits // 32)
+        call("qk_dmemset", self.cbloom.ptr, 0, c_u64(self.cbloom_bits // 8))
         call("qk_fill_i64", sh, self.cust_keys.ptr,
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cust_cap))
         call("qk_build_keyval_i32", sh, c_u64(ncust),
              cust_cols["c_custkey"].ptr, cust_cols["c_nationkey"].ptr,
              ctypes.c_uint32(mask), self.cust_keys.ptr, self.cust_val.ptr,
-             c_u64(self.cust_cap))
+             c_u64(self.cust_cap), self.cbloom.ptr,
+             c_u64(self.cbloom_bits - 1))
         # supplier table: all rows
         self.supp_cap = ops._pow2_at_least(max(16, 2 * nsupp))
         self.supp_keys = DevColumn(np.int64, self.supp_cap)
@@ -472,7 +485,7 @@ class Q5Fused:
         call("qk_build_keyval_i32", sh, c_u64(nsupp),
              supp_cols["s_suppkey"].ptr, supp_cols["s_nationkey"].ptr,
              ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
-             self.supp_val.ptr, c_u64(self.supp_cap))
+             self.supp_val.ptr, c_u64(self.supp_cap), None, c_u64(0))
         # orders: count survivors -> tight table -> build
         cnt = ops._count_buf()
         call("qk_q5_build_orders", sh, c_u64(nord),
@@ -480,7 +493,7 @@ class Q5Fused:
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), None, None, c_u64(16), cnt.ptr, None,
-             c_u64(0))
+             c_u64(0), self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         if stream:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
@@ -499,7 +512,8 @@ class Q5Fused:
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
              c_u64(self.ord_cap), None, self.bloom.ptr,
-             c_u64(self.bloom_bits - 1))
+             c_u64(self.bloom_bits - 1), self.cbloom.ptr,
+             c_u64(self.cbloom_bits - 1))
         self.out25 = DevBuffer(32 * 8)
         call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
         self._ord_cols = ord_cols
@@ -520,18 +534,20 @@ class Q5Fused:
             call("qk_fill_i64", sh, keys.ptr, c_i64(int(shim.JOIN_EMPTY)),
                  c_u64(cap))
         call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
+        call("qk_dmemset", self.cbloom.ptr, 0, c_u64(self.cbloom_bits // 8))
         call("qk_build_keyval_i32", sh,
              c_u64(self._cust_cols["c_custkey"].n),
              self._cust_cols["c_custkey"].ptr,
              self._cust_cols["c_nationkey"].ptr,
              ctypes.c_uint32(self.asia_mask), self.cust_keys.ptr,
-             self.cust_val.ptr, c_u64(self.cust_cap))
+             self.cust_val.ptr, c_u64(self.cust_cap),
+             self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         call("qk_build_keyval_i32", sh,
              c_u64(self._supp_cols["s_suppkey"].n),
              self._supp_cols["s_suppkey"].ptr,
              self._supp_cols["s_nationkey"].ptr,
              ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
-             self.supp_val.ptr, c_u64(self.supp_cap))
+             self.supp_val.ptr, c_u64(self.supp_cap), None, c_u64(0))
         call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
         call("qk_q5_build_orders", sh, c_u64(self._ord_cols["o_orderkey"].n),
              self._ord_cols["o_orderkey"].ptr,
@@ -540,7 +556,8 @@ class Q5Fused:
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
              c_u64(self.ord_cap), None, self.bloom.ptr,
-             c_u64(self.bloom_bits - 1))
+             c_u64(self.bloom_bits - 1), self.cbloom.ptr,
+             c_u64(self.cbloom_bits - 1))
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         sh = self.stream.handle if self.stream else None
@@ -578,7 +595,8 @@ class Q5Fused:
 
     def free(self):
         for c in (self.cust_keys, self.cust_val, self.supp_keys,
-                  self.supp_val, self.ord_keys, self.ord_val, self.bloom):
+                  self.supp_val, self.ord_keys, self.ord_val, self.bloom,
+                  self.cbloom):
             c.free()
         self.out25.free()
 

@@ -70,17 +70,19 @@ _SIGS = {
     "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp],
     "qk_gen_supplier": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
     "qk_build_keyval_i32": [c_vp, c_u64, c_vp, c_vp, c_u32, c_vp, c_vp,
-                            c_u64],
+                            c_u64, c_vp, c_u64],
     "qk_q5_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_i32,
                            c_vp, c_vp, c_u64, c_vp, c_vp, c_u64, c_vp, c_vp,
-                           c_u64],
+                           c_u64, c_vp, c_u64],
     "qk_q5_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                         c_u64, c_vp, c_vp, c_u64, c_vp, c_vp],
-    "qk_build_u8eq": [c_vp, c_u64, c_vp, c_vp, c_u8, c_vp, c_vp, c_u64],
+    "qk_build_u8eq": [c_vp, c_u64, c_vp, c_vp, c_u8, c_vp, c_vp, c_u64,
+                      c_vp, c_u64],
     "qk_q3_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_vp, c_vp,
-                           c_u64, c_vp, c_vp, c_u64, c_vp, c_u64],
+                           c_u64, c_vp, c_vp, c_u64, c_vp, c_u64, c_vp,
+                           c_u64],
     "qk_q3_count_orders": [c_vp, c_u64, c_vp, c_vp, c_i32, c_vp, c_vp,
-                           c_u64, c_vp],
+                           c_u64, c_vp, c_vp, c_u64],
     "qk_q3_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
                         c_vp, c_u64, c_vp, c_vp],
     "qk_q3_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
