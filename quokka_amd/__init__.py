@@ -15,5 +15,7 @@ __version__ = "0.1.0"
 # execute, per the registration contract quokka_runtime.py:325); the shim
 # import inside them fails loudly at first execute if the library is absent.
 from .executors import (Executor, GPUAggExecutor,  # noqa: F401
-                        GPUBuildProbeJoinExecutor, GPUTopKExecutor,
+                        GPUBroadcastJoinExecutor,
+                        GPUBuildProbeJoinExecutor, GPUCountExecutor,
+                        GPUDistinctExecutor, GPUTopKExecutor,
                         gpu_partition_fn)

@@ -372,6 +372,126 @@ class GPUAggExecutor(Executor):
         return _to_table(out)
 
 
+class GPUCountExecutor(Executor):
+    """= CountExecutor (sql_executors.py:69-86): running row count,
+    done() returns a one-row table."""
+
+    def __init__(self):
+        self.state = 0
+
+    def execute(self, batches, stream_id, executor_id):
+        self.state += sum(len(b) for b in batches
+                          if b is not None)
+
+    def done(self, executor_id):
+        import pyarrow as pa
+        return pa.table({"count": np.array([self.state], dtype=np.int64)})
+
+
+class GPUBroadcastJoinExecutor(GPUBuildProbeJoinExecutor):
+    """= BroadcastJoinExecutor (sql_executors.py:275-319): the small build
+    side is given at CONSTRUCTION (broadcast to every channel) instead of
+    arriving as stream 1; every execute() batch is a probe. Same device
+    hash-table machinery as GPUBuildProbeJoinExecutor; picklable because
+    the table is built lazily from the stored host columns."""
+
+    def __init__(self, small_table, on=None, small_on=None, big_on=None,
+                 suffix="_small", how="inner"):
+        if on is not None:
+            assert small_on is None and big_on is None
+            small_on = big_on = on
+        super().__init__(left_on=big_on, right_on=small_on, how=how)
+        self.suffix = suffix
+        import pyarrow as pa
+        if not isinstance(small_table, pa.Table):
+            small_table = pa.table(small_table)
+        self._small_host = {c: small_table.column(c).to_numpy(
+            zero_copy_only=False) for c in small_table.column_names}
+        assert self.right_on in self._small_host
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if not batches:
+            return
+        if self._table is None:
+            self._build_names = list(self._small_host.keys())
+            self._build_cols = {c: [np.asarray(v)]
+                                for c, v in self._small_host.items()}
+            self._finish_build()
+        self.phase = "probe"
+        return self._probe(pa.concat_tables(batches))
+
+
+class GPUDistinctExecutor(Executor):
+    """= DistinctExecutor (sql_executors.py:517-554): per batch, emit the
+    rows whose key was never seen before (batch.unique() then anti-join
+    against the accumulated state, :529-543). Device hash table keyed i64;
+    done() returns None like the reference (state already emitted)."""
+
+    def __init__(self, keys):
+        self.keys = keys if isinstance(keys, str) else keys[0]
+        assert isinstance(self.keys, str), "round 1: single i64 key"
+        self._table = None
+        self._n_rows = 0
+
+    def __getstate__(self):
+        assert self._table is None, "pickle before first execute"
+        return dict(self.__dict__)
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        ops, shim, staging = _lazy_gpu()
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if not batches:
+            return
+        batch = pa.concat_tables(batches)
+        keys = staging.column_to_numpy(batch.column(self.keys))
+        if keys.dtype != np.int64:
+            raise TypeError("GPUDistinctExecutor requires i64 keys")
+        # within-batch unique (first occurrence), as batch.unique() does
+        _, first_idx = np.unique(keys, return_index=True)
+        first_idx.sort()
+        batch = batch.take(first_idx)
+        keys = keys[first_idx]
+        if self._table is None:
+            self._table = ops.JoinTable(max(1 << 16, 4 * len(keys)))
+            self._seen = []
+        if self._n_rows + len(keys) > min(self._table.chain_cap,
+                                          self._table.cap // 2):
+            self._grow(len(keys))
+        kcol = shim.DevColumn.from_numpy(keys)
+        pidx, _, nm = self._table.probe(kcol, mode=2)   # anti: unseen rows
+        sel = np.sort(pidx.to_numpy(nm))
+        contribution = batch.take(sel)
+        new_keys = keys[sel]
+        if len(new_keys):
+            ncol = shim.DevColumn.from_numpy(new_keys)
+            self._table.build(ncol)
+            ncol.free()
+            self._seen.append(new_keys)
+            self._n_rows += len(new_keys)
+        kcol.free()
+        pidx.free()
+        return contribution if len(contribution) else None
+
+    def _grow(self, incoming):
+        """Rebuild a larger table from the accumulated distinct keys
+        (the reference's vstack state just grows, :542)."""
+        ops, shim, staging = _lazy_gpu()
+        self._table.free()
+        allk = np.concatenate(self._seen) if self._seen else             np.empty(0, np.int64)
+        self._table = ops.JoinTable(max(1 << 16,
+                                        4 * (self._n_rows + incoming)))
+        if len(allk):
+            kcol = shim.DevColumn.from_numpy(allk)
+            self._table.build(kcol)
+            kcol.free()
+
+    def done(self, executor_id):
+        return None
+
+
 class GPUTopKExecutor(Executor):
     """Mirror of ConcatThenSQLExecutor as lowered by DataStream.top_k
     (sql_executors.py:45-67; datastream.py:1746-1767): concatenate incoming
