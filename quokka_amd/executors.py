@@ -319,7 +319,13 @@ class GPUAggExecutor(Executor):
         for k, codes in zip(self.groupby_keys, outs):
             cb = self._key_state["codebooks"][k]
             inv = {v: kk for kk, v in cb.items()}
-            cols[k] = np.array([inv[c] for c in codes.astype(np.int64)])
+            vals = [inv[c] for c in codes.astype(np.int64)]
+            # string columns were StringDict-coded before the codebook:
+            # unwind that layer too so callers get the original strings
+            sd = self._key_state["dicts"].get(k)
+            if sd is not None and sd.values:
+                vals = [sd.values[v] for v in vals]
+            cols[k] = np.array(vals)
         return cols
 
     def execute(self, batches, stream_id, executor_id):
