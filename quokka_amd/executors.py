@@ -538,27 +538,48 @@ class GPUTopKExecutor(Executor):
         return self.state.take(order)
 
 
-def gpu_partition_fn(data, source_channel, num_target_channels, key=None):
-    """GPU hash partitioner mirroring partition_key_str
-    (quokka_runtime.py:217-231): int key -> key % N, bit-exact with :222.
+def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
+                     predicate=None, string_dicts=None, projection=None):
+    """GPU partition function mirroring the reference's full partition_fn
+    (core.py:152-195): optional PREDICATE (filter_sql grammar, JIT-fused
+    on device via quokka_amd.jit) -> hash partition (int key -> key % N,
+    bit-exact with quokka_runtime.py:222) -> optional column projection
+    (sorted order, core.py:191-193).
     `data`: pyarrow Table; returns dict target_channel -> pyarrow Table."""
     import pyarrow as pa
     ops, shim, staging = _lazy_gpu()
-    keys = staging.column_to_numpy(data.column(key))
+    host_cols = {c: staging.column_to_numpy(
+        data.column(c), (string_dicts or {}).get(c))
+        for c in data.column_names}
+
+    if predicate is not None:
+        from . import jit
+        dcols = {c: shim.DevColumn.from_numpy(v)
+                 for c, v in host_cols.items()}
+        schema = {c: v.dtype for c, v in dcols.items()}
+        f = jit.JitFilter(predicate, schema, string_dicts)
+        fidx, k = f.run(dcols)
+        sel0 = fidx.to_numpy(k)
+        host_cols = {c: v[sel0] for c, v in host_cols.items()}
+        f.free()
+        fidx.free()
+        for c in dcols.values():
+            c.free()
+
+    keys = host_cols[key]
     if keys.dtype != np.int64:
         raise TypeError("gpu_partition_fn: int64 keys only in round 1")
     kcol = shim.DevColumn.from_numpy(keys)
     offsets, idx = ops.partition_i64(kcol, num_target_channels)
     sel = idx.to_numpy(len(keys))
-    host_cols = {c: staging.column_to_numpy(data.column(c))
-                 for c in data.column_names}
+    names = sorted(projection) if projection else sorted(host_cols)
     out = {}
     for p in range(num_target_channels):
         lo, hi = int(offsets[p]), int(offsets[p + 1])
         if hi == lo:
             continue
         rows = sel[lo:hi]
-        out[p] = pa.table({c: v[rows] for c, v in host_cols.items()})
+        out[p] = pa.table({c: host_cols[c][rows] for c in names})
     kcol.free()
     idx.free()
     return out

@@ -809,3 +809,45 @@ def test_executor_distinct_and_broadcast_and_count(gpu):
         seen |= set(ck)
     assert seen == set(all_keys.tolist())
     assert de.done(0) is None
+
+
+def test_jit_filter_parity(gpu, data):
+    """hiprtc-JIT'd predicates == oracle row sets, bit-exact & ordered:
+    Q6's compound predicate, a string-dict equality, OR/NOT forms."""
+    from quokka_amd import jit, staging
+    li = data["lineitem"]
+    cols = staging.stage_columns(li)
+    schema = {k: v.dtype for k, v in cols.items()}
+
+    f = jit.JitFilter(
+        "l_shipdate >= date '1994-01-01' and l_shipdate < "
+        "date '1994-01-01' + interval '1' year and l_discount between "
+        "0.06 - 0.01 and 0.06 + 0.01 and l_quantity < 24", schema)
+    idx, k = f.run(cols)
+    lo, hi = 0.06 - 0.01, 0.06 + 0.01
+    want = np.nonzero((li["l_shipdate"] >= G.Q5_LO)
+                      & (li["l_shipdate"] < G.Q5_HI)
+                      & (li["l_discount"] >= lo) & (li["l_discount"] <= hi)
+                      & (li["l_quantity"] < 24))[0]
+    assert np.array_equal(idx.to_numpy(k), want)
+    f.free(); idx.free()
+
+    f2 = jit.JitFilter(
+        "not (l_returnflag = 'A' or l_returnflag = 'R') and "
+        "l_quantity >= 25", schema,
+        string_dicts={"l_returnflag": type("SD", (), {
+            "codes": {"A": 0, "N": 1, "R": 2}})()})
+    idx2, k2 = f2.run(cols)
+    want2 = np.nonzero(~((li["l_returnflag"] == 0)
+                         | (li["l_returnflag"] == 2))
+                       & (li["l_quantity"] >= 25))[0]
+    assert np.array_equal(idx2.to_numpy(k2), want2)
+    f2.free(); idx2.free()
+
+    # empty result
+    f3 = jit.JitFilter("l_quantity > 1000", schema)
+    idx3, k3 = f3.run(cols)
+    assert k3 == 0
+    f3.free(); idx3.free()
+    for c in cols.values():
+        c.free()
