@@ -237,6 +237,154 @@ extern "C" int qk_jit_filter_run(void *prog, void *stream, uint64_t n,
   return 0;
 }
 
+// ---- JIT fused scan + small-cardinality group-by partial aggregate ----
+// Generates the k_q1_agg structure (per-thread REGISTER
+// accumulators statically unrolled per group, wave shuffle-reduce, block
+// LDS reduce, one f64 atomicAdd per accumulator per block) for ARBITRARY
+// (predicate, group expression, aggregate expressions). This is how the
+// reference's folded map-side DuckDB batch aggregates
+// (datastream.py:795-801 + df.py:1354-1394) generalize on device.
+
+static std::string gen_agg_source(const char *pred, const char *group_expr,
+                                  int ngroups, int naggs,
+                                  const char *const *agg_exprs, int ncols,
+                                  const int *coltypes) {
+  std::string s;
+  s += "#define BLOCK 256\n#define WAVE 64\n";
+  s += "typedef unsigned long long u64; typedef unsigned u32;\n";
+  s += "extern \"C\" __global__ __launch_bounds__(BLOCK) void jit_agg(\n";
+  s += "    u64 n, double* __restrict__ out";
+  for (int c = 0; c < ncols; c++) {
+    s += ", const ";
+    s += type_name(coltypes[c]);
+    s += "* __restrict__ col";
+    s += std::to_string(c);
+  }
+  s += ") {\n";
+  for (int g = 0; g < ngroups; g++)
+    for (int a = 0; a < naggs; a++)
+      s += "  double acc_" + std::to_string(g) + "_" + std::to_string(a) +
+           " = 0.0;\n";
+  s += "  u64 stride = (u64)gridDim.x * blockDim.x;\n"
+       "  for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < n;\n"
+       "       i += stride) {\n";
+  for (int c = 0; c < ncols; c++) {
+    s += "    ";
+    s += type_name(coltypes[c]);
+    s += " v" + std::to_string(c) + " = __builtin_nontemporal_load(&col" +
+         std::to_string(c) + "[i]);\n";
+  }
+  if (pred && pred[0]) {
+    s += "    if (!(";
+    s += pred;
+    s += ")) continue;\n";
+  }
+  s += "    int gid = (int)(";
+  s += group_expr;
+  s += ");\n";
+  for (int g = 0; g < ngroups; g++) {
+    s += g == 0 ? "    if (gid == 0) {\n"
+                : "    else if (gid == " + std::to_string(g) + ") {\n";
+    for (int a = 0; a < naggs; a++) {
+      s += "      acc_" + std::to_string(g) + "_" + std::to_string(a) +
+           " += (double)(";
+      s += agg_exprs[a];
+      s += ");\n";
+    }
+    s += "    }\n";
+  }
+  s += "  }\n";
+  // wave reduce -> LDS -> block reduce -> atomicAdd
+  s += "  __shared__ double lds[BLOCK / WAVE][" +
+       std::to_string(ngroups * naggs) + "];\n"
+       "  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;\n";
+  for (int g = 0; g < ngroups; g++)
+    for (int a = 0; a < naggs; a++) {
+      std::string acc = "acc_" + std::to_string(g) + "_" + std::to_string(a);
+      s += "  { double t = " + acc + ";\n"
+           "    for (int off = WAVE / 2; off > 0; off >>= 1)\n"
+           "      t += __shfl_down(t, off);\n"
+           "    if (lane == 0) lds[wid][" +
+           std::to_string(g * naggs + a) + "] = t; }\n";
+    }
+  s += "  __syncthreads();\n"
+       "  if (threadIdx.x < " + std::to_string(ngroups * naggs) + ") {\n"
+       "    double t = 0;\n"
+       "    for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w][threadIdx.x];\n"
+       "    if (t != 0.0) atomicAdd(&out[threadIdx.x], t);\n  }\n}\n";
+  return s;
+}
+
+extern "C" int qk_jit_agg_build(const char *pred, const char *group_expr,
+                                int ngroups, int naggs,
+                                const char *const *agg_exprs, int ncols,
+                                const int *coltypes, void **prog_out) {
+  if (ncols < 1 || ncols > QK_JIT_MAX_COLS)
+    return j_fail("qk_jit_agg_build", "ncols out of range");
+  if (ngroups < 1 || naggs < 1 || ngroups * naggs > 64)
+    return j_fail("qk_jit_agg_build",
+                  "ngroups*naggs out of range (register accumulators; "
+                  "use the hash group-by for high cardinality)");
+  for (int c = 0; c < ncols; c++)
+    if (!type_name(coltypes[c]))
+      return j_fail("qk_jit_agg_build", "bad column type");
+  std::string src = gen_agg_source(pred, group_expr, ngroups, naggs,
+                                   agg_exprs, ncols, coltypes);
+  hiprtcProgram prog;
+  if (hiprtcCreateProgram(&prog, src.c_str(), "qk_jit_agg.cu", 0, nullptr,
+                          nullptr) != HIPRTC_SUCCESS)
+    return j_fail("qk_jit_agg_build", "hiprtcCreateProgram failed");
+  const char *opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17",
+                        "-munsafe-fp-atomics"};
+  hiprtcResult rc = hiprtcCompileProgram(prog, 4, opts);
+  if (rc != HIPRTC_SUCCESS) {
+    size_t lsz = 0;
+    hiprtcGetProgramLogSize(prog, &lsz);
+    std::string log(lsz, '\0');
+    if (lsz) hiprtcGetProgramLog(prog, &log[0]);
+    hiprtcDestroyProgram(&prog);
+    snprintf(j_err, sizeof(j_err), "qk_jit_agg_build: compile failed: %s",
+             log.c_str());
+    return 2;
+  }
+  size_t csz = 0;
+  hiprtcGetCodeSize(prog, &csz);
+  QkJitProg *p = new QkJitProg();
+  p->code.resize(csz);
+  hiprtcGetCode(prog, &p->code[0]);
+  hiprtcDestroyProgram(&prog);
+  p->ncols = ncols;
+  memcpy(p->coltypes, coltypes, ncols * sizeof(int));
+  *prog_out = p;
+  return 0;
+}
+
+extern "C" int qk_jit_agg_run(void *prog, void *stream, uint64_t n,
+                              const void *const *col_ptrs,
+                              double *out_dev /* accumulates */) {
+  QkJitProg *p = (QkJitProg *)prog;
+  if (!p->mod) {
+    if (hipModuleLoadData(&p->mod, p->code.data()) != hipSuccess)
+      return j_fail("qk_jit_agg_run", "hipModuleLoadData failed");
+  }
+  if (!p->f_count &&
+      hipModuleGetFunction(&p->f_count, p->mod, "jit_agg") != hipSuccess)
+    return j_fail("qk_jit_agg_run", "hipModuleGetFunction failed");
+  if (!n) return 0;
+  uint64_t a_n = n;
+  void *a_out = out_dev;
+  std::vector<const void *> cols(col_ptrs, col_ptrs + p->ncols);
+  std::vector<void *> args = {&a_n, &a_out};
+  for (int c = 0; c < p->ncols; c++) args.push_back((void *)&cols[c]);
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  if (hipModuleLaunchKernel(p->f_count, blocks, 1, 1, 256, 1, 1, 0,
+                            (hipStream_t)stream, args.data(),
+                            nullptr) != hipSuccess)
+    return j_fail("qk_jit_agg_run", "launch failed");
+  return 0;
+}
+
 extern "C" int qk_jit_filter_free(void *prog) {
   QkJitProg *p = (QkJitProg *)prog;
   if (p->scratch) hipFree(p->scratch);

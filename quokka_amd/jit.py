@@ -282,6 +282,122 @@ def translate(predicate, schema, string_dicts=None):
     return tr.translate(predicate), tr.cols
 
 
+def translate_arith(expr, schema, translator=None):
+    """Arithmetic expression (no comparisons) -> (c_expr, columns)."""
+    tr = translator or Translator(schema)
+    tr.toks = _tokenize(expr)
+    tr.i = 0
+    e, _ = tr.p_arith()
+    if tr.i != len(tr.toks):
+        raise ValueError("trailing tokens in %r" % expr)
+    return e, tr.cols
+
+
+_AGG_SQL = re.compile(r"^\s*(sum|count)\s*\(\s*(.*?)\s*\)\s*"
+                      r"(?:as\s+([A-Za-z_][A-Za-z0-9_]*))?\s*$",
+                      re.I | re.S)
+
+
+class JitAggregate:
+    """JIT-fused scan + small-cardinality group-by partial aggregate:
+    the generalization of the hand-written Q1 kernel (csrc qk_jit_agg_*).
+
+    group_keys: list of (column_name, cardinality) over u8 code columns;
+    aggs: list of 'SUM(expr)' / 'COUNT(*)' strings (the map-side partial
+    forms the two-phase rewrite emits, sql_utils.py:299-413);
+    predicate: optional filter_sql string fused into the same pass.
+    run() ACCUMULATES into a DevBuffer of ngroups*naggs f64 (executor
+    state semantics, like qk_q1_agg)."""
+
+    def __init__(self, schema, group_keys, aggs, predicate=None,
+                 string_dicts=None):
+        tr = Translator(schema, string_dicts)
+        pred_expr = ""
+        if predicate:
+            pred_expr = tr.translate(predicate)
+        self.group_keys = list(group_keys)
+        self.ngroups = 1
+        gparts = []
+        for name, card in self.group_keys:
+            ref, dt = tr.col_ref(name)
+            if dt != np.dtype(np.uint8):
+                raise TypeError("group keys must be u8 code columns")
+            gparts.append((ref, card))
+            self.ngroups *= int(card)
+        gexpr = "0"
+        for ref, card in gparts:
+            gexpr = "(%s) * %d + (int)%s" % (gexpr, card, ref)
+        self.agg_names = []
+        agg_exprs = []
+        for a in aggs:
+            m = _AGG_SQL.match(a)
+            if not m:
+                raise ValueError("unsupported aggregate %r "
+                                 "(SUM(expr) / COUNT(*))" % a)
+            fn, inner, alias = m.group(1).lower(), m.group(2), m.group(3)
+            if fn == "count":
+                agg_exprs.append("1.0")
+            else:
+                e, _ = translate_arith(inner, schema, tr)
+                agg_exprs.append(e)
+            self.agg_names.append(alias or a.strip())
+        self.naggs = len(agg_exprs)
+        self.cols = tr.cols
+        self.dtypes = [np.dtype(schema[c]) for c in self.cols]
+
+        lib = shim._lib
+        lib.qk_jit_agg_build.argtypes = [
+            ctypes.c_char_p, ctypes.c_char_p, ctypes.c_int, ctypes.c_int,
+            ctypes.POINTER(ctypes.c_char_p), ctypes.c_int,
+            ctypes.POINTER(ctypes.c_int), c_vp]
+        lib.qk_jit_last_error.restype = ctypes.c_char_p
+        exprs_c = (ctypes.c_char_p * self.naggs)(
+            *[e.encode() for e in agg_exprs])
+        types_c = (ctypes.c_int * len(self.cols))(
+            *[_TYPE_CODE[d] for d in self.dtypes])
+        prog = c_vp(0)
+        rc = lib.qk_jit_agg_build(pred_expr.encode(), gexpr.encode(),
+                                  self.ngroups, self.naggs, exprs_c,
+                                  len(self.cols), types_c,
+                                  ctypes.byref(prog))
+        if rc != 0:
+            raise shim.QkError("jit agg build failed: %s"
+                               % lib.qk_jit_last_error().decode())
+        self.prog = prog
+
+    def make_acc(self):
+        from . import ops
+        from .shim import DevBuffer
+        b = DevBuffer(self.ngroups * self.naggs * 8)
+        shim.call("qk_dmemset", b.ptr, 0,
+                  c_u64(self.ngroups * self.naggs * 8))
+        return b
+
+    def run(self, cols, acc, stream=None):
+        lib = shim._lib
+        lib.qk_jit_agg_run.argtypes = [c_vp, c_vp, c_u64,
+                                       ctypes.POINTER(c_vp), c_vp]
+        n = cols[self.cols[0]].n
+        ptrs = (c_vp * len(self.cols))(*[cols[c].ptr for c in self.cols])
+        sh = stream.handle if stream else None
+        rc = lib.qk_jit_agg_run(self.prog, sh, c_u64(n), ptrs, acc.ptr)
+        if rc != 0:
+            raise shim.QkError("jit agg run failed: %s"
+                               % lib.qk_jit_last_error().decode())
+
+    def read(self, acc):
+        """d2h -> (ngroups, naggs) f64."""
+        host = np.zeros(self.ngroups * self.naggs, dtype=np.float64)
+        shim.call("qk_d2h", host.ctypes.data_as(c_vp), acc.ptr,
+                  c_u64(host.nbytes))
+        return host.reshape(self.ngroups, self.naggs)
+
+    def free(self):
+        if self.prog:
+            shim._lib.qk_jit_filter_free(self.prog)
+            self.prog = None
+
+
 class JitFilter:
     """Compiled fused filter for one predicate over a fixed schema."""
 

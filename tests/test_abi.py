@@ -163,3 +163,18 @@ def test_jit_translate_and_compile_cpu():
     import pytest as _pytest
     with _pytest.raises(ValueError):
         jit.translate("no_such_col < 5", schema)
+
+
+def test_jit_aggregate_compiles_cpu():
+    import numpy as np
+    from quokka_amd import jit
+    schema = {"d": np.dtype(np.int32), "x": np.dtype(np.float64),
+              "g": np.dtype(np.uint8)}
+    agg = jit.JitAggregate(schema, [("g", 4)],
+                           ["sum(x) as sx", "count(*) as n"],
+                           predicate="d < 100")
+    assert agg.ngroups == 4 and agg.naggs == 2
+    agg.free()
+    import pytest as _p
+    with _p.raises(Exception):
+        jit.JitAggregate(schema, [("g", 200)], ["sum(x)"])  # >64 accums
