@@ -16,6 +16,7 @@
 #include <stdlib.h>
 #include <string.h>
 #include <string>
+#include <ctype.h>
 #include <vector>
 
 #define QK_JIT_MAX_COLS 8
@@ -261,38 +262,103 @@ static std::string gen_agg_source(const char *pred, const char *group_expr,
     s += std::to_string(c);
   }
   s += ") {\n";
+  s += "  typedef int v2i __attribute__((ext_vector_type(2)));\n"
+       "  typedef double v2d __attribute__((ext_vector_type(2)));\n"
+       "  typedef long long v2l __attribute__((ext_vector_type(2)));\n";
   for (int g = 0; g < ngroups; g++)
     for (int a = 0; a < naggs; a++)
       s += "  double acc_" + std::to_string(g) + "_" + std::to_string(a) +
            " = 0.0;\n";
-  s += "  u64 stride = (u64)gridDim.x * blockDim.x;\n"
-       "  for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < n;\n"
-       "       i += stride) {\n";
+  // one row's predicate + accumulate, parameterized by value suffix
+  auto body = [&](const char *suf) {
+    std::string b;
+    std::string pfx = "    ";
+    if (pred && pred[0]) {
+      std::string p = pred;
+      // rename v<k> -> v<k><suf>
+      std::string q;
+      for (size_t k = 0; k < p.size(); k++) {
+        if (p[k] == 'v' && k + 1 < p.size() && isdigit(p[k + 1])) {
+          size_t e = k + 1;
+          while (e < p.size() && isdigit(p[e])) e++;
+          q += p.substr(k, e - k) + suf;
+          k = e - 1;
+        } else q += p[k];
+      }
+      b += pfx + "if ((" + q + ")) {\n";
+    } else
+      b += pfx + "{\n";
+    auto ren = [&](const char *src) {
+      std::string p = src, q;
+      for (size_t k = 0; k < p.size(); k++) {
+        if (p[k] == 'v' && k + 1 < p.size() && isdigit(p[k + 1])) {
+          size_t e = k + 1;
+          while (e < p.size() && isdigit(p[e])) e++;
+          q += p.substr(k, e - k) + suf;
+          k = e - 1;
+        } else q += p[k];
+      }
+      return q;
+    };
+    b += pfx + "  int gid = (int)(" + ren(group_expr) + ");\n";
+    for (int g = 0; g < ngroups; g++) {
+      b += pfx + (g == 0 ? std::string("  if (gid == 0) {\n")
+                         : "  else if (gid == " + std::to_string(g) +
+                               ") {\n");
+      for (int a = 0; a < naggs; a++)
+        b += pfx + "    acc_" + std::to_string(g) + "_" +
+             std::to_string(a) + " += (double)(" + ren(agg_exprs[a]) +
+             ");\n";
+      b += pfx + "  }\n";
+    }
+    b += pfx + "}\n";
+    return b;
+  };
+  // 2 rows per thread: vector nt loads (16 B/lane on f64/i64 columns)
+  s += "  u64 npairs = n / 2;\n"
+       "  u64 stride = (u64)gridDim.x * blockDim.x;\n"
+       "  for (u64 p = (u64)blockIdx.x * blockDim.x + threadIdx.x;\n"
+       "       p < npairs; p += stride) {\n"
+       "    u64 i = 2 * p;\n";
+  for (int c = 0; c < ncols; c++) {
+    std::string cn = std::to_string(c);
+    switch (coltypes[c]) {
+    case 0:
+      s += "    v2i w" + cn + " = __builtin_nontemporal_load("
+           "(const v2i*)(col" + cn + " + i));\n"
+           "    int v" + cn + "_0 = w" + cn + ".x, v" + cn + "_1 = w" +
+           cn + ".y;\n";
+      break;
+    case 1:
+      s += "    v2d w" + cn + " = __builtin_nontemporal_load("
+           "(const v2d*)(col" + cn + " + i));\n"
+           "    double v" + cn + "_0 = w" + cn + ".x, v" + cn + "_1 = w" +
+           cn + ".y;\n";
+      break;
+    case 3:
+      s += "    v2l w" + cn + " = __builtin_nontemporal_load("
+           "(const v2l*)(col" + cn + " + i));\n"
+           "    long long v" + cn + "_0 = w" + cn + ".x, v" + cn +
+           "_1 = w" + cn + ".y;\n";
+      break;
+    default: // u8: two scalar loads (sub-word vectors not worth it)
+      s += "    unsigned char v" + cn + "_0 = col" + cn + "[i];\n"
+           "    unsigned char v" + cn + "_1 = col" + cn + "[i + 1];\n";
+    }
+  }
+  s += body("_0");
+  s += body("_1");
+  s += "  }\n";
+  // odd tail handled by one thread with scalar loads
+  s += "  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {\n"
+       "    u64 i = n - 1;\n";
   for (int c = 0; c < ncols; c++) {
     s += "    ";
     s += type_name(coltypes[c]);
-    s += " v" + std::to_string(c) + " = __builtin_nontemporal_load(&col" +
-         std::to_string(c) + "[i]);\n";
+    s += " v" + std::to_string(c) + "_t = col" + std::to_string(c) +
+         "[i];\n";
   }
-  if (pred && pred[0]) {
-    s += "    if (!(";
-    s += pred;
-    s += ")) continue;\n";
-  }
-  s += "    int gid = (int)(";
-  s += group_expr;
-  s += ");\n";
-  for (int g = 0; g < ngroups; g++) {
-    s += g == 0 ? "    if (gid == 0) {\n"
-                : "    else if (gid == " + std::to_string(g) + ") {\n";
-    for (int a = 0; a < naggs; a++) {
-      s += "      acc_" + std::to_string(g) + "_" + std::to_string(a) +
-           " += (double)(";
-      s += agg_exprs[a];
-      s += ");\n";
-    }
-    s += "    }\n";
-  }
+  s += body("_t");
   s += "  }\n";
   // wave reduce -> LDS -> block reduce -> atomicAdd
   s += "  __shared__ double lds[BLOCK / WAVE][" +
