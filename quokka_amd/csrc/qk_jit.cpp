@@ -451,6 +451,99 @@ extern "C" int qk_jit_agg_run(void *prog, void *stream, uint64_t n,
   return 0;
 }
 
+// ---- JIT elementwise transform (transform_sql map side) ---------------
+// out[i] = (double)(EXPR over typed columns) — generalizes qk_mul_1md;
+// the reference's with_columns_sql / transform_sql per-batch expressions
+// (datastream.py:652-815).
+static std::string gen_map_source(const char *expr, int ncols,
+                                  const int *coltypes) {
+  std::string s;
+  s += "#define BLOCK 256\n";
+  s += "typedef unsigned long long u64;\n";
+  s += "extern \"C\" __global__ __launch_bounds__(BLOCK) void jit_map(\n";
+  s += "    u64 n, double* __restrict__ out";
+  for (int c = 0; c < ncols; c++) {
+    s += ", const ";
+    s += type_name(coltypes[c]);
+    s += "* __restrict__ col";
+    s += std::to_string(c);
+  }
+  s += ") {\n"
+       "  u64 stride = (u64)gridDim.x * blockDim.x;\n"
+       "  for (u64 i = (u64)blockIdx.x * blockDim.x + threadIdx.x; i < n;\n"
+       "       i += stride) {\n";
+  for (int c = 0; c < ncols; c++) {
+    s += "    ";
+    s += type_name(coltypes[c]);
+    s += " v" + std::to_string(c) + " = col" + std::to_string(c) +
+         "[i];\n";
+  }
+  s += "    out[i] = (double)(";
+  s += expr;
+  s += ");\n  }\n}\n";
+  return s;
+}
+
+extern "C" int qk_jit_map_build(const char *expr, int ncols,
+                                const int *coltypes, void **prog_out) {
+  if (ncols < 1 || ncols > QK_JIT_MAX_COLS)
+    return j_fail("qk_jit_map_build", "ncols out of range");
+  for (int c = 0; c < ncols; c++)
+    if (!type_name(coltypes[c]))
+      return j_fail("qk_jit_map_build", "bad column type");
+  std::string src = gen_map_source(expr, ncols, coltypes);
+  hiprtcProgram prog;
+  if (hiprtcCreateProgram(&prog, src.c_str(), "qk_jit_map.cu", 0, nullptr,
+                          nullptr) != HIPRTC_SUCCESS)
+    return j_fail("qk_jit_map_build", "hiprtcCreateProgram failed");
+  const char *opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17"};
+  if (hiprtcCompileProgram(prog, 3, opts) != HIPRTC_SUCCESS) {
+    size_t lsz = 0;
+    hiprtcGetProgramLogSize(prog, &lsz);
+    std::string log(lsz, '\0');
+    if (lsz) hiprtcGetProgramLog(prog, &log[0]);
+    hiprtcDestroyProgram(&prog);
+    snprintf(j_err, sizeof(j_err), "qk_jit_map_build: compile failed: %s",
+             log.c_str());
+    return 2;
+  }
+  size_t csz = 0;
+  hiprtcGetCodeSize(prog, &csz);
+  QkJitProg *p = new QkJitProg();
+  p->code.resize(csz);
+  hiprtcGetCode(prog, &p->code[0]);
+  hiprtcDestroyProgram(&prog);
+  p->ncols = ncols;
+  memcpy(p->coltypes, coltypes, ncols * sizeof(int));
+  *prog_out = p;
+  return 0;
+}
+
+extern "C" int qk_jit_map_run(void *prog, void *stream, uint64_t n,
+                              const void *const *col_ptrs, double *out_dev) {
+  QkJitProg *p = (QkJitProg *)prog;
+  if (!p->mod) {
+    if (hipModuleLoadData(&p->mod, p->code.data()) != hipSuccess)
+      return j_fail("qk_jit_map_run", "hipModuleLoadData failed");
+  }
+  if (!p->f_count &&
+      hipModuleGetFunction(&p->f_count, p->mod, "jit_map") != hipSuccess)
+    return j_fail("qk_jit_map_run", "hipModuleGetFunction failed");
+  if (!n) return 0;
+  uint64_t a_n = n;
+  void *a_out = out_dev;
+  std::vector<const void *> cols(col_ptrs, col_ptrs + p->ncols);
+  std::vector<void *> args = {&a_n, &a_out};
+  for (int c = 0; c < p->ncols; c++) args.push_back((void *)&cols[c]);
+  uint32_t blocks = (uint32_t)((n + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  if (hipModuleLaunchKernel(p->f_count, blocks, 1, 1, 256, 1, 1, 0,
+                            (hipStream_t)stream, args.data(),
+                            nullptr) != hipSuccess)
+    return j_fail("qk_jit_map_run", "launch failed");
+  return 0;
+}
+
 extern "C" int qk_jit_filter_free(void *prog) {
   QkJitProg *p = (QkJitProg *)prog;
   if (p->scratch) hipFree(p->scratch);

@@ -398,6 +398,50 @@ class JitAggregate:
             self.prog = None
 
 
+class JitMap:
+    """JIT elementwise transform: out f64 column = expr(columns) per row
+    (the reference's transform_sql / with_columns_sql batch expressions,
+    datastream.py:652-815; generalizes the static qk_mul_1md)."""
+
+    def __init__(self, expr, schema):
+        self.expr, self.cols = translate_arith(expr, schema)
+        self.dtypes = [np.dtype(schema[c]) for c in self.cols]
+        lib = shim._lib
+        lib.qk_jit_map_build.argtypes = [ctypes.c_char_p, ctypes.c_int,
+                                         ctypes.POINTER(ctypes.c_int), c_vp]
+        lib.qk_jit_last_error.restype = ctypes.c_char_p
+        types_c = (ctypes.c_int * len(self.cols))(
+            *[_TYPE_CODE[d] for d in self.dtypes])
+        prog = c_vp(0)
+        rc = lib.qk_jit_map_build(self.expr.encode(), len(self.cols),
+                                  types_c, ctypes.byref(prog))
+        if rc != 0:
+            raise shim.QkError("jit map build failed: %s"
+                               % lib.qk_jit_last_error().decode())
+        self.prog = prog
+
+    def run(self, cols, stream=None, out=None):
+        lib = shim._lib
+        lib.qk_jit_map_run.argtypes = [c_vp, c_vp, c_u64,
+                                       ctypes.POINTER(c_vp), c_vp]
+        n = cols[self.cols[0]].n
+        if out is None:
+            out = DevColumn(np.float64, max(1, n))
+            out.n = n
+        ptrs = (c_vp * len(self.cols))(*[cols[c].ptr for c in self.cols])
+        sh = stream.handle if stream else None
+        rc = lib.qk_jit_map_run(self.prog, sh, c_u64(n), ptrs, out.ptr)
+        if rc != 0:
+            raise shim.QkError("jit map run failed: %s"
+                               % lib.qk_jit_last_error().decode())
+        return out
+
+    def free(self):
+        if self.prog:
+            shim._lib.qk_jit_filter_free(self.prog)
+            self.prog = None
+
+
 class JitFilter:
     """Compiled fused filter for one predicate over a fixed schema."""
 

@@ -925,3 +925,33 @@ def test_jit_aggregate_adhoc(gpu, data):
     acc.free(); agg.free()
     for c in cols.values():
         c.free()
+
+
+def test_jit_map_parity(gpu, data):
+    """JIT transform == numpy on the revenue expression and a compound
+    arithmetic expression; equals the static qk_mul_1md kernel exactly."""
+    from quokka_amd import jit, staging, queries as DQ
+    li = data["lineitem"]
+    cols = staging.stage_columns(li, names=["l_extendedprice",
+                                            "l_discount", "l_tax",
+                                            "l_quantity"])
+    schema = {k: v.dtype for k, v in cols.items()}
+    m = jit.JitMap("l_extendedprice * (1 - l_discount)", schema)
+    out = m.run(cols)
+    want = li["l_extendedprice"] * (1.0 - li["l_discount"])
+    got = out.to_numpy(out.n)
+    np.testing.assert_array_equal(got, want)      # identical f64 ops
+    # same as the static kernel
+    rev2 = DQ._mul_1md(cols["l_extendedprice"], cols["l_discount"], None)
+    np.testing.assert_array_equal(got, rev2.to_numpy(rev2.n))
+    m.free(); out.free(); rev2.free()
+    m2 = jit.JitMap(
+        "l_extendedprice * (1 - l_discount) * (1 + l_tax) / l_quantity",
+        schema)
+    out2 = m2.run(cols)
+    want2 = (li["l_extendedprice"] * (1.0 - li["l_discount"])
+             * (1.0 + li["l_tax"]) / li["l_quantity"])
+    np.testing.assert_array_equal(out2.to_numpy(out2.n), want2)
+    m2.free(); out2.free()
+    for c in cols.values():
+        c.free()
