@@ -293,6 +293,35 @@ int qk_partition_scatter(void *stream, uint64_t n, const int64_t *keys,
                          uint32_t nparts, uint64_t *cursors_dev,
                          uint32_t *out_idx);
 
+/* ---- hiprtc JIT (arbitrary predicates / transforms / partial aggs) --- *
+ * Runtime-compiled gfx950 kernels for the reference's ARBITRARY
+ * filter_sql predicates (core.py:157-170), transform_sql expressions
+ * (datastream.py:652-815) and folded map-side batch aggregates
+ * (df.py:1354-1394) — quokka_amd/jit.py translates the SQL grammar to
+ * the C expressions these take. Compilation needs no GPU; module load
+ * is lazy at first run. coltypes: 0=i32, 1=f64, 2=u8, 3=i64.
+ * qk_jit_last_error() returns the hiprtc log on compile failure. */
+const char *qk_jit_last_error(void);
+int qk_jit_filter_build(const char *c_expr, int ncols, const int *coltypes,
+                        void **prog_out);
+int qk_jit_filter_run(void *prog, void *stream, uint64_t n,
+                      const void *const *col_ptrs, uint32_t *out_idx,
+                      uint64_t *count_dev);
+int qk_jit_map_build(const char *c_expr, int ncols, const int *coltypes,
+                     void **prog_out);
+int qk_jit_map_run(void *prog, void *stream, uint64_t n,
+                   const void *const *col_ptrs, double *out_dev);
+/* group_expr: C int expression over v<i>; ngroups*naggs <= 64 (register
+ * accumulators; use qk_groupby_* for high cardinality). out accumulates
+ * ngroups*naggs f64 (row-major [group][agg]). */
+int qk_jit_agg_build(const char *pred_or_empty, const char *group_expr,
+                     int ngroups, int naggs, const char *const *agg_exprs,
+                     int ncols, const int *coltypes, void **prog_out);
+int qk_jit_agg_run(void *prog, void *stream, uint64_t n,
+                   const void *const *col_ptrs, double *out_dev);
+int qk_jit_filter_free(void *prog);   /* frees any qk_jit_* program */
+uint64_t qk_jit_code_size(void *prog);
+
 /* ---- RCCL exchange (multi-GPU hash repartition) ----------------------- *
  * Replaces the reference's shuffle data plane (core.py:276-376 push ->
  * Arrow Flight do_put/do_get, flight.py) for the join/group-by
