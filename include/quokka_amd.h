@@ -293,6 +293,22 @@ int qk_partition_scatter(void *stream, uint64_t n, const int64_t *keys,
                          uint32_t nparts, uint64_t *cursors_dev,
                          uint32_t *out_idx);
 
+/* ---- stable radix sort ------------------------------------------------ *
+ * Replaces the sorts the reference delegates to polars
+ * (SuperFastSortExecutor sql_executors.py:88-187; build-side sort :369;
+ * order-by tails). Stable ascending LSD sort of (u64 key, u32 payload);
+ * keys_tmp/pay_tmp: caller scratch of same sizes; npasses -1 derives the
+ * pass count from the max key. Order-preserving key maps for f64/i64 and
+ * an iota payload helper included. */
+int qk_sort_pairs_u64(void *stream, uint64_t n, uint64_t *keys,
+                      uint32_t *payload, uint64_t *keys_tmp,
+                      uint32_t *pay_tmp, int npasses);
+int qk_map_f64_u64(void *stream, uint64_t n, const double *in, uint64_t *out);
+int qk_map_i64_u64(void *stream, uint64_t n, const int64_t *in,
+                   uint64_t *out);
+int qk_iota_u32(void *stream, uint64_t n, uint32_t *out);
+int qk_bnot_u64(void *stream, uint64_t n, uint64_t *x); /* stable desc sort */
+
 /* ---- hiprtc JIT (arbitrary predicates / transforms / partial aggs) --- *
  * Runtime-compiled gfx950 kernels for the reference's ARBITRARY
  * filter_sql predicates (core.py:157-170), transform_sql expressions

@@ -17,5 +17,6 @@ __version__ = "0.1.0"
 from .executors import (Executor, GPUAggExecutor,  # noqa: F401
                         GPUBroadcastJoinExecutor,
                         GPUBuildProbeJoinExecutor, GPUCountExecutor,
-                        GPUDistinctExecutor, GPUTopKExecutor,
+                        GPUDistinctExecutor, GPUSortExecutor,
+                        GPUTopKExecutor,
                         gpu_partition_fn)

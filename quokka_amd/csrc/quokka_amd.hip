@@ -1628,6 +1628,251 @@ extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
   return 0;
 }
 
+// ---- stable radix sort (u64 keys, u32 payload) -------------------------
+// Replaces the sort the reference delegates to polars (SuperFastSort,
+// sql_executors.py:88-187; build-side sort :369; order-by tails). LSD,
+// 8 bits/pass, pass count from the key range; stability per pass via
+// (block-major offsets) x (wave-serialized in-wave ballot ranks).
+
+#define RADIX 256
+
+__global__ void __launch_bounds__(BLOCK) k_radix_count(
+    uint64_t n, const uint64_t *__restrict__ keys, int shift, uint64_t chunk,
+    uint32_t *__restrict__ counts /* [RADIX][nblocks] */, uint32_t nblocks) {
+  __shared__ uint32_t hist[RADIX];
+  for (int d = threadIdx.x; d < RADIX; d += BLOCK) hist[d] = 0;
+  __syncthreads();
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  for (uint64_t i = lo + threadIdx.x; i < hi; i += BLOCK)
+    atomicAdd(&hist[(uint32_t)(keys[i] >> shift) & 255u], 1u);
+  __syncthreads();
+  for (int d = threadIdx.x; d < RADIX; d += BLOCK)
+    counts[(uint64_t)d * nblocks + blockIdx.x] = hist[d];
+}
+
+// one 256-thread block: thread d owns digit d's row of per-block counts;
+// produces exclusive global bases per (digit, block)
+__global__ void __launch_bounds__(RADIX) k_radix_scan(
+    uint32_t nblocks, uint32_t *__restrict__ counts,
+    uint64_t *__restrict__ bases /* [RADIX][nblocks] */) {
+  __shared__ uint64_t tot[RADIX];
+  int d = threadIdx.x;
+  uint64_t acc = 0;
+  for (uint32_t b = 0; b < nblocks; b++) {
+    uint32_t c = counts[(uint64_t)d * nblocks + b];
+    bases[(uint64_t)d * nblocks + b] = acc;
+    acc += c;
+  }
+  tot[d] = acc;
+  __syncthreads();
+  // exclusive scan of 256 digit totals (simple doubling scan)
+  for (int off = 1; off < RADIX; off <<= 1) {
+    uint64_t v = d >= off ? tot[d - off] : 0;
+    __syncthreads();
+    tot[d] += v;
+    __syncthreads();
+  }
+  uint64_t base = d == 0 ? 0 : tot[d - 1];
+  for (uint32_t b = 0; b < nblocks; b++)
+    bases[(uint64_t)d * nblocks + b] += base;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_radix_scatter(
+    uint64_t n, const uint64_t *__restrict__ keys_in,
+    const uint32_t *__restrict__ pay_in, int shift, uint64_t chunk,
+    const uint64_t *__restrict__ bases, uint32_t nblocks,
+    uint64_t *__restrict__ keys_out, uint32_t *__restrict__ pay_out) {
+  __shared__ uint64_t cursor[RADIX];
+  for (int d = threadIdx.x; d < RADIX; d += BLOCK)
+    cursor[d] = bases[(uint64_t)d * nblocks + blockIdx.x];
+  __syncthreads();
+  uint64_t lo = (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  for (uint64_t t0 = lo; t0 < hi; t0 += BLOCK) {
+    uint64_t i = t0 + threadIdx.x;
+    bool valid = i < hi;
+    uint64_t k = valid ? keys_in[i] : ~0ULL;
+    uint32_t pay = valid ? pay_in[i] : 0;
+    uint32_t d8 = (uint32_t)(k >> shift) & 255u;
+    // waves take turns IN ORDER: block stability = wave order x lane order
+    for (int w = 0; w < BLOCK / WAVE; w++) {
+      if (wid == w) {
+        uint64_t same = __ballot(valid);
+        for (int j = 0; j < 8; j++) {
+          uint64_t bj = __ballot((d8 >> j) & 1u);
+          same &= ((d8 >> j) & 1u) ? bj : ~bj;
+        }
+        uint32_t before = __popcll(same & ((1ULL << lane) - 1));
+        uint64_t base_leader = 0;
+        if (valid && before == 0)
+          base_leader = atomicAdd((unsigned long long *)&cursor[d8],
+                                  (unsigned long long)__popcll(same));
+        int leader = __ffsll((unsigned long long)same) - 1;
+        uint64_t base = __shfl(base_leader, leader < 0 ? 0 : leader);
+        if (valid) {
+          keys_out[base + before] = k;
+          pay_out[base + before] = pay;
+        }
+      }
+      __syncthreads();
+    }
+  }
+}
+
+__global__ void k_reduce_max_u64(uint64_t n, const uint64_t *__restrict__ in,
+                                 unsigned long long *__restrict__ out) {
+  uint64_t m = 0;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    m = in[i] > m ? in[i] : m;
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    uint64_t o = __shfl_down(m, off);
+    m = o > m ? o : m;
+  }
+  __shared__ uint64_t lds[BLOCK / WAVE];
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < BLOCK / WAVE; w++) m = lds[w] > m ? lds[w] : m;
+    atomicMax(out, (unsigned long long)m);
+  }
+}
+
+// order-preserving key maps: sort ascending on the u64 image == ascending
+// on the source type
+__global__ void k_map_f64_u64(uint64_t n, const double *__restrict__ in,
+                              uint64_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t b = (uint64_t)__double_as_longlong(in[i]);
+    out[i] = (b & 0x8000000000000000ULL) ? ~b : (b | 0x8000000000000000ULL);
+  }
+}
+__global__ void k_map_i64_u64(uint64_t n, const int64_t *__restrict__ in,
+                              uint64_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (uint64_t)in[i] ^ 0x8000000000000000ULL;
+}
+__global__ void k_bnot_u64(uint64_t n, uint64_t *__restrict__ x) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    x[i] = ~x[i];
+}
+__global__ void k_iota_u32(uint64_t n, uint32_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (uint32_t)i;
+}
+
+extern "C" int qk_map_f64_u64(void *stream, uint64_t n, const double *in,
+                              uint64_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_map_f64_u64, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, in, out);
+  QK_TRY("qk_map_f64_u64", hipGetLastError());
+  return 0;
+}
+extern "C" int qk_map_i64_u64(void *stream, uint64_t n, const int64_t *in,
+                              uint64_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_map_i64_u64, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, in, out);
+  QK_TRY("qk_map_i64_u64", hipGetLastError());
+  return 0;
+}
+extern "C" int qk_bnot_u64(void *stream, uint64_t n, uint64_t *x) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_bnot_u64, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, x);
+  QK_TRY("qk_bnot_u64", hipGetLastError());
+  return 0;
+}
+extern "C" int qk_iota_u32(void *stream, uint64_t n, uint32_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_iota_u32, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, out);
+  QK_TRY("qk_iota_u32", hipGetLastError());
+  return 0;
+}
+
+/* Stable ascending sort of (keys, payload) in place. keys_tmp/pay_tmp are
+ * caller scratch of the same sizes. npasses -1 = derive from max key. */
+extern "C" int qk_sort_pairs_u64(void *stream, uint64_t n, uint64_t *keys,
+                                 uint32_t *pay, uint64_t *keys_tmp,
+                                 uint32_t *pay_tmp, int npasses) {
+  if (n < 2) return 0;
+  uint64_t chunk = (n + MAX_BLOCKS - 1) / MAX_BLOCKS;
+  chunk = ((chunk + BLOCK - 1) / BLOCK) * BLOCK;
+  uint32_t nblocks = (uint32_t)((n + chunk - 1) / chunk);
+
+  static __thread uint32_t *counts = nullptr;
+  static __thread uint64_t *bases = nullptr;
+  static __thread unsigned long long *dmax = nullptr;
+  if (!counts) {
+    QK_TRY("qk_sort_pairs_u64",
+           hipMalloc(&counts, (uint64_t)RADIX * MAX_BLOCKS * 4));
+    QK_TRY("qk_sort_pairs_u64",
+           hipMalloc(&bases, (uint64_t)RADIX * MAX_BLOCKS * 8));
+    QK_TRY("qk_sort_pairs_u64", hipMalloc(&dmax, 8));
+  }
+  if (npasses < 0) {
+    QK_TRY("qk_sort_pairs_u64",
+           hipMemsetAsync(dmax, 0, 8, (hipStream_t)stream));
+    uint32_t b2 = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+    hipLaunchKernelGGL(k_reduce_max_u64, dim3(b2), dim3(BLOCK), 0,
+                       (hipStream_t)stream, n, keys, dmax);
+    unsigned long long hmax = 0;
+    QK_TRY("qk_sort_pairs_u64",
+           hipMemcpyAsync(&hmax, dmax, 8, hipMemcpyDeviceToHost,
+                          (hipStream_t)stream));
+    QK_TRY("qk_sort_pairs_u64", hipStreamSynchronize((hipStream_t)stream));
+    npasses = 0;
+    while (hmax) {
+      npasses++;
+      hmax >>= 8;
+    }
+    if (!npasses) npasses = 1;
+  }
+  uint64_t *ki = keys, *ko = keys_tmp;
+  uint32_t *pi = pay, *po = pay_tmp;
+  for (int p = 0; p < npasses; p++) {
+    int shift = 8 * p;
+    hipLaunchKernelGGL(k_radix_count, dim3(nblocks), dim3(BLOCK), 0,
+                       (hipStream_t)stream, n, ki, shift, chunk, counts,
+                       nblocks);
+    hipLaunchKernelGGL(k_radix_scan, dim3(1), dim3(RADIX), 0,
+                       (hipStream_t)stream, nblocks, counts, bases);
+    hipLaunchKernelGGL(k_radix_scatter, dim3(nblocks), dim3(BLOCK), 0,
+                       (hipStream_t)stream, n, ki, pi, shift, chunk, bases,
+                       nblocks, ko, po);
+    uint64_t *tk = ki; ki = ko; ko = tk;
+    uint32_t *tp = pi; pi = po; po = tp;
+  }
+  QK_TRY("qk_sort_pairs_u64", hipGetLastError());
+  if (ki != keys) {  // odd pass count: copy back
+    QK_TRY("qk_sort_pairs_u64",
+           hipMemcpyAsync(keys, ki, n * 8, hipMemcpyDeviceToDevice,
+                          (hipStream_t)stream));
+    QK_TRY("qk_sort_pairs_u64",
+           hipMemcpyAsync(pay, pi, n * 4, hipMemcpyDeviceToDevice,
+                          (hipStream_t)stream));
+  }
+  return 0;
+}
+
 // ---- RCCL exchange ----------------------------------------------------
 // Grouped send/recv all-to-allv over xGMI (direct per-peer, not a ring);
 // replaces the reference's Flight-based shuffle (core.py:276-376) for the

@@ -498,6 +498,49 @@ class GPUDistinctExecutor(Executor):
         return None
 
 
+class GPUSortExecutor(Executor):
+    """Mirror of SuperFastSortExecutor (sql_executors.py:88-187): the
+    operator that makes its output stream SORTED by `key`. The reference
+    implements it as a disk-backed external merge sort (spill files +
+    k-way merge) because its build ran on 64 GB hosts; with 288 GB of HBM
+    the state fits device memory, so this mirror accumulates batches and
+    done() returns the fully sorted table via the device radix sort
+    (stable, ascending — the reference's output contract)."""
+
+    def __init__(self, key, record_batch_rows=None, output_batch_rows=None,
+                 file_prefix=None):
+        # extra args accepted for signature compatibility (:89); the
+        # spill tuning knobs have no meaning on the device path
+        self.key = key
+        self._batches = None
+
+    def __getstate__(self):
+        assert self._batches is None, "pickle before first execute"
+        return dict(self.__dict__)
+
+    def execute(self, batches, stream_id, executor_id):
+        import pyarrow as pa
+        batches = [b for b in batches if b is not None and len(b) > 0]
+        if not batches:
+            return
+        b = pa.concat_tables(batches)
+        self._batches = b if self._batches is None             else pa.concat_tables([self._batches, b])
+        return None
+
+    def done(self, executor_id):
+        if self._batches is None:
+            return None
+        ops, shim, staging = _lazy_gpu()
+        t = self._batches
+        kcol = shim.DevColumn.from_numpy(
+            staging.column_to_numpy(t.column(self.key)))
+        perm = ops.sort_permutation(kcol)
+        order = perm.to_numpy(perm.n)
+        kcol.free()
+        perm.free()
+        return t.take(order)
+
+
 class GPUTopKExecutor(Executor):
     """Mirror of ConcatThenSQLExecutor as lowered by DataStream.top_k
     (sql_executors.py:45-67; datastream.py:1746-1767): concatenate incoming

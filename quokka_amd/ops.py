@@ -231,6 +231,43 @@ class GroupByI64:
             self._ops_dev.free()
 
 
+def sort_permutation(col, stream=None, descending=False):
+    """Stable sort permutation of a DevColumn (i64/f64/i32/u32 keys):
+    returns a u32 DevColumn `perm` with rows in ascending (or descending)
+    key order; equal keys keep their original relative order (stable,
+    like polars sort — sql_executors.py:369 build-side sort semantics).
+    Device LSD radix sort (qk_sort_pairs_u64)."""
+    n = col.n
+    sh = stream.handle if stream else None
+    keys = DevColumn(np.uint64, max(1, n))
+    if col.dtype == np.dtype(np.float64):
+        shim.call("qk_map_f64_u64", sh, c_u64(n), col.ptr, keys.ptr)
+    elif col.dtype == np.dtype(np.int64):
+        shim.call("qk_map_i64_u64", sh, c_u64(n), col.ptr, keys.ptr)
+    elif col.dtype in (np.dtype(np.int32), np.dtype(np.uint32)):
+        tmp = DevColumn(np.int64, max(1, n))
+        host = col.to_numpy(n).astype(np.int64)   # widen via host (i32 cols are small in this path)
+        tmp2 = DevColumn.from_numpy(host)
+        shim.call("qk_map_i64_u64", sh, c_u64(n), tmp2.ptr, keys.ptr)
+        tmp.free(); tmp2.free()
+    else:
+        raise TypeError("sort_permutation: unsupported dtype %s" % col.dtype)
+    if descending:
+        # stable descending == stable ascending on complemented key image
+        shim.call("qk_bnot_u64", sh, c_u64(n), keys.ptr)
+    perm = DevColumn(np.uint32, max(1, n))
+    shim.call("qk_iota_u32", sh, c_u64(n), perm.ptr)
+    ktmp = DevColumn(np.uint64, max(1, n))
+    ptmp = DevColumn(np.uint32, max(1, n))
+    shim.call("qk_sort_pairs_u64", sh, c_u64(n), keys.ptr, perm.ptr,
+              ktmp.ptr, ptmp.ptr, -1)
+    if stream:
+        stream.sync()
+    keys.free(); ktmp.free(); ptmp.free()
+    perm.n = n
+    return perm
+
+
 def partition_i64(keys_col, nparts, stream=None, n=None):
     """Hash partition, int-key semantics key % nparts
     (quokka_runtime.py:222). Returns (offsets np.uint64[nparts+1],

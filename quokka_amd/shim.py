@@ -97,6 +97,12 @@ _SIGS = {
     "qk_fill_f64": [c_vp, c_vp, c_f64, c_u64],
     "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,
                            c_u64, c_vp],
+    "qk_sort_pairs_u64": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp,
+                          ctypes.c_int],
+    "qk_map_f64_u64": [c_vp, c_u64, c_vp, c_vp],
+    "qk_map_i64_u64": [c_vp, c_u64, c_vp, c_vp],
+    "qk_iota_u32": [c_vp, c_u64, c_vp],
+    "qk_bnot_u64": [c_vp, c_u64, c_vp],
     "qk_partition_hist": [c_vp, c_u64, c_vp, c_u32, c_vp],
     "qk_partition_scatter": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp],
 }
