@@ -227,7 +227,9 @@ class Q3Fused:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
         cnt.free()
-        self.ord_cap = ops._pow2_at_least(max(16, 4 * self.n_build))
+        import os
+        lf = int(os.environ.get("QK_TABLE_LF", "4"))
+        self.ord_cap = ops._pow2_at_least(max(16, lf * self.n_build))
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_head = DevColumn(np.int32, self.ord_cap)
         self.ord_sums = DevColumn(np.float64, self.ord_cap)
@@ -498,7 +500,9 @@ class Q5Fused:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
         cnt.free()
-        self.ord_cap = ops._pow2_at_least(max(16, 4 * self.n_build))
+        import os
+        lf = int(os.environ.get("QK_TABLE_LF", "4"))
+        self.ord_cap = ops._pow2_at_least(max(16, lf * self.n_build))
         self.ord_keys = DevColumn(np.int64, self.ord_cap)
         self.ord_val = DevColumn(np.int32, self.ord_cap)
         self.bloom_bits = ops._pow2_at_least(max(1 << 16, 8 * self.n_build))
