@@ -63,7 +63,9 @@ def build_probe_join(build_keys, probe_keys, how="inner"):
 
 def partition_int(keys, num_target_channels):
     """quokka_runtime.py:222: int key -> key % num_target_channels.
-    Non-negative key domain (TPC-H keys are >= 0)."""
+    numpy % is the mathematical (non-negative) mod, matching the GPU
+    kernels for negative keys too (the reference's polars % yields
+    negative partition ids there and its runtime breaks — DESIGN.md)."""
     return np.asarray(keys, dtype=np.int64) % num_target_channels
 
 

@@ -1940,6 +1940,14 @@ extern "C" int qk_allreduce_f64(void *stream, void *comm, double *buf,
 }
 
 // ---- hash partition (int key: part = key % nparts, quokka_runtime:222) --
+// Negative keys: the reference's polars `%` yields NEGATIVE partition ids
+// and its runtime breaks on them; we keep equal-keys-colocate semantics
+// with the mathematical (non-negative) mod instead of corrupting memory.
+__device__ inline uint32_t part_of(int64_t key, uint32_t nparts) {
+  int64_t p = key % (int64_t)nparts;
+  return (uint32_t)(p < 0 ? p + (int64_t)nparts : p);
+}
+
 __global__ void __launch_bounds__(BLOCK) k_partition_hist(
     uint64_t n, const int64_t *__restrict__ keys, uint32_t nparts,
     uint64_t *__restrict__ hist) {
@@ -1950,7 +1958,7 @@ __global__ void __launch_bounds__(BLOCK) k_partition_hist(
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride)
-    atomicAdd(&lh[(uint32_t)(keys[i] % nparts)], 1u);
+    atomicAdd(&lh[part_of(keys[i], nparts)], 1u);
   __syncthreads();
   for (uint32_t p = threadIdx.x; p < nparts; p += BLOCK)
     if (lh[p]) atomicAdd((unsigned long long *)&hist[p], (unsigned long long)lh[p]);
@@ -1984,7 +1992,7 @@ __global__ void __launch_bounds__(BLOCK) k_partition_scatter(
     for (uint32_t p = threadIdx.x; p < nparts; p += BLOCK) lcnt[p] = 0;
     __syncthreads();
     for (uint64_t i = c0 + threadIdx.x; i < c1; i += BLOCK)
-      atomicAdd(&lcnt[(uint32_t)(keys[i] % nparts)], 1u);
+      atomicAdd(&lcnt[part_of(keys[i], nparts)], 1u);
     __syncthreads();
     for (uint32_t p = threadIdx.x; p < nparts; p += BLOCK) {
       lbase[p] = lcnt[p]
@@ -1995,7 +2003,7 @@ __global__ void __launch_bounds__(BLOCK) k_partition_scatter(
     }
     __syncthreads();
     for (uint64_t i = c0 + threadIdx.x; i < c1; i += BLOCK) {
-      uint32_t p = (uint32_t)(keys[i] % nparts);
+      uint32_t p = part_of(keys[i], nparts);
       uint32_t r = atomicAdd(&lcnt[p], 1u);
       out_idx[lbase[p] + r] = (uint32_t)i;
     }

@@ -254,6 +254,7 @@ def test_partition_parity_bitexact(gpu, nparts):
     from quokka_amd import ops, shim
     rng = np.random.default_rng(23)
     keys = rng.integers(0, 1 << 40, 300000).astype(np.int64)
+    keys[::97] = -keys[::97]          # negative keys: no OOB, colocated
     kcol = shim.DevColumn.from_numpy(keys)
     offsets, idx = ops.partition_i64(kcol, nparts)
     sel = idx.to_numpy(len(keys))
