@@ -71,7 +71,7 @@ def partition_int(keys, num_target_channels):
 
 def splitmix64(x):
     """The 64-bit finalizer quokka_amd uses for non-int partition keys and
-    for hash-table slots (kernel parity reference; see csrc/qk_common.h)."""
+    for hash-table slots (kernel parity reference; device twin in csrc/quokka_amd.hip)."""
     x = np.asarray(x, dtype=np.uint64).copy()
     x += np.uint64(0x9E3779B97F4A7C15)
     x ^= x >> np.uint64(30)
