@@ -859,6 +859,33 @@ __device__ inline int32_t probe_unique(const int64_t *__restrict__ slot_keys,
 
 // orders: insert o_orderkey[i] where o_orderdate[i] < date_lt AND
 // o_custkey[i] hits the customer table (fused filter + semi join + build)
+__device__ inline void q3_build_row(
+    int64_t ck, bool pass, uint64_t i,
+    const int64_t *__restrict__ o_orderkey,
+    const int64_t *__restrict__ cust_keys,
+    const int32_t *__restrict__ cust_head, uint64_t cust_cap,
+    int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
+    uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
+  if (!pass) return;
+  if (probe_unique(cust_keys, cust_head, cust_cap, ck, nullptr) < 0) return;
+  int64_t key = __builtin_nontemporal_load(&o_orderkey[i]);
+  uint64_t s = slot_of(key, cap);
+  for (;;) {
+    int64_t cur = slot_keys[s];
+    if (cur == key) break;
+    if (cur == QK_JOIN_EMPTY) {
+      int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+                                        (unsigned long long)QK_JOIN_EMPTY,
+                                        (unsigned long long)key);
+      if (prev == QK_JOIN_EMPTY || prev == key) break;
+    }
+    s = (s + 1) & (cap - 1);
+  }
+  slot_head[s] = (int32_t)i;
+  if (bloom) bloom_set(bloom, bloom_mask, key);
+}
+
+// 2 rows/thread (vector nt loads, two cust lookups/inserts in flight)
 __global__ void __launch_bounds__(BLOCK) k_q3_build_orders(
     uint64_t n, const int64_t *__restrict__ o_orderkey,
     const int64_t *__restrict__ o_custkey,
@@ -868,30 +895,35 @@ __global__ void __launch_bounds__(BLOCK) k_q3_build_orders(
     int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
     uint64_t cap, uint32_t *__restrict__ bloom, uint64_t bloom_mask,
     const uint32_t *__restrict__ cbloom, uint64_t cbloom_mask) {
+  typedef int v2i __attribute__((ext_vector_type(2)));
+  typedef long long v2l __attribute__((ext_vector_type(2)));
+  uint64_t npairs = n / 2;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    // nt loads: stream the orders columns without evicting the tables
-    if (__builtin_nontemporal_load(&o_orderdate[i]) >= date_lt) continue;
-    int64_t ck = __builtin_nontemporal_load(&o_custkey[i]);
-    if (cbloom && !bloom_test(cbloom, cbloom_mask, ck)) continue;
-    if (probe_unique(cust_keys, cust_head, cust_cap, ck, nullptr) < 0)
-      continue;
-    int64_t key = __builtin_nontemporal_load(&o_orderkey[i]);
-    uint64_t s = slot_of(key, cap);
-    for (;;) {
-      int64_t cur = slot_keys[s];
-      if (cur == key) break;
-      if (cur == QK_JOIN_EMPTY) {
-        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
-                                          (unsigned long long)QK_JOIN_EMPTY,
-                                          (unsigned long long)key);
-        if (prev == QK_JOIN_EMPTY || prev == key) break;
-      }
-      s = (s + 1) & (cap - 1);
+  for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       p < npairs; p += stride) {
+    uint64_t i = 2 * p;
+    v2i d2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2i *>(o_orderdate + i));
+    v2l c2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2l *>(o_custkey + i));
+    bool p0 = d2.x < date_lt, p1 = d2.y < date_lt;
+    if (cbloom) {
+      if (p0) p0 = bloom_test(cbloom, cbloom_mask, c2.x);
+      if (p1) p1 = bloom_test(cbloom, cbloom_mask, c2.y);
     }
-    slot_head[s] = (int32_t)i;
-    if (bloom) bloom_set(bloom, bloom_mask, key);
+    q3_build_row(c2.x, p0, i, o_orderkey, cust_keys, cust_head, cust_cap,
+                 slot_keys, slot_head, cap, bloom, bloom_mask);
+    q3_build_row(c2.y, p1, i + 1, o_orderkey, cust_keys, cust_head,
+                 cust_cap, slot_keys, slot_head, cap, bloom, bloom_mask);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    uint64_t i = n - 1;
+    if (o_orderdate[i] < date_lt) {
+      int64_t ck = o_custkey[i];
+      bool pass = !cbloom || bloom_test(cbloom, cbloom_mask, ck);
+      q3_build_row(ck, pass, i, o_orderkey, cust_keys, cust_head, cust_cap,
+                   slot_keys, slot_head, cap, bloom, bloom_mask);
+    }
   }
 }
 extern "C" int qk_q3_build_orders(void *stream, uint64_t n,
