@@ -90,15 +90,22 @@ def _tokenize(s):
     return out
 
 
+def _last_dom(y, m):
+    if m == 12:
+        return 31
+    return (datetime.date(y, m + 1, 1) - datetime.timedelta(days=1)).day
+
+
 def _date_add(days, n, unit):
     d = _EPOCH + datetime.timedelta(days=days)
     if unit == "day":
         d = d + datetime.timedelta(days=n)
-    elif unit == "month":
-        mo = d.month - 1 + n
-        d = d.replace(year=d.year + mo // 12, month=mo % 12 + 1)
     else:
-        d = d.replace(year=d.year + n)
+        months = n if unit == "month" else 12 * n
+        mo = d.month - 1 + months
+        y, m = d.year + mo // 12, mo % 12 + 1
+        # SQL interval semantics: clamp the day to the target month's end
+        d = datetime.date(y, m, min(d.day, _last_dom(y, m)))
     return (d - _EPOCH).days
 
 
@@ -186,8 +193,11 @@ class Translator:
             lo, _ = self.p_arith()
             self.take("AND")
             hi, _ = self.p_arith()
-            return "((%s) >= (%s) && (%s) <= (%s))" % (left, lo, left, hi)
+            return "(((%s) >= (%s)) && ((%s) <= (%s)))" % (left, lo,
+                                                            left, hi)
         op = self.take("op").val
+        if op not in ("<", "<=", ">", ">=", "=", "==", "<>", "!="):
+            raise ValueError("expected comparison operator, got %r" % op)
         cop = {"=": "==", "<>": "!="}.get(op, op)
         right, _ = self.p_arith()
         return "(%s) %s (%s)" % (left, cop, right)
@@ -223,6 +233,15 @@ class Translator:
 
     def p_atom(self):
         t = self.take()
+        if t.kind == "op" and t.val in ("-", "+"):
+            e, et = self.p_atom()
+            if t.val == "-":
+                if et == "lit":
+                    v = eval("-(%s)" % e)
+                    return (repr(float(v)) if isinstance(v, float)
+                            else repr(v)), "lit"
+                return "(-(%s))" % e, "mixed"
+            return e, et
         if t.kind == "op" and t.val == "(":
             e, et = self.p_arith()
             c = self.take("op")
