@@ -334,7 +334,7 @@ class JitAggregate:
         pred_expr = ""
         if predicate:
             pred_expr = tr.translate(predicate)
-        self.group_keys = list(group_keys)
+        self.group_keys = list(group_keys or [])
         self.ngroups = 1
         gparts = []
         for name, card in self.group_keys:
@@ -343,7 +343,7 @@ class JitAggregate:
                 raise TypeError("group keys must be u8 code columns")
             gparts.append((ref, card))
             self.ngroups *= int(card)
-        gexpr = "0"
+        gexpr = "0"   # [] group keys -> one grand-aggregate group
         for ref, card in gparts:
             gexpr = "(%s) * %d + (int)%s" % (gexpr, card, ref)
         self.agg_names = []
