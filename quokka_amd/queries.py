@@ -495,8 +495,7 @@ class Q5Fused:
              ord_cols["o_orderdate"].ptr, ctypes.c_int32(Q5_LO),
              ctypes.c_int32(Q5_HI), self.cust_keys.ptr, self.cust_val.ptr,
              c_u64(self.cust_cap), None, None, c_u64(16), cnt.ptr, None,
-             c_u64(0), self.cbloom.ptr, c_u64(self.cbloom_bits - 1),
-             None, c_u64(0))
+             c_u64(0), self.cbloom.ptr, c_u64(self.cbloom_bits - 1))
         if stream:
             stream.sync()
         self.n_build = ops._read_u64(cnt)
