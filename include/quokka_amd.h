@@ -235,8 +235,7 @@ int qk_q5_build_orders(void *stream, uint64_t n, const int64_t *o_orderkey,
                        int32_t *slot_val, uint64_t capacity,
                        uint64_t *count_dev, uint32_t *bloom,
                        uint64_t bloom_mask, const uint32_t *cust_bloom,
-                       uint64_t cust_bloom_mask, uint32_t *bloom_l1,
-                       uint64_t bloom_l1_mask);
+                       uint64_t cust_bloom_mask);
 /* Fused probe: join lineitem to orders (-> customer nation) and supplier
  * (-> supplier nation); where equal accumulate revenue into out25[nation].
  * out25: f64[32], zeroed (slots 25..31 unused). */
@@ -256,9 +255,7 @@ int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
                        const int64_t *supp_keys, const int32_t *supp_val,
                        uint64_t supp_cap, double *out25,
                        uint64_t *match_count_dev, const uint32_t *bloom,
-                       uint64_t bloom_mask, const uint32_t *bloom_l1,
-                       uint64_t bloom_l1_mask /* small L2-resident k=1
-                       pre-filter, nullable */);
+                       uint64_t bloom_mask);
 
 /* ---- group-by (i64 key) sum ------------------------------------------- *
  * Replaces SQLAggExecutor's DuckDB group-by (sql_executors.py:592-599) for

@@ -686,23 +686,6 @@ __device__ inline bool bloom_test(const uint32_t *bloom, uint64_t bloom_mask,
   return (bloom[w] & m) == m;
 }
 
-// Level-1 mini-bloom (k=1, sized to stay L2-resident on every XCD): a
-// cheap line-local pre-test in front of the main blocked Bloom. High FP
-// is fine — the L3-resident bloom2 still filters; what it buys is that
-// most rows touch only an L2 line instead of an L3 line.
-__device__ inline void bloom1_set(uint32_t *b1, uint64_t mask1,
-                                  int64_t key) {
-  uint64_t h = splitmix64((uint64_t)key ^ 0x1B10001B10001B10ULL);
-  uint64_t b = h & mask1;
-  atomicOr(&b1[b >> 5], 1u << (b & 31));
-}
-__device__ inline bool bloom1_test(const uint32_t *b1, uint64_t mask1,
-                                   int64_t key) {
-  uint64_t h = splitmix64((uint64_t)key ^ 0x1B10001B10001B10ULL);
-  uint64_t b = h & mask1;
-  return (b1[b >> 5] >> (b & 31)) & 1u;
-}
-
 __global__ void __launch_bounds__(BLOCK) k_join_build2(
     uint64_t n, const int64_t *__restrict__ keys, uint32_t row_offset,
     int64_t *__restrict__ slot_keys, int32_t *__restrict__ slot_head,
@@ -1324,8 +1307,7 @@ __global__ void __launch_bounds__(BLOCK) k_q5_build_orders(
     int32_t *__restrict__ slot_val, uint64_t cap,
     uint64_t *__restrict__ count, uint32_t *__restrict__ bloom,
     uint64_t bloom_mask, const uint32_t *__restrict__ cbloom,
-    uint64_t cbloom_mask, uint32_t *__restrict__ bloom1,
-    uint64_t bloom1_mask) {
+    uint64_t cbloom_mask) {
   uint32_t cnt = 0;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -1353,7 +1335,6 @@ __global__ void __launch_bounds__(BLOCK) k_q5_build_orders(
     }
     slot_val[s] = nat;
     if (bloom) bloom_set(bloom, bloom_mask, key);
-    if (bloom1) bloom1_set(bloom1, bloom1_mask, key);
   }
   if (count) {
     __shared__ uint32_t lds[BLOCK / WAVE];
@@ -1378,8 +1359,7 @@ extern "C" int qk_q5_build_orders(void *stream, uint64_t n,
                                   uint64_t cap, uint64_t *count_dev,
                                   uint32_t *bloom, uint64_t bloom_mask,
                                   const uint32_t *cbloom,
-                                  uint64_t cbloom_mask, uint32_t *bloom1,
-                                  uint64_t bloom1_mask) {
+                                  uint64_t cbloom_mask) {
   if (!n) return 0;
   if ((cust_cap & (cust_cap - 1)) || (slot_keys && (cap & (cap - 1))))
     return qk_fail("qk_q5_build_orders.cap_pow2", hipErrorInvalidValue);
@@ -1388,7 +1368,7 @@ extern "C" int qk_q5_build_orders(void *stream, uint64_t n,
                      (hipStream_t)stream, n, o_orderkey, o_custkey,
                      o_orderdate, date_lo, date_hi, cust_keys, cust_val,
                      cust_cap, slot_keys, slot_val, cap, count_dev, bloom,
-                     bloom_mask, cbloom, cbloom_mask, bloom1, bloom1_mask);
+                     bloom_mask, cbloom, cbloom_mask);
   QK_TRY("qk_q5_build_orders", hipGetLastError());
   return 0;
 }
@@ -1455,8 +1435,7 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
     uint64_t ord_cap, const int64_t *__restrict__ supp_keys,
     const int32_t *__restrict__ supp_val, uint64_t supp_cap,
     double *__restrict__ out25, uint64_t *__restrict__ match_count,
-    const uint32_t *__restrict__ bloom, uint64_t bloom_mask,
-    const uint32_t *__restrict__ bloom1, uint64_t bloom1_mask) {
+    const uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
   __shared__ double lsum[32];
   __shared__ uint32_t lcnt;
   if (threadIdx.x < 32) lsum[threadIdx.x] = 0.0;
@@ -1471,12 +1450,8 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
     typedef long long v2l __attribute__((ext_vector_type(2)));
     v2l k2 = __builtin_nontemporal_load(
         reinterpret_cast<const v2l *>(l_orderkey + i));
-    bool pass0 = !bloom1 || bloom1_test(bloom1, bloom1_mask, k2.x);
-    bool pass1 = !bloom1 || bloom1_test(bloom1, bloom1_mask, k2.y);
-    if (bloom) {
-      if (pass0) pass0 = bloom_test(bloom, bloom_mask, k2.x);
-      if (pass1) pass1 = bloom_test(bloom, bloom_mask, k2.y);
-    }
+    bool pass0 = !bloom || bloom_test(bloom, bloom_mask, k2.x);
+    bool pass1 = !bloom || bloom_test(bloom, bloom_mask, k2.y);
     q5_probe_row(k2.x, pass0, i, l_suppkey, l_price, l_disc, ord_keys,
                  ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
                  &lcnt, count);
@@ -1487,8 +1462,7 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
   if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
     uint64_t i = n - 1;
     int64_t okey = l_orderkey[i];
-    bool pass = (!bloom1 || bloom1_test(bloom1, bloom1_mask, okey)) &&
-                (!bloom || bloom_test(bloom, bloom_mask, okey));
+    bool pass = !bloom || bloom_test(bloom, bloom_mask, okey);
     q5_probe_row(okey, pass, i, l_suppkey, l_price, l_disc, ord_keys,
                  ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
                  &lcnt, count);
@@ -1529,9 +1503,7 @@ extern "C" int qk_q5_probe_agg_nt(void *stream, uint64_t n,
                                   const int32_t *supp_val, uint64_t supp_cap,
                                   double *out25, uint64_t *match_count,
                                   const uint32_t *bloom,
-                                  uint64_t bloom_mask,
-                                  const uint32_t *bloom1,
-                                  uint64_t bloom1_mask) {
+                                  uint64_t bloom_mask) {
   if (!n) return 0;
   if ((ord_cap & (ord_cap - 1)) || (supp_cap & (supp_cap - 1)))
     return qk_fail("qk_q5_probe_agg_nt.cap_pow2", hipErrorInvalidValue);
@@ -1539,8 +1511,7 @@ extern "C" int qk_q5_probe_agg_nt(void *stream, uint64_t n,
   hipLaunchKernelGGL(k_q5_probe_agg_nt, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, l_orderkey, l_suppkey, l_price,
                      l_disc, ord_keys, ord_val, ord_cap, supp_keys, supp_val,
-                     supp_cap, out25, match_count, bloom, bloom_mask, bloom1,
-                     bloom1_mask);
+                     supp_cap, out25, match_count, bloom, bloom_mask);
   QK_TRY("qk_q5_probe_agg_nt", hipGetLastError());
   return 0;
 }

@@ -508,16 +508,6 @@ class Q5Fused:
         self.bloom_bits = ops._pow2_at_least(max(1 << 16, 8 * self.n_build))
         self.bloom = DevColumn(np.uint32, self.bloom_bits // 32)
         call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
-        # level-1 mini-bloom: <= 4 MB so it stays resident in every XCD's
-        # L2 (the main bitmap's 64B line reads were L3-bandwidth bound)
-        import os as _os
-        self.bloom1_bits = min(1 << 25, self.bloom_bits)  # <= 4 MiB
-        if _os.environ.get("QK_BLOOM1", "1") == "0":
-            self.bloom1 = None
-        else:
-            self.bloom1 = DevColumn(np.uint32, self.bloom1_bits // 32)
-            call("qk_dmemset", self.bloom1.ptr, 0,
-                 c_u64(self.bloom1_bits // 8))
         call("qk_fill_i64", sh, self.ord_keys.ptr,
              c_i64(int(shim.JOIN_EMPTY)), c_u64(self.ord_cap))
         call("qk_q5_build_orders", sh, c_u64(nord),
@@ -527,9 +517,7 @@ class Q5Fused:
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
              c_u64(self.ord_cap), None, self.bloom.ptr,
              c_u64(self.bloom_bits - 1), self.cbloom.ptr,
-             c_u64(self.cbloom_bits - 1),
-             self.bloom1.ptr if self.bloom1 else None,
-             c_u64(self.bloom1_bits - 1))
+             c_u64(self.cbloom_bits - 1))
         self.out25 = DevBuffer(32 * 8)
         call("qk_dmemset", self.out25.ptr, 0, c_u64(32 * 8))
         self._ord_cols = ord_cols
@@ -565,9 +553,6 @@ class Q5Fused:
              ctypes.c_uint32(0xFFFFFFFF), self.supp_keys.ptr,
              self.supp_val.ptr, c_u64(self.supp_cap), None, c_u64(0))
         call("qk_dmemset", self.bloom.ptr, 0, c_u64(self.bloom_bits // 8))
-        if self.bloom1:
-            call("qk_dmemset", self.bloom1.ptr, 0,
-                 c_u64(self.bloom1_bits // 8))
         call("qk_q5_build_orders", sh, c_u64(self._ord_cols["o_orderkey"].n),
              self._ord_cols["o_orderkey"].ptr,
              self._ord_cols["o_custkey"].ptr,
@@ -576,9 +561,7 @@ class Q5Fused:
              c_u64(self.cust_cap), self.ord_keys.ptr, self.ord_val.ptr,
              c_u64(self.ord_cap), None, self.bloom.ptr,
              c_u64(self.bloom_bits - 1), self.cbloom.ptr,
-             c_u64(self.cbloom_bits - 1),
-             self.bloom1.ptr if self.bloom1 else None,
-             c_u64(self.bloom1_bits - 1))
+             c_u64(self.cbloom_bits - 1))
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         sh = self.stream.handle if self.stream else None
@@ -591,9 +574,7 @@ class Q5Fused:
                  self.supp_keys.ptr, self.supp_val.ptr,
                  c_u64(self.supp_cap), self.out25.ptr,
                  match_count_buf.ptr if match_count_buf else None,
-                 self.bloom.ptr, c_u64(self.bloom_bits - 1),
-                 self.bloom1.ptr if self.bloom1 else None,
-                 c_u64(self.bloom1_bits - 1))
+                 self.bloom.ptr, c_u64(self.bloom_bits - 1))
             return
         call("qk_q5_probe_agg", sh,
              c_u64(n), li_cols["l_orderkey"].ptr,
@@ -621,8 +602,6 @@ class Q5Fused:
                   self.supp_val, self.ord_keys, self.ord_val, self.bloom,
                   self.cbloom):
             c.free()
-        if self.bloom1:
-            self.bloom1.free()
         self.out25.free()
 
 

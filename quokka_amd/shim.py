@@ -73,7 +73,7 @@ _SIGS = {
                             c_u64, c_vp, c_u64],
     "qk_q5_build_orders": [c_vp, c_u64, c_vp, c_vp, c_vp, c_i32, c_i32,
                            c_vp, c_vp, c_u64, c_vp, c_vp, c_u64, c_vp, c_vp,
-                           c_u64, c_vp, c_u64, c_vp, c_u64],
+                           c_u64, c_vp, c_u64],
     "qk_q5_probe_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                         c_u64, c_vp, c_vp, c_u64, c_vp, c_vp],
     "qk_build_u8eq": [c_vp, c_u64, c_vp, c_vp, c_u8, c_vp, c_vp, c_u64,
@@ -89,7 +89,7 @@ _SIGS = {
                            c_vp, c_u64, c_vp, c_vp, c_vp, c_u64],
     "qk_q5_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                            c_u64, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
-                           c_u64, c_vp, c_u64],
+                           c_u64],
     "qk_q3_extract": [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
                       c_u64, c_vp],
     "qk_groupby_i64_sum": [c_vp, c_u64, c_vp, c_vp, c_vp, ctypes.c_int,
