@@ -1442,41 +1442,31 @@ __global__ void __launch_bounds__(BLOCK) k_q5_probe_agg_nt(
   if (threadIdx.x == 0) lcnt = 0;
   __syncthreads();
   bool count = match_count != nullptr;
-  uint64_t nquads = n / 4;
+  uint64_t npairs = n / 2;
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       p < nquads; p += stride) {
-    uint64_t i = 4 * p;
-    typedef long long v4l __attribute__((ext_vector_type(4)));
-    v4l k4 = __builtin_nontemporal_load(
-        reinterpret_cast<const v4l *>(l_orderkey + i));
-    // issue all four bloom lookups before any table walk (4x MLP on the
-    // random-load latency chain)
-    bool pass0 = !bloom || bloom_test(bloom, bloom_mask, k4.x);
-    bool pass1 = !bloom || bloom_test(bloom, bloom_mask, k4.y);
-    bool pass2 = !bloom || bloom_test(bloom, bloom_mask, k4.z);
-    bool pass3 = !bloom || bloom_test(bloom, bloom_mask, k4.w);
-    q5_probe_row(k4.x, pass0, i, l_suppkey, l_price, l_disc, ord_keys,
+       p < npairs; p += stride) {
+    uint64_t i = 2 * p;
+    typedef long long v2l __attribute__((ext_vector_type(2)));
+    v2l k2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2l *>(l_orderkey + i));
+    bool pass0 = !bloom || bloom_test(bloom, bloom_mask, k2.x);
+    bool pass1 = !bloom || bloom_test(bloom, bloom_mask, k2.y);
+    q5_probe_row(k2.x, pass0, i, l_suppkey, l_price, l_disc, ord_keys,
                  ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
                  &lcnt, count);
-    q5_probe_row(k4.y, pass1, i + 1, l_suppkey, l_price, l_disc, ord_keys,
-                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
-                 &lcnt, count);
-    q5_probe_row(k4.z, pass2, i + 2, l_suppkey, l_price, l_disc, ord_keys,
-                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
-                 &lcnt, count);
-    q5_probe_row(k4.w, pass3, i + 3, l_suppkey, l_price, l_disc, ord_keys,
+    q5_probe_row(k2.y, pass1, i + 1, l_suppkey, l_price, l_disc, ord_keys,
                  ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
                  &lcnt, count);
   }
-  if (blockIdx.x == 0 && threadIdx.x == 0)
-    for (uint64_t i = n & ~3ULL; i < n; i++) {
-      int64_t okey = l_orderkey[i];
-      bool pass = !bloom || bloom_test(bloom, bloom_mask, okey);
-      q5_probe_row(okey, pass, i, l_suppkey, l_price, l_disc, ord_keys,
-                   ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
-                   &lcnt, count);
-    }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    uint64_t i = n - 1;
+    int64_t okey = l_orderkey[i];
+    bool pass = !bloom || bloom_test(bloom, bloom_mask, okey);
+    q5_probe_row(okey, pass, i, l_suppkey, l_price, l_disc, ord_keys,
+                 ord_val, ord_cap, supp_keys, supp_val, supp_cap, lsum,
+                 &lcnt, count);
+  }
   __syncthreads();
   if (threadIdx.x < 32 && lsum[threadIdx.x] != 0.0)
     atomicAdd(&out25[threadIdx.x], lsum[threadIdx.x]);
