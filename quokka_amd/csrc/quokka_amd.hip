@@ -51,6 +51,10 @@ extern "C" int qk_device_count(int *out) {
 }
 
 // ---- memory -----------------------------------------------------------
+extern "C" int qk_hmalloc_impl(uint64_t nbytes, void **hptr) {
+  QK_TRY("qk_hmalloc", hipHostMalloc(hptr, nbytes ? nbytes : 1));
+  return 0;
+}
 extern "C" int qk_dmalloc(uint64_t nbytes, void **dptr) {
   QK_TRY("qk_dmalloc", hipMalloc(dptr, nbytes ? nbytes : 1));
   return 0;

@@ -36,6 +36,7 @@ _SIGS = {
     "qk_init": [ctypes.c_int],
     "qk_device_count": [c_vp],
     "qk_dmalloc": [c_u64, c_vp],
+    "qk_hmalloc_impl": [c_u64, c_vp],
     "qk_dfree": [c_vp],
     "qk_h2d": [c_vp, c_vp, c_u64],
     "qk_d2h": [c_vp, c_vp, c_u64],
@@ -145,6 +146,56 @@ def device_count():
 
 JOIN_EMPTY = np.int64(-(2 ** 63))  # QK_JOIN_EMPTY
 
+class _PinnedBounce:
+    """One lazily-allocated pinned host buffer used as a bounce for
+    pageable h2d/d2h copies (PCIe runs ~2-3x faster from pinned memory;
+    the executor boundary hands us pageable numpy arrays)."""
+
+    CHUNK = 64 << 20
+
+    def __init__(self):
+        self.buf = None
+
+    def _ensure(self):
+        if self.buf is None:
+            p = c_vp(0)
+            rc = _lib.qk_hmalloc_impl(c_u64(self.CHUNK), ctypes.byref(p))
+            if rc != 0:
+                return None
+            self.buf = p
+        return self.buf
+
+    def h2d(self, dst_dev, src_arr):
+        nbytes = src_arr.nbytes
+        if nbytes < (8 << 20) or self._ensure() is None:
+            call("qk_h2d", dst_dev, src_arr.ctypes.data_as(c_vp),
+                 c_u64(nbytes))
+            return
+        off = 0
+        while off < nbytes:
+            m = min(self.CHUNK, nbytes - off)
+            ctypes.memmove(self.buf,
+                           src_arr.ctypes.data_as(c_vp).value + off, m)
+            call("qk_h2d", c_vp(dst_dev.value + off), self.buf, c_u64(m))
+            off += m
+
+    def d2h(self, dst_arr, src_dev):
+        nbytes = dst_arr.nbytes
+        if nbytes < (8 << 20) or self._ensure() is None:
+            call("qk_d2h", dst_arr.ctypes.data_as(c_vp), src_dev,
+                 c_u64(nbytes))
+            return
+        off = 0
+        while off < nbytes:
+            m = min(self.CHUNK, nbytes - off)
+            call("qk_d2h", self.buf, c_vp(src_dev.value + off), c_u64(m))
+            ctypes.memmove(dst_arr.ctypes.data_as(c_vp).value + off,
+                           self.buf, m)
+            off += m
+
+
+_bounce = _PinnedBounce()
+
 _DTYPE_GATHER = {
     np.dtype(np.int64): "qk_gather_i64",
     np.dtype(np.float64): "qk_gather_f64",
@@ -238,16 +289,14 @@ class DevColumn:
         arr = np.ascontiguousarray(arr)
         col = cls(arr.dtype, len(arr))
         if len(arr):
-            call("qk_h2d", col.ptr, arr.ctypes.data_as(c_vp),
-                 c_u64(arr.nbytes))
+            _bounce.h2d(col.ptr, arr)
         return col
 
     def to_numpy(self, n=None):
         n = self.n if n is None else int(n)
         out = np.empty(n, dtype=self.dtype)
         if n:
-            call("qk_d2h", out.ctypes.data_as(c_vp), self.ptr,
-                 c_u64(out.nbytes))
+            _bounce.d2h(out, self.ptr)
         return out
 
     def gather(self, idx_col, n_idx, stream=None):
