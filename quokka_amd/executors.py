@@ -110,7 +110,11 @@ class GPUBuildProbeJoinExecutor(Executor):
                             for c, v in self._build_cols.items()}
         keys = self._host_build[self.right_on]
         if keys.dtype != np.int64:
-            raise TypeError("GPU join requires i64 keys, got %s" % keys.dtype)
+            if keys.dtype.kind in "iu":   # any int key, as the reference
+                keys = keys.astype(np.int64)
+            else:
+                raise TypeError("GPU join requires integer keys, got %s"
+                                % keys.dtype)
         n = len(keys)
         self._table = ops.JoinTable(max(16, n))
         if n:
@@ -129,7 +133,11 @@ class GPUBuildProbeJoinExecutor(Executor):
         ops, shim, staging = _lazy_gpu()
         probe_keys = staging.column_to_numpy(batch.column(self.left_on))
         if probe_keys.dtype != np.int64:
-            raise TypeError("GPU join requires i64 keys")
+            if probe_keys.dtype.kind in "iu":
+                probe_keys = probe_keys.astype(np.int64)
+            else:
+                raise TypeError("GPU join requires integer keys, got %s"
+                                % probe_keys.dtype)
         kcol = shim.DevColumn.from_numpy(probe_keys)
         mode = {"inner": 0, "left": 0, "semi": 1, "anti": 2}[self.how]
         pidx, bidx, nm = self._table.probe(kcol, mode=mode)

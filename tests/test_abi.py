@@ -178,3 +178,13 @@ def test_jit_aggregate_compiles_cpu():
     import pytest as _p
     with _p.raises(Exception):
         jit.JitAggregate(schema, [("g", 200)], ["sum(x)"])  # >64 accums
+
+
+def test_jit_filter_col_cap_clean_error():
+    import numpy as np
+    import pytest as _p
+    from quokka_amd import jit, shim
+    schema = {("c%d" % i): np.dtype(np.float64) for i in range(10)}
+    pred = " and ".join("c%d > 0" % i for i in range(10))
+    with _p.raises(shim.QkError, match="ncols"):
+        jit.JitFilter(pred, schema)

@@ -1019,3 +1019,18 @@ def test_sort_executor_plugin_api(gpu):
     order = np.argsort(k, kind="stable")
     assert out.column("k").to_pylist() == k[order].tolist()
     assert out.column("v").to_pylist() == v[order].tolist()
+
+
+def test_executor_join_i32_keys_widened(gpu):
+    """Integer keys narrower than i64 are accepted (the reference joins on
+    any int dtype); widened at the boundary."""
+    import pyarrow as pa
+    from quokka_amd import GPUBuildProbeJoinExecutor
+    bk = np.arange(0, 50, dtype=np.int32)
+    pk = np.array([1, 7, 7, 99, 3], dtype=np.int32)
+    ex = GPUBuildProbeJoinExecutor(left_on="pk", right_on="bk", how="inner")
+    ex.execute([pa.table({"bk": bk, "t": bk * 2})], 1, 0)
+    out = ex.execute([pa.table({"pk": pk})], 0, 0)
+    got = sorted(zip(out.column("pk").to_pylist(),
+                     out.column("t").to_pylist()))
+    assert got == [(1, 2), (3, 6), (7, 14), (7, 14)]
