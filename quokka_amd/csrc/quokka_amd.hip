@@ -1520,6 +1520,47 @@ extern "C" int qk_q5_probe_agg_nt(void *stream, uint64_t n,
   return 0;
 }
 
+// diagnostic: bloom-test-only pass (counts survivors) — isolates the
+// Bloom lookup cost from the table walk for the split-probe decision
+__global__ void __launch_bounds__(BLOCK) k_bloom_count(
+    uint64_t n, const int64_t *__restrict__ keys,
+    const uint32_t *__restrict__ bloom, uint64_t bloom_mask,
+    uint64_t *__restrict__ out) {
+  uint32_t cnt = 0;
+  uint64_t npairs = n / 2;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       p < npairs; p += stride) {
+    uint64_t i = 2 * p;
+    typedef long long v2l __attribute__((ext_vector_type(2)));
+    v2l k2 = __builtin_nontemporal_load(
+        reinterpret_cast<const v2l *>(keys + i));
+    cnt += bloom_test(bloom, bloom_mask, k2.x) ? 1u : 0u;
+    cnt += bloom_test(bloom, bloom_mask, k2.y) ? 1u : 0u;
+  }
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t t = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+    if (t) atomicAdd((unsigned long long *)out, (unsigned long long)t);
+  }
+}
+extern "C" int qk_bloom_count(void *stream, uint64_t n, const int64_t *keys,
+                              const uint32_t *bloom, uint64_t bloom_mask,
+                              uint64_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n / 2 + BLOCK - 1) / BLOCK);
+  if (!blocks) blocks = 1;
+  hipLaunchKernelGGL(k_bloom_count, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, bloom, bloom_mask, out);
+  QK_TRY("qk_bloom_count", hipGetLastError());
+  return 0;
+}
+
 __global__ void k_gen_supplier(uint64_t n, uint64_t row_offset, uint64_t seed,
                                int64_t *s_suppkey, int32_t *s_nationkey) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;

@@ -98,6 +98,7 @@ _SIGS = {
     "qk_fill_f64": [c_vp, c_vp, c_f64, c_u64],
     "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,
                            c_u64, c_vp],
+    "qk_bloom_count": [c_vp, c_u64, c_vp, c_vp, c_u64, c_vp],
     "qk_sort_pairs_u64": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp,
                           ctypes.c_int],
     "qk_map_f64_u64": [c_vp, c_u64, c_vp, c_vp],
