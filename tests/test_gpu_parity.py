@@ -1034,3 +1034,27 @@ def test_executor_join_i32_keys_widened(gpu):
     got = sorted(zip(out.column("pk").to_pylist(),
                      out.column("t").to_pylist()))
     assert got == [(1, 2), (3, 6), (7, 14), (7, 14)]
+
+
+def test_parquet_scan_to_gpu_q1(gpu, tmp_path, data):
+    """End-to-end scan side: lineitem written to
+    Parquet, read back with column selection (as the reference's
+    InputParquetDataset does, unordered_readers.py:73-99), staged to HBM,
+    fused Q1 == oracle."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from quokka_amd import staging, queries as DQ
+    li = data["lineitem"]
+    path = str(tmp_path / "lineitem.parquet")
+    pq.write_table(pa.table({k: v for k, v in li.items()}), path)
+    cols_needed = ["l_shipdate", "l_quantity", "l_extendedprice",
+                   "l_discount", "l_tax", "l_returnflag", "l_linestatus"]
+    t = pq.read_table(path, columns=cols_needed)   # column pushdown
+    dcols = staging.stage_columns(t)
+    got = DQ.q1(dcols)
+    want = OQ.q1(li)
+    assert np.array_equal(got["count_order"], want["count_order"])
+    for c in ("sum_qty", "sum_charge", "avg_disc"):
+        np.testing.assert_allclose(got[c], want[c], rtol=1e-9)
+    for c in dcols.values():
+        c.free()
