@@ -100,24 +100,35 @@ def cpu_baseline(sample_rows, target_seconds=12.0):
     }
 
 
-def gen_device_q3_tables(shim, n, rank):
-    """Lineitem (4 cols) + orders + customer shards for the fused Q3."""
+def gen_device_q3_tables(shim, n, rank, world=1, n_total=None):
+    """Lineitem (4 cols) + orders + customer for the fused Q3.
+
+    world == 1: full tables on this rank. world > 1 (exchange mode,
+    configs[3]): lineitem and orders are SHARDED by row range across
+    ranks over the TOTAL key space (n_total rows overall); customer is
+    replicated (broadcast side)."""
     from quokka_amd.shim import DevColumn, c_u64, c_i64
-    n_ord = max(1, n // 4)
-    n_cust = max(1, n_ord // 10)
+    n_total = n_total or n * world
+    n_ord_total = max(1, n_total // 4)
+    n_cust = max(1, n_ord_total // 10)
     li = {k: DevColumn(dt, n) for k, dt in [
         ("l_orderkey", np.int64), ("l_shipdate", np.int32),
         ("l_extendedprice", np.float64), ("l_discount", np.float64)]}
     shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
-              c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord_total),
               li["l_orderkey"].ptr, None, None, li["l_extendedprice"].ptr,
               li["l_discount"].ptr, None, None, None, li["l_shipdate"].ptr)
-    od = {k: DevColumn(dt, n_ord) for k, dt in [
+    n_ord_local = n_ord_total // world if world > 1 else n_ord_total
+    ord_off = rank * n_ord_local if world > 1 else 0
+    if world > 1 and rank == world - 1:      # last rank takes the remainder
+        n_ord_local = n_ord_total - ord_off
+    od = {k: DevColumn(dt, n_ord_local) for k, dt in [
         ("o_orderkey", np.int64), ("o_custkey", np.int64),
         ("o_orderdate", np.int32), ("o_shippriority", np.int32)]}
-    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
-              c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
-              od["o_orderdate"].ptr, od["o_shippriority"].ptr)
+    shim.call("qk_gen_orders", None, c_u64(n_ord_local), c_u64(ord_off),
+              c_u64(42), c_i64(n_cust), od["o_orderkey"].ptr,
+              od["o_custkey"].ptr, od["o_orderdate"].ptr,
+              od["o_shippriority"].ptr)
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_mktsegment": DevColumn(np.uint8, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
@@ -241,7 +252,8 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         n_local = (n // world) & ~3
     else:
         n_local = n
-    li, od, cu = gen_device_q3_tables(shim, n_local, rank)
+    li, od, cu = gen_device_q3_tables(shim, n_local, rank, world,
+                                      n_total=n if use_exchange else None)
     stream = shim.Stream()
     comm = exchange.Comm(rank, world, dist) if use_exchange else None
 
