@@ -1058,3 +1058,25 @@ def test_parquet_scan_to_gpu_q1(gpu, tmp_path, data):
         np.testing.assert_allclose(got[c], want[c], rtol=1e-9)
     for c in dcols.values():
         c.free()
+
+
+def test_executor_left_join_plugin_api(gpu):
+    """LEFT join through the executor: matched pairs + unmatched probe
+    rows with null build payload (sql_executors.py:341 how set)."""
+    import pyarrow as pa
+    from quokka_amd import GPUBuildProbeJoinExecutor
+    bk = np.array([1, 2, 2, 5], dtype=np.int64)
+    bv = np.array([10.0, 20.0, 21.0, 50.0])
+    pk = np.array([2, 3, 1, 7, 2], dtype=np.int64)
+    ex = GPUBuildProbeJoinExecutor(left_on="pk", right_on="bk", how="left")
+    ex.execute([pa.table({"bk": bk, "bv": bv})], 1, 0)
+    out = ex.execute([pa.table({"pk": pk, "row": np.arange(5)})], 0, 0)
+    got = sorted(
+        (r["pk"], r["row"], r["bv"])
+        for r in out.to_pylist())
+    want = sorted([
+        (2, 0, 20.0), (2, 0, 21.0), (1, 2, 10.0),
+        (2, 4, 20.0), (2, 4, 21.0),
+        (3, 1, None), (7, 3, None),
+    ])
+    assert got == want
