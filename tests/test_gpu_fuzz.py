@@ -120,11 +120,11 @@ def test_fuzz_jit_filter(gpu):
                    "b": np.round(rng.uniform(-10, 10, n), 3),
                    "k": rng.integers(-1000, 1000, n).astype(np.int64)}
         t1 = "a %s %d" % (ops_sql[rng.integers(0, 6)],
-                          rng.integers(-120, 120))
+                          int(rng.integers(-120, 120)))
         t2 = "b between %r and %r" % tuple(
-            sorted(np.round(rng.uniform(-12, 12, 2), 3)))
+            float(x) for x in sorted(np.round(rng.uniform(-12, 12, 2), 3)))
         t3 = "k %s %d" % (ops_sql[rng.integers(0, 6)],
-                          rng.integers(-1200, 1200))
+                          int(rng.integers(-1200, 1200)))
         conn = [" and ", " or "][rng.integers(0, 2)]
         pred = ("not (%s)%s(%s) and %s" % (t1, conn, t2, t3)
                 if rng.integers(0, 2) else "(%s)%s(%s)" % (t1, conn, t3))
