@@ -91,3 +91,37 @@ def test_repartition_count_bookkeeping():
                    for k in per_rank_keys)
         assert got == want
     assert send.sum() == sum(len(k) for k in per_rank_keys)
+
+
+def test_top10_merge_from_disjoint_rank_shards():
+    """bench.py main_q3 merges per-rank Q3 top-10s by concatenating the
+    candidates and re-running _topk; with disjoint per-rank orderkey
+    groups (`orderkey % world` sharding) that must equal the global
+    top-10 (revenue desc, orderdate asc, orderkey asc). Pure CPU check
+    of the merge rule."""
+    from quokka_amd.queries import _topk
+    rng = np.random.default_rng(11)
+    world = 4
+    n = 5_000
+    keys = rng.permutation(n).astype(np.int64)
+    full = {
+        "l_orderkey": keys,
+        "revenue": np.round(rng.uniform(0, 1e6, n), 2),
+        "o_orderdate": rng.integers(9000, 9300, n).astype(np.int32),
+        "o_shippriority": np.zeros(n, dtype=np.int32),
+    }
+    # force some revenue ties across ranks to exercise the tie-break
+    full["revenue"][keys % 17 == 0] = 999999.0
+    sel = _topk(full, 10)
+    want = {c: v[sel] for c, v in full.items()}
+    per_rank = []
+    for r in range(world):
+        m = keys % world == r
+        shard = {c: v[m] for c, v in full.items()}
+        s = _topk(shard, 10)
+        per_rank.append({c: v[s] for c, v in shard.items()})
+    cand = {c: np.concatenate([g[c] for g in per_rank]) for c in full}
+    msel = _topk(cand, 10)
+    got = {c: v[msel] for c, v in cand.items()}
+    for c in full:
+        assert np.array_equal(got[c], want[c]), c
