@@ -1986,24 +1986,55 @@ extern "C" int qk_comm_destroy(void *comm) {
   QK_NCCL("qk_comm_destroy", ncclCommDestroy((ncclComm_t)comm));
   return 0;
 }
+// RCCL 2.27 p2p SILENTLY DELIVERS ONLY THE FIRST HALF of a send whose
+// byte count exceeds 1 GiB (measured on MI355X: exact at 2^30 B, exactly
+// half the rows at every size above — scripts/diag_exchange.py). Two
+// defenses: the self-partition never touches RCCL (plain async D2D copy),
+// and peer pieces are split into <=QK_P2P_CHUNK-byte sends/recvs — both
+// ranks derive the identical split from the exchanged counts, so the
+// grouped send/recv matching stays aligned.
+static const uint64_t QK_P2P_CHUNK = 256ull << 20;
+
 extern "C" int qk_alltoallv(void *stream, void *comm, int world,
                             uint32_t elem_size, const void *send_buf,
                             const uint64_t *send_offsets,
                             const uint64_t *send_counts, void *recv_buf,
                             const uint64_t *recv_offsets,
                             const uint64_t *recv_counts) {
+  int rank = 0;
+  QK_NCCL("qk_alltoallv.rank",
+          ncclCommUserRank((ncclComm_t)comm, &rank));
+  if (send_counts[rank]) {
+    if (send_counts[rank] != recv_counts[rank])
+      return qk_fail("qk_alltoallv.self_count", hipErrorInvalidValue);
+    QK_TRY("qk_alltoallv.self_copy",
+           hipMemcpyAsync(
+               (char *)recv_buf + recv_offsets[rank] * elem_size,
+               (const char *)send_buf + send_offsets[rank] * elem_size,
+               send_counts[rank] * elem_size, hipMemcpyDeviceToDevice,
+               (hipStream_t)stream));
+  }
   QK_NCCL("qk_alltoallv.group_start", ncclGroupStart());
   for (int p = 0; p < world; p++) {
-    if (send_counts[p])
+    if (p == rank) continue;
+    for (uint64_t off = 0, nb = send_counts[p] * elem_size; off < nb;
+         off += QK_P2P_CHUNK) {
+      uint64_t m = nb - off < QK_P2P_CHUNK ? nb - off : QK_P2P_CHUNK;
       QK_NCCL("qk_alltoallv.send",
-              ncclSend((const char *)send_buf + send_offsets[p] * elem_size,
-                       send_counts[p] * elem_size, ncclUint8, p,
-                       (ncclComm_t)comm, (hipStream_t)stream));
-    if (recv_counts[p])
+              ncclSend((const char *)send_buf +
+                           send_offsets[p] * elem_size + off,
+                       m, ncclUint8, p, (ncclComm_t)comm,
+                       (hipStream_t)stream));
+    }
+    for (uint64_t off = 0, nb = recv_counts[p] * elem_size; off < nb;
+         off += QK_P2P_CHUNK) {
+      uint64_t m = nb - off < QK_P2P_CHUNK ? nb - off : QK_P2P_CHUNK;
       QK_NCCL("qk_alltoallv.recv",
-              ncclRecv((char *)recv_buf + recv_offsets[p] * elem_size,
-                       recv_counts[p] * elem_size, ncclUint8, p,
-                       (ncclComm_t)comm, (hipStream_t)stream));
+              ncclRecv((char *)recv_buf +
+                           recv_offsets[p] * elem_size + off,
+                       m, ncclUint8, p, (ncclComm_t)comm,
+                       (hipStream_t)stream));
+    }
   }
   QK_NCCL("qk_alltoallv.group_end", ncclGroupEnd());
   return 0;

@@ -587,6 +587,58 @@ def test_exchange_world1_roundtrip(gpu):
     comm.destroy()
 
 
+def test_exchange_world1_over_1gib(gpu):
+    """Regression: RCCL 2.27 p2p silently delivers only the FIRST HALF of
+    any single send whose byte count exceeds 1 GiB (measured on MI355X,
+    scripts/diag_exchange.py — exact at 2^30 B, half the rows above).
+    qk_alltoallv must route the self-partition through a D2D copy and
+    chunk peer pieces, so a >1 GiB column survives the exchange intact."""
+    from quokka_amd import exchange, shim
+    from quokka_amd.shim import DevColumn, c_u64
+    n = 300_000_000                      # u32 -> 1.2 GB, over the 1 GiB cliff
+    col = DevColumn(np.uint32, n)
+    shim.call("qk_iota_u32", None, c_u64(n), col.ptr)
+    comm = exchange.Comm(0, 1)
+    so = np.zeros(1, dtype=np.uint64)
+    sc = np.array([n], dtype=np.uint64)
+    recv = comm.alltoallv_column(col, so, sc, sc)
+    got = recv.to_numpy(n)
+    bad = np.nonzero(got != np.arange(n, dtype=np.uint32))[0]
+    assert bad.size == 0, "first bad idx %d of %d" % (bad[0], bad.size)
+    recv.free()
+    col.free()
+    # and through the full repartition (partition -> gather -> exchange)
+    # with device-generated i64 keys (1.28 GiB key column)
+    from quokka_amd.shim import c_i64
+    nk = 160_000_000
+    kcol = DevColumn(np.int64, nk)
+    dat = DevColumn(np.int32, nk)
+    pr = DevColumn(np.float64, nk)
+    di = DevColumn(np.float64, nk)
+    shim.call("qk_gen_lineitem", None, c_u64(nk), c_u64(0), c_u64(42),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(nk // 4),
+              kcol.ptr, None, None, pr.ptr, di.ptr, None, None, None,
+              dat.ptr)
+    pr.free()
+    di.free()
+    want_k = kcol.to_numpy(nk)
+    want_d = dat.to_numpy(nk)
+    rk, rp = exchange.repartition(comm, kcol, {"d": dat})
+    hk = rk.to_numpy(nk)
+    hd = rp["d"].to_numpy(nk)
+    # permutation-invariant checks (sum/xor/min/max) on both columns
+    for got, want in ((hk, want_k), (hd, want_d)):
+        assert int(np.sum(got, dtype=np.uint64)) == \
+            int(np.sum(want, dtype=np.uint64))
+        assert int(got.min()) == int(want.min())
+        assert int(got.max()) == int(want.max())
+    rk.free()
+    rp["d"].free()
+    kcol.free()
+    dat.free()
+    comm.destroy()
+
+
 def test_exchange_allreduce_world1(gpu):
     from quokka_amd import exchange, shim
     from quokka_amd.shim import DevColumn
