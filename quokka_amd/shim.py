@@ -215,7 +215,14 @@ class _DevPool:
 
     def __init__(self, cap_bytes=None):
         if cap_bytes is None:
-            cap_bytes = int(os.environ.get("QK_POOL_CAP_GB", "24")) << 30
+            # Cap bounds CACHED FREE bytes only (live allocations are not
+            # counted). 288 GB HBM per GPU, one process per GPU: a large
+            # cap is safe, and it matters — the SF100 exchange step frees/
+            # reallocs ~20 GB; with a small cap those become real
+            # hipFree/hipMalloc calls, which on some boxes cost ~50 ms per
+            # multi-GB buffer (measured 298 vs 40 ms/step box-to-box until
+            # the working set fit the pool).
+            cap_bytes = int(os.environ.get("QK_POOL_CAP_GB", "128")) << 30
         self.cap = cap_bytes
         self.cached = 0
         self.buckets = {}
