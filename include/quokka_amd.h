@@ -364,6 +364,31 @@ int qk_alltoallv(void *stream, void *comm, int world, uint32_t elem_size,
 /* All-reduce SUM of an f64 device buffer (tiny partial-agg combine). */
 int qk_allreduce_f64(void *stream, void *comm, double *buf, uint64_t n);
 
+/* ---- GPU Parquet page decode ------------------------------------------ *
+ * The reference scans Parquet with pyarrow's CPU reader
+ * (pyquokka/dataset.py InputParquetDataset.get_next_batch ->
+ * pyarrow.parquet). Here the host parses only METADATA (footer via
+ * pyarrow; per-page Thrift headers and RLE run boundaries in
+ * quokka_amd/parquet_thrift.py + parquet_gpu.py) and the raw column
+ * chunk bytes are uploaded once; these kernels do all VALUE decoding in
+ * HBM. Host-built tile tables (uploaded as u64 arrays) give each
+ * workgroup one bounded slice of work.
+ *
+ * PLAIN pages (fixed-width): tiles = ntiles x 3 u64
+ * [src_byte_off, dst_elem_off, count]; elem-wise copy of `count` values
+ * of `elem_size` bytes from src_bytes+src_byte_off (byte-aligned, may be
+ * unaligned) to dst + dst_elem_off. */
+int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
+                     const void *src_bytes, void *dst, uint32_t elem_size);
+/* RLE/bit-packed hybrid dictionary indices (parquet encoding.md): ents =
+ * nents x 5 u64 [kind, dst_off, count, a, b]; kind 0 = RLE fill run
+ * (a = the index value), kind 1 = bit-packed slice (a = ABSOLUTE BIT
+ * offset into src_bytes of the slice's first value, b = bit width;
+ * value i is the b bits at bit a + i*b). Emits u32 indices. src_bytes
+ * must have >= 8 bytes of slack after the last referenced byte. */
+int qk_pq_rle_expand(void *stream, uint64_t nents, const uint64_t *ents,
+                     const uint8_t *src_bytes, uint32_t *out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1986,6 +1986,80 @@ extern "C" int qk_comm_destroy(void *comm) {
   QK_NCCL("qk_comm_destroy", ncclCommDestroy((ncclComm_t)comm));
   return 0;
 }
+// ---- GPU Parquet page decode (include/quokka_amd.h for the contract) --
+// Host-built tile tables give each workgroup one bounded slice; all
+// value bytes stay in HBM (the host touched only Thrift headers and RLE
+// run descriptors).
+template <typename T>
+__device__ inline void pq_copy_vals(const uint8_t *__restrict__ s,
+                                    uint8_t *__restrict__ d, uint64_t cnt) {
+  for (uint64_t i = threadIdx.x; i < cnt; i += BLOCK) {
+    T v;
+    __builtin_memcpy(&v, s + i * sizeof(T), sizeof(T));  // src unaligned
+    *reinterpret_cast<T *>(d + i * sizeof(T)) = v;       // dst aligned
+  }
+}
+__global__ void __launch_bounds__(BLOCK) k_pq_plain_copy(
+    uint64_t ntiles, const uint64_t *__restrict__ tiles,
+    const uint8_t *__restrict__ src, uint8_t *__restrict__ dst,
+    uint32_t es) {
+  for (uint64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    uint64_t so = tiles[t * 3], eo = tiles[t * 3 + 1], cnt = tiles[t * 3 + 2];
+    const uint8_t *s = src + so;
+    uint8_t *d = dst + eo * es;
+    switch (es) {
+      case 8: pq_copy_vals<uint64_t>(s, d, cnt); break;
+      case 4: pq_copy_vals<uint32_t>(s, d, cnt); break;
+      case 2: pq_copy_vals<uint16_t>(s, d, cnt); break;
+      default: pq_copy_vals<uint8_t>(s, d, cnt); break;
+    }
+  }
+}
+extern "C" int qk_pq_plain_copy(void *stream, uint64_t ntiles,
+                                const uint64_t *tiles, const void *src_bytes,
+                                void *dst, uint32_t elem_size) {
+  if (!ntiles) return 0;
+  if (elem_size != 1 && elem_size != 2 && elem_size != 4 && elem_size != 8)
+    return qk_fail("qk_pq_plain_copy.elem_size", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, ntiles);
+  hipLaunchKernelGGL(k_pq_plain_copy, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, ntiles, tiles,
+                     (const uint8_t *)src_bytes, (uint8_t *)dst, elem_size);
+  QK_TRY("qk_pq_plain_copy", hipGetLastError());
+  return 0;
+}
+__global__ void __launch_bounds__(BLOCK) k_pq_rle_expand(
+    uint64_t nents, const uint64_t *__restrict__ ents,
+    const uint8_t *__restrict__ src, uint32_t *__restrict__ out) {
+  for (uint64_t e = blockIdx.x; e < nents; e += gridDim.x) {
+    const uint64_t *E = ents + e * 5;
+    uint64_t kind = E[0], dst = E[1], cnt = E[2], a = E[3], b = E[4];
+    if (kind == 0) {
+      uint32_t v = (uint32_t)a;
+      for (uint64_t i = threadIdx.x; i < cnt; i += BLOCK) out[dst + i] = v;
+    } else {
+      uint32_t bw = (uint32_t)b;                 // <= 32 (parquet indices)
+      uint64_t mask = (1ull << bw) - 1;
+      for (uint64_t i = threadIdx.x; i < cnt; i += BLOCK) {
+        uint64_t bit = a + i * (uint64_t)bw;
+        uint64_t w;
+        __builtin_memcpy(&w, src + (bit >> 3), 8);  // unaligned, 8B slack
+        out[dst + i] = (uint32_t)((w >> (bit & 7)) & mask);
+      }
+    }
+  }
+}
+extern "C" int qk_pq_rle_expand(void *stream, uint64_t nents,
+                                const uint64_t *ents, const uint8_t *src_bytes,
+                                uint32_t *out) {
+  if (!nents) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, nents);
+  hipLaunchKernelGGL(k_pq_rle_expand, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, nents, ents, src_bytes, out);
+  QK_TRY("qk_pq_rle_expand", hipGetLastError());
+  return 0;
+}
+
 // RCCL 2.27 p2p SILENTLY DELIVERS ONLY THE FIRST HALF of a send whose
 // byte count exceeds 1 GiB (measured on MI355X: exact at 2^30 B, exactly
 // half the rows at every size above — scripts/diag_exchange.py). Two

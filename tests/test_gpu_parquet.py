@@ -1,0 +1,129 @@
+"""GPU parity for the Parquet decode kernels: the same file shapes the
+CPU plan tests pin (tests/test_parquet_cpu.py) decoded by the real
+qk_pq_plain_copy / qk_pq_rle_expand kernels via parquet_gpu.read_table,
+compared against pyarrow's decode (the reference's reader,
+pyquokka/dataset.py). Ends with Parquet -> HBM -> Q1 vs the oracle."""
+import io
+
+import numpy as np
+import pytest
+
+pa = pytest.importorskip("pyarrow")
+import pyarrow.parquet as pq  # noqa: E402
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def gpu():
+    from quokka_amd import shim
+    shim.init(0)
+    return shim
+
+
+def write(table, **kw):
+    buf = io.BytesIO()
+    kw.setdefault("compression", "NONE")
+    kw.setdefault("data_page_version", "1.0")
+    pq.write_table(table, buf, **kw)
+    return buf.getvalue()
+
+
+def roundtrip(gpu, table, **kw):
+    from quokka_amd import parquet_gpu as P
+    raw = write(table, **kw)
+    cols = P.read_table(raw)
+    try:
+        for name in table.schema.names:
+            want = table.column(name).to_numpy()
+            got = cols[name]
+            if isinstance(got, tuple):
+                codes, cb = got
+                vals = np.asarray(cb)[codes.to_numpy(codes.n)]
+                np.testing.assert_array_equal(vals, want, err_msg=name)
+            else:
+                np.testing.assert_array_equal(got.to_numpy(got.n), want,
+                                              err_msg=name)
+    finally:
+        for c in cols.values():
+            (c[0] if isinstance(c, tuple) else c).free()
+
+
+def test_plain_mixed_types(gpu):
+    rng = np.random.default_rng(0)
+    n = 300_000
+    roundtrip(gpu, pa.table({
+        "i64": rng.integers(-1 << 40, 1 << 40, n),
+        "f64": rng.random(n),
+        "i32": rng.integers(-1 << 30, 1 << 30, n).astype(np.int32)}),
+        use_dictionary=False)
+
+
+def test_dict_small_and_wide_bitwidths(gpu):
+    rng = np.random.default_rng(1)
+    n = 400_000
+    roundtrip(gpu, pa.table({
+        "narrow": rng.integers(0, 7, n),          # bw 3
+        "mid": rng.integers(0, 300, n),           # bw 9
+        "wide": rng.integers(0, 20_000, n)}),     # bw 15
+        use_dictionary=True)
+
+
+def test_dict_fallback_mid_chunk(gpu):
+    rng = np.random.default_rng(2)
+    roundtrip(gpu, pa.table({"k": rng.integers(0, 1 << 30, 150_000)}),
+              use_dictionary=True, dictionary_pagesize_limit=4096)
+
+
+def test_multi_row_group_and_v2(gpu):
+    rng = np.random.default_rng(3)
+    t = pa.table({"a": rng.integers(0, 1 << 40, 250_000),
+                  "k": rng.integers(0, 37, 250_000)})
+    roundtrip(gpu, t, use_dictionary=["k"], row_group_size=40_000)
+    roundtrip(gpu, t, use_dictionary=["k"], data_page_version="2.0")
+
+
+def test_string_dict_codes(gpu):
+    rng = np.random.default_rng(4)
+    vals = np.array(["BUILDING", "AUTOMOBILE", "MACHINERY", "HOUSEHOLD",
+                     "FURNITURE"])
+    s = vals[rng.integers(0, 5, 200_000)]
+    roundtrip(gpu, pa.table({"seg": s}), use_dictionary=True)
+
+
+def test_parquet_to_q1_vs_oracle(gpu):
+    """End-to-end scan row: TPC-H lineitem written as
+    Parquet, decoded by the GPU kernels straight into HBM columns, fused
+    Q1 on them, parity vs the CPU oracle on the same host data."""
+    from quokka_amd import parquet_gpu as P, queries as DQ
+    from oracle import tpch_gen as G, queries as OQ
+    li = G.gen_lineitem(0.01, seed=11)
+    table = pa.table({
+        "l_shipdate": li["l_shipdate"],
+        "l_quantity": li["l_quantity"],
+        "l_extendedprice": li["l_extendedprice"],
+        "l_discount": li["l_discount"],
+        "l_tax": li["l_tax"],
+        "l_returnflag": li["l_returnflag"].astype(np.int32),
+        "l_linestatus": li["l_linestatus"].astype(np.int32)})
+    raw = write(table, use_dictionary=False)
+    cols = P.read_table(raw)
+    # Q1 wants u8 flag/status codes; decode returns i32 — narrow on host
+    # copy for the two tiny dict-code columns (u8 staging is the staging
+    # module's job; this test pins the decode+compute path)
+    from quokka_amd import staging
+    dev = {k: cols[k] for k in ("l_shipdate", "l_quantity",
+                                "l_extendedprice", "l_discount", "l_tax")}
+    for k in ("l_returnflag", "l_linestatus"):
+        narrowed = cols[k].to_numpy(cols[k].n).astype(np.uint8)
+        dev[k] = gpu.DevColumn.from_numpy(narrowed)
+        cols[k].free()
+    got = DQ.q1(dev)
+    want = OQ.q1(li)
+    assert list(got["l_returnflag"]) == list(want["l_returnflag"])
+    np.testing.assert_array_equal(got["count_order"], want["count_order"])
+    for c in ("sum_qty", "sum_base_price", "sum_disc_price", "sum_charge",
+              "avg_qty", "avg_price", "avg_disc"):
+        np.testing.assert_allclose(got[c], want[c], rtol=1e-9)
+    for c in dev.values():
+        c.free()
