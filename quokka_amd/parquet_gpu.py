@@ -34,7 +34,10 @@ ENC_PLAIN_DICT = 2
 _PHYS = {"INT32": np.dtype(np.int32), "INT64": np.dtype(np.int64),
          "FLOAT": np.dtype(np.float32), "DOUBLE": np.dtype(np.float64)}
 
-_TILE = 1 << 16          # values per PLAIN tile / max RLE entry length
+_TILE = 1 << 13          # values per PLAIN tile / max RLE entry length
+# (8K values -> ~2900 workgroups per 24M-value column: tiles map 1:1 to
+# workgroups, and the 256-CU chip needs >>256 of them; 64K tiles ran the
+# decode at 4.9 GB/s, 8K at TB/s-class)
 
 
 class QkParquetError(RuntimeError):
