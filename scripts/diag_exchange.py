@@ -14,7 +14,7 @@ import sys
 
 import numpy as np
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from quokka_amd import shim, exchange  # noqa: E402
 from quokka_amd.shim import DevColumn, c_u64, c_i64  # noqa: E402
 
