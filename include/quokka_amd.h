@@ -380,14 +380,15 @@ int qk_allreduce_f64(void *stream, void *comm, double *buf, uint64_t n);
  * unaligned) to dst + dst_elem_off. */
 int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
                      const void *src_bytes, void *dst, uint32_t elem_size);
-/* RLE/bit-packed hybrid dictionary indices (parquet encoding.md): ents =
- * nents x 5 u64 [kind, dst_off, count, a, b]; kind 0 = RLE fill run
- * (a = the index value), kind 1 = bit-packed slice (a = ABSOLUTE BIT
- * offset into src_bytes of the slice's first value, b = bit width;
- * value i is the b bits at bit a + i*b). Emits u32 indices. src_bytes
- * must have >= 8 bytes of slack after the last referenced byte. */
-int qk_pq_rle_expand(void *stream, uint64_t nents, const uint64_t *ents,
-                     const uint8_t *src_bytes, uint32_t *out);
+/* RLE/bit-packed hybrid dictionary indices (parquet encoding.md),
+ * parsed AND expanded on-device, one wave per page (low-cardinality
+ * columns emit millions of tiny runs — host-side run parsing cost
+ * seconds, so the host ships one descriptor per page): ents = npages x
+ * 5 u64 [src_byte_off (first run header, after the bit-width byte),
+ * src_byte_end, dst_off, count, bit_width]. Emits u32 indices; bw 0 =>
+ * all zeros. src_bytes needs >= 8 bytes of slack after the end. */
+int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
+                    const uint8_t *src_bytes, uint32_t *out);
 
 #ifdef __cplusplus
 }

@@ -128,7 +128,7 @@ class _Chunk:
         if col.dictionary_page_offset is not None:
             start = min(start, col.dictionary_page_offset)
         self.plain_tiles = []
-        self.rle_ents = []
+        self.rle_pages = []
         self.dict_vals = None          # np array (fixed) or list (strings)
         self.n = col.num_values
         row = dst0
@@ -164,11 +164,12 @@ class _Chunk:
                 bw = buf[data]
                 if bw > 32:
                     raise QkParquetError("index bit width %d" % bw)
-                if bw == 0:
-                    self.rle_ents.append((0, row, p.num_values, 0, 0))
-                else:
-                    _walk_rle(buf, data + 1, pend, bw, p.num_values, row,
-                              self.rle_ents)
+                # one descriptor per PAGE; the run stream is parsed on
+                # the GPU (qk_pq_rle_pages) — low-cardinality columns
+                # emit millions of tiny runs and walking them in host
+                # Python measured seconds per file
+                self.rle_pages.append((data + 1, pend, row,
+                                       p.num_values, bw))
             else:
                 raise QkParquetError("data page encoding %d" % p.encoding)
             row += p.num_values
@@ -258,7 +259,7 @@ def read_table(source, columns=None):
                           ctypes.c_uint32(dt.itemsize))
                 dtile.free()
             for ch in chunks:
-                if ch.rle_ents:
+                if ch.rle_pages:
                     _decode_dict_fixed(shim, dev_file, ch, col_out)
             shim.call("qk_stream_sync", None)
             out[names[ci]] = col_out
@@ -267,27 +268,27 @@ def read_table(source, columns=None):
         dev_file.free()
 
 
-def _expand_indices(shim, dev_file, rle_ents):
-    """Upload RLE entries (rebased to their first output row) and expand
-    to a device u32 index buffer. The covered rows must be one contiguous
-    span — pyarrow's dictionary->PLAIN fallback writes all dictionary
-    pages before any PLAIN page, so a chunk's RLE entries are always a
-    prefix; anything else would leave holes that a full-span gather would
-    fill with garbage, so it raises instead."""
-    import ctypes
+def _expand_indices(shim, dev_file, rle_pages):
+    """Upload per-page descriptors (rebased to their first output row)
+    and parse+expand the RLE/bit-packed index streams on the GPU. The
+    covered rows must be one contiguous span — pyarrow's
+    dictionary->PLAIN fallback writes all dictionary pages before any
+    PLAIN page, so a chunk's dictionary pages are always a prefix;
+    anything else would leave holes that a full-span gather would fill
+    with garbage, so it raises instead."""
     from .shim import DevBuffer, DevColumn, c_u64, c_vp
-    ents = np.asarray(rle_ents, dtype=np.uint64)
-    base = int(ents[:, 1].min())
-    nv = int(ents[:, 2].sum())
-    if int((ents[:, 1] + ents[:, 2]).max()) - base != nv:
+    pages = np.asarray(rle_pages, dtype=np.uint64)
+    base = int(pages[:, 2].min())
+    nv = int(pages[:, 3].sum())
+    if int((pages[:, 2] + pages[:, 3]).max()) - base != nv:
         raise QkParquetError("non-contiguous dictionary-page coverage")
-    ents_local = ents.copy()
-    ents_local[:, 1] -= base
-    dents = DevBuffer(ents_local.nbytes)
-    shim.call("qk_h2d", dents.ptr, ents_local.ctypes.data_as(c_vp),
-              c_u64(ents_local.nbytes))
+    pages_local = pages.copy()
+    pages_local[:, 2] -= base
+    dents = DevBuffer(pages_local.nbytes)
+    shim.call("qk_h2d", dents.ptr, pages_local.ctypes.data_as(c_vp),
+              c_u64(pages_local.nbytes))
     idx = DevColumn(np.uint32, max(1, nv))
-    shim.call("qk_pq_rle_expand", None, c_u64(len(ents)), dents.ptr,
+    shim.call("qk_pq_rle_pages", None, c_u64(len(pages)), dents.ptr,
               dev_file.ptr, idx.ptr)
     return base, nv, dents, idx
 
@@ -297,7 +298,7 @@ def _decode_dict_fixed(shim, dev_file, ch, col_out):
     from .shim import DevBuffer, DevColumn, c_u64, c_vp
     if ch.dict_vals is None:
         raise QkParquetError("RLE_DICTIONARY page without dictionary page")
-    base, nv, dents, idx = _expand_indices(shim, dev_file, ch.rle_ents)
+    base, nv, dents, idx = _expand_indices(shim, dev_file, ch.rle_pages)
     dvals = shim.DevColumn.from_numpy(ch.dict_vals)
     es = ch.dtype.itemsize
     gather = {8: "qk_gather_i64", 4: "qk_gather_i32"}[es]
@@ -313,7 +314,7 @@ def _decode_ba(shim, dev_file, ch, glob, codes):
     from .shim import DevBuffer, DevColumn, c_u64, c_vp
     if ch.dict_vals is None:
         raise QkParquetError("BYTE_ARRAY without dictionary encoding")
-    base, nv, dents, idx = _expand_indices(shim, dev_file, ch.rle_ents)
+    base, nv, dents, idx = _expand_indices(shim, dev_file, ch.rle_pages)
     remap = np.asarray([glob[v] for v in ch.dict_vals], dtype=np.int32)
     dremap = DevColumn.from_numpy(remap if len(remap) else
                                   np.zeros(1, np.int32))

@@ -46,7 +46,7 @@ def sim_column(raw, ci):
                 glob.setdefault(v, len(glob))
         out = np.empty(total, dtype=np.uint32)
         for ch in chunks:
-            idx, base = sim_rle(raw, ch.rle_ents)
+            idx, base = sim_rle_pages(raw, ch.rle_pages)
             remap = np.asarray([glob[v] for v in ch.dict_vals],
                                dtype=np.uint32)
             out[base:base + len(idx)] = remap[idx]
@@ -57,10 +57,23 @@ def sim_column(raw, ci):
         for src, dst, cnt in ch.plain_tiles:
             out[dst:dst + cnt] = np.frombuffer(raw, dtype=dt, count=cnt,
                                                offset=src)
-        if ch.rle_ents:
-            idx, base = sim_rle(raw, ch.rle_ents)
+        if ch.rle_pages:
+            idx, base = sim_rle_pages(raw, ch.rle_pages)
             out[base:base + len(idx)] = ch.dict_vals[idx]
     return out
+
+
+def sim_rle_pages(raw, pages):
+    """Walk each page's run stream with the (test-retained) host walker,
+    then simulate the expansion — pins the per-page descriptors the GPU
+    kernel consumes against run-level semantics."""
+    ents = []
+    for src, end, dst, cnt, bw in pages:
+        if bw == 0:
+            ents.append((0, dst, cnt, 0, 0))
+        else:
+            P._walk_rle(raw, src, end, bw, cnt, dst, ents)
+    return sim_rle(raw, ents)
 
 
 def sim_rle(raw, ents):
