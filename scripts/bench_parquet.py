@@ -106,7 +106,7 @@ def main():
                           ctypes.c_uint32(dt.itemsize))
                 dtile.free()
             for ch in chs:
-                if ch.rle_ents:
+                if ch.rle_pages:
                     P._decode_dict_fixed(shim, dev_file, ch, col_out)
             outs.append(col_out)
         shim.call("qk_stream_sync", None)
