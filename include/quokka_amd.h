@@ -42,6 +42,15 @@ int qk_hmalloc_impl(uint64_t nbytes, void **hptr); /* pinned host memory */
 int qk_dfree(void *dptr);
 int qk_h2d(void *dst_dev, const void *src_host, uint64_t nbytes);
 int qk_d2h(void *dst_host, const void *src_dev, uint64_t nbytes);
+/* Async variants on a stream; src/dst host memory must be PINNED
+ * (qk_hmalloc_impl) or the copy silently degrades to staged+sync. Used
+ * by the double-buffered staging bounce (shim._PinnedBounce): host
+ * memmove of chunk k+1 overlaps the DMA of chunk k on alternating
+ * streams. */
+int qk_h2d_async(void *stream, void *dst_dev, const void *src_host,
+                 uint64_t nbytes);
+int qk_d2h_async(void *stream, void *dst_host, const void *src_dev,
+                 uint64_t nbytes);
 int qk_dmemset(void *dst_dev, int value, uint64_t nbytes);
 int qk_fill_i64(void *stream, int64_t *dst_dev, int64_t value, uint64_t n);
 

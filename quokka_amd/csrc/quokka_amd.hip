@@ -71,6 +71,18 @@ extern "C" int qk_d2h(void *dst, const void *src, uint64_t n) {
   QK_TRY("qk_d2h", hipMemcpy(dst, src, n, hipMemcpyDeviceToHost));
   return 0;
 }
+extern "C" int qk_h2d_async(void *stream, void *dst, const void *src,
+                            uint64_t n) {
+  QK_TRY("qk_h2d_async", hipMemcpyAsync(dst, src, n, hipMemcpyHostToDevice,
+                                        (hipStream_t)stream));
+  return 0;
+}
+extern "C" int qk_d2h_async(void *stream, void *dst, const void *src,
+                            uint64_t n) {
+  QK_TRY("qk_d2h_async", hipMemcpyAsync(dst, src, n, hipMemcpyDeviceToHost,
+                                        (hipStream_t)stream));
+  return 0;
+}
 extern "C" int qk_dmemset(void *dst, int value, uint64_t n) {
   QK_TRY("qk_dmemset", hipMemset(dst, value, n));
   return 0;
