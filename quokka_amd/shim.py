@@ -158,9 +158,30 @@ class _PinnedBounce:
 
     CHUNK = 64 << 20
 
+    MEMMOVE_THREADS = 4
+
     def __init__(self):
         self.bufs = None
         self.streams = None
+        self.pool = None
+
+    def _mm(self, dst, src, n):
+        """Parallel memmove: ctypes.memmove releases the GIL, and one
+        core moves only ~10 GB/s — slice across a small thread pool."""
+        if n < (8 << 20):
+            ctypes.memmove(dst, src, n)
+            return
+        if self.pool is None:
+            from concurrent.futures import ThreadPoolExecutor
+            self.pool = ThreadPoolExecutor(self.MEMMOVE_THREADS)
+        t = self.MEMMOVE_THREADS
+        step = (n + t - 1) // t
+        futs = [self.pool.submit(ctypes.memmove, dst + i * step,
+                                 src + i * step,
+                                 min(step, n - i * step))
+                for i in range(t) if i * step < n]
+        for f in futs:
+            f.result()
 
     def _ensure(self):
         if self.bufs is None:
@@ -198,7 +219,7 @@ class _PinnedBounce:
             i = k & 1
             # wait for the DMA issued from this buffer two chunks ago
             call("qk_stream_sync", self.streams[i])
-            ctypes.memmove(self.bufs[i], src + off, m)
+            self._mm(self.bufs[i].value, src + off, m)
             call("qk_h2d_async", self.streams[i],
                  c_vp(dst_dev.value + off), self.bufs[i], c_u64(m))
         call("qk_stream_sync", self.streams[0])
@@ -225,7 +246,7 @@ class _PinnedBounce:
                      c_u64(sizes[k + 1]))
             i = k & 1
             call("qk_stream_sync", self.streams[i])
-            ctypes.memmove(dst + k * self.CHUNK, self.bufs[i], sizes[k])
+            self._mm(dst + k * self.CHUNK, self.bufs[i].value, sizes[k])
 
 
 _bounce = _PinnedBounce()
