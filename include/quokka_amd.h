@@ -393,9 +393,12 @@ int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
  * parsed AND expanded on-device, one wave per page (low-cardinality
  * columns emit millions of tiny runs — host-side run parsing cost
  * seconds, so the host ships one descriptor per page): ents = npages x
- * 5 u64 [src_byte_off (first run header, after the bit-width byte),
- * src_byte_end, dst_off, count, bit_width]. Emits u32 indices; bw 0 =>
- * all zeros. src_bytes needs >= 8 bytes of slack after the end. */
+ * 6 u64 [src_byte_off (first run header, after the bit-width byte),
+ * src_byte_end, dst_off, count, bit_width, idx_off]. Emits u32
+ * (index + idx_off) — idx_off rebases each chunk's dictionary into a
+ * column-global concatenated dictionary so a whole multi-row-group
+ * column expands in ONE launch. bw 0 => all idx_off. src_bytes needs
+ * >= 8 bytes of slack after the end. */
 int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
                     const uint8_t *src_bytes, uint32_t *out);
 

@@ -2050,12 +2050,13 @@ __global__ void __launch_bounds__(WAVE) k_pq_rle_pages(
     uint64_t npages, const uint64_t *__restrict__ ents,
     const uint8_t *__restrict__ src, uint32_t *__restrict__ out) {
   for (uint64_t pg = blockIdx.x; pg < npages; pg += gridDim.x) {
-    const uint64_t *E = ents + pg * 5;
+    const uint64_t *E = ents + pg * 6;
     uint64_t pos = E[0], end = E[1], dst = E[2], remaining = E[3];
     uint32_t bw = (uint32_t)E[4];
+    uint32_t idx_off = (uint32_t)E[5];
     if (bw == 0) {                       // all indices are 0, no stream
       for (uint64_t i = threadIdx.x; i < remaining; i += WAVE)
-        out[dst + i] = 0;
+        out[dst + i] = idx_off;
       continue;
     }
     uint64_t mask = (1ull << bw) - 1;
@@ -2086,14 +2087,14 @@ __global__ void __launch_bounds__(WAVE) k_pq_rle_pages(
           uint64_t bit = bitbase + i * (uint64_t)bw;
           uint64_t w;
           __builtin_memcpy(&w, src + (bit >> 3), 8);  // unaligned, slack
-          out[dst + i] = (uint32_t)((w >> (bit & 7)) & mask);
+          out[dst + i] = (uint32_t)((w >> (bit & 7)) & mask) + idx_off;
         }
         npos += ngroups * bw;
       } else {
         val = (uint64_t)__shfl((long long)val, 0);
         n = qk_min_u64(h >> 1, remaining);
         for (uint64_t i = threadIdx.x; i < n; i += WAVE)
-          out[dst + i] = (uint32_t)val;
+          out[dst + i] = (uint32_t)val + idx_off;
       }
       dst += n;
       remaining -= n;

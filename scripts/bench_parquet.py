@@ -83,32 +83,8 @@ def main():
     def kernels_once():
         outs = []
         for name, total, chs in plans:
-            if chs[0].is_ba:
-                glob = {}
-                for ch in chs:
-                    for v in ch.dict_vals or []:
-                        glob.setdefault(v, len(glob))
-                codes = DevColumn(np.uint32, max(1, total))
-                for ch in chs:
-                    P._decode_ba(shim, dev_file, ch, glob, codes)
-                outs.append(codes)
-                continue
-            dt = chs[0].dtype
-            col_out = DevColumn(dt, max(1, total))
-            tiles = np.asarray([t for ch in chs for t in ch.plain_tiles],
-                               dtype=np.uint64)
-            if len(tiles):
-                dtile = DevBuffer(tiles.nbytes)
-                shim.call("qk_h2d", dtile.ptr,
-                          tiles.ctypes.data_as(c_vp), c_u64(tiles.nbytes))
-                shim.call("qk_pq_plain_copy", None, c_u64(len(tiles)),
-                          dtile.ptr, dev_file.ptr, col_out.ptr,
-                          ctypes.c_uint32(dt.itemsize))
-                dtile.free()
-            for ch in chs:
-                if ch.rle_pages:
-                    P._decode_dict_fixed(shim, dev_file, ch, col_out)
-            outs.append(col_out)
+            c = P._decode_column(shim, dev_file, chs, total)
+            outs.append(c[0] if isinstance(c, tuple) else c)
         shim.call("qk_stream_sync", None)
         for c in outs:
             c.free()
