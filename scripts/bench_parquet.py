@@ -76,8 +76,7 @@ def main():
             row += ch.n
             chs.append(ch)
         plans.append((name, row, chs))
-    from quokka_amd.shim import DevBuffer, DevColumn, c_u64, c_vp, Timer
-    import ctypes
+    from quokka_amd.shim import Timer
     timer = Timer()
 
     def kernels_once():

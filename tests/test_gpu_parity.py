@@ -352,7 +352,7 @@ def test_gen_lineitem_device_vs_oracle_shape(gpu):
     (same filter selectivities within tolerance) and the Q1 kernel over it
     matches the oracle partials computed on the d2h copy of the SAME data."""
     from quokka_amd import shim, ops, queries as DQ
-    from quokka_amd.shim import DevColumn, DevBuffer, c_u64
+    from quokka_amd.shim import DevColumn, c_u64
     n = 1_000_000
     cols = {name: DevColumn(dt, n) for name, dt in [
         ("l_quantity", np.float64), ("l_extendedprice", np.float64),

@@ -5,7 +5,6 @@ numpy evaluation of the predicate AST on random column data."""
 import datetime
 
 import numpy as np
-import pytest
 from hypothesis import given, settings, strategies as st
 
 from quokka_amd import jit
