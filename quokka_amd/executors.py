@@ -590,18 +590,73 @@ class GPUTopKExecutor(Executor):
 
 
 def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
-                     predicate=None, string_dicts=None, projection=None):
+                     predicate=None, string_dicts=None, projection=None,
+                     batch_agg=None):
     """GPU partition function mirroring the reference's full partition_fn
     (core.py:152-195): optional PREDICATE (filter_sql grammar, JIT-fused
-    on device via quokka_amd.jit) -> hash partition (int key -> key % N,
-    bit-exact with quokka_runtime.py:222) -> optional column projection
-    (sorted order, core.py:191-193).
+    on device via quokka_amd.jit) -> optional map-side PARTIAL AGGREGATE
+    (`batch_agg`, the reference's folded batch_funcs, core.py:173-176 +
+    df.py:1354-1394) -> hash partition (int key -> key % N, bit-exact
+    with quokka_runtime.py:222) -> optional column projection (sorted
+    order, core.py:191-193).
+
+    batch_agg = (group_keys, aggs) with group_keys a list of
+    (u8_code_column, cardinality) and aggs the two-phase PARTIAL forms
+    ('SUM(expr) as x' / 'COUNT(*) as n', sql_utils.py:299-413). The
+    predicate and the aggregate run as ONE fused device pass
+    (jit.JitAggregate — the Q1-shaped kernel), and what gets partitioned
+    (by group id % N) is the tiny partial table, not rows: the map-side
+    fusion that shrinks shuffle bytes from O(rows) to O(groups).
     `data`: pyarrow Table; returns dict target_channel -> pyarrow Table."""
     import pyarrow as pa
     ops, shim, staging = _lazy_gpu()
     host_cols = {c: staging.column_to_numpy(
         data.column(c), (string_dicts or {}).get(c))
         for c in data.column_names}
+
+    if batch_agg is not None:
+        from . import jit
+        group_keys, aggs = batch_agg
+        # COUNT(*) identifies which groups were actually observed (the
+        # reference's partial tables contain only observed groups);
+        # append one internally if the caller didn't ask for it
+        have_count = any(a.lower().replace(" ", "").startswith("count(")
+                         for a in aggs)
+        run_aggs = list(aggs) + ([] if have_count
+                                 else ["COUNT(*) as __qk_presence"])
+        dcols = {c: shim.DevColumn.from_numpy(v)
+                 for c, v in host_cols.items()}
+        schema = {c: v.dtype for c, v in dcols.items()}
+        agg = jit.JitAggregate(schema, group_keys, run_aggs, predicate,
+                               string_dicts)
+        acc = agg.make_acc()
+        agg.run(dcols, acc)
+        partials = agg.read(acc)               # (ngroups, naggs)
+        acc.free()
+        agg.free()
+        for c in dcols.values():
+            c.free()
+        gid = np.arange(partials.shape[0])
+        count_col = (len(aggs) if not have_count else
+                     next(i for i, a in enumerate(aggs)
+                          if a.lower().replace(" ", "")
+                          .startswith("count(")))
+        observed = partials[:, count_col] > 0
+        cols_out = {}
+        rem = gid
+        for name, card in reversed(list(group_keys)):
+            cols_out[name] = (rem % card).astype(np.uint8)
+            rem = rem // card
+        for j, aname in enumerate(agg.agg_names):
+            if aname != "__qk_presence":
+                cols_out[aname] = partials[:, j]
+        parts = gid % num_target_channels      # group id is the int key
+        out = {}
+        for p in range(num_target_channels):
+            m = (parts == p) & observed
+            if m.any():
+                out[p] = pa.table({c: v[m] for c, v in cols_out.items()})
+        return out
 
     if predicate is not None:
         from . import jit

@@ -814,6 +814,45 @@ def test_gpu_partition_fn_plugin_api(gpu):
     assert total == len(keys)
 
 
+def test_gpu_partition_fn_batch_agg_fusion(gpu):
+    """Map-side partial-agg fusion (the reference
+    folds batch partial aggs into partition_fn, core.py:173-176 +
+    df.py:1354-1394): predicate + group-by partial aggregate run as one
+    fused device pass and the PARTIAL table (one row per observed
+    group), not rows, is what gets partitioned — by group id % N."""
+    import pyarrow as pa
+    from quokka_amd import gpu_partition_fn
+    rng = np.random.default_rng(63)
+    n = 200_000
+    g1 = rng.integers(0, 3, n).astype(np.uint8)
+    g2 = rng.integers(0, 2, n).astype(np.uint8)
+    x = np.round(rng.random(n), 6)
+    d = rng.integers(0, 1000, n).astype(np.int32)
+    t = pa.table({"flag": g1, "status": g2, "x": x, "d": d})
+    out = gpu_partition_fn(
+        t, source_channel=0, num_target_channels=4, key=None,
+        predicate="d < 700",
+        batch_agg=([("flag", 3), ("status", 2)],
+                   ["SUM(x) as sx", "COUNT(*) as cnt"]))
+    mask = d < 700
+    seen_groups = 0
+    for ch, tbl in out.items():
+        gf = np.asarray(tbl.column("flag"))
+        gs = np.asarray(tbl.column("status"))
+        sx = np.asarray(tbl.column("sx"))
+        cnt = np.asarray(tbl.column("cnt"))
+        for i in range(len(tbl)):
+            gid = int(gf[i]) * 2 + int(gs[i])
+            assert gid % 4 == ch                     # int-key bucketing
+            m = mask & (g1 == gf[i]) & (g2 == gs[i])
+            np.testing.assert_allclose(sx[i], x[m].sum(), rtol=1e-9)
+            assert cnt[i] == m.sum()
+            seen_groups += 1
+    assert seen_groups == 6                          # all groups observed
+    # shuffle payload is O(groups), not O(rows)
+    assert sum(len(tbl) for tbl in out.values()) == 6
+
+
 def test_executor_distinct_and_broadcast_and_count(gpu):
     """The remaining hot-path executor family (sql_executors.py:69
     CountExecutor, :275 BroadcastJoinExecutor, :517 DistinctExecutor)
