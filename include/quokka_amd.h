@@ -402,6 +402,29 @@ int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
 int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
                     const uint8_t *src_bytes, uint32_t *out);
 
+/* ---- GPU CSV parse ---------------------------------------------------- *
+ * The reference's CSV scan splits files into byte ranges and decodes on
+ * the CPU with polars.read_csv (unordered_readers.py:273-442, :438).
+ * Here the raw bytes live in HBM: qk_csv_newlines builds the ORDERED
+ * newline index (u64 positions; count written to out_count_dev), then
+ * qk_csv_parse decodes one row per thread. coltypes: 0=i64, 1=f64
+ * (decimal, <=15 significant digits — parsed as exact-int / exact
+ * power-of-ten division, BIT-EXACT vs strtod; more digits = row error),
+ * 2=date32 ("YYYY-MM-DD" -> days since epoch), 3=u8 dictionary code
+ * (candidates as first-8-bytes-LE u64 + length, <=32 per column,
+ * unknown value = row error), 4=skip. Trailing '\r' stripped; trailing
+ * separators (dbgen .tbl) tolerated. err_row: device u64 initialized to
+ * UINT64_MAX; the LOWEST failing row index lands there. */
+#define QK_CSV_MAX_DICT 32
+int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
+                    const uint8_t *bytes, uint64_t *out_pos,
+                    uint64_t *out_count_dev);
+int qk_csv_parse(void *stream, uint64_t nrows, const uint8_t *bytes,
+                 uint64_t data_start, const uint64_t *nl_pos, uint8_t sep,
+                 int ncols, const int *coltypes, void *const *out_ptrs,
+                 const uint64_t *dict_cands, const uint8_t *dict_lens,
+                 const int *ncands, uint64_t *err_row);
+
 #ifdef __cplusplus
 }
 #endif

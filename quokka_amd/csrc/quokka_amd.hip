@@ -2113,6 +2113,237 @@ extern "C" int qk_pq_rle_pages(void *stream, uint64_t npages,
   return 0;
 }
 
+// ---- GPU CSV parse (include/quokka_amd.h for the contract) ------------
+// The reference's CSV scan reads byte ranges and hands them to
+// polars.read_csv on the CPU (unordered_readers.py:273-442, :438). Here
+// the raw bytes go to HBM and two kernels do the work: an ordered
+// newline index (same count/scan/scatter shape as the filter), then a
+// thread-per-row typed field parser.
+
+__global__ void __launch_bounds__(BLOCK) k_csv_nl_count(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
+    uint64_t *__restrict__ block_counts) {
+  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  uint32_t cnt = 0;
+  for (uint64_t r = lo + threadIdx.x; r < hi; r += BLOCK)
+    cnt += b[r] == '\n' ? 1u : 0u;
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w];
+    block_counts[blockIdx.x] = s;
+  }
+}
+
+__global__ void __launch_bounds__(BLOCK) k_csv_nl_scatter(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
+    const uint64_t *__restrict__ block_offsets, uint64_t *__restrict__ out) {
+  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  __shared__ uint64_t base;
+  __shared__ uint32_t wave_tot[BLOCK / WAVE];
+  if (threadIdx.x == 0) base = block_offsets[blockIdx.x];
+  __syncthreads();
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  for (uint64_t r0 = lo; r0 < hi; r0 += BLOCK) {
+    uint64_t r = r0 + threadIdx.x;
+    bool pass = r < hi && b[r] == '\n';
+    uint64_t mask = __ballot(pass);
+    uint32_t rank = __popcll(mask & ((1ULL << lane) - 1));
+    uint32_t wtot = __popcll(mask);
+    if (lane == 0) wave_tot[wid] = wtot;
+    __syncthreads();
+    uint32_t wbase = 0;
+    for (int w = 0; w < wid; w++) wbase += wave_tot[w];
+    if (pass) out[base + wbase + rank] = r;
+    uint32_t btot = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) btot += wave_tot[w];
+    __syncthreads();
+    if (threadIdx.x == 0) base += btot;
+    __syncthreads();
+  }
+}
+
+extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
+                               const uint8_t *bytes, uint64_t *out_pos,
+                               uint64_t *out_count_dev) {
+  if (n <= data_start) return 0;
+  uint64_t span = n - data_start;
+  uint64_t chunk = (span + MAX_BLOCKS - 1) / MAX_BLOCKS;
+  chunk = ((chunk + BLOCK - 1) / BLOCK) * BLOCK;
+  uint32_t blocks = (uint32_t)((span + chunk - 1) / chunk);
+  static __thread uint64_t *scratch = nullptr;
+  if (!scratch)
+    QK_TRY("qk_csv_newlines",
+           hipMalloc(&scratch, (MAX_BLOCKS + 1) * sizeof(uint64_t)));
+  hipLaunchKernelGGL(k_csv_nl_count, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     scratch);
+  hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, (uint64_t)blocks, scratch,
+                     out_count_dev);
+  hipLaunchKernelGGL(k_csv_nl_scatter, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     scratch, out_pos);
+  QK_TRY("qk_csv_newlines", hipGetLastError());
+  return 0;
+}
+
+// field parsers. Numeric parses are BIT-EXACT vs strtod for <= 15
+// significant digits: the digit string becomes an exact int64, the
+// scale 10^k is an exact double (k <= 22), and one correctly-rounded
+// division yields the correctly-rounded decimal value — the same double
+// strtod returns. More digits -> parse error (row reported), never a
+// silently different value.
+__device__ inline bool csv_i64(const uint8_t *b, uint64_t s, uint64_t e,
+                               int64_t *out) {
+  if (s >= e) return false;
+  bool neg = b[s] == '-';
+  if (neg || b[s] == '+') s++;
+  if (s >= e) return false;
+  int64_t v = 0;
+  int nd = 0;
+  for (; s < e; s++) {
+    uint8_t c = b[s] - '0';
+    if (c > 9) return false;
+    v = v * 10 + c;
+    if (++nd > 18) return false;
+  }
+  *out = neg ? -v : v;
+  return true;
+}
+__device__ inline bool csv_f64(const uint8_t *b, uint64_t s, uint64_t e,
+                               double *out) {
+  static const double P10[19] = {1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7,
+                                 1e8, 1e9, 1e10, 1e11, 1e12, 1e13, 1e14,
+                                 1e15, 1e16, 1e17, 1e18};
+  if (s >= e) return false;
+  bool neg = b[s] == '-';
+  if (neg || b[s] == '+') s++;
+  if (s >= e) return false;
+  int64_t m = 0;
+  int nd = 0, k = 0;
+  bool dot = false;
+  for (; s < e; s++) {
+    uint8_t c = b[s];
+    if (c == '.') {
+      if (dot) return false;
+      dot = true;
+      continue;
+    }
+    c -= '0';
+    if (c > 9) return false;
+    if (nd || c) nd++;                    // significant digits
+    m = m * 10 + c;
+    if (dot) k++;
+    if (nd > 15 || k > 18) return false;  // beyond bit-exact range
+  }
+  double v = (double)m / P10[k];
+  *out = neg ? -v : v;
+  return true;
+}
+__device__ inline bool csv_date32(const uint8_t *b, uint64_t s, uint64_t e,
+                                  int32_t *out) {
+  if (e - s != 10 || b[s + 4] != '-' || b[s + 7] != '-') return false;
+  int y = 0, mo = 0, d = 0;
+  for (int i = 0; i < 4; i++) {
+    uint8_t c = b[s + i] - '0';
+    if (c > 9) return false;
+    y = y * 10 + c;
+  }
+  for (int i = 5; i < 7; i++) {
+    uint8_t c = b[s + i] - '0';
+    if (c > 9) return false;
+    mo = mo * 10 + c;
+  }
+  for (int i = 8; i < 10; i++) {
+    uint8_t c = b[s + i] - '0';
+    if (c > 9) return false;
+    d = d * 10 + c;
+  }
+  if (mo < 1 || mo > 12 || d < 1 || d > 31) return false;
+  // Howard Hinnant days_from_civil (public-domain algorithm)
+  int yy = y - (mo <= 2);
+  int era = (yy >= 0 ? yy : yy - 399) / 400;
+  unsigned yoe = (unsigned)(yy - era * 400);
+  unsigned doy = (153u * (unsigned)(mo + (mo > 2 ? -3 : 9)) + 2u) / 5u +
+                 (unsigned)d - 1u;
+  unsigned doe = yoe * 365u + yoe / 4u - yoe / 100u + doy;
+  *out = (int32_t)(era * 146097 + (int)doe - 719468);
+  return true;
+}
+
+#define QK_CSV_MAX_DICT 32
+
+__global__ void __launch_bounds__(BLOCK) k_csv_parse(
+    uint64_t nrows, const uint8_t *__restrict__ b, uint64_t data_start,
+    const uint64_t *__restrict__ nl, uint8_t sep, int ncols,
+    const int *__restrict__ coltypes, void *const *__restrict__ outs,
+    const uint64_t *__restrict__ dict_cands,
+    const uint8_t *__restrict__ dict_lens, const int *__restrict__ ncands,
+    unsigned long long *__restrict__ err_row) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t r = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       r < nrows; r += stride) {
+    uint64_t pos = r == 0 ? data_start : nl[r - 1] + 1;
+    uint64_t row_end = nl[r];
+    if (row_end > pos && b[row_end - 1] == '\r') row_end--;
+    bool ok = true;
+    for (int c = 0; c < ncols && ok; c++) {
+      uint64_t fs = pos, fe = pos;
+      while (fe < row_end && b[fe] != sep) fe++;
+      pos = fe < row_end ? fe + 1 : row_end;
+      switch (coltypes[c]) {
+        case 0: ok = csv_i64(b, fs, fe, (int64_t *)outs[c] + r); break;
+        case 1: ok = csv_f64(b, fs, fe, (double *)outs[c] + r); break;
+        case 2: ok = csv_date32(b, fs, fe, (int32_t *)outs[c] + r); break;
+        case 3: {
+          uint64_t len = fe - fs;
+          uint64_t w = 0;
+          for (uint64_t i = 0; i < 8 && i < len; i++)
+            w |= (uint64_t)b[fs + i] << (8 * i);
+          int code = -1;
+          for (int j = 0; j < ncands[c]; j++)
+            if (dict_cands[c * QK_CSV_MAX_DICT + j] == w &&
+                dict_lens[c * QK_CSV_MAX_DICT + j] ==
+                    (uint8_t)(len > 255 ? 255 : len)) {
+              code = j;
+              break;
+            }
+          if (code < 0) { ok = false; break; }
+          ((uint8_t *)outs[c])[r] = (uint8_t)code;
+          break;
+        }
+        default: break;                  // 4 = skip
+      }
+    }
+    if (!ok) atomicMin(err_row, (unsigned long long)r);
+  }
+}
+
+extern "C" int qk_csv_parse(void *stream, uint64_t nrows,
+                            const uint8_t *bytes, uint64_t data_start,
+                            const uint64_t *nl_pos, uint8_t sep, int ncols,
+                            const int *coltypes, void *const *out_ptrs,
+                            const uint64_t *dict_cands,
+                            const uint8_t *dict_lens, const int *ncands,
+                            uint64_t *err_row) {
+  if (!nrows) return 0;
+  uint32_t blocks =
+      (uint32_t)qk_min_u64(MAX_BLOCKS, (nrows + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_csv_parse, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, nrows, bytes, data_start, nl_pos,
+                     sep, ncols, coltypes, out_ptrs, dict_cands, dict_lens,
+                     ncands, (unsigned long long *)err_row);
+  QK_TRY("qk_csv_parse", hipGetLastError());
+  return 0;
+}
+
 // RCCL 2.27 p2p SILENTLY DELIVERS ONLY THE FIRST HALF of a send whose
 // byte count exceeds 1 GiB (measured on MI355X: exact at 2^30 B, exactly
 // half the rows at every size above — scripts/diag_exchange.py). Two

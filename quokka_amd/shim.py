@@ -108,6 +108,11 @@ _SIGS = {
     "qk_iota_u32": [c_vp, c_u64, c_vp],
     "qk_bnot_u64": [c_vp, c_u64, c_vp],
     "qk_partition_hist": [c_vp, c_u64, c_vp, c_u32, c_vp],
+    "qk_csv_newlines": [c_vp, c_u64, c_u64, c_vp, c_vp, c_vp],
+    "qk_csv_parse": [c_vp, c_u64, c_vp, c_u64, c_vp, c_u8, ctypes.c_int,
+                     c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
+    "qk_pq_plain_copy": [c_vp, c_u64, c_vp, c_vp, c_vp, c_u32],
+    "qk_pq_rle_pages": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_partition_scatter": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp],
 }
 for name, argtypes in _SIGS.items():
