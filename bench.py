@@ -215,13 +215,18 @@ def cpu_baseline_q3(sample_rows, target_seconds=12.0):
     }
 
 
-def read_traffic():
+def read_traffic(sf=None):
     """Per-launch HBM bytes from the committed rocprofv3 PMC measurement
-    (profiles/traffic_q1.json), or None before one exists."""
+    (profiles/traffic_<q>.json), or None before one exists. The PMC run
+    was taken at the measured SF recorded in the file; at any other
+    --sf the figure would be wrong, so return None instead."""
     path = os.path.join(ROOT, "profiles", "traffic_%s.json" % QUERY)
     if os.path.exists(path):
         with open(path) as f:
             d = json.load(f)
+        measured_sf = d.get("sf", 100)
+        if sf is not None and abs(float(sf) - float(measured_sf)) > 1e-9:
+            return None
         return d.get("traffic_bytes_per_launch")
     return None
 
@@ -399,7 +404,7 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
                 "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s",
                 "frac": achieved_gbps / HBM_PEAK_GBPS,
-                "traffic": read_traffic(),
+                "traffic": read_traffic(args.sf),
             },
             "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
                              else cpu_baseline_q3(args.cpu_sample_rows)),
@@ -506,7 +511,7 @@ def main_q5(args, n, world, rank, dist, shim, DQ):
                 "bound": "hbm", "achieved": achieved_gbps,
                 "peak": HBM_PEAK_GBPS, "unit": "GB/s",
                 "frac": achieved_gbps / HBM_PEAK_GBPS,
-                "traffic": read_traffic(),
+                "traffic": read_traffic(args.sf),
             },
             "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
                              else cpu_baseline_q5(args.cpu_sample_rows)),
@@ -702,7 +707,7 @@ def main():
         total_rows = n * world * args.steps
         avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
         achieved_gbps = n * Q1_BYTES_PER_ROW / avg_kernel_s / 1e9
-        traffic = read_traffic()
+        traffic = read_traffic(args.sf)
         out = {
             "metric": "rows/s",
             "value": total_rows / elapsed,
