@@ -92,15 +92,19 @@ class GPUCSVReader:
         limit = end - base
         if base + skip >= end:      # first complete row starts past end
             return None, {}
-        if base + len(buf) < size:
+        if end >= size:
+            buf = buf[skip:]           # the final range: runs to EOF
+        else:
             tail = buf.find(b"\n", max(skip, limit - 1))
             if tail < 0:
-                raise csv_gpu.QkCsvError(
-                    "row longer than window=%d at %s:%d"
-                    % (self.window, f, end))
-            buf = buf[skip:tail + 1]
-        else:
-            buf = buf[skip:]
+                if base + len(buf) >= size:
+                    buf = buf[skip:]   # straddling last row, no final \n
+                else:
+                    raise csv_gpu.QkCsvError(
+                        "row longer than window=%d at %s:%d"
+                        % (self.window, f, end))
+            else:
+                buf = buf[skip:tail + 1]
         if not buf:
             return None, {}
         return None, csv_gpu.read_csv(buf, self.schema, sep=self.sep)
