@@ -20,3 +20,8 @@ from .executors import (Executor, GPUAggExecutor,  # noqa: F401
                         GPUDistinctExecutor, GPUSortExecutor,
                         GPUTopKExecutor,
                         gpu_partition_fn)
+
+# Scan-reader mirrors of the reference dataset API (readers.py imports
+# the decoders lazily inside execute(), so this too is .so-free at
+# registration time).
+from .readers import GPUParquetReader, GPUCSVReader  # noqa: F401
