@@ -339,10 +339,14 @@ int qk_jit_map_run(void *prog, void *stream, uint64_t n,
                    const void *const *col_ptrs, double *out_dev);
 /* group_expr: C int expression over v<i>; ngroups*naggs <= 64 (register
  * accumulators; use qk_groupby_* for high cardinality). out accumulates
- * ngroups*naggs f64 (row-major [group][agg]). */
+ * ngroups*naggs f64 (row-major [group][agg]). agg_ops (or NULL = all
+ * SUM): 0=SUM, 1=MIN, 2=MAX per aggregate — the caller must initialize
+ * `out` to the matching identity (0 / +inf / -inf) before the first
+ * run; untouched MIN/MAX groups keep the identity. */
 int qk_jit_agg_build(const char *pred_or_empty, const char *group_expr,
                      int ngroups, int naggs, const char *const *agg_exprs,
-                     int ncols, const int *coltypes, void **prog_out);
+                     const int *agg_ops, int ncols, const int *coltypes,
+                     void **prog_out);
 int qk_jit_agg_run(void *prog, void *stream, uint64_t n,
                    const void *const *col_ptrs, double *out_dev);
 int qk_jit_filter_free(void *prog);   /* frees any qk_jit_* program */

@@ -58,6 +58,16 @@ static std::string gen_source(const char *expr, int ncols,
   std::string s;
   s += "#define BLOCK 256\n#define WAVE 64\n";
   s += "typedef unsigned long long u64; typedef unsigned u32;\n";
+  s += "__device__ inline void qk_atomic_mm(double* a, double v, int mx) {\n"
+       "  unsigned long long* p = (unsigned long long*)a;\n"
+       "  unsigned long long old = *p;\n"
+       "  for (;;) {\n"
+       "    double cur = __longlong_as_double(old);\n"
+       "    if (mx ? (v <= cur) : (v >= cur)) return;\n"
+       "    unsigned long long prev = atomicCAS(p, old,"
+       " __double_as_longlong(v));\n"
+       "    if (prev == old) return;\n"
+       "    old = prev;\n  }\n}\n";
   s += "extern \"C\" __global__ __launch_bounds__(BLOCK) void jit_count(\n";
   s += "    u64 n, u64 chunk, u64* block_counts";
   for (int c = 0; c < ncols; c++) {
@@ -248,11 +258,26 @@ extern "C" int qk_jit_filter_run(void *prog, void *stream, uint64_t n,
 
 static std::string gen_agg_source(const char *pred, const char *group_expr,
                                   int ngroups, int naggs,
-                                  const char *const *agg_exprs, int ncols,
+                                  const char *const *agg_exprs,
+                                  const int *agg_ops, int ncols,
                                   const int *coltypes) {
+  auto op = [&](int a) { return agg_ops ? agg_ops[a] : 0; };
+  auto init_of = [&](int a) {
+    return op(a) == 1 ? "(1.0/0.0)" : op(a) == 2 ? "(-1.0/0.0)" : "0.0";
+  };
   std::string s;
   s += "#define BLOCK 256\n#define WAVE 64\n";
   s += "typedef unsigned long long u64; typedef unsigned u32;\n";
+  s += "__device__ inline void qk_atomic_mm(double* a, double v, int mx) {\n"
+       "  unsigned long long* p = (unsigned long long*)a;\n"
+       "  unsigned long long old = *p;\n"
+       "  for (;;) {\n"
+       "    double cur = __longlong_as_double(old);\n"
+       "    if (mx ? (v <= cur) : (v >= cur)) return;\n"
+       "    unsigned long long prev = atomicCAS(p, old,"
+       " __double_as_longlong(v));\n"
+       "    if (prev == old) return;\n"
+       "    old = prev;\n  }\n}\n";
   s += "extern \"C\" __global__ __launch_bounds__(BLOCK) void jit_agg(\n";
   s += "    u64 n, double* __restrict__ out";
   for (int c = 0; c < ncols; c++) {
@@ -268,7 +293,7 @@ static std::string gen_agg_source(const char *pred, const char *group_expr,
   for (int g = 0; g < ngroups; g++)
     for (int a = 0; a < naggs; a++)
       s += "  double acc_" + std::to_string(g) + "_" + std::to_string(a) +
-           " = 0.0;\n";
+           " = " + init_of(a) + ";\n";
   // one row's predicate + accumulate, parameterized by value suffix
   auto body = [&](const char *suf) {
     std::string b;
@@ -305,10 +330,17 @@ static std::string gen_agg_source(const char *pred, const char *group_expr,
       b += pfx + (g == 0 ? std::string("  if (gid == 0) {\n")
                          : "  else if (gid == " + std::to_string(g) +
                                ") {\n");
-      for (int a = 0; a < naggs; a++)
-        b += pfx + "    acc_" + std::to_string(g) + "_" +
-             std::to_string(a) + " += (double)(" + ren(agg_exprs[a]) +
-             ");\n";
+      for (int a = 0; a < naggs; a++) {
+        std::string acc = "acc_" + std::to_string(g) + "_" +
+                          std::to_string(a);
+        std::string val = "(double)(" + ren(agg_exprs[a]) + ")";
+        if (op(a) == 1)
+          b += pfx + "    " + acc + " = fmin(" + acc + ", " + val + ");\n";
+        else if (op(a) == 2)
+          b += pfx + "    " + acc + " = fmax(" + acc + ", " + val + ");\n";
+        else
+          b += pfx + "    " + acc + " += " + val + ";\n";
+      }
       b += pfx + "  }\n";
     }
     b += pfx + "}\n";
@@ -367,23 +399,49 @@ static std::string gen_agg_source(const char *pred, const char *group_expr,
   for (int g = 0; g < ngroups; g++)
     for (int a = 0; a < naggs; a++) {
       std::string acc = "acc_" + std::to_string(g) + "_" + std::to_string(a);
+      const char *comb = op(a) == 1 ? "t = fmin(t, __shfl_down(t, off))"
+                         : op(a) == 2 ? "t = fmax(t, __shfl_down(t, off))"
+                                      : "t += __shfl_down(t, off)";
       s += "  { double t = " + acc + ";\n"
            "    for (int off = WAVE / 2; off > 0; off >>= 1)\n"
-           "      t += __shfl_down(t, off);\n"
+           "      " + std::string(comb) + ";\n"
            "    if (lane == 0) lds[wid][" +
            std::to_string(g * naggs + a) + "] = t; }\n";
     }
-  s += "  __syncthreads();\n"
-       "  if (threadIdx.x < " + std::to_string(ngroups * naggs) + ") {\n"
-       "    double t = 0;\n"
-       "    for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w][threadIdx.x];\n"
-       "    if (t != 0.0) atomicAdd(&out[threadIdx.x], t);\n  }\n}\n";
+  s += "  __syncthreads();\n";
+  // per-accumulator block combine + atomic tail (op-specific)
+  for (int g = 0; g < ngroups; g++)
+    for (int a = 0; a < naggs; a++) {
+      std::string idx = std::to_string(g * naggs + a);
+      std::string body;
+      if (op(a) == 0)
+        body = "    double t = 0;\n"
+               "    for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w][" +
+               idx + "];\n"
+               "    if (t != 0.0) atomicAdd(&out[" + idx + "], t);\n";
+      else if (op(a) == 1)
+        body = "    double t = (1.0/0.0);\n"
+               "    for (int w = 0; w < BLOCK / WAVE; w++)"
+               " t = fmin(t, lds[w][" + idx + "]);\n"
+               "    if (t < (1.0/0.0)) qk_atomic_mm(&out[" + idx +
+               "], t, 0);\n";
+      else
+        body = "    double t = (-1.0/0.0);\n"
+               "    for (int w = 0; w < BLOCK / WAVE; w++)"
+               " t = fmax(t, lds[w][" + idx + "]);\n"
+               "    if (t > (-1.0/0.0)) qk_atomic_mm(&out[" + idx +
+               "], t, 1);\n";
+      s += "  if (threadIdx.x == " + std::to_string(g * naggs + a) +
+           " % BLOCK && threadIdx.x < BLOCK) {\n" + body + "  }\n";
+    }
+  s += "}\n";
   return s;
 }
 
 extern "C" int qk_jit_agg_build(const char *pred, const char *group_expr,
                                 int ngroups, int naggs,
-                                const char *const *agg_exprs, int ncols,
+                                const char *const *agg_exprs,
+                                const int *agg_ops, int ncols,
                                 const int *coltypes, void **prog_out) {
   if (ncols < 1 || ncols > QK_JIT_MAX_COLS)
     return j_fail("qk_jit_agg_build", "ncols out of range");
@@ -394,8 +452,12 @@ extern "C" int qk_jit_agg_build(const char *pred, const char *group_expr,
   for (int c = 0; c < ncols; c++)
     if (!type_name(coltypes[c]))
       return j_fail("qk_jit_agg_build", "bad column type");
+  if (agg_ops)
+    for (int a = 0; a < naggs; a++)
+      if (agg_ops[a] < 0 || agg_ops[a] > 2)
+        return j_fail("qk_jit_agg_build", "bad agg op");
   std::string src = gen_agg_source(pred, group_expr, ngroups, naggs,
-                                   agg_exprs, ncols, coltypes);
+                                   agg_exprs, agg_ops, ncols, coltypes);
   hiprtcProgram prog;
   if (hiprtcCreateProgram(&prog, src.c_str(), "qk_jit_agg.cu", 0, nullptr,
                           nullptr) != HIPRTC_SUCCESS)

@@ -312,7 +312,7 @@ def translate_arith(expr, schema, translator=None):
     return e, tr.cols
 
 
-_AGG_SQL = re.compile(r"^\s*(sum|count)\s*\(\s*(.*?)\s*\)\s*"
+_AGG_SQL = re.compile(r"^\s*(sum|count|min|max)\s*\(\s*(.*?)\s*\)\s*"
                       r"(?:as\s+([A-Za-z_][A-Za-z0-9_]*))?\s*$",
                       re.I | re.S)
 
@@ -347,18 +347,21 @@ class JitAggregate:
         for ref, card in gparts:
             gexpr = "(%s) * %d + (int)%s" % (gexpr, card, ref)
         self.agg_names = []
+        self.agg_ops = []            # 0=SUM, 1=MIN, 2=MAX (COUNT -> SUM 1)
         agg_exprs = []
         for a in aggs:
             m = _AGG_SQL.match(a)
             if not m:
-                raise ValueError("unsupported aggregate %r "
-                                 "(SUM(expr) / COUNT(*))" % a)
+                raise ValueError("unsupported aggregate %r (SUM/MIN/MAX"
+                                 "(expr) / COUNT(*))" % a)
             fn, inner, alias = m.group(1).lower(), m.group(2), m.group(3)
             if fn == "count":
                 agg_exprs.append("1.0")
+                self.agg_ops.append(0)
             else:
                 e, _ = translate_arith(inner, schema, tr)
                 agg_exprs.append(e)
+                self.agg_ops.append({"sum": 0, "min": 1, "max": 2}[fn])
             self.agg_names.append(alias or a.strip())
         self.naggs = len(agg_exprs)
         self.cols = tr.cols
@@ -367,17 +370,18 @@ class JitAggregate:
         lib = shim._lib
         lib.qk_jit_agg_build.argtypes = [
             ctypes.c_char_p, ctypes.c_char_p, ctypes.c_int, ctypes.c_int,
-            ctypes.POINTER(ctypes.c_char_p), ctypes.c_int,
-            ctypes.POINTER(ctypes.c_int), c_vp]
+            ctypes.POINTER(ctypes.c_char_p), ctypes.POINTER(ctypes.c_int),
+            ctypes.c_int, ctypes.POINTER(ctypes.c_int), c_vp]
         lib.qk_jit_last_error.restype = ctypes.c_char_p
         exprs_c = (ctypes.c_char_p * self.naggs)(
             *[e.encode() for e in agg_exprs])
         types_c = (ctypes.c_int * len(self.cols))(
             *[_TYPE_CODE[d] for d in self.dtypes])
+        ops_c = (ctypes.c_int * self.naggs)(*self.agg_ops)
         prog = c_vp(0)
         rc = lib.qk_jit_agg_build(pred_expr.encode(), gexpr.encode(),
                                   self.ngroups, self.naggs, exprs_c,
-                                  len(self.cols), types_c,
+                                  ops_c, len(self.cols), types_c,
                                   ctypes.byref(prog))
         if rc != 0:
             raise shim.QkError("jit agg build failed: %s"
@@ -385,11 +389,19 @@ class JitAggregate:
         self.prog = prog
 
     def make_acc(self):
-        from . import ops
-        from .shim import DevBuffer
+        """Accumulator initialized to each aggregate's identity
+        (0 / +inf / -inf) — MIN/MAX groups never touched keep it."""
+        from .shim import DevBuffer, c_vp as _vp
         b = DevBuffer(self.ngroups * self.naggs * 8)
-        shim.call("qk_dmemset", b.ptr, 0,
-                  c_u64(self.ngroups * self.naggs * 8))
+        init = np.zeros((self.ngroups, self.naggs), dtype=np.float64)
+        for a, o in enumerate(self.agg_ops):
+            if o == 1:
+                init[:, a] = np.inf
+            elif o == 2:
+                init[:, a] = -np.inf
+        flat = np.ascontiguousarray(init.reshape(-1))
+        shim.call("qk_h2d", b.ptr, flat.ctypes.data_as(_vp),
+                  c_u64(flat.nbytes))
         return b
 
     def run(self, cols, acc, stream=None):

@@ -1171,3 +1171,53 @@ def test_executor_left_join_plugin_api(gpu):
         (3, 1, None), (7, 3, None),
     ])
     assert got == want
+
+
+def test_jit_aggregate_min_max(gpu, data):
+    """JIT aggregate with MIN/MAX ops (generalizing GroupByI64's op set
+    to the fused register-accumulator kernel): mixed
+    sum/min/max/count per group vs numpy on the same rows, including a
+    group the predicate filters out entirely (keeps the identity)."""
+    from quokka_amd import jit, staging
+    li = data["lineitem"]
+    cols = staging.stage_columns(li)
+    schema = {k: v.dtype for k, v in cols.items()}
+    agg = jit.JitAggregate(
+        schema,
+        group_keys=[("l_returnflag", 3)],
+        aggs=["min(l_extendedprice) as mn",
+              "max(l_extendedprice * (1 - l_discount)) as mx",
+              "sum(l_quantity) as s",
+              "count(*) as n"],
+        predicate="l_shipdate > date '1995-06-17'")
+    acc = agg.make_acc()
+    agg.run(cols, acc)
+    got = agg.read(acc)                      # (3 groups, 4 aggs)
+    mask = li["l_shipdate"] > (np.datetime64("1995-06-17")
+                               - np.datetime64("1970-01-01")).astype(int)
+    for g in range(3):
+        m = mask & (li["l_returnflag"] == g)
+        if m.sum() == 0:
+            assert got[g, 0] == np.inf and got[g, 1] == -np.inf
+            assert got[g, 2] == 0 and got[g, 3] == 0
+            continue
+        np.testing.assert_allclose(got[g, 0], li["l_extendedprice"][m].min(),
+                                   rtol=1e-12)
+        rev = li["l_extendedprice"][m] * (1 - li["l_discount"][m])
+        np.testing.assert_allclose(got[g, 1], rev.max(), rtol=1e-12)
+        np.testing.assert_allclose(got[g, 2], li["l_quantity"][m].sum(),
+                                   rtol=1e-9)
+        assert got[g, 3] == m.sum()
+    # batched accumulation preserves min/max identity semantics
+    acc2 = agg.make_acc()
+    half = len(li["l_shipdate"]) // 2
+    for lo, hi in ((0, half), (half, len(li["l_shipdate"]))):
+        chunk = {k: v[lo:hi] for k, v in li.items()}
+        ccols = staging.stage_columns(chunk)
+        agg.run(ccols, acc2)
+        for c in ccols.values():
+            c.free()
+    np.testing.assert_allclose(agg.read(acc2), got, rtol=1e-12)
+    acc.free(); acc2.free(); agg.free()
+    for c in cols.values():
+        c.free()
