@@ -1221,3 +1221,73 @@ def test_jit_aggregate_min_max(gpu, data):
     acc.free(); acc2.free(); agg.free()
     for c in cols.values():
         c.free()
+
+
+def test_q10_shaped_composed_plan(gpu, data):
+    """A query we did NOT hand-fuse, composed exactly as pyquokka lowers
+    TPC-H Q10 (apps/tpc-h/tpch.py Q10 plan shape: filter -> join chain ->
+    group-by -> top-k): returned-lineitem revenue per customer inside an
+    order-date window, through gpu_partition_fn (JIT predicate) ->
+    GPUBuildProbeJoinExecutor x2 -> GPUAggExecutor -> GPUTopKExecutor,
+    all via the execute/done plugin contract, vs a numpy restatement."""
+    import pyarrow as pa
+    from quokka_amd import (GPUAggExecutor, GPUBuildProbeJoinExecutor,
+                            GPUTopKExecutor, gpu_partition_fn)
+    li, od, cu = data["lineitem"], data["orders"], data["customer"]
+    D1, D2 = 8800, 9100                       # order-date window (date32)
+    R = 2                                     # l_returnflag == 'R' code
+
+    # --- numpy restatement (same dense-key layout as the oracle) ---
+    om = (od["o_orderdate"] >= D1) & (od["o_orderdate"] < D2)
+    okeys = od["o_orderkey"][om]
+    ocust = od["o_custkey"][om]
+    key2cust = dict(zip(okeys.tolist(), ocust.tolist()))
+    lm = li["l_returnflag"] == R
+    rev = {}
+    for k, p, d in zip(li["l_orderkey"][lm].tolist(),
+                       li["l_extendedprice"][lm].tolist(),
+                       li["l_discount"][lm].tolist()):
+        c = key2cust.get(k)
+        if c is not None:
+            rev[c] = rev.get(c, 0.0) + p * (1 - d)
+    want = sorted(rev.items(), key=lambda kv: (-kv[1], kv[0]))[:20]
+
+    # --- GPU plan ---
+    t_li = pa.table({k: li[k] for k in ("l_orderkey", "l_extendedprice",
+                                        "l_discount", "l_returnflag")})
+    fl = gpu_partition_fn(t_li, 0, 1, key="l_orderkey",
+                          predicate="l_returnflag = %d" % R)[0]
+    t_od = pa.table({k: od[k] for k in ("o_orderkey", "o_custkey",
+                                        "o_orderdate")})
+    fo = gpu_partition_fn(
+        t_od, 0, 1, key="o_orderkey",
+        predicate="o_orderdate >= %d and o_orderdate < %d" % (D1, D2))[0]
+    t_cu = pa.table({"c_custkey": cu["c_custkey"]})
+    j1 = GPUBuildProbeJoinExecutor(left_on="o_custkey",
+                                   right_on="c_custkey", how="inner")
+    j1.execute([t_cu], 1, 0)
+    r1 = j1.execute([fo], 0, 0)
+    j1.done(0)
+    j2 = GPUBuildProbeJoinExecutor(left_on="l_orderkey",
+                                   right_on="o_orderkey", how="inner")
+    j2.execute([r1], 1, 0)
+    r2 = j2.execute([fl], 0, 0)
+    j2.done(0)
+    # map-side partial (the two-phase rewrite's e0_agg_0 column)
+    partial = pa.table({
+        "c_custkey": r2.column("o_custkey"),
+        "e0_agg_0": np.asarray(r2.column("l_extendedprice")) *
+        (1 - np.asarray(r2.column("l_discount")))})
+    agg = GPUAggExecutor(["c_custkey"], [], "sum(e0_agg_0) as revenue")
+    agg.execute([partial], 0, 0)
+    groups = agg.done(0)
+    topk = GPUTopKExecutor(["revenue", "c_custkey"], 20,
+                           descending=[True, False])
+    topk.execute([groups], 0, 0)
+    out = topk.done(0)
+    got = list(zip(out.column("c_custkey").to_pylist(),
+                   out.column("revenue").to_pylist()))
+    assert len(got) == len(want)
+    for (gc, gr), (wc, wr) in zip(got, want):
+        assert gc == wc
+        np.testing.assert_allclose(gr, wr, rtol=1e-9)
