@@ -173,3 +173,18 @@ def test_fuzz_random_files_gpu(gpu):
         finally:
             for c in got.values():
                 (c[0] if isinstance(c, tuple) else c).free()
+
+
+def test_column_selection(gpu):
+    from quokka_amd import parquet_gpu as P
+    rng = np.random.default_rng(9)
+    t = pa.table({"a": rng.integers(0, 1000, 5000),
+                  "b": rng.random(5000),
+                  "c": rng.integers(0, 9, 5000)})
+    raw = write(t, use_dictionary=False)
+    cols = P.read_table(raw, columns=["b", "c"])
+    assert set(cols) == {"b", "c"}
+    np.testing.assert_array_equal(cols["b"].to_numpy(5000),
+                                  t.column("b").to_numpy())
+    for c in cols.values():
+        c.free()
