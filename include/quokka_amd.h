@@ -419,6 +419,14 @@ int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
  * unknown value = row error), 4=skip. Trailing '\r' stripped; trailing
  * separators (dbgen .tbl) tolerated. err_row: device u64 initialized to
  * UINT64_MAX; the LOWEST failing row index lands there. */
+/* Range-partition ids (quokka_runtime.py:234-243 partition_key_range):
+ * id = (key - 1) / (total_range / nparts), floor division, clamped into
+ * [0, nparts-1] (the reference misroutes keys outside [1, total_range];
+ * we clamp — documented divergence, same key->same channel). Feed the
+ * ids to qk_partition_hist/scatter (id % nparts == id). */
+int qk_range_part_ids(void *stream, uint64_t n, const int64_t *keys,
+                      int64_t per_range, uint32_t nparts, int64_t *out);
+
 #define QK_CSV_MAX_DICT 32
 int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
                     const uint8_t *bytes, uint64_t *out_pos,

@@ -2113,6 +2113,34 @@ extern "C" int qk_pq_rle_pages(void *stream, uint64_t npages,
   return 0;
 }
 
+// ---- range partition ids (include/quokka_amd.h for the contract) ------
+__global__ void __launch_bounds__(BLOCK) k_range_part_ids(
+    uint64_t n, const int64_t *__restrict__ keys, int64_t per_range,
+    uint32_t nparts, int64_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    int64_t k = keys[i] - 1;
+    int64_t q = k / per_range;
+    if (k % per_range < 0) q--;               // floor division
+    if (q < 0) q = 0;
+    if (q >= (int64_t)nparts) q = (int64_t)nparts - 1;
+    out[i] = q;
+  }
+}
+extern "C" int qk_range_part_ids(void *stream, uint64_t n,
+                                 const int64_t *keys, int64_t per_range,
+                                 uint32_t nparts, int64_t *out) {
+  if (!n) return 0;
+  if (per_range <= 0 || !nparts)
+    return qk_fail("qk_range_part_ids", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_range_part_ids, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, keys, per_range, nparts, out);
+  QK_TRY("qk_range_part_ids", hipGetLastError());
+  return 0;
+}
+
 // ---- GPU CSV parse (include/quokka_amd.h for the contract) ------------
 // The reference's CSV scan reads byte ranges and hands them to
 // polars.read_csv on the CPU (unordered_readers.py:273-442, :438). Here

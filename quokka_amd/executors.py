@@ -591,7 +591,7 @@ class GPUTopKExecutor(Executor):
 
 def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
                      predicate=None, string_dicts=None, projection=None,
-                     batch_agg=None):
+                     batch_agg=None, partitioner="hash", total_range=None):
     """GPU partition function mirroring the reference's full partition_fn
     (core.py:152-195): optional PREDICATE (filter_sql grammar, JIT-fused
     on device via quokka_amd.jit) -> optional map-side PARTIAL AGGREGATE
@@ -607,9 +607,18 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     (jit.JitAggregate — the Q1-shaped kernel), and what gets partitioned
     (by group id % N) is the tiny partial table, not rows: the map-side
     fusion that shrinks shuffle bytes from O(rows) to O(groups).
+    partitioner: "hash" (key % N, quokka_runtime.py:222), "range"
+    (id = (key-1) // (total_range // N), quokka_runtime.py:234-243 —
+    keys outside [1, total_range] are CLAMPED to the edge channels where
+    the reference would misroute them; same key -> same channel either
+    way), or "broadcast" (every channel gets the full table,
+    quokka_runtime.py:245-246).
     `data`: pyarrow Table; returns dict target_channel -> pyarrow Table."""
     import pyarrow as pa
     ops, shim, staging = _lazy_gpu()
+    if partitioner == "broadcast":
+        return {i: data for i in range(num_target_channels)}
+    assert partitioner in ("hash", "range"), partitioner
     host_cols = {c: staging.column_to_numpy(
         data.column(c), (string_dicts or {}).get(c))
         for c in data.column_names}
@@ -676,6 +685,15 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     if keys.dtype != np.int64:
         raise TypeError("gpu_partition_fn: int64 keys only in round 1")
     kcol = shim.DevColumn.from_numpy(keys)
+    if partitioner == "range":
+        assert total_range, "range partitioner needs total_range"
+        from .shim import DevColumn, c_u64, c_i64, c_u32
+        per = max(1, int(total_range) // num_target_channels)
+        ids = DevColumn(np.int64, max(1, len(keys)))
+        shim.call("qk_range_part_ids", None, c_u64(len(keys)), kcol.ptr,
+                  c_i64(per), c_u32(num_target_channels), ids.ptr)
+        kcol.free()
+        kcol = ids
     offsets, idx = ops.partition_i64(kcol, num_target_channels)
     sel = idx.to_numpy(len(keys))
     names = sorted(projection) if projection else sorted(host_cols)

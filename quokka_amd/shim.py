@@ -108,6 +108,7 @@ _SIGS = {
     "qk_iota_u32": [c_vp, c_u64, c_vp],
     "qk_bnot_u64": [c_vp, c_u64, c_vp],
     "qk_partition_hist": [c_vp, c_u64, c_vp, c_u32, c_vp],
+    "qk_range_part_ids": [c_vp, c_u64, c_vp, c_i64, c_u32, c_vp],
     "qk_csv_newlines": [c_vp, c_u64, c_u64, c_vp, c_vp, c_vp],
     "qk_csv_parse": [c_vp, c_u64, c_vp, c_u64, c_vp, c_u8, ctypes.c_int,
                      c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],

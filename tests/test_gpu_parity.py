@@ -1291,3 +1291,41 @@ def test_q10_shaped_composed_plan(gpu, data):
     for (gc, gr), (wc, wr) in zip(got, want):
         assert gc == wc
         np.testing.assert_allclose(gr, wr, rtol=1e-9)
+
+
+def test_gpu_partition_fn_range_and_broadcast(gpu):
+    """The reference's other two partitioners (quokka_runtime.py:234-246):
+    range = (key-1) // (total_range // N) (clamped at the edges where the
+    reference would misroute out-of-range keys), broadcast = full table
+    to every channel."""
+    import pyarrow as pa
+    from quokka_amd import gpu_partition_fn
+    rng = np.random.default_rng(65)
+    keys = rng.integers(1, 10_001, 30_000).astype(np.int64)
+    vals = rng.random(30_000)
+    t = pa.table({"k": keys, "v": vals})
+    nch = 4
+    out = gpu_partition_fn(t, 0, nch, key="k", partitioner="range",
+                           total_range=10_000)
+    per = 10_000 // nch
+    want = np.clip((keys - 1) // per, 0, nch - 1)
+    total = 0
+    for ch, tbl in out.items():
+        gk = np.asarray(tbl.column("k"))
+        assert np.all(np.clip((gk - 1) // per, 0, nch - 1) == ch)
+        total += len(tbl)
+    assert total == len(keys)
+    assert sorted(out) == sorted(set(want.tolist()))
+    # edge clamping: keys outside [1, total_range]
+    t2 = pa.table({"k": np.array([-5, 0, 1, 10_000, 10_001, 99_999],
+                                 dtype=np.int64)})
+    out2 = gpu_partition_fn(t2, 0, nch, key="k", partitioner="range",
+                            total_range=10_000)
+    assert np.asarray(out2[0].column("k")).tolist() == [-5, 0, 1]
+    assert np.asarray(out2[nch - 1].column("k")).tolist() == [
+        10_000, 10_001, 99_999]
+    # broadcast
+    outb = gpu_partition_fn(t, 0, 3, partitioner="broadcast")
+    assert sorted(outb) == [0, 1, 2]
+    for ch in range(3):
+        assert outb[ch] is t
