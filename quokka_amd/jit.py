@@ -8,6 +8,9 @@ reference's TPC-H workload uses (apps/tpc-h/tpch.py filter_sql calls):
   comparisons  = == != <> < <= > >=
   connectives  AND OR NOT, parentheses
   BETWEEN x AND y (inclusive, as DuckDB)
+  expr IN (v1, v2, ...) / expr NOT IN (...) — numeric lists, or string
+  lists on dict-coded columns (each element resolved through the
+  column's StringDict; absent values never match)
   arithmetic   + - * / on columns and literals
   literals     ints, floats, date 'YYYY-MM-DD' (+/- interval 'N'
                day/month/year, evaluated host-side to date32 days),
@@ -40,7 +43,7 @@ _TOKEN = re.compile(r"""
     | (?P<interval>interval\s*'(\d+)'\s*(day|month|year)s?)
     | (?P<num>\d+\.\d+|\.\d+|\d+)
     | (?P<str>'[^']*')
-    | (?P<op><>|!=|>=|<=|==|=|<|>|\+|-|\*|/|\(|\))
+    | (?P<op><>|!=|>=|<=|==|=|<|>|\+|-|\*|/|\(|\)|,)
     | (?P<word>[A-Za-z_][A-Za-z0-9_]*)
     )""", re.X | re.I)
 
@@ -83,7 +86,7 @@ def _tokenize(s):
             out.append(_Tok("op", m.group("op")))
         else:
             w = m.group("word").upper()
-            if w in ("AND", "OR", "NOT", "BETWEEN"):
+            if w in ("AND", "OR", "NOT", "BETWEEN", "IN"):
                 out.append(_Tok(w, w))
             else:
                 out.append(_Tok("ident", m.group("word")))
@@ -188,6 +191,17 @@ class Translator:
                 self.i = save
         left, _ = self.p_arith()
         t = self.peek()
+        if t and (t.kind == "IN" or
+                  (t.kind == "NOT" and self.i + 1 < len(self.toks)
+                   and self.toks[self.i + 1].kind == "IN")):
+            neg = t.kind == "NOT"
+            self.take()
+            if neg:
+                self.take("IN")
+            els = self._p_in_list(allow_str=False)
+            body = " || ".join("((%s) == (%s))" % (left, repr(e))
+                               for e in els)
+            return ("!(%s)" if neg else "(%s)") % body
         if t and t.kind == "BETWEEN":
             self.take()
             lo, _ = self.p_arith()
@@ -201,6 +215,36 @@ class Translator:
         cop = {"=": "==", "<>": "!="}.get(op, op)
         right, _ = self.p_arith()
         return "(%s) %s (%s)" % (left, cop, right)
+
+    def _p_in_list(self, allow_str):
+        """Parse IN's '(v1, v2, ...)' — literal numbers (and strings when
+        allow_str) only."""
+        p = self.take("op")   # caller already consumed IN (and NOT)
+        if p.val != "(":
+            raise ValueError("expected ( after IN")
+        out = []
+        while True:
+            t = self.take()
+            neg = False
+            if t.kind == "op" and t.val in ("-", "+"):
+                neg = t.val == "-"
+                t = self.take()
+            if t.kind == "num":
+                out.append(-t.val if neg else t.val)
+            elif t.kind == "date" and not neg:
+                out.append(t.val)
+            elif t.kind == "str" and allow_str and not neg:
+                out.append(t.val)
+            else:
+                raise ValueError("bad IN list element %r" % (t,))
+            t = self.take("op")
+            if t.val == ")":
+                break
+            if t.val != ",":
+                raise ValueError("expected , or ) in IN list")
+        if not out:
+            raise ValueError("empty IN list")
+        return out
 
     def p_arith(self):
         left, lt = self.p_term()
@@ -266,6 +310,28 @@ class Translator:
             ref, dtype = self.col_ref(t.val)
             # string equality: partner literal becomes the dict code
             nxt = self.peek()
+            if dtype == np.dtype(np.uint8) and nxt and (
+                    nxt.kind == "IN" or
+                    (nxt.kind == "NOT" and self.i + 1 < len(self.toks)
+                     and self.toks[self.i + 1].kind == "IN")):
+                neg = nxt.kind == "NOT"
+                self.take()
+                if neg:
+                    self.take("IN")
+                codes = []
+                for el in self._p_in_list(allow_str=True):
+                    if isinstance(el, str):
+                        sd = self.string_dicts.get(t.val)
+                        if sd is None:
+                            raise ValueError(
+                                "string IN list on %r needs its "
+                                "StringDict" % t.val)
+                        codes.append(sd.codes.get(el, 255))
+                    else:
+                        codes.append(int(el))
+                body = " || ".join("((%s) == (%d))" % (ref, c)
+                                   for c in codes)
+                raise _Folded(("!(%s)" if neg else "(%s)") % body)
             if dtype == np.dtype(np.uint8) and nxt and nxt.kind == "op" \
                     and nxt.val in ("=", "==", "!=", "<>"):
                 save = self.i

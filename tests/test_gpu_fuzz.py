@@ -141,3 +141,27 @@ def test_fuzz_jit_filter(gpu):
         idx.free()
         for c in dcols.values():
             c.free()
+
+
+def test_jit_filter_in_lists(gpu):
+    """Compiled IN / NOT IN predicates on device vs numpy."""
+    from quokka_amd import jit, shim
+    rng = np.random.default_rng(4100)
+    n = 40_000
+    a = rng.integers(-50, 50, n).astype(np.int32)
+    k = rng.integers(-500, 500, n).astype(np.int64)
+    schema = {"a": np.dtype(np.int32), "k": np.dtype(np.int64)}
+    for pred, want in [
+        ("a in (3, -7, 11, 42)", np.isin(a, [3, -7, 11, 42])),
+        ("k not in (0, 250, -250)", ~np.isin(k, [0, 250, -250])),
+        ("a in (1,2,3) and k > 0", np.isin(a, [1, 2, 3]) & (k > 0)),
+    ]:
+        f = jit.JitFilter(pred, schema)
+        dcols = {"a": shim.DevColumn.from_numpy(a),
+                 "k": shim.DevColumn.from_numpy(k)}
+        idx, m = f.run(dcols)
+        assert np.array_equal(idx.to_numpy(m), np.nonzero(want)[0]), pred
+        f.free()
+        idx.free()
+        for c in dcols.values():
+            c.free()

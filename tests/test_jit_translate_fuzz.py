@@ -130,3 +130,49 @@ def test_translate_date_intervals_exhaustive():
                     want = datetime.date(wy, wm, min(base.day, ld))
                 days = (want - epoch).days
                 assert expr == "(v0) < (%d)" % days, (sql, expr)
+
+
+def test_in_lists():
+    """IN / NOT IN (numeric and dict-string lists) against numpy."""
+    import numpy as np
+    from quokka_amd import jit
+    schema = {"a": np.dtype(np.int32), "b": np.dtype(np.float64)}
+    rng = np.random.default_rng(5)
+    env = {"v0": rng.integers(-10, 10, 2000).astype(np.int32),
+           "v1": np.round(rng.uniform(-2, 2, 2000), 3)}
+    cases = [
+        ("a in (1, 2, 5)", np.isin(env["v0"], [1, 2, 5])),
+        ("a not in (0, -3)", ~np.isin(env["v0"], [0, -3])),
+        ("b in (0.5, -1.25)", np.isin(env["v1"], [0.5, -1.25])),
+        ("a in (1,2) and b < 0.5",
+         np.isin(env["v0"], [1, 2]) & (env["v1"] < 0.5)),
+        ("not (a in (7))", ~np.isin(env["v0"], [7])),
+    ]
+    for pred, want in cases:
+        e, cols = jit.translate(pred, schema)
+        py = e.replace("&&", "&").replace("||", "|").replace("!(", "~(")
+        got = eval(py, {}, env)
+        assert np.array_equal(got, want), pred
+    for bad in ("a in ()", "a in (1,'x')", "a in (1", "a in 1"):
+        try:
+            jit.translate(bad, schema)
+            raise AssertionError("accepted %r" % bad)
+        except ValueError:
+            pass
+
+
+def test_in_lists_dict_strings():
+    import numpy as np
+    from quokka_amd import jit
+
+    class SD:
+        codes = {"MAIL": 0, "SHIP": 1, "RAIL": 2}
+    tr = jit.Translator({"m": np.dtype(np.uint8)}, {"m": SD()})
+    e = tr.translate("m in ('MAIL', 'RAIL')")
+    v0 = np.array([0, 1, 2, 2, 0], dtype=np.uint8)
+    got = eval(e.replace("||", "|"), {}, {"v0": v0})
+    assert np.array_equal(got, np.isin(v0, [0, 2]))
+    # absent value never matches
+    e2 = jit.Translator({"m": np.dtype(np.uint8)}, {"m": SD()}) \
+        .translate("m in ('TRUCK')")
+    assert not eval(e2.replace("||", "|"), {}, {"v0": v0}).any()
