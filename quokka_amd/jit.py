@@ -11,6 +11,10 @@ reference's TPC-H workload uses (apps/tpc-h/tpch.py filter_sql calls):
   expr IN (v1, v2, ...) / expr NOT IN (...) — numeric lists, or string
   lists on dict-coded columns (each element resolved through the
   column's StringDict; absent values never match)
+  col LIKE 'pat' / NOT LIKE — dict-coded columns only: the SQL pattern
+  (% = any run, _ = any char) is matched against the dictionary VALUES
+  host-side and becomes a code-set test (no per-row string work on
+  device — low-cardinality dictionaries make LIKE a set membership)
   arithmetic   + - * / on columns and literals
   literals     ints, floats, date 'YYYY-MM-DD' (+/- interval 'N'
                day/month/year, evaluated host-side to date32 days),
@@ -86,7 +90,7 @@ def _tokenize(s):
             out.append(_Tok("op", m.group("op")))
         else:
             w = m.group("word").upper()
-            if w in ("AND", "OR", "NOT", "BETWEEN", "IN"):
+            if w in ("AND", "OR", "NOT", "BETWEEN", "IN", "LIKE"):
                 out.append(_Tok(w, w))
             else:
                 out.append(_Tok("ident", m.group("word")))
@@ -310,6 +314,30 @@ class Translator:
             ref, dtype = self.col_ref(t.val)
             # string equality: partner literal becomes the dict code
             nxt = self.peek()
+            if dtype == np.dtype(np.uint8) and nxt and (
+                    nxt.kind == "LIKE" or
+                    (nxt.kind == "NOT" and self.i + 1 < len(self.toks)
+                     and self.toks[self.i + 1].kind == "LIKE")):
+                neg = nxt.kind == "NOT"
+                self.take()
+                if neg:
+                    self.take("LIKE")
+                p = self.take("str")
+                sd = self.string_dicts.get(t.val)
+                if sd is None:
+                    raise ValueError("LIKE on %r needs its StringDict"
+                                     % t.val)
+                import re as _re
+                rx = _re.compile(
+                    "^" + "".join(
+                        ".*" if ch == "%" else "." if ch == "_"
+                        else _re.escape(ch) for ch in p.val) + "$", _re.S)
+                codes = [c for v, c in sd.codes.items() if rx.match(v)]
+                if not codes:
+                    raise _Folded("(0)" if not neg else "(1)")
+                body = " || ".join("((%s) == (%d))" % (ref, c)
+                                   for c in sorted(codes))
+                raise _Folded(("!(%s)" if neg else "(%s)") % body)
             if dtype == np.dtype(np.uint8) and nxt and (
                     nxt.kind == "IN" or
                     (nxt.kind == "NOT" and self.i + 1 < len(self.toks)

@@ -177,3 +177,37 @@ def test_in_lists_dict_strings():
     e2 = jit.Translator({"m": np.dtype(np.uint8)}, {"m": SD()}) \
         .translate("m in ('TRUCK')")
     assert not eval(e2.replace("||", "|"), {}, {"v0": v0}).any()
+
+
+def test_like_on_dict_columns():
+    """LIKE/NOT LIKE resolve the SQL pattern against the dictionary
+    VALUES host-side (low-cardinality dicts make LIKE a code-set test)."""
+    import numpy as np
+    from quokka_amd import jit
+
+    class SD:
+        codes = {"MAIL": 0, "SHIP": 1, "RAIL": 2, "REG AIR": 3, "AIR": 4}
+
+    vals = np.array(["MAIL", "SHIP", "RAIL", "REG AIR", "AIR"])
+    v0 = np.array([0, 1, 2, 3, 4, 3, 0], dtype=np.uint8)
+
+    def run(pred):
+        tr = jit.Translator({"m": np.dtype(np.uint8)}, {"m": SD()})
+        e = tr.translate(pred)
+        r = eval(e.replace("||", "|").replace("!(", "~("), {},
+                 {"v0": v0})
+        # "(0)" / "(1)" constant folds broadcast like the C scalar would
+        return np.broadcast_to(np.asarray(r, dtype=bool), v0.shape)
+
+    import fnmatch
+    for pat, sql in [("*AIR", "%AIR"), ("R*", "R%"), ("?AIL", "_AIL"),
+                     ("*A*", "%A%"), ("TRUCK", "TRUCK")]:
+        want = np.array([fnmatch.fnmatchcase(vals[c], pat) for c in v0])
+        assert np.array_equal(run("m like '%s'" % sql), want), sql
+        assert np.array_equal(run("m not like '%s'" % sql), ~want), sql
+    try:
+        jit.Translator({"x": np.dtype(np.int64)}, {}).translate(
+            "x like '5'")
+        raise AssertionError("LIKE accepted on non-dict column")
+    except ValueError:
+        pass
