@@ -28,12 +28,14 @@ _cmp = st.sampled_from(["<", "<=", ">", ">=", "=", "<>"])
 @st.composite
 def comparison(draw):
     col = draw(_col)
-    op = draw(_cmp)
-    lit = draw(_num)
-    if draw(st.booleans()):
-        return ("cmp", col, op, lit)
-    lo, hi = sorted([draw(_num), draw(_num)])
-    return ("between", col, lo, hi)
+    kind = draw(st.integers(0, 3))
+    if kind == 0:
+        return ("cmp", col, draw(_cmp), draw(_num))
+    if kind == 1:
+        lo, hi = sorted([draw(_num), draw(_num)])
+        return ("between", col, lo, hi)
+    els = draw(st.lists(_num, min_size=1, max_size=5))
+    return ("in" if kind == 2 else "notin", col, els)
 
 
 @st.composite
@@ -55,6 +57,10 @@ def to_sql(p):
     if k == "between":
         _, col, lo, hi = p
         return "%s between %r and %r" % (col, lo, hi)
+    if k in ("in", "notin"):
+        _, col, els = p
+        return "%s %sin (%s)" % (col, "not " if k == "notin" else "",
+                                 ", ".join(repr(e) for e in els))
     if k == "not":
         return "not (%s)" % to_sql(p[1])
     return "(%s) %s (%s)" % (to_sql(p[1]), p[0], to_sql(p[2]))
@@ -70,6 +76,12 @@ def np_eval(p, cols):
     if k == "between":
         _, col, lo, hi = p
         return (cols[col] >= lo) & (cols[col] <= hi)
+    if k in ("in", "notin"):
+        _, col, els = p
+        m = np.zeros(len(cols[col]), dtype=bool)
+        for e in els:
+            m |= cols[col] == e
+        return ~m if k == "notin" else m
     if k == "not":
         return ~np_eval(p[1], cols)
     a, b = np_eval(p[1], cols), np_eval(p[2], cols)
