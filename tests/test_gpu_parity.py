@@ -1329,3 +1329,46 @@ def test_gpu_partition_fn_range_and_broadcast(gpu):
     assert sorted(outb) == [0, 1, 2]
     for ch in range(3):
         assert outb[ch] is t
+
+
+def test_distributed_sort_via_range_partition(gpu):
+    """The reference's global-sort recipe (sort_info 'range' mode,
+    unordered_readers.py:301-309 + SuperFastSortExecutor
+    sql_executors.py:88): range-partition rows by key so channel i holds
+    keys below channel i+1, sort each channel on device, concatenate in
+    channel order -> globally sorted."""
+    import pyarrow as pa
+    from quokka_amd import gpu_partition_fn, ops, shim
+    rng = np.random.default_rng(67)
+    n = 200_000
+    keys = rng.integers(1, 1_000_001, n).astype(np.int64)
+    vals = rng.random(n)
+    t = pa.table({"k": keys, "v": vals})
+    nch = 4
+    parts = gpu_partition_fn(t, 0, nch, key="k", partitioner="range",
+                             total_range=1_000_000)
+    out_k, out_v = [], []
+    for ch in range(nch):
+        if ch not in parts:
+            continue
+        ck = np.asarray(parts[ch].column("k"))
+        cv = np.asarray(parts[ch].column("v"))
+        kcol = shim.DevColumn.from_numpy(ck)
+        perm = ops.sort_permutation(kcol)
+        sel = perm.to_numpy(perm.n)
+        out_k.append(ck[sel])
+        out_v.append(cv[sel])
+        perm.free()
+        kcol.free()
+    gk = np.concatenate(out_k)
+    gv = np.concatenate(out_v)
+    order = np.argsort(keys, kind="stable")
+    assert np.array_equal(gk, keys[order])        # globally sorted
+    # stability within equal keys is preserved per channel; values must
+    # be a permutation attached to the right keys
+    assert np.array_equal(np.sort(gv), np.sort(vals))
+    lookup = {}
+    for k_, v_ in zip(keys.tolist(), vals.tolist()):
+        lookup.setdefault(k_, set()).add(round(v_, 12))
+    for k_, v_ in zip(gk[:200].tolist(), gv[:200].tolist()):
+        assert round(v_, 12) in lookup[k_]
