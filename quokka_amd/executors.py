@@ -591,7 +591,8 @@ class GPUTopKExecutor(Executor):
 
 def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
                      predicate=None, string_dicts=None, projection=None,
-                     batch_agg=None, partitioner="hash", total_range=None):
+                     batch_agg=None, partitioner="hash", total_range=None,
+                     transforms=None):
     """GPU partition function mirroring the reference's full partition_fn
     (core.py:152-195): optional PREDICATE (filter_sql grammar, JIT-fused
     on device via quokka_amd.jit) -> optional map-side PARTIAL AGGREGATE
@@ -607,6 +608,12 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     (jit.JitAggregate — the Q1-shaped kernel), and what gets partitioned
     (by group id % N) is the tiny partial table, not rows: the map-side
     fusion that shrinks shuffle bytes from O(rows) to O(groups).
+    transforms: list of (out_name, arith_expr) applied per batch AFTER
+    the predicate (the reference's folded transform_sql maps inside
+    batch_funcs, core.py:173-176 + datastream.py:652-815): each
+    expression JIT-compiles to a device elementwise kernel (quokka_amd
+    .jit.JitMap, f64 out) and the result column joins the partition +
+    projection like any input column.
     partitioner: "hash" (key % N, quokka_runtime.py:222), "range"
     (id = (key-1) // (total_range // N), quokka_runtime.py:234-243 —
     keys outside [1, total_range] are CLAMPED to the edge channels where
@@ -678,6 +685,20 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
         host_cols = {c: v[sel0] for c, v in host_cols.items()}
         f.free()
         fidx.free()
+        for c in dcols.values():
+            c.free()
+
+    if transforms:
+        from . import jit
+        dcols = {c: shim.DevColumn.from_numpy(v)
+                 for c, v in host_cols.items()}
+        schema = {c: v.dtype for c, v in dcols.items()}
+        for out_name, expr in transforms:
+            m = jit.JitMap(expr, schema)
+            o = m.run(dcols)
+            host_cols[out_name] = o.to_numpy(o.n)
+            o.free()
+            m.free()
         for c in dcols.values():
             c.free()
 

@@ -1372,3 +1372,36 @@ def test_distributed_sort_via_range_partition(gpu):
         lookup.setdefault(k_, set()).add(round(v_, 12))
     for k_, v_ in zip(gk[:200].tolist(), gv[:200].tolist()):
         assert round(v_, 12) in lookup[k_]
+
+
+def test_gpu_partition_fn_transforms(gpu):
+    """Folded map transforms inside partition_fn (the reference's
+    transform_sql batch_funcs, core.py:173-176 + datastream.py:652-815):
+    predicate -> JIT elementwise transform -> partition; the computed
+    column rides along like any input column."""
+    import pyarrow as pa
+    from quokka_amd import gpu_partition_fn
+    rng = np.random.default_rng(69)
+    n = 40_000
+    k = rng.integers(0, 10_000, n).astype(np.int64)
+    p = np.round(rng.uniform(1, 100, n), 2)
+    d = np.round(rng.uniform(0, 0.1, n), 2)
+    t = pa.table({"k": k, "p": p, "d": d})
+    out = gpu_partition_fn(
+        t, 0, 3, key="k", predicate="d < 0.05",
+        transforms=[("revenue", "p * (1 - d)")],
+        projection=["k", "revenue"])
+    mask = d < 0.05
+    want = {}
+    for kk, rr in zip(k[mask].tolist(), (p[mask] * (1 - d[mask])).tolist()):
+        want.setdefault(kk % 3, []).append((kk, round(rr, 9)))
+    total = 0
+    for ch, tbl in out.items():
+        assert tbl.column_names == ["k", "revenue"]
+        gk = np.asarray(tbl.column("k"))
+        gr = np.asarray(tbl.column("revenue"))
+        assert np.all(gk % 3 == ch)
+        got = sorted(zip(gk.tolist(), [round(v, 9) for v in gr.tolist()]))
+        assert got == sorted(want.get(ch, [])), ch
+        total += len(tbl)
+    assert total == int(mask.sum())
