@@ -614,6 +614,10 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     expression JIT-compiles to a device elementwise kernel (quokka_amd
     .jit.JitMap, f64 out) and the result column joins the partition +
     projection like any input column.
+    Row ORDER within a partition is unspecified (the device scatter
+    assigns block-chunk ranges; the reference's polars partition_by is
+    input-ordered) — every consumer on this path is order-insensitive
+    (joins/aggregates; sorts order explicitly).
     partitioner: "hash" (key % N, quokka_runtime.py:222), "range"
     (id = (key-1) // (total_range // N), quokka_runtime.py:234-243 —
     keys outside [1, total_range] are CLAMPED to the edge channels where
@@ -631,6 +635,10 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
         for c in data.column_names}
 
     if batch_agg is not None:
+        if transforms:
+            raise ValueError("batch_agg already aggregates arbitrary "
+                             "arithmetic expressions; transforms cannot "
+                             "be combined with it")
         from . import jit
         group_keys, aggs = batch_agg
         # COUNT(*) identifies which groups were actually observed (the
