@@ -71,10 +71,14 @@ def read_csv(source, schema, sep="|", header=False):
     if nrows == 0:
         pos.free()
         dev.free()
-        return {name: DevColumn(_OUT_DTYPE.get(TYPE_CODE[spec[1]],
-                                               np.int64), 1)
-                for spec in schema for name in [spec[0]]
-                if spec[1] != "skip"}
+        out = {}
+        for spec in schema:
+            if spec[1] == "skip":
+                continue
+            col = DevColumn(_OUT_DTYPE[TYPE_CODE[spec[1]]], 1)
+            col.n = 0                      # no rows; buffer is a stub
+            out[spec[0]] = col
+        return out
 
     ncols = len(schema)
     coltypes = np.zeros(ncols, dtype=np.int32)
