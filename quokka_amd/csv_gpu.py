@@ -55,7 +55,12 @@ def read_csv(source, schema, sep="|", header=False):
     n = len(arr)
     data_start = 0
     if header:
-        data_start = raw.index(b"\n") + 1
+        nl = raw.find(b"\n")
+        if nl < 0:                      # empty / header-only input
+            raw = b""
+            n = 0
+        else:
+            data_start = nl + 1
     dev = DevBuffer(max(1, n + 8))
     shim._bounce.h2d(dev.ptr, arr)
 
