@@ -51,16 +51,15 @@ def read_csv(source, schema, sep="|", header=False):
     raw = open(source, "rb").read() if isinstance(source, str) else source
     if raw and not raw.endswith(b"\n"):
         raw = raw + b"\n"
-    arr = np.frombuffer(raw, dtype=np.uint8)
-    n = len(arr)
     data_start = 0
     if header:
         nl = raw.find(b"\n")
         if nl < 0:                      # empty / header-only input
             raw = b""
-            n = 0
         else:
             data_start = nl + 1
+    arr = np.frombuffer(raw, dtype=np.uint8)
+    n = len(arr)
     dev = DevBuffer(max(1, n + 8))
     shim._bounce.h2d(dev.ptr, arr)
 
