@@ -189,3 +189,15 @@ def test_csv_reader_header_and_parquet_reader(gpu, tmp_path):
             got.append(cols["x"].to_numpy(cols["x"].n))
             cols["x"].free()
     assert np.array_equal(np.sort(np.concatenate(got)), np.arange(30))
+
+
+def test_empty_and_header_only_inputs(gpu):
+    from quokka_amd import csv_gpu
+    for raw, kw in [(b"", {}), (b"a|b\n", {"header": True}),
+                    (b"junk-no-newline", {"header": True})]:
+        cols = csv_gpu.read_csv(raw, [("a", "i64"), ("b", "f64")],
+                                sep="|", **kw)
+        assert set(cols) == {"a", "b"}
+        for c in cols.values():
+            assert c.n == 0
+            c.free()
