@@ -63,15 +63,19 @@ def read_csv(source, schema, sep="|", header=False):
     dev = DevBuffer(max(1, n + 8))
     shim._bounce.h2d(dev.ptr, arr)
 
-    # newline index
+    # newline index (counter zeroed first: the kernel early-returns
+    # without writing it when the data range is empty, and pool-recycled
+    # buffers hold stale bytes)
     pos = DevColumn(np.uint64, max(1, n - data_start))
     cnt = DevBuffer(8)
+    shim.call("qk_dmemset", cnt.ptr, 0, c_u64(8))
     shim.call("qk_csv_newlines", None, c_u64(data_start), c_u64(n),
               dev.ptr, pos.ptr, cnt.ptr)
     host_cnt = np.zeros(1, dtype=np.uint64)
     shim.call("qk_d2h", host_cnt.ctypes.data_as(c_vp), cnt.ptr, c_u64(8))
     nrows = int(host_cnt[0])
     cnt.free()
+    assert nrows <= n, (nrows, n)
     if nrows == 0:
         pos.free()
         dev.free()
