@@ -99,18 +99,23 @@ def repartition(comm, key_col, payload_cols, stream=None):
     send_counts = np.diff(offsets).astype(np.uint64)
     recv_counts = comm.exchange_counts(send_counts)
 
+    # per-column gather -> alltoallv are STREAM-ORDERED (RCCL grouped ops
+    # and the D2D self-copy launch on `stream`), so the whole exchange
+    # pipelines with ONE sync at the end; temps are freed only after it
+    # (the pool would otherwise hand a buffer still being read by an
+    # in-flight send to the next allocation)
+    temps = []
+
     def exch(col):
         ordered = col.gather(idx, n, stream)
-        if stream:
-            stream.sync()
-        recv = comm.alltoallv_column(ordered, offsets, send_counts,
+        temps.append(ordered)
+        return comm.alltoallv_column(ordered, offsets, send_counts,
                                      recv_counts, stream)
-        if stream:
-            stream.sync()
-        ordered.free()
-        return recv
 
     recv_key = exch(key_col)
     recv_payload = {name: exch(col) for name, col in payload_cols.items()}
+    shim.call("qk_stream_sync", stream.handle if stream else None)
+    for t in temps:
+        t.free()
     idx.free()
     return recv_key, recv_payload
