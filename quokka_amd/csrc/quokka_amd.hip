@@ -1723,12 +1723,7 @@ extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
 // 8 bits/pass, pass count from the key range; stability per pass via
 // (block-major offsets) x (wave-serialized in-wave ballot ranks).
 
-// 11-bit digits: 6 passes over u64 instead of 8 (25% less key+payload
-// traffic); LDS per block stays small (count hist 8 KB, scatter cursors
-// 16 KB of the 160 KB budget)
-#define RADIX_BITS 11
-#define RADIX (1 << RADIX_BITS)
-#define RADIX_MASK ((uint32_t)(RADIX - 1))
+#define RADIX 256
 
 __global__ void __launch_bounds__(BLOCK) k_radix_count(
     uint64_t n, const uint64_t *__restrict__ keys, int shift, uint64_t chunk,
@@ -1739,44 +1734,37 @@ __global__ void __launch_bounds__(BLOCK) k_radix_count(
   uint64_t lo = (uint64_t)blockIdx.x * chunk;
   uint64_t hi = qk_min_u64(n, lo + chunk);
   for (uint64_t i = lo + threadIdx.x; i < hi; i += BLOCK)
-    atomicAdd(&hist[(uint32_t)(keys[i] >> shift) & RADIX_MASK], 1u);
+    atomicAdd(&hist[(uint32_t)(keys[i] >> shift) & 255u], 1u);
   __syncthreads();
   for (int d = threadIdx.x; d < RADIX; d += BLOCK)
     counts[(uint64_t)d * nblocks + blockIdx.x] = hist[d];
 }
 
-// one block: each thread strides over digits, owning a digit's row of
-// per-block counts; produces exclusive global bases per (digit, block).
-// The cross-digit exclusive scan is serial on thread 0 (RADIX iterations
-// of add — microseconds, launched once per pass).
-__global__ void __launch_bounds__(BLOCK) k_radix_scan(
+// one 256-thread block: thread d owns digit d's row of per-block counts;
+// produces exclusive global bases per (digit, block)
+__global__ void __launch_bounds__(RADIX) k_radix_scan(
     uint32_t nblocks, uint32_t *__restrict__ counts,
     uint64_t *__restrict__ bases /* [RADIX][nblocks] */) {
   __shared__ uint64_t tot[RADIX];
-  for (int d = threadIdx.x; d < RADIX; d += BLOCK) {
-    uint64_t acc = 0;
-    for (uint32_t b = 0; b < nblocks; b++) {
-      uint32_t c = counts[(uint64_t)d * nblocks + b];
-      bases[(uint64_t)d * nblocks + b] = acc;
-      acc += c;
-    }
-    tot[d] = acc;
+  int d = threadIdx.x;
+  uint64_t acc = 0;
+  for (uint32_t b = 0; b < nblocks; b++) {
+    uint32_t c = counts[(uint64_t)d * nblocks + b];
+    bases[(uint64_t)d * nblocks + b] = acc;
+    acc += c;
   }
+  tot[d] = acc;
   __syncthreads();
-  if (threadIdx.x == 0) {
-    uint64_t acc = 0;
-    for (int d = 0; d < RADIX; d++) {
-      uint64_t c = tot[d];
-      tot[d] = acc;
-      acc += c;
-    }
+  // exclusive scan of 256 digit totals (simple doubling scan)
+  for (int off = 1; off < RADIX; off <<= 1) {
+    uint64_t v = d >= off ? tot[d - off] : 0;
+    __syncthreads();
+    tot[d] += v;
+    __syncthreads();
   }
-  __syncthreads();
-  for (int d = threadIdx.x; d < RADIX; d += BLOCK) {
-    uint64_t base = tot[d];
-    for (uint32_t b = 0; b < nblocks; b++)
-      bases[(uint64_t)d * nblocks + b] += base;
-  }
+  uint64_t base = d == 0 ? 0 : tot[d - 1];
+  for (uint32_t b = 0; b < nblocks; b++)
+    bases[(uint64_t)d * nblocks + b] += base;
 }
 
 __global__ void __launch_bounds__(BLOCK) k_radix_scatter(
@@ -1796,12 +1784,12 @@ __global__ void __launch_bounds__(BLOCK) k_radix_scatter(
     bool valid = i < hi;
     uint64_t k = valid ? keys_in[i] : ~0ULL;
     uint32_t pay = valid ? pay_in[i] : 0;
-    uint32_t d8 = (uint32_t)(k >> shift) & RADIX_MASK;
+    uint32_t d8 = (uint32_t)(k >> shift) & 255u;
     // waves take turns IN ORDER: block stability = wave order x lane order
     for (int w = 0; w < BLOCK / WAVE; w++) {
       if (wid == w) {
         uint64_t same = __ballot(valid);
-        for (int j = 0; j < RADIX_BITS; j++) {
+        for (int j = 0; j < 8; j++) {
           uint64_t bj = __ballot((d8 >> j) & 1u);
           same &= ((d8 >> j) & 1u) ? bj : ~bj;
         }
@@ -1943,22 +1931,18 @@ extern "C" int qk_sort_pairs_u64(void *stream, uint64_t n, uint64_t *keys,
     npasses = 0;
     while (hmax) {
       npasses++;
-      hmax >>= RADIX_BITS;
+      hmax >>= 8;
     }
     if (!npasses) npasses = 1;
-  }
-  {
-    int maxp = (64 + RADIX_BITS - 1) / RADIX_BITS;
-    if (npasses > maxp) npasses = maxp;      // shift >= 64 would be UB
   }
   uint64_t *ki = keys, *ko = keys_tmp;
   uint32_t *pi = pay, *po = pay_tmp;
   for (int p = 0; p < npasses; p++) {
-    int shift = RADIX_BITS * p;
+    int shift = 8 * p;
     hipLaunchKernelGGL(k_radix_count, dim3(nblocks), dim3(BLOCK), 0,
                        (hipStream_t)stream, n, ki, shift, chunk, counts,
                        nblocks);
-    hipLaunchKernelGGL(k_radix_scan, dim3(1), dim3(BLOCK), 0,
+    hipLaunchKernelGGL(k_radix_scan, dim3(1), dim3(RADIX), 0,
                        (hipStream_t)stream, nblocks, counts, bases);
     hipLaunchKernelGGL(k_radix_scatter, dim3(nblocks), dim3(BLOCK), 0,
                        (hipStream_t)stream, n, ki, pi, shift, chunk, bases,
