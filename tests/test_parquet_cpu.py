@@ -233,3 +233,38 @@ def test_fuzz_random_files():
             want = t.column(name).to_numpy()
             np.testing.assert_array_equal(
                 got, want, err_msg="case %d col %s kw %r" % (case, name, kw))
+
+
+def test_thrift_primitives_handcrafted():
+    """Compact-protocol primitives against handcrafted byte sequences
+    (pyarrow only ever exercises a writer-specific subset)."""
+    from quokka_amd.parquet_thrift import (_varint, _zigzag, _skip,
+                                           _read_struct)
+    assert _varint(b"\x00", 0) == (0, 1)
+    assert _varint(b"\x7f", 0) == (127, 1)
+    assert _varint(b"\x80\x01", 0) == (128, 2)
+    assert _varint(b"\xff\xff\x03", 0) == (0xFFFF, 3)
+    assert _zigzag(b"\x00", 0) == (0, 1)
+    assert _zigzag(b"\x01", 0) == (-1, 1)
+    assert _zigzag(b"\x02", 0) == (1, 1)
+    assert _zigzag(b"\x03", 0) == (-2, 1)
+    # skip: bool(no payload), byte, i32 varint, double, binary, list
+    assert _skip(b"", 0, 1) == 0
+    assert _skip(b"\x42", 0, 3) == 1
+    assert _skip(b"\x80\x01", 0, 5) == 2
+    assert _skip(b"\x00" * 8, 0, 7) == 8
+    assert _skip(b"\x03abc", 0, 8) == 4
+    assert _skip(b"\x25\x02\x04", 0, 9) == 3   # list: 2 elems of i32
+    # struct with field ids via delta and long-form, nested struct skip
+    # field 1 (i32 = 5), field 3 (struct{field 1 bool true}), stop
+    buf = bytes([0x15, 0x0A,                   # delta1 type5 -> zigzag 10
+                 0x2C, 0x11, 0x00,             # delta2 type12: {f1 booltrue}
+                 0x00])
+    out, pos = _read_struct(buf, 0, {1: "i", 3: {1: "i"}})
+    assert out == {1: 5, 3: {1: True}} and pos == len(buf)
+    # unknown field types are skipped without corrupting position
+    buf2 = bytes([0x17, 0, 0, 0, 0, 0, 0, 0, 0,   # f1 double, skipped
+                  0x15, 0x06,                      # f2 i32 = 3
+                  0x00])
+    out2, _ = _read_struct(buf2, 0, {2: "i"})
+    assert out2 == {2: 3}
