@@ -223,3 +223,18 @@ def test_like_on_dict_columns():
         raise AssertionError("LIKE accepted on non-dict column")
     except ValueError:
         pass
+
+
+def test_in_list_date_literals():
+    import numpy as np
+    from quokka_amd import jit
+    schema = {"d": np.dtype(np.int32)}
+    e, cols = jit.translate(
+        "d in (date '1994-01-01', date '1995-06-17')", schema)
+    d0 = (np.datetime64("1994-01-01") -
+          np.datetime64("1970-01-01")).astype(int)
+    d1 = (np.datetime64("1995-06-17") -
+          np.datetime64("1970-01-01")).astype(int)
+    v0 = np.array([d0, d1, d0 + 1, 0], dtype=np.int32)
+    got = eval(e.replace("||", "|"), {}, {"v0": v0})
+    assert np.array_equal(got, np.isin(v0, [d0, d1]))
