@@ -277,11 +277,15 @@ int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
 /* agg_ops_dev (device i32[nvals], nullable -> all SUM): per value column
  * 0 = SUM (slot init 0), 1 = MIN (init +inf), 2 = MAX (init -inf) — the
  * distributive ops the two-phase rewrite emits (sql_utils.py:299-413). */
+/* n_inserted_dev (u64, nullable): incremented once per NEW group claimed —
+ * the host reads it after each batch and rebuilds into a larger table
+ * before cumulative distinct keys approach capacity (the find-or-insert
+ * probe loop would otherwise spin forever on a full table). */
 int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
                        const double *const *vals_dev,
                        const int32_t *agg_ops_dev, int nvals,
                        int64_t *slot_keys, double *slot_sums,
-                       uint64_t capacity);
+                       uint64_t capacity, uint64_t *n_inserted_dev);
 int qk_fill_f64(void *stream, double *dst_dev, double value, uint64_t n);
 /* Compact occupied slots to out_keys/out_sums (unordered); out_cursor_dev
  * (u64, zeroed) = number of groups. out capacity must be >= group count. */

@@ -1625,7 +1625,7 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
     uint64_t n, const int64_t *__restrict__ keys, const double *const *vals,
     const int32_t *__restrict__ agg_ops, int nvals,
     int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
-    uint64_t cap) {
+    uint64_t cap, uint64_t *__restrict__ n_inserted) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -1638,7 +1638,14 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
         int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
                                           (unsigned long long)QK_JOIN_EMPTY,
                                           (unsigned long long)key);
-        if (prev == QK_JOIN_EMPTY || prev == key) break;
+        if (prev == QK_JOIN_EMPTY) {
+          // this lane claimed a fresh slot: count the new group so the
+          // host can grow the table before cumulative distinct keys
+          // approach capacity (a full table would spin this loop forever)
+          if (n_inserted) atomicAdd((unsigned long long *)n_inserted, 1ULL);
+          break;
+        }
+        if (prev == key) break;
       }
       s = (s + 1) & (cap - 1);
     }
@@ -1659,14 +1666,14 @@ extern "C" int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
                                   const double *const *vals_dev,
                                   const int32_t *agg_ops_dev, int nvals,
                                   int64_t *slot_keys, double *slot_sums,
-                                  uint64_t cap) {
+                                  uint64_t cap, uint64_t *n_inserted_dev) {
   if (!n) return 0;
   if (cap & (cap - 1))
     return qk_fail("qk_groupby_i64_sum.cap_pow2", hipErrorInvalidValue);
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_groupby_sum, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, keys, vals_dev, agg_ops_dev,
-                     nvals, slot_keys, slot_sums, cap);
+                     nvals, slot_keys, slot_sums, cap, n_inserted_dev);
   QK_TRY("qk_groupby_i64_sum", hipGetLastError());
   return 0;
 }

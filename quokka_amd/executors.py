@@ -152,6 +152,14 @@ class GPUBuildProbeJoinExecutor(Executor):
             g.free()
             return res
 
+        # the reference renames the kept key column after EVERY join emit
+        # (sql_executors.py:372-373), regardless of how/path
+        def rename_key(cols):
+            if self.key_to_keep == "right" and self.left_on != self.right_on \
+                    and self.left_on in cols:
+                cols[self.right_on] = cols.pop(self.left_on)
+            return cols
+
         out = {}
         if self.how in ("semi", "anti"):
             for c in batch.column_names:
@@ -176,8 +184,10 @@ class GPUBuildProbeJoinExecutor(Executor):
                     for c in batch.column_names:
                         out[c] = np.concatenate(
                             [out[c], staging.column_to_numpy(batch.column(c))[un]])
-                    tbl = _to_table({k: v for k, v in out.items()
-                                     if k in batch.column_names})
+                    probe_cols = rename_key(
+                        {k: v for k, v in out.items()
+                         if k in batch.column_names})
+                    tbl = _to_table(probe_cols)
                     pay = {}
                     for c in self._build_payload_dev:
                         pay[c] = pa.chunked_array([
@@ -191,8 +201,7 @@ class GPUBuildProbeJoinExecutor(Executor):
                         bidx.free()
                     kcol.free()
                     return tbl
-        if self.key_to_keep == "right" and self.how in ("inner", "left"):
-            out[self.right_on] = out.pop(self.left_on)
+        rename_key(out)
         pidx.free()
         if bidx:
             bidx.free()
@@ -297,6 +306,10 @@ class GPUAggExecutor(Executor):
             return arrs[0]
         # composite: each non-i64 key coded via np codebook, packed base-N
         self._key_state["mode"] = "composite"
+        if len(self.groupby_keys) > 3:
+            raise ValueError(
+                "composite group keys: at most 3 keys fit 63 bits at 21 "
+                "bits/key (got %d)" % len(self.groupby_keys))
         packed = np.zeros(len(arrs[0]), dtype=np.int64)
         widths = []
         comps = []
@@ -307,6 +320,11 @@ class GPUAggExecutor(Executor):
             for i, v in enumerate(vals):
                 key = v.item() if hasattr(v, "item") else v
                 if key not in cb:
+                    if len(cb) >= (1 << 21):
+                        raise ValueError(
+                            "composite group key %r exceeds 2**21 distinct "
+                            "values; exceeding the 21-bit field would "
+                            "silently merge distinct groups" % k)
                     cb[key] = len(cb)
                 codes[i] = cb[key]
             comps.append(codes[inv])
@@ -377,8 +395,16 @@ class GPUAggExecutor(Executor):
             cols = []
             for item in reversed(self.orderby_keys):
                 k, d = item if isinstance(item, (tuple, list)) else (item, "asc")
-                v = out[k]
-                cols.append(-v if d == "desc" else v)
+                v = np.asarray(out[k])
+                if d == "desc":
+                    if v.dtype.kind in "biuf":
+                        v = -v
+                    else:
+                        # non-numeric desc (e.g. decoded string group keys):
+                        # invert factorized rank instead of numeric negation
+                        _, inv = np.unique(v, return_inverse=True)
+                        v = inv.max() - inv
+                cols.append(v)
             order = np.lexsort(cols)
             out = {k: np.asarray(v)[order] for k, v in out.items()}
         self._gb.free()

@@ -354,9 +354,16 @@ class Translator:
                             raise ValueError(
                                 "string IN list on %r needs its "
                                 "StringDict" % t.val)
-                        codes.append(sd.codes.get(el, 255))
+                        # absent literal: no code can ever equal it — drop
+                        # it (a sentinel code could collide with a later
+                        # legitimate dictionary entry)
+                        c = sd.codes.get(el)
+                        if c is not None:
+                            codes.append(c)
                     else:
                         codes.append(int(el))
+                if not codes:
+                    raise _Folded("(1)" if neg else "(0)")
                 body = " || ".join("((%s) == (%d))" % (ref, c)
                                    for c in codes)
                 raise _Folded(("!(%s)" if neg else "(%s)") % body)
@@ -374,7 +381,10 @@ class Translator:
                             "StringDict" % t.val)
                     code = sd.codes.get(p.val)
                     if code is None:
-                        code = 255  # absent value: never-matching code
+                        # absent value: constant false for =, true for != —
+                        # never a sentinel code (it could collide with a
+                        # later legitimate dictionary entry)
+                        raise _Folded("(1)" if op in ("!=", "<>") else "(0)")
                     cop = {"=": "==", "<>": "!="}.get(op, op)
                     raise _Folded("(%s) %s (%d)" % (ref, cop, code))
                 self.i = save

@@ -183,12 +183,20 @@ class GroupByI64:
         if any(self.agg_ops):
             self._ops_dev = DevColumn.from_numpy(
                 np.asarray(self.agg_ops, dtype=np.int32))
+        self.n_groups = 0           # exact, from the device insert counter
+        self._nins = _count_buf()
 
     def update(self, keys_col, val_cols, n=None):
         n = keys_col.n if n is None else n
         if not n:
             return
         assert len(val_cols) == self.nvals
+        # cumulative distinct keys can exceed the first batch's sizing: a
+        # full table makes the kernel's find-or-insert loop spin forever,
+        # so grow FIRST whenever this batch could push the exact group
+        # count (device insert counter) past half capacity
+        if 2 * (self.n_groups + n) > self.cap:
+            self._grow(self.n_groups + n)
         sh = self.stream.handle if self.stream else None
         ptrs = np.array([c.ptr.value if hasattr(c.ptr, "value") else c.ptr
                          for c in val_cols], dtype=np.uint64)
@@ -198,10 +206,42 @@ class GroupByI64:
         shim.call("qk_groupby_i64_sum", sh, c_u64(n), keys_col.ptr,
                   dptrs.ptr, self._ops_dev.ptr if self._ops_dev else None,
                   self.nvals, self.slot_keys.ptr,
-                  self.slot_sums.ptr, c_u64(self.cap))
+                  self.slot_sums.ptr, c_u64(self.cap), self._nins.ptr)
         if self.stream:
             self.stream.sync()
+        self.n_groups = _read_u64(self._nins)
         dptrs.free()
+
+    def _grow(self, need_groups):
+        """Rebuild into a larger table (>= 4x the needed group count):
+        extract the accumulated (keys, sums) and re-insert them — correct
+        for SUM (adds into the zero identity) and MIN/MAX (single value vs
+        the op identity) alike. Mirrors GPUDistinctExecutor._grow."""
+        keys, sums = self.extract()
+        old_groups = len(keys)
+        for c in (self.slot_keys, self.slot_sums):
+            c.free()
+        self.cap = _pow2_at_least(max(16, 4 * int(need_groups)))
+        self.slot_keys = DevColumn(np.int64, self.cap)
+        self.slot_sums = DevColumn(np.float64, self.cap * self.nvals)
+        sh = self.stream.handle if self.stream else None
+        shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
+                  c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
+        import ctypes as _ct
+        for c, op in enumerate(self.agg_ops):
+            shim.call("qk_fill_f64", sh,
+                      shim.c_vp(self.slot_sums.ptr.value + c * self.cap * 8),
+                      _ct.c_double(self.AGG_INIT[op]), c_u64(self.cap))
+        shim.call("qk_dmemset", self._nins.ptr, 0, c_u64(8))
+        self.n_groups = 0
+        if old_groups:
+            kcol = DevColumn.from_numpy(keys)
+            vcols = [DevColumn.from_numpy(np.ascontiguousarray(sums[i]))
+                     for i in range(self.nvals)]
+            self.update(kcol, vcols, old_groups)
+            kcol.free()
+            for v in vcols:
+                v.free()
 
     def extract(self):
         """-> (keys np.int64[K], sums np.float64[nvals, K]); unordered."""
@@ -227,6 +267,7 @@ class GroupByI64:
     def free(self):
         self.slot_keys.free()
         self.slot_sums.free()
+        self._nins.free()
         if self._ops_dev is not None:
             self._ops_dev.free()
 

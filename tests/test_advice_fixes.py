@@ -1,0 +1,206 @@
+"""Tests for the round-1 advisor findings:
+
+1. (high) GroupByI64 grows when cumulative distinct keys across batches
+   exceed the first batch's sizing (the kernel's find-or-insert loop would
+   otherwise spin forever on a full table).
+2. (medium) join key rename (key_to_keep='right') applied on EVERY emitted
+   table — including the left-join unmatched path and semi/anti — matching
+   the reference's unconditional rename (sql_executors.py:372-373).
+3. (medium) composite group-key packing bounds: >3 keys or a 2**21-entry
+   codebook raise instead of silently merging distinct groups.
+4. (low) absent string literals in JIT predicates fold to constant
+   false/true, never to a sentinel code that a later dictionary entry
+   could legitimately take (code 255 is a valid StringDict code).
+5. (low) GPUAggExecutor order-by desc works on non-numeric (decoded
+   string) key columns.
+"""
+import numpy as np
+import pytest
+
+from quokka_amd.executors import GPUAggExecutor, GPUBuildProbeJoinExecutor
+
+
+# ---------- CPU-side checks ---------------------------------------------
+
+def test_composite_key_too_many_keys_raises():
+    import pyarrow as pa
+    ex = GPUAggExecutor(["a", "b", "c", "d"], [], "sum(x) as s")
+    t = pa.table({"a": ["u"], "b": ["v"], "c": ["w"], "d": ["y"],
+                  "x": [1.0]})
+    with pytest.raises(ValueError, match="at most 3 keys"):
+        ex._encode_keys(t)
+
+
+def test_composite_key_codebook_overflow_raises():
+    import pyarrow as pa
+    ex = GPUAggExecutor(["a", "b"], [], "sum(x) as s")
+    t = pa.table({"a": ["u", "v"], "b": ["p", "q"], "x": [1.0, 2.0]})
+    ex._encode_keys(t)   # initialise state
+    # simulate a codebook that has already hit the 21-bit field limit
+    # (keys offset so the incoming StringDict code is NOT already present)
+    ex._key_state["codebooks"]["a"] = {i + 1000: i for i in range(1 << 21)}
+    t2 = pa.table({"a": ["brand-new"], "b": ["p"], "x": [3.0]})
+    with pytest.raises(ValueError, match="2\\*\\*21"):
+        ex._encode_keys(t2)
+
+
+def test_jit_absent_string_literal_folds_constant():
+    """A string literal absent from the StringDict can never match — and
+    code 255 is a legitimate dictionary code (StringDict holds up to 256
+    values), so the fold must be a constant, not a sentinel compare."""
+    from quokka_amd import jit
+    from quokka_amd.staging import StringDict
+
+    sd = StringDict()
+    # fill all 256 codes; the value at code 255 is "v255"
+    sd.encode(np.array(["v%d" % i for i in range(256)]))
+    assert sd.codes["v255"] == 255
+    schema = {"m": np.dtype(np.uint8)}
+
+    e, _ = jit.translate("m = 'NOT-IN-DICT'", schema, {"m": sd})
+    assert "255" not in e and "(0)" in e
+    e, _ = jit.translate("m != 'NOT-IN-DICT'", schema, {"m": sd})
+    assert "255" not in e and "(1)" in e
+    # IN list: absent members dropped, present ones kept
+    e, _ = jit.translate("m in ('NOT-IN-DICT', 'v3')", schema, {"m": sd})
+    assert "== (3)" in e and "255" not in e
+    e, _ = jit.translate("m in ('NOT-IN-DICT', 'NOPE')", schema, {"m": sd})
+    assert e.strip() in ("(0)",)
+    e, _ = jit.translate("m not in ('NOT-IN-DICT', 'NOPE')", schema,
+                         {"m": sd})
+    assert e.strip() in ("(1)",)
+    # and the legitimate code-255 value still matches exactly
+    e, _ = jit.translate("m = 'v255'", schema, {"m": sd})
+    assert "== (255)" in e or "(255)" in e
+
+
+# ---------- GPU checks --------------------------------------------------
+
+@pytest.fixture(scope="module")
+def gpu():
+    from quokka_amd import shim
+    shim.init(0)
+    return shim
+
+
+@pytest.mark.gpu
+def test_groupby_grows_across_batches(gpu):
+    """Cumulative distinct keys 64x the initial sizing; sums == numpy."""
+    from quokka_amd import ops, shim
+    rng = np.random.default_rng(0)
+    gb = ops.GroupByI64(expected_groups=16, nvals=1)
+    want = {}
+    for b in range(8):
+        keys = rng.integers(0, 50_000, size=4096).astype(np.int64)
+        vals = rng.random(4096)
+        for k, v in zip(keys, vals):
+            want[k] = want.get(k, 0.0) + v
+        kc = shim.DevColumn.from_numpy(keys)
+        vc = shim.DevColumn.from_numpy(vals)
+        gb.update(kc, [vc])
+        kc.free(); vc.free()
+    keys_out, sums_out = gb.extract()
+    assert gb.n_groups == len(want) == len(keys_out)
+    got = dict(zip(keys_out.tolist(), sums_out[0].tolist()))
+    for k, v in want.items():
+        assert abs(got[k] - v) <= 1e-9 * max(1.0, abs(v)), k
+    gb.free()
+
+
+@pytest.mark.gpu
+def test_groupby_grows_min_max(gpu):
+    """Growth preserves MIN/MAX partials (rebuild re-inserts one value per
+    group against the op identity)."""
+    from quokka_amd import ops, shim
+    rng = np.random.default_rng(1)
+    gb = ops.GroupByI64(expected_groups=4, nvals=3, agg_ops=[0, 1, 2])
+    wsum, wmin, wmax = {}, {}, {}
+    for b in range(6):
+        keys = rng.integers(0, 3000, size=2048).astype(np.int64)
+        vals = rng.standard_normal(2048)
+        for k, v in zip(keys, vals):
+            wsum[k] = wsum.get(k, 0.0) + v
+            wmin[k] = min(wmin.get(k, np.inf), v)
+            wmax[k] = max(wmax.get(k, -np.inf), v)
+        kc = shim.DevColumn.from_numpy(keys)
+        vc = shim.DevColumn.from_numpy(vals)
+        gb.update(kc, [vc, vc, vc])
+        kc.free(); vc.free()
+    keys_out, sums_out = gb.extract()
+    assert gb.n_groups == len(wsum) == len(keys_out)
+    for i, k in enumerate(keys_out.tolist()):
+        np.testing.assert_allclose(sums_out[0][i], wsum[k], rtol=1e-9)
+        assert sums_out[1][i] == wmin[k]
+        assert sums_out[2][i] == wmax[k]
+    gb.free()
+
+
+@pytest.mark.gpu
+def test_executor_agg_cardinality_beyond_first_batch(gpu):
+    """GPUAggExecutor sized from a small first batch must survive a large
+    second batch (the advisor's GPU-hang scenario)."""
+    import pyarrow as pa
+    ex = GPUAggExecutor(["k"], [], "sum(v) as s")
+    b1 = pa.table({"k": np.arange(10, dtype=np.int64),
+                   "v": np.ones(10)})
+    n2 = 200_000
+    b2 = pa.table({"k": np.arange(n2, dtype=np.int64),
+                   "v": np.full(n2, 2.0)})
+    ex.execute([b1], 0, 0)
+    ex.execute([b2], 0, 0)
+    out = ex.done(0)
+    assert out.num_rows == n2
+    d = dict(zip(out.column("k").to_pylist(), out.column("s").to_pylist()))
+    assert d[5] == 3.0 and d[n2 - 1] == 2.0
+
+
+@pytest.mark.gpu
+def test_join_rename_every_path(gpu):
+    """key_to_keep='right' renames on every emitted table: inner, the
+    left-join unmatched path, and semi/anti."""
+    import pyarrow as pa
+
+    def build_probe(how, probe_keys):
+        ex = GPUBuildProbeJoinExecutor(left_on="lk", right_on="rk", how=how,
+                                       key_to_keep="right")
+        build = pa.table({"rk": np.array([1, 2, 3], dtype=np.int64),
+                          "pay": np.array([10., 20., 30.])})
+        ex.execute([build], 1, 0)
+        probe = pa.table({"lk": np.asarray(probe_keys, dtype=np.int64),
+                          "x": np.arange(len(probe_keys), dtype=np.float64)})
+        return ex.execute([probe], 0, 0)
+
+    t = build_probe("inner", [1, 2, 9])
+    assert "rk" in t.column_names and "lk" not in t.column_names
+    # left join WITH unmatched rows (the path that skipped the rename)
+    t = build_probe("left", [1, 9, 9])
+    assert "rk" in t.column_names and "lk" not in t.column_names
+    assert t.num_rows == 3
+    # left join with no unmatched rows (late path)
+    t = build_probe("left", [1, 2])
+    assert "rk" in t.column_names and "lk" not in t.column_names
+    for how in ("semi", "anti"):
+        t = build_probe(how, [1, 9])
+        assert "rk" in t.column_names and "lk" not in t.column_names, how
+        assert t.num_rows == 1
+
+
+@pytest.mark.gpu
+def test_agg_orderby_desc_string_key(gpu):
+    """order-by desc on a decoded string group key (numeric negation would
+    raise TypeError)."""
+    import pyarrow as pa
+    ex = GPUAggExecutor(["name", "grp"],
+                        [("name", "desc"), ("s", "asc")],
+                        "sum(v) as s")
+    t = pa.table({"name": ["b", "a", "c", "a"],
+                  "grp": ["x", "x", "y", "y"],
+                  "v": [1.0, 2.0, 3.0, 4.0]})
+    ex.execute([t], 0, 0)
+    out = ex.done(0)
+    names = out.column("name").to_pylist()
+    assert names == sorted(names, reverse=True)
+    d = {(n, g): s for n, g, s in zip(names, out.column("grp").to_pylist(),
+                                      out.column("s").to_pylist())}
+    assert d[("a", "x")] == 2.0 and d[("a", "y")] == 4.0
+    assert d[("b", "x")] == 1.0 and d[("c", "y")] == 3.0
