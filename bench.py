@@ -49,10 +49,19 @@ def parse_args():
     p.add_argument("--exchange", choices=["auto", "none", "rccl"],
                    default="auto",
                    help="q3 multi-rank repartition (auto: rccl when world>1)")
+    p.add_argument("--chunks", type=int, default=4,
+                   help="exchange pipeline depth: probe chunk j overlaps "
+                        "the RCCL transfer of chunk j+1 (north_star)")
+    p.add_argument("--no-overlap", action="store_true",
+                   help="A/B: sequential exchange -> probe (round-1 path)")
     p.add_argument("--nt", type=int, default=1,
                    help="non-temporal probe loads (Q3/Q5)")
     p.add_argument("--cpu-sample-rows", type=int, default=12_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--skip-subbench", action="store_true",
+                   help="q1 default run: skip the folded q3/q6/e2e legs")
+    p.add_argument("--e2e-sf", type=float, default=10.0,
+                   help="scale factor of the end-to-end Parquet leg")
     return p.parse_args()
 
 
@@ -215,12 +224,12 @@ def cpu_baseline_q3(sample_rows, target_seconds=12.0):
     }
 
 
-def read_traffic(sf=None):
+def read_traffic(sf=None, query=None):
     """Per-launch HBM bytes from the committed rocprofv3 PMC measurement
     (profiles/traffic_<q>.json), or None before one exists. The PMC run
     was taken at the measured SF recorded in the file; at any other
     --sf the figure would be wrong, so return None instead."""
-    path = os.path.join(ROOT, "profiles", "traffic_%s.json" % QUERY)
+    path = os.path.join(ROOT, "profiles", "traffic_%s.json" % (query or QUERY))
     if os.path.exists(path):
         with open(path) as f:
             d = json.load(f)
@@ -234,7 +243,95 @@ def read_traffic(sf=None):
 QUERY = "q1"
 
 
-def main_q3(args, n, world, rank, dist, shim, DQ):
+def run_e2e(args, shim, DQ):
+    """End-to-end leg: Parquet file ON DISK -> host metadata parse +
+    file-bytes upload -> GPU decode (qk_pq_* kernels) -> fused Q1 ->
+    result. The reference's hot loop includes the scan
+    (unordered_readers.py:42-99); this is the number that exposes the
+    PCIe-inclusive rate next to the HBM-resident headline.
+
+    The file is WRITTEN untimed (synthetic lineitem, flags as
+    dictionary-encoded strings — the realistic shape); each timed pass
+    re-reads it from the page cache, so the measured path is
+    host RAM -> HBM -> kernels, like the reference's warm scan."""
+    import os
+    import tempfile
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from quokka_amd import parquet_gpu
+    from quokka_amd.shim import DevColumn, c_u64
+
+    n = int(round(args.e2e_sf / 100.0 * SF100_LINEITEM_ROWS)) & ~3
+    cols = gen_device_lineitem(shim, n, 0)
+    host = {k: c.to_numpy() for k, c in cols.items()}
+    for c in cols.values():
+        c.free()
+    RET = np.array(["A", "N", "R"])
+    LIN = np.array(["F", "O"])
+    t = pa.table({
+        "l_quantity": host["l_quantity"],
+        "l_extendedprice": host["l_extendedprice"],
+        "l_discount": host["l_discount"],
+        "l_tax": host["l_tax"],
+        "l_shipdate": pa.array(host["l_shipdate"], type=pa.int32()),
+        "l_returnflag": pa.array(RET[host["l_returnflag"]]).dictionary_encode(),
+        "l_linestatus": pa.array(LIN[host["l_linestatus"]]).dictionary_encode(),
+    })
+    del host
+    tmpdir = tempfile.mkdtemp(prefix="qk_e2e_")
+    path = os.path.join(tmpdir, "lineitem.parquet")
+    pq.write_table(t, path, compression="NONE", data_page_version="1.0",
+                   use_dictionary=["l_returnflag", "l_linestatus"])
+    file_bytes = os.path.getsize(path)
+    del t
+
+    def one_pass():
+        dec = parquet_gpu.read_table(path)
+        # dict flag columns -> canonical u8 codes via a 3-entry device
+        # gather (values order in the file is writer-dependent)
+        def canon(name, order):
+            codes_u32, values = dec[name]
+            m = np.zeros(max(1, len(values)), dtype=np.uint8)
+            for i, v in enumerate(values):
+                m[i] = order.index(v)
+            mcol = DevColumn.from_numpy(m)
+            out = DevColumn(np.uint8, codes_u32.n)
+            shim.call("qk_gather_u8", None, c_u64(codes_u32.n),
+                      codes_u32.ptr, mcol.ptr, out.ptr)
+            out.n = codes_u32.n
+            mcol.free()
+            codes_u32.free()
+            return out
+        dec["l_returnflag"] = canon("l_returnflag", ["A", "N", "R"])
+        dec["l_linestatus"] = canon("l_linestatus", ["F", "O"])
+        res = DQ.q1(dec)
+        for c in dec.values():
+            c.free()
+        return res
+
+    res = one_pass()                       # warm (page cache + pool)
+    passes = 3
+    t0 = time.time()
+    for _ in range(passes):
+        res = one_pass()
+    dt = (time.time() - t0) / passes
+    os.unlink(path)
+    os.rmdir(tmpdir)
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "ms_per_pass": dt * 1e3,
+        "file_gb_per_s": file_bytes / dt / 1e9,
+        "file_bytes": file_bytes,
+        "rows": n,
+        "sf": args.e2e_sf,
+        "q1_result_rows": len(res["count_order"]),
+        "path": "parquet on disk (page-cache warm) -> upload -> "
+                "qk_pq_* GPU decode -> fused Q1",
+    }
+
+
+def main_q3(args, n, world, rank, dist, shim, DQ, standalone=True):
     """TPC-H Q3 on the fused device path.
 
     world == 1 (BASELINE.json configs[2]): a step = rebuild customer+orders
@@ -253,6 +350,7 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
 
     use_exchange = args.exchange == "rccl" or (args.exchange == "auto"
                                                and world > 1)
+    overlap = use_exchange and not args.no_overlap
     if use_exchange:
         n_local = (n // world) & ~3
     else:
@@ -260,6 +358,7 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
     li, od, cu = gen_device_q3_tables(shim, n_local, rank, world,
                                       n_total=n if use_exchange else None)
     stream = shim.Stream()
+    comm_stream = shim.Stream() if use_exchange else None
     comm = exchange.Comm(rank, world, dist) if use_exchange else None
 
     def do_exchange():
@@ -274,6 +373,23 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
             {k: v for k, v in od.items() if k != "o_orderkey"}, stream)
         od2 = {"o_orderkey": ok, **op}
         return li2, od2
+
+    def do_step_overlapped(fused_obj):
+        """One exchange step with the lineitem RCCL repartition OVERLAPPED
+        against the fused probe, chunk by chunk, on comm_stream vs stream
+        (exchange.repartition_overlapped — the north_star side-stream
+        overlap; replaces the sequential round-1 exchange->probe)."""
+        def consume(views, start, nrows, j):
+            fused_obj.probe({
+                "l_orderkey": views["__key__"],
+                "l_shipdate": views["l_shipdate"],
+                "l_extendedprice": views["l_extendedprice"],
+                "l_discount": views["l_discount"]}, nt=bool(args.nt))
+        rk, rp, _, _ = exchange.repartition_overlapped(
+            comm, li["l_orderkey"],
+            {k: v for k, v in li.items() if k != "l_orderkey"},
+            stream, comm_stream, consume, nchunks=max(1, args.chunks))
+        return {"l_orderkey": rk, **rp}
 
     if use_exchange:
         li_x, od_x = do_exchange()
@@ -305,16 +421,35 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
         if use_exchange:
             for c in list(li_x.values()) + list(od_x.values()):
                 c.free()
-            li_x, od_x = do_exchange()
+            # orders (build side) must be fully received before any probe:
+            # exchange + rebuild first, then the lineitem repartition
+            # overlapped with the probe chunk pipeline
+            ok, op = exchange.repartition(
+                comm, od["o_orderkey"],
+                {k: v for k, v in od.items() if k != "o_orderkey"}, stream)
+            od_x = {"o_orderkey": ok, **op}
             fused.free()
             fused = DQ.Q3Fused(od_x, cu, stream)
+            if timed:
+                timer.start(stream)
+            if overlap:
+                li_x = do_step_overlapped(fused)
+            else:
+                rk, rp = exchange.repartition(
+                    comm, li["l_orderkey"],
+                    {k: v for k, v in li.items() if k != "l_orderkey"},
+                    stream)
+                li_x = {"l_orderkey": rk, **rp}
+                fused.probe(li_x, nt=bool(args.nt))
+            if timed:
+                timer.stop(stream)
         else:
             fused.rebuild()
-        if timed:
-            timer.start(stream)
-        fused.probe(li_x, nt=bool(args.nt))
-        if timed:
-            timer.stop(stream)
+            if timed:
+                timer.start(stream)
+            fused.probe(li_x, nt=bool(args.nt))
+            if timed:
+                timer.stop(stream)
         n_groups, top10 = fused.extract_top10(10)
         if dist is not None:
             # per-rank orderkey partitions are disjoint: merge top-10s
@@ -393,6 +528,9 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
                 "rows_per_gpu": n_local,
                 "query": "Q3",
                 "exchange": "rccl" if use_exchange else "none",
+                "overlap": (bool(overlap) if use_exchange else None),
+                "exchange_chunks": (max(1, args.chunks) if overlap
+                                    else None),
                 "n_groups": int(n_groups),
                 "orders_build_rows": int(fused.n_build),
                 "lineitem_ship_pass": int(n_pass),
@@ -404,20 +542,38 @@ def main_q3(args, n, world, rank, dist, shim, DQ):
                 "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s",
                 "frac": achieved_gbps / HBM_PEAK_GBPS,
-                "traffic": read_traffic(args.sf),
+                "traffic": read_traffic(args.sf, "q3"),
+                # in exchange mode the HIP-event window covers the
+                # OVERLAPPED exchange+probe pipeline, not the bare probe
+                # kernel: `achieved` is then a lower bound on kernel rate
+                "window": ("exchange+probe overlapped" if use_exchange and
+                           overlap else
+                           "exchange+probe sequential" if use_exchange
+                           else "probe kernel"),
             },
             "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
                              else cpu_baseline_q3(args.cpu_sample_rows)),
         }
-        print(json.dumps(out))
+        if standalone:
+            print(json.dumps(out))
+    else:
+        out = None
     timer.destroy()
     fused.free()
+    # non-exchange mode aliases li_x/od_x to the base tables (free() is
+    # idempotent, so the union below is safe either way)
+    for c in (list(li.values()) + list(od.values()) + list(cu.values()) +
+              list(li_x.values()) + list(od_x.values())):
+        c.free()
+    if comm_stream is not None:
+        comm_stream.destroy()
+    if comm is not None:
+        comm.destroy()
     stream.destroy()
-    if dist is not None:
-        dist.destroy_process_group()
+    return out
 
 
-def main_q5(args, n, world, rank, dist, shim, DQ):
+def main_q5(args, n, world, rank, dist, shim, DQ, standalone=True):
     """TPC-H Q5 on the fused device path (BASELINE.json configs[4] query
     shape; single-node). A step = rebuild the three key->nation tables +
     fused probe + per-nation result; weak scaling like Q3."""
@@ -511,27 +667,35 @@ def main_q5(args, n, world, rank, dist, shim, DQ):
                 "bound": "hbm", "achieved": achieved_gbps,
                 "peak": HBM_PEAK_GBPS, "unit": "GB/s",
                 "frac": achieved_gbps / HBM_PEAK_GBPS,
-                "traffic": read_traffic(args.sf),
+                "traffic": read_traffic(args.sf, "q5"),
             },
             "cpu_baseline": (None if args.skip_cpu_baseline or world > 1
                              else cpu_baseline_q5(args.cpu_sample_rows)),
         }
-        print(json.dumps(out))
+        if standalone:
+            print(json.dumps(out))
+    else:
+        out = None
     timer.destroy()
     fused.free()
+    for c in (list(li.values()) + list(od.values()) + list(cu.values()) +
+              list(su.values())):
+        c.free()
     stream.destroy()
-    if dist is not None:
-        dist.destroy_process_group()
+    return out
 
 
-def main_q6(args, n, world, rank, dist, shim, DQ):
+def main_q6(args, n, world, rank, dist, shim, DQ, cols=None,
+            standalone=True):
     """TPC-H Q6 entirely through the hiprtc JIT: the predicate AND the
     aggregate are runtime-compiled from the reference's SQL strings
     (tpch_ref.py:171-183) — no hand-written kernel on this path."""
     from quokka_amd import jit, ops
     import numpy as _np
 
-    cols = gen_device_lineitem(shim, n, rank)
+    own_cols = cols is None
+    if own_cols:
+        cols = gen_device_lineitem(shim, n, rank)
     schema = {k: v.dtype for k, v in cols.items()}
     agg = jit.JitAggregate(
         schema, group_keys=[],
@@ -595,11 +759,18 @@ def main_q6(args, n, world, rank, dist, shim, DQ):
                          "traffic": None},
             "cpu_baseline": None,
         }
-        print(json.dumps(out))
+        if standalone:
+            print(json.dumps(out))
+    else:
+        out = None
     acc.free()
     agg.free()
+    if own_cols:
+        for c in cols.values():
+            c.free()
     timer.destroy()
     stream.destroy()
+    return out
 
 
 def main():
@@ -624,12 +795,12 @@ def main():
     QUERY = args.query
     n = int(round(args.sf / 100.0 * SF100_LINEITEM_ROWS))
     n &= ~3  # multiple of 4 -> vectorized Q1 path, 4 lines/order for Q3
-    if args.query == "q3":
-        return main_q3(args, n, world, rank, dist, shim, DQ)
-    if args.query == "q5":
-        return main_q5(args, n, world, rank, dist, shim, DQ)
-    if args.query == "q6":
-        return main_q6(args, n, world, rank, dist, shim, DQ)
+    if args.query in ("q3", "q5", "q6"):
+        fn = {"q3": main_q3, "q5": main_q5, "q6": main_q6}[args.query]
+        fn(args, n, world, rank, dist, shim, DQ)
+        if dist is not None:
+            dist.destroy_process_group()
+        return
     cols = gen_device_lineitem(shim, n, rank)
     stream = shim.Stream()
     timer = shim.Timer()
@@ -703,6 +874,32 @@ def main():
             print("# verify ok: q1 group-count sum == independent filter "
                   "count (%d)" % n_pass_indep, flush=True)
 
+    # folded sub-measurements (driver-timed evidence for the other graded
+    # configs in the SAME record — VERDICT r01 item 3): Q6 reuses the
+    # resident Q1 columns; Q3 (and the RCCL exchange at world>1) and the
+    # end-to-end Parquet leg run after the Q1 columns are freed.
+    subs = {}
+    if args.query == "q1" and not args.skip_subbench:
+        import argparse as _ap
+        sub = _ap.Namespace(**vars(args))
+        sub.steps = max(3, min(10, args.steps))
+        sub.warmup = max(1, min(3, args.warmup))
+        sub.cpu_sample_rows = min(args.cpu_sample_rows, 6_000_000)
+        r = main_q6(sub, n, world, rank, dist, shim, DQ, cols=cols,
+                    standalone=False)
+        if r:
+            subs["q6"] = r
+        for c in cols.values():
+            c.free()
+        r = main_q3(sub, n, world, rank, dist, shim, DQ, standalone=False)
+        if r:
+            subs["q3"] = r
+        if world == 1:
+            try:
+                subs["e2e"] = run_e2e(sub, shim, DQ)
+            except Exception as e:  # the headline line must still print
+                subs["e2e"] = {"error": "%s: %s" % (type(e).__name__, e)}
+
     if rank == 0:
         total_rows = n * world * args.steps
         avg_kernel_s = float(np.mean(kernel_ms)) / 1e3
@@ -741,6 +938,10 @@ def main():
                              else cpu_baseline(args.cpu_sample_rows)),
             "q1_result_rows": len(result["count_order"]) if result else 0,
         }
+        # folded legs: Q6 (JIT scan), Q3 (join+group-by, the north_star
+        # core; RCCL-exchanged and overlapped at world>1), e2e Parquet —
+        # each with its own HIP-event roofline, same record
+        out.update(subs)
         print(json.dumps(out))
 
     timer.destroy()

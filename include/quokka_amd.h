@@ -58,6 +58,15 @@ int qk_fill_i64(void *stream, int64_t *dst_dev, int64_t value, uint64_t n);
 int qk_stream_create(void **stream);
 int qk_stream_destroy(void *stream);
 int qk_stream_sync(void *stream);
+/* Bare (non-timing) HIP events for cross-stream ordering: the overlapped
+ * shuffle records an event on the comm stream after each received chunk and
+ * the compute stream waits on it before probing that chunk (north_star:
+ * RCCL repartition overlapped with probe on a side HIP stream; replaces the
+ * reference's 8-thread Flight do_put pool overlap, core.py:324-371). */
+int qk_event_create(void **ev);
+int qk_event_destroy(void *ev);
+int qk_event_record(void *ev, void *stream);
+int qk_stream_wait_event(void *stream, void *ev);
 /* HIP-event timer pair on the launching stream (roofline measurement). */
 int qk_timer_create(void **timer);
 int qk_timer_destroy(void *timer);

@@ -117,6 +117,30 @@ extern "C" int qk_stream_sync(void *stream) {
   return 0;
 }
 
+// ---- bare events: cross-stream ordering for the overlapped exchange ----
+// (the comm stream records an event after each received chunk; the compute
+// stream waits on it before probing that chunk — north_star's "shuffle
+// overlapped with probe on a side HIP stream")
+extern "C" int qk_event_create(void **ev) {
+  QK_TRY("qk_event_create",
+         hipEventCreateWithFlags((hipEvent_t *)ev, hipEventDisableTiming));
+  return 0;
+}
+extern "C" int qk_event_destroy(void *ev) {
+  QK_TRY("qk_event_destroy", hipEventDestroy((hipEvent_t)ev));
+  return 0;
+}
+extern "C" int qk_event_record(void *ev, void *stream) {
+  QK_TRY("qk_event_record",
+         hipEventRecord((hipEvent_t)ev, (hipStream_t)stream));
+  return 0;
+}
+extern "C" int qk_stream_wait_event(void *stream, void *ev) {
+  QK_TRY("qk_stream_wait_event",
+         hipStreamWaitEvent((hipStream_t)stream, (hipEvent_t)ev, 0));
+  return 0;
+}
+
 struct QkTimer {
   hipEvent_t start, stop;
 };

@@ -84,6 +84,21 @@ class Comm:
         sh = stream.handle if stream else None
         shim.call("qk_allreduce_f64", sh, self.handle, buf_ptr, c_u64(n))
 
+    def alltoallv_into(self, col, recv, so, sc, ro, rc, stream=None):
+        """alltoallv one column slice-set into a PREALLOCATED recv column
+        (the overlapped exchange reuses one recv allocation across chunks;
+        so/sc/ro/rc are per-peer element offsets/counts, uint64[world])."""
+        sh = stream.handle if stream else None
+        so = np.ascontiguousarray(so, dtype=np.uint64)
+        sc = np.ascontiguousarray(sc, dtype=np.uint64)
+        ro = np.ascontiguousarray(ro, dtype=np.uint64)
+        rc = np.ascontiguousarray(rc, dtype=np.uint64)
+        shim.call("qk_alltoallv", sh, self.handle, self.world,
+                  ctypes.c_uint32(col.dtype.itemsize), col.ptr,
+                  so.ctypes.data_as(c_vp), sc.ctypes.data_as(c_vp),
+                  recv.ptr, ro.ctypes.data_as(c_vp),
+                  rc.ctypes.data_as(c_vp))
+
 
 def repartition(comm, key_col, payload_cols, stream=None):
     """Hash-repartition rows by key % world across ranks.
@@ -119,3 +134,111 @@ def repartition(comm, key_col, payload_cols, stream=None):
         t.free()
     idx.free()
     return recv_key, recv_payload
+
+
+def plan_chunks(send_offsets, send_counts, recv_counts, nchunks):
+    """Deterministic chunk plan both sides of the exchange derive from the
+    SAME exchanged counts (keeping RCCL grouped send/recv matching aligned
+    across ranks, like the 256 MiB split inside qk_alltoallv).
+
+    Peer p's count c splits into nchunks pieces of size
+    c*(j+1)//nchunks - c*j//nchunks. The receive layout is CHUNK-MAJOR so
+    each received chunk is one contiguous row range (probe-able while the
+    next chunk is still in flight):
+        chunk j rows = [chunk_start[j], chunk_start[j] + chunk_rows[j])
+    Returns (per_chunk, chunk_start, chunk_rows) with per_chunk[j] =
+    (so_j, sc_j, ro_j, rc_j) uint64[world] element offsets/counts."""
+    world = len(send_counts)
+    sc = np.asarray(send_counts, dtype=np.uint64)
+    rc = np.asarray(recv_counts, dtype=np.uint64)
+    so_base = np.asarray(send_offsets[:world], dtype=np.uint64)
+
+    def split(c, j):
+        return c * (j + 1) // nchunks - c * j // nchunks
+
+    per_chunk = []
+    chunk_start = np.zeros(nchunks, dtype=np.uint64)
+    chunk_rows = np.zeros(nchunks, dtype=np.uint64)
+    sent_so_far = np.zeros(world, dtype=np.uint64)
+    recv_cursor = np.uint64(0)
+    for j in range(nchunks):
+        sc_j = np.array([split(int(sc[p]), j) for p in range(world)],
+                        dtype=np.uint64)
+        rc_j = np.array([split(int(rc[p]), j) for p in range(world)],
+                        dtype=np.uint64)
+        so_j = so_base + sent_so_far
+        ro_j = recv_cursor + np.concatenate(
+            ([np.uint64(0)], np.cumsum(rc_j[:-1])))
+        per_chunk.append((so_j, sc_j, ro_j.astype(np.uint64), rc_j))
+        chunk_start[j] = recv_cursor
+        chunk_rows[j] = rc_j.sum()
+        recv_cursor += chunk_rows[j]
+        sent_so_far += sc_j
+    return per_chunk, chunk_start, chunk_rows
+
+
+def repartition_overlapped(comm, key_col, payload_cols, comp_stream,
+                           comm_stream, consume=None, nchunks=4):
+    """Hash-repartition by key % world with the RCCL exchange OVERLAPPED
+    against `consume` (the probe) on the compute stream — the north_star
+    requirement the reference meets with its 8-thread Flight do_put pool
+    (core.py:324-371).
+
+    Pipeline (all enqueued, the GPU overlaps):
+      comp_stream:  partition -> per-column gather -> [record ready]
+                    [wait chunk j done] -> consume(chunk j) ...
+      comm_stream:  [wait ready] -> alltoallv chunk 0 -> [record done 0]
+                    -> alltoallv chunk 1 -> [record done 1] -> ...
+
+    consume(cols, start_row, n_rows, chunk_idx): cols maps the key column
+    name '__key__' and each payload name to a DevColumnView of that
+    received chunk. Returns (recv_key, recv_payload, chunk_start,
+    chunk_rows); both streams are synced and temporaries freed before
+    returning (one host sync per step, like `repartition`)."""
+    n = key_col.n
+    world = comm.world
+    offsets, idx = ops.partition_i64(key_col, world, comp_stream, n)
+    send_counts = np.diff(offsets).astype(np.uint64)
+    recv_counts = comm.exchange_counts(send_counts)
+    total_recv = int(recv_counts.sum())
+
+    named = [("__key__", key_col)] + list(payload_cols.items())
+    ordered = {}
+    for name, col in named:
+        ordered[name] = col.gather(idx, n, comp_stream)
+    ready = shim.Event()
+    ready.record(comp_stream)
+    ready.wait(comm_stream)
+
+    recvs = {name: DevColumn(col.dtype, max(1, total_recv))
+             for name, col in named}
+    for name, col in named:
+        recvs[name].n = total_recv
+    per_chunk, chunk_start, chunk_rows = plan_chunks(
+        offsets, send_counts, recv_counts, nchunks)
+
+    done_evs = [shim.Event() for _ in range(nchunks)]
+    for j, (so_j, sc_j, ro_j, rc_j) in enumerate(per_chunk):
+        for name, _ in named:
+            comm.alltoallv_into(ordered[name], recvs[name],
+                                so_j, sc_j, ro_j, rc_j, comm_stream)
+        done_evs[j].record(comm_stream)
+    if consume is not None:
+        for j in range(nchunks):
+            if not int(chunk_rows[j]):
+                continue
+            done_evs[j].wait(comp_stream)
+            views = {name: recvs[name].view(int(chunk_start[j]),
+                                            int(chunk_rows[j]))
+                     for name, _ in named}
+            consume(views, int(chunk_start[j]), int(chunk_rows[j]), j)
+    comp_stream.sync()
+    comm_stream.sync()
+    for ev in done_evs:
+        ev.destroy()
+    ready.destroy()
+    for t in ordered.values():
+        t.free()
+    idx.free()
+    recv_key = recvs.pop("__key__")
+    return recv_key, recvs, chunk_start, chunk_rows

@@ -47,6 +47,10 @@ _SIGS = {
     "qk_stream_create": [c_vp],
     "qk_stream_destroy": [c_vp],
     "qk_stream_sync": [c_vp],
+    "qk_event_create": [c_vp],
+    "qk_event_destroy": [c_vp],
+    "qk_event_record": [c_vp, c_vp],
+    "qk_stream_wait_event": [c_vp, c_vp],
     "qk_timer_create": [c_vp],
     "qk_timer_destroy": [c_vp],
     "qk_timer_start": [c_vp, c_vp],
@@ -377,8 +381,43 @@ class DevColumn:
             call(fn, sh, c_u64(n_idx), idx_col.ptr, self.ptr, out.ptr)
         return out
 
+    def view(self, offset_rows, n_rows):
+        """Non-owning row-range view (the overlapped exchange probes a
+        received CHUNK of a column while the next chunk is in flight)."""
+        return DevColumnView(self, offset_rows, n_rows)
+
     def free(self):
         self.buf.free()
+
+
+class DevColumnView:
+    """Non-owning slice of a DevColumn: same duck type (ptr/dtype/n) for
+    kernel calls; free() is a no-op (the base column owns the memory)."""
+
+    def __init__(self, base, offset_rows, n_rows):
+        self.dtype = base.dtype
+        self.n = int(n_rows)
+        self._base = base
+        self.ptr = c_vp(base.ptr.value + int(offset_rows) *
+                        base.dtype.itemsize)
+
+    def gather(self, idx_col, n_idx, stream=None):
+        fn = _DTYPE_GATHER[self.dtype]
+        sh = stream.handle if isinstance(stream, Stream) else stream
+        out = DevColumn(self.dtype, n_idx)
+        if n_idx:
+            call(fn, sh, c_u64(n_idx), idx_col.ptr, self.ptr, out.ptr)
+        return out
+
+    def to_numpy(self, n=None):
+        n = self.n if n is None else int(n)
+        out = np.empty(n, dtype=self.dtype)
+        if n:
+            _bounce.d2h(out, self.ptr)
+        return out
+
+    def free(self):
+        pass
 
 
 class Stream:
@@ -393,6 +432,29 @@ class Stream:
     def destroy(self):
         if self.handle is not None:
             call("qk_stream_destroy", self.handle)
+            self.handle = None
+
+
+class Event:
+    """Bare HIP event (no timing) for cross-stream ordering."""
+
+    def __init__(self):
+        p = c_vp(0)
+        call("qk_event_create", ctypes.byref(p))
+        self.handle = p
+
+    def record(self, stream):
+        call("qk_event_record", self.handle,
+             stream.handle if stream else None)
+
+    def wait(self, stream):
+        """Make `stream` wait until this event's recorded point."""
+        call("qk_stream_wait_event",
+             stream.handle if stream else None, self.handle)
+
+    def destroy(self):
+        if self.handle is not None:
+            call("qk_event_destroy", self.handle)
             self.handle = None
 
 

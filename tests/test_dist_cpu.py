@@ -125,3 +125,91 @@ def test_top10_merge_from_disjoint_rank_shards():
     got = {c: v[msel] for c, v in cand.items()}
     for c in full:
         assert np.array_equal(got[c], want[c]), c
+
+
+def test_plan_chunks_conservation_and_layout():
+    """exchange.plan_chunks: both sides derive the SAME split from the
+    exchanged counts. Check (a) per-peer chunk counts sum back to the
+    full counts, (b) send offsets tile each peer's partition range
+    exactly, (c) the chunk-major receive layout is contiguous and
+    non-overlapping."""
+    from quokka_amd.exchange import plan_chunks
+    rng = np.random.default_rng(5)
+    world, nchunks = 8, 4
+    send_counts = rng.integers(0, 10_000, world).astype(np.uint64)
+    send_offsets = np.zeros(world + 1, dtype=np.uint64)
+    np.cumsum(send_counts, out=send_offsets[1:])
+    recv_counts = rng.integers(0, 10_000, world).astype(np.uint64)
+    per_chunk, chunk_start, chunk_rows = plan_chunks(
+        send_offsets, send_counts, recv_counts, nchunks)
+
+    sc_sum = np.zeros(world, dtype=np.uint64)
+    rc_sum = np.zeros(world, dtype=np.uint64)
+    covered = []
+    for j, (so_j, sc_j, ro_j, rc_j) in enumerate(per_chunk):
+        sc_sum += sc_j
+        rc_sum += rc_j
+        # send ranges tile the peer's partition contiguously
+        for p in range(world):
+            prior = sum(int(per_chunk[k][1][p]) for k in range(j))
+            assert int(so_j[p]) == int(send_offsets[p]) + prior
+        # recv ranges: chunk j occupies [chunk_start[j], +chunk_rows[j])
+        lo = int(chunk_start[j])
+        cur = lo
+        for p in range(world):
+            assert int(ro_j[p]) == cur
+            cur += int(rc_j[p])
+        assert cur - lo == int(chunk_rows[j])
+        covered.append((lo, cur))
+    assert np.array_equal(sc_sum, send_counts)
+    assert np.array_equal(rc_sum, recv_counts)
+    # chunk recv ranges are contiguous, in order, and cover the total
+    assert covered[0][0] == 0
+    for (a, b), (c, d) in zip(covered, covered[1:]):
+        assert b == c
+    assert covered[-1][1] == int(recv_counts.sum())
+
+
+def test_plan_chunks_numpy_emulation_multiset():
+    """Emulate the chunked overlapped exchange with numpy copies across a
+    4-rank world: applying each rank's send plan into each peer's
+    chunk-major recv layout must deliver every row exactly once, with
+    per-rank received multisets equal to the unchunked exchange."""
+    from quokka_amd.exchange import plan_chunks
+    rng = np.random.default_rng(13)
+    world, nchunks = 4, 3
+    data = [rng.integers(0, 997, rng.integers(2_000, 4_000)).astype(np.int64)
+            for _ in range(world)]
+    ordered, offsets, send_counts = [], [], []
+    for r in range(world):
+        parts = E.partition_int(data[r], world)
+        order = np.argsort(parts, kind="stable")
+        ordered.append(data[r][order])
+        sc = np.bincount(parts, minlength=world).astype(np.uint64)
+        off = np.zeros(world + 1, dtype=np.uint64)
+        np.cumsum(sc, out=off[1:])
+        offsets.append(off)
+        send_counts.append(sc)
+    recv_counts = [np.array([send_counts[p][r] for p in range(world)],
+                            dtype=np.uint64) for r in range(world)]
+    recv = [np.full(int(rc.sum()), -1, dtype=np.int64)
+            for rc in recv_counts]
+    plans = [plan_chunks(offsets[r], send_counts[r], recv_counts[r],
+                         nchunks) for r in range(world)]
+    for j in range(nchunks):
+        for r in range(world):              # r sends chunk j to every peer
+            so_j, sc_j, _, _ = plans[r][0][j]
+            for p in range(world):
+                # receiver p's plan for chunk j names where r's piece lands
+                _, _, ro_p, rc_p = plans[p][0][j]
+                assert int(rc_p[r]) == int(sc_j[p])
+                lo_s = int(so_j[p]); n = int(sc_j[p])
+                lo_r = int(ro_p[r])
+                recv[p][lo_r:lo_r + n] = ordered[r][lo_s:lo_s + n]
+    for r in range(world):
+        assert not np.any(recv[r] == -1)
+        # multiset equality with the unchunked exchange result
+        want = np.concatenate([data[p][E.partition_int(data[p], world) == r]
+                               for p in range(world)])
+        assert np.array_equal(np.sort(recv[r]), np.sort(want))
+        assert np.all(recv[r] % world == r)

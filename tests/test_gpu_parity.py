@@ -1405,3 +1405,92 @@ def test_gpu_partition_fn_transforms(gpu):
         assert got == sorted(want.get(ch, [])), ch
         total += len(tbl)
     assert total == int(mask.sum())
+
+
+def test_repartition_overlapped_world1_multiset(gpu):
+    """Chunked overlapped self-exchange (exchange.repartition_overlapped):
+    every row delivered exactly once across the chunk-major layout, and
+    each consume() chunk view holds exactly its slice."""
+    from quokka_amd import exchange, shim
+    rng = np.random.default_rng(43)
+    n = 123_457
+    keys = rng.integers(0, 10_000, n).astype(np.int64)
+    vals = rng.random(n)
+    kcol = shim.DevColumn.from_numpy(keys)
+    vcol = shim.DevColumn.from_numpy(vals)
+    comm = exchange.Comm(0, 1)
+    comp, cstr = shim.Stream(), shim.Stream()
+    seen_chunks = []
+
+    def consume(views, start, nrows, j):
+        comp.sync()                     # test-only: read views host-side
+        seen_chunks.append((start, nrows, views["__key__"].to_numpy(),
+                            views["v"].to_numpy()))
+
+    rk, rp, chunk_start, chunk_rows = exchange.repartition_overlapped(
+        comm, kcol, {"v": vcol}, comp, cstr, consume, nchunks=5)
+    got_k = rk.to_numpy(rk.n)
+    got_v = rp["v"].to_numpy(rp["v"].n)
+    # full multiset preserved
+    og, ow = np.lexsort((got_v, got_k)), np.lexsort((vals, keys))
+    assert np.array_equal(got_k[og], keys[ow])
+    np.testing.assert_allclose(got_v[og], vals[ow], rtol=0)
+    # chunks tile the receive buffer and the views match it
+    assert len(seen_chunks) == 5
+    cursor = 0
+    for (start, nrows, ck, cv), cs, cr in zip(seen_chunks, chunk_start,
+                                              chunk_rows):
+        assert start == int(cs) == cursor and nrows == int(cr)
+        assert np.array_equal(ck, got_k[start:start + nrows])
+        np.testing.assert_allclose(cv, got_v[start:start + nrows], rtol=0)
+        cursor += nrows
+    assert cursor == n
+    rk.free(); rp["v"].free(); kcol.free(); vcol.free()
+    comp.destroy(); cstr.destroy()
+    comm.destroy()
+
+
+def test_q3_overlapped_world1_parity(gpu, data):
+    """The bench's overlapped exchange step at world 1: orders exchanged
+    plain + rebuilt, lineitem repartition overlapped with the fused probe
+    chunk pipeline == oracle Q3 (the configs[3] overlap path on one GPU)."""
+    from quokka_amd import queries as DQ, staging, exchange, shim
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipdate",
+                                             "l_extendedprice", "l_discount"])
+    ocols = staging.stage_columns(orders)
+    ccols = staging.stage_columns(cust, names=["c_custkey", "c_mktsegment"])
+    comm = exchange.Comm(0, 1)
+    comp, cstr = shim.Stream(), shim.Stream()
+    ok, op = exchange.repartition(
+        comm, ocols["o_orderkey"],
+        {k: v for k, v in ocols.items() if k != "o_orderkey"}, comp)
+    od_x = {"o_orderkey": ok, **op}
+    fused = DQ.Q3Fused(od_x, ccols, comp)
+
+    def consume(views, start, nrows, j):
+        fused.probe({"l_orderkey": views["__key__"],
+                     "l_shipdate": views["l_shipdate"],
+                     "l_extendedprice": views["l_extendedprice"],
+                     "l_discount": views["l_discount"]})
+
+    rk, rp, _, _ = exchange.repartition_overlapped(
+        comm, lcols["l_orderkey"],
+        {k: v for k, v in lcols.items() if k != "l_orderkey"},
+        comp, cstr, consume, nchunks=4)
+    full, top10 = fused.extract()
+    wfull, wtop = OQ.q3(li, orders, cust)
+    og = np.argsort(full["l_orderkey"])
+    ow = np.argsort(wfull["l_orderkey"])
+    assert np.array_equal(full["l_orderkey"][og], wfull["l_orderkey"][ow])
+    np.testing.assert_allclose(full["revenue"][og], wfull["revenue"][ow],
+                               rtol=1e-9)
+    assert np.array_equal(top10["l_orderkey"], wtop["l_orderkey"])
+    np.testing.assert_allclose(top10["revenue"], wtop["revenue"], rtol=1e-9)
+    fused.free()
+    for c in ([rk] + list(rp.values()) + list(od_x.values()) +
+              list(lcols.values()) + list(ocols.values()) +
+              list(ccols.values())):
+        c.free()
+    comp.destroy(); cstr.destroy()
+    comm.destroy()
