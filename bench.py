@@ -278,6 +278,10 @@ def run_e2e(args, shim, DQ):
         "l_linestatus": pa.array(LIN[host["l_linestatus"]]).dictionary_encode(),
     })
     del host
+    # TPC-H columns are non-null: write them as such so the decoder skips
+    # the per-page definition-level walk (host-side cost on warm scans)
+    t = t.cast(pa.schema([pa.field(f.name, f.type, nullable=False)
+                          for f in t.schema]))
     tmpdir = tempfile.mkdtemp(prefix="qk_e2e_")
     path = os.path.join(tmpdir, "lineitem.parquet")
     pq.write_table(t, path, compression="NONE", data_page_version="1.0",

@@ -418,6 +418,18 @@ int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
  * >= 8 bytes of slack after the end. */
 int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
                     const uint8_t *src_bytes, uint32_t *out);
+/* Host-side Thrift compact-protocol walk of one column chunk's page
+ * headers (parquet-format PageHeader; the metadata side of the decode —
+ * pyarrow exposes only the footer, and walking thousands of headers in
+ * Python measured ~0.2 s/GB of file, dominating warm end-to-end scans).
+ * out: max_pages x 10 i64 [kind(0 data v1/2 dict/3 data v2), num_values,
+ * encoding, def_level_encoding(-1 none/3 RLE), data_off(abs),
+ * data_len(compressed), v2_levels_len, num_nulls, uncompressed_len, 0];
+ * *n_out = pages written. Walks until num_values data values covered. */
+#define QK_PQ_PAGE_FIELDS 10
+int qk_pq_walk_pages(const uint8_t *buf, uint64_t start,
+                     uint64_t total_len, int64_t num_values, int64_t *out,
+                     int64_t max_pages, int64_t *n_out);
 
 /* ---- GPU CSV parse ---------------------------------------------------- *
  * The reference's CSV scan splits files into byte ranges and decodes on

@@ -126,7 +126,8 @@ PAGE_DATA_V2 = 3
 
 class PageInfo:
     __slots__ = ("kind", "num_values", "encoding", "def_enc",
-                 "data_off", "data_len", "v2_levels_len", "num_nulls")
+                 "data_off", "data_len", "v2_levels_len", "num_nulls",
+                 "uncompressed_len")
 
     def __repr__(self):
         return ("PageInfo(kind=%d n=%d enc=%d off=%d len=%d)"
@@ -138,7 +139,49 @@ def walk_pages(buf, start, total_len, num_values):
     """Walk a column chunk's byte range; yield PageInfo per page until
     `num_values` data values are covered. `buf` is the whole file (or
     chunk) as bytes/memoryview; offsets in the returned PageInfo are
-    absolute into `buf`."""
+    absolute into `buf`.
+
+    The walk itself runs in host C (qk_pq_walk_pages in the .so) — the
+    pure-Python walk below (_walk_pages_py) is the documented reference
+    of the format subset and the parity pin for the C parser
+    (tests/test_parquet_cpu.py)."""
+    import ctypes
+    import numpy as np
+    from . import shim
+
+    if not isinstance(buf, (bytes, bytearray)):
+        buf = bytes(buf)
+    cap = 4096
+    while True:
+        out = np.empty((cap, 10), dtype=np.int64)
+        n_out = ctypes.c_int64(0)
+        try:
+            shim.call(
+                "qk_pq_walk_pages", buf,
+                shim.c_u64(start), shim.c_u64(total_len),
+                shim.c_i64(num_values),
+                out.ctypes.data_as(shim.c_vp), shim.c_i64(cap),
+                ctypes.byref(n_out))
+        except shim.QkError as e:
+            if "max_pages" in str(e) and cap < (1 << 24):
+                cap *= 8
+                continue
+            raise
+        break
+    pages = []
+    for row in out[: n_out.value]:
+        p = PageInfo()
+        (p.kind, p.num_values, p.encoding, p.def_enc, p.data_off,
+         p.data_len, p.v2_levels_len, p.num_nulls,
+         p.uncompressed_len) = (int(row[0]), int(row[1]), int(row[2]),
+                                int(row[3]), int(row[4]), int(row[5]),
+                                int(row[6]), int(row[7]), int(row[8]))
+        pages.append(p)
+    return pages
+
+
+def _walk_pages_py(buf, start, total_len, num_values):
+    """Pure-Python reference walk (see walk_pages)."""
     pages = []
     pos = start
     end = start + total_len
@@ -149,6 +192,7 @@ def walk_pages(buf, start, total_len, num_values):
         p.kind = h[1]
         p.data_off = after
         p.data_len = h[3]                      # compressed size == plain
+        p.uncompressed_len = h.get(2, h[3])
         p.v2_levels_len = 0
         p.num_nulls = 0
         if p.kind == PAGE_DATA:
