@@ -237,7 +237,11 @@ __global__ void k_gen_lineitem(uint64_t n, uint64_t row_offset, uint64_t seed,
     if (l_discount) l_discount[i] = (double)(h6 % 11) / 100.0;
     if (l_tax) l_tax[i] = (double)(h7 % 9) / 100.0;
     if (l_suppkey) l_suppkey[i] = 1 + (int64_t)(h8 % (uint64_t)n_suppliers);
-    if (l_orderkey) l_orderkey[i] = 1 + (int64_t)(row / 4 % (uint64_t)n_orders);
+    if (l_orderkey) {
+      // the ORDER row's spec-sparse key (matches k_gen_orders)
+      uint64_t orow = row / 4 % (uint64_t)n_orders;
+      l_orderkey[i] = (int64_t)((orow / 8) * 32 + orow % 8) + 1;
+    }
   }
 }
 extern "C" int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset,
@@ -270,9 +274,15 @@ __global__ void k_gen_orders(uint64_t n, uint64_t row_offset, uint64_t seed,
     uint64_t base = splitmix64(seed ^ 0x0DE50DE50DE50DE5ULL) ^
                     (row * 0x9E3779B97F4A7C15ULL);
     uint64_t h0 = splitmix64(base + 0), h1 = splitmix64(base + 1);
-    if (o_orderkey) o_orderkey[i] = (int64_t)row + 1;
-    if (o_custkey)
-      o_custkey[i] = 1 + (int64_t)(h0 % (uint64_t)n_customers);
+    // TPC-H spec 4.2.3 sparse orderkey (8 keys per 32-key bucket) —
+    // matches oracle/tpch_gen.py sparse_orderkeys
+    if (o_orderkey) o_orderkey[i] = (int64_t)((row / 8) * 32 + row % 8) + 1;
+    if (o_custkey) {
+      int64_t ck = 1 + (int64_t)(h0 % (uint64_t)n_customers);
+      // spec: custkey % 3 != 0 (customer-mortality hole); step down one
+      if (ck % 3 == 0) ck -= 1;
+      o_custkey[i] = ck;
+    }
     if (o_orderdate)
       o_orderdate[i] = QK_ORDERDATE_LO +
           (int32_t)(h1 % (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));

@@ -1,0 +1,188 @@
+"""External-engine anchor for the oracle.
+
+The reference defines results via DuckDB SQL (apps/tpc-h/tpch_ref.py) and
+computes with polars — neither is installable here, and dbgen's per-stream
+seed table (dbgen rnd.c) is not vendored in /root/reference, so published
+TPC-H answer sets cannot be asserted bit-for-bit (oracle/tpch_gen.py
+docstring). The strongest independent engine REACHABLE in this container
+is pyarrow Acero (C++ hash join / hash aggregation — the same Arrow
+stack the reference itself hands its executors). These tests restate
+Q1/Q3/Q5/Q6 clause-for-clause on Acero and assert the numpy oracle
+matches it on the same generated inputs: two independent implementations
+of the reference SQL agreeing, rather than the oracle checking itself.
+"""
+import numpy as np
+import pyarrow as pa
+import pyarrow.compute as pc
+import pytest
+
+from oracle import tpch_gen as G, queries as OQ
+
+SF = 0.05
+SEED = 42
+
+
+@pytest.fixture(scope="module")
+def data():
+    return G.gen_all(SF, SEED)
+
+
+def _li_table(li):
+    return pa.table({
+        "l_orderkey": li["l_orderkey"],
+        "l_suppkey": li["l_suppkey"],
+        "l_quantity": li["l_quantity"],
+        "l_extendedprice": li["l_extendedprice"],
+        "l_discount": li["l_discount"],
+        "l_tax": li["l_tax"],
+        "l_returnflag": np.array(G.RETURNFLAG)[li["l_returnflag"]],
+        "l_linestatus": np.array(G.LINESTATUS)[li["l_linestatus"]],
+        "l_shipdate": li["l_shipdate"],
+    })
+
+
+def test_q1_oracle_equals_acero(data):
+    """tpch_ref.py:16-38 on Acero: filter + grouped sums/means/count."""
+    t = _li_table(data["lineitem"])
+    t = t.filter(pc.less_equal(t["l_shipdate"], G.Q1_CUTOFF))
+    disc_price = pc.multiply(t["l_extendedprice"],
+                             pc.subtract(pa.scalar(1.0), t["l_discount"]))
+    charge = pc.multiply(disc_price,
+                         pc.add(pa.scalar(1.0), t["l_tax"]))
+    t = t.append_column("disc_price", disc_price)
+    t = t.append_column("charge", charge)
+    g = t.group_by(["l_returnflag", "l_linestatus"]).aggregate([
+        ("l_quantity", "sum"), ("l_extendedprice", "sum"),
+        ("disc_price", "sum"), ("charge", "sum"),
+        ("l_quantity", "mean"), ("l_extendedprice", "mean"),
+        ("l_discount", "mean"), ("l_orderkey", "count"),
+    ])
+    g = g.sort_by([("l_returnflag", "ascending"),
+                   ("l_linestatus", "ascending")])
+    want = OQ.q1(data["lineitem"])
+    assert g.num_rows == len(want["count_order"])
+    assert g.column("l_returnflag").to_pylist() == \
+        list(want["l_returnflag"])
+    assert g.column("l_linestatus").to_pylist() == \
+        list(want["l_linestatus"])
+    assert np.array_equal(np.asarray(g.column("l_orderkey_count")),
+                          want["count_order"])
+    for acol, wcol in [("l_quantity_sum", "sum_qty"),
+                       ("l_extendedprice_sum", "sum_base_price"),
+                       ("disc_price_sum", "sum_disc_price"),
+                       ("charge_sum", "sum_charge"),
+                       ("l_quantity_mean", "avg_qty"),
+                       ("l_extendedprice_mean", "avg_price"),
+                       ("l_discount_mean", "avg_disc")]:
+        np.testing.assert_allclose(np.asarray(g.column(acol)), want[wcol],
+                                   rtol=1e-9, err_msg=acol)
+
+
+def test_q6_oracle_equals_acero(data):
+    li = data["lineitem"]
+    t = pa.table({k: li[k] for k in ("l_shipdate", "l_quantity",
+                                     "l_extendedprice", "l_discount")})
+    lo, hi = 0.06 - 0.01, 0.06 + 0.01
+    m = pc.and_(
+        pc.and_(pc.greater_equal(t["l_shipdate"], G.Q5_LO),
+                pc.less(t["l_shipdate"], G.Q5_HI)),
+        pc.and_(pc.and_(pc.greater_equal(t["l_discount"], lo),
+                        pc.less_equal(t["l_discount"], hi)),
+                pc.less(t["l_quantity"], 24.0)))
+    ft = t.filter(m)
+    rev = pc.sum(pc.multiply(ft["l_extendedprice"],
+                             ft["l_discount"])).as_py()
+    want = OQ.q6(li)
+    np.testing.assert_allclose(rev, want["revenue"], rtol=1e-9)
+    assert ft.num_rows == want["rows_passed"]
+
+
+def test_q3_oracle_equals_acero(data):
+    """tpch_ref.py:89-115 on Acero: two hash joins + hash group-by."""
+    li, orders, cust = data["lineitem"], data["orders"], data["customer"]
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "seg": cust["c_mktsegment"]})
+    c = c.filter(pc.equal(c["seg"], G.MKTSEGMENT.index("BUILDING")))
+    o = pa.table({k: orders[k] for k in ("o_orderkey", "o_custkey",
+                                         "o_orderdate", "o_shippriority")})
+    o = o.filter(pc.less(o["o_orderdate"], G.Q3_DATE))
+    o = o.join(c.select(["c_custkey"]), keys="o_custkey",
+               right_keys="c_custkey", join_type="left semi")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_shipdate",
+                                     "l_extendedprice", "l_discount")})
+    l = l.filter(pc.greater(l["l_shipdate"], G.Q3_DATE))
+    j = l.join(o, keys="l_orderkey", right_keys="o_orderkey",
+               join_type="inner")
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    j = j.append_column("revenue", rev)
+    g = j.group_by(["l_orderkey", "o_orderdate",
+                    "o_shippriority"]).aggregate([("revenue", "sum")])
+    g = g.sort_by([("l_orderkey", "ascending")])
+    full, top10 = OQ.q3(li, orders, cust)
+    order = np.argsort(full["l_orderkey"])
+    assert g.num_rows == len(full["l_orderkey"])
+    assert np.array_equal(np.asarray(g.column("l_orderkey")),
+                          full["l_orderkey"][order])
+    assert np.array_equal(np.asarray(g.column("o_orderdate")),
+                          full["o_orderdate"][order])
+    np.testing.assert_allclose(np.asarray(g.column("revenue_sum")),
+                               full["revenue"][order], rtol=1e-9)
+    # and the top-10 rule (revenue desc, orderdate asc) on Acero's result
+    gs = g.sort_by([("revenue_sum", "descending"),
+                    ("o_orderdate", "ascending"),
+                    ("l_orderkey", "ascending")]).slice(0, 10)
+    assert np.array_equal(np.asarray(gs.column("l_orderkey")),
+                          top10["l_orderkey"])
+    np.testing.assert_allclose(np.asarray(gs.column("revenue_sum")),
+                               top10["revenue"], rtol=1e-9)
+
+
+def test_q5_oracle_equals_acero(data):
+    """tpch_ref.py:142-169 on Acero: 6-table chain with the extra
+    c_nationkey = s_nationkey equi-predicate."""
+    li, orders = data["lineitem"], data["orders"]
+    cust, supp = data["customer"], data["supplier"]
+    nat, reg = data["nation"], data["region"]
+    asia = G.REGIONS.index("ASIA")
+    n = pa.table({"n_nationkey": nat["n_nationkey"],
+                  "n_regionkey": nat["n_regionkey"]})
+    n = n.filter(pc.equal(n["n_regionkey"], asia))
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "c_nationkey": cust["c_nationkey"]})
+    c = c.join(n.select(["n_nationkey"]), keys="c_nationkey",
+               right_keys="n_nationkey", join_type="left semi")
+    o = pa.table({k: orders[k] for k in ("o_orderkey", "o_custkey",
+                                         "o_orderdate")})
+    o = o.filter(pc.and_(pc.greater_equal(o["o_orderdate"], G.Q5_LO),
+                         pc.less(o["o_orderdate"], G.Q5_HI)))
+    oc = o.join(c, keys="o_custkey", right_keys="c_custkey",
+                join_type="inner")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_suppkey",
+                                     "l_extendedprice", "l_discount")})
+    j = l.join(oc.select(["o_orderkey", "c_nationkey"]),
+               keys="l_orderkey", right_keys="o_orderkey",
+               join_type="inner")
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "s_nationkey": supp["s_nationkey"]})
+    j = j.join(s, keys="l_suppkey", right_keys="s_suppkey",
+               join_type="inner")
+    j = j.filter(pc.equal(j["c_nationkey"], j["s_nationkey"]))
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    j = j.append_column("revenue", rev)
+    g = j.group_by(["c_nationkey"]).aggregate([("revenue", "sum")])
+    got = {int(k): v for k, v in zip(g.column("c_nationkey").to_pylist(),
+                                     g.column("revenue_sum").to_pylist())}
+    want = OQ.q5(li, orders, cust, supp, nat, reg)
+    names = [nm for nm, _ in G.NATIONS]
+    for nm, wrev in want:
+        nk = names.index(nm)
+        if wrev == 0.0:
+            assert got.get(nk, 0.0) == 0.0
+        else:
+            np.testing.assert_allclose(got[nk], wrev, rtol=1e-9,
+                                       err_msg=nm)
+    # no non-ASIA nation appears
+    asia_keys = {i for i in range(25) if nat["n_regionkey"][i] == asia}
+    assert set(got) <= asia_keys
