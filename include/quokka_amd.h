@@ -418,6 +418,18 @@ int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
  * >= 8 bytes of slack after the end. */
 int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
                     const uint8_t *src_bytes, uint32_t *out);
+/* GPU snappy decompression of Parquet pages (the reference reads
+ * compressed files transparently through pyarrow,
+ * unordered_readers.py:51). One wave per page, pages independent.
+ * descs_dev: npages x 8 u64 [src_off, src_len, dst_off,
+ * uncompressed_len, mode (0 raw / 1 verify v1 def-levels max_def==1),
+ * num_values, 0, 0]; out_dev: npages x 4 i64 [data_off_rel (start of
+ * values after the in-page levels block), err (0 ok / 1 corrupt /
+ * 2 length mismatch / 3 nulls present), first_byte_after_levels (the
+ * RLE bit-width byte, -1 if none), 0]. */
+int qk_snappy_pages(void *stream, uint64_t npages, const uint64_t *descs_dev,
+                    const uint8_t *src_dev, uint8_t *dst_dev,
+                    int64_t *out_dev);
 /* Host-side Thrift compact-protocol walk of one column chunk's page
  * headers (parquet-format PageHeader; the metadata side of the decode —
  * pyarrow exposes only the footer, and walking thousands of headers in

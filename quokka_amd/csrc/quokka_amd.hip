@@ -2154,6 +2154,185 @@ extern "C" int qk_pq_rle_pages(void *stream, uint64_t npages,
   return 0;
 }
 
+// ---- GPU snappy page decompression -------------------------------------
+// Parquet SNAPPY-compressed pages (the reference reads them transparently
+// through pyarrow, unordered_readers.py:51; real TPC-H datasets ship
+// compressed). One WAVE per page: lane 0 walks the snappy element stream
+// (format: varint uncompressed length, then tagged elements — tag low
+// 2 bits: 0 literal, 1 copy with 11-bit offset, 2 copy with 2-byte
+// offset, 3 copy with 4-byte offset) and broadcasts each element; all 64
+// lanes execute the copy cooperatively. Match copies may read bytes this
+// wave wrote earlier, so a single-wave __syncthreads() (block == one
+// wave) orders them; literals read only the source and skip the barrier.
+// Pages decode independently — thousands in flight fill the chip.
+//
+// After decompression, v1 data pages carry their definition-levels block
+// INSIDE the decompressed bytes; for max_def == 1 lane 0 verifies
+// "no nulls" (bit-packed groups all 0xFF / fill runs == 1, mirroring
+// parquet_gpu._check_levels_v1) and reports where the values start.
+//
+// desc per page (8 u64): [src_off, src_len, dst_off, uncompressed_len,
+//   mode (0 = raw, 1 = verify v1 def-levels for max_def 1), num_values,
+//   0, 0]
+// out per page (4 i64): [data_off_rel (after levels), err (0 ok,
+//   1 corrupt stream, 2 length mismatch, 3 nulls present),
+//   first_byte_after_levels (the RLE bit-width byte), 0]
+__global__ void k_snappy_pages(uint64_t npages,
+                               const uint64_t *__restrict__ descs,
+                               const uint8_t *__restrict__ src,
+                               uint8_t *__restrict__ dst,
+                               int64_t *__restrict__ out) {
+  uint64_t page = blockIdx.x;
+  if (page >= npages) return;
+  const uint64_t *d = descs + page * 8;
+  const uint8_t *ip = src + d[0];
+  const uint8_t *iend = ip + d[1];
+  uint8_t *op0 = dst + d[2];
+  uint64_t expect = d[3];
+  int lane = threadIdx.x;
+  int64_t *o = out + page * 4;
+
+  // varint preamble: uncompressed length (lane 0 parses; all lanes
+  // recompute — cheap, keeps them in lockstep without a broadcast)
+  uint64_t ulen = 0;
+  {
+    int shift = 0;
+    const uint8_t *p = ip;
+    while (p < iend) {
+      uint8_t b = *p++;
+      ulen |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    ip = p;
+  }
+  if (ulen != expect) {
+    if (lane == 0) { o[0] = 0; o[1] = 2; o[2] = 0; }
+    return;
+  }
+
+  uint64_t opos = 0;
+  int err = 0;
+  while (ip < iend && opos < ulen) {
+    // every lane parses the tag identically (uniform scalar work) — no
+    // broadcast needed, the wave stays converged
+    uint8_t tag = *ip++;
+    uint64_t len;
+    uint64_t cof = 0;       // copy offset (0 => literal)
+    if ((tag & 3) == 0) {                         // literal
+      len = (tag >> 2) + 1;
+      if (len > 60) {
+        int nb = (int)len - 60;
+        len = 0;
+        for (int k = 0; k < nb; k++) len |= (uint64_t)ip[k] << (8 * k);
+        len += 1;
+        ip += nb;
+      }
+      if (ip + len > iend || opos + len > ulen) { err = 1; break; }
+      for (uint64_t k = lane; k < len; k += 64) op0[opos + k] = ip[k];
+      ip += len;
+    } else {
+      if ((tag & 3) == 1) {                       // copy1: 4..11 bytes
+        len = 4 + ((tag >> 2) & 7);
+        cof = ((uint64_t)(tag >> 5) << 8) | *ip;
+        ip += 1;
+      } else if ((tag & 3) == 2) {                // copy2
+        len = (tag >> 2) + 1;
+        cof = (uint64_t)ip[0] | ((uint64_t)ip[1] << 8);
+        ip += 2;
+      } else {                                    // copy4
+        len = (tag >> 2) + 1;
+        cof = (uint64_t)ip[0] | ((uint64_t)ip[1] << 8) |
+              ((uint64_t)ip[2] << 16) | ((uint64_t)ip[3] << 24);
+        ip += 4;
+      }
+      if (cof == 0 || cof > opos || opos + len > ulen) { err = 1; break; }
+      // the match may reference bytes other lanes wrote: order them
+      __threadfence_block();
+      __syncthreads();
+      if (cof >= len) {
+        for (uint64_t k = lane; k < len; k += 64)
+          op0[opos + k] = op0[opos - cof + k];
+      } else {
+        // overlapped match = repeating pattern of period cof: every
+        // output byte maps to the PRE-MATCH region, no intra-copy RAW
+        for (uint64_t k = lane; k < len; k += 64)
+          op0[opos + k] = op0[opos - cof + (k % cof)];
+      }
+      __threadfence_block();
+      __syncthreads();
+    }
+    opos += len;
+  }
+  if (!err && opos != ulen) err = 1;
+
+  if (lane == 0) {
+    int64_t data_off = 0;
+    int64_t first = -1;
+    if (!err && d[4] == 1) {
+      // v1 definition-levels block: [u32 len][RLE runs], max_def == 1
+      if (ulen < 4) {
+        err = 1;
+      } else {
+        uint64_t ln = (uint64_t)op0[0] | ((uint64_t)op0[1] << 8) |
+                      ((uint64_t)op0[2] << 16) | ((uint64_t)op0[3] << 24);
+        uint64_t p = 4, lend = 4 + ln;
+        uint64_t seen = 0, nvals = d[5];
+        if (lend > ulen) err = 1;
+        while (!err && seen < nvals && p < lend) {
+          // varint run header (mirrors _check_levels_v1 exactly)
+          uint64_t h = 0;
+          int shift = 0;
+          while (p < lend) {
+            uint8_t b = op0[p++];
+            h |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+          }
+          if (h & 1) {                    // bit-packed groups, width 1
+            uint64_t ngroups = h >> 1;
+            uint64_t nv = ngroups * 8;
+            if (nv > nvals - seen) nv = nvals - seen;
+            uint64_t full = nv / 8, rem = nv % 8;
+            if (p + ngroups > lend) { err = 1; break; }
+            for (uint64_t g = 0; g < full; g++)
+              if (op0[p + g] != 0xFF) { err = 3; break; }
+            if (!err && rem &&
+                (op0[p + full] & ((1u << rem) - 1)) != ((1u << rem) - 1))
+              err = 3;
+            p += ngroups;
+            seen += nv;
+          } else {                        // fill run
+            if (p >= lend) { err = 1; break; }
+            uint64_t nv = h >> 1;
+            if (nv > nvals - seen) nv = nvals - seen;
+            if (nv && op0[p] != 1) err = 3;
+            p += 1;
+            seen += nv;
+          }
+        }
+        data_off = (int64_t)lend;
+      }
+    }
+    if (!err && (uint64_t)data_off < ulen) first = op0[data_off];
+    o[0] = data_off;
+    o[1] = err;
+    o[2] = first;
+    o[3] = 0;
+  }
+}
+extern "C" int qk_snappy_pages(void *stream, uint64_t npages,
+                               const uint64_t *descs_dev,
+                               const uint8_t *src_dev, uint8_t *dst_dev,
+                               int64_t *out_dev) {
+  if (!npages) return 0;
+  hipLaunchKernelGGL(k_snappy_pages, dim3((uint32_t)npages), dim3(64), 0,
+                     (hipStream_t)stream, npages, descs_dev, src_dev,
+                     dst_dev, out_dev);
+  QK_TRY("qk_snappy_pages", hipGetLastError());
+  return 0;
+}
+
 // ---- host-side Thrift page-header walker -------------------------------
 // Parquet page headers are Thrift compact-protocol structs between pages
 // (parquet-format PageHeader). pyarrow does not expose them, and walking

@@ -116,8 +116,9 @@ class _Chunk:
 
     def __init__(self, buf, col, max_def, dst0):
         if col.compression != "UNCOMPRESSED":
-            raise QkParquetError("compressed chunks unsupported (%s); "
-                                 "rewrite with compression='NONE'"
+            raise QkParquetError("compressed chunks unsupported here (%s); "
+                                 "SNAPPY goes through the GPU decompressor "
+                                 "(_snappy_column), others are rejected"
                                  % col.compression)
         self.dtype = _PHYS.get(col.physical_type)
         self.is_ba = col.physical_type == "BYTE_ARRAY"
@@ -190,6 +191,218 @@ class _Chunk:
                 offset=p.data_off).copy()
 
 
+class _PlanChunk:
+    """Duck-typed like _Chunk for _decode_column: a decode plan whose
+    offsets point into the DECOMPRESSED scratch buffer instead of the
+    file bytes."""
+
+    def __init__(self, col, dtype, is_ba, n):
+        self.dtype = dtype
+        self.is_ba = is_ba
+        self.n = n
+        self.plain_tiles = []
+        self.rle_pages = []
+        self.dict_vals = None
+
+
+def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
+    """Plan + decompress one column whose chunks are SNAPPY-compressed
+    (or a mix of SNAPPY and UNCOMPRESSED pages): every page's
+    decompressed bytes land in one scratch DevBuffer via qk_snappy_pages
+    (one wave per page; v1 definition levels verified in-kernel), then
+    the normal qk_pq_plain_copy / qk_pq_rle_pages decode runs against
+    the scratch. Returns (list of _PlanChunk, scratch DevBuffer) —
+    caller passes scratch as the decode source and frees it.
+
+    The reference reads compressed files transparently through pyarrow
+    (unordered_readers.py:51); real TPC-H datasets usually ship snappy."""
+    import ctypes
+    from .shim import DevBuffer, c_u64, c_vp
+
+    if max_def > 1 or (max_def and max_def.bit_length() != 1):
+        raise QkParquetError("nested schemas unsupported (max_def=%d)"
+                             % max_def)
+    phys = cols_meta[0].physical_type
+    dtype = _PHYS.get(phys)
+    is_ba = phys == "BYTE_ARRAY"
+    if dtype is None and not is_ba:
+        raise QkParquetError("unsupported physical type %s" % phys)
+
+    chunk_pages = []
+    row = dst0
+    for col in cols_meta:
+        if col.compression not in ("SNAPPY", "UNCOMPRESSED"):
+            raise QkParquetError("compression %s unsupported (SNAPPY and "
+                                 "UNCOMPRESSED only)" % col.compression)
+        start = col.data_page_offset
+        if col.dictionary_page_offset is not None:
+            start = min(start, col.dictionary_page_offset)
+        pl = T.walk_pages(raw, start, col.total_compressed_size,
+                          col.num_values)
+        chunk_pages.append((col, pl, row))
+        row += col.num_values
+
+    descs = []          # qk_snappy_pages descriptors
+    copies = []         # byte-copy tiles (uncompressed pages / v2 levels)
+    plans = []          # per page: post-processing info
+    off = 0
+    for ci, (col, pl, row0) in enumerate(chunk_pages):
+        rows = row0
+        for p in pl:
+            snappy = col.compression == "SNAPPY"
+            ent = {"chunk": ci, "page": p, "off": off, "row": rows,
+                   "desc": None}
+            if not snappy:
+                copies.append((p.data_off, off, p.data_len))
+                ent["kind"] = "raw"
+                ent["size"] = p.data_len
+            elif p.kind == T.PAGE_DATA:            # v1: levels inside
+                descs.append((p.data_off, p.data_len, off,
+                              p.uncompressed_len,
+                              1 if max_def > 0 else 0, p.num_values, 0, 0))
+                ent["kind"] = "v1"
+                ent["desc"] = len(descs) - 1
+                ent["size"] = p.uncompressed_len
+            elif p.kind == T.PAGE_DICT:
+                descs.append((p.data_off, p.data_len, off,
+                              p.uncompressed_len, 0, p.num_values, 0, 0))
+                ent["kind"] = "dict"
+                ent["desc"] = len(descs) - 1
+                ent["size"] = p.uncompressed_len
+            else:                                  # v2: levels NOT compressed
+                if p.num_nulls:
+                    raise QkParquetError("page contains nulls")
+                lv = p.v2_levels_len
+                if lv:
+                    copies.append((p.data_off, off, lv))
+                descs.append((p.data_off + lv, p.data_len - lv, off + lv,
+                              p.uncompressed_len - lv, 0, p.num_values,
+                              0, 0))
+                ent["kind"] = "v2"
+                ent["desc"] = len(descs) - 1
+                ent["size"] = p.uncompressed_len
+            if p.kind != T.PAGE_DICT:
+                rows += p.num_values
+            plans.append(ent)
+            off += ent["size"]
+
+    scratch = DevBuffer(off + 8)
+    if copies:
+        tiles = np.asarray(copies, dtype=np.uint64)
+        dt = DevBuffer(tiles.nbytes)
+        shim.call("qk_h2d", dt.ptr, tiles.ctypes.data_as(c_vp),
+                  c_u64(tiles.nbytes))
+        shim.call("qk_pq_plain_copy", None, c_u64(len(tiles)), dt.ptr,
+                  dev_file.ptr, scratch.ptr, ctypes.c_uint32(1))
+        dt.free()
+    results = None
+    if descs:
+        da = np.asarray(descs, dtype=np.uint64)
+        dd = DevBuffer(da.nbytes)
+        shim.call("qk_h2d", dd.ptr, da.ctypes.data_as(c_vp),
+                  c_u64(da.nbytes))
+        dout = DevBuffer(len(descs) * 4 * 8)
+        shim.call("qk_snappy_pages", None, c_u64(len(descs)), dd.ptr,
+                  dev_file.ptr, scratch.ptr, dout.ptr)
+        shim.call("qk_stream_sync", None)
+        results = np.zeros(len(descs) * 4, dtype=np.int64)
+        shim.call("qk_d2h", results.ctypes.data_as(c_vp), dout.ptr,
+                  c_u64(results.nbytes))
+        results = results.reshape(-1, 4)
+        dd.free()
+        dout.free()
+        bad = np.nonzero(results[:, 1])[0]
+        if bad.size:
+            code = int(results[bad[0], 1])
+            raise QkParquetError(
+                "snappy page %d failed: %s" % (
+                    bad[0], {1: "corrupt stream",
+                             2: "uncompressed-length mismatch",
+                             3: "page contains nulls"}.get(code, code)))
+
+    out_chunks = []
+    cur = None
+    scratch_host = {}       # chunk dict pages d2h'd lazily
+
+    def _scratch_bytes(o, ln):
+        arr = np.empty(ln, dtype=np.uint8)
+        shim.call("qk_d2h", arr.ctypes.data_as(c_vp),
+                  ctypes.c_void_p(scratch.ptr.value + o), c_u64(ln))
+        return arr.tobytes()
+
+    for ci, (col, pl, row0) in enumerate(chunk_pages):
+        cur = _PlanChunk(col, dtype, is_ba, col.num_values)
+        out_chunks.append(cur)
+    for ent in plans:
+        p = ent["page"]
+        cur = out_chunks[ent["chunk"]]
+        o = ent["off"]
+        if ent["kind"] == "dict" or (ent["kind"] == "raw"
+                                     and p.kind == T.PAGE_DICT):
+            if p.encoding not in (ENC_PLAIN, ENC_PLAIN_DICT):
+                raise QkParquetError("dict page encoding %d" % p.encoding)
+            if ent["kind"] == "dict":
+                db = _scratch_bytes(o, ent["size"])
+            else:
+                db = bytes(raw[p.data_off:p.data_off + p.data_len])
+            if is_ba:
+                vals, pos = [], 0
+                for _ in range(p.num_values):
+                    ln = int.from_bytes(db[pos:pos + 4], "little")
+                    pos += 4
+                    vals.append(db[pos:pos + ln].decode())
+                    pos += ln
+                cur.dict_vals = vals
+            else:
+                cur.dict_vals = np.frombuffer(
+                    db, dtype=dtype, count=p.num_values).copy()
+            continue
+        # data pages: find where values start + the RLE bit-width byte
+        if ent["kind"] == "v1":
+            r = results[ent["desc"]]
+            data = o + int(r[0])
+            first = int(r[2])
+            end = o + ent["size"]
+        elif ent["kind"] == "v2":
+            data = o + p.v2_levels_len
+            r = results[ent["desc"]]
+            first = int(r[2])
+            end = o + ent["size"]
+        else:                                       # raw page in scratch
+            data_abs = p.data_off
+            if p.kind == T.PAGE_DATA:
+                if max_def > 0:
+                    data_abs = _check_levels_v1(raw, data_abs,
+                                                p.num_values, max_def)
+            else:
+                if p.num_nulls:
+                    raise QkParquetError("page contains nulls")
+                data_abs = p.data_off + p.v2_levels_len
+            data = o + (data_abs - p.data_off)
+            first = raw[data_abs] if data_abs < p.data_off + p.data_len \
+                else -1
+            end = o + ent["size"]
+        if p.encoding == ENC_PLAIN:
+            if is_ba:
+                raise QkParquetError("PLAIN BYTE_ARRAY unsupported; "
+                                     "write with use_dictionary=True")
+            es = dtype.itemsize
+            s = 0
+            while s < p.num_values:
+                m = min(_TILE, p.num_values - s)
+                cur.plain_tiles.append((data + s * es, ent["row"] + s, m))
+                s += m
+        elif p.encoding == ENC_RLE_DICT:
+            bw = first
+            if bw < 0 or bw > 32:
+                raise QkParquetError("index bit width %d" % bw)
+            cur.rle_pages.append((data + 1, end, ent["row"],
+                                  p.num_values, bw))
+        else:
+            raise QkParquetError("data page encoding %d" % p.encoding)
+    return out_chunks, scratch
+
+
 def _upload(shim, host_bytes):
     """File bytes -> device, with the 8-byte slack qk_pq_rle_expand needs."""
     from .shim import DevBuffer, c_u64, c_vp
@@ -224,10 +437,22 @@ def read_table(source, columns=None):
     try:
         for ci in want:
             max_def = md.schema.column(ci).max_definition_level
+            cols_meta = [md.row_group(rg).column(ci)
+                         for rg in range(md.num_row_groups)]
+            comps = {c.compression for c in cols_meta}
+            if "SNAPPY" in comps:
+                # GPU decompression path: pages -> scratch -> normal decode
+                chunks, scratch = _snappy_column(shim, raw, dev_file,
+                                                 cols_meta, max_def, 0)
+                assert sum(ch.n for ch in chunks) == total
+                out[names[ci]] = _decode_column(shim, scratch, chunks,
+                                                total)
+                shim.call("qk_stream_sync", None)
+                scratch.free()
+                continue
             chunks = []
             row = 0
-            for rg in range(md.num_row_groups):
-                col = md.row_group(rg).column(ci)
+            for col in cols_meta:
                 ch = _Chunk(raw, col, max_def, row)
                 row += ch.n
                 chunks.append(ch)
