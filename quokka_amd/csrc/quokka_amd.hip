@@ -2459,7 +2459,7 @@ extern "C" int qk_pq_walk_pages(const uint8_t *buf, uint64_t start,
     o[6] = 0;                       // v2_levels_len
     o[7] = 0;                       // num_nulls
     o[8] = top.has[2] ? top.v[2] : top.v[3];  // uncompressed size
-    o[9] = 0;
+    o[9] = 1;  // page data compressed under the chunk codec (v2 may opt out)
     if (kind == 0) {                // data page v1
       if (!d5.has[1] || !d5.has[2])
         return qk_fail("qk_pq_walk_pages.v1", hipErrorInvalidValue);
@@ -2481,6 +2481,9 @@ extern "C" int qk_pq_walk_pages(const uint8_t *buf, uint64_t start,
       o[3] = 3;
       o[6] = (d8.has[5] ? d8.v[5] : 0) + (d8.has[6] ? d8.v[6] : 0);
       o[7] = d8.has[2] ? d8.v[2] : 0;
+      // DataPageHeaderV2.is_compressed (optional bool, default true):
+      // writers leave incompressible pages raw and clear this flag
+      o[9] = d8.has[7] ? d8.v[7] : 1;
       seen += o[1];
     } else {
       return qk_fail("qk_pq_walk_pages.page_type", hipErrorInvalidValue);

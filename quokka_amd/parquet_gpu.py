@@ -249,7 +249,9 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
     for ci, (col, pl, row0) in enumerate(chunk_pages):
         rows = row0
         for p in pl:
-            snappy = col.compression == "SNAPPY"
+            # v2 pages carry is_compressed (writers leave incompressible
+            # pages raw and clear it) — those route to the byte-copy path
+            snappy = col.compression == "SNAPPY" and p.is_compressed
             ent = {"chunk": ci, "page": p, "off": off, "row": rows,
                    "desc": None}
             if not snappy:
@@ -425,8 +427,17 @@ def read_table(source, columns=None):
     import ctypes
     import io
 
-    raw = open(source, "rb").read() if isinstance(source, str) else source
-    f = pq.ParquetFile(io.BytesIO(raw))
+    if isinstance(source, str):
+        # mmap instead of read(): the page-cache copy then happens inside
+        # the pinned upload bounce (4-thread memmove overlapped with DMA)
+        # rather than as a separate full-file pass
+        import mmap
+        fh = open(source, "rb")
+        raw = mmap.mmap(fh.fileno(), 0, prot=mmap.PROT_READ)
+        f = pq.ParquetFile(source)
+    else:
+        raw = source
+        f = pq.ParquetFile(io.BytesIO(raw))
     md = f.metadata
     names = [md.schema.column(i).name for i in range(md.num_columns)]
     want = [i for i, nm in enumerate(names)
@@ -462,6 +473,9 @@ def read_table(source, columns=None):
         return out
     finally:
         dev_file.free()
+        if isinstance(source, str):
+            raw.close()
+            fh.close()
 
 
 def _decode_column(shim, dev_file, chunks, total):

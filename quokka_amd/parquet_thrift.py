@@ -127,7 +127,7 @@ PAGE_DATA_V2 = 3
 class PageInfo:
     __slots__ = ("kind", "num_values", "encoding", "def_enc",
                  "data_off", "data_len", "v2_levels_len", "num_nulls",
-                 "uncompressed_len")
+                 "uncompressed_len", "is_compressed")
 
     def __repr__(self):
         return ("PageInfo(kind=%d n=%d enc=%d off=%d len=%d)"
@@ -149,15 +149,16 @@ def walk_pages(buf, start, total_len, num_values):
     import numpy as np
     from . import shim
 
-    if not isinstance(buf, (bytes, bytearray)):
-        buf = bytes(buf)
+    # zero-copy pointer for bytes / mmap / any buffer-protocol object
+    barr = np.frombuffer(buf, dtype=np.uint8)
+    bptr = barr.ctypes.data_as(shim.c_vp)
     cap = 4096
     while True:
         out = np.empty((cap, 10), dtype=np.int64)
         n_out = ctypes.c_int64(0)
         try:
             shim.call(
-                "qk_pq_walk_pages", buf,
+                "qk_pq_walk_pages", bptr,
                 shim.c_u64(start), shim.c_u64(total_len),
                 shim.c_i64(num_values),
                 out.ctypes.data_as(shim.c_vp), shim.c_i64(cap),
@@ -172,10 +173,11 @@ def walk_pages(buf, start, total_len, num_values):
     for row in out[: n_out.value]:
         p = PageInfo()
         (p.kind, p.num_values, p.encoding, p.def_enc, p.data_off,
-         p.data_len, p.v2_levels_len, p.num_nulls,
-         p.uncompressed_len) = (int(row[0]), int(row[1]), int(row[2]),
-                                int(row[3]), int(row[4]), int(row[5]),
-                                int(row[6]), int(row[7]), int(row[8]))
+         p.data_len, p.v2_levels_len, p.num_nulls, p.uncompressed_len,
+         p.is_compressed) = (int(row[0]), int(row[1]), int(row[2]),
+                             int(row[3]), int(row[4]), int(row[5]),
+                             int(row[6]), int(row[7]), int(row[8]),
+                             bool(row[9]))
         pages.append(p)
     return pages
 
@@ -193,6 +195,7 @@ def _walk_pages_py(buf, start, total_len, num_values):
         p.data_off = after
         p.data_len = h[3]                      # compressed size == plain
         p.uncompressed_len = h.get(2, h[3])
+        p.is_compressed = True
         p.v2_levels_len = 0
         p.num_nulls = 0
         if p.kind == PAGE_DATA:
@@ -210,6 +213,9 @@ def _walk_pages_py(buf, start, total_len, num_values):
             p.encoding = h[8][4]
             p.def_enc = 3
             p.v2_levels_len = h[8].get(5, 0) + h[8].get(6, 0)
+            # optional bool, default true; writers clear it on pages they
+            # chose to leave uncompressed
+            p.is_compressed = bool(h[8].get(7, True))
             seen += p.num_values
         else:
             raise ValueError("unsupported page type %d" % p.kind)

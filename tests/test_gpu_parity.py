@@ -459,10 +459,17 @@ def test_q3_fused_on_device_generated(gpu):
     li_h = {k: v.to_numpy() for k, v in li.items()}
     od_h = {k: v.to_numpy() for k, v in od.items()}
     cu_h = {k: v.to_numpy() for k, v in cu.items()}
-    # shipdate must correlate with the order's orderdate (1..121 ahead)
-    omap = od_h["o_orderdate"][(li_h["l_orderkey"] - 1).astype(np.int64)]
+    # shipdate must correlate with the order's orderdate (1..121 ahead);
+    # invert the spec 4.2.3 sparse orderkey (8 keys per 32-key bucket)
+    # back to the order ROW to index the generated orders table
+    k0 = (li_h["l_orderkey"] - 1).astype(np.int64)
+    assert np.all(k0 % 32 < 8), "orderkeys must be spec-sparse"
+    orow = (k0 // 32) * 8 + (k0 % 32)
+    omap = od_h["o_orderdate"][orow]
     delta = li_h["l_shipdate"] - omap
     assert delta.min() >= 1 and delta.max() <= 121
+    # and the custkey mortality hole (spec: custkey % 3 != 0)
+    assert np.all(od_h["o_custkey"] % 3 != 0)
     full, top10 = DQ.q3_fused(li, od, cu)
     wfull, wtop = OQ.q3(li_h, od_h, cu_h)
     og = np.argsort(full["l_orderkey"])
