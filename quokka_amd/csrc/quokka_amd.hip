@@ -2154,6 +2154,160 @@ extern "C" int qk_pq_rle_pages(void *stream, uint64_t npages,
   return 0;
 }
 
+// ---- device string dictionary ------------------------------------------
+// General variable-width string keys for join/group-by (the reference
+// joins/groups on arbitrary key types via polars, sql_executors.py:
+// 325-377/:556-599; round 1 supported only <=256-entry host dicts).
+// Design: an open-addressing table keyed by a 64-bit byte hash with EXACT
+// byte verification against an on-device string ARENA (first occurrence
+// of each distinct string is copied in), assigning dense u32 codes that
+// stay consistent across batches because the table and arena persist.
+// Equal strings => equal codes, distinct => distinct (bytes verified, no
+// hash-collision false sharing), so string joins/group-bys reduce to the
+// existing integer kernels over the codes.
+//
+// slot_hash: u64[cap], 0 = empty (real hash 0 remapped to 1)
+// slot_code: i32[cap], -1 until the owning lane publishes
+// code_off/code_len: per-code arena offset/length (cap_codes entries)
+// arena/arena_cursor: byte arena; HOST guarantees capacity >= cursor +
+//   batch bytes before each launch (no mid-kernel growth path needed)
+__device__ inline uint64_t qk_str_hash(const uint8_t *p, uint32_t len) {
+  uint64_t h = 0x9E3779B97F4A7C15ULL ^ len;
+  uint32_t i = 0;
+  for (; i + 8 <= len; i += 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, p + i, 8);
+    h = splitmix64(h ^ w);
+  }
+  uint64_t tail = 0;
+  for (uint32_t k = 0; i < len; i++, k++)
+    tail |= (uint64_t)p[i] << (8 * k);
+  h = splitmix64(h ^ tail);
+  return h ? h : 1;
+}
+
+__global__ void __launch_bounds__(BLOCK) k_str_dict_encode(
+    uint64_t n, const int64_t *__restrict__ offsets,
+    const uint8_t *__restrict__ bytes, uint64_t *slot_hash,
+    int32_t *slot_code, uint64_t cap, uint64_t *code_off,
+    uint32_t *code_len, uint8_t *arena, uint64_t *arena_cursor,
+    uint32_t *counter, uint32_t *__restrict__ out_codes) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t o0 = offsets[i], o1 = offsets[i + 1];
+    uint32_t len = (uint32_t)(o1 - o0);
+    const uint8_t *p = bytes + o0;
+    uint64_t h = qk_str_hash(p, len);
+    uint64_t s = h & (cap - 1);
+    for (;;) {
+      uint64_t cur = __atomic_load_n((unsigned long long *)&slot_hash[s],
+                                     __ATOMIC_RELAXED);
+      if (cur == 0) {
+        uint64_t prev = atomicCAS((unsigned long long *)&slot_hash[s],
+                                  0ULL, (unsigned long long)h);
+        if (prev == 0) {
+          // this lane owns the slot: copy bytes, assign the code, then
+          // PUBLISH via slot_code (readers spin until != -1)
+          uint64_t aoff =
+              atomicAdd((unsigned long long *)arena_cursor,
+                        (unsigned long long)len);
+          for (uint32_t k = 0; k < len; k++) arena[aoff + k] = p[k];
+          uint32_t code = atomicAdd(counter, 1u);
+          code_off[code] = aoff;
+          code_len[code] = len;
+          __threadfence();
+          __atomic_store_n(&slot_code[s], (int32_t)code,
+                           __ATOMIC_RELEASE);
+          out_codes[i] = code;
+          break;
+        }
+        cur = prev;
+      }
+      if (cur == h) {
+        int32_t code;
+        do {
+          code = __atomic_load_n(&slot_code[s], __ATOMIC_ACQUIRE);
+        } while (code < 0);
+        if (code_len[code] == len) {
+          const uint8_t *q = arena + code_off[code];
+          uint32_t k = 0;
+          while (k < len && q[k] == p[k]) k++;
+          if (k == len) {
+            out_codes[i] = (uint32_t)code;
+            break;
+          }
+        }
+        // same 64-bit hash, different bytes: keep probing
+      }
+      s = (s + 1) & (cap - 1);
+    }
+  }
+}
+extern "C" int qk_str_dict_encode(void *stream, uint64_t n,
+                                  const int64_t *offsets,
+                                  const uint8_t *bytes, uint64_t *slot_hash,
+                                  int32_t *slot_code, uint64_t cap,
+                                  uint64_t *code_off, uint32_t *code_len,
+                                  uint8_t *arena, uint64_t *arena_cursor,
+                                  uint32_t *counter, uint32_t *out_codes) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_str_dict_encode.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_str_dict_encode, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, offsets, bytes, slot_hash,
+                     slot_code, cap, code_off, code_len, arena,
+                     arena_cursor, counter, out_codes);
+  QK_TRY("qk_str_dict_encode", hipGetLastError());
+  return 0;
+}
+
+// Growth path: re-seat (hash, code) pairs into a larger table. Codes and
+// the arena are immutable — previously returned codes stay valid.
+__global__ void __launch_bounds__(BLOCK) k_str_dict_rehash(
+    uint32_t ncodes, const uint64_t *__restrict__ code_off,
+    const uint32_t *__restrict__ code_len, const uint8_t *__restrict__ arena,
+    uint64_t *slot_hash, int32_t *slot_code, uint64_t cap) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t c = blockIdx.x * blockDim.x + threadIdx.x; c < ncodes;
+       c += stride) {
+    uint64_t h = qk_str_hash(arena + code_off[c], code_len[c]);
+    uint64_t s = h & (cap - 1);
+    for (;;) {
+      uint64_t prev = atomicCAS((unsigned long long *)&slot_hash[s], 0ULL,
+                                (unsigned long long)h);
+      if (prev == 0) {
+        __threadfence();
+        __atomic_store_n(&slot_code[s], (int32_t)c, __ATOMIC_RELEASE);
+        break;
+      }
+      s = (s + 1) & (cap - 1);
+    }
+  }
+}
+extern "C" int qk_str_dict_rehash(void *stream, uint32_t ncodes,
+                                  const uint64_t *code_off,
+                                  const uint32_t *code_len,
+                                  const uint8_t *arena, uint64_t *slot_hash,
+                                  int32_t *slot_code, uint64_t cap) {
+  if (!ncodes) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_str_dict_rehash.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks =
+      (uint32_t)qk_min_u64(MAX_BLOCKS, (ncodes + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_str_dict_rehash, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, ncodes, code_off, code_len, arena,
+                     slot_hash, slot_code, cap);
+  QK_TRY("qk_str_dict_rehash", hipGetLastError());
+  return 0;
+}
+
+extern "C" int qk_d2d(void *dst, const void *src, uint64_t nbytes) {
+  QK_TRY("qk_d2d", hipMemcpy(dst, src, nbytes, hipMemcpyDeviceToDevice));
+  return 0;
+}
+
 // ---- GPU snappy page decompression -------------------------------------
 // Parquet SNAPPY-compressed pages (the reference reads them transparently
 // through pyarrow, unordered_readers.py:51; real TPC-H datasets ship
