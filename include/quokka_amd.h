@@ -418,6 +418,34 @@ int qk_pq_plain_copy(void *stream, uint64_t ntiles, const uint64_t *tiles,
  * >= 8 bytes of slack after the end. */
 int qk_pq_rle_pages(void *stream, uint64_t npages, const uint64_t *ents,
                     const uint8_t *src_bytes, uint32_t *out);
+/* ---- device string dictionary ---------------------------------------- *
+ * General variable-width string keys for join/group-by (the reference
+ * joins/groups on arbitrary key types via polars, sql_executors.py:
+ * 325-377/:556-599). Open-addressing table keyed by a 64-bit byte hash
+ * with EXACT byte verification against an on-device arena; assigns dense
+ * u32 codes, consistent across calls because table+arena persist. Equal
+ * strings => equal codes (bytes verified), so string joins/group-bys run
+ * on the existing integer kernels over the codes.
+ * slot_hash u64[cap] zeroed (0 = empty), slot_code i32[cap] filled -1,
+ * code_off u64 / code_len u32 per code, arena + arena_cursor_dev (u64),
+ * counter_dev (u32) = number of codes. HOST must guarantee, before each
+ * call: table load < 1/2 after worst-case n new codes, code arrays hold
+ * counter + n, arena holds cursor + sum(len). */
+int qk_str_dict_encode(void *stream, uint64_t n, const int64_t *offsets,
+                       const uint8_t *bytes, uint64_t *slot_hash,
+                       int32_t *slot_code, uint64_t capacity,
+                       uint64_t *code_off, uint32_t *code_len,
+                       uint8_t *arena, uint64_t *arena_cursor_dev,
+                       uint32_t *counter_dev, uint32_t *out_codes);
+/* Growth: re-seat (hash, code) pairs into a larger zeroed/-1 table;
+ * codes and arena unchanged (previously returned codes stay valid). */
+int qk_str_dict_rehash(void *stream, uint32_t ncodes,
+                       const uint64_t *code_off, const uint32_t *code_len,
+                       const uint8_t *arena, uint64_t *slot_hash,
+                       int32_t *slot_code, uint64_t capacity);
+/* Synchronous device-to-device copy (dict growth, arena relocation). */
+int qk_d2d(void *dst_dev, const void *src_dev, uint64_t nbytes);
+
 /* GPU snappy decompression of Parquet pages (the reference reads
  * compressed files transparently through pyarrow,
  * unordered_readers.py:51). One wave per page, pages independent.

@@ -41,6 +41,23 @@ def _to_table(d):
     return pa.table({k: v for k, v in d.items()})
 
 
+def _is_stringish(col):
+    """pyarrow string / large_string / dictionary-of-string column."""
+    import pyarrow as pa
+    t = col.type
+    if pa.types.is_dictionary(t):
+        t = t.value_type
+    return pa.types.is_string(t) or pa.types.is_large_string(t)
+
+
+def _concat_col(parts):
+    import pyarrow as pa
+    chunks = []
+    for p in parts:
+        chunks.extend(p.chunks if isinstance(p, pa.ChunkedArray) else [p])
+    return pa.chunked_array(chunks)
+
+
 class GPUBuildProbeJoinExecutor(Executor):
     """GPU replacement for BuildProbeJoinExecutor (sql_executors.py:325-377).
 
@@ -70,6 +87,8 @@ class GPUBuildProbeJoinExecutor(Executor):
         self._table = None
         self._build_cols = None     # dict name -> list of numpy arrays
         self._build_names = None
+        self._key_dict = None       # DeviceStringDict for string keys
+        self._build_payload_host = {}
 
     def __getstate__(self):
         assert self._table is None, "executor must be pickled before first execute"
@@ -90,8 +109,15 @@ class GPUBuildProbeJoinExecutor(Executor):
                 self._build_names = [c for c in batch.column_names]
                 self._build_cols = {c: [] for c in self._build_names}
             for c in self._build_names:
-                self._build_cols[c].append(
-                    staging.column_to_numpy(batch.column(c)))
+                col = batch.column(c)
+                if _is_stringish(col):
+                    # string columns stay Arrow until _finish_build:
+                    # KEYS go through the device string dictionary,
+                    # payloads are host-gathered at emit
+                    self._build_cols[c].append(col)
+                else:
+                    self._build_cols[c].append(
+                        staging.column_to_numpy(col))
             return
 
         # probe
@@ -106,15 +132,43 @@ class GPUBuildProbeJoinExecutor(Executor):
 
     def _finish_build(self):
         ops, shim, staging = _lazy_gpu()
-        self._host_build = {c: np.concatenate(v)
-                            for c, v in self._build_cols.items()}
+        self._host_build = {}
+        self._build_payload_host = {}     # string payloads, host-gathered
+        for c, v in self._build_cols.items():
+            if v and not isinstance(v[0], np.ndarray):
+                col = _concat_col(v)
+                if c == self.right_on:
+                    # string join key: device string dictionary -> dense
+                    # codes; probe side encodes into the SAME dict so
+                    # equal strings get equal codes (bytes verified)
+                    self._key_dict = ops.DeviceStringDict(
+                        expected=max(1024, col.length()))
+                    self._host_build[c] = self._key_dict.encode_column(
+                        col).astype(np.int64)
+                else:
+                    self._build_payload_host[c] = np.asarray(
+                        col.to_pylist(), dtype=object)
+            else:
+                arr = np.concatenate(v)
+                if arr.dtype == object:   # broadcast-join host strings
+                    import pyarrow as pa
+                    if c == self.right_on:
+                        self._key_dict = ops.DeviceStringDict(
+                            expected=max(1024, len(arr)))
+                        self._host_build[c] = self._key_dict.encode_column(
+                            pa.chunked_array([pa.array(arr)])
+                        ).astype(np.int64)
+                    else:
+                        self._build_payload_host[c] = arr
+                else:
+                    self._host_build[c] = arr
         keys = self._host_build[self.right_on]
         if keys.dtype != np.int64:
             if keys.dtype.kind in "iu":   # any int key, as the reference
                 keys = keys.astype(np.int64)
             else:
-                raise TypeError("GPU join requires integer keys, got %s"
-                                % keys.dtype)
+                raise TypeError("GPU join requires integer or string keys,"
+                                " got %s" % keys.dtype)
         n = len(keys)
         self._table = ops.JoinTable(max(16, n))
         if n:
@@ -123,7 +177,7 @@ class GPUBuildProbeJoinExecutor(Executor):
             kcol.free()
         self._build_payload_dev = {}
         for c in self._build_names:
-            if c == self.right_on:
+            if c == self.right_on or c in self._build_payload_host:
                 continue
             self._build_payload_dev[c] = shim.DevColumn.from_numpy(
                 self._host_build[c])
@@ -131,13 +185,23 @@ class GPUBuildProbeJoinExecutor(Executor):
 
     def _probe(self, batch):
         ops, shim, staging = _lazy_gpu()
-        probe_keys = staging.column_to_numpy(batch.column(self.left_on))
+        key_col = batch.column(self.left_on)
+        if _is_stringish(key_col):
+            if self._key_dict is None:
+                raise TypeError("string probe keys against a non-string "
+                                "build side")
+            # same device dictionary as the build side: equal strings ==
+            # equal codes (bytes verified in-kernel)
+            probe_keys = self._key_dict.encode_column(
+                key_col).astype(np.int64)
+        else:
+            probe_keys = staging.column_to_numpy(key_col)
         if probe_keys.dtype != np.int64:
             if probe_keys.dtype.kind in "iu":
                 probe_keys = probe_keys.astype(np.int64)
             else:
-                raise TypeError("GPU join requires integer keys, got %s"
-                                % probe_keys.dtype)
+                raise TypeError("GPU join requires integer or string keys,"
+                                " got %s" % probe_keys.dtype)
         kcol = shim.DevColumn.from_numpy(probe_keys)
         mode = {"inner": 0, "left": 0, "semi": 1, "anti": 2}[self.how]
         pidx, bidx, nm = self._table.probe(kcol, mode=mode)
@@ -160,20 +224,31 @@ class GPUBuildProbeJoinExecutor(Executor):
                 cols[self.right_on] = cols.pop(self.left_on)
             return cols
 
+        sel = pidx.to_numpy(nm)
+
+        def probe_col_rows(c, rows_dev, rows_host):
+            """One probe-side column at the selected rows: strings gather
+            host-side (emit is host Arrow anyway), numerics through the
+            device gather like the reference's polars emit."""
+            col = batch.column(c)
+            if _is_stringish(col):
+                return np.asarray(col.to_pylist(), dtype=object)[rows_host]
+            return dev_gather(staging.column_to_numpy(col), rows_dev)
+
         out = {}
         if self.how in ("semi", "anti"):
             for c in batch.column_names:
-                out[c] = dev_gather(
-                    staging.column_to_numpy(batch.column(c)), pidx)
+                out[c] = probe_col_rows(c, pidx, sel)
         else:
-            sel = pidx.to_numpy(nm)
             for c in batch.column_names:
-                out[c] = dev_gather(
-                    staging.column_to_numpy(batch.column(c)), pidx)
+                out[c] = probe_col_rows(c, pidx, sel)
+            bsel = bidx.to_numpy(nm)
             for c, dev in self._build_payload_dev.items():
                 g = dev.gather(bidx, nm)
                 out[c] = g.to_numpy(nm)
                 g.free()
+            for c, arr in self._build_payload_host.items():
+                out[c] = arr[bsel]
             if self.how == "left":
                 # unmatched probe rows appended with null build payload
                 matched = np.zeros(len(probe_keys), dtype=bool)
@@ -182,18 +257,23 @@ class GPUBuildProbeJoinExecutor(Executor):
                 if len(un):
                     import pyarrow as pa
                     for c in batch.column_names:
-                        out[c] = np.concatenate(
-                            [out[c], staging.column_to_numpy(batch.column(c))[un]])
+                        col = batch.column(c)
+                        if _is_stringish(col):
+                            tailv = np.asarray(col.to_pylist(),
+                                               dtype=object)[un]
+                        else:
+                            tailv = staging.column_to_numpy(col)[un]
+                        out[c] = np.concatenate([out[c], tailv])
                     probe_cols = rename_key(
                         {k: v for k, v in out.items()
                          if k in batch.column_names})
                     tbl = _to_table(probe_cols)
                     pay = {}
-                    for c in self._build_payload_dev:
+                    for c in (list(self._build_payload_dev) +
+                              list(self._build_payload_host)):
+                        a = pa.array(out[c])
                         pay[c] = pa.chunked_array([
-                            pa.array(out[c]),
-                            pa.nulls(len(un), pa.from_numpy_dtype(out[c].dtype)),
-                        ])
+                            a, pa.nulls(len(un), a.type)])
                     for c, v in pay.items():
                         tbl = tbl.append_column(c, v)
                     pidx.free()
@@ -295,14 +375,22 @@ class GPUAggExecutor(Executor):
         arrs = []
         for k in self.groupby_keys:
             col = batch.column(k)
-            sd = self._key_state["dicts"].setdefault(k, staging.StringDict())
-            try:
-                arr = staging.column_to_numpy(col, sd)
-            except TypeError:
+            if _is_stringish(col):
+                # device string dictionary: unbounded cardinality, codes
+                # consistent across batches (replaces the <=256-entry
+                # host StringDict of round 1)
+                sd = self._key_state["dicts"].get(k)
+                if sd is None:
+                    sd = ops.DeviceStringDict(expected=4096)
+                    self._key_state["dicts"][k] = sd
+                arr = sd.encode_column(col).astype(np.int64)
+            else:
                 arr = staging.column_to_numpy(col)
             arrs.append(arr)
         if len(arrs) == 1 and arrs[0].dtype == np.int64:
             self._key_state["mode"] = "passthrough"
+            self._key_state["pass_dict"] = self._key_state["dicts"].get(
+                self.groupby_keys[0])
             return arrs[0]
         # composite: each non-i64 key coded via np codebook, packed base-N
         self._key_state["mode"] = "composite"
@@ -388,7 +476,9 @@ class GPUAggExecutor(Executor):
             if self._key_state and self._key_state.get("mode") == "composite":
                 out.update(self._decode_keys(keys))
             else:
-                out[self.groupby_keys[0]] = keys
+                sd = (self._key_state or {}).get("pass_dict")
+                out[self.groupby_keys[0]] = (sd.decode(keys) if sd
+                                             else keys)
         for alias, expr in self.exprs:
             out[alias] = eval(expr, {"s": s})  # noqa: S307 — expr built above
         if self.orderby_keys:

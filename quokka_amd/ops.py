@@ -272,6 +272,166 @@ class GroupByI64:
             self._ops_dev.free()
 
 
+class DeviceStringDict:
+    """Device-resident accumulating string dictionary: arbitrary-width,
+    unbounded-cardinality string keys -> dense u32 codes, consistent
+    across batches (the table and byte arena persist; growth re-seats
+    hashes without changing codes). Replaces the round-1 host
+    `staging.StringDict` (<=256 values) for join/group-by keys — the
+    reference handles such keys inside polars (sql_executors.py:325-377,
+    :556-599).
+
+    encode(offsets, bytes) takes Arrow-layout string data (int64 offsets
+    of n+1 entries + the byte buffer) and returns np.uint32 codes[n].
+    `values` materializes decoded strings lazily (host list, index ==
+    code)."""
+
+    def __init__(self, expected=1024, stream=None):
+        self.stream = stream
+        self.cap = _pow2_at_least(max(64, 4 * int(expected)))
+        self._alloc_table(self.cap)
+        self.code_cap = max(1024, self.cap // 2)
+        self.code_off = DevColumn(np.uint64, self.code_cap)
+        self.code_len = DevColumn(np.uint32, self.code_cap)
+        self.arena_cap = 1 << 20
+        self.arena = DevBuffer(self.arena_cap)
+        self._ctr = DevBuffer(16)           # [arena_cursor u64, counter u32]
+        shim.call("qk_dmemset", self._ctr.ptr, 0, c_u64(16))
+        self.n_codes = 0
+        self.arena_used = 0
+        self._values = []                   # host cache, len == decoded so far
+
+    def _alloc_table(self, cap):
+        self.slot_hash = DevColumn(np.uint64, cap)
+        self.slot_code = DevColumn(np.int32, cap)
+        shim.call("qk_dmemset", self.slot_hash.ptr, 0, c_u64(cap * 8))
+        shim.call("qk_dmemset", self.slot_code.ptr, 0xFF, c_u64(cap * 4))
+
+    @property
+    def _cursor_ptrs(self):
+        import ctypes as _ct
+        return (self._ctr.ptr,
+                shim.c_vp(self._ctr.ptr.value + 8))
+
+    def encode(self, offsets, data, stream=None):
+        """offsets: np.int64[n+1] (or any int np array), data: np.uint8
+        bytes buffer. Returns np.uint32[n] codes."""
+        st = stream or self.stream
+        sh = st.handle if st else None
+        offsets = np.ascontiguousarray(offsets, dtype=np.int64)
+        n = len(offsets) - 1
+        if n <= 0:
+            return np.empty(0, dtype=np.uint32)
+        data = np.ascontiguousarray(data, dtype=np.uint8)
+        # worst case: every row is a new distinct string
+        self._ensure(n, int(offsets[-1] - offsets[0]), sh)
+        doff = DevColumn.from_numpy(offsets - offsets[0])
+        dbytes = DevColumn.from_numpy(
+            data[int(offsets[0]):int(offsets[-1])] if offsets[0] else
+            data[: int(offsets[-1])])
+        out = DevColumn(np.uint32, n)
+        acur, cnt = self._cursor_ptrs
+        shim.call("qk_str_dict_encode", sh, c_u64(n), doff.ptr, dbytes.ptr,
+                  self.slot_hash.ptr, self.slot_code.ptr, c_u64(self.cap),
+                  self.code_off.ptr, self.code_len.ptr, self.arena.ptr,
+                  acur, cnt, out.ptr)
+        if st:
+            st.sync()
+        else:
+            shim.call("qk_stream_sync", None)
+        host = np.zeros(2, dtype=np.uint64)
+        shim.call("qk_d2h", host.ctypes.data_as(c_vp), self._ctr.ptr,
+                  c_u64(16))
+        self.arena_used = int(host[0])
+        self.n_codes = int(np.uint64(host[1]) & np.uint64(0xFFFFFFFF))
+        codes = out.to_numpy(n)
+        doff.free(); dbytes.free(); out.free()
+        return codes
+
+    def encode_column(self, col, stream=None):
+        """pyarrow string/large_string column -> np.uint32 codes."""
+        import pyarrow as pa
+        if isinstance(col, pa.ChunkedArray):
+            col = col.combine_chunks()
+        if pa.types.is_dictionary(col.type):
+            col = col.dictionary_decode() if hasattr(col, "dictionary_decode") \
+                else col.cast(col.type.value_type)
+        if col.null_count:
+            raise TypeError("null string keys unsupported")
+        bufs = col.buffers()
+        width = 8 if pa.types.is_large_string(col.type) else 4
+        odt = np.int64 if width == 8 else np.int32
+        off = np.frombuffer(bufs[1], dtype=odt,
+                            count=len(col) + 1 + col.offset)[col.offset:]
+        data = np.frombuffer(bufs[2], dtype=np.uint8)
+        return self.encode(off.astype(np.int64), data, stream)
+
+    def _ensure(self, n_new, nbytes_new, sh):
+        need_codes = self.n_codes + n_new
+        if 2 * need_codes > self.cap:
+            new_cap = _pow2_at_least(4 * need_codes)
+            self.slot_hash.free(); self.slot_code.free()
+            self._alloc_table(new_cap)
+            self.cap = new_cap
+            shim.call("qk_str_dict_rehash", sh, c_u32(self.n_codes),
+                      self.code_off.ptr, self.code_len.ptr, self.arena.ptr,
+                      self.slot_hash.ptr, self.slot_code.ptr,
+                      c_u64(self.cap))
+        if need_codes > self.code_cap:
+            new_cc = _pow2_at_least(2 * need_codes)
+            for name in ("code_off", "code_len"):
+                old = getattr(self, name)
+                new = DevColumn(old.dtype, new_cc)
+                if self.n_codes:
+                    shim.call("qk_d2d", new.ptr, old.ptr,
+                              c_u64(self.n_codes * old.dtype.itemsize))
+                old.free()
+                setattr(self, name, new)
+            self.code_cap = new_cc
+        if self.arena_used + nbytes_new > self.arena_cap:
+            new_ac = max(2 * self.arena_cap,
+                         self.arena_used + nbytes_new)
+            new_ar = DevBuffer(new_ac)
+            if self.arena_used:
+                shim.call("qk_d2d", new_ar.ptr, self.arena.ptr,
+                          c_u64(self.arena_used))
+            self.arena.free()
+            self.arena = new_ar
+            self.arena_cap = new_ac
+
+    @property
+    def values(self):
+        """Decoded strings, index == code (lazily materialized)."""
+        if len(self._values) < self.n_codes:
+            k0 = len(self._values)
+            k = self.n_codes - k0
+            offs = self.code_off.to_numpy(self.n_codes)[k0:]
+            lens = self.code_len.to_numpy(self.n_codes)[k0:]
+            lo = int(offs.min()) if k else 0
+            hi = int((offs + lens).max()) if k else 0
+            blob = np.empty(max(1, hi - lo), dtype=np.uint8)
+            if hi > lo:
+                shim.call("qk_d2h", blob.ctypes.data_as(c_vp),
+                          shim.c_vp(self.arena.ptr.value + lo),
+                          c_u64(hi - lo))
+            raw = blob.tobytes()
+            for o, ln in zip(offs, lens):
+                s = int(o) - lo
+                self._values.append(raw[s:s + int(ln)].decode())
+        return self._values
+
+    def decode(self, codes):
+        vals = np.asarray(self.values, dtype=object)
+        return vals[np.asarray(codes, dtype=np.int64)]
+
+    def free(self):
+        for c in (self.slot_hash, self.slot_code, self.code_off,
+                  self.code_len):
+            c.free()
+        self.arena.free()
+        self._ctr.free()
+
+
 def sort_permutation(col, stream=None, descending=False):
     """Stable sort permutation of a DevColumn (i64/f64/i32/u32 keys):
     returns a u32 DevColumn `perm` with rows in ascending (or descending)
