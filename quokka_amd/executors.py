@@ -370,6 +370,10 @@ class GPUAggExecutor(Executor):
         """Multiple group keys -> one i64 composite via accumulated host
         codebooks (DESIGN.md §Group keys). Single i64 key passes through."""
         ops, shim, staging = _lazy_gpu()
+        if len(self.groupby_keys) > 3:
+            raise ValueError(
+                "composite group keys: at most 3 keys fit 63 bits at 21 "
+                "bits/key (got %d)" % len(self.groupby_keys))
         if self._key_state is None:
             self._key_state = {"dicts": {}, "decode": []}
         arrs = []
@@ -394,10 +398,6 @@ class GPUAggExecutor(Executor):
             return arrs[0]
         # composite: each non-i64 key coded via np codebook, packed base-N
         self._key_state["mode"] = "composite"
-        if len(self.groupby_keys) > 3:
-            raise ValueError(
-                "composite group keys: at most 3 keys fit 63 bits at 21 "
-                "bits/key (got %d)" % len(self.groupby_keys))
         packed = np.zeros(len(arrs[0]), dtype=np.int64)
         widths = []
         comps = []

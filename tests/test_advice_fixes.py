@@ -32,14 +32,19 @@ def test_composite_key_too_many_keys_raises():
 
 
 def test_composite_key_codebook_overflow_raises():
+    # int32 keys take the composite-codebook path without the (GPU-only)
+    # device string dictionary; the 21-bit bound under test is the same
     import pyarrow as pa
     ex = GPUAggExecutor(["a", "b"], [], "sum(x) as s")
-    t = pa.table({"a": ["u", "v"], "b": ["p", "q"], "x": [1.0, 2.0]})
+    t = pa.table({"a": pa.array([7, 8], type=pa.int32()),
+                  "b": pa.array([1, 2], type=pa.int32()),
+                  "x": [1.0, 2.0]})
     ex._encode_keys(t)   # initialise state
     # simulate a codebook that has already hit the 21-bit field limit
-    # (keys offset so the incoming StringDict code is NOT already present)
+    # (keys offset so the incoming value is NOT already present)
     ex._key_state["codebooks"]["a"] = {i + 1000: i for i in range(1 << 21)}
-    t2 = pa.table({"a": ["brand-new"], "b": ["p"], "x": [3.0]})
+    t2 = pa.table({"a": pa.array([999], type=pa.int32()),
+                   "b": pa.array([1], type=pa.int32()), "x": [3.0]})
     with pytest.raises(ValueError, match="2\\*\\*21"):
         ex._encode_keys(t2)
 
