@@ -20,7 +20,7 @@ def gpu():
 
 
 def _rand_strings(rng, n, n_distinct, minlen=1, maxlen=24):
-    pool = np.array(["s%0*d" % (rng.integers(minlen, maxlen), i)
+    pool = np.array(["s%0*d" % (int(rng.integers(minlen, maxlen)), i)
                      for i in range(n_distinct)], dtype=object)
     return pool[rng.integers(0, n_distinct, n)]
 
