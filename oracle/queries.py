@@ -144,6 +144,95 @@ def q3(li, orders, customer, limit=10):
     return full, top10
 
 
+Q4_LO = G.days(1993, 7, 1)
+Q4_HI = G.days(1993, 10, 1)     # date + interval '3' month
+Q10_LO = G.days(1993, 10, 1)
+Q10_HI = G.days(1994, 1, 1)     # date + interval '3' month
+
+
+def q4(li, orders):
+    """tpch_ref.py:117-140: orders in [1993-07-01, +3mo) with EXISTS a
+    line l_orderkey = o_orderkey and l_commitdate < l_receiptdate;
+    count(*) group by o_orderpriority, order by o_orderpriority.
+    Returns dict o_orderpriority(str) -> count."""
+    lmask = li["l_commitdate"] < li["l_receiptdate"]
+    nkey = int(orders["o_orderkey"].max()) + 2
+    has_late = np.zeros(nkey, dtype=bool)
+    has_late[li["l_orderkey"][lmask]] = True
+    omask = ((orders["o_orderdate"] >= Q4_LO)
+             & (orders["o_orderdate"] < Q4_HI)
+             & has_late[orders["o_orderkey"]])
+    pr = orders["o_orderpriority"][omask]
+    counts = np.bincount(pr, minlength=len(G.ORDERPRIORITY))
+    return {G.ORDERPRIORITY[i]: int(counts[i])
+            for i in range(len(G.ORDERPRIORITY)) if counts[i]}
+
+
+def q18(li, orders, customer, limit=100):
+    """tpch_ref.py:544-580: orders whose lines sum(l_quantity) > 300,
+    joined to customer; group by (c_name, c_custkey, o_orderkey,
+    o_orderdate, o_totalprice) sum(l_quantity); order by o_totalprice
+    desc, o_orderdate asc, limit 100. The group key includes o_orderkey,
+    so each group IS one qualifying order. Ties beyond the reference's
+    ordering broken by o_orderkey asc for fixture determinism."""
+    nkey = int(orders["o_orderkey"].max()) + 2
+    qty_by_key = np.bincount(li["l_orderkey"], weights=li["l_quantity"],
+                             minlength=nkey)
+    ok = orders["o_orderkey"]
+    qual = qty_by_key[ok] > 300.0
+    rows = np.nonzero(qual)[0]
+    cust = orders["o_custkey"][rows]
+    names = (customer["c_name"][cust - 1] if "c_name" in customer
+             else np.array(["Customer#%09d" % c for c in cust],
+                           dtype=object))
+    out = {
+        "c_name": names,
+        "c_custkey": cust,
+        "o_orderkey": ok[rows],
+        "o_orderdate": orders["o_orderdate"][rows],
+        "o_totalprice": orders["o_totalprice"][rows],
+        "sum_qty": qty_by_key[ok[rows]],
+    }
+    order = np.lexsort((out["o_orderkey"], out["o_orderdate"],
+                        -out["o_totalprice"]))
+    top = order[:limit]
+    return {k: v[top] for k, v in out.items()}
+
+
+def q10(li, orders, customer, nation, limit=20):
+    """tpch_ref.py:306-342: returned lines (l_returnflag = 'R') of orders
+    in [1993-10-01, +3mo), revenue per customer with the customer's
+    nation name and string attributes; order by revenue desc limit 20.
+    The 7-column group key is functionally dependent on c_custkey.
+    Ties broken by c_custkey asc for fixture determinism."""
+    rflag_r = G.RETURNFLAG.index("R")
+    omask = ((orders["o_orderdate"] >= Q10_LO)
+             & (orders["o_orderdate"] < Q10_HI))
+    nkey = int(orders["o_orderkey"].max()) + 2
+    order_cust = np.zeros(nkey, dtype=np.int64)
+    order_cust[orders["o_orderkey"][omask]] = orders["o_custkey"][omask]
+    lmask = (li["l_returnflag"] == rflag_r) & \
+        (order_cust[li["l_orderkey"]] > 0)
+    cust = order_cust[li["l_orderkey"][lmask]]
+    rev = li["l_extendedprice"][lmask] * (1.0 - li["l_discount"][lmask])
+    ncust = int(customer["c_custkey"].max()) + 2
+    rev_by_cust = np.bincount(cust, weights=rev, minlength=ncust)
+    ck = np.nonzero(rev_by_cust > 0)[0]
+    row = ck - 1                                  # custkey is dense 1..N
+    out = {
+        "c_custkey": ck.astype(np.int64),
+        "revenue": rev_by_cust[ck],
+        "c_acctbal": customer["c_acctbal"][row],
+        "n_name": nation["n_name"][customer["c_nationkey"][row]],
+    }
+    for c in ("c_name", "c_address", "c_phone", "c_comment"):
+        if c in customer:
+            out[c] = customer[c][row]
+    order = np.lexsort((out["c_custkey"], -out["revenue"]))
+    top = order[:limit]
+    return {k: v[top] for k, v in out.items()}
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

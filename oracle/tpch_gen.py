@@ -56,6 +56,8 @@ def days(y, m, d):
 RETURNFLAG = ["A", "N", "R"]          # l_returnflag codes 0,1,2
 LINESTATUS = ["F", "O"]               # l_linestatus codes 0,1
 MKTSEGMENT = ["AUTOMOBILE", "BUILDING", "FURNITURE", "HOUSEHOLD", "MACHINERY"]
+ORDERPRIORITY = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED",
+                 "5-LOW"]             # spec 4.2.3 uniform, sorted order
 
 # TPC-H spec nation table: (name, regionkey)
 NATIONS = [
@@ -124,6 +126,12 @@ def gen_orders(sf, seed=42):
         "o_custkey": ck,
         "o_orderdate": rng.integers(ORDERDATE_LO, ORDERDATE_HI + 1, n).astype(np.int32),
         "o_shippriority": np.zeros(n, dtype=np.int32),
+        # spec 4.2.3: O_ORDERPRIORITY random over the 5 priorities (u8
+        # codes into ORDERPRIORITY). O_TOTALPRICE is derived from the
+        # order's lines; gen_lineitem fills it in when given this dict
+        # (zeros until then).
+        "o_orderpriority": rng.integers(0, 5, n).astype(np.uint8),
+        "o_totalprice": np.zeros(n, dtype=np.float64),
     }
 
 
@@ -146,12 +154,19 @@ def gen_lineitem(sf, seed=42, orders=None):
     discount = rng.integers(0, 11, n).astype(np.float64) / 100.0
     tax = rng.integers(0, 9, n).astype(np.float64) / 100.0
     shipdate = o_date_rep + rng.integers(1, 122, n)
+    commitdate = o_date_rep + rng.integers(30, 91, n)  # spec: +U[30,90]
     receiptdate = shipdate + rng.integers(1, 31, n)
 
     returned = receiptdate <= RECEIPT_CUTOFF
     ra = rng.integers(0, 2, n)                        # 0 -> 'R', 1 -> 'A'
     returnflag = np.where(returned, np.where(ra == 0, 2, 0), 1).astype(np.uint8)
     linestatus = (shipdate > RECEIPT_CUTOFF).astype(np.uint8)  # 1='O', 0='F'
+
+    # spec 4.2.3: O_TOTALPRICE = sum over the order's lines of
+    # extendedprice * (1 + tax) * (1 - discount)
+    line_total = extendedprice * (1.0 + tax) * (1.0 - discount)
+    orders["o_totalprice"] = np.add.reduceat(
+        line_total, np.concatenate(([0], np.cumsum(lines_per_order)[:-1])))
 
     return {
         "l_orderkey": l_orderkey,
@@ -163,19 +178,41 @@ def gen_lineitem(sf, seed=42, orders=None):
         "l_returnflag": returnflag,
         "l_linestatus": linestatus,
         "l_shipdate": shipdate.astype(np.int32),
+        "l_commitdate": commitdate.astype(np.int32),
+        "l_receiptdate": receiptdate.astype(np.int32),
     }
 
 
-def gen_customer(sf, seed=42):
+def gen_customer(sf, seed=42, strings=False):
     """customer columns: c_custkey i64 (dense 1..N), c_mktsegment u8 code,
-    c_nationkey i32."""
+    c_nationkey i32, c_acctbal f64 (spec U[-999.99, 9999.99]). With
+    strings=True also c_name ('Customer#%09d', the spec format), c_phone
+    (spec 4.2.2.9 country-code format) and placeholder c_address/
+    c_comment (spec text grammar out of scope, tpch_gen header) — the
+    Q10 output attributes, all functionally dependent on c_custkey."""
     n = n_customers(sf)
     rng = np.random.default_rng([seed, 3])
-    return {
+    out = {
         "c_custkey": np.arange(1, n + 1, dtype=np.int64),
         "c_mktsegment": rng.integers(0, len(MKTSEGMENT), n).astype(np.uint8),
         "c_nationkey": rng.integers(0, 25, n).astype(np.int32),
+        "c_acctbal": rng.integers(-99999, 1000000, n) / 100.0,
     }
+    if strings:
+        ck = out["c_custkey"]
+        nk = out["c_nationkey"]
+        local = rng.integers(100, 1000, (n, 3))
+        out["c_name"] = np.array(
+            ["Customer#%09d" % k for k in ck], dtype=object)
+        out["c_phone"] = np.array(
+            ["%d-%d-%d-%d" % (10 + nk[i], local[i, 0], local[i, 1],
+                              local[i, 2]) for i in range(n)], dtype=object)
+        out["c_address"] = np.array(
+            ["addr#%d.%d" % (k, int(rng.integers(0, 1 << 30)))
+             for k in ck], dtype=object)
+        out["c_comment"] = np.array(
+            ["comment#%d" % k for k in ck], dtype=object)
+    return out
 
 
 def gen_supplier(sf, seed=42):
@@ -191,6 +228,7 @@ def gen_nation():
     return {
         "n_nationkey": np.arange(25, dtype=np.int32),
         "n_regionkey": np.array([r for _, r in NATIONS], dtype=np.int32),
+        "n_name": np.array([n for n, _ in NATIONS], dtype=object),
     }
 
 

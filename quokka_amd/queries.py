@@ -613,6 +613,184 @@ def q5_fused(li_cols, ord_cols, cust_cols, supp_cols, stream=None):
     return out
 
 
+Q4_LO = 8582           # 1993-07-01
+Q4_HI = 8674           # 1993-10-01 (+3 months)
+Q10_LO = 8674          # 1993-10-01
+Q10_HI = 8766          # 1994-01-01 (+3 months)
+ORDERPRIORITY = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED",
+                 "5-LOW"]
+
+
+def q4(li_cols, ord_cols, stream=None):
+    """Device Q4 (tpch_ref.py:117-140) composed from the generic
+    operators + JIT: EXISTS == semi-join of the date-filtered orders
+    against lines with l_commitdate < l_receiptdate; count(*) by
+    o_orderpriority runs as a JIT fused group-aggregate over the semi
+    survivors. Returns dict priority(str) -> count, priority ascending."""
+    from . import jit, ops
+    st = stream
+    # lines with commitdate < receiptdate -> build table on l_orderkey
+    lf = jit.JitFilter("l_commitdate < l_receiptdate",
+                       {k: v.dtype for k, v in li_cols.items()})
+    lidx, ln = lf.run(li_cols, st)
+    lkeys = li_cols["l_orderkey"].gather(lidx, ln, st)
+    tab = ops.JoinTable(max(16, ln), st)
+    if ln:
+        tab.build(lkeys)
+    # orders date window
+    of = jit.JitFilter(
+        "o_orderdate >= date '1993-07-01' and "
+        "o_orderdate < date '1993-07-01' + interval '3' month",
+        {k: v.dtype for k, v in ord_cols.items()})
+    oidx, on = of.run(ord_cols, st)
+    okeys = ord_cols["o_orderkey"].gather(oidx, on, st)
+    sidx, _, ns = tab.probe(okeys, mode=1, n=on)       # semi
+    oprio_f = ord_cols["o_orderpriority"].gather(oidx, on, st)
+    oprio_s = oprio_f.gather(sidx, ns, st)
+    agg = jit.JitAggregate({"o_orderpriority": np.dtype(np.uint8)},
+                           [("o_orderpriority", 5)],
+                           ["COUNT(*) as order_count"])
+    acc = agg.make_acc()
+    if ns:
+        agg.run({"o_orderpriority": oprio_s}, acc, st)
+    if st:
+        st.sync()
+    counts = agg.read(acc)[:, 0]
+    out = {ORDERPRIORITY[i]: int(counts[i]) for i in range(5)
+           if counts[i] > 0}
+    for c in (lidx, lkeys, oidx, okeys, sidx, oprio_f, oprio_s):
+        c.free()
+    acc.free()
+    agg.free()
+    lf.free()
+    of.free()
+    tab.free()
+    return out
+
+
+def q18(li_cols, ord_cols, cust_names=None, stream=None, limit=100):
+    """Device Q18 (tpch_ref.py:544-580): group lineitem by l_orderkey
+    sum(l_quantity) on the device group-by table (unbounded cardinality,
+    grows), qualify sum > 300, then attach the order row (the group key
+    contains o_orderkey, so each group IS one order) by probing the
+    qualifying keys against a device table over the filtered orders;
+    top-100 by (o_totalprice desc, o_orderdate asc). cust_names: optional
+    np array custkey-1 -> name (Q18's c_name via dense custkey)."""
+    from . import ops
+    st = stream
+    n = li_cols["l_orderkey"].n
+    gb = ops.GroupByI64(expected_groups=max(1024, n // 4), nvals=1,
+                        stream=st)
+    gb.update(li_cols["l_orderkey"], [li_cols["l_quantity"]], n)
+    keys, sums = gb.extract()
+    qual = sums[0] > 300.0
+    qkeys = keys[qual]
+    qsums = sums[0][qual]
+    gb.free()
+    # join qualifying orderkeys -> order rows (device probe over orders)
+    otab = ops.JoinTable(max(16, len(qkeys)), st)
+    if len(qkeys):
+        kcol = DevColumn.from_numpy(qkeys)
+        otab.build(kcol)
+        kcol.free()
+    pidx, bidx, nm = otab.probe(ord_cols["o_orderkey"], mode=1)
+    rows = np.sort(pidx.to_numpy(nm))
+    rcol = DevColumn.from_numpy(rows.astype(np.uint32))
+    ok = ord_cols["o_orderkey"].gather(rcol, nm, st)
+    od = ord_cols["o_orderdate"].gather(rcol, nm, st)
+    tp = ord_cols["o_totalprice"].gather(rcol, nm, st)
+    ck = ord_cols["o_custkey"].gather(rcol, nm, st)
+    if st:
+        st.sync()
+    okeys_h = ok.to_numpy(nm)
+    qmap = {int(k): s for k, s in zip(qkeys, qsums)}
+    cust_h = ck.to_numpy(nm)
+    out = {
+        "c_name": (cust_names[cust_h - 1] if cust_names is not None else
+                   np.array(["Customer#%09d" % c for c in cust_h],
+                            dtype=object)),
+        "c_custkey": cust_h,
+        "o_orderkey": okeys_h,
+        "o_orderdate": od.to_numpy(nm),
+        "o_totalprice": tp.to_numpy(nm),
+        "sum_qty": np.array([qmap[int(k)] for k in okeys_h]),
+    }
+    order = np.lexsort((out["o_orderkey"], out["o_orderdate"],
+                        -out["o_totalprice"]))
+    top = order[:limit]
+    for c in (pidx, rcol, ok, od, tp, ck):
+        c.free()
+    otab.free()
+    return {k: v[top] for k, v in out.items()}
+
+
+def q10(li_cols, ord_cols, cust_cols, cust_host, nation_names,
+        stream=None, limit=20):
+    """Device Q10 (tpch_ref.py:306-342): date-filtered orders build a
+    key->custkey table; returned lines (l_returnflag == 'R') probe it;
+    revenue accumulates per CUSTKEY on the device group-by (the 7-column
+    reference group key is functionally dependent on c_custkey); the
+    top-20 customers' string attributes (c_name/address/phone/comment,
+    n_name) attach host-side from `cust_host` (dense custkey -> row).
+    li_cols needs l_orderkey/l_returnflag/l_extendedprice/l_discount;
+    ord_cols o_orderkey/o_custkey/o_orderdate; cust_cols c_nationkey,
+    c_acctbal (device, dense by custkey)."""
+    from . import jit, ops
+    st = stream
+    # orders in the window -> table orderkey -> custkey
+    of = jit.JitFilter(
+        "o_orderdate >= date '1993-10-01' and "
+        "o_orderdate < date '1993-10-01' + interval '3' month",
+        {k: v.dtype for k, v in ord_cols.items()})
+    oidx, on = of.run(ord_cols, st)
+    okeys = ord_cols["o_orderkey"].gather(oidx, on, st)
+    ocust = ord_cols["o_custkey"].gather(oidx, on, st)
+    otab = ops.JoinTable(max(16, on), st)
+    if on:
+        otab.build(okeys)
+    # returned lines
+    lf = jit.JitFilter("l_returnflag = 2",   # code of 'R' (sorted dict)
+                       {k: v.dtype for k, v in li_cols.items()})
+    lidx, ln = lf.run(li_cols, st)
+    lkeys = li_cols["l_orderkey"].gather(lidx, ln, st)
+    pidx, bidx, nm = otab.probe(lkeys, mode=0, n=ln)
+    # revenue per custkey on the device group-by
+    lprice = li_cols["l_extendedprice"].gather(lidx, ln, st)
+    ldisc = li_cols["l_discount"].gather(lidx, ln, st)
+    mprice = lprice.gather(pidx, nm, st)
+    mdisc = ldisc.gather(pidx, nm, st)
+    rev = _mul_1md(mprice, mdisc, st)
+    mcust = ocust.gather(bidx, nm, st)
+    gb = ops.GroupByI64(expected_groups=max(1024, nm), nvals=1, stream=st)
+    gb.update(mcust, [rev], nm)
+    ck, sums = gb.extract()
+    gb.free()
+    out = {"c_custkey": ck, "revenue": sums[0]}
+    order = np.lexsort((out["c_custkey"], -out["revenue"]))
+    top = order[:limit]
+    out = {k: v[top] for k, v in out.items()}
+    row = out["c_custkey"].astype(np.int64) - 1       # dense custkey
+    # attach numeric attrs from device customer columns, strings host-side
+    rcol = DevColumn.from_numpy(row.astype(np.uint32))
+    ab = cust_cols["c_acctbal"].gather(rcol, len(row), st)
+    nk = cust_cols["c_nationkey"].gather(rcol, len(row), st)
+    if st:
+        st.sync()
+    out["c_acctbal"] = ab.to_numpy(len(row))
+    out["n_name"] = np.asarray(nation_names, dtype=object)[
+        nk.to_numpy(len(row))]
+    for c in ("c_name", "c_address", "c_phone", "c_comment"):
+        if cust_host and c in cust_host:
+            out[c] = cust_host[c][row]
+    for c in (oidx, okeys, ocust, lidx, lkeys, pidx, bidx, lprice, ldisc,
+              mprice, mdisc, rev, mcust, rcol, ab, nk):
+        c.free()
+    of.free()
+    lf.free()
+    otab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

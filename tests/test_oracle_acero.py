@@ -186,3 +186,85 @@ def test_q5_oracle_equals_acero(data):
     # no non-ASIA nation appears
     asia_keys = {i for i in range(25) if nat["n_regionkey"][i] == asia}
     assert set(got) <= asia_keys
+
+
+def test_q4_oracle_equals_acero(data):
+    """tpch_ref.py:117-140 on Acero: EXISTS == left-semi join."""
+    li, orders = data["lineitem"], data["orders"]
+    l = pa.table({"l_orderkey": li["l_orderkey"],
+                  "l_commitdate": li["l_commitdate"],
+                  "l_receiptdate": li["l_receiptdate"]})
+    l = l.filter(pc.less(l["l_commitdate"], l["l_receiptdate"]))
+    o = pa.table({"o_orderkey": orders["o_orderkey"],
+                  "o_orderdate": orders["o_orderdate"],
+                  "o_orderpriority": orders["o_orderpriority"]})
+    o = o.filter(pc.and_(pc.greater_equal(o["o_orderdate"], OQ.Q4_LO),
+                         pc.less(o["o_orderdate"], OQ.Q4_HI)))
+    o = o.join(l.select(["l_orderkey"]), keys="o_orderkey",
+               right_keys="l_orderkey", join_type="left semi")
+    g = o.group_by(["o_orderpriority"]).aggregate([("o_orderkey", "count")])
+    got = {G.ORDERPRIORITY[int(k)]: int(v) for k, v in
+           zip(g.column("o_orderpriority").to_pylist(),
+               g.column("o_orderkey_count").to_pylist())}
+    want = OQ.q4(li, orders)
+    assert got == want
+
+
+def test_q18_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    cust = G.gen_customer(SF, SEED, strings=True)
+    l = pa.table({"l_orderkey": li["l_orderkey"],
+                  "l_quantity": li["l_quantity"]})
+    g = l.group_by("l_orderkey").aggregate([("l_quantity", "sum")])
+    g = g.filter(pc.greater(g["l_quantity_sum"], 300.0))
+    o = pa.table({k: orders[k] for k in ("o_orderkey", "o_custkey",
+                                         "o_orderdate", "o_totalprice")})
+    j = o.join(g, keys="o_orderkey", right_keys="l_orderkey",
+               join_type="inner")
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "c_name": pa.array(list(cust["c_name"]))})
+    j = j.join(c, keys="o_custkey", right_keys="c_custkey",
+               join_type="inner")
+    j = j.sort_by([("o_totalprice", "descending"),
+                   ("o_orderdate", "ascending"),
+                   ("o_orderkey", "ascending")]).slice(0, 100)
+    want = OQ.q18(li, orders, cust)
+    assert j.num_rows == len(want["o_orderkey"])
+    assert np.array_equal(np.asarray(j.column("o_orderkey")),
+                          want["o_orderkey"])
+    assert j.column("c_name").to_pylist() == list(want["c_name"])
+    np.testing.assert_allclose(np.asarray(j.column("l_quantity_sum")),
+                               want["sum_qty"], rtol=1e-12)
+    np.testing.assert_allclose(np.asarray(j.column("o_totalprice")),
+                               want["o_totalprice"], rtol=0)
+
+
+def test_q10_oracle_equals_acero(data):
+    li, orders, nat = data["lineitem"], data["orders"], data["nation"]
+    cust = G.gen_customer(SF, SEED, strings=True)
+    rcode = G.RETURNFLAG.index("R")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_returnflag",
+                                     "l_extendedprice", "l_discount")})
+    l = l.filter(pc.equal(l["l_returnflag"], rcode))
+    o = pa.table({k: orders[k] for k in ("o_orderkey", "o_custkey",
+                                         "o_orderdate")})
+    o = o.filter(pc.and_(pc.greater_equal(o["o_orderdate"], OQ.Q10_LO),
+                         pc.less(o["o_orderdate"], OQ.Q10_HI)))
+    j = l.join(o, keys="l_orderkey", right_keys="o_orderkey",
+               join_type="inner")
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    j = j.append_column("revenue", rev)
+    g = j.group_by("o_custkey").aggregate([("revenue", "sum")])
+    g = g.sort_by([("revenue_sum", "descending"),
+                   ("o_custkey", "ascending")]).slice(0, 20)
+    want = OQ.q10(li, orders, cust, nat)
+    assert np.array_equal(np.asarray(g.column("o_custkey")),
+                          want["c_custkey"])
+    np.testing.assert_allclose(np.asarray(g.column("revenue_sum")),
+                               want["revenue"], rtol=1e-9)
+    # attribute attachment (functionally dependent on c_custkey)
+    row = want["c_custkey"] - 1
+    assert list(want["c_name"]) == list(cust["c_name"][row])
+    assert list(want["n_name"]) == \
+        [nat["n_name"][k] for k in cust["c_nationkey"][row]]
