@@ -39,6 +39,7 @@ const char *qk_build_arch(void);         /* "gfx950"                       */
 /* ---- device memory -------------------------------------------------- */
 int qk_dmalloc(uint64_t nbytes, void **dptr);
 int qk_hmalloc_impl(uint64_t nbytes, void **hptr); /* pinned host memory */
+int qk_hfree(void *hptr);                          /* free pinned memory */
 int qk_dfree(void *dptr);
 int qk_h2d(void *dst_dev, const void *src_host, uint64_t nbytes);
 int qk_d2h(void *dst_host, const void *src_dev, uint64_t nbytes);

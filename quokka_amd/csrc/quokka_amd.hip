@@ -51,6 +51,10 @@ extern "C" int qk_device_count(int *out) {
 }
 
 // ---- memory -----------------------------------------------------------
+extern "C" int qk_hfree(void *hptr) {
+  QK_TRY("qk_hfree", hipHostFree(hptr));
+  return 0;
+}
 extern "C" int qk_hmalloc_impl(uint64_t nbytes, void **hptr) {
   QK_TRY("qk_hmalloc", hipHostMalloc(hptr, nbytes ? nbytes : 1));
   return 0;

@@ -207,11 +207,14 @@ class GPUBuildProbeJoinExecutor(Executor):
         pidx, bidx, nm = self._table.probe(kcol, mode=mode)
 
         def dev_gather(host_arr, idx_col):
-            """stage -> device gather by idx -> d2h (the payload gather is
-            the polars-join emit step, sql_executors.py:371)."""
+            """stage -> device gather by idx -> one DMA into a PINNED
+            buffer (the payload gather is the polars-join emit step,
+            sql_executors.py:371; pyarrow wraps the pinned array
+            zero-copy — quokka_amd.bridge, §8f row 3)."""
+            from . import bridge
             src = shim.DevColumn.from_numpy(host_arr)
             g = src.gather(idx_col, nm)
-            res = g.to_numpy(nm)
+            res = bridge.to_pinned_numpy(g, nm)
             src.free()
             g.free()
             return res
@@ -243,9 +246,10 @@ class GPUBuildProbeJoinExecutor(Executor):
             for c in batch.column_names:
                 out[c] = probe_col_rows(c, pidx, sel)
             bsel = bidx.to_numpy(nm)
+            from . import bridge
             for c, dev in self._build_payload_dev.items():
                 g = dev.gather(bidx, nm)
-                out[c] = g.to_numpy(nm)
+                out[c] = bridge.to_pinned_numpy(g, nm)
                 g.free()
             for c, arr in self._build_payload_host.items():
                 out[c] = arr[bsel]

@@ -37,6 +37,7 @@ _SIGS = {
     "qk_device_count": [c_vp],
     "qk_dmalloc": [c_u64, c_vp],
     "qk_hmalloc_impl": [c_u64, c_vp],
+    "qk_hfree": [c_vp],
     "qk_dfree": [c_vp],
     "qk_h2d": [c_vp, c_vp, c_u64],
     "qk_d2h": [c_vp, c_vp, c_u64],
