@@ -388,7 +388,7 @@ def main_q3(args, n, world, rank, dist, shim, DQ, standalone=True):
                 "l_orderkey": views["__key__"],
                 "l_shipdate": views["l_shipdate"],
                 "l_extendedprice": views["l_extendedprice"],
-                "l_discount": views["l_discount"]}, nt=bool(args.nt))
+                "l_discount": views["l_discount"]}, nt=(4 if args.nt == 4 else bool(args.nt)))
         rk, rp, _, _ = exchange.repartition_overlapped(
             comm, li["l_orderkey"],
             {k: v for k, v in li.items() if k != "l_orderkey"},
@@ -444,14 +444,14 @@ def main_q3(args, n, world, rank, dist, shim, DQ, standalone=True):
                     {k: v for k, v in li.items() if k != "l_orderkey"},
                     stream)
                 li_x = {"l_orderkey": rk, **rp}
-                fused.probe(li_x, nt=bool(args.nt))
+                fused.probe(li_x, nt=(4 if args.nt == 4 else bool(args.nt)))
             if timed:
                 timer.stop(stream)
         else:
             fused.rebuild()
             if timed:
                 timer.start(stream)
-            fused.probe(li_x, nt=bool(args.nt))
+            fused.probe(li_x, nt=(4 if args.nt == 4 else bool(args.nt)))
             if timed:
                 timer.stop(stream)
         n_groups, top10 = fused.extract_top10(10)
@@ -600,7 +600,7 @@ def main_q5(args, n, world, rank, dist, shim, DQ, standalone=True):
         fused.rebuild()
         if timed:
             timer.start(stream)
-        fused.probe(li, nt=bool(args.nt))
+        fused.probe(li, nt=(4 if args.nt == 4 else bool(args.nt)))
         if timed:
             timer.stop(stream)
         res = fused.result()

@@ -88,10 +88,12 @@ int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed
                     uint8_t *l_returnflag, uint8_t *l_linestatus,
                     int32_t *l_shipdate);
 
-/* Orders columns for rows [row_offset, row_offset+n): o_orderkey dense
- * row+1, o_custkey uniform 1..n_customers, o_orderdate uniform spec range,
- * o_shippriority 0. Consistent with qk_gen_lineitem's l_orderkey
- * (= 1 + row/4 % n_orders: every order has exactly 4 lines). */
+/* Orders columns for rows [row_offset, row_offset+n): o_orderkey SPARSE
+ * per TPC-H spec 4.2.3 (8 keys per 32-key bucket, = oracle/tpch_gen
+ * sparse_orderkeys), o_custkey uniform 1..n_customers with the spec's
+ * custkey%3 != 0 mortality hole, o_orderdate uniform spec range,
+ * o_shippriority 0. Consistent with qk_gen_lineitem's l_orderkey (order
+ * row = lineitem row/4: every order has exactly 4 lines). */
 int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
                   int64_t n_customers, int64_t *o_orderkey, int64_t *o_custkey,
                   int32_t *o_orderdate, int32_t *o_shippriority);
@@ -227,6 +229,16 @@ int qk_q3_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
                        uint64_t capacity, double *slot_sums,
                        uint64_t *match_count_dev, const uint32_t *bloom,
                        uint64_t bloom_mask);
+/* 4-rows-per-lane ILP variant: doubles the independent bloom/table load
+ * chains per lane (the probe is random-load latency-bound at full
+ * occupancy — r01 stall anatomy); same results, A/B in profiles/r02. */
+int qk_q3_probe_agg_nt4(void *stream, uint64_t n, const int64_t *l_orderkey,
+                        const int32_t *l_shipdate, const double *l_price,
+                        const double *l_disc, int32_t date_gt,
+                        const int64_t *slot_keys, const int32_t *slot_head,
+                        uint64_t capacity, double *slot_sums,
+                        uint64_t *match_count_dev, const uint32_t *bloom,
+                        uint64_t bloom_mask);
 /* Emit (orderkey, orders_build_row, revenue) for slots with sum != 0.
  * cursor (u64, zeroed) = group count (counted even past out_cap). */
 int qk_q3_extract(void *stream, const int64_t *slot_keys,

@@ -1169,6 +1169,90 @@ __global__ void __launch_bounds__(BLOCK) k_q3_probe_agg_nt(
   }
 }
 
+// 4 rows/thread experiment: the probe is random-load latency-bound at
+// full occupancy (r01 stall anatomy), so the remaining lever is MORE
+// independent load chains per lane — 4 bloom tests + up to 4 table
+// walks in flight instead of 2. A/B'd against the 2-row kernel on
+// MI355X; see profiles/r02 notes for the verdict.
+__global__ void __launch_bounds__(BLOCK) k_q3_probe_agg_nt4(
+    uint64_t n, const int64_t *__restrict__ l_orderkey,
+    const int32_t *__restrict__ l_shipdate,
+    const double *__restrict__ l_price, const double *__restrict__ l_disc,
+    int32_t date_gt, const int64_t *__restrict__ slot_keys,
+    const int32_t *__restrict__ slot_head, uint64_t cap,
+    double *__restrict__ slot_sums, uint64_t *__restrict__ match_count,
+    const uint32_t *__restrict__ bloom, uint64_t bloom_mask) {
+  uint32_t matches = 0;
+  uint64_t nquads = n / 4;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t q = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       q < nquads; q += stride) {
+    uint64_t i = 4 * q;
+    typedef int v4i __attribute__((ext_vector_type(4)));
+    typedef long long v4l __attribute__((ext_vector_type(4)));
+    v4i s4 = __builtin_nontemporal_load(
+        reinterpret_cast<const v4i *>(l_shipdate + i));
+    v4l k4 = __builtin_nontemporal_load(
+        reinterpret_cast<const v4l *>(l_orderkey + i));
+    bool pass[4] = {s4.x > date_gt, s4.y > date_gt, s4.z > date_gt,
+                    s4.w > date_gt};
+    int64_t keys[4] = {k4.x, k4.y, k4.z, k4.w};
+    if (bloom) {
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        if (pass[j]) pass[j] = bloom_test(bloom, bloom_mask, keys[j]);
+    }
+#pragma unroll
+    for (int j = 0; j < 4; j++)
+      q3_probe_row(keys[j], pass[j], i + j, l_price, l_disc, slot_keys,
+                   slot_head, cap, slot_sums, matches);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (uint64_t i = nquads * 4; i < n; i++)
+      if (l_shipdate[i] > date_gt) {
+        int64_t key = l_orderkey[i];
+        bool pass = !bloom || bloom_test(bloom, bloom_mask, key);
+        q3_probe_row(key, pass, i, l_price, l_disc, slot_keys, slot_head,
+                     cap, slot_sums, matches);
+      }
+  if (match_count) {
+    __shared__ uint32_t lds[BLOCK / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      matches += __shfl_down(matches, off);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = matches;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint64_t t = 0;
+      for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+      if (t) atomicAdd((unsigned long long *)match_count,
+                       (unsigned long long)t);
+    }
+  }
+}
+extern "C" int qk_q3_probe_agg_nt4(void *stream, uint64_t n,
+                                   const int64_t *l_orderkey,
+                                   const int32_t *l_shipdate,
+                                   const double *l_price,
+                                   const double *l_disc, int32_t date_gt,
+                                   const int64_t *slot_keys,
+                                   const int32_t *slot_head, uint64_t cap,
+                                   double *slot_sums, uint64_t *match_count,
+                                   const uint32_t *bloom,
+                                   uint64_t bloom_mask) {
+  if (!n) return 0;
+  if (cap & (cap - 1))
+    return qk_fail("qk_q3_probe_agg_nt4.cap_pow2", hipErrorInvalidValue);
+  uint32_t blocks =
+      (uint32_t)qk_min_u64(MAX_BLOCKS, (n / 4 + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_q3_probe_agg_nt4, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, l_orderkey, l_shipdate,
+                     l_price, l_disc, date_gt, slot_keys, slot_head, cap,
+                     slot_sums, match_count, bloom, bloom_mask);
+  QK_TRY("qk_q3_probe_agg_nt4", hipGetLastError());
+  return 0;
+}
+
 extern "C" int qk_q3_probe_agg(void *stream, uint64_t n,
                                const int64_t *l_orderkey,
                                const int32_t *l_shipdate,

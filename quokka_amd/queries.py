@@ -297,10 +297,19 @@ class Q3Fused:
 
     def probe(self, li_cols, match_count_buf=None, nt=True):
         """The fused filter+probe+group-by-aggregate pass (one kernel).
-        nt=True streams the lineitem columns with non-temporal loads."""
+        nt=True streams the lineitem columns with non-temporal loads;
+        nt=4 uses the 4-rows-per-lane ILP variant (A/B, profiles/r02)."""
         sh = self.stream.handle if self.stream else None
         n = li_cols["l_orderkey"].n
-        if nt:
+        if nt == 4:
+            call("qk_q3_probe_agg_nt4", sh, c_u64(n),
+                 li_cols["l_orderkey"].ptr, li_cols["l_shipdate"].ptr,
+                 li_cols["l_extendedprice"].ptr, li_cols["l_discount"].ptr,
+                 ctypes.c_int32(Q3_DATE), self.ord_keys.ptr,
+                 self.ord_head.ptr, c_u64(self.ord_cap), self.ord_sums.ptr,
+                 match_count_buf.ptr if match_count_buf else None,
+                 self.bloom.ptr, c_u64(self.bloom_bits - 1))
+        elif nt:
             call("qk_q3_probe_agg_nt", sh, c_u64(n),
                  li_cols["l_orderkey"].ptr, li_cols["l_shipdate"].ptr,
                  li_cols["l_extendedprice"].ptr, li_cols["l_discount"].ptr,

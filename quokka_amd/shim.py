@@ -95,6 +95,8 @@ _SIGS = {
                         c_vp, c_u64, c_vp, c_vp],
     "qk_q3_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_vp,
                            c_vp, c_u64, c_vp, c_vp, c_vp, c_u64],
+    "qk_q3_probe_agg_nt4": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32,
+                            c_vp, c_vp, c_u64, c_vp, c_vp, c_vp, c_u64],
     "qk_q5_probe_agg_nt": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                            c_u64, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
                            c_u64],
