@@ -40,7 +40,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--sf", type=float, default=100.0,
                    help="scale factor per GPU (rows = SF/100 * 600037902)")
-    p.add_argument("--query", choices=["q1", "q3", "q5", "q6"],
+    p.add_argument("--query", choices=["q1", "q3", "q4", "q5", "q6",
+                                       "q18"],
                    default="q1")
     p.add_argument("--verify", action="store_true",
                    help="full-size property cross-checks (SURVEY.md §8c): "
@@ -79,7 +80,7 @@ def gen_device_lineitem(shim, n, rank):
               cols["l_quantity"].ptr, cols["l_extendedprice"].ptr,
               cols["l_discount"].ptr, cols["l_tax"].ptr,
               cols["l_returnflag"].ptr, cols["l_linestatus"].ptr,
-              cols["l_shipdate"].ptr)
+              cols["l_shipdate"].ptr, None, None)
     return cols
 
 
@@ -126,7 +127,8 @@ def gen_device_q3_tables(shim, n, rank, world=1, n_total=None):
     shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
               c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord_total),
               li["l_orderkey"].ptr, None, None, li["l_extendedprice"].ptr,
-              li["l_discount"].ptr, None, None, None, li["l_shipdate"].ptr)
+              li["l_discount"].ptr, None, None, None, li["l_shipdate"].ptr,
+              None, None)
     n_ord_local = n_ord_total // world if world > 1 else n_ord_total
     ord_off = rank * n_ord_local if world > 1 else 0
     if world > 1 and rank == world - 1:      # last rank takes the remainder
@@ -137,7 +139,7 @@ def gen_device_q3_tables(shim, n, rank, world=1, n_total=None):
     shim.call("qk_gen_orders", None, c_u64(n_ord_local), c_u64(ord_off),
               c_u64(42), c_i64(n_cust), od["o_orderkey"].ptr,
               od["o_custkey"].ptr, od["o_orderdate"].ptr,
-              od["o_shippriority"].ptr)
+              od["o_shippriority"].ptr, None, None, c_i64(1))
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_mktsegment": DevColumn(np.uint8, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
@@ -156,13 +158,13 @@ def gen_device_q5_tables(shim, n, rank):
               c_i64(20_000_000), c_i64(n_supp), c_i64(n_ord),
               li["l_orderkey"].ptr, li["l_suppkey"].ptr, None,
               li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
-              None, None, None)
+              None, None, None, None, None)
     od = {k: DevColumn(dt, n_ord) for k, dt in [
         ("o_orderkey", np.int64), ("o_custkey", np.int64),
         ("o_orderdate", np.int32)]}
     shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
               c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
-              od["o_orderdate"].ptr, None)
+              od["o_orderdate"].ptr, None, None, None, c_i64(1))
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_nationkey": DevColumn(np.int32, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(42),
@@ -777,6 +779,106 @@ def main_q6(args, n, world, rank, dist, shim, DQ, cols=None,
     return out
 
 
+def main_q4(args, n, world, rank, dist, shim, DQ, standalone=True):
+    """TPC-H Q4 at scale on the composed generic-operator pipeline
+    (queries.q4: JIT col-vs-col filter -> dup-key build -> semi probe ->
+    JIT grouped count). A step = the whole query; JIT programs are
+    plan-time (cached). Weak scaling like Q1."""
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n_ord = max(1, n // 4)
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_commitdate", np.int32),
+        ("l_receiptdate", np.int32)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord),
+              li["l_orderkey"].ptr, None, None, None, None, None,
+              None, None, None, li["l_commitdate"].ptr,
+              li["l_receiptdate"].ptr)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_orderdate", np.int32),
+        ("o_orderpriority", np.uint8)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
+              c_i64(max(1, n_ord // 10)), od["o_orderkey"].ptr, None,
+              od["o_orderdate"].ptr, None, od["o_orderpriority"].ptr,
+              None, c_i64(1))
+    res = DQ.q4(li, od)                 # warm (JIT compile, pool)
+    t0 = time.time()
+    for _ in range(args.steps):
+        res = DQ.q4(li, od)
+    elapsed = time.time() - t0
+    out = None
+    if rank == 0:
+        out = {
+            "metric": "rows/s", "value": n * args.steps / elapsed,
+            "unit": "rows/s", "n_gpus": world, "steps": args.steps,
+            "warmup": 1, "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "f64", "data": "synthetic",
+            "config": {"workload": "TPC-H SF%g Q4 (EXISTS semi-join + "
+                                   "grouped count) via the generic "
+                                   "operator pipeline, %d lineitem "
+                                   "rows/GPU" % (args.sf, n),
+                       "sf_per_gpu": args.sf, "rows_per_gpu": n,
+                       "query": "Q4", "result": res},
+        }
+        if standalone:
+            print(json.dumps(out))
+    for c in list(li.values()) + list(od.values()):
+        c.free()
+    return out
+
+
+def main_q18(args, n, world, rank, dist, shim, DQ, standalone=True):
+    """TPC-H Q18 at scale: 600M-row group-by into ~n/4 groups on the
+    growing device table, HAVING evaluated on device
+    (qk_groupby_extract_gt), qualifying orders attached by device probe.
+    The high-cardinality group-by benchmark."""
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    n_ord = max(1, n // 4)
+    li = {k: DevColumn(dt, n) for k, dt in [
+        ("l_orderkey", np.int64), ("l_quantity", np.float64)]}
+    shim.call("qk_gen_lineitem", None, c_u64(n), c_u64(rank * n), c_u64(42),
+              c_i64(20_000_000), c_i64(1_000_000), c_i64(n_ord),
+              li["l_orderkey"].ptr, None, li["l_quantity"].ptr, None,
+              None, None, None, None, None, None, None)
+    od = {k: DevColumn(dt, n_ord) for k, dt in [
+        ("o_orderkey", np.int64), ("o_custkey", np.int64),
+        ("o_orderdate", np.int32), ("o_totalprice", np.float64)]}
+    shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(42),
+              c_i64(max(1, n_ord // 10)), od["o_orderkey"].ptr,
+              od["o_custkey"].ptr, od["o_orderdate"].ptr, None, None,
+              od["o_totalprice"].ptr, c_i64(20_000_000))
+    # the device generator emits exactly 4 lines/order (max sum 200), so
+    # the reference's 300 threshold would qualify nothing; bench at 150
+    # (~5% of orders) to exercise the HAVING path at scale
+    res = DQ.q18(li, od, threshold=150.0)       # warm
+    t0 = time.time()
+    for _ in range(args.steps):
+        res = DQ.q18(li, od, threshold=150.0)
+    elapsed = time.time() - t0
+    out = None
+    if rank == 0:
+        out = {
+            "metric": "rows/s", "value": n * args.steps / elapsed,
+            "unit": "rows/s", "n_gpus": world, "steps": args.steps,
+            "warmup": 1, "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "f64", "data": "synthetic",
+            "config": {"workload": "TPC-H SF%g Q18 (HAVING over a ~%dM-"
+                                   "group device group-by), %d lineitem "
+                                   "rows/GPU"
+                                   % (args.sf, n_ord // 1_000_000, n),
+                       "sf_per_gpu": args.sf, "rows_per_gpu": n,
+                       "query": "Q18", "having_threshold": 150.0,
+                       "qualifying_orders": len(res["o_orderkey"])},
+        }
+        if standalone:
+            print(json.dumps(out))
+    for c in list(li.values()) + list(od.values()):
+        c.free()
+    return out
+
+
 def main():
     args = parse_args()
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -799,8 +901,9 @@ def main():
     QUERY = args.query
     n = int(round(args.sf / 100.0 * SF100_LINEITEM_ROWS))
     n &= ~3  # multiple of 4 -> vectorized Q1 path, 4 lines/order for Q3
-    if args.query in ("q3", "q5", "q6"):
-        fn = {"q3": main_q3, "q5": main_q5, "q6": main_q6}[args.query]
+    if args.query in ("q3", "q4", "q5", "q6", "q18"):
+        fn = {"q3": main_q3, "q4": main_q4, "q5": main_q5,
+              "q6": main_q6, "q18": main_q18}[args.query]
         fn(args, n, world, rank, dist, shim, DQ)
         if dist is not None:
             dist.destroy_process_group()

@@ -86,7 +86,8 @@ int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed
                     double *l_quantity, double *l_extendedprice,
                     double *l_discount, double *l_tax,
                     uint8_t *l_returnflag, uint8_t *l_linestatus,
-                    int32_t *l_shipdate);
+                    int32_t *l_shipdate, int32_t *l_commitdate,
+                    int32_t *l_receiptdate);
 
 /* Orders columns for rows [row_offset, row_offset+n): o_orderkey SPARSE
  * per TPC-H spec 4.2.3 (8 keys per 32-key bucket, = oracle/tpch_gen
@@ -94,9 +95,14 @@ int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed
  * custkey%3 != 0 mortality hole, o_orderdate uniform spec range,
  * o_shippriority 0. Consistent with qk_gen_lineitem's l_orderkey (order
  * row = lineitem row/4: every order has exactly 4 lines). */
+/* o_orderpriority: uniform u8 code 0..4; o_totalprice: derived from the
+ * order's 4 lines by re-deriving their qty/partkey/disc/tax with
+ * qk_gen_lineitem's counter-based formulas (pass the same li_n_parts). */
 int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
                   int64_t n_customers, int64_t *o_orderkey, int64_t *o_custkey,
-                  int32_t *o_orderdate, int32_t *o_shippriority);
+                  int32_t *o_orderdate, int32_t *o_shippriority,
+                  uint8_t *o_orderpriority, double *o_totalprice,
+                  int64_t li_n_parts);
 /* Customer columns: c_custkey dense row+1, c_mktsegment uniform u8 0..4,
  * c_nationkey uniform i32 0..24. Any output may be NULL. */
 int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
@@ -315,6 +321,14 @@ int qk_groupby_extract(void *stream, const int64_t *slot_keys,
                        const double *slot_sums, int nvals, uint64_t capacity,
                        int64_t *out_keys, double *out_sums,
                        uint64_t out_capacity, uint64_t *out_cursor_dev);
+/* Thresholded extract: only groups whose sums[col] > threshold are
+ * compacted (HAVING clauses, e.g. Q18's sum(l_quantity) > 300 over
+ * ~n_orders groups — d2h of only the qualifying handful). */
+int qk_groupby_extract_gt(void *stream, const int64_t *slot_keys,
+                          const double *slot_sums, int nvals,
+                          uint64_t capacity, int col, double threshold,
+                          int64_t *out_keys, double *out_sums,
+                          uint64_t out_capacity, uint64_t *out_cursor_dev);
 
 /* ---- hash partition (shuffle map side) -------------------------------- *
  * Replaces partition_key_str (quokka_runtime.py:217-231). Int key semantics

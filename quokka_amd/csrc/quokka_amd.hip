@@ -202,7 +202,9 @@ __global__ void k_gen_lineitem(uint64_t n, uint64_t row_offset, uint64_t seed,
                                int64_t *l_suppkey, double *l_quantity,
                                double *l_extendedprice, double *l_discount,
                                double *l_tax, uint8_t *l_returnflag,
-                               uint8_t *l_linestatus, int32_t *l_shipdate) {
+                               uint8_t *l_linestatus, int32_t *l_shipdate,
+                               int32_t *l_commitdate,
+                               int32_t *l_receiptdate) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -225,6 +227,10 @@ __global__ void k_gen_lineitem(uint64_t n, uint64_t row_offset, uint64_t seed,
                   (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
     int32_t ship = odate + 1 + (int32_t)(h1 % 121);
     int32_t receipt = ship + 1 + (int32_t)(h2 % 30);
+    // spec: L_COMMITDATE = O_ORDERDATE + U[30,90] (independent stream)
+    if (l_commitdate)
+      l_commitdate[i] = odate + 30 + (int32_t)(splitmix64(base + 9) % 61);
+    if (l_receiptdate) l_receiptdate[i] = receipt;
     if (l_shipdate) l_shipdate[i] = ship;
     if (l_returnflag)
       l_returnflag[i] =
@@ -255,14 +261,15 @@ extern "C" int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset,
                                double *l_quantity, double *l_extendedprice,
                                double *l_discount, double *l_tax,
                                uint8_t *l_returnflag, uint8_t *l_linestatus,
-                               int32_t *l_shipdate) {
+                               int32_t *l_shipdate, int32_t *l_commitdate,
+                               int32_t *l_receiptdate) {
   if (!n) return 0;
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_gen_lineitem, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, row_offset, seed, n_parts,
                      n_suppliers, n_orders, l_orderkey, l_suppkey, l_quantity,
                      l_extendedprice, l_discount, l_tax, l_returnflag,
-                     l_linestatus, l_shipdate);
+                     l_linestatus, l_shipdate, l_commitdate, l_receiptdate);
   QK_TRY("qk_gen_lineitem", hipGetLastError());
   return 0;
 }
@@ -270,7 +277,9 @@ extern "C" int qk_gen_lineitem(void *stream, uint64_t n, uint64_t row_offset,
 __global__ void k_gen_orders(uint64_t n, uint64_t row_offset, uint64_t seed,
                              int64_t n_customers, int64_t *o_orderkey,
                              int64_t *o_custkey, int32_t *o_orderdate,
-                             int32_t *o_shippriority) {
+                             int32_t *o_shippriority,
+                             uint8_t *o_orderpriority, double *o_totalprice,
+                             int64_t li_n_parts) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -291,17 +300,43 @@ __global__ void k_gen_orders(uint64_t n, uint64_t row_offset, uint64_t seed,
       o_orderdate[i] = QK_ORDERDATE_LO +
           (int32_t)(h1 % (QK_ORDERDATE_HI - QK_ORDERDATE_LO + 1));
     if (o_shippriority) o_shippriority[i] = 0;
+    if (o_orderpriority)
+      o_orderpriority[i] = (uint8_t)(splitmix64(base + 2) % 5);
+    if (o_totalprice) {
+      // spec 4.2.3: derived from the order's lines — re-derive the 4
+      // lines' qty/partkey/disc/tax with k_gen_lineitem's exact
+      // counter-based formulas (line rows 4*row .. 4*row+3)
+      double tp = 0.0;
+      for (int j = 0; j < 4; j++) {
+        uint64_t lrow = 4 * row + (uint64_t)j;
+        uint64_t lbase = splitmix64(seed ^ 0x51A2B3C4D5E6F708ULL) ^
+                         (lrow * 0x9E3779B97F4A7C15ULL);
+        uint64_t h4 = splitmix64(lbase + 4), h5 = splitmix64(lbase + 5),
+                 h6 = splitmix64(lbase + 6), h7 = splitmix64(lbase + 7);
+        double qty = (double)(1 + (int32_t)(h4 % 50));
+        int64_t pk = 1 + (int64_t)(h5 % (uint64_t)li_n_parts);
+        int64_t cents = 90000 + (pk / 10) % 20001 + 100 * (pk % 1000);
+        double price = qty * ((double)cents / 100.0);
+        double disc = (double)(h6 % 11) / 100.0;
+        double tax = (double)(h7 % 9) / 100.0;
+        tp += price * (1.0 + tax) * (1.0 - disc);
+      }
+      o_totalprice[i] = tp;
+    }
   }
 }
 extern "C" int qk_gen_orders(void *stream, uint64_t n, uint64_t row_offset,
                              uint64_t seed, int64_t n_customers,
                              int64_t *o_orderkey, int64_t *o_custkey,
-                             int32_t *o_orderdate, int32_t *o_shippriority) {
+                             int32_t *o_orderdate, int32_t *o_shippriority,
+                             uint8_t *o_orderpriority, double *o_totalprice,
+                             int64_t li_n_parts) {
   if (!n) return 0;
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_gen_orders, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, n, row_offset, seed, n_customers,
-                     o_orderkey, o_custkey, o_orderdate, o_shippriority);
+                     o_orderkey, o_custkey, o_orderdate, o_shippriority,
+                     o_orderpriority, o_totalprice, li_n_parts);
   QK_TRY("qk_gen_orders", hipGetLastError());
   return 0;
 }
@@ -1843,6 +1878,42 @@ extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
                      (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
                      out_keys, out_sums, out_cap, cursor);
   QK_TRY("qk_groupby_extract", hipGetLastError());
+  return 0;
+}
+
+// Thresholded extract (HAVING clauses): only groups with
+// sums[col] > threshold are compacted — Q18 qualifies a handful of its
+// ~n_orders groups, so the d2h stays tiny instead of GBs.
+__global__ void __launch_bounds__(BLOCK) k_groupby_extract_gt(
+    const int64_t *__restrict__ slot_keys, const double *__restrict__ slot_sums,
+    int nvals, uint64_t cap, int col, double threshold,
+    int64_t *__restrict__ out_keys, double *__restrict__ out_sums,
+    uint64_t out_cap, uint64_t *__restrict__ cursor) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += stride) {
+    int64_t key = slot_keys[s];
+    if (key == QK_JOIN_EMPTY) continue;
+    if (!(slot_sums[(uint64_t)col * cap + s] > threshold)) continue;
+    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
+    if (pos < out_cap) {
+      out_keys[pos] = key;
+      for (int c = 0; c < nvals; c++)
+        out_sums[(uint64_t)c * out_cap + pos] =
+            slot_sums[(uint64_t)c * cap + s];
+    }
+  }
+}
+extern "C" int qk_groupby_extract_gt(void *stream, const int64_t *slot_keys,
+                                     const double *slot_sums, int nvals,
+                                     uint64_t cap, int col, double threshold,
+                                     int64_t *out_keys, double *out_sums,
+                                     uint64_t out_cap, uint64_t *cursor) {
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_groupby_extract_gt, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
+                     col, threshold, out_keys, out_sums, out_cap, cursor);
+  QK_TRY("qk_groupby_extract_gt", hipGetLastError());
   return 0;
 }
 

@@ -264,6 +264,38 @@ class GroupByI64:
         out_sums.free()
         return keys, sums
 
+    def extract_where_gt(self, col, threshold, out_guess=1 << 20):
+        """Extract only groups whose sums[col] > threshold (HAVING — e.g.
+        Q18's sum(l_quantity) > 300): the d2h stays proportional to the
+        QUALIFYING groups, not the table. Returns (keys, sums) like
+        extract(); reruns with a larger buffer if the guess undershoots
+        (counted, never truncated)."""
+        import ctypes as _ct
+        sh = self.stream.handle if self.stream else None
+        out_cap = min(self.cap, max(16, int(out_guess)))
+        while True:
+            out_keys = DevColumn(np.int64, out_cap)
+            out_sums = DevColumn(np.float64, out_cap * self.nvals)
+            cur = _count_buf()
+            shim.call("qk_groupby_extract_gt", sh, self.slot_keys.ptr,
+                      self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+                      int(col), _ct.c_double(float(threshold)),
+                      out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
+            if self.stream:
+                self.stream.sync()
+            k = _read_u64(cur)
+            cur.free()
+            if k <= out_cap:
+                keys = out_keys.to_numpy(k)
+                sums = out_sums.to_numpy(out_cap * self.nvals).reshape(
+                    self.nvals, out_cap)[:, :k].copy()
+                out_keys.free()
+                out_sums.free()
+                return keys, sums
+            out_keys.free()
+            out_sums.free()
+            out_cap = int(k)
+
     def free(self):
         self.slot_keys.free()
         self.slot_sums.free()

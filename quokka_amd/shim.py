@@ -58,7 +58,8 @@ _SIGS = {
     "qk_timer_stop": [c_vp, c_vp],
     "qk_timer_elapsed_ms": [c_vp, c_vp],
     "qk_gen_lineitem": [c_vp, c_u64, c_u64, c_u64, c_i64, c_i64, c_i64,
-                        c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
+                        c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
+                        c_vp, c_vp],
     "qk_q1_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp,
                   c_i32, c_vp],
     "qk_q6_agg": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_i32, c_i32,
@@ -74,7 +75,7 @@ _SIGS = {
     "qk_join_probe": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_u64,
                       ctypes.c_int, c_vp, c_vp, c_u64, c_vp],
     "qk_gen_orders": [c_vp, c_u64, c_u64, c_u64, c_i64, c_vp, c_vp, c_vp,
-                      c_vp],
+                      c_vp, c_vp, c_vp, c_i64],
     "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp],
     "qk_gen_supplier": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
     "qk_build_keyval_i32": [c_vp, c_u64, c_vp, c_vp, c_u32, c_vp, c_vp,
@@ -107,6 +108,8 @@ _SIGS = {
     "qk_fill_f64": [c_vp, c_vp, c_f64, c_u64],
     "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,
                            c_u64, c_vp],
+    "qk_groupby_extract_gt": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64,
+                              ctypes.c_int, c_f64, c_vp, c_vp, c_u64, c_vp],
     "qk_bloom_count": [c_vp, c_u64, c_vp, c_vp, c_u64, c_vp],
     "qk_sort_pairs_u64": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp,
                           ctypes.c_int],

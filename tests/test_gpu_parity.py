@@ -365,7 +365,7 @@ def test_gen_lineitem_device_vs_oracle_shape(gpu):
               cols["l_quantity"].ptr, cols["l_extendedprice"].ptr,
               cols["l_discount"].ptr, cols["l_tax"].ptr,
               cols["l_returnflag"].ptr, cols["l_linestatus"].ptr,
-              cols["l_shipdate"].ptr)
+              cols["l_shipdate"].ptr, None, None)
     host = {k: v.to_numpy() for k, v in cols.items()}
     got = DQ.q1(cols)
     want = OQ.q1(host)
@@ -445,13 +445,14 @@ def test_q3_fused_on_device_generated(gpu):
               c_i64(20000), c_i64(1000), c_i64(n_ord),
               li["l_orderkey"].ptr, None, None,
               li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
-              None, None, li["l_shipdate"].ptr)
+              None, None, li["l_shipdate"].ptr, None, None)
     od = {k: DevColumn(dt, n_ord) for k, dt in [
         ("o_orderkey", np.int64), ("o_custkey", np.int64),
         ("o_orderdate", np.int32), ("o_shippriority", np.int32)]}
     shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(5),
               c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
-              od["o_orderdate"].ptr, od["o_shippriority"].ptr)
+              od["o_orderdate"].ptr, od["o_shippriority"].ptr,
+              None, None, c_i64(1))
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_mktsegment": DevColumn(np.uint8, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(5),
@@ -540,13 +541,13 @@ def test_q5_fused_on_device_generated(gpu):
               c_i64(20000), c_i64(n_supp), c_i64(n_ord),
               li["l_orderkey"].ptr, li["l_suppkey"].ptr, None,
               li["l_extendedprice"].ptr, li["l_discount"].ptr, None,
-              None, None, None)
+              None, None, None, None, None)
     od = {k: DevColumn(dt, n_ord) for k, dt in [
         ("o_orderkey", np.int64), ("o_custkey", np.int64),
         ("o_orderdate", np.int32)]}
     shim.call("qk_gen_orders", None, c_u64(n_ord), c_u64(0), c_u64(9),
               c_i64(n_cust), od["o_orderkey"].ptr, od["o_custkey"].ptr,
-              od["o_orderdate"].ptr, None)
+              od["o_orderdate"].ptr, None, None, None, c_i64(1))
     cu = {"c_custkey": DevColumn(np.int64, n_cust),
           "c_nationkey": DevColumn(np.int32, n_cust)}
     shim.call("qk_gen_customer", None, c_u64(n_cust), c_u64(0), c_u64(9),
@@ -625,7 +626,7 @@ def test_exchange_world1_over_1gib(gpu):
     shim.call("qk_gen_lineitem", None, c_u64(nk), c_u64(0), c_u64(42),
               c_i64(20_000_000), c_i64(1_000_000), c_i64(nk // 4),
               kcol.ptr, None, None, pr.ptr, di.ptr, None, None, None,
-              dat.ptr)
+              dat.ptr, None, None)
     pr.free()
     di.free()
     want_k = kcol.to_numpy(nk)
@@ -1501,3 +1502,27 @@ def test_q3_overlapped_world1_parity(gpu, data):
         c.free()
     comp.destroy(); cstr.destroy()
     comm.destroy()
+
+
+def test_groupby_extract_where_gt(gpu):
+    """Thresholded (HAVING) extract: only groups with sums[col] > t come
+    back, values exact, rerun-on-undershoot counted not truncated."""
+    from quokka_amd import ops, shim
+    rng = np.random.default_rng(61)
+    keys = rng.integers(0, 5000, 100_000).astype(np.int64)
+    vals = rng.random(100_000)
+    gb = ops.GroupByI64(expected_groups=5000, nvals=2)
+    kc = shim.DevColumn.from_numpy(keys)
+    vc = shim.DevColumn.from_numpy(vals)
+    gb.update(kc, [vc, vc])
+    want_k, want_s = gb.extract()
+    t = float(np.median(want_s[0]))
+    # tiny out_guess forces the rerun path
+    got_k, got_s = gb.extract_where_gt(0, t, out_guess=16)
+    m = want_s[0] > t
+    ow, og = np.argsort(want_k[m]), np.argsort(got_k)
+    assert np.array_equal(got_k[og], want_k[m][ow])
+    np.testing.assert_allclose(got_s[0][og], want_s[0][m][ow], rtol=0)
+    np.testing.assert_allclose(got_s[1][og], want_s[1][m][ow], rtol=0)
+    kc.free(); vc.free()
+    gb.free()
