@@ -851,10 +851,10 @@ def main_q18(args, n, world, rank, dist, shim, DQ, standalone=True):
     # the device generator emits exactly 4 lines/order (max sum 200), so
     # the reference's 300 threshold would qualify nothing; bench at 150
     # (~5% of orders) to exercise the HAVING path at scale
-    res = DQ.q18(li, od, threshold=150.0)       # warm
+    res = DQ.q18(li, od, threshold=150.0, n_groups_hint=n_ord)   # warm
     t0 = time.time()
     for _ in range(args.steps):
-        res = DQ.q18(li, od, threshold=150.0)
+        res = DQ.q18(li, od, threshold=150.0, n_groups_hint=n_ord)
     elapsed = time.time() - t0
     out = None
     if rank == 0:

@@ -1784,6 +1784,7 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
     int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
     uint64_t cap, uint64_t *__restrict__ n_inserted) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  uint32_t my_inserts = 0;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
     int64_t key = keys[i];
@@ -1798,8 +1799,12 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
         if (prev == QK_JOIN_EMPTY) {
           // this lane claimed a fresh slot: count the new group so the
           // host can grow the table before cumulative distinct keys
-          // approach capacity (a full table would spin this loop forever)
-          if (n_inserted) atomicAdd((unsigned long long *)n_inserted, 1ULL);
+          // approach capacity (a full table would spin this loop
+          // forever). Counted per-thread and block-reduced at the end —
+          // a single cursor word takes ~88 atomics/us, and a
+          // high-cardinality batch (Q18: 150M groups) would serialize
+          // ~1.7 s on per-insert atomics.
+          my_inserts++;
           break;
         }
         if (prev == key) break;
@@ -1816,6 +1821,20 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
         atomic_max_f64(dst, v);
       else
         atomicAdd(dst, v);
+    }
+  }
+  if (n_inserted) {
+    __shared__ uint32_t lds[BLOCK / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      my_inserts += __shfl_down(my_inserts, off);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = my_inserts;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint64_t t = 0;
+      for (int w = 0; w < BLOCK / WAVE; w++) t += lds[w];
+      if (t) atomicAdd((unsigned long long *)n_inserted,
+                       (unsigned long long)t);
     }
   }
 }

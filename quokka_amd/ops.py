@@ -186,7 +186,10 @@ class GroupByI64:
         self.n_groups = 0           # exact, from the device insert counter
         self._nins = _count_buf()
 
-    def update(self, keys_col, val_cols, n=None):
+    def update(self, keys_col, val_cols, n=None, max_new_groups=None):
+        """max_new_groups: optional caller-known bound on DISTINCT keys
+        this batch can introduce (e.g. the key domain size) — without it
+        the growth check conservatively assumes every row is new."""
         n = keys_col.n if n is None else n
         if not n:
             return
@@ -195,8 +198,9 @@ class GroupByI64:
         # full table makes the kernel's find-or-insert loop spin forever,
         # so grow FIRST whenever this batch could push the exact group
         # count (device insert counter) past half capacity
-        if 2 * (self.n_groups + n) > self.cap:
-            self._grow(self.n_groups + n)
+        bound = n if max_new_groups is None else min(n, int(max_new_groups))
+        if 2 * (self.n_groups + bound) > self.cap:
+            self._grow(self.n_groups + bound)
         sh = self.stream.handle if self.stream else None
         ptrs = np.array([c.ptr.value if hasattr(c.ptr, "value") else c.ptr
                          for c in val_cols], dtype=np.uint64)

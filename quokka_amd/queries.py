@@ -694,7 +694,7 @@ def q4(li_cols, ord_cols, stream=None):
 
 
 def q18(li_cols, ord_cols, cust_names=None, stream=None, limit=100,
-        threshold=300.0):
+        threshold=300.0, n_groups_hint=None):
     """Device Q18 (tpch_ref.py:544-580): group lineitem by l_orderkey
     sum(l_quantity) on the device group-by table (unbounded cardinality,
     grows), qualify sum > 300, then attach the order row (the group key
@@ -705,9 +705,10 @@ def q18(li_cols, ord_cols, cust_names=None, stream=None, limit=100,
     from . import ops
     st = stream
     n = li_cols["l_orderkey"].n
-    gb = ops.GroupByI64(expected_groups=max(1024, n // 4), nvals=1,
-                        stream=st)
-    gb.update(li_cols["l_orderkey"], [li_cols["l_quantity"]], n)
+    hint = n_groups_hint or max(1024, n // 4)
+    gb = ops.GroupByI64(expected_groups=hint, nvals=1, stream=st)
+    gb.update(li_cols["l_orderkey"], [li_cols["l_quantity"]], n,
+              max_new_groups=(n_groups_hint if n_groups_hint else None))
     # HAVING sum > 300 evaluated ON DEVICE: d2h only the qualifying
     # groups (a handful of ~n_orders), not the whole table
     qkeys, qsums = gb.extract_where_gt(0, float(threshold))
@@ -729,25 +730,38 @@ def q18(li_cols, ord_cols, cust_names=None, stream=None, limit=100,
     if st:
         st.sync()
     okeys_h = ok.to_numpy(nm)
-    qmap = {int(k): s for k, s in zip(qkeys, qsums)}
+    od_h = od.to_numpy(nm)
+    tp_h = tp.to_numpy(nm)
     cust_h = ck.to_numpy(nm)
+    # top-limit FIRST (o_totalprice desc, o_orderdate asc, o_orderkey asc)
+    # — name/qty attachment is then O(limit), not O(qualifying orders)
+    if nm > 4 * limit + 64:
+        cand = np.argpartition(-tp_h, 2 * limit)[: 2 * limit]
+        cutoff = tp_h[cand].min()
+        cand = np.nonzero(tp_h >= cutoff)[0]
+    else:
+        cand = np.arange(nm)
+    order = np.lexsort((okeys_h[cand], od_h[cand], -tp_h[cand]))
+    top = cand[order[:limit]]
+    sel_keys = okeys_h[top]
+    qorder = np.argsort(qkeys)
+    qpos = np.searchsorted(qkeys, sel_keys, sorter=qorder)
+    sum_sel = qsums[qorder[qpos]]
+    cust_sel = cust_h[top]
     out = {
-        "c_name": (cust_names[cust_h - 1] if cust_names is not None else
-                   np.array(["Customer#%09d" % c for c in cust_h],
+        "c_name": (cust_names[cust_sel - 1] if cust_names is not None else
+                   np.array(["Customer#%09d" % c for c in cust_sel],
                             dtype=object)),
-        "c_custkey": cust_h,
-        "o_orderkey": okeys_h,
-        "o_orderdate": od.to_numpy(nm),
-        "o_totalprice": tp.to_numpy(nm),
-        "sum_qty": np.array([qmap[int(k)] for k in okeys_h]),
+        "c_custkey": cust_sel,
+        "o_orderkey": sel_keys,
+        "o_orderdate": od_h[top],
+        "o_totalprice": tp_h[top],
+        "sum_qty": sum_sel,
     }
-    order = np.lexsort((out["o_orderkey"], out["o_orderdate"],
-                        -out["o_totalprice"]))
-    top = order[:limit]
     for c in (pidx, rcol, ok, od, tp, ck):
         c.free()
     otab.free()
-    return {k: v[top] for k, v in out.items()}
+    return out
 
 
 def q10(li_cols, ord_cols, cust_cols, cust_host, nation_names,
