@@ -72,14 +72,20 @@ class GPUBuildProbeJoinExecutor(Executor):
     def __init__(self, on=None, left_on=None, right_on=None, how="inner",
                  key_to_keep="left"):
         if on is not None:
-            assert left_on is None and right_on is None
+            if left_on is not None or right_on is not None:
+                raise ValueError("pass either on= or left_on=/right_on=, "
+                                 "not both (sql_executors.py:327-343)")
             self.left_on = on
             self.right_on = on
         else:
-            assert left_on is not None and right_on is not None
+            if left_on is None or right_on is None:
+                raise ValueError("join needs on= or both left_on= and "
+                                 "right_on=")
             self.left_on = left_on
             self.right_on = right_on
-        assert how in {"inner", "left", "semi", "anti"}
+        if how not in {"inner", "left", "semi", "anti"}:
+            raise ValueError("unsupported join how=%r (inner/left/semi/"
+                             "anti, as the reference)" % (how,))
         self.how = how
         self.key_to_keep = key_to_keep
         self.phase = "build"
@@ -318,7 +324,9 @@ class GPUAggExecutor(Executor):
     """
 
     def __init__(self, groupby_keys, orderby_keys, sql_statement):
-        assert isinstance(groupby_keys, list)
+        if not isinstance(groupby_keys, list):
+            raise TypeError("groupby_keys must be a list (may be empty "
+                            "for a grand aggregate)")
         self.groupby_keys = groupby_keys
         self.orderby_keys = orderby_keys or []
         self.sql_statement = sql_statement
@@ -332,11 +340,25 @@ class GPUAggExecutor(Executor):
             if m:
                 alias = m.group(1)
                 part = part[: m.start()]
+            low = part.lower()
+            if re.search(r"count\s*\(\s*distinct", low):
+                raise ValueError(
+                    "count(distinct ...) is not distributive and cannot "
+                    "appear in the FINAL aggregate this executor computes; "
+                    "the reference's two-phase rewrite (sql_utils.py:299-"
+                    "413) does not emit it either — pre-aggregate "
+                    "distinct keys upstream (e.g. GPUDistinctExecutor)")
+            if re.search(r"\bavg\s*\(", low):
+                raise ValueError(
+                    "avg(...) must arrive REWRITTEN as sum(partial_sum)/"
+                    "sum(partial_count) (the two-phase rewrite, "
+                    "sql_utils.py:379-413); avg over partials would be "
+                    "wrong: %r" % part)
             hits = _AGG_RE.findall(part)
             if not hits:
                 raise ValueError("unsupported aggregate (supported: "
-                                 "expressions over SUM/MIN/MAX(col)): %r"
-                                 % part)
+                                 "expressions over SUM/MIN/MAX(col) of "
+                                 "partial columns): %r" % part)
             for fn, c in hits:
                 if c not in self.sum_cols:
                     self.sum_cols.append(c)

@@ -209,3 +209,23 @@ def test_agg_orderby_desc_string_key(gpu):
                                       out.column("s").to_pylist())}
     assert d[("a", "x")] == 2.0 and d[("a", "y")] == 4.0
     assert d[("b", "x")] == 1.0 and d[("c", "y")] == 3.0
+
+
+def test_agg_executor_typed_errors():
+    """Narrow-surface failure modes raise typed, actionable errors
+    instead of asserts (advisor weak item): count(distinct) and
+    un-rewritten avg() name the two-phase rewrite; bad join args raise
+    ValueError."""
+    with pytest.raises(ValueError, match="two-phase"):
+        GPUAggExecutor(["k"], [], "count(distinct x) as n")
+    with pytest.raises(ValueError, match="two-phase|REWRITTEN"):
+        GPUAggExecutor(["k"], [], "avg(x) as a")
+    with pytest.raises(TypeError, match="list"):
+        GPUAggExecutor("k", [], "sum(x) as s")
+    with pytest.raises(ValueError, match="left_on"):
+        GPUBuildProbeJoinExecutor(on="k", left_on="a", right_on="k")
+    with pytest.raises(ValueError, match="how"):
+        GPUBuildProbeJoinExecutor(on="k", how="outer")
+    # the supported form still parses
+    ex = GPUAggExecutor(["k"], [], "sum(s0) / sum(c0) as avg_x")
+    assert ex.sum_cols == ["s0", "c0"]
