@@ -17,6 +17,7 @@ __version__ = "0.1.0"
 from .executors import (Executor, GPUAggExecutor,  # noqa: F401
                         GPUBroadcastJoinExecutor,
                         GPUBuildProbeJoinExecutor, GPUCountExecutor,
+                        GPUDiskBuildProbeJoinExecutor,
                         GPUDistinctExecutor, GPUSortExecutor,
                         GPUTopKExecutor,
                         gpu_partition_fn)
