@@ -233,6 +233,51 @@ def q10(li, orders, customer, nation, limit=20):
     return {k: v[top] for k, v in out.items()}
 
 
+Q12_LO = G.days(1994, 1, 1)
+Q12_HI = G.days(1995, 1, 1)     # + interval '1' year
+Q14_LO = G.days(1995, 9, 1)
+Q14_HI = G.days(1995, 10, 1)    # + interval '1' month
+
+
+def q12(li, orders):
+    """tpch_ref.py:376-407: lines with shipmode in (MAIL, SHIP),
+    ship < commit < receipt, receipt in [1994-01-01, +1y), joined to
+    orders; per shipmode count lines with HIGH (1-URGENT/2-HIGH) vs low
+    priority. Returns dict shipmode(str) -> (high_count, low_count),
+    shipmode ascending."""
+    mail, ship_m = G.SHIPMODE.index("MAIL"), G.SHIPMODE.index("SHIP")
+    m = (np.isin(li["l_shipmode"], [mail, ship_m])
+         & (li["l_commitdate"] < li["l_receiptdate"])
+         & (li["l_shipdate"] < li["l_commitdate"])
+         & (li["l_receiptdate"] >= Q12_LO)
+         & (li["l_receiptdate"] < Q12_HI))
+    nkey = int(orders["o_orderkey"].max()) + 2
+    prio_by_key = np.full(nkey, -1, dtype=np.int8)
+    prio_by_key[orders["o_orderkey"]] = orders["o_orderpriority"]
+    pr = prio_by_key[li["l_orderkey"][m]]
+    sm = li["l_shipmode"][m]
+    assert (pr >= 0).all(), "lineitem orderkey missing from orders"
+    high = pr <= 1                  # codes 0/1 = 1-URGENT / 2-HIGH
+    out = {}
+    for code in sorted({mail, ship_m}):
+        sel = sm == code
+        out[G.SHIPMODE[code]] = (int((sel & high).sum()),
+                                 int((sel & ~high).sum()))
+    return out
+
+
+def q14(li, part):
+    """tpch_ref.py:434-450: promo revenue ratio over the shipdate month
+    window, lineitem x part on partkey. Returns the percentage."""
+    m = (li["l_shipdate"] >= Q14_LO) & (li["l_shipdate"] < Q14_HI)
+    pk = li["l_partkey"][m]
+    promo_flag = (part["p_type"] // 25) == G.PTYPE_PROMO_SYL1
+    is_promo = promo_flag[pk - 1]        # partkey dense 1..N
+    rev = li["l_extendedprice"][m] * (1.0 - li["l_discount"][m])
+    total = rev.sum()
+    return 100.0 * rev[is_promo].sum() / total if total else 0.0
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

@@ -58,6 +58,13 @@ LINESTATUS = ["F", "O"]               # l_linestatus codes 0,1
 MKTSEGMENT = ["AUTOMOBILE", "BUILDING", "FURNITURE", "HOUSEHOLD", "MACHINERY"]
 ORDERPRIORITY = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED",
                  "5-LOW"]             # spec 4.2.3 uniform, sorted order
+SHIPMODE = ["AIR", "FOB", "MAIL", "RAIL", "REG AIR", "SHIP", "TRUCK"]
+# p_type = syllable1 x syllable2 x syllable3 (spec 4.2.2.13); code =
+# ((s1 * 5) + s2) * 5 + s3; LIKE 'PROMO%' == s1 == index of "PROMO"
+PTYPE_SYL1 = ["ECONOMY", "LARGE", "MEDIUM", "PROMO", "SMALL", "STANDARD"]
+PTYPE_SYL2 = ["ANODIZED", "BRUSHED", "BURNISHED", "PLATED", "POLISHED"]
+PTYPE_SYL3 = ["BRASS", "COPPER", "NICKEL", "STEEL", "TIN"]
+PTYPE_PROMO_SYL1 = PTYPE_SYL1.index("PROMO")
 
 # TPC-H spec nation table: (name, regionkey)
 NATIONS = [
@@ -168,7 +175,7 @@ def gen_lineitem(sf, seed=42, orders=None):
     orders["o_totalprice"] = np.add.reduceat(
         line_total, np.concatenate(([0], np.cumsum(lines_per_order)[:-1])))
 
-    return {
+    out = {
         "l_orderkey": l_orderkey,
         "l_suppkey": rng.integers(1, n_suppliers(sf) + 1, n, dtype=np.int64),
         "l_quantity": quantity,
@@ -180,7 +187,11 @@ def gen_lineitem(sf, seed=42, orders=None):
         "l_shipdate": shipdate.astype(np.int32),
         "l_commitdate": commitdate.astype(np.int32),
         "l_receiptdate": receiptdate.astype(np.int32),
+        "l_partkey": partkey,
     }
+    # drawn LAST so earlier columns' RNG streams stay fixture-stable
+    out["l_shipmode"] = rng.integers(0, len(SHIPMODE), n).astype(np.uint8)
+    return out
 
 
 def gen_customer(sf, seed=42, strings=False):
@@ -224,6 +235,19 @@ def gen_supplier(sf, seed=42):
     }
 
 
+def gen_part(sf, seed=42):
+    """part columns: p_partkey i64 (dense 1..N — the key space
+    gen_lineitem's l_partkey draws from), p_type u8 code into the 150
+    spec type combinations (PTYPE_SYL* above); LIKE 'PROMO%' == code //
+    25 == PTYPE_PROMO_SYL1."""
+    n = n_parts(sf)
+    rng = np.random.default_rng([seed, 5])
+    return {
+        "p_partkey": np.arange(1, n + 1, dtype=np.int64),
+        "p_type": rng.integers(0, 150, n).astype(np.uint8),
+    }
+
+
 def gen_nation():
     return {
         "n_nationkey": np.arange(25, dtype=np.int32),
@@ -243,6 +267,7 @@ def gen_all(sf, seed=42):
         "lineitem": gen_lineitem(sf, seed, orders),
         "customer": gen_customer(sf, seed),
         "supplier": gen_supplier(sf, seed),
+        "part": gen_part(sf, seed),
         "nation": gen_nation(),
         "region": gen_region(),
     }

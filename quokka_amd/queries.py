@@ -832,6 +832,104 @@ def q10(li_cols, ord_cols, cust_cols, cust_host, nation_names,
     return out
 
 
+SHIPMODE = ["AIR", "FOB", "MAIL", "RAIL", "REG AIR", "SHIP", "TRUCK"]
+PTYPE_PROMO_SYL1 = 3        # index of "PROMO" (oracle.tpch_gen.PTYPE_SYL1)
+
+
+def q12(li_cols, ord_cols, stream=None):
+    """Device Q12 (tpch_ref.py:376-407): one JIT pass fuses the 5-clause
+    lineitem predicate (IN-list on the shipmode dict codes + two
+    col-vs-col date comparisons + the receipt window); survivors probe
+    the orders table; priority counts per shipmode come from a JIT
+    grouped count over the (shipmode, priority) product. Returns dict
+    shipmode(str) -> (high_count, low_count)."""
+    from . import jit, ops
+    st = stream
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_shipmode in (2, 5) and l_commitdate < l_receiptdate and "
+        "l_shipdate < l_commitdate and "
+        "l_receiptdate >= date '1994-01-01' and "
+        "l_receiptdate < date '1994-01-01' + interval '1' year", lsch),
+        "q12_pred", _schema_key(lsch))
+    lidx, ln = lf.run(li_cols, st)
+    lkeys = li_cols["l_orderkey"].gather(lidx, ln, st)
+    lmode = li_cols["l_shipmode"].gather(lidx, ln, st)
+    otab = ops.JoinTable(max(16, ord_cols["o_orderkey"].n), st)
+    otab.build(ord_cols["o_orderkey"])
+    pidx, bidx, nm = otab.probe(lkeys, mode=0, n=ln)
+    m_mode = lmode.gather(pidx, nm, st)
+    m_prio = ord_cols["o_orderpriority"].gather(bidx, nm, st)
+    agg = _cached_jit("a", lambda: jit.JitAggregate(
+        {"l_shipmode": np.dtype(np.uint8),
+         "o_orderpriority": np.dtype(np.uint8)},
+        [("l_shipmode", 7), ("o_orderpriority", 5)],
+        ["COUNT(*) as n"]), "q12_counts")
+    acc = agg.make_acc()
+    if nm:
+        agg.run({"l_shipmode": m_mode, "o_orderpriority": m_prio}, acc, st)
+    if st:
+        st.sync()
+    counts = agg.read(acc)[:, 0].reshape(7, 5)
+    out = {}
+    for code in (2, 5):                 # MAIL, SHIP
+        high = int(counts[code, 0] + counts[code, 1])
+        low = int(counts[code, 2:].sum())
+        out[SHIPMODE[code]] = (high, low)
+    for c in (lidx, lkeys, lmode, pidx, bidx, m_mode, m_prio):
+        c.free()
+    acc.free()
+    otab.free()
+    return out
+
+
+def q14(li_cols, part_cols, stream=None):
+    """Device Q14 (tpch_ref.py:434-450): JIT date-window filter ->
+    probe the part table -> JIT grouped SUM of revenue by the promo
+    flag; the ratio finishes host-side over two doubles. part_cols:
+    p_partkey i64 + p_promo u8 (p_type LIKE 'PROMO%' resolved to the
+    flag at staging, the same trick the JIT uses for LIKE on dict
+    columns)."""
+    from . import jit, ops
+    st = stream
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_shipdate >= date '1995-09-01' and "
+        "l_shipdate < date '1995-09-01' + interval '1' month", lsch),
+        "q14_window", _schema_key(lsch))
+    lidx, ln = lf.run(li_cols, st)
+    lpart = li_cols["l_partkey"].gather(lidx, ln, st)
+    lprice = li_cols["l_extendedprice"].gather(lidx, ln, st)
+    ldisc = li_cols["l_discount"].gather(lidx, ln, st)
+    ptab = ops.JoinTable(max(16, part_cols["p_partkey"].n), st)
+    ptab.build(part_cols["p_partkey"])
+    pidx, bidx, nm = ptab.probe(lpart, mode=0, n=ln)
+    m_price = lprice.gather(pidx, nm, st)
+    m_disc = ldisc.gather(pidx, nm, st)
+    m_promo = part_cols["p_promo"].gather(bidx, nm, st)
+    agg = _cached_jit("a", lambda: jit.JitAggregate(
+        {"p_promo": np.dtype(np.uint8),
+         "l_extendedprice": np.dtype(np.float64),
+         "l_discount": np.dtype(np.float64)},
+        [("p_promo", 2)],
+        ["SUM(l_extendedprice * (1 - l_discount)) as rev"]), "q14_rev")
+    acc = agg.make_acc()
+    if nm:
+        agg.run({"p_promo": m_promo, "l_extendedprice": m_price,
+                 "l_discount": m_disc}, acc, st)
+    if st:
+        st.sync()
+    rev = agg.read(acc)[:, 0]
+    total = rev.sum()
+    out = 100.0 * rev[1] / total if total else 0.0
+    for c in (lidx, lpart, lprice, ldisc, pidx, bidx, m_price, m_disc,
+              m_promo):
+        c.free()
+    acc.free()
+    ptab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

@@ -91,3 +91,38 @@ def test_q10_device_vs_oracle(gpu, data):
     for cs in (lcols, ocols, ccols):
         for c in cs.values():
             c.free()
+
+
+def test_q12_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, od = data["lineitem"], data["orders"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_shipmode",
+                                             "l_shipdate", "l_commitdate",
+                                             "l_receiptdate"])
+    ocols = staging.stage_columns(od, names=["o_orderkey",
+                                             "o_orderpriority"])
+    got = DQ.q12(lcols, ocols)
+    want = OQ.q12(li, od)
+    assert got == want
+    for cs in (lcols, ocols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q14_device_vs_oracle(gpu, data):
+    import numpy as _np
+    from quokka_amd import staging, queries as DQ
+    li, part = data["lineitem"], data["part"]
+    lcols = staging.stage_columns(li, names=["l_partkey", "l_shipdate",
+                                             "l_extendedprice",
+                                             "l_discount"])
+    pcols = staging.stage_columns({
+        "p_partkey": part["p_partkey"],
+        "p_promo": ((part["p_type"] // 25) ==
+                    G.PTYPE_PROMO_SYL1).astype(_np.uint8)})
+    got = DQ.q14(lcols, pcols)
+    want = OQ.q14(li, part)
+    _np.testing.assert_allclose(got, want, rtol=1e-9)
+    for cs in (lcols, pcols):
+        for c in cs.values():
+            c.free()

@@ -268,3 +268,50 @@ def test_q10_oracle_equals_acero(data):
     assert list(want["c_name"]) == list(cust["c_name"][row])
     assert list(want["n_name"]) == \
         [nat["n_name"][k] for k in cust["c_nationkey"][row]]
+
+
+def test_q12_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    mail, shipm = G.SHIPMODE.index("MAIL"), G.SHIPMODE.index("SHIP")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_shipmode",
+                                     "l_shipdate", "l_commitdate",
+                                     "l_receiptdate")})
+    m = pc.and_(
+        pc.and_(pc.is_in(l["l_shipmode"],
+                         value_set=pa.array([mail, shipm],
+                                            type=pa.uint8())),
+                pc.less(l["l_commitdate"], l["l_receiptdate"])),
+        pc.and_(pc.less(l["l_shipdate"], l["l_commitdate"]),
+                pc.and_(pc.greater_equal(l["l_receiptdate"], OQ.Q12_LO),
+                        pc.less(l["l_receiptdate"], OQ.Q12_HI))))
+    l = l.filter(m)
+    o = pa.table({"o_orderkey": orders["o_orderkey"],
+                  "o_orderpriority": orders["o_orderpriority"]})
+    j = l.join(o, keys="l_orderkey", right_keys="o_orderkey",
+               join_type="inner")
+    high = pc.less_equal(j["o_orderpriority"], 1)
+    want = OQ.q12(li, orders)
+    for code, name in ((mail, "MAIL"), (shipm, "SHIP")):
+        sel = pc.equal(j["l_shipmode"], code)
+        h = j.filter(pc.and_(sel, high)).num_rows
+        lo = j.filter(pc.and_(sel, pc.invert(high))).num_rows
+        assert (h, lo) == want[name], name
+
+
+def test_q14_oracle_equals_acero(data):
+    li, part = data["lineitem"], data["part"]
+    l = pa.table({k: li[k] for k in ("l_partkey", "l_shipdate",
+                                     "l_extendedprice", "l_discount")})
+    l = l.filter(pc.and_(pc.greater_equal(l["l_shipdate"], OQ.Q14_LO),
+                         pc.less(l["l_shipdate"], OQ.Q14_HI)))
+    p = pa.table({"p_partkey": part["p_partkey"],
+                  "promo": (part["p_type"] // 25) == G.PTYPE_PROMO_SYL1})
+    j = l.join(p, keys="l_partkey", right_keys="p_partkey",
+               join_type="inner")
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    total = pc.sum(rev).as_py()
+    promo = pc.sum(pc.if_else(j["promo"], rev, 0.0)).as_py()
+    got = 100.0 * promo / total
+    want = OQ.q14(li, part)
+    np.testing.assert_allclose(got, want, rtol=1e-9)
