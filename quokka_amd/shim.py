@@ -181,7 +181,7 @@ class _PinnedBounce:
 
     CHUNK = 64 << 20
 
-    MEMMOVE_THREADS = 4
+    MEMMOVE_THREADS = int(os.environ.get("QK_MEMMOVE_THREADS", "4"))
 
     def __init__(self):
         self.bufs = None
