@@ -278,6 +278,86 @@ def q14(li, part):
     return 100.0 * rev[is_promo].sum() / total if total else 0.0
 
 
+Q7_LO = G.days(1995, 1, 1)
+Q7_HI = G.days(1996, 12, 31)    # BETWEEN is inclusive
+Y1996 = G.days(1996, 1, 1)
+
+
+def q7(li, orders, customer, supplier, nation):
+    """tpch_ref.py:185-227: FRANCE<->GERMANY shipping volume by
+    (supp_nation, cust_nation, year of l_shipdate), shipdate BETWEEN
+    1995-01-01 AND 1996-12-31 (inclusive). Returns dict
+    (supp_nation, cust_nation, year) -> revenue, sorted key order."""
+    names = list(nation["n_name"])
+    fr, de = names.index("FRANCE"), names.index("GERMANY")
+    cust_nat = np.full(int(customer["c_custkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    cust_nat[customer["c_custkey"]] = customer["c_nationkey"]
+    nkey = int(orders["o_orderkey"].max()) + 2
+    order_cnat = np.full(nkey, -1, dtype=np.int32)
+    order_cnat[orders["o_orderkey"]] = cust_nat[orders["o_custkey"]]
+    supp_nat = np.full(int(supplier["s_suppkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    supp_nat[supplier["s_suppkey"]] = supplier["s_nationkey"]
+
+    m = (li["l_shipdate"] >= Q7_LO) & (li["l_shipdate"] <= Q7_HI)
+    cn = order_cnat[li["l_orderkey"][m]]
+    sn = supp_nat[li["l_suppkey"][m]]
+    pair = ((sn == fr) & (cn == de)) | ((sn == de) & (cn == fr))
+    rev = (li["l_extendedprice"][m] * (1.0 - li["l_discount"][m]))[pair]
+    year = np.where(li["l_shipdate"][m][pair] >= Y1996, 1996, 1995)
+    snp, cnp = sn[pair], cn[pair]
+    out = {}
+    for s, c in ((fr, de), (de, fr)):
+        for y in (1995, 1996):
+            v = rev[(snp == s) & (cnp == c) & (year == y)].sum()
+            out[(names[s], names[c], y)] = float(v)
+    return dict(sorted(out.items()))
+
+
+def q8(li, orders, customer, supplier, part, nation, region):
+    """tpch_ref.py:229-268: BRAZIL market share among AMERICA-region
+    customers for p_type = 'ECONOMY ANODIZED STEEL', by order year.
+    Returns dict year -> mkt_share."""
+    america = G.REGIONS.index("AMERICA")
+    nat_in_america = nation["n_regionkey"] == america
+    # p_type code for 'ECONOMY ANODIZED STEEL'
+    code = ((G.PTYPE_SYL1.index("ECONOMY") * 5 +
+             G.PTYPE_SYL2.index("ANODIZED")) * 5 +
+            G.PTYPE_SYL3.index("STEEL"))
+    part_ok = np.zeros(int(part["p_partkey"].max()) + 2, dtype=bool)
+    part_ok[part["p_partkey"][part["p_type"] == code]] = True
+
+    cust_nat = np.full(int(customer["c_custkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    cust_nat[customer["c_custkey"]] = customer["c_nationkey"]
+    omask = ((orders["o_orderdate"] >= Q7_LO)
+             & (orders["o_orderdate"] <= Q7_HI))
+    ocn = cust_nat[orders["o_custkey"][omask]]
+    okeep = nat_in_america[np.clip(ocn, 0, 24)] & (ocn >= 0)
+    nkey = int(orders["o_orderkey"].max()) + 2
+    order_year = np.zeros(nkey, dtype=np.int16)
+    ok = orders["o_orderkey"][omask][okeep]
+    order_year[ok] = np.where(
+        orders["o_orderdate"][omask][okeep] >= Y1996, 1996, 1995)
+    supp_nat = np.full(int(supplier["s_suppkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    supp_nat[supplier["s_suppkey"]] = supplier["s_nationkey"]
+
+    m = part_ok[li["l_partkey"]] & (order_year[li["l_orderkey"]] > 0)
+    rev = li["l_extendedprice"][m] * (1.0 - li["l_discount"][m])
+    yr = order_year[li["l_orderkey"][m]]
+    sn = supp_nat[li["l_suppkey"][m]]
+    brazil = list(nation["n_name"]).index("BRAZIL")
+    out = {}
+    for y in (1995, 1996):
+        sel = yr == y
+        tot = rev[sel].sum()
+        br = rev[sel & (sn == brazil)].sum()
+        out[y] = float(br / tot) if tot else 0.0
+    return out
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

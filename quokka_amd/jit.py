@@ -443,8 +443,10 @@ class JitAggregate:
         gparts = []
         for name, card in self.group_keys:
             ref, dt = tr.col_ref(name)
-            if dt != np.dtype(np.uint8):
-                raise TypeError("group keys must be u8 code columns")
+            if dt not in (np.dtype(np.uint8), np.dtype(np.int32)):
+                raise TypeError("group keys must be u8 code or small-"
+                                "cardinality i32 columns (values in "
+                                "[0, cardinality), e.g. nation keys)")
             gparts.append((ref, card))
             self.ngroups *= int(card)
         gexpr = "0"   # [] group keys -> one grand-aggregate group

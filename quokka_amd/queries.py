@@ -930,6 +930,182 @@ def q14(li_cols, part_cols, stream=None):
     return out
 
 
+def q7(li_cols, ord_cols, cust_cols, supp_cols, nation_names,
+       nation1="FRANCE", nation2="GERMANY", stream=None):
+    """Device Q7 (tpch_ref.py:185-227): 6-table chain composed from
+    generic join probes — customer->orders attaches cust_nation,
+    orders->lineitem, supplier->lineitem attaches supp_nation; the
+    nation-pair condition is a JIT filter over the two i32 nation
+    columns and the per-(supp_nation, year) revenue comes from two JIT
+    grouped sums (one per shipdate year; BETWEEN bounds inclusive).
+    Returns dict (supp_nation, cust_nation, year) -> revenue."""
+    from . import jit, ops
+    st = stream
+    names = list(nation_names)
+    fr, de = names.index(nation1), names.index(nation2)
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_shipdate >= date '1995-01-01' and "
+        "l_shipdate <= date '1996-12-31'", lsch),
+        "q7_window", _schema_key(lsch))
+    lidx, ln = lf.run(li_cols, st)
+    g = {c: li_cols[c].gather(lidx, ln, st)
+         for c in ("l_orderkey", "l_suppkey", "l_extendedprice",
+                   "l_discount", "l_shipdate")}
+    # customer -> orders: pairs give (orders row, customer row)
+    ctab = ops.JoinTable(max(16, cust_cols["c_custkey"].n), st)
+    ctab.build(cust_cols["c_custkey"])
+    opidx, cbidx, ncm = ctab.probe(ord_cols["o_custkey"], mode=0)
+    ord_keys_j = ord_cols["o_orderkey"].gather(opidx, ncm, st)
+    ord_cnat_j = cust_cols["c_nationkey"].gather(cbidx, ncm, st)
+    # orders -> lineitem
+    otab = ops.JoinTable(max(16, ncm), st)
+    if ncm:
+        otab.build(ord_keys_j)
+    lpidx, obidx, nlm = otab.probe(g["l_orderkey"], mode=0, n=ln)
+    cn = ord_cnat_j.gather(obidx, nlm, st)
+    m = {c: g[c].gather(lpidx, nlm, st)
+         for c in ("l_suppkey", "l_extendedprice", "l_discount",
+                   "l_shipdate")}
+    # supplier -> matched lines (re-align everything by the pair order)
+    stab = ops.JoinTable(max(16, supp_cols["s_suppkey"].n), st)
+    stab.build(supp_cols["s_suppkey"])
+    spidx, sbidx, nsm = stab.probe(m["l_suppkey"], mode=0, n=nlm)
+    sn = supp_cols["s_nationkey"].gather(sbidx, nsm, st)
+    al = {c: m[c].gather(spidx, nsm, st)
+          for c in ("l_extendedprice", "l_discount", "l_shipdate")}
+    al["cn"] = cn.gather(spidx, nsm, st)
+    al["sn"] = sn
+    psch = {"sn": np.dtype(np.int32), "cn": np.dtype(np.int32),
+            "l_shipdate": np.dtype(np.int32),
+            "l_extendedprice": np.dtype(np.float64),
+            "l_discount": np.dtype(np.float64)}
+    pf = _cached_jit("f", lambda: jit.JitFilter(
+        "(sn = %d and cn = %d) or (sn = %d and cn = %d)"
+        % (fr, de, de, fr), psch),
+        "q7_pair_%d_%d" % (fr, de), _schema_key(psch))
+    fidx, nf = pf.run(al, st)
+    fin = {c: al[c].gather(fidx, nf, st) for c in psch}
+    out = {}
+    for y, pred in ((1995, "l_shipdate < date '1996-01-01'"),
+                    (1996, "l_shipdate >= date '1996-01-01'")):
+        agg = _cached_jit("a", lambda p=pred: jit.JitAggregate(
+            psch, [("sn", 25)],
+            ["SUM(l_extendedprice * (1 - l_discount)) as rev"],
+            predicate=p), "q7_rev_%d" % y)
+        acc = agg.make_acc()
+        if nf:
+            agg.run(fin, acc, st)
+        if st:
+            st.sync()
+        rev = agg.read(acc)[:, 0]
+        out[(names[fr], names[de], y)] = float(rev[fr])
+        out[(names[de], names[fr], y)] = float(rev[de])
+        acc.free()
+    for c in ([lidx, opidx, cbidx, ord_keys_j, ord_cnat_j, lpidx, obidx,
+               cn, spidx, sbidx, fidx] + list(g.values()) +
+              list(m.values()) + list(al.values()) + list(fin.values())):
+        c.free()
+    ctab.free()
+    otab.free()
+    stab.free()
+    return dict(sorted(out.items()))
+
+
+def q8(li_cols, ord_cols, cust_cols, supp_cols, part_cols,
+       nation_region, region_idx=1, target_nation=2, ptype_code=None,
+       stream=None):
+    """Device Q8 (tpch_ref.py:229-268): BRAZIL market share among
+    AMERICA-region customers for one part type, by order year.
+    part_cols: p_partkey + p_type (u8 code); nation_region: 25-entry
+    host array nationkey -> regionkey. Returns dict year -> share."""
+    from . import jit, ops
+    st = stream
+    # 'ECONOMY ANODIZED STEEL' = ((0*5)+0)*5+3 in the sorted-syllable
+    # code space (oracle.tpch_gen.PTYPE_SYL*)
+    code = ptype_code if ptype_code is not None else 3
+    # part rows of the target type -> semi table on p_partkey
+    pidx0, npf = ops.filter_col(part_cols["p_type"], ops.EQ, code, st)
+    pkeys = part_cols["p_partkey"].gather(pidx0, npf, st)
+    ptab = ops.JoinTable(max(16, npf), st)
+    if npf:
+        ptab.build(pkeys)
+    # orders in the window with AMERICA-region customers
+    osch = {k: v.dtype for k, v in ord_cols.items()}
+    of = _cached_jit("f", lambda: jit.JitFilter(
+        "o_orderdate >= date '1995-01-01' and "
+        "o_orderdate <= date '1996-12-31'", osch),
+        "q8_window", _schema_key(osch))
+    oidx, on = of.run(ord_cols, st)
+    o_key = ord_cols["o_orderkey"].gather(oidx, on, st)
+    o_cust = ord_cols["o_custkey"].gather(oidx, on, st)
+    o_date = ord_cols["o_orderdate"].gather(oidx, on, st)
+    ctab = ops.JoinTable(max(16, cust_cols["c_custkey"].n), st)
+    ctab.build(cust_cols["c_custkey"])
+    opidx, cbidx, ncm = ctab.probe(o_cust, mode=0, n=on)
+    cnat = cust_cols["c_nationkey"].gather(cbidx, ncm, st)
+    okey2 = o_key.gather(opidx, ncm, st)
+    odate2 = o_date.gather(opidx, ncm, st)
+    amer = [i for i in range(25) if nation_region[i] == region_idx]
+    csch = {"cnat": np.dtype(np.int32)}
+    cf = _cached_jit("f", lambda: jit.JitFilter(
+        " or ".join("cnat = %d" % a for a in amer), csch),
+        "q8_america_%d" % region_idx, _schema_key(csch))
+    aidx, na = cf.run({"cnat": cnat}, st)
+    okey3 = okey2.gather(aidx, na, st)
+    odate3 = odate2.gather(aidx, na, st)
+    otab = ops.JoinTable(max(16, na), st)
+    if na:
+        otab.build(okey3)
+    # lineitem: part semi, then orders join, then supplier nation
+    sp, _, nsp = ptab.probe(li_cols["l_partkey"], mode=1)
+    l_ok = li_cols["l_orderkey"].gather(sp, nsp, st)
+    l_sk = li_cols["l_suppkey"].gather(sp, nsp, st)
+    l_pr = li_cols["l_extendedprice"].gather(sp, nsp, st)
+    l_di = li_cols["l_discount"].gather(sp, nsp, st)
+    lp, ob, nlm = otab.probe(l_ok, mode=0, n=nsp)
+    odate4 = odate3.gather(ob, nlm, st)
+    m_sk = l_sk.gather(lp, nlm, st)
+    m_pr = l_pr.gather(lp, nlm, st)
+    m_di = l_di.gather(lp, nlm, st)
+    stab = ops.JoinTable(max(16, supp_cols["s_suppkey"].n), st)
+    stab.build(supp_cols["s_suppkey"])
+    spx, sbx, nsm = stab.probe(m_sk, mode=0, n=nlm)
+    sn = supp_cols["s_nationkey"].gather(sbx, nsm, st)
+    fin = {"sn": sn, "o_orderdate": odate4.gather(spx, nsm, st),
+           "l_extendedprice": m_pr.gather(spx, nsm, st),
+           "l_discount": m_di.gather(spx, nsm, st)}
+    fsch = {"sn": np.dtype(np.int32), "o_orderdate": np.dtype(np.int32),
+            "l_extendedprice": np.dtype(np.float64),
+            "l_discount": np.dtype(np.float64)}
+    out = {}
+    for y, pred in ((1995, "o_orderdate < date '1996-01-01'"),
+                    (1996, "o_orderdate >= date '1996-01-01'")):
+        agg = _cached_jit("a", lambda p=pred: jit.JitAggregate(
+            fsch, [("sn", 25)],
+            ["SUM(l_extendedprice * (1 - l_discount)) as rev"],
+            predicate=p), "q8_rev_%d" % y)
+        acc = agg.make_acc()
+        if nsm:
+            agg.run(fin, acc, st)
+        if st:
+            st.sync()
+        rev = agg.read(acc)[:, 0]
+        tot = rev.sum()
+        out[y] = float(rev[target_nation] / tot) if tot else 0.0
+        acc.free()
+    for c in ([pidx0, pkeys, oidx, o_key, o_cust, o_date, opidx, cbidx,
+               cnat, okey2, odate2, aidx, okey3, odate3, sp, l_ok, l_sk,
+               l_pr, l_di, lp, ob, odate4, m_sk, m_pr, m_di, spx, sbx] +
+              list(fin.values())):
+        c.free()
+    ptab.free()
+    ctab.free()
+    otab.free()
+    stab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

@@ -126,3 +126,48 @@ def test_q14_device_vs_oracle(gpu, data):
     for cs in (lcols, pcols):
         for c in cs.values():
             c.free()
+
+
+def test_q7_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, od = data["lineitem"], data["orders"]
+    cu, su, nat = data["customer"], data["supplier"], data["nation"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_suppkey",
+                                             "l_shipdate",
+                                             "l_extendedprice",
+                                             "l_discount"])
+    ocols = staging.stage_columns(od, names=["o_orderkey", "o_custkey"])
+    ccols = staging.stage_columns(cu, names=["c_custkey", "c_nationkey"])
+    scols = staging.stage_columns(su, names=["s_suppkey", "s_nationkey"])
+    got = DQ.q7(lcols, ocols, ccols, scols, nat["n_name"])
+    want = OQ.q7(li, od, cu, su, nat)
+    assert set(got) == set(want)
+    for k in want:
+        np.testing.assert_allclose(got[k], want[k], rtol=1e-9,
+                                   err_msg=str(k))
+    for cs in (lcols, ocols, ccols, scols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q8_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, od = data["lineitem"], data["orders"]
+    cu, su = data["customer"], data["supplier"]
+    part, nat, reg = data["part"], data["nation"], data["region"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_suppkey",
+                                             "l_partkey",
+                                             "l_extendedprice",
+                                             "l_discount"])
+    ocols = staging.stage_columns(od, names=["o_orderkey", "o_custkey",
+                                             "o_orderdate"])
+    ccols = staging.stage_columns(cu, names=["c_custkey", "c_nationkey"])
+    scols = staging.stage_columns(su, names=["s_suppkey", "s_nationkey"])
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_type"])
+    got = DQ.q8(lcols, ocols, ccols, scols, pcols, nat["n_regionkey"])
+    want = OQ.q8(li, od, cu, su, part, nat, reg)
+    for y in (1995, 1996):
+        np.testing.assert_allclose(got[y], want[y], rtol=1e-9, err_msg=y)
+    for cs in (lcols, ocols, ccols, scols, pcols):
+        for c in cs.values():
+            c.free()

@@ -315,3 +315,93 @@ def test_q14_oracle_equals_acero(data):
     got = 100.0 * promo / total
     want = OQ.q14(li, part)
     np.testing.assert_allclose(got, want, rtol=1e-9)
+
+
+def test_q7_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    cust, supp, nat = data["customer"], data["supplier"], data["nation"]
+    names = list(nat["n_name"])
+    fr, de = names.index("FRANCE"), names.index("GERMANY")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_suppkey",
+                                     "l_shipdate", "l_extendedprice",
+                                     "l_discount")})
+    l = l.filter(pc.and_(pc.greater_equal(l["l_shipdate"], OQ.Q7_LO),
+                         pc.less_equal(l["l_shipdate"], OQ.Q7_HI)))
+    o = pa.table({"o_orderkey": orders["o_orderkey"],
+                  "o_custkey": orders["o_custkey"]})
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "cn": cust["c_nationkey"]})
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"]})
+    j = l.join(o, keys="l_orderkey", right_keys="o_orderkey")
+    j = j.join(c, keys="o_custkey", right_keys="c_custkey")
+    j = j.join(s, keys="l_suppkey", right_keys="s_suppkey")
+    pair = pc.or_(pc.and_(pc.equal(j["sn"], fr), pc.equal(j["cn"], de)),
+                  pc.and_(pc.equal(j["sn"], de), pc.equal(j["cn"], fr)))
+    j = j.filter(pair)
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    j = j.append_column("rev", rev)
+    yr = pc.if_else(pc.greater_equal(j["l_shipdate"], OQ.Y1996),
+                    1996, 1995)
+    j = j.append_column("yr", yr)
+    g = j.group_by(["sn", "cn", "yr"]).aggregate([("rev", "sum")])
+    got = {}
+    for snv, cnv, yv, rv in zip(g.column("sn").to_pylist(),
+                                g.column("cn").to_pylist(),
+                                g.column("yr").to_pylist(),
+                                g.column("rev_sum").to_pylist()):
+        got[(names[snv], names[cnv], yv)] = rv
+    want = OQ.q7(li, orders, cust, supp, nat)
+    for k, v in want.items():
+        np.testing.assert_allclose(got.get(k, 0.0), v, rtol=1e-9,
+                                   err_msg=str(k))
+
+
+def test_q8_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    cust, supp = data["customer"], data["supplier"]
+    part, nat, reg = data["part"], data["nation"], data["region"]
+    america = G.REGIONS.index("AMERICA")
+    code = ((G.PTYPE_SYL1.index("ECONOMY") * 5 +
+             G.PTYPE_SYL2.index("ANODIZED")) * 5 +
+            G.PTYPE_SYL3.index("STEEL"))
+    p = pa.table({"p_partkey": part["p_partkey"],
+                  "p_type": part["p_type"]})
+    p = p.filter(pc.equal(p["p_type"], code))
+    o = pa.table({k: orders[k] for k in ("o_orderkey", "o_custkey",
+                                         "o_orderdate")})
+    o = o.filter(pc.and_(pc.greater_equal(o["o_orderdate"], OQ.Q7_LO),
+                         pc.less_equal(o["o_orderdate"], OQ.Q7_HI)))
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "cn": cust["c_nationkey"]})
+    amer = pa.table({"nk": np.nonzero(nat["n_regionkey"] == america)[0]
+                     .astype(np.int32)})
+    c = c.join(amer, keys="cn", right_keys="nk", join_type="left semi")
+    o = o.join(c.select(["c_custkey"]), keys="o_custkey",
+               right_keys="c_custkey", join_type="left semi")
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_suppkey",
+                                     "l_partkey", "l_extendedprice",
+                                     "l_discount")})
+    l = l.join(p.select(["p_partkey"]), keys="l_partkey",
+               right_keys="p_partkey", join_type="left semi")
+    j = l.join(o, keys="l_orderkey", right_keys="o_orderkey")
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"]})
+    j = j.join(s, keys="l_suppkey", right_keys="s_suppkey")
+    rev = pc.multiply(j["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), j["l_discount"]))
+    j = j.append_column("rev", rev)
+    brazil = list(nat["n_name"]).index("BRAZIL")
+    want = OQ.q8(li, orders, cust, supp, part, nat, reg)
+    for y, lo_ok in ((1995, True), (1996, True)):
+        if y == 1995:
+            m = pc.less(j["o_orderdate"], OQ.Y1996)
+        else:
+            m = pc.greater_equal(j["o_orderdate"], OQ.Y1996)
+        sel = j.filter(m)
+        tot = pc.sum(sel["rev"]).as_py() or 0.0
+        br = pc.sum(sel.filter(pc.equal(sel["sn"],
+                                        brazil))["rev"]).as_py() or 0.0
+        got = br / tot if tot else 0.0
+        np.testing.assert_allclose(got, want[y], rtol=1e-9, err_msg=y)
