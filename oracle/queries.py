@@ -358,6 +358,27 @@ def q8(li, orders, customer, supplier, part, nation, region):
     return out
 
 
+def q17(li, part, brand_code=12, container_code=17):
+    """tpch_ref.py:522-542: small-quantity ('below 20% of that part's
+    average quantity') revenue for one brand+container, / 7.0. Codes
+    default to arbitrary members of the generated code spaces (the
+    reference's 'Brand#23' / 'MED BOX' are dbgen strings; selection
+    logic, not the label, is what is under test). Returns the scalar."""
+    sel_part = (part["p_brand"] == brand_code) & \
+        (part["p_container"] == container_code)
+    nkey = int(part["p_partkey"].max()) + 2
+    qty_sum = np.bincount(li["l_partkey"], weights=li["l_quantity"],
+                          minlength=nkey)
+    qty_cnt = np.bincount(li["l_partkey"], minlength=nkey)
+    with np.errstate(invalid="ignore", divide="ignore"):
+        thr = 0.2 * qty_sum / qty_cnt
+    ok = np.zeros(nkey, dtype=bool)
+    ok[part["p_partkey"][sel_part]] = True
+    m = ok[li["l_partkey"]] & \
+        (li["l_quantity"] < thr[li["l_partkey"]])
+    return float(li["l_extendedprice"][m].sum() / 7.0)
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

@@ -242,10 +242,16 @@ def gen_part(sf, seed=42):
     25 == PTYPE_PROMO_SYL1."""
     n = n_parts(sf)
     rng = np.random.default_rng([seed, 5])
-    return {
+    out = {
         "p_partkey": np.arange(1, n + 1, dtype=np.int64),
         "p_type": rng.integers(0, 150, n).astype(np.uint8),
     }
+    # Q17/Q19 attributes, drawn after p_type (stream-append stable):
+    # p_brand 'Brand#MN' M,N in 1..5 -> code 0..24; p_container
+    # 4x10 spec combinations -> code 0..39
+    out["p_brand"] = rng.integers(0, 25, n).astype(np.uint8)
+    out["p_container"] = rng.integers(0, 40, n).astype(np.uint8)
+    return out
 
 
 def gen_nation():

@@ -1106,6 +1106,75 @@ def q8(li_cols, ord_cols, cust_cols, supp_cols, part_cols,
     return out
 
 
+def q17(li_cols, part_cols, brand_code=12, container_code=17,
+        stream=None):
+    """Device Q17 (tpch_ref.py:522-542): the correlated scalar subquery
+    (0.2 * avg quantity per part) becomes a device group-by over
+    l_partkey (sum + count), whose per-part thresholds ride along a
+    generic join table built on the qualifying parts; survivors of the
+    JIT col-vs-col filter (quantity < threshold) sum on device.
+    part_cols: p_partkey i64, p_brand u8, p_container u8."""
+    from . import jit, ops
+    st = stream
+    n = li_cols["l_partkey"].n
+    # per-part avg over the WHOLE lineitem: sum + count on the device
+    # group-by (the correlated subquery's aggregate)
+    ones = DevColumn(np.float64, max(1, n))
+    call("qk_fill_f64", st.handle if st else None, ones.ptr,
+         ctypes.c_double(1.0), c_u64(n))
+    gb = ops.GroupByI64(expected_groups=max(1024, n // 4), nvals=2,
+                        stream=st)
+    gb.update(li_cols["l_partkey"], [li_cols["l_quantity"], ones], n)
+    pk_all, sums = gb.extract()
+    gb.free()
+    ones.free()
+    thr_all = 0.2 * sums[0] / np.maximum(sums[1], 1.0)
+    # qualifying parts (brand AND container) -> join table + aligned
+    # threshold column
+    bidx0, nb = ops.filter_col(part_cols["p_brand"], ops.EQ, brand_code,
+                               st)
+    bkeys = part_cols["p_partkey"].gather(bidx0, nb, st)
+    bcont = part_cols["p_container"].gather(bidx0, nb, st)
+    cidx0, ncp = ops.filter_col(bcont, ops.EQ, container_code, st)
+    qkeys_col = bkeys.gather(cidx0, ncp, st)
+    if st:
+        st.sync()
+    qkeys = qkeys_col.to_numpy(ncp)
+    # threshold per qualifying part via searchsorted over the group keys
+    order = np.argsort(pk_all)
+    pos = np.searchsorted(pk_all, qkeys, sorter=order)
+    have = (pos < len(pk_all)) & \
+        (pk_all[order[np.minimum(pos, len(pk_all) - 1)]] == qkeys)
+    qkeys = qkeys[have]
+    qthr = thr_all[order[pos[have]]]
+    ptab = ops.JoinTable(max(16, len(qkeys)), st)
+    dthr = DevColumn.from_numpy(qthr)
+    dkeys = DevColumn.from_numpy(qkeys)
+    if len(qkeys):
+        ptab.build(dkeys)
+    pidx, bidx, nm = ptab.probe(li_cols["l_partkey"], mode=0)
+    m_qty = li_cols["l_quantity"].gather(pidx, nm, st)
+    m_pr = li_cols["l_extendedprice"].gather(pidx, nm, st)
+    m_thr = dthr.gather(bidx, nm, st)
+    fin = {"l_quantity": m_qty, "thr": m_thr, "l_extendedprice": m_pr}
+    fsch = {k: np.dtype(np.float64) for k in fin}
+    agg = _cached_jit("a", lambda: jit.JitAggregate(
+        fsch, [], ["SUM(l_extendedprice) as s"],
+        predicate="l_quantity < thr"), "q17_sum")
+    acc = agg.make_acc()
+    if nm:
+        agg.run(fin, acc, st)
+    if st:
+        st.sync()
+    out = float(agg.read(acc)[0, 0] / 7.0)
+    for c in (bidx0, bkeys, bcont, cidx0, qkeys_col, dthr, dkeys, pidx,
+              bidx, m_qty, m_pr, m_thr):
+        c.free()
+    acc.free()
+    ptab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

@@ -171,3 +171,18 @@ def test_q8_device_vs_oracle(gpu, data):
     for cs in (lcols, ocols, ccols, scols, pcols):
         for c in cs.values():
             c.free()
+
+
+def test_q17_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, part = data["lineitem"], data["part"]
+    lcols = staging.stage_columns(li, names=["l_partkey", "l_quantity",
+                                             "l_extendedprice"])
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_brand",
+                                               "p_container"])
+    got = DQ.q17(lcols, pcols)
+    want = OQ.q17(li, part)
+    np.testing.assert_allclose(got, want, rtol=1e-9)
+    for cs in (lcols, pcols):
+        for c in cs.values():
+            c.free()

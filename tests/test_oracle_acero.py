@@ -405,3 +405,25 @@ def test_q8_oracle_equals_acero(data):
                                         brazil))["rev"]).as_py() or 0.0
         got = br / tot if tot else 0.0
         np.testing.assert_allclose(got, want[y], rtol=1e-9, err_msg=y)
+
+
+def test_q17_oracle_equals_acero(data):
+    li, part = data["lineitem"], data["part"]
+    g = pa.table({"l_partkey": li["l_partkey"],
+                  "l_quantity": li["l_quantity"]}) \
+        .group_by("l_partkey").aggregate([("l_quantity", "mean")])
+    p = pa.table({"p_partkey": part["p_partkey"],
+                  "b": part["p_brand"], "c": part["p_container"]})
+    p = p.filter(pc.and_(pc.equal(p["b"], 12), pc.equal(p["c"], 17)))
+    l = pa.table({k: li[k] for k in ("l_partkey", "l_quantity",
+                                     "l_extendedprice")})
+    j = l.join(p.select(["p_partkey"]), keys="l_partkey",
+               right_keys="p_partkey", join_type="left semi")
+    j = j.join(g, keys="l_partkey", right_keys="l_partkey",
+               join_type="inner", right_suffix="_avg")
+    j = j.filter(pc.less(j["l_quantity"],
+                         pc.multiply(pa.scalar(0.2),
+                                     j["l_quantity_mean"])))
+    got = (pc.sum(j["l_extendedprice"]).as_py() or 0.0) / 7.0
+    want = OQ.q17(li, part)
+    np.testing.assert_allclose(got, want, rtol=1e-9)
