@@ -379,6 +379,63 @@ def q17(li, part, brand_code=12, container_code=17):
     return float(li["l_extendedprice"][m].sum() / 7.0)
 
 
+Q15_LO = G.days(1996, 1, 1)
+Q15_HI = G.days(1996, 4, 1)     # + interval '3' month
+
+
+def q15(li, supplier):
+    """tpch_ref.py:452-485: revenue per supplier over the 3-month window;
+    suppliers whose revenue equals the maximum. Returns (suppkeys asc,
+    max_revenue)."""
+    m = (li["l_shipdate"] >= Q15_LO) & (li["l_shipdate"] < Q15_HI)
+    nkey = int(supplier["s_suppkey"].max()) + 2
+    rev = np.bincount(li["l_suppkey"][m],
+                      weights=(li["l_extendedprice"][m]
+                               * (1.0 - li["l_discount"][m])),
+                      minlength=nkey)
+    mx = rev.max()
+    winners = np.nonzero(rev == mx)[0]
+    return winners.astype(np.int64), float(mx)
+
+
+# Q19 branch constants (tpch_ref.py:582-620), resolved to dictionary
+# codes via the committed code tables (oracle.tpch_gen)
+Q19_BRANCHES = [
+    ("Brand#12", ["SM CASE", "SM BOX", "SM PACK", "SM PKG"], 1, 11, 1, 5),
+    ("Brand#23", ["MED BAG", "MED BOX", "MED PKG", "MED PACK"], 10, 20, 1, 10),
+    ("Brand#34", ["LG CASE", "LG BOX", "LG PACK", "LG PKG"], 20, 30, 1, 15),
+]
+
+
+def q19(li, part):
+    """tpch_ref.py:582-620: sum of revenue over the OR of three
+    brand/container/quantity/size/shipmode/shipinstruct branches."""
+    air = [G.SHIPMODE.index("AIR"), G.SHIPMODE.index("REG AIR")]
+    deliver = G.SHIPINSTRUCT.index("DELIVER IN PERSON")
+    nkey = int(part["p_partkey"].max()) + 2
+    brand = np.zeros(nkey, dtype=np.int16)
+    cont = np.zeros(nkey, dtype=np.int16)
+    size = np.zeros(nkey, dtype=np.int16)
+    brand[part["p_partkey"]] = part["p_brand"]
+    cont[part["p_partkey"]] = part["p_container"]
+    size[part["p_partkey"]] = part["p_size"]
+    lb = brand[li["l_partkey"]]
+    lc = cont[li["l_partkey"]]
+    ls = size[li["l_partkey"]]
+    common = (np.isin(li["l_shipmode"], air)
+              & (li["l_shipinstruct"] == deliver))
+    m = np.zeros(len(lb), dtype=bool)
+    for bname, conts, qlo, qhi, slo, shi in Q19_BRANCHES:
+        bc = G.brand_code(bname)
+        cc = [G.container_code(c) for c in conts]
+        m |= ((lb == bc) & np.isin(lc, cc)
+              & (li["l_quantity"] >= qlo) & (li["l_quantity"] <= qhi)
+              & (ls >= slo) & (ls <= shi))
+    m &= common
+    return float((li["l_extendedprice"][m]
+                  * (1.0 - li["l_discount"][m])).sum())
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

@@ -59,6 +59,21 @@ MKTSEGMENT = ["AUTOMOBILE", "BUILDING", "FURNITURE", "HOUSEHOLD", "MACHINERY"]
 ORDERPRIORITY = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED",
                  "5-LOW"]             # spec 4.2.3 uniform, sorted order
 SHIPMODE = ["AIR", "FOB", "MAIL", "RAIL", "REG AIR", "SHIP", "TRUCK"]
+SHIPINSTRUCT = ["COLLECT COD", "DELIVER IN PERSON", "NONE",
+                "TAKE BACK RETURN"]
+# p_container = size x kind (spec 4.2.2.13); code = c1 * 8 + c2
+CONTAINER1 = ["JUMBO", "LG", "MED", "SM", "WRAP"]
+CONTAINER2 = ["BAG", "BOX", "CAN", "CASE", "DRUM", "JAR", "PACK", "PKG"]
+
+
+def container_code(name):
+    a, b = name.split(" ", 1)
+    return CONTAINER1.index(a) * 8 + CONTAINER2.index(b)
+
+
+def brand_code(name):
+    mn = name.split("#", 1)[1]           # 'Brand#MN', M,N in 1..5
+    return (int(mn[0]) - 1) * 5 + (int(mn[1]) - 1)
 # p_type = syllable1 x syllable2 x syllable3 (spec 4.2.2.13); code =
 # ((s1 * 5) + s2) * 5 + s3; LIKE 'PROMO%' == s1 == index of "PROMO"
 PTYPE_SYL1 = ["ECONOMY", "LARGE", "MEDIUM", "PROMO", "SMALL", "STANDARD"]
@@ -191,6 +206,8 @@ def gen_lineitem(sf, seed=42, orders=None):
     }
     # drawn LAST so earlier columns' RNG streams stay fixture-stable
     out["l_shipmode"] = rng.integers(0, len(SHIPMODE), n).astype(np.uint8)
+    out["l_shipinstruct"] = rng.integers(
+        0, len(SHIPINSTRUCT), n).astype(np.uint8)
     return out
 
 
@@ -247,10 +264,12 @@ def gen_part(sf, seed=42):
         "p_type": rng.integers(0, 150, n).astype(np.uint8),
     }
     # Q17/Q19 attributes, drawn after p_type (stream-append stable):
-    # p_brand 'Brand#MN' M,N in 1..5 -> code 0..24; p_container
-    # 4x10 spec combinations -> code 0..39
+    # p_brand 'Brand#MN' M,N in 1..5 -> code 0..24 (brand_code);
+    # p_container 5x8 spec combinations -> code 0..39 (container_code);
+    # p_size uniform 1..50 (spec 4.2.3)
     out["p_brand"] = rng.integers(0, 25, n).astype(np.uint8)
     out["p_container"] = rng.integers(0, 40, n).astype(np.uint8)
+    out["p_size"] = rng.integers(1, 51, n).astype(np.uint8)
     return out
 
 

@@ -1175,6 +1175,85 @@ def q17(li_cols, part_cols, brand_code=12, container_code=17,
     return out
 
 
+def q15(li_cols, stream=None):
+    """Device Q15 (tpch_ref.py:452-485): per-supplier revenue over the
+    window on the device group-by; the max + winner set resolve host-side
+    over the extracted (tiny) per-supplier totals. Returns
+    (winner suppkeys asc, max_revenue)."""
+    from . import jit, ops
+    st = stream
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_shipdate >= date '1996-01-01' and "
+        "l_shipdate < date '1996-01-01' + interval '3' month", lsch),
+        "q15_window", _schema_key(lsch))
+    lidx, ln = lf.run(li_cols, st)
+    sk = li_cols["l_suppkey"].gather(lidx, ln, st)
+    pr = li_cols["l_extendedprice"].gather(lidx, ln, st)
+    di = li_cols["l_discount"].gather(lidx, ln, st)
+    rev = _mul_1md(pr, di, st)
+    gb = ops.GroupByI64(expected_groups=max(1024, ln // 16), nvals=1,
+                        stream=st)
+    gb.update(sk, [rev], ln)
+    keys, sums = gb.extract()
+    gb.free()
+    mx = sums[0].max() if len(keys) else 0.0
+    winners = np.sort(keys[sums[0] == mx])
+    for c in (lidx, sk, pr, di, rev):
+        c.free()
+    return winners, float(mx)
+
+
+def q19(li_cols, part_cols, stream=None):
+    """Device Q19 (tpch_ref.py:582-620): the three-branch OR predicate
+    over mixed lineitem/part attributes compiles to ONE fused JIT
+    scan-aggregate (predicate + SUM(revenue) in the same kernel) after a
+    generic join attaches the part attributes to each line."""
+    from . import jit, ops
+    st = stream
+    ptab = ops.JoinTable(max(16, part_cols["p_partkey"].n), st)
+    ptab.build(part_cols["p_partkey"])
+    pidx, bidx, nm = ptab.probe(li_cols["l_partkey"], mode=0)
+    fin = {}
+    for c in ("l_quantity", "l_extendedprice", "l_discount",
+              "l_shipmode", "l_shipinstruct"):
+        fin[c] = li_cols[c].gather(pidx, nm, st)
+    for c in ("p_brand", "p_container", "p_size"):
+        fin[c] = part_cols[c].gather(bidx, nm, st)
+    fsch = {k: v.dtype for k, v in fin.items()}
+    # branch constants resolved to the committed code tables
+    # (oracle.tpch_gen brand_code/container_code); AIR=0, REG AIR=4,
+    # DELIVER IN PERSON=1
+    branches = [
+        (1, (27, 25, 30, 31), 1, 11, 1, 5),     # Brand#12, SM *
+        (7, (16, 17, 23, 22), 10, 20, 1, 10),   # Brand#23, MED *
+        (13, (11, 9, 14, 15), 20, 30, 1, 15),   # Brand#34, LG *
+    ]
+    parts = []
+    for bc, cc, qlo, qhi, slo, shi in branches:
+        parts.append(
+            "(p_brand = %d and p_container in (%s) and "
+            "l_quantity >= %d and l_quantity <= %d and "
+            "p_size >= %d and p_size <= %d)"
+            % (bc, ", ".join(str(c) for c in cc), qlo, qhi, slo, shi))
+    pred = ("(%s) and l_shipinstruct = 1 and l_shipmode in (0, 4)"
+            % " or ".join(parts))
+    agg = _cached_jit("a", lambda: jit.JitAggregate(
+        fsch, [], ["SUM(l_extendedprice * (1 - l_discount)) as rev"],
+        predicate=pred), "q19_rev", _schema_key(fsch))
+    acc = agg.make_acc()
+    if nm:
+        agg.run(fin, acc, st)
+    if st:
+        st.sync()
+    out = float(agg.read(acc)[0, 0])
+    for c in [pidx, bidx] + list(fin.values()):
+        c.free()
+    acc.free()
+    ptab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

@@ -186,3 +186,34 @@ def test_q17_device_vs_oracle(gpu, data):
     for cs in (lcols, pcols):
         for c in cs.values():
             c.free()
+
+
+def test_q15_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, supp = data["lineitem"], data["supplier"]
+    lcols = staging.stage_columns(li, names=["l_suppkey", "l_shipdate",
+                                             "l_extendedprice",
+                                             "l_discount"])
+    wk, wmx = OQ.q15(li, supp)
+    gk, gmx = DQ.q15(lcols)
+    assert list(gk) == list(wk)
+    np.testing.assert_allclose(gmx, wmx, rtol=1e-9)
+    for c in lcols.values():
+        c.free()
+
+
+def test_q19_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, part = data["lineitem"], data["part"]
+    lcols = staging.stage_columns(li, names=["l_partkey", "l_quantity",
+                                             "l_extendedprice",
+                                             "l_discount", "l_shipmode",
+                                             "l_shipinstruct"])
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_brand",
+                                               "p_container", "p_size"])
+    got = DQ.q19(lcols, pcols)
+    want = OQ.q19(li, part)
+    np.testing.assert_allclose(got, want, rtol=1e-9)
+    for cs in (lcols, pcols):
+        for c in cs.values():
+            c.free()

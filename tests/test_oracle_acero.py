@@ -427,3 +427,58 @@ def test_q17_oracle_equals_acero(data):
     got = (pc.sum(j["l_extendedprice"]).as_py() or 0.0) / 7.0
     want = OQ.q17(li, part)
     np.testing.assert_allclose(got, want, rtol=1e-9)
+
+
+def test_q15_oracle_equals_acero(data):
+    li, supp = data["lineitem"], data["supplier"]
+    l = pa.table({k: li[k] for k in ("l_suppkey", "l_shipdate",
+                                     "l_extendedprice", "l_discount")})
+    l = l.filter(pc.and_(pc.greater_equal(l["l_shipdate"], OQ.Q15_LO),
+                         pc.less(l["l_shipdate"], OQ.Q15_HI)))
+    rev = pc.multiply(l["l_extendedprice"],
+                      pc.subtract(pa.scalar(1.0), l["l_discount"]))
+    l = l.append_column("rev", rev)
+    g = l.group_by("l_suppkey").aggregate([("rev", "sum")])
+    mx = pc.max(g.column("rev_sum")).as_py()
+    winners = sorted(g.filter(pc.equal(g["rev_sum"], mx))
+                     .column("l_suppkey").to_pylist())
+    wk, wmx = OQ.q15(li, supp)
+    assert winners == list(wk)
+    np.testing.assert_allclose(mx, wmx, rtol=1e-9)
+
+
+def test_q19_oracle_equals_acero(data):
+    li, part = data["lineitem"], data["part"]
+    air = [G.SHIPMODE.index("AIR"), G.SHIPMODE.index("REG AIR")]
+    deliver = G.SHIPINSTRUCT.index("DELIVER IN PERSON")
+    l = pa.table({k: li[k] for k in ("l_partkey", "l_quantity",
+                                     "l_extendedprice", "l_discount",
+                                     "l_shipmode", "l_shipinstruct")})
+    p = pa.table({k: part[k] for k in ("p_partkey", "p_brand",
+                                       "p_container", "p_size")})
+    j = l.join(p, keys="l_partkey", right_keys="p_partkey",
+               join_type="inner")
+    m = pa.array(np.zeros(j.num_rows, dtype=bool))
+    for bname, conts, qlo, qhi, slo, shi in OQ.Q19_BRANCHES:
+        bc = G.brand_code(bname)
+        cc = [G.container_code(c) for c in conts]
+        b = pc.and_(
+            pc.and_(pc.equal(j["p_brand"], bc),
+                    pc.is_in(j["p_container"],
+                             value_set=pa.array(cc, type=pa.uint8()))),
+            pc.and_(
+                pc.and_(pc.greater_equal(j["l_quantity"], qlo),
+                        pc.less_equal(j["l_quantity"], qhi)),
+                pc.and_(pc.greater_equal(j["p_size"], slo),
+                        pc.less_equal(j["p_size"], shi))))
+        m = pc.or_(m, b)
+    m = pc.and_(m, pc.and_(
+        pc.is_in(j["l_shipmode"], value_set=pa.array(air,
+                                                     type=pa.uint8())),
+        pc.equal(j["l_shipinstruct"], deliver)))
+    sel = j.filter(m)
+    got = pc.sum(pc.multiply(sel["l_extendedprice"],
+                             pc.subtract(pa.scalar(1.0),
+                                         sel["l_discount"]))).as_py() or 0.0
+    want = OQ.q19(li, part)
+    np.testing.assert_allclose(got, want, rtol=1e-9)
