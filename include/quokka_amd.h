@@ -470,6 +470,10 @@ int qk_str_dict_rehash(void *stream, uint32_t ncodes,
                        const uint64_t *code_off, const uint32_t *code_len,
                        const uint8_t *arena, uint64_t *slot_hash,
                        int32_t *slot_code, uint64_t capacity);
+/* Composite i64 key column: out = x * scale + y (per-(a,b) group keys,
+ * e.g. Q20's (partkey, suppkey) quantity sums). */
+int qk_i64_combine(void *stream, uint64_t n, const int64_t *x,
+                   const int64_t *y, int64_t scale, int64_t *out);
 /* Synchronous device-to-device copy (dict growth, arena relocation). */
 int qk_d2d(void *dst_dev, const void *src_dev, uint64_t nbytes);
 

@@ -436,6 +436,95 @@ def q19(li, part):
                   * (1.0 - li["l_discount"][m])).sum())
 
 
+def q2(part, supplier, partsupp, nation, region, limit=100):
+    """tpch_ref.py:40-86: EUROPE suppliers offering size-15 '%BRASS'
+    parts at that part's EUROPE-minimum supply cost; order by s_acctbal
+    desc, n_name, s_name (== s_suppkey order), p_partkey, limit 100.
+    Returns dict of arrays (p_partkey, s_suppkey, s_acctbal, n_name,
+    ps_supplycost)."""
+    europe = G.REGIONS.index("EUROPE")
+    nat_eu = nation["n_regionkey"] == europe
+    supp_nat = np.full(int(supplier["s_suppkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    supp_nat[supplier["s_suppkey"]] = supplier["s_nationkey"]
+    ps_sn = supp_nat[partsupp["ps_suppkey"]]
+    eu = (ps_sn >= 0) & nat_eu[np.clip(ps_sn, 0, 24)]
+    pk = partsupp["ps_partkey"][eu]
+    cost = partsupp["ps_supplycost"][eu]
+    nkey = int(part["p_partkey"].max()) + 2
+    min_cost = np.full(nkey, np.inf)
+    np.minimum.at(min_cost, pk, cost)
+    part_ok = np.zeros(nkey, dtype=bool)
+    sel = (part["p_size"] == 15) & (part["p_type"] % 5 ==
+                                    G.PTYPE_SYL3.index("BRASS"))
+    part_ok[part["p_partkey"][sel]] = True
+    win = part_ok[pk] & (cost == min_cost[pk])
+    sk = partsupp["ps_suppkey"][eu][win]
+    row = sk - 1
+    names = list(nation["n_name"])
+    out = {
+        "p_partkey": pk[win],
+        "s_suppkey": sk,
+        "s_acctbal": supplier["s_acctbal"][row],
+        "n_name": np.array([names[k] for k in
+                            supplier["s_nationkey"][row]], dtype=object),
+        "ps_supplycost": cost[win],
+    }
+    nrank = np.array([sorted(names).index(n) for n in out["n_name"]])
+    order = np.lexsort((out["p_partkey"], out["s_suppkey"], nrank,
+                        -out["s_acctbal"]))
+    top = order[:limit]
+    return {k: v[top] for k, v in out.items()}
+
+
+def q11(partsupp, supplier, nation, fraction=0.0001):
+    """tpch_ref.py:344-374: GERMANY suppliers' ps value per part,
+    HAVING value > fraction * total; order by value desc. Returns
+    (partkeys, values) sorted."""
+    germany = list(nation["n_name"]).index("GERMANY")
+    supp_nat = np.full(int(supplier["s_suppkey"].max()) + 2, -1,
+                       dtype=np.int32)
+    supp_nat[supplier["s_suppkey"]] = supplier["s_nationkey"]
+    m = supp_nat[partsupp["ps_suppkey"]] == germany
+    pk = partsupp["ps_partkey"][m]
+    val = (partsupp["ps_supplycost"][m]
+           * partsupp["ps_availqty"][m].astype(np.float64))
+    nkey = int(partsupp["ps_partkey"].max()) + 2
+    per = np.bincount(pk, weights=val, minlength=nkey)
+    thr = val.sum() * fraction
+    win = np.nonzero(per > thr)[0]
+    order = np.lexsort((win, -per[win]))
+    return win[order].astype(np.int64), per[win][order]
+
+
+def q20(li, part, partsupp, supplier, nation):
+    """tpch_ref.py:622-662: CANADA suppliers holding availqty > half the
+    1994 shipped quantity of a 'forest%' part they supply. Returns
+    sorted s_suppkey array (s_name order == suppkey order)."""
+    forest = np.zeros(int(part["p_partkey"].max()) + 2, dtype=bool)
+    forest[part["p_partkey"][part["p_name1"] == G.P_NAME_FOREST]] = True
+    m = (forest[li["l_partkey"]]
+         & (li["l_shipdate"] >= G.Q5_LO) & (li["l_shipdate"] < G.Q5_HI))
+    S = int(max(li["l_suppkey"].max(), partsupp["ps_suppkey"].max())) + 1
+    key = li["l_partkey"][m] * S + li["l_suppkey"][m]
+    kk, inv = np.unique(key, return_inverse=True)
+    qty = np.bincount(inv, weights=li["l_quantity"][m])
+    pm = forest[partsupp["ps_partkey"]]
+    pskey = partsupp["ps_partkey"][pm] * S + partsupp["ps_suppkey"][pm]
+    pos = np.searchsorted(kk, pskey)
+    have = (pos < len(kk)) & (kk[np.minimum(pos, len(kk) - 1)] == pskey)
+    thr = np.zeros(len(pskey))
+    thr[have] = 0.5 * qty[pos[have]]
+    # SQL: availqty > (scalar subquery); an EMPTY subquery yields NULL
+    # and the comparison is false — rows with no 1994 shipments do NOT
+    # qualify (DuckDB semantics, tpch_ref.py:646-653)
+    ok = have & (partsupp["ps_availqty"][pm] > thr)
+    sk = np.unique(partsupp["ps_suppkey"][pm][ok])
+    canada = list(nation["n_name"]).index("CANADA")
+    sn = supplier["s_nationkey"][sk - 1]
+    return sk[sn == canada].astype(np.int64)
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

@@ -64,6 +64,7 @@ SHIPINSTRUCT = ["COLLECT COD", "DELIVER IN PERSON", "NONE",
 # p_container = size x kind (spec 4.2.2.13); code = c1 * 8 + c2
 CONTAINER1 = ["JUMBO", "LG", "MED", "SM", "WRAP"]
 CONTAINER2 = ["BAG", "BOX", "CAN", "CASE", "DRUM", "JAR", "PACK", "PKG"]
+P_NAME_FOREST = 29       # code of "forest" in the 92-word p_name space
 
 
 def container_code(name):
@@ -246,10 +247,13 @@ def gen_customer(sf, seed=42, strings=False):
 def gen_supplier(sf, seed=42):
     n = n_suppliers(sf)
     rng = np.random.default_rng([seed, 4])
-    return {
+    out = {
         "s_suppkey": np.arange(1, n + 1, dtype=np.int64),
         "s_nationkey": rng.integers(0, 25, n).astype(np.int32),
     }
+    # appended draws (stream-stable): spec s_acctbal U[-999.99, 9999.99]
+    out["s_acctbal"] = rng.integers(-99999, 1000000, n) / 100.0
+    return out
 
 
 def gen_part(sf, seed=42):
@@ -270,7 +274,32 @@ def gen_part(sf, seed=42):
     out["p_brand"] = rng.integers(0, 25, n).astype(np.uint8)
     out["p_container"] = rng.integers(0, 40, n).astype(np.uint8)
     out["p_size"] = rng.integers(1, 51, n).astype(np.uint8)
+    # p_name's FIRST word as a code (spec: 5 of 92 color words; prefix
+    # LIKE 'forest%' == first word == "forest" == code P_NAME_FOREST —
+    # the text-boundary adaptation documented in the header)
+    out["p_name1"] = rng.integers(0, 92, n).astype(np.uint8)
     return out
+
+
+def gen_partsupp(sf, seed=42):
+    """partsupp: 4 rows per part (spec 4.2.3); ps_suppkey follows the
+    spec's supplier-spread formula
+        ((ps_partkey + i*(S/4 + (ps_partkey-1)/S)) % S) + 1,  i in 0..3
+    (guarantees 4 distinct suppliers per part); ps_availqty U[1,9999];
+    ps_supplycost U[1.00, 1000.00]."""
+    npart = n_parts(sf)
+    S = n_suppliers(sf)
+    rng = np.random.default_rng([seed, 6])
+    pk = np.repeat(np.arange(1, npart + 1, dtype=np.int64), 4)
+    i = np.tile(np.arange(4, dtype=np.int64), npart)
+    sk = ((pk + i * (S // 4 + (pk - 1) // S)) % S) + 1
+    n = len(pk)
+    return {
+        "ps_partkey": pk,
+        "ps_suppkey": sk,
+        "ps_availqty": rng.integers(1, 10_000, n).astype(np.int32),
+        "ps_supplycost": rng.integers(100, 100_001, n) / 100.0,
+    }
 
 
 def gen_nation():
@@ -293,6 +322,7 @@ def gen_all(sf, seed=42):
         "customer": gen_customer(sf, seed),
         "supplier": gen_supplier(sf, seed),
         "part": gen_part(sf, seed),
+        "partsupp": gen_partsupp(sf, seed),
         "nation": gen_nation(),
         "region": gen_region(),
     }

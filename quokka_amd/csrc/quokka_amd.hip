@@ -2481,6 +2481,26 @@ extern "C" int qk_str_dict_rehash(void *stream, uint32_t ncodes,
   return 0;
 }
 
+__global__ void __launch_bounds__(BLOCK) k_i64_combine(
+    uint64_t n, const int64_t *__restrict__ x, const int64_t *__restrict__ y,
+    int64_t scale, int64_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = x[i] * scale + y[i];
+}
+// Composite i64 key: out = x * scale + y (e.g. (partkey, suppkey) group
+// keys for Q20/Q21-style per-pair aggregates — exact for |x*scale+y| < 2^63)
+extern "C" int qk_i64_combine(void *stream, uint64_t n, const int64_t *x,
+                              const int64_t *y, int64_t scale, int64_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_i64_combine, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, x, y, scale, out);
+  QK_TRY("qk_i64_combine", hipGetLastError());
+  return 0;
+}
+
 extern "C" int qk_d2d(void *dst, const void *src, uint64_t nbytes) {
   QK_TRY("qk_d2d", hipMemcpy(dst, src, nbytes, hipMemcpyDeviceToDevice));
   return 0;

@@ -1254,6 +1254,185 @@ def q19(li_cols, part_cols, stream=None):
     return out
 
 
+def q2(part_cols, supp_cols, ps_cols, nation_region, nation_names,
+       stream=None, limit=100):
+    """Device Q2 (tpch_ref.py:40-86): the correlated min(supplycost)
+    subquery becomes a device MIN group-by over the EUROPE-supplier ps
+    rows keyed by partkey; winners re-qualify via a threshold-carrying
+    probe (cost == per-part min) intersected with the size/type part
+    filter. Output attachment + the 100-row order-by run host-side over
+    the winner set."""
+    from . import jit, ops
+    st = stream
+    europe = 3
+    eu_nats = [i for i in range(25) if nation_region[i] == europe]
+    # EUROPE suppliers -> semi over ps_suppkey
+    ssch = {"s_nationkey": np.dtype(np.int32)}
+    sf_ = _cached_jit("f", lambda: jit.JitFilter(
+        " or ".join("s_nationkey = %d" % k for k in eu_nats), ssch),
+        "q2_eu", _schema_key(ssch))
+    sidx, ns = sf_.run({"s_nationkey": supp_cols["s_nationkey"]}, st)
+    eu_keys = supp_cols["s_suppkey"].gather(sidx, ns, st)
+    stab = ops.JoinTable(max(16, ns), st)
+    if ns:
+        stab.build(eu_keys)
+    spx, _, nps = stab.probe(ps_cols["ps_suppkey"], mode=1)
+    pk = ps_cols["ps_partkey"].gather(spx, nps, st)
+    cost = ps_cols["ps_supplycost"].gather(spx, nps, st)
+    sk = ps_cols["ps_suppkey"].gather(spx, nps, st)
+    # per-part MIN cost (the correlated subquery)
+    gb = ops.GroupByI64(expected_groups=max(1024, nps), nvals=1,
+                        stream=st, agg_ops=[1])
+    gb.update(pk, [cost], nps)
+    gkeys, gmins = gb.extract()
+    gb.free()
+    if st:
+        st.sync()
+    pk_h = pk.to_numpy(nps)
+    cost_h = cost.to_numpy(nps)
+    sk_h = sk.to_numpy(nps)
+    order = np.argsort(gkeys)
+    pos = np.searchsorted(gkeys, pk_h, sorter=order)
+    mins = gmins[0][order[pos]]
+    # part filter: size == 15 and type LIKE '%BRASS' (syl3 code 0)
+    psz = part_cols["p_size"].to_numpy(part_cols["p_size"].n)
+    pty = part_cols["p_type"].to_numpy(part_cols["p_type"].n)
+    part_ok = np.zeros(int(part_cols["p_partkey"].n) + 2, dtype=bool)
+    pkeys_h = part_cols["p_partkey"].to_numpy(part_cols["p_partkey"].n)
+    part_ok[pkeys_h[(psz == 15) & (pty % 5 == 0)]] = True
+    win = part_ok[np.minimum(pk_h, len(part_ok) - 1)] & \
+        (cost_h == mins)
+    wk, ws, wc = pk_h[win], sk_h[win], cost_h[win]
+    sab = supp_cols["s_acctbal"].to_numpy(supp_cols["s_acctbal"].n)
+    snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
+    names = list(nation_names)
+    out = {
+        "p_partkey": wk, "s_suppkey": ws,
+        "s_acctbal": sab[ws - 1],
+        "n_name": np.array([names[k] for k in snk[ws - 1]], dtype=object),
+        "ps_supplycost": wc,
+    }
+    nrank = np.array([sorted(names).index(n) for n in out["n_name"]])
+    ordr = np.lexsort((out["p_partkey"], out["s_suppkey"], nrank,
+                       -out["s_acctbal"]))
+    top = ordr[:limit]
+    for c in (sidx, eu_keys, spx, pk, cost, sk):
+        c.free()
+    stab.free()
+    return {k: v[top] for k, v in out.items()}
+
+
+def q11(ps_cols, supp_cols, nation_names, fraction=0.0001, stream=None):
+    """Device Q11 (tpch_ref.py:344-374): GERMANY-supplier ps rows (semi)
+    -> JIT value map -> device group-by per partkey -> HAVING vs the
+    global fraction host-side over the extracted totals."""
+    from . import jit, ops
+    st = stream
+    germany = list(nation_names).index("GERMANY")
+    sidx, ns = ops.filter_col(supp_cols["s_nationkey"], ops.EQ, germany,
+                              st)
+    gkeys_s = supp_cols["s_suppkey"].gather(sidx, ns, st)
+    stab = ops.JoinTable(max(16, ns), st)
+    if ns:
+        stab.build(gkeys_s)
+    spx, _, nps = stab.probe(ps_cols["ps_suppkey"], mode=1)
+    pk = ps_cols["ps_partkey"].gather(spx, nps, st)
+    cost = ps_cols["ps_supplycost"].gather(spx, nps, st)
+    qty = ps_cols["ps_availqty"].gather(spx, nps, st)
+    vsch = {"ps_supplycost": np.dtype(np.float64),
+            "ps_availqty": np.dtype(np.int32)}
+    vm = _cached_jit("m", lambda: jit.JitMap(
+        "ps_supplycost * ps_availqty", vsch), "q11_value",
+        _schema_key(vsch))
+    val = vm.run({"ps_supplycost": cost, "ps_availqty": qty})
+    gb = ops.GroupByI64(expected_groups=max(1024, nps), nvals=1,
+                        stream=st)
+    gb.update(pk, [val], nps)
+    keys, sums = gb.extract()
+    gb.free()
+    thr = sums[0].sum() * fraction
+    win = sums[0] > thr
+    order = np.lexsort((keys[win], -sums[0][win]))
+    for c in (sidx, gkeys_s, spx, pk, cost, qty, val):
+        c.free()
+    stab.free()
+    return keys[win][order], sums[0][win][order]
+
+
+def q20(li_cols, part_cols, ps_cols, supp_cols, nation_names,
+        forest_code=29, stream=None):
+    """Device Q20 (tpch_ref.py:622-662): forest parts semi over ps; the
+    correlated half-of-1994-quantity subquery becomes a device group-by
+    over the composite (partkey, suppkey) key (qk_i64_combine); CANADA
+    + ordering resolve host-side over the small winner set. An EMPTY
+    subquery yields SQL NULL -> the comparison is false (pairs with no
+    1994 shipments do not qualify)."""
+    from . import jit, ops, shim
+    from .shim import c_i64
+    st = stream
+    sh = st.handle if st else None
+    # forest parts -> semi table
+    fidx, nf = ops.filter_col(part_cols["p_name1"], ops.EQ, forest_code,
+                              st)
+    fkeys = part_cols["p_partkey"].gather(fidx, nf, st)
+    ftab = ops.JoinTable(max(16, nf), st)
+    if nf:
+        ftab.build(fkeys)
+    # 1994 lineitem of forest parts -> per-(partkey,suppkey) qty sums
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_shipdate >= date '1994-01-01' and "
+        "l_shipdate < date '1994-01-01' + interval '1' year", lsch),
+        "q20_window", _schema_key(lsch))
+    lidx, ln = lf.run(li_cols, st)
+    lpk = li_cols["l_partkey"].gather(lidx, ln, st)
+    lsk = li_cols["l_suppkey"].gather(lidx, ln, st)
+    lqty = li_cols["l_quantity"].gather(lidx, ln, st)
+    fpx, _, nfm = ftab.probe(lpk, mode=1, n=ln)
+    m_pk = lpk.gather(fpx, nfm, st)
+    m_sk = lsk.gather(fpx, nfm, st)
+    m_q = lqty.gather(fpx, nfm, st)
+    S = int(supp_cols["s_suppkey"].n) + 1
+    ckey = DevColumn(np.int64, max(1, nfm))
+    call("qk_i64_combine", sh, c_u64(nfm), m_pk.ptr, m_sk.ptr,
+         c_i64(S), ckey.ptr)
+    ckey.n = nfm
+    gb = ops.GroupByI64(expected_groups=max(1024, nfm), nvals=1,
+                        stream=st)
+    gb.update(ckey, [m_q], nfm)
+    kk, qsums = gb.extract()
+    gb.free()
+    # eligible ps rows (forest parts) with their composite keys
+    ppx, _, npm = ftab.probe(ps_cols["ps_partkey"], mode=1)
+    e_pk = ps_cols["ps_partkey"].gather(ppx, npm, st)
+    e_sk = ps_cols["ps_suppkey"].gather(ppx, npm, st)
+    e_av = ps_cols["ps_availqty"].gather(ppx, npm, st)
+    eck = DevColumn(np.int64, max(1, npm))
+    call("qk_i64_combine", sh, c_u64(npm), e_pk.ptr, e_sk.ptr,
+         c_i64(S), eck.ptr)
+    if st:
+        st.sync()
+    eck_h = eck.to_numpy(npm)
+    av_h = e_av.to_numpy(npm)
+    sk_h = e_sk.to_numpy(npm)
+    order = np.argsort(kk)
+    pos = np.searchsorted(kk, eck_h, sorter=order)
+    have = (pos < len(kk)) & \
+        (kk[order[np.minimum(pos, len(kk) - 1)]] == eck_h)
+    thr = np.zeros(npm)
+    thr[have] = 0.5 * qsums[0][order[pos[have]]]
+    ok = have & (av_h > thr)
+    winners = np.unique(sk_h[ok])
+    canada = list(nation_names).index("CANADA")
+    snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
+    winners = winners[snk[winners - 1] == canada]
+    for c in (fidx, fkeys, lidx, lpk, lsk, lqty, fpx, m_pk, m_sk, m_q,
+              ckey, ppx, e_pk, e_sk, e_av, eck):
+        c.free()
+    ftab.free()
+    return winners.astype(np.int64)
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

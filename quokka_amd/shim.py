@@ -131,6 +131,7 @@ _SIGS = {
     "qk_str_dict_rehash": [c_vp, c_u32, c_vp, c_vp, c_vp, c_vp, c_vp,
                            c_u64],
     "qk_d2d": [c_vp, c_vp, c_u64],
+    "qk_i64_combine": [c_vp, c_u64, c_vp, c_vp, c_i64, c_vp],
     "qk_partition_scatter": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp],
 }
 for name, argtypes in _SIGS.items():

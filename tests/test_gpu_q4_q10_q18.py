@@ -217,3 +217,63 @@ def test_q19_device_vs_oracle(gpu, data):
     for cs in (lcols, pcols):
         for c in cs.values():
             c.free()
+
+
+def test_q2_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    part, supp = data["part"], data["supplier"]
+    ps, nat, reg = data["partsupp"], data["nation"], data["region"]
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_size",
+                                               "p_type"])
+    scols = staging.stage_columns(supp, names=["s_suppkey", "s_nationkey",
+                                               "s_acctbal"])
+    pscols = staging.stage_columns(ps, names=["ps_partkey", "ps_suppkey",
+                                              "ps_supplycost"])
+    got = DQ.q2(pcols, scols, pscols, nat["n_regionkey"], nat["n_name"])
+    want = OQ.q2(part, supp, ps, nat, reg)
+    assert np.array_equal(got["p_partkey"], want["p_partkey"])
+    assert np.array_equal(got["s_suppkey"], want["s_suppkey"])
+    np.testing.assert_allclose(got["s_acctbal"], want["s_acctbal"],
+                               rtol=0)
+    np.testing.assert_allclose(got["ps_supplycost"],
+                               want["ps_supplycost"], rtol=0)
+    assert list(got["n_name"]) == list(want["n_name"])
+    for cs in (pcols, scols, pscols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q11_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    ps, supp, nat = data["partsupp"], data["supplier"], data["nation"]
+    pscols = staging.stage_columns(ps, names=["ps_partkey", "ps_suppkey",
+                                              "ps_supplycost",
+                                              "ps_availqty"])
+    scols = staging.stage_columns(supp, names=["s_suppkey",
+                                               "s_nationkey"])
+    gk, gv = DQ.q11(pscols, scols, nat["n_name"])
+    wk, wv = OQ.q11(ps, supp, nat)
+    assert np.array_equal(gk, wk)
+    np.testing.assert_allclose(gv, wv, rtol=1e-9)
+    for cs in (pscols, scols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q20_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, part = data["lineitem"], data["part"]
+    ps, supp, nat = data["partsupp"], data["supplier"], data["nation"]
+    lcols = staging.stage_columns(li, names=["l_partkey", "l_suppkey",
+                                             "l_quantity", "l_shipdate"])
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_name1"])
+    pscols = staging.stage_columns(ps, names=["ps_partkey", "ps_suppkey",
+                                              "ps_availqty"])
+    scols = staging.stage_columns(supp, names=["s_suppkey",
+                                               "s_nationkey"])
+    got = DQ.q20(lcols, pcols, pscols, scols, nat["n_name"])
+    want = OQ.q20(li, part, ps, supp, nat)
+    assert np.array_equal(got, want)
+    for cs in (lcols, pcols, pscols, scols):
+        for c in cs.values():
+            c.free()

@@ -482,3 +482,93 @@ def test_q19_oracle_equals_acero(data):
                                          sel["l_discount"]))).as_py() or 0.0
     want = OQ.q19(li, part)
     np.testing.assert_allclose(got, want, rtol=1e-9)
+
+
+def test_q2_oracle_equals_acero(data):
+    part, supp = data["part"], data["supplier"]
+    ps, nat = data["partsupp"], data["nation"]
+    europe = G.REGIONS.index("EUROPE")
+    eu = pa.table({"nk": np.nonzero(nat["n_regionkey"] == europe)[0]
+                   .astype(np.int32)})
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"],
+                  "s_acctbal": supp["s_acctbal"]})
+    s_eu = s.join(eu, keys="sn", right_keys="nk", join_type="left semi")
+    p = pa.table({k: ps[k] for k in ("ps_partkey", "ps_suppkey",
+                                     "ps_supplycost")})
+    p = p.join(s_eu.select(["s_suppkey"]), keys="ps_suppkey",
+               right_keys="s_suppkey", join_type="left semi")
+    mins = p.group_by("ps_partkey").aggregate([("ps_supplycost", "min")])
+    pt = pa.table({"p_partkey": part["p_partkey"],
+                   "p_size": part["p_size"], "p_type": part["p_type"]})
+    pt = pt.filter(pc.and_(pc.equal(pt["p_size"], 15),
+                           pc.equal(pc.bit_wise_and(pa.chunked_array(
+                               [pa.array(np.asarray(part["p_type"]) % 5)]),
+                               255), 0)))
+    j = p.join(mins, keys="ps_partkey", right_keys="ps_partkey")
+    j = j.filter(pc.equal(j["ps_supplycost"], j["ps_supplycost_min"]))
+    j = j.join(pt.select(["p_partkey"]), keys="ps_partkey",
+               right_keys="p_partkey", join_type="left semi")
+    want = OQ.q2(part, supp, ps, nat, data["region"])
+    got = set(zip(j.column("ps_partkey").to_pylist(),
+                  j.column("ps_suppkey").to_pylist()))
+    # the oracle's top-100 rows must all be Acero winners, counts match
+    assert len(got) >= len(want["p_partkey"])
+    for a, b in zip(want["p_partkey"], want["s_suppkey"]):
+        assert (a, b) in got
+
+
+def test_q11_oracle_equals_acero(data):
+    ps, supp, nat = data["partsupp"], data["supplier"], data["nation"]
+    germany = list(nat["n_name"]).index("GERMANY")
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"]})
+    s = s.filter(pc.equal(s["sn"], germany))
+    p = pa.table({k: ps[k] for k in ("ps_partkey", "ps_suppkey",
+                                     "ps_supplycost", "ps_availqty")})
+    p = p.join(s.select(["s_suppkey"]), keys="ps_suppkey",
+               right_keys="s_suppkey", join_type="left semi")
+    val = pc.multiply(p["ps_supplycost"],
+                      pc.cast(p["ps_availqty"], pa.float64()))
+    p = p.append_column("val", val)
+    g = p.group_by("ps_partkey").aggregate([("val", "sum")])
+    thr = pc.sum(p["val"]).as_py() * 0.0001
+    g = g.filter(pc.greater(g["val_sum"], thr))
+    g = g.sort_by([("val_sum", "descending"), ("ps_partkey", "ascending")])
+    wk, wv = OQ.q11(ps, supp, nat)
+    assert np.array_equal(np.asarray(g.column("ps_partkey")), wk)
+    np.testing.assert_allclose(np.asarray(g.column("val_sum")), wv,
+                               rtol=1e-9)
+
+
+def test_q20_oracle_equals_acero(data):
+    li, part = data["lineitem"], data["part"]
+    ps, supp, nat = data["partsupp"], data["supplier"], data["nation"]
+    forest = pa.table({"pk": part["p_partkey"][
+        part["p_name1"] == G.P_NAME_FOREST]})
+    l = pa.table({k: li[k] for k in ("l_partkey", "l_suppkey",
+                                     "l_quantity", "l_shipdate")})
+    l = l.filter(pc.and_(pc.greater_equal(l["l_shipdate"], G.Q5_LO),
+                         pc.less(l["l_shipdate"], G.Q5_HI)))
+    l = l.join(forest, keys="l_partkey", right_keys="pk",
+               join_type="left semi")
+    g = l.group_by(["l_partkey", "l_suppkey"]).aggregate(
+        [("l_quantity", "sum")])
+    p = pa.table({k: ps[k] for k in ("ps_partkey", "ps_suppkey",
+                                     "ps_availqty")})
+    p = p.join(forest, keys="ps_partkey", right_keys="pk",
+               join_type="left semi")
+    j = p.join(g, keys=["ps_partkey", "ps_suppkey"],
+               right_keys=["l_partkey", "l_suppkey"], join_type="inner")
+    j = j.filter(pc.greater(pc.cast(j["ps_availqty"], pa.float64()),
+                            pc.multiply(pa.scalar(0.5),
+                                        j["l_quantity_sum"])))
+    canada = list(nat["n_name"]).index("CANADA")
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"]})
+    s = s.filter(pc.equal(s["sn"], canada))
+    winners = s.join(j.select(["ps_suppkey"]), keys="s_suppkey",
+                     right_keys="ps_suppkey", join_type="left semi")
+    got = sorted(winners.column("s_suppkey").to_pylist())
+    want = OQ.q20(li, part, ps, supp, nat)
+    assert got == list(want)
