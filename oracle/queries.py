@@ -525,6 +525,144 @@ def q20(li, part, partsupp, supplier, nation):
     return sk[sn == canada].astype(np.int64)
 
 
+def q9(li, orders, supplier, part, partsupp, nation):
+    """tpch_ref.py (do_9): profit = revenue - supplycost*qty for 'green'
+    parts, by (supplier nation, order year); order by nation asc, year
+    desc. Returns dict (n_name, year) -> profit."""
+    green = np.zeros(int(part["p_partkey"].max()) + 2, dtype=bool)
+    green[part["p_partkey"][part["p_name_green"] == 1]] = True
+    S = int(max(li["l_suppkey"].max(), partsupp["ps_suppkey"].max())) + 1
+    pskey = partsupp["ps_partkey"] * S + partsupp["ps_suppkey"]
+    order_ps = np.argsort(pskey)
+    m = green[li["l_partkey"]]
+    lkey = li["l_partkey"][m] * S + li["l_suppkey"][m]
+    pos = np.searchsorted(pskey, lkey, sorter=order_ps)
+    pos = np.minimum(pos, len(pskey) - 1)
+    have = pskey[order_ps[pos]] == lkey
+    # inner-join semantics: only (part, supp) pairs present in partsupp
+    # (the generator maintains the spec invariant, so this keeps all)
+    idx = np.nonzero(m)[0][have]
+    m = np.zeros(len(li["l_partkey"]), dtype=bool)
+    m[idx] = True
+    cost = partsupp["ps_supplycost"][order_ps[pos][have]]
+    amount = (li["l_extendedprice"][m] * (1.0 - li["l_discount"][m])
+              - cost * li["l_quantity"][m])
+    nkey = int(orders["o_orderkey"].max()) + 2
+    oyear = np.zeros(nkey, dtype=np.int16)
+    od = orders["o_orderdate"]
+    oyear[orders["o_orderkey"]] = 1970 + (
+        np.asarray(od, dtype="datetime64[D]").astype(
+            "datetime64[Y]").astype(np.int64)).astype(np.int16)
+    yr = oyear[li["l_orderkey"][m]]
+    sn = supplier["s_nationkey"][li["l_suppkey"][m] - 1]
+    names = list(nation["n_name"])
+    out = {}
+    for nk in range(25):
+        for y in np.unique(yr):
+            v = amount[(sn == nk) & (yr == y)].sum()
+            if v != 0.0:
+                out[(names[nk], int(y))] = float(v)
+    return out
+
+
+def q13(orders, customer):
+    """tpch_ref.py:409-432: distribution of per-customer order counts,
+    orders with the 'special requests' comment excluded (text-boundary
+    flag), customers with zero orders included. Returns dict
+    c_count -> custdist."""
+    keep = orders["o_comment_special"] == 0
+    ncust = len(customer["c_custkey"])
+    cnt = np.bincount(orders["o_custkey"][keep], minlength=ncust + 2)
+    per_cust = cnt[1:ncust + 1]
+    counts = np.bincount(per_cust)
+    return {int(c): int(v) for c, v in enumerate(counts) if v}
+
+
+def q16(part, partsupp, supplier):
+    """tpch_ref.py:487-520: DISTINCT supplier count per (brand, type,
+    size) over filtered parts, complained suppliers excluded. Returns
+    dict (brand, type, size) -> count, ordered by count desc then key."""
+    med_pol = (part["p_type"] // 25 == G.PTYPE_SYL1.index("MEDIUM")) & \
+        ((part["p_type"] // 5) % 5 == G.PTYPE_SYL2.index("POLISHED"))
+    sel = ((part["p_brand"] != G.brand_code("Brand#45"))
+           & ~med_pol
+           & np.isin(part["p_size"], [49, 14, 23, 45, 19, 3, 36, 9]))
+    ok = np.zeros(int(part["p_partkey"].max()) + 2, dtype=bool)
+    ok[part["p_partkey"][sel]] = True
+    attr = {}
+    for k, b, t, z in zip(part["p_partkey"], part["p_brand"],
+                          part["p_type"], part["p_size"]):
+        attr[int(k)] = (int(b), int(t), int(z))
+    bad_supp = np.zeros(int(supplier["s_suppkey"].max()) + 2, dtype=bool)
+    bad_supp[supplier["s_suppkey"][
+        supplier["s_comment_complaints"] == 1]] = True
+    m = ok[partsupp["ps_partkey"]] & ~bad_supp[partsupp["ps_suppkey"]]
+    seen = set()
+    for pk, sk in zip(partsupp["ps_partkey"][m],
+                      partsupp["ps_suppkey"][m]):
+        seen.add(attr[int(pk)] + (int(sk),))
+    out = {}
+    for b, t, z, sk in seen:
+        out[(b, t, z)] = out.get((b, t, z), 0) + 1
+    return dict(sorted(out.items(), key=lambda kv: (-kv[1], kv[0])))
+
+
+def q21(li, orders, supplier, nation, limit=100):
+    """tpch_ref.py (do_21): suppliers (SAUDI ARABIA) that were the ONLY
+    late supplier on a multi-supplier 'F' order. Returns dict
+    s_suppkey -> numwait, ordered by numwait desc then suppkey."""
+    nkey = int(orders["o_orderkey"].max()) + 2
+    status_f = np.zeros(nkey, dtype=bool)
+    status_f[orders["o_orderkey"][orders["o_orderstatus"] == 0]] = True
+    S = int(li["l_suppkey"].max()) + 1
+    # distinct suppliers per order (all lines)
+    all_pairs = np.unique(li["l_orderkey"] * S + li["l_suppkey"])
+    nsupp_all = np.bincount((all_pairs // S).astype(np.int64),
+                            minlength=nkey)
+    # distinct suppliers per order among LATE lines
+    late = li["l_receiptdate"] > li["l_commitdate"]
+    late_pairs = np.unique(li["l_orderkey"][late] * S
+                           + li["l_suppkey"][late])
+    lo = (late_pairs // S).astype(np.int64)
+    nsupp_late = np.bincount(lo, minlength=nkey)
+    # qualifying: F order, >=2 suppliers total, exactly 1 late supplier
+    qual_orders = status_f & (nsupp_all >= 2) & (nsupp_late == 1)
+    qual = qual_orders[lo]
+    wait_supp = (late_pairs % S)[qual]
+    saudi = list(nation["n_name"]).index("SAUDI ARABIA")
+    sn = supplier["s_nationkey"][wait_supp - 1]
+    wait_supp = wait_supp[sn == saudi]
+    cnt = np.bincount(wait_supp, minlength=S + 1)
+    sk = np.nonzero(cnt)[0]
+    order = np.lexsort((sk, -cnt[sk]))
+    top = order[:limit]
+    return {int(sk[i]): int(cnt[sk[i]]) for i in top}
+
+
+def q22(customer, orders):
+    """tpch_ref.py (do_22): country code = first two phone digits
+    (= 10 + c_nationkey in the generator, spec 4.2.2.9); codes
+    {13,31,23,29,30,18,17}, acctbal above the positive average of those
+    codes, and NO orders. Returns dict cntrycode -> (numcust, total)."""
+    codes = np.array([13, 31, 23, 29, 30, 18, 17])
+    cc = 10 + customer["c_nationkey"]
+    in_list = np.isin(cc, codes)
+    pos = in_list & (customer["c_acctbal"] > 0.0)
+    avg = customer["c_acctbal"][pos].mean()
+    ncust = int(customer["c_custkey"].max()) + 2
+    has_order = np.zeros(ncust, dtype=bool)
+    has_order[orders["o_custkey"]] = True
+    m = in_list & (customer["c_acctbal"] > avg) & \
+        ~has_order[customer["c_custkey"]]
+    out = {}
+    for code in sorted(codes):
+        sel = m & (cc == code)
+        if sel.any():
+            out[str(code)] = (int(sel.sum()),
+                              float(customer["c_acctbal"][sel].sum()))
+    return out
+
+
 def q5(li, orders, customer, supplier, nation, region):
     """tpch_ref.py:142-169: 6-table chain, r_name='ASIA',
     o_orderdate in [1994-01-01, 1995-01-01), extra equi-predicate

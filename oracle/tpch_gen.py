@@ -155,6 +155,15 @@ def gen_orders(sf, seed=42):
         # (zeros until then).
         "o_orderpriority": rng.integers(0, 5, n).astype(np.uint8),
         "o_totalprice": np.zeros(n, dtype=np.float64),
+        # text-boundary adaptation (header): the Q13 predicate
+        # o_comment NOT LIKE '%special%requests%' becomes this flag
+        # (dbgen plants the phrase in a small fraction of comments);
+        # drawn after the columns above (stream-append stable).
+        # o_orderstatus is DERIVED from the order's line statuses
+        # (spec 4.2.3: F if all lines F, O if all O, else P) —
+        # gen_lineitem fills it in (2 = P until then).
+        "o_comment_special": (rng.random(n) < 0.019).astype(np.uint8),
+        "o_orderstatus": np.full(n, 2, dtype=np.uint8),
     }
 
 
@@ -188,12 +197,25 @@ def gen_lineitem(sf, seed=42, orders=None):
     # spec 4.2.3: O_TOTALPRICE = sum over the order's lines of
     # extendedprice * (1 + tax) * (1 - discount)
     line_total = extendedprice * (1.0 + tax) * (1.0 - discount)
-    orders["o_totalprice"] = np.add.reduceat(
-        line_total, np.concatenate(([0], np.cumsum(lines_per_order)[:-1])))
+    starts = np.concatenate(([0], np.cumsum(lines_per_order)[:-1]))
+    orders["o_totalprice"] = np.add.reduceat(line_total, starts)
+    # spec 4.2.3: O_ORDERSTATUS = F if every line F, O if every line O,
+    # else P (codes 0/1/2 into ["F", "O", "P"])
+    f_cnt = np.add.reduceat((linestatus == 0).astype(np.int64), starts)
+    status = np.full(len(starts), 2, dtype=np.uint8)
+    status[f_cnt == lines_per_order] = 0
+    status[f_cnt == 0] = 1
+    orders["o_orderstatus"] = status
 
+    # spec 4.2.3: L_SUPPKEY is one of the part's FOUR partsupp suppliers
+    # (the same spread formula as gen_partsupp), picked uniformly — the
+    # invariant Q9's ps join relies on
+    S = n_suppliers(sf)
+    i4 = rng.integers(0, 4, n)
+    l_suppkey = ((partkey + i4 * (S // 4 + (partkey - 1) // S)) % S) + 1
     out = {
         "l_orderkey": l_orderkey,
-        "l_suppkey": rng.integers(1, n_suppliers(sf) + 1, n, dtype=np.int64),
+        "l_suppkey": l_suppkey,
         "l_quantity": quantity,
         "l_extendedprice": extendedprice,
         "l_discount": discount,
@@ -251,8 +273,11 @@ def gen_supplier(sf, seed=42):
         "s_suppkey": np.arange(1, n + 1, dtype=np.int64),
         "s_nationkey": rng.integers(0, 25, n).astype(np.int32),
     }
-    # appended draws (stream-stable): spec s_acctbal U[-999.99, 9999.99]
+    # appended draws (stream-stable): spec s_acctbal U[-999.99, 9999.99];
+    # s_comment LIKE '%Customer%Complaints%' as a flag (text boundary,
+    # header; dbgen plants it in 10/SF/10000 suppliers)
     out["s_acctbal"] = rng.integers(-99999, 1000000, n) / 100.0
+    out["s_comment_complaints"] = (rng.random(n) < 0.0013).astype(np.uint8)
     return out
 
 
@@ -278,6 +303,9 @@ def gen_part(sf, seed=42):
     # LIKE 'forest%' == first word == "forest" == code P_NAME_FOREST —
     # the text-boundary adaptation documented in the header)
     out["p_name1"] = rng.integers(0, 92, n).astype(np.uint8)
+    # p_name LIKE '%green%' as a flag: 5 words drawn from 92, so
+    # P(contains a given word) = 1 - C(91,5)/C(92,5) = 5/92
+    out["p_name_green"] = (rng.random(n) < 5 / 92).astype(np.uint8)
     return out
 
 

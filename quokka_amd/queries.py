@@ -1433,6 +1433,316 @@ def q20(li_cols, part_cols, ps_cols, supp_cols, nation_names,
     return winners.astype(np.int64)
 
 
+def q9(li_cols, ord_cols, supp_cols, part_cols, ps_cols, nation_names,
+       stream=None):
+    """Device Q9: green-part lines pick up ps_supplycost through a
+    composite-key (partkey*S + suppkey) device join, order year through
+    the orders join, supplier nation through the supplier join; profit
+    sums run as one JIT grouped aggregate per year over the 25-nation
+    key. Returns dict (n_name, year) -> profit."""
+    from . import jit, ops
+    from .shim import c_i64
+    st = stream
+    sh = st.handle if st else None
+    S = int(supp_cols["s_suppkey"].n) + 1
+    # green parts semi over lineitem
+    gidx, ng = ops.filter_col(part_cols["p_name_green"], ops.EQ, 1, st)
+    gkeys = part_cols["p_partkey"].gather(gidx, ng, st)
+    gtab = ops.JoinTable(max(16, ng), st)
+    if ng:
+        gtab.build(gkeys)
+    lpx, _, nl = gtab.probe(li_cols["l_partkey"], mode=1)
+    g = {c: li_cols[c].gather(lpx, nl, st)
+         for c in ("l_partkey", "l_suppkey", "l_orderkey", "l_quantity",
+                   "l_extendedprice", "l_discount")}
+    # ps composite join for supplycost
+    nps = ps_cols["ps_partkey"].n
+    psk = DevColumn(np.int64, max(1, nps))
+    call("qk_i64_combine", sh, c_u64(nps), ps_cols["ps_partkey"].ptr,
+         ps_cols["ps_suppkey"].ptr, c_i64(S), psk.ptr)
+    pst = ops.JoinTable(max(16, nps), st)
+    psk.n = nps
+    pst.build(psk)
+    lck = DevColumn(np.int64, max(1, nl))
+    call("qk_i64_combine", sh, c_u64(nl), g["l_partkey"].ptr,
+         g["l_suppkey"].ptr, c_i64(S), lck.ptr)
+    lck.n = nl
+    cpx, cbx, nc = pst.probe(lck, mode=0, n=nl)
+    cost = ps_cols["ps_supplycost"].gather(cbx, nc, st)
+    a1 = {c: g[c].gather(cpx, nc, st)
+          for c in ("l_orderkey", "l_suppkey", "l_quantity",
+                    "l_extendedprice", "l_discount")}
+    # orders join for the year
+    otab = ops.JoinTable(max(16, ord_cols["o_orderkey"].n), st)
+    otab.build(ord_cols["o_orderkey"])
+    opx, obx, no = otab.probe(a1["l_orderkey"], mode=0, n=nc)
+    odate = ord_cols["o_orderdate"].gather(obx, no, st)
+    a2 = {c: a1[c].gather(opx, no, st)
+          for c in ("l_suppkey", "l_quantity", "l_extendedprice",
+                    "l_discount")}
+    a2["cost"] = cost.gather(opx, no, st)
+    # supplier join for the nation
+    stab = ops.JoinTable(max(16, supp_cols["s_suppkey"].n), st)
+    stab.build(supp_cols["s_suppkey"])
+    spx, sbx, ns = stab.probe(a2["l_suppkey"], mode=0, n=no)
+    fin = {"sn": supp_cols["s_nationkey"].gather(sbx, ns, st),
+           "o_orderdate": odate.gather(spx, ns, st)}
+    for c in ("l_quantity", "l_extendedprice", "l_discount", "cost"):
+        fin[c] = a2[c].gather(spx, ns, st)
+    fsch = {k: v.dtype for k, v in fin.items()}
+    names = list(nation_names)
+    out = {}
+    for y in range(1992, 1999):
+        agg = _cached_jit("a", lambda yy=y: jit.JitAggregate(
+            fsch, [("sn", 25)],
+            ["SUM(l_extendedprice * (1 - l_discount) - cost * "
+             "l_quantity) as profit"],
+            predicate="o_orderdate >= date '%d-01-01' and o_orderdate "
+                      "< date '%d-01-01'" % (yy, yy + 1)),
+            "q9_prof_%d" % y, _schema_key(fsch))
+        acc = agg.make_acc()
+        if ns:
+            agg.run(fin, acc, st)
+        if st:
+            st.sync()
+        prof = agg.read(acc)[:, 0]
+        for nk in range(25):
+            if prof[nk] != 0.0:
+                out[(names[nk], y)] = float(prof[nk])
+        acc.free()
+    for c in ([gidx, gkeys, lpx, psk, lck, cpx, cbx, cost, opx, obx,
+               odate, spx, sbx] + list(g.values()) + list(a1.values()) +
+              list(a2.values()) + list(fin.values())):
+        c.free()
+    gtab.free()
+    pst.free()
+    otab.free()
+    stab.free()
+    return out
+
+
+def q13(ord_cols, n_customers, stream=None):
+    """Device Q13: orders without the 'special requests' comment flag
+    count per customer on the device group-by; the count-of-counts
+    histogram (including zero-order customers) finishes host-side over
+    the extracted per-customer totals. Returns dict c_count -> custdist."""
+    from . import ops
+    st = stream
+    kidx, nk = ops.filter_col(ord_cols["o_comment_special"], ops.EQ, 0,
+                              st)
+    ck = ord_cols["o_custkey"].gather(kidx, nk, st)
+    ones = DevColumn(np.float64, max(1, nk))
+    call("qk_fill_f64", st.handle if st else None, ones.ptr,
+         ctypes.c_double(1.0), c_u64(nk))
+    gb = ops.GroupByI64(expected_groups=max(1024, n_customers), nvals=1,
+                        stream=st)
+    gb.update(ck, [ones], nk)
+    keys, sums = gb.extract()
+    gb.free()
+    per = sums[0].astype(np.int64)
+    counts = np.bincount(per, minlength=1)
+    counts[0] += n_customers - len(keys)       # customers with no orders
+    for c in (kidx, ck, ones):
+        c.free()
+    return {int(c): int(v) for c, v in enumerate(counts) if v}
+
+
+def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
+    """Device Q16: the part filter (brand<>45, NOT 'MEDIUM POLISHED%'
+    == type-code range 70..74, size IN list) compiles to one JIT pass;
+    complained suppliers drop via an ANTI probe; DISTINCT
+    (part, supplier) pairs dedupe on the device group-by over the
+    composite key; the (brand,type,size) attachment + ordering finish
+    host-side over the deduped pairs. part_host: dict with p_partkey/
+    p_brand/p_type/p_size host arrays (output attributes). Returns dict
+    (brand, type, size) -> distinct supplier count."""
+    from . import jit, ops
+    from .shim import c_i64
+    st = stream
+    sh = st.handle if st else None
+    psch = {k: part_cols[k].dtype for k in ("p_brand", "p_type",
+                                            "p_size")}
+    pf = _cached_jit("f", lambda: jit.JitFilter(
+        "p_brand != 19 and not (p_type >= 70 and p_type <= 74) and "
+        "p_size in (49, 14, 23, 45, 19, 3, 36, 9)", psch),
+        "q16_parts", _schema_key(psch))
+    pidx, npq = pf.run({k: part_cols[k] for k in psch}, st)
+    qkeys = part_cols["p_partkey"].gather(pidx, npq, st)
+    qtab = ops.JoinTable(max(16, npq), st)
+    if npq:
+        qtab.build(qkeys)
+    ppx, _, nps = qtab.probe(ps_cols["ps_partkey"], mode=1)
+    e_pk = ps_cols["ps_partkey"].gather(ppx, nps, st)
+    e_sk = ps_cols["ps_suppkey"].gather(ppx, nps, st)
+    # complained suppliers -> ANTI
+    bidx, nb = ops.filter_col(supp_cols["s_comment_complaints"], ops.EQ,
+                              1, st)
+    if nb:
+        bkeys = supp_cols["s_suppkey"].gather(bidx, nb, st)
+        btab = ops.JoinTable(max(16, nb), st)
+        btab.build(bkeys)
+        apx, _, na = btab.probe(e_sk, mode=2, n=nps)
+        f_pk = e_pk.gather(apx, na, st)
+        f_sk = e_sk.gather(apx, na, st)
+        btab.free()
+        bkeys.free()
+        apx.free()
+    else:
+        f_pk, f_sk, na = e_pk, e_sk, nps
+    S = int(supp_cols["s_suppkey"].n) + 1
+    ckey = DevColumn(np.int64, max(1, na))
+    call("qk_i64_combine", sh, c_u64(na), f_pk.ptr, f_sk.ptr, c_i64(S),
+         ckey.ptr)
+    ckey.n = na
+    ones = DevColumn(np.float64, max(1, na))
+    call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0), c_u64(na))
+    gb = ops.GroupByI64(expected_groups=max(1024, na), nvals=1, stream=st)
+    gb.update(ckey, [ones], na)
+    keys, _ = gb.extract()
+    gb.free()
+    pk = (keys // S).astype(np.int64)
+    attr = {}
+    for k, b, t, z in zip(part_host["p_partkey"], part_host["p_brand"],
+                          part_host["p_type"], part_host["p_size"]):
+        attr[int(k)] = (int(b), int(t), int(z))
+    out = {}
+    for p in pk:
+        key = attr[int(p)]
+        out[key] = out.get(key, 0) + 1
+    for c in [pidx, qkeys, ppx, e_pk, e_sk, bidx, ckey, ones] + \
+            ([f_pk, f_sk] if nb else []):
+        c.free()
+    qtab.free()
+    return dict(sorted(out.items(), key=lambda kv: (-kv[1], kv[0])))
+
+
+def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
+        stream=None):
+    """Device Q21: the exists/not-exists pair reduces to per-order
+    DISTINCT supplier counts — deduped on the device group-by over
+    (orderkey*S + suppkey) for all lines and for LATE lines — with the
+    qualification (F status, >=2 suppliers, exactly 1 late) and the
+    SAUDI ARABIA numwait tally finishing host-side over the deduped
+    pairs. Returns dict s_suppkey -> numwait."""
+    from . import jit, ops
+    from .shim import c_i64
+    st = stream
+    sh = st.handle if st else None
+    S = int(supp_cols["s_suppkey"].n) + 1
+    n = li_cols["l_orderkey"].n
+
+    def distinct_pairs(idx=None, nn=None):
+        if idx is None:
+            kx, ky, cnt = li_cols["l_orderkey"], li_cols["l_suppkey"], n
+        else:
+            kx = li_cols["l_orderkey"].gather(idx, nn, st)
+            ky = li_cols["l_suppkey"].gather(idx, nn, st)
+            cnt = nn
+        ck = DevColumn(np.int64, max(1, cnt))
+        call("qk_i64_combine", sh, c_u64(cnt), kx.ptr, ky.ptr, c_i64(S),
+             ck.ptr)
+        ck.n = cnt
+        ones = DevColumn(np.float64, max(1, cnt))
+        call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0),
+             c_u64(cnt))
+        gb = ops.GroupByI64(expected_groups=max(1024, cnt), nvals=1,
+                            stream=st)
+        gb.update(ck, [ones], cnt)
+        keys, _ = gb.extract()
+        gb.free()
+        ck.free()
+        ones.free()
+        if idx is not None:
+            kx.free()
+            ky.free()
+        return keys
+
+    all_pairs = distinct_pairs()
+    lsch = {k: v.dtype for k, v in li_cols.items()}
+    lf = _cached_jit("f", lambda: jit.JitFilter(
+        "l_receiptdate > l_commitdate", lsch),
+        "q21_late", _schema_key(lsch))
+    lidx, nl = lf.run(li_cols, st)
+    late_pairs = distinct_pairs(lidx, nl)
+    lidx.free()
+    if st:
+        st.sync()
+    # host tail over the deduped pairs (O(pairs), not O(rows))
+    ok_h = ord_cols["o_orderkey"].to_numpy(ord_cols["o_orderkey"].n)
+    os_h = ord_cols["o_orderstatus"].to_numpy(
+        ord_cols["o_orderstatus"].n)
+    nkey = int(ok_h.max()) + 2
+    status_f = np.zeros(nkey, dtype=bool)
+    status_f[ok_h[os_h == 0]] = True
+    nsupp_all = np.bincount((all_pairs // S), minlength=nkey)
+    lo = (late_pairs // S).astype(np.int64)
+    nsupp_late = np.bincount(lo, minlength=nkey)
+    qual_orders = status_f & (nsupp_all >= 2) & (nsupp_late == 1)
+    wait_supp = (late_pairs % S)[qual_orders[lo]]
+    saudi = list(nation_names).index("SAUDI ARABIA")
+    snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
+    wait_supp = wait_supp[snk[wait_supp - 1] == saudi]
+    cnt = np.bincount(wait_supp, minlength=S + 1)
+    sk = np.nonzero(cnt)[0]
+    order = np.lexsort((sk, -cnt[sk]))[:limit]
+    return {int(sk[i]): int(cnt[sk[i]]) for i in order}
+
+
+def q22(cust_cols, ord_cols, stream=None):
+    """Device Q22: the country-code membership + positive-balance
+    average run as one JIT grand aggregate; customers with no orders
+    drop out via an ANTI probe against the orders custkeys; the final
+    per-code count/sum is a JIT grouped aggregate over c_nationkey
+    (cntrycode == 10 + nationkey in the generator's spec 4.2.2.9 phone
+    format). Returns dict cntrycode(str) -> (numcust, totacctbal)."""
+    from . import jit, ops
+    st = stream
+    codes = [13, 31, 23, 29, 30, 18, 17]
+    nats = [c - 10 for c in codes]
+    in_list = " or ".join("c_nationkey = %d" % k for k in nats)
+    csch = {"c_nationkey": np.dtype(np.int32),
+            "c_acctbal": np.dtype(np.float64)}
+    avg_agg = _cached_jit("a", lambda: jit.JitAggregate(
+        csch, [], ["SUM(c_acctbal) as s", "COUNT(*) as n"],
+        predicate="c_acctbal > 0 and (%s)" % in_list),
+        "q22_avg", _schema_key(csch))
+    acc = avg_agg.make_acc()
+    avg_agg.run({k: cust_cols[k] for k in csch}, acc, st)
+    if st:
+        st.sync()
+    s, npos = avg_agg.read(acc)[0]
+    acc.free()
+    avg = s / npos if npos else 0.0
+    # customers with NO orders: ANTI against the orders custkeys
+    otab = ops.JoinTable(max(16, ord_cols["o_custkey"].n), st)
+    otab.build(ord_cols["o_custkey"])
+    apx, _, na = otab.probe(cust_cols["c_custkey"], mode=2)
+    fin = {"c_nationkey": cust_cols["c_nationkey"].gather(apx, na, st),
+           "c_acctbal": cust_cols["c_acctbal"].gather(apx, na, st)}
+    fagg = _cached_jit("a", lambda: jit.JitAggregate(
+        {"c_nationkey": np.dtype(np.int32),
+         "c_acctbal": np.dtype(np.float64)}, [("c_nationkey", 25)],
+        ["COUNT(*) as n", "SUM(c_acctbal) as s"],
+        predicate="c_acctbal > %r and (%s)" % (avg, in_list)),
+        "q22_final_%r" % avg)
+    acc = fagg.make_acc()
+    if na:
+        fagg.run(fin, acc, st)
+    if st:
+        st.sync()
+    res = fagg.read(acc)
+    out = {}
+    for c in sorted(codes):
+        nk = c - 10
+        if res[nk, 0] > 0:
+            out[str(c)] = (int(res[nk, 0]), float(res[nk, 1]))
+    for c in [apx] + list(fin.values()):
+        c.free()
+    acc.free()
+    otab.free()
+    return out
+
+
 def _topk(full, limit):
     """Indices of the top-`limit` rows by (revenue desc, o_orderdate asc,
     l_orderkey asc). O(n) candidate selection, then an exact sort over the

@@ -277,3 +277,93 @@ def test_q20_device_vs_oracle(gpu, data):
     for cs in (lcols, pcols, pscols, scols):
         for c in cs.values():
             c.free()
+
+
+def test_q9_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, od = data["lineitem"], data["orders"]
+    su, part, ps, nat = (data["supplier"], data["part"],
+                         data["partsupp"], data["nation"])
+    lcols = staging.stage_columns(li, names=["l_partkey", "l_suppkey",
+                                             "l_orderkey", "l_quantity",
+                                             "l_extendedprice",
+                                             "l_discount"])
+    ocols = staging.stage_columns(od, names=["o_orderkey",
+                                             "o_orderdate"])
+    scols = staging.stage_columns(su, names=["s_suppkey",
+                                             "s_nationkey"])
+    pcols = staging.stage_columns(part, names=["p_partkey",
+                                               "p_name_green"])
+    pscols = staging.stage_columns(ps, names=["ps_partkey", "ps_suppkey",
+                                              "ps_supplycost"])
+    got = DQ.q9(lcols, ocols, scols, pcols, pscols, nat["n_name"])
+    want = OQ.q9(li, od, su, part, ps, nat)
+    assert set(got) == set(want)
+    for k in want:
+        np.testing.assert_allclose(got[k], want[k], rtol=1e-9,
+                                   err_msg=str(k))
+    for cs in (lcols, ocols, scols, pcols, pscols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q13_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    od, cu = data["orders"], data["customer"]
+    ocols = staging.stage_columns(od, names=["o_custkey",
+                                             "o_comment_special"])
+    got = DQ.q13(ocols, len(cu["c_custkey"]))
+    assert got == OQ.q13(od, cu)
+    for c in ocols.values():
+        c.free()
+
+
+def test_q16_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    part, ps, su = data["part"], data["partsupp"], data["supplier"]
+    pcols = staging.stage_columns(part, names=["p_partkey", "p_brand",
+                                               "p_type", "p_size"])
+    pscols = staging.stage_columns(ps, names=["ps_partkey",
+                                              "ps_suppkey"])
+    scols = staging.stage_columns(su, names=["s_suppkey",
+                                             "s_comment_complaints"])
+    got = DQ.q16(pcols, pscols, scols, part)
+    assert got == dict(OQ.q16(part, ps, su))
+    for cs in (pcols, pscols, scols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q21_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    li, od = data["lineitem"], data["orders"]
+    su, nat = data["supplier"], data["nation"]
+    lcols = staging.stage_columns(li, names=["l_orderkey", "l_suppkey",
+                                             "l_receiptdate",
+                                             "l_commitdate"])
+    ocols = staging.stage_columns(od, names=["o_orderkey",
+                                             "o_orderstatus"])
+    scols = staging.stage_columns(su, names=["s_suppkey",
+                                             "s_nationkey"])
+    got = DQ.q21(lcols, ocols, scols, nat["n_name"])
+    assert got == OQ.q21(li, od, su, nat)
+    for cs in (lcols, ocols, scols):
+        for c in cs.values():
+            c.free()
+
+
+def test_q22_device_vs_oracle(gpu, data):
+    from quokka_amd import staging, queries as DQ
+    cu, od = data["customer"], data["orders"]
+    ccols = staging.stage_columns(cu, names=["c_custkey", "c_nationkey",
+                                             "c_acctbal"])
+    ocols = staging.stage_columns(od, names=["o_custkey"])
+    got = DQ.q22(ccols, ocols)
+    want = OQ.q22(cu, od)
+    assert set(got) == set(want)
+    for k in want:
+        assert got[k][0] == want[k][0], k
+        np.testing.assert_allclose(got[k][1], want[k][1], rtol=1e-9)
+    for cs in (ccols, ocols):
+        for c in cs.values():
+            c.free()

@@ -572,3 +572,143 @@ def test_q20_oracle_equals_acero(data):
     got = sorted(winners.column("s_suppkey").to_pylist())
     want = OQ.q20(li, part, ps, supp, nat)
     assert got == list(want)
+
+
+def test_q9_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    supp, part, ps, nat = (data["supplier"], data["part"],
+                           data["partsupp"], data["nation"])
+    g = pa.table({"pk": part["p_partkey"][part["p_name_green"] == 1]})
+    l = pa.table({k: li[k] for k in ("l_partkey", "l_suppkey",
+                                     "l_orderkey", "l_quantity",
+                                     "l_extendedprice", "l_discount")})
+    l = l.join(g, keys="l_partkey", right_keys="pk", join_type="left semi")
+    p = pa.table({k: ps[k] for k in ("ps_partkey", "ps_suppkey",
+                                     "ps_supplycost")})
+    j = l.join(p, keys=["l_partkey", "l_suppkey"],
+               right_keys=["ps_partkey", "ps_suppkey"],
+               join_type="inner")
+    o = pa.table({"o_orderkey": orders["o_orderkey"],
+                  "o_orderdate": orders["o_orderdate"]})
+    j = j.join(o, keys="l_orderkey", right_keys="o_orderkey")
+    s = pa.table({"s_suppkey": supp["s_suppkey"],
+                  "sn": supp["s_nationkey"]})
+    j = j.join(s, keys="l_suppkey", right_keys="s_suppkey")
+    amount = pc.subtract(
+        pc.multiply(j["l_extendedprice"],
+                    pc.subtract(pa.scalar(1.0), j["l_discount"])),
+        pc.multiply(j["ps_supplycost"], j["l_quantity"]))
+    yr = pc.year(pc.cast(pc.multiply(pc.cast(j["o_orderdate"],
+                                             pa.int64()),
+                                     86400000), pa.timestamp("ms")))
+    j = j.append_column("amount", amount).append_column("yr", yr)
+    gr = j.group_by(["sn", "yr"]).aggregate([("amount", "sum")])
+    names = list(nat["n_name"])
+    got = {(names[s_], int(y)): v for s_, y, v in
+           zip(gr.column("sn").to_pylist(), gr.column("yr").to_pylist(),
+               gr.column("amount_sum").to_pylist())}
+    want = OQ.q9(li, orders, supp, part, ps, nat)
+    assert set(got) >= set(want)
+    for k, v in want.items():
+        np.testing.assert_allclose(got[k], v, rtol=1e-9, err_msg=str(k))
+
+
+def test_q13_oracle_equals_acero(data):
+    orders, cust = data["orders"], data["customer"]
+    o = pa.table({"o_custkey": orders["o_custkey"],
+                  "flag": orders["o_comment_special"]})
+    o = o.filter(pc.equal(o["flag"], 0))
+    g = o.group_by("o_custkey").aggregate([("flag", "count")])
+    per = np.zeros(len(cust["c_custkey"]) + 2, dtype=np.int64)
+    per[np.asarray(g.column("o_custkey"))] = \
+        np.asarray(g.column("flag_count"))
+    counts = np.bincount(per[1:len(cust["c_custkey"]) + 1])
+    got = {int(c): int(v) for c, v in enumerate(counts) if v}
+    assert got == OQ.q13(orders, cust)
+
+
+def test_q16_oracle_equals_acero(data):
+    part, ps, supp = data["part"], data["partsupp"], data["supplier"]
+    med_pol = ((np.asarray(part["p_type"]) // 25 ==
+                G.PTYPE_SYL1.index("MEDIUM")) &
+               ((np.asarray(part["p_type"]) // 5) % 5 ==
+                G.PTYPE_SYL2.index("POLISHED")))
+    sel = ((part["p_brand"] != G.brand_code("Brand#45")) & ~med_pol &
+           np.isin(part["p_size"], [49, 14, 23, 45, 19, 3, 36, 9]))
+    p = pa.table({"p_partkey": part["p_partkey"][sel],
+                  "b": part["p_brand"][sel], "t": part["p_type"][sel],
+                  "z": part["p_size"][sel]})
+    bad = pa.table({"sk": supp["s_suppkey"][
+        supp["s_comment_complaints"] == 1]})
+    j = pa.table({k: ps[k] for k in ("ps_partkey", "ps_suppkey")})
+    j = j.join(bad, keys="ps_suppkey", right_keys="sk",
+               join_type="left anti")
+    j = j.join(p, keys="ps_partkey", right_keys="p_partkey",
+               join_type="inner")
+    dd = j.group_by(["b", "t", "z", "ps_suppkey"]).aggregate([])
+    g = dd.group_by(["b", "t", "z"]).aggregate([("ps_suppkey", "count")])
+    got = {(int(b), int(t), int(z)): int(c) for b, t, z, c in
+           zip(g.column("b").to_pylist(), g.column("t").to_pylist(),
+               g.column("z").to_pylist(),
+               g.column("ps_suppkey_count").to_pylist())}
+    assert got == dict(OQ.q16(part, ps, supp))
+
+
+def test_q21_oracle_equals_acero(data):
+    li, orders = data["lineitem"], data["orders"]
+    supp, nat = data["supplier"], data["nation"]
+    l = pa.table({k: li[k] for k in ("l_orderkey", "l_suppkey",
+                                     "l_receiptdate", "l_commitdate")})
+    dd = l.group_by(["l_orderkey", "l_suppkey"]).aggregate([])
+    nall = dd.group_by("l_orderkey").aggregate([("l_suppkey", "count")])
+    nall = nall.rename_columns(["l_orderkey", "n_all"])
+    late = l.filter(pc.greater(l["l_receiptdate"], l["l_commitdate"]))
+    dl = late.group_by(["l_orderkey", "l_suppkey"]).aggregate([])
+    nlate = dl.group_by("l_orderkey").aggregate(
+        [("l_suppkey", "count"), ("l_suppkey", "max")])
+    nlate = nlate.rename_columns(["l_orderkey", "n_late",
+                                  "l_suppkey_max"])
+    o = pa.table({"o_orderkey": orders["o_orderkey"],
+                  "st": orders["o_orderstatus"]})
+    o = o.filter(pc.equal(o["st"], 0))
+    j = o.join(nall, keys="o_orderkey", right_keys="l_orderkey")
+    j = j.join(nlate, keys="o_orderkey", right_keys="l_orderkey")
+    j = j.filter(pc.and_(pc.greater_equal(j["n_all"], 2),
+                         pc.equal(j["n_late"], 1)))
+    wait = np.asarray(j.column("l_suppkey_max"))
+    saudi = list(nat["n_name"]).index("SAUDI ARABIA")
+    sn = supp["s_nationkey"][wait - 1]
+    cnt = np.bincount(wait[sn == saudi],
+                      minlength=int(supp["s_suppkey"].max()) + 2)
+    sk = np.nonzero(cnt)[0]
+    order = np.lexsort((sk, -cnt[sk]))[:100]
+    got = {int(sk[i]): int(cnt[sk[i]]) for i in order}
+    assert got == OQ.q21(li, orders, supp, nat)
+
+
+def test_q22_oracle_equals_acero(data):
+    cust, orders = data["customer"], data["orders"]
+    codes = np.array([13, 31, 23, 29, 30, 18, 17])
+    cc = 10 + np.asarray(cust["c_nationkey"])
+    c = pa.table({"c_custkey": cust["c_custkey"],
+                  "cc": cc.astype(np.int32),
+                  "c_acctbal": cust["c_acctbal"]})
+    inl = c.filter(pc.is_in(c["cc"], value_set=pa.array(
+        codes, type=pa.int32())))
+    posv = inl.filter(pc.greater(inl["c_acctbal"], 0.0))
+    avg = pc.mean(posv["c_acctbal"]).as_py()
+    o = pa.table({"ck": np.unique(orders["o_custkey"])})
+    sel = inl.filter(pc.greater(inl["c_acctbal"], avg))
+    sel = sel.join(o, keys="c_custkey", right_keys="ck",
+                   join_type="left anti")
+    g = sel.group_by("cc").aggregate([("c_acctbal", "count"),
+                                      ("c_acctbal", "sum")])
+    got = {str(int(k)): (int(n), v) for k, n, v in
+           zip(g.column("cc").to_pylist(),
+               g.column("c_acctbal_count").to_pylist(),
+               g.column("c_acctbal_sum").to_pylist())}
+    want = OQ.q22(cust, orders)
+    assert set(got) == set(want)
+    for k in want:
+        assert got[k][0] == want[k][0]
+        np.testing.assert_allclose(got[k][1], want[k][1], rtol=1e-9)
