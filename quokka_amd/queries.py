@@ -1601,14 +1601,19 @@ def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
     keys, _ = gb.extract()
     gb.free()
     pk = (keys // S).astype(np.int64)
+    sk2 = (keys % S).astype(np.int64)
     attr = {}
     for k, b, t, z in zip(part_host["p_partkey"], part_host["p_brand"],
                           part_host["p_type"], part_host["p_size"]):
         attr[int(k)] = (int(b), int(t), int(z))
+    # DISTINCT suppliers per (brand, type, size): a supplier supplying
+    # two parts with the SAME attributes must count once
+    seen = set()
+    for p, s_ in zip(pk, sk2):
+        seen.add(attr[int(p)] + (int(s_),))
     out = {}
-    for p in pk:
-        key = attr[int(p)]
-        out[key] = out.get(key, 0) + 1
+    for b, t, z, _ in seen:
+        out[(b, t, z)] = out.get((b, t, z), 0) + 1
     for c in [pidx, qkeys, ppx, e_pk, e_sk, bidx, ckey, ones] + \
             ([f_pk, f_sk] if nb else []):
         c.free()
@@ -1723,8 +1728,9 @@ def q22(cust_cols, ord_cols, stream=None):
         {"c_nationkey": np.dtype(np.int32),
          "c_acctbal": np.dtype(np.float64)}, [("c_nationkey", 25)],
         ["COUNT(*) as n", "SUM(c_acctbal) as s"],
-        predicate="c_acctbal > %r and (%s)" % (avg, in_list)),
-        "q22_final_%r" % avg)
+        predicate="c_acctbal > %s and (%s)" % (repr(float(avg)),
+                                                in_list)),
+        "q22_final_%s" % repr(float(avg)))
     acc = fagg.make_acc()
     if na:
         fagg.run(fin, acc, st)
