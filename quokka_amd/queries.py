@@ -1602,18 +1602,26 @@ def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
     gb.free()
     pk = (keys // S).astype(np.int64)
     sk2 = (keys % S).astype(np.int64)
-    attr = {}
-    for k, b, t, z in zip(part_host["p_partkey"], part_host["p_brand"],
-                          part_host["p_type"], part_host["p_size"]):
-        attr[int(k)] = (int(b), int(t), int(z))
     # DISTINCT suppliers per (brand, type, size): a supplier supplying
-    # two parts with the SAME attributes must count once
-    seen = set()
-    for p, s_ in zip(pk, sk2):
-        seen.add(attr[int(p)] + (int(s_),))
+    # two parts with the SAME attributes counts once — vectorized:
+    # composite (attr, supplier) ids -> unique -> counts per attr id
+    maxpk = int(part_host["p_partkey"].max())
+    b_by = np.zeros(maxpk + 2, dtype=np.int64)
+    t_by = np.zeros(maxpk + 2, dtype=np.int64)
+    z_by = np.zeros(maxpk + 2, dtype=np.int64)
+    b_by[part_host["p_partkey"]] = part_host["p_brand"]
+    t_by[part_host["p_partkey"]] = part_host["p_type"]
+    z_by[part_host["p_partkey"]] = part_host["p_size"]
+    attr_id = (b_by[pk] * 150 + t_by[pk]) * 51 + z_by[pk]
+    pair_id = attr_id * S + sk2
+    uniq = np.unique(pair_id)
+    aid, cnts = np.unique(uniq // S, return_counts=True)
     out = {}
-    for b, t, z, _ in seen:
-        out[(b, t, z)] = out.get((b, t, z), 0) + 1
+    for a, c in zip(aid, cnts):
+        z = int(a % 51)
+        t = int((a // 51) % 150)
+        b = int(a // (51 * 150))
+        out[(b, t, z)] = int(c)
     for c in [pidx, qkeys, ppx, e_pk, e_sk, bidx, ckey, ones] + \
             ([f_pk, f_sk] if nb else []):
         c.free()
