@@ -1589,7 +1589,8 @@ def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
         apx.free()
     else:
         f_pk, f_sk, na = e_pk, e_sk, nps
-    S = int(supp_cols["s_suppkey"].n) + 1
+    S = 1 << (int(supp_cols["s_suppkey"].n) + 1).bit_length()
+    sbits = S.bit_length() - 1
     ckey = DevColumn(np.int64, max(1, na))
     call("qk_i64_combine", sh, c_u64(na), f_pk.ptr, f_sk.ptr, c_i64(S),
          ckey.ptr)
@@ -1600,8 +1601,8 @@ def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
     gb.update(ckey, [ones], na)
     keys, _ = gb.extract()
     gb.free()
-    pk = (keys // S).astype(np.int64)
-    sk2 = (keys % S).astype(np.int64)
+    pk = (keys >> sbits).astype(np.int64)
+    sk2 = (keys & (S - 1)).astype(np.int64)
     # DISTINCT suppliers per (brand, type, size): a supplier supplying
     # two parts with the SAME attributes counts once — vectorized:
     # composite (attr, supplier) ids -> unique -> counts per attr id
@@ -1641,7 +1642,8 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     from .shim import c_i64
     st = stream
     sh = st.handle if st else None
-    S = int(supp_cols["s_suppkey"].n) + 1
+    S = 1 << (int(supp_cols["s_suppkey"].n) + 1).bit_length()
+    sbits = S.bit_length() - 1          # pow2 scale: decompose = shift/mask
     n = li_cols["l_orderkey"].n
 
     def distinct_pairs(idx=None, nn=None):
@@ -1687,11 +1689,11 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     nkey = int(ok_h.max()) + 2
     status_f = np.zeros(nkey, dtype=bool)
     status_f[ok_h[os_h == 0]] = True
-    nsupp_all = np.bincount((all_pairs // S), minlength=nkey)
-    lo = (late_pairs // S).astype(np.int64)
+    nsupp_all = np.bincount(all_pairs >> sbits, minlength=nkey)
+    lo = (late_pairs >> sbits).astype(np.int64)
     nsupp_late = np.bincount(lo, minlength=nkey)
     qual_orders = status_f & (nsupp_all >= 2) & (nsupp_late == 1)
-    wait_supp = (late_pairs % S)[qual_orders[lo]]
+    wait_supp = (late_pairs & (S - 1))[qual_orders[lo]]
     saudi = list(nation_names).index("SAUDI ARABIA")
     snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
     wait_supp = wait_supp[snk[wait_supp - 1] == saudi]
