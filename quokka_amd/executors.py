@@ -713,9 +713,11 @@ class GPUDistinctExecutor(Executor):
 
     def __init__(self, keys):
         self.keys = keys if isinstance(keys, str) else keys[0]
-        assert isinstance(self.keys, str), "round 1: single i64 key"
+        if not isinstance(self.keys, str):
+            raise TypeError("single distinct key column expected")
         self._table = None
         self._n_rows = 0
+        self._key_dict = None       # DeviceStringDict for string keys
 
     def __getstate__(self):
         assert self._table is None, "pickle before first execute"
@@ -728,9 +730,22 @@ class GPUDistinctExecutor(Executor):
         if not batches:
             return
         batch = pa.concat_tables(batches)
-        keys = staging.column_to_numpy(batch.column(self.keys))
+        key_col = batch.column(self.keys)
+        if _is_stringish(key_col):
+            # device string dictionary -> dense codes (consistent across
+            # batches), distinct-on-codes == distinct-on-strings
+            if self._key_dict is None:
+                self._key_dict = ops.DeviceStringDict(
+                    expected=max(1024, len(batch)))
+            keys = self._key_dict.encode_column(key_col).astype(np.int64)
+        else:
+            keys = staging.column_to_numpy(key_col)
         if keys.dtype != np.int64:
-            raise TypeError("GPUDistinctExecutor requires i64 keys")
+            if keys.dtype.kind in "iu":
+                keys = keys.astype(np.int64)
+            else:
+                raise TypeError("GPUDistinctExecutor requires integer or "
+                                "string keys, got %s" % keys.dtype)
         # within-batch unique (first occurrence), as batch.unique() does
         _, first_idx = np.unique(keys, return_index=True)
         first_idx.sort()
@@ -898,9 +913,17 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     if partitioner == "broadcast":
         return {i: data for i in range(num_target_channels)}
     assert partitioner in ("hash", "range"), partitioner
-    host_cols = {c: staging.column_to_numpy(
-        data.column(c), (string_dicts or {}).get(c))
-        for c in data.column_names}
+    def _col(c):
+        try:
+            return staging.column_to_numpy(data.column(c),
+                                           (string_dicts or {}).get(c))
+        except TypeError:
+            col = data.column(c)
+            if _is_stringish(col):       # general strings: host object
+                return np.asarray(col.to_pylist(), dtype=object)
+            raise
+
+    host_cols = {c: _col(c) for c in data.column_names}
 
     if batch_agg is not None:
         if transforms:
@@ -917,7 +940,7 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
         run_aggs = list(aggs) + ([] if have_count
                                  else ["COUNT(*) as __qk_presence"])
         dcols = {c: shim.DevColumn.from_numpy(v)
-                 for c, v in host_cols.items()}
+                 for c, v in host_cols.items() if v.dtype != object}
         schema = {c: v.dtype for c, v in dcols.items()}
         agg = jit.JitAggregate(schema, group_keys, run_aggs, predicate,
                                string_dicts)
@@ -953,7 +976,7 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     if predicate is not None:
         from . import jit
         dcols = {c: shim.DevColumn.from_numpy(v)
-                 for c, v in host_cols.items()}
+                 for c, v in host_cols.items() if v.dtype != object}
         schema = {c: v.dtype for c, v in dcols.items()}
         f = jit.JitFilter(predicate, schema, string_dicts)
         fidx, k = f.run(dcols)
@@ -967,7 +990,7 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
     if transforms:
         from . import jit
         dcols = {c: shim.DevColumn.from_numpy(v)
-                 for c, v in host_cols.items()}
+                 for c, v in host_cols.items() if v.dtype != object}
         schema = {c: v.dtype for c, v in dcols.items()}
         for out_name, expr in transforms:
             m = jit.JitMap(expr, schema)
@@ -979,8 +1002,25 @@ def gpu_partition_fn(data, source_channel, num_target_channels, key=None,
             c.free()
 
     keys = host_cols[key]
+    if keys.dtype == object:
+        # string partition keys: device string dictionary -> dense codes
+        # (the reference hashes strings inside polars,
+        # quokka_runtime.py:224 — an un-vendored implementation detail;
+        # the preserved contract is equal keys -> same channel, which
+        # code % N gives. Range partitioning needs an ordered key.)
+        if partitioner != "hash":
+            raise TypeError("string keys support hash partitioning only")
+        import pyarrow as _pa
+        sd = ops.DeviceStringDict(expected=max(1024, len(keys)))
+        keys = sd.encode_column(
+            _pa.chunked_array([_pa.array(keys)])).astype(np.int64)
+        sd.free()
     if keys.dtype != np.int64:
-        raise TypeError("gpu_partition_fn: int64 keys only in round 1")
+        if keys.dtype.kind in "iu":
+            keys = keys.astype(np.int64)
+        else:
+            raise TypeError("gpu_partition_fn: integer or string keys, "
+                            "got %s" % keys.dtype)
     kcol = shim.DevColumn.from_numpy(keys)
     if partitioner == "range":
         assert total_range, "range partitioner needs total_range"

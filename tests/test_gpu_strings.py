@@ -207,3 +207,47 @@ def test_disk_spill_join_multi_chunk(gpu):
     assert not [f for f in os.listdir(d) if f.startswith("build_")]
     with pytest.raises(ValueError, match="inner"):
         GPUDiskBuildProbeJoinExecutor(on="k", how="left", spill_dir=d)
+
+
+def test_distinct_executor_string_keys(gpu):
+    """GPUDistinctExecutor with unbounded-cardinality string keys:
+    first-occurrence rows across batches, like the reference's
+    batch.unique + anti-join state (sql_executors.py:517-554)."""
+    from quokka_amd.executors import GPUDistinctExecutor
+    rng = np.random.default_rng(81)
+    ex = GPUDistinctExecutor("k")
+    seen = set()
+    total = 0
+    for b in range(4):
+        keys = _rand_strings(rng, 3000, 900)
+        t = pa.table({"k": pa.array(keys),
+                      "v": np.arange(3000.0) + b * 10000})
+        out = ex.execute([t], 0, 0)
+        new = [k for k in dict.fromkeys(keys) if k not in seen]
+        if out is not None:
+            got = out.column("k").to_pylist()
+            assert got == new, b
+            total += out.num_rows
+        else:
+            assert not new
+        seen.update(keys)
+    assert total == len(seen)
+
+
+def test_partition_fn_string_keys_colocate(gpu):
+    """gpu_partition_fn with a string key: equal keys land on one
+    channel (the reference's contract; bucket ids differ from polars'
+    internal hash, as documented)."""
+    from quokka_amd import gpu_partition_fn
+    rng = np.random.default_rng(82)
+    keys = _rand_strings(rng, 20_000, 500)
+    t = pa.table({"k": pa.array(keys), "v": rng.random(20_000)})
+    parts = gpu_partition_fn(t, 0, 4, key="k")
+    where = {}
+    n_out = 0
+    for ch, tbl in parts.items():
+        n_out += tbl.num_rows
+        for k in set(tbl.column("k").to_pylist()):
+            assert where.setdefault(k, ch) == ch, k
+    assert n_out == 20_000
+    assert len(where) == len(set(keys))
