@@ -823,8 +823,23 @@ class GPUSortExecutor(Executor):
             return None
         ops, shim, staging = _lazy_gpu()
         t = self._batches
-        kcol = shim.DevColumn.from_numpy(
-            staging.column_to_numpy(t.column(self.key)))
+        key_col = t.column(self.key)
+        if _is_stringish(key_col):
+            # lexicographic string sort: dictionary codes are
+            # arrival-ordered, so remap code -> lexicographic rank of
+            # its value (host sorts the DISTINCT values only), then the
+            # stable device radix sort runs on the ranks
+            sd = ops.DeviceStringDict(expected=max(1024, t.num_rows))
+            codes = sd.encode_column(key_col)
+            vals = sd.values
+            rank_of = np.empty(len(vals), dtype=np.int64)
+            rank_of[np.argsort(np.asarray(vals, dtype=object))] = \
+                np.arange(len(vals))
+            keys = rank_of[codes]
+            sd.free()
+        else:
+            keys = staging.column_to_numpy(key_col)
+        kcol = shim.DevColumn.from_numpy(keys)
         perm = ops.sort_permutation(kcol)
         order = perm.to_numpy(perm.n)
         kcol.free()

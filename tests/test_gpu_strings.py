@@ -251,3 +251,24 @@ def test_partition_fn_string_keys_colocate(gpu):
             assert where.setdefault(k, ch) == ch, k
     assert n_out == 20_000
     assert len(where) == len(set(keys))
+
+
+def test_sort_executor_string_key(gpu):
+    """GPUSortExecutor sorts string keys LEXICOGRAPHICALLY (codes
+    remapped to value ranks; stable device radix sort on the ranks)."""
+    from quokka_amd.executors import GPUSortExecutor
+    rng = np.random.default_rng(83)
+    keys = _rand_strings(rng, 30_000, 5000)
+    v = np.arange(30_000.0)
+    ex = GPUSortExecutor("k")
+    t = pa.table({"k": pa.array(keys), "v": v})
+    ex.execute([t.slice(0, 15_000)], 0, 0)
+    ex.execute([t.slice(15_000)], 0, 0)
+    out = ex.done(0)
+    got = out.column("k").to_pylist()
+    assert got == sorted(keys)
+    # stability: equal keys keep input order
+    gv = np.asarray(out.column("v"))
+    for i in range(1, len(got)):
+        if got[i] == got[i - 1]:
+            assert gv[i] > gv[i - 1]
