@@ -44,6 +44,17 @@ class QkParquetError(RuntimeError):
     pass
 
 
+def _tile_block(data_off, dst_row, num_values, elem_size):
+    """Vectorized per-page tile table (ntiles x 3 u64) — the former
+    per-tile Python loop measured ~45 ms of the warm SF10 e2e pass."""
+    starts = np.arange(0, num_values, _TILE, dtype=np.uint64)
+    out = np.empty((len(starts), 3), dtype=np.uint64)
+    out[:, 0] = data_off + starts * elem_size
+    out[:, 1] = dst_row + starts
+    out[:, 2] = np.minimum(_TILE, num_values - starts)
+    return out
+
+
 def _check_levels_v1(buf, pos, num_values, max_def):
     """Definition-levels block of a v1 data page ([u32 len][RLE runs]).
     Verify every level == max_def (no nulls) without expanding; return
@@ -156,11 +167,8 @@ class _Chunk:
                     raise QkParquetError("PLAIN BYTE_ARRAY unsupported; "
                                          "write with use_dictionary=True")
                 es = self.dtype.itemsize
-                s = 0
-                while s < p.num_values:
-                    m = min(_TILE, p.num_values - s)
-                    self.plain_tiles.append((data + s * es, row + s, m))
-                    s += m
+                self.plain_tiles.append(
+                    _tile_block(data, row, p.num_values, es))
             elif p.encoding == ENC_RLE_DICT:
                 bw = buf[data]
                 if bw > 32:
@@ -388,12 +396,9 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
             if is_ba:
                 raise QkParquetError("PLAIN BYTE_ARRAY unsupported; "
                                      "write with use_dictionary=True")
-            es = dtype.itemsize
-            s = 0
-            while s < p.num_values:
-                m = min(_TILE, p.num_values - s)
-                cur.plain_tiles.append((data + s * es, ent["row"] + s, m))
-                s += m
+            cur.plain_tiles.append(
+                _tile_block(data, ent["row"], p.num_values,
+                            dtype.itemsize))
         elif p.encoding == ENC_RLE_DICT:
             bw = first
             if bw < 0 or bw > 32:
@@ -569,9 +574,10 @@ def _decode_column(shim, dev_file, chunks, total):
         dents.free()
         idx.free()
         dcat.free()
-    all_tiles = [t for ch in chunks for t in ch.plain_tiles]
+    all_tiles = [np.atleast_2d(np.asarray(t, dtype=np.uint64))
+                 for ch in chunks for t in ch.plain_tiles]
     if all_tiles:
-        tiles = np.asarray(all_tiles, dtype=np.uint64)
+        tiles = np.ascontiguousarray(np.vstack(all_tiles))
         dtile = DevBuffer(tiles.nbytes)
         shim.call("qk_h2d", dtile.ptr, tiles.ctypes.data_as(c_vp),
                   c_u64(tiles.nbytes))

@@ -54,9 +54,13 @@ def sim_column(raw, ci):
     dt = chunks[0].dtype
     out = np.empty(total, dtype=dt)
     for ch in chunks:
-        for src, dst, cnt in ch.plain_tiles:
-            out[dst:dst + cnt] = np.frombuffer(raw, dtype=dt, count=cnt,
-                                               offset=src)
+        # plain_tiles holds vectorized per-page blocks (ntiles x 3 u64)
+        for blk in ch.plain_tiles:
+            for src, dst, cnt in np.atleast_2d(
+                    np.asarray(blk, dtype=np.int64)):
+                src, dst, cnt = int(src), int(dst), int(cnt)
+                out[dst:dst + cnt] = np.frombuffer(raw, dtype=dt,
+                                                   count=cnt, offset=src)
         if ch.rle_pages:
             idx, base = sim_rle_pages(raw, ch.rle_pages)
             out[base:base + len(idx)] = ch.dict_vals[idx]
