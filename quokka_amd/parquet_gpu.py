@@ -45,8 +45,11 @@ class QkParquetError(RuntimeError):
 
 
 def _tile_block(data_off, dst_row, num_values, elem_size):
-    """Vectorized per-page tile table (ntiles x 3 u64) — the former
-    per-tile Python loop measured ~45 ms of the warm SF10 e2e pass."""
+    """Vectorized per-page tile table (ntiles x 3 u64). A/B'd against
+    the per-tile Python loop: roughly neutral at 1 MB pages (~5 ms per
+    300 pages either way — planning cost is spread across per-page
+    overheads), clearly ahead for small-page files; kept for the
+    cheaper downstream vstack (0.8 vs 2.7 ms per 4800 tiles)."""
     starts = np.arange(0, num_values, _TILE, dtype=np.uint64)
     out = np.empty((len(starts), 3), dtype=np.uint64)
     out[:, 0] = data_off + starts * elem_size
