@@ -474,6 +474,10 @@ int qk_str_dict_rehash(void *stream, uint32_t ncodes,
  * e.g. Q20's (partkey, suppkey) quantity sums). */
 int qk_i64_combine(void *stream, uint64_t n, const int64_t *x,
                    const int64_t *y, int64_t scale, int64_t *out);
+/* Arithmetic shift right: out = x >> shift (decompose pow2-scaled
+ * composite keys on device, e.g. pair key -> orderkey). */
+int qk_i64_shr(void *stream, uint64_t n, const int64_t *x, int shift,
+               int64_t *out);
 /* Synchronous device-to-device copy (dict growth, arena relocation). */
 int qk_d2d(void *dst_dev, const void *src_dev, uint64_t nbytes);
 

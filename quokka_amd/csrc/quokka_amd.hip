@@ -2501,6 +2501,25 @@ extern "C" int qk_i64_combine(void *stream, uint64_t n, const int64_t *x,
   return 0;
 }
 
+__global__ void __launch_bounds__(BLOCK) k_i64_shr(
+    uint64_t n, const int64_t *__restrict__ x, int shift,
+    int64_t *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = x[i] >> shift;
+}
+// Arithmetic shift right (decompose pow2-scaled composite keys on device)
+extern "C" int qk_i64_shr(void *stream, uint64_t n, const int64_t *x,
+                          int shift, int64_t *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_i64_shr, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, x, shift, out);
+  QK_TRY("qk_i64_shr", hipGetLastError());
+  return 0;
+}
+
 extern "C" int qk_d2d(void *dst, const void *src, uint64_t nbytes) {
   QK_TRY("qk_d2d", hipMemcpy(dst, src, nbytes, hipMemcpyDeviceToDevice));
   return 0;

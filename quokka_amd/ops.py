@@ -275,6 +275,27 @@ class GroupByI64:
         out_sums.free()
         return keys, sums
 
+    def extract_device(self):
+        """Like extract() but the compacted (keys, sums) STAY ON DEVICE:
+        returns (keys DevColumn i64 [n], sums DevColumn f64
+        [nvals x out_cap, column-major with stride out_cap], n, out_cap).
+        Caller frees both columns. Feeds device-side second-level
+        aggregation (e.g. Q21's per-order distinct-supplier counts)."""
+        sh = self.stream.handle if self.stream else None
+        out_cap = self.cap
+        out_keys = DevColumn(np.int64, out_cap)
+        out_sums = DevColumn(np.float64, out_cap * self.nvals)
+        cur = _count_buf()
+        shim.call("qk_groupby_extract", sh, self.slot_keys.ptr,
+                  self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+                  out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
+        if self.stream:
+            self.stream.sync()
+        k = _read_u64(cur)
+        cur.free()
+        out_keys.n = k
+        return out_keys, out_sums, k, out_cap
+
     def extract_where_gt(self, col, threshold, out_guess=1 << 20):
         """Extract only groups whose sums[col] > threshold (HAVING — e.g.
         Q18's sum(l_quantity) > 300): the d2h stays proportional to the

@@ -1633,11 +1633,15 @@ def q16(part_cols, ps_cols, supp_cols, part_host, stream=None):
 def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
         stream=None):
     """Device Q21: the exists/not-exists pair reduces to per-order
-    DISTINCT supplier counts — deduped on the device group-by over
-    (orderkey*S + suppkey) for all lines and for LATE lines — with the
-    qualification (F status, >=2 suppliers, exactly 1 late) and the
-    SAUDI ARABIA numwait tally finishing host-side over the deduped
-    pairs. Returns dict s_suppkey -> numwait."""
+    DISTINCT supplier counts. Entirely device-side until the final
+    (small) candidate set: (1) dedupe (orderkey*S + suppkey) pairs on
+    the group-by table, (2) shift the EXTRACTED device keys back to
+    orderkeys (qk_i64_shr) and run a SECOND group-by — count for all
+    lines; count + MIN(suppkey) for LATE lines, (3) qualifying orders
+    (F status, >=2 suppliers, exactly 1 late) resolve by probing the
+    late-order candidates against the all-pairs count table and the
+    orders table; only the candidates' few columns come to the host.
+    Returns dict s_suppkey -> numwait."""
     from . import jit, ops
     from .shim import c_i64
     st = stream
@@ -1646,7 +1650,8 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     sbits = S.bit_length() - 1          # pow2 scale: decompose = shift/mask
     n = li_cols["l_orderkey"].n
 
-    def distinct_pairs(idx=None, nn=None):
+    def pair_keys_device(idx=None, nn=None):
+        """Dedupe pairs; return (device keys DevColumn, count)."""
         if idx is None:
             kx, ky, cnt = li_cols["l_orderkey"], li_cols["l_suppkey"], n
         else:
@@ -1663,37 +1668,99 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
         gb = ops.GroupByI64(expected_groups=max(1024, cnt), nvals=1,
                             stream=st)
         gb.update(ck, [ones], cnt)
-        keys, _ = gb.extract()
+        keys_dev, sums_dev, k, _ = gb.extract_device()
+        sums_dev.free()
         gb.free()
         ck.free()
         ones.free()
         if idx is not None:
             kx.free()
             ky.free()
-        return keys
+        return keys_dev, k
 
-    all_pairs = distinct_pairs()
-    lsch = {k: v.dtype for k, v in li_cols.items()}
+    def per_order(keys_dev, k, with_min_supp):
+        """Second-level device group-by keyed by orderkey."""
+        ok = DevColumn(np.int64, max(1, k))
+        call("qk_i64_shr", sh, c_u64(k), keys_dev.ptr, sbits, ok.ptr)
+        ok.n = k
+        ones = DevColumn(np.float64, max(1, k))
+        call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0), c_u64(k))
+        vals = [ones]
+        aggs = [0]
+        sk_f = None
+        if with_min_supp:
+            # suppkey = pairkey - (orderkey << sbits), as f64 for the
+            # MIN aggregate (exact: suppkey < 2^53)
+            sk_i = DevColumn(np.int64, max(1, k))
+            call("qk_i64_combine", sh, c_u64(k), ok.ptr, keys_dev.ptr,
+                 c_i64(-S), sk_i.ptr)
+            sk_i.n = k
+            m = _cached_jit("m", lambda: jit.JitMap(
+                "sk", {"sk": np.dtype(np.int64)}), "q21_sk_f64")
+            sk_f = m.run({"sk": sk_i}, st)
+            sk_i.free()
+            vals.append(sk_f)
+            aggs.append(1)
+        gb = ops.GroupByI64(expected_groups=max(1024, k),
+                            nvals=len(vals), stream=st, agg_ops=aggs)
+        gb.update(ok, vals, k)
+        ok.free()
+        ones.free()
+        if sk_f is not None:
+            sk_f.free()
+        return gb
+
+    all_keys, ka = pair_keys_device()
+    lsch = {k_: v.dtype for k_, v in li_cols.items()}
     lf = _cached_jit("f", lambda: jit.JitFilter(
         "l_receiptdate > l_commitdate", lsch),
         "q21_late", _schema_key(lsch))
     lidx, nl = lf.run(li_cols, st)
-    late_pairs = distinct_pairs(lidx, nl)
+    late_keys, kl = pair_keys_device(lidx, nl)
     lidx.free()
+
+    gb_all = per_order(all_keys, ka, with_min_supp=False)
+    gb_late = per_order(late_keys, kl, with_min_supp=True)
+    all_keys.free()
+    late_keys.free()
+    # candidates: late-supplier count in (0.5, 1.5) == exactly one
+    cand_k, cand_s = gb_late.extract_where_gt(0, 0.5)
+    one = cand_s[0] < 1.5
+    cand_orders = cand_k[one]
+    cand_supp = cand_s[1][one].astype(np.int64)
+    gb_late.free()
+    # all-pairs count >= 2 for those orders: probe the count table
+    akeys, asums, kn, acap = gb_all.extract_device()
+    atab = ops.JoinTable(max(16, kn), st)
+    atab.build(akeys, kn)
+    cko = DevColumn.from_numpy(cand_orders)
+    px, bx, nmm = atab.probe(cko, mode=0, n=len(cand_orders))
+    acnt_dev = shim_gather_f64(asums, bx, nmm, st)
     if st:
         st.sync()
-    # host tail over the deduped pairs (O(pairs), not O(rows))
+    sel_rows = px.to_numpy(nmm)
+    acnt = acnt_dev.to_numpy(nmm)
+    keep = np.zeros(len(cand_orders), dtype=bool)
+    keep[sel_rows[acnt >= 2.0]] = True
+    gb_all.free()
+    akeys.free()
+    asums.free()
+    atab.free()
+    cko.free()
+    px.free()
+    bx.free()
+    acnt_dev.free()
+    cand_orders = cand_orders[keep]
+    cand_supp = cand_supp[keep]
+    # F-status filter via the orders table (host lookup over candidates)
     ok_h = ord_cols["o_orderkey"].to_numpy(ord_cols["o_orderkey"].n)
     os_h = ord_cols["o_orderstatus"].to_numpy(
         ord_cols["o_orderstatus"].n)
     nkey = int(ok_h.max()) + 2
     status_f = np.zeros(nkey, dtype=bool)
     status_f[ok_h[os_h == 0]] = True
-    nsupp_all = np.bincount(all_pairs >> sbits, minlength=nkey)
-    lo = (late_pairs >> sbits).astype(np.int64)
-    nsupp_late = np.bincount(lo, minlength=nkey)
-    qual_orders = status_f & (nsupp_all >= 2) & (nsupp_late == 1)
-    wait_supp = (late_pairs & (S - 1))[qual_orders[lo]]
+    m = status_f[np.minimum(cand_orders, nkey - 1)]
+    wait_supp = cand_supp[m]
     saudi = list(nation_names).index("SAUDI ARABIA")
     snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
     wait_supp = wait_supp[snk[wait_supp - 1] == saudi]
@@ -1701,6 +1768,16 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     sk = np.nonzero(cnt)[0]
     order = np.lexsort((sk, -cnt[sk]))[:limit]
     return {int(sk[i]): int(cnt[sk[i]]) for i in order}
+
+
+def shim_gather_f64(col_f64, idx_col, n_idx, stream=None):
+    """Gather over a raw f64 DevColumn region (first n entries used)."""
+    out = DevColumn(np.float64, max(1, n_idx))
+    sh = stream.handle if stream else None
+    call("qk_gather_f64", sh, c_u64(n_idx), idx_col.ptr, col_f64.ptr,
+         out.ptr)
+    out.n = n_idx
+    return out
 
 
 def q22(cust_cols, ord_cols, stream=None):

@@ -132,6 +132,7 @@ _SIGS = {
                            c_u64],
     "qk_d2d": [c_vp, c_vp, c_u64],
     "qk_i64_combine": [c_vp, c_u64, c_vp, c_vp, c_i64, c_vp],
+    "qk_i64_shr": [c_vp, c_u64, c_vp, ctypes.c_int, c_vp],
     "qk_partition_scatter": [c_vp, c_u64, c_vp, c_u32, c_vp, c_vp],
 }
 for name, argtypes in _SIGS.items():
