@@ -493,6 +493,16 @@ int qk_d2d(void *dst_dev, const void *src_dev, uint64_t nbytes);
 int qk_snappy_pages(void *stream, uint64_t npages, const uint64_t *descs_dev,
                     const uint8_t *src_dev, uint8_t *dst_dev,
                     int64_t *out_dev);
+/* GPU gzip (DEFLATE) decompression of Parquet pages: entropy decode is
+ * sequential, so lane 0 decodes a page's bit stream (canonical Huffman
+ * tables built in LDS) and pages decode independently in parallel.
+ * descs_dev: npages x 8 u64 [src_off, src_len, dst_off,
+ * uncompressed_len, mode (1 = gzip wrapper / 0 = raw deflate), 0,0,0];
+ * out_dev: npages x 4 i64 [written, err (0 ok/1 corrupt/2 length
+ * mismatch/3 unsupported), first_byte, 0]. */
+int qk_gzip_pages(void *stream, uint64_t npages, const uint64_t *descs_dev,
+                  const uint8_t *src_dev, uint8_t *dst_dev,
+                  int64_t *out_dev);
 /* Host-side Thrift compact-protocol walk of one column chunk's page
  * headers (parquet-format PageHeader; the metadata side of the decode —
  * pyarrow exposes only the footer, and walking thousands of headers in

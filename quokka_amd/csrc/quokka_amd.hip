@@ -2704,6 +2704,254 @@ extern "C" int qk_snappy_pages(void *stream, uint64_t npages,
   return 0;
 }
 
+// ---- GPU gzip (DEFLATE, RFC 1951/1952) page decompression --------------
+// Parquet's GZIP codec wraps each page in a gzip stream. Entropy decode
+// is inherently sequential, so ONE LANE decodes a page's bit stream
+// (canonical Huffman via per-length first/count tables built in LDS);
+// pages decode independently and hundreds in flight fill the chip —
+// the same division of labor as qk_snappy_pages, with the sequential
+// part larger. Supported: stored/fixed/dynamic blocks, standard gzip
+// headers (FEXTRA/FNAME/FCOMMENT/FHCRC skipped); errors are reported
+// per page, never guessed.
+namespace qkgz {
+struct BitReader {
+  const uint8_t *p, *end;
+  uint32_t bitbuf;
+  int bitcnt;
+  bool fail;
+  __device__ void init(const uint8_t *s, const uint8_t *e) {
+    p = s; end = e; bitbuf = 0; bitcnt = 0; fail = false;
+  }
+  __device__ uint32_t bits(int n) {       // n <= 24, LSB-first
+    while (bitcnt < n) {
+      if (p >= end) { fail = true; return 0; }
+      bitbuf |= (uint32_t)(*p++) << bitcnt;
+      bitcnt += 8;
+    }
+    uint32_t v = bitbuf & ((1u << n) - 1);
+    bitbuf >>= n;
+    bitcnt -= n;
+    return v;
+  }
+  __device__ void align_byte() {
+    bitbuf = 0; bitcnt = 0;
+  }
+};
+
+// canonical Huffman decode tables (per length 1..15)
+struct Huff {
+  uint16_t count[16];     // codes of each length
+  uint16_t first[16];     // first canonical code of each length
+  uint16_t offset[16];    // index into syms of first code of each length
+  uint16_t syms[288];
+  __device__ bool build(const uint8_t *lens, int n) {
+    for (int i = 0; i < 16; i++) count[i] = 0;
+    for (int i = 0; i < n; i++) count[lens[i]]++;
+    count[0] = 0;
+    uint32_t code = 0;
+    int total = 0;
+    for (int l = 1; l < 16; l++) {
+      code = (code + count[l - 1]) << 1;
+      first[l] = (uint16_t)code;
+      offset[l] = total;
+      total += count[l];
+      if (code + count[l] > (1u << l)) return false;   // over-subscribed
+      code += 0;                                        // (first advanced below)
+    }
+    // recompute properly: first[l] must accumulate counts of SHORTER
+    // lengths only (the loop above already does via code shifting)
+    uint16_t idx[16];
+    for (int l = 0; l < 16; l++) idx[l] = offset[l];
+    for (int i = 0; i < n; i++)
+      if (lens[i]) syms[idx[lens[i]]++] = (uint16_t)i;
+    return true;
+  }
+  __device__ int decode(BitReader &br) {
+    uint32_t code = 0;
+    for (int l = 1; l < 16; l++) {
+      code |= br.bits(1);
+      if (br.fail) return -1;
+      if (count[l] && code < (uint32_t)first[l] + count[l] &&
+          code >= first[l])
+        return syms[offset[l] + (code - first[l])];
+      code <<= 1;
+    }
+    return -1;
+  }
+};
+
+__constant__ uint16_t LEN_BASE[29] = {
+    3, 4, 5, 6, 7, 8, 9, 10, 11, 13, 15, 17, 19, 23, 27, 31,
+    35, 43, 51, 59, 67, 83, 99, 115, 131, 163, 195, 227, 258};
+__constant__ uint8_t LEN_EXTRA[29] = {
+    0, 0, 0, 0, 0, 0, 0, 0, 1, 1, 1, 1, 2, 2, 2, 2,
+    3, 3, 3, 3, 4, 4, 4, 4, 5, 5, 5, 5, 0};
+__constant__ uint16_t DIST_BASE[30] = {
+    1, 2, 3, 4, 5, 7, 9, 13, 17, 25, 33, 49, 65, 97, 129, 193,
+    257, 385, 513, 769, 1025, 1537, 2049, 3073, 4097, 6145, 8193,
+    12289, 16385, 24577};
+__constant__ uint8_t DIST_EXTRA[30] = {
+    0, 0, 0, 0, 1, 1, 2, 2, 3, 3, 4, 4, 5, 5, 6, 6,
+    7, 7, 8, 8, 9, 9, 10, 10, 11, 11, 12, 12, 13, 13};
+__constant__ uint8_t CLC_ORDER[19] = {
+    16, 17, 18, 0, 8, 7, 9, 6, 10, 5, 11, 4, 12, 3, 13, 2, 14, 1, 15};
+}  // namespace qkgz
+
+// LDS scratch per workgroup (one wave): the two Huffman tables + the
+// code-length array for dynamic blocks
+struct QkGzLds {
+  qkgz::Huff lit, dist;
+  uint8_t lens[320];
+};
+
+// desc per page (8 u64): [src_off, src_len, dst_off, uncompressed_len,
+//   mode (0 raw deflate / 1 gzip wrapper), 0, 0, 0]
+// out per page (4 i64): [written, err, 0, 0]; err 0 ok, 1 corrupt,
+//   2 length mismatch, 3 unsupported
+__global__ void __launch_bounds__(64) k_gzip_pages(
+    uint64_t npages, const uint64_t *__restrict__ descs,
+    const uint8_t *__restrict__ src, uint8_t *__restrict__ dst,
+    int64_t *__restrict__ out) {
+  using namespace qkgz;
+  __shared__ QkGzLds L;
+  uint64_t page = blockIdx.x;
+  if (page >= npages) return;
+  if (threadIdx.x != 0) return;   // entropy decode is sequential; other
+                                  // lanes idle (pages fill the chip)
+  const uint64_t *d = descs + page * 8;
+  const uint8_t *ip = src + d[0];
+  const uint8_t *iend = ip + d[1];
+  uint8_t *op = dst + d[2];
+  uint64_t ocap = d[3];
+  int64_t *o = out + page * 4;
+  o[0] = 0; o[1] = 0; o[2] = 0; o[3] = 0;
+
+  if (d[4] == 1) {                       // gzip wrapper (RFC 1952)
+    if (iend - ip < 10 || ip[0] != 0x1f || ip[1] != 0x8b || ip[2] != 8) {
+      o[1] = 3; return;
+    }
+    uint8_t flg = ip[3];
+    ip += 10;
+    if (flg & 4) {                       // FEXTRA
+      if (iend - ip < 2) { o[1] = 1; return; }
+      uint32_t xl = ip[0] | ((uint32_t)ip[1] << 8);
+      ip += 2 + xl;
+    }
+    if (flg & 8)  while (ip < iend && *ip++) {}   // FNAME
+    if (flg & 16) while (ip < iend && *ip++) {}   // FCOMMENT
+    if (flg & 2)  ip += 2;                        // FHCRC
+    if (ip >= iend) { o[1] = 1; return; }
+  }
+
+  BitReader br;
+  br.init(ip, iend);
+  uint64_t w = 0;
+  for (;;) {
+    uint32_t bfinal = br.bits(1);
+    uint32_t btype = br.bits(2);
+    if (br.fail) { o[1] = 1; return; }
+    if (btype == 0) {                    // stored
+      br.align_byte();
+      // br.p already consumed whole bytes; realign: p points past
+      // consumed bytes (bitbuf dropped) — p is correct since bits()
+      // only advances p per byte
+      if (br.end - br.p < 4) { o[1] = 1; return; }
+      uint32_t len = br.p[0] | ((uint32_t)br.p[1] << 8);
+      uint32_t nlen = br.p[2] | ((uint32_t)br.p[3] << 8);
+      br.p += 4;
+      if ((len ^ 0xFFFF) != nlen || br.end - br.p < (long)len ||
+          w + len > ocap) { o[1] = 1; return; }
+      for (uint32_t i = 0; i < len; i++) op[w + i] = br.p[i];
+      br.p += len;
+      w += len;
+    } else if (btype == 1 || btype == 2) {
+      if (btype == 1) {                  // fixed tables
+        for (int i = 0; i < 144; i++) L.lens[i] = 8;
+        for (int i = 144; i < 256; i++) L.lens[i] = 9;
+        for (int i = 256; i < 280; i++) L.lens[i] = 7;
+        for (int i = 280; i < 288; i++) L.lens[i] = 8;
+        if (!L.lit.build(L.lens, 288)) { o[1] = 1; return; }
+        for (int i = 0; i < 30; i++) L.lens[i] = 5;
+        if (!L.dist.build(L.lens, 30)) { o[1] = 1; return; }
+      } else {                           // dynamic tables
+        uint32_t hlit = br.bits(5) + 257;
+        uint32_t hdist = br.bits(5) + 1;
+        uint32_t hclen = br.bits(4) + 4;
+        if (br.fail || hlit > 288 || hdist > 30) { o[1] = 1; return; }
+        uint8_t cl_lens[19];
+        for (int i = 0; i < 19; i++) cl_lens[i] = 0;
+        for (uint32_t i = 0; i < hclen; i++)
+          cl_lens[CLC_ORDER[i]] = (uint8_t)br.bits(3);
+        Huff clh;
+        if (br.fail || !clh.build(cl_lens, 19)) { o[1] = 1; return; }
+        uint32_t nl = hlit + hdist;
+        uint32_t i = 0;
+        while (i < nl) {
+          int s = clh.decode(br);
+          if (s < 0) { o[1] = 1; return; }
+          if (s < 16) {
+            L.lens[i++] = (uint8_t)s;
+          } else if (s == 16) {
+            if (i == 0) { o[1] = 1; return; }
+            uint32_t r = 3 + br.bits(2);
+            uint8_t v = L.lens[i - 1];
+            while (r-- && i < nl) L.lens[i++] = v;
+          } else if (s == 17) {
+            uint32_t r = 3 + br.bits(3);
+            while (r-- && i < nl) L.lens[i++] = 0;
+          } else {
+            uint32_t r = 11 + br.bits(7);
+            while (r-- && i < nl) L.lens[i++] = 0;
+          }
+          if (br.fail) { o[1] = 1; return; }
+        }
+        if (!L.lit.build(L.lens, hlit)) { o[1] = 1; return; }
+        if (!L.dist.build(L.lens + hlit, hdist)) { o[1] = 1; return; }
+      }
+      for (;;) {                         // decode symbols
+        int s = L.lit.decode(br);
+        if (s < 0) { o[1] = 1; return; }
+        if (s < 256) {
+          if (w >= ocap) { o[1] = 1; return; }
+          op[w++] = (uint8_t)s;
+        } else if (s == 256) {
+          break;
+        } else {
+          s -= 257;
+          if (s >= 29) { o[1] = 1; return; }
+          uint32_t len = LEN_BASE[s] + br.bits(LEN_EXTRA[s]);
+          int ds = L.dist.decode(br);
+          if (ds < 0 || ds >= 30) { o[1] = 1; return; }
+          uint32_t distv = DIST_BASE[ds] + br.bits(DIST_EXTRA[ds]);
+          if (br.fail || distv > w || w + len > ocap) {
+            o[1] = 1; return;
+          }
+          for (uint32_t k = 0; k < len; k++, w++)
+            op[w] = op[w - distv];
+        }
+      }
+    } else {
+      o[1] = 3; return;                  // reserved btype
+    }
+    if (bfinal) break;
+  }
+  if (w != d[3]) { o[1] = 2; return; }
+  o[0] = (int64_t)w;
+  o[2] = w ? (int64_t)op[0] : -1;   // the RLE bit-width byte for
+                                    // dictionary-index pages
+}
+extern "C" int qk_gzip_pages(void *stream, uint64_t npages,
+                             const uint64_t *descs_dev,
+                             const uint8_t *src_dev, uint8_t *dst_dev,
+                             int64_t *out_dev) {
+  if (!npages) return 0;
+  hipLaunchKernelGGL(k_gzip_pages, dim3((uint32_t)npages), dim3(64), 0,
+                     (hipStream_t)stream, npages, descs_dev, src_dev,
+                     dst_dev, out_dev);
+  QK_TRY("qk_gzip_pages", hipGetLastError());
+  return 0;
+}
+
 // ---- host-side Thrift page-header walker -------------------------------
 // Parquet page headers are Thrift compact-protocol structs between pages
 // (parquet-format PageHeader). pyarrow does not expose them, and walking

@@ -242,9 +242,10 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
     chunk_pages = []
     row = dst0
     for col in cols_meta:
-        if col.compression not in ("SNAPPY", "UNCOMPRESSED"):
-            raise QkParquetError("compression %s unsupported (SNAPPY and "
-                                 "UNCOMPRESSED only)" % col.compression)
+        if col.compression not in ("SNAPPY", "GZIP", "UNCOMPRESSED"):
+            raise QkParquetError("compression %s unsupported (SNAPPY, "
+                                 "GZIP and UNCOMPRESSED only)"
+                                 % col.compression)
         start = col.data_page_offset
         if col.dictionary_page_offset is not None:
             start = min(start, col.dictionary_page_offset)
@@ -254,6 +255,7 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
         row += col.num_values
 
     descs = []          # qk_snappy_pages descriptors
+    gdescs = []         # qk_gzip_pages descriptors
     copies = []         # byte-copy tiles (uncompressed pages / v2 levels)
     plans = []          # per page: post-processing info
     off = 0
@@ -262,38 +264,64 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
         for p in pl:
             # v2 pages carry is_compressed (writers leave incompressible
             # pages raw and clear it) — those route to the byte-copy path
-            snappy = col.compression == "SNAPPY" and p.is_compressed
+            codec = col.compression if p.is_compressed else "UNCOMPRESSED"
+            if codec == "GZIP" and p.kind == T.PAGE_DATA and max_def > 0:
+                # a nullable v1 page keeps its def-levels INSIDE the
+                # compressed stream; the gzip kernel (unlike snappy's)
+                # does not parse them — write non-nullable or v2 pages
+                raise QkParquetError(
+                    "GZIP v1 pages with definition levels unsupported; "
+                    "write the column non-nullable or data_page_version"
+                    "='2.0'")
             ent = {"chunk": ci, "page": p, "off": off, "row": rows,
-                   "desc": None}
-            if not snappy:
+                   "desc": None, "codec": codec}
+            if codec == "UNCOMPRESSED":
                 copies.append((p.data_off, off, p.data_len))
                 ent["kind"] = "raw"
                 ent["size"] = p.data_len
             elif p.kind == T.PAGE_DATA:            # v1: levels inside
-                descs.append((p.data_off, p.data_len, off,
-                              p.uncompressed_len,
-                              1 if max_def > 0 else 0, p.num_values, 0, 0))
                 ent["kind"] = "v1"
-                ent["desc"] = len(descs) - 1
                 ent["size"] = p.uncompressed_len
+                if codec == "SNAPPY":
+                    descs.append((p.data_off, p.data_len, off,
+                                  p.uncompressed_len,
+                                  1 if max_def > 0 else 0,
+                                  p.num_values, 0, 0))
+                    ent["desc"] = len(descs) - 1
+                else:
+                    gdescs.append((p.data_off, p.data_len, off,
+                                   p.uncompressed_len, 1, 0, 0, 0))
+                    ent["gdesc"] = len(gdescs) - 1
             elif p.kind == T.PAGE_DICT:
-                descs.append((p.data_off, p.data_len, off,
-                              p.uncompressed_len, 0, p.num_values, 0, 0))
                 ent["kind"] = "dict"
-                ent["desc"] = len(descs) - 1
                 ent["size"] = p.uncompressed_len
+                if codec == "SNAPPY":
+                    descs.append((p.data_off, p.data_len, off,
+                                  p.uncompressed_len, 0, p.num_values,
+                                  0, 0))
+                    ent["desc"] = len(descs) - 1
+                else:
+                    gdescs.append((p.data_off, p.data_len, off,
+                                   p.uncompressed_len, 1, 0, 0, 0))
+                    ent["gdesc"] = len(gdescs) - 1
             else:                                  # v2: levels NOT compressed
                 if p.num_nulls:
                     raise QkParquetError("page contains nulls")
                 lv = p.v2_levels_len
                 if lv:
                     copies.append((p.data_off, off, lv))
-                descs.append((p.data_off + lv, p.data_len - lv, off + lv,
-                              p.uncompressed_len - lv, 0, p.num_values,
-                              0, 0))
                 ent["kind"] = "v2"
-                ent["desc"] = len(descs) - 1
                 ent["size"] = p.uncompressed_len
+                if codec == "SNAPPY":
+                    descs.append((p.data_off + lv, p.data_len - lv,
+                                  off + lv, p.uncompressed_len - lv, 0,
+                                  p.num_values, 0, 0))
+                    ent["desc"] = len(descs) - 1
+                else:
+                    gdescs.append((p.data_off + lv, p.data_len - lv,
+                                   off + lv, p.uncompressed_len - lv, 1,
+                                   0, 0, 0))
+                    ent["gdesc"] = len(gdescs) - 1
             if p.kind != T.PAGE_DICT:
                 rows += p.num_values
             plans.append(ent)
@@ -308,6 +336,30 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
         shim.call("qk_pq_plain_copy", None, c_u64(len(tiles)), dt.ptr,
                   dev_file.ptr, scratch.ptr, ctypes.c_uint32(1))
         dt.free()
+    gresults = None
+    if gdescs:
+        ga = np.asarray(gdescs, dtype=np.uint64)
+        gd = DevBuffer(ga.nbytes)
+        shim.call("qk_h2d", gd.ptr, ga.ctypes.data_as(c_vp),
+                  c_u64(ga.nbytes))
+        gout = DevBuffer(len(gdescs) * 4 * 8)
+        shim.call("qk_gzip_pages", None, c_u64(len(gdescs)), gd.ptr,
+                  dev_file.ptr, scratch.ptr, gout.ptr)
+        shim.call("qk_stream_sync", None)
+        gresults = np.zeros(len(gdescs) * 4, dtype=np.int64)
+        shim.call("qk_d2h", gresults.ctypes.data_as(c_vp), gout.ptr,
+                  c_u64(gresults.nbytes))
+        gresults = gresults.reshape(-1, 4)
+        gd.free()
+        gout.free()
+        bad = np.nonzero(gresults[:, 1])[0]
+        if bad.size:
+            code = int(gresults[bad[0], 1])
+            raise QkParquetError(
+                "gzip page %d failed: %s" % (
+                    bad[0], {1: "corrupt stream",
+                             2: "uncompressed-length mismatch",
+                             3: "unsupported feature"}.get(code, code)))
     results = None
     if descs:
         da = np.asarray(descs, dtype=np.uint64)
@@ -372,14 +424,22 @@ def _snappy_column(shim, raw, dev_file, cols_meta, max_def, dst0):
             continue
         # data pages: find where values start + the RLE bit-width byte
         if ent["kind"] == "v1":
-            r = results[ent["desc"]]
-            data = o + int(r[0])
-            first = int(r[2])
+            if ent.get("gdesc") is not None:
+                # GZIP v1 reaches here only with max_def == 0 (no
+                # levels inside): values start at the page start
+                data = o
+                first = int(gresults[ent["gdesc"]][2])
+            else:
+                r = results[ent["desc"]]
+                data = o + int(r[0])
+                first = int(r[2])
             end = o + ent["size"]
         elif ent["kind"] == "v2":
             data = o + p.v2_levels_len
-            r = results[ent["desc"]]
-            first = int(r[2])
+            if ent.get("gdesc") is not None:
+                first = int(gresults[ent["gdesc"]][2])
+            else:
+                first = int(results[ent["desc"]][2])
             end = o + ent["size"]
         else:                                       # raw page in scratch
             data_abs = p.data_off
@@ -484,7 +544,7 @@ def read_table(source, columns=None):
             cols_meta = [md.row_group(rg).column(ci)
                          for rg in range(md.num_row_groups)]
             comps = {c.compression for c in cols_meta}
-            if "SNAPPY" in comps:
+            if comps & {"SNAPPY", "GZIP"}:
                 plans[ci] = ("snappy", cols_meta, max_def)
                 continue
             chunks = []

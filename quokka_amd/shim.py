@@ -126,6 +126,7 @@ _SIGS = {
     "qk_pq_rle_pages": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_pq_walk_pages": [c_vp, c_u64, c_u64, c_i64, c_vp, c_i64, c_vp],
     "qk_snappy_pages": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp],
+    "qk_gzip_pages": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp],
     "qk_str_dict_encode": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp, c_u64,
                            c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
     "qk_str_dict_rehash": [c_vp, c_u32, c_vp, c_vp, c_vp, c_vp, c_vp,
