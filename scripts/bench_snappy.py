@@ -24,6 +24,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--sf", type=float, default=3.0)
     ap.add_argument("--passes", type=int, default=3)
+    ap.add_argument("--codec", default="SNAPPY", choices=["SNAPPY", "GZIP"])
     args = ap.parse_args()
 
     li = G.gen_lineitem(args.sf, seed=5)
@@ -40,12 +41,12 @@ def main():
     t = t.cast(pa.schema([pa.field(f.name, f.type, nullable=False)
                           for f in t.schema]))
     buf = io.BytesIO()
-    pq.write_table(t, buf, compression="SNAPPY",
+    pq.write_table(t, buf, compression=args.codec,
                    use_dictionary=["l_returnflag"])
     raw = buf.getvalue()
     unc = sum(t.column(c).nbytes for c in t.column_names)
-    print("rows=%d file=%.2f GB (snappy), uncompressed columns=%.2f GB"
-          % (n, len(raw) / 1e9, unc / 1e9))
+    print("rows=%d file=%.2f GB (%s), uncompressed columns=%.2f GB"
+          % (n, len(raw) / 1e9, args.codec, unc / 1e9))
 
     cols = P.read_table(raw)          # warm (pool, hiprtc none needed)
     for c in cols.values():
