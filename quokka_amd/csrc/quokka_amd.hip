@@ -2736,7 +2736,21 @@ struct BitReader {
   __device__ void align_byte() {
     bitbuf = 0; bitcnt = 0;
   }
+  __device__ uint32_t peek(int n) {      // pads zeros past the end
+    while (bitcnt < n && p < end) {
+      bitbuf |= (uint32_t)(*p++) << bitcnt;
+      bitcnt += 8;
+    }
+    return bitbuf & ((1u << n) - 1);
+  }
+  __device__ void consume(int n) {
+    bitbuf >>= n;
+    bitcnt -= n;
+    if (bitcnt < 0) { fail = true; bitcnt = 0; }
+  }
 };
+
+#define QK_GZ_LUT_BITS 10
 
 // canonical Huffman decode tables (per length 1..15)
 struct Huff {
@@ -2778,6 +2792,33 @@ struct Huff {
     }
     return -1;
   }
+  // 10-bit direct lookup (zlib-style): entry = sym | (len << 9), 0 =
+  // escape to the bit-by-bit path (codes longer than 10 bits). The
+  // stream is LSB-first while Huffman codes are MSB-first, so each
+  // code is bit-reversed into the index space.
+  __device__ void build_lut(uint16_t *lut) {
+    for (int i = 0; i < (1 << QK_GZ_LUT_BITS); i++) lut[i] = 0;
+    for (int l = 1; l <= QK_GZ_LUT_BITS; l++)
+      for (int k = 0; k < count[l]; k++) {
+        uint32_t code = first[l] + k;
+        uint16_t sym = syms[offset[l] + k];
+        uint32_t rev = 0;
+        for (int b = 0; b < l; b++)
+          rev |= ((code >> b) & 1u) << (l - 1 - b);
+        uint16_t e = (uint16_t)(sym | (l << 9));
+        for (uint32_t idx = rev; idx < (1u << QK_GZ_LUT_BITS);
+             idx += (1u << l))
+          lut[idx] = e;
+      }
+  }
+  __device__ int decode_fast(BitReader &br, const uint16_t *lut) {
+    uint16_t e = lut[br.peek(QK_GZ_LUT_BITS)];
+    if (e) {
+      br.consume(e >> 9);
+      return br.fail ? -1 : (e & 0x1FF);
+    }
+    return decode(br);
+  }
 };
 
 __constant__ uint16_t LEN_BASE[29] = {
@@ -2802,6 +2843,8 @@ __constant__ uint8_t CLC_ORDER[19] = {
 struct QkGzLds {
   qkgz::Huff lit, dist;
   uint8_t lens[320];
+  uint16_t lut_lit[1 << QK_GZ_LUT_BITS];
+  uint16_t lut_dist[1 << QK_GZ_LUT_BITS];
 };
 
 // desc per page (8 u64): [src_off, src_len, dst_off, uncompressed_len,
@@ -2873,6 +2916,8 @@ __global__ void __launch_bounds__(64) k_gzip_pages(
         if (!L.lit.build(L.lens, 288)) { o[1] = 1; return; }
         for (int i = 0; i < 30; i++) L.lens[i] = 5;
         if (!L.dist.build(L.lens, 30)) { o[1] = 1; return; }
+        L.lit.build_lut(L.lut_lit);
+        L.dist.build_lut(L.lut_dist);
       } else {                           // dynamic tables
         uint32_t hlit = br.bits(5) + 257;
         uint32_t hdist = br.bits(5) + 1;
@@ -2907,9 +2952,11 @@ __global__ void __launch_bounds__(64) k_gzip_pages(
         }
         if (!L.lit.build(L.lens, hlit)) { o[1] = 1; return; }
         if (!L.dist.build(L.lens + hlit, hdist)) { o[1] = 1; return; }
+        L.lit.build_lut(L.lut_lit);
+        L.dist.build_lut(L.lut_dist);
       }
       for (;;) {                         // decode symbols
-        int s = L.lit.decode(br);
+        int s = L.lit.decode_fast(br, L.lut_lit);
         if (s < 0) { o[1] = 1; return; }
         if (s < 256) {
           if (w >= ocap) { o[1] = 1; return; }
@@ -2920,7 +2967,7 @@ __global__ void __launch_bounds__(64) k_gzip_pages(
           s -= 257;
           if (s >= 29) { o[1] = 1; return; }
           uint32_t len = LEN_BASE[s] + br.bits(LEN_EXTRA[s]);
-          int ds = L.dist.decode(br);
+          int ds = L.dist.decode_fast(br, L.lut_dist);
           if (ds < 0 || ds >= 30) { o[1] = 1; return; }
           uint32_t distv = DIST_BASE[ds] + br.bits(DIST_EXTRA[ds]);
           if (br.fail || distv > w || w + len > ocap) {
