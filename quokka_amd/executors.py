@@ -425,7 +425,6 @@ class GPUAggExecutor(Executor):
         # composite: each non-i64 key coded via np codebook, packed base-N
         self._key_state["mode"] = "composite"
         packed = np.zeros(len(arrs[0]), dtype=np.int64)
-        widths = []
         comps = []
         for k, arr in zip(self.groupby_keys, arrs):
             cb = self._key_state.setdefault("codebooks", {}).setdefault(k, {})
