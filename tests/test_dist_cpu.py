@@ -213,3 +213,22 @@ def test_plan_chunks_numpy_emulation_multiset():
                                for p in range(world)])
         assert np.array_equal(np.sort(recv[r]), np.sort(want))
         assert np.all(recv[r] % world == r)
+
+
+def test_plan_chunks_more_chunks_than_rows():
+    """nchunks larger than some peers' counts: zero-size chunk pieces
+    are planned consistently (conservation + layout hold)."""
+    from quokka_amd.exchange import plan_chunks
+    world, nchunks = 4, 7
+    send_counts = np.array([3, 0, 1, 5], dtype=np.uint64)
+    send_offsets = np.zeros(world + 1, dtype=np.uint64)
+    np.cumsum(send_counts, out=send_offsets[1:])
+    recv_counts = np.array([0, 2, 9, 1], dtype=np.uint64)
+    per_chunk, chunk_start, chunk_rows = plan_chunks(
+        send_offsets, send_counts, recv_counts, nchunks)
+    sc_sum = sum(c[1] for c in per_chunk)
+    rc_sum = sum(c[3] for c in per_chunk)
+    assert np.array_equal(sc_sum, send_counts)
+    assert np.array_equal(rc_sum, recv_counts)
+    assert int(chunk_rows.sum()) == int(recv_counts.sum())
+    assert int(chunk_start[0]) == 0

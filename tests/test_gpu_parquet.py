@@ -357,3 +357,15 @@ def test_gzip_lineitem_q6(gpu):
     assert got["rows_passed"] == want["rows_passed"]
     for c in dec.values():
         c.free()
+
+
+@pytest.mark.parametrize("codec", ["NONE", "SNAPPY", "GZIP"])
+def test_single_row_and_tiny_tables(gpu, codec):
+    """Degenerate sizes through every codec path."""
+    t = pa.table({"a": np.array([3.25]), "k": np.array([7], np.int64)})
+    t = t.cast(pa.schema([pa.field(f.name, f.type, nullable=False)
+                          for f in t.schema]))
+    roundtrip(gpu, t, compression=codec, use_dictionary=["k"])
+    t2 = pa.table({"x": np.arange(3, dtype=np.int64)})
+    t2 = t2.cast(pa.schema([pa.field("x", pa.int64(), nullable=False)]))
+    roundtrip(gpu, t2, compression=codec, use_dictionary=False)
