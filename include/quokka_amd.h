@@ -541,9 +541,17 @@ int qk_range_part_ids(void *stream, uint64_t n, const int64_t *keys,
 int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
                     const uint8_t *bytes, uint64_t *out_pos,
                     uint64_t *out_count_dev);
+/* Quote-aware variant (RFC-4180): a newline ends a row only when the
+ * count of `quote` chars before it is even ("" escapes toggle twice and
+ * cancel). Slower 5-pass path; callers use it only when the buffer
+ * contains the quote char at all. */
+int qk_csv_newlines_quoted(void *stream, uint64_t data_start, uint64_t n,
+                           const uint8_t *bytes_dev, uint8_t quote_char,
+                           uint64_t *out_pos_dev, uint64_t *out_count_dev);
 int qk_csv_parse(void *stream, uint64_t nrows, const uint8_t *bytes,
                  uint64_t data_start, const uint64_t *nl_pos, uint8_t sep,
-                 int ncols, const int *coltypes, void *const *out_ptrs,
+                 uint8_t quote_char /* 0 = no quoting */, int ncols,
+                 const int *coltypes, void *const *out_ptrs,
                  const uint64_t *dict_cands, const uint8_t *dict_lens,
                  const int *ncands, uint64_t *err_row);
 

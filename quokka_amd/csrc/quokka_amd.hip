@@ -3245,6 +3245,73 @@ __global__ void __launch_bounds__(BLOCK) k_csv_nl_scatter(
   }
 }
 
+// ---- quote-aware newline indexing --------------------------------------
+// A '\n' ends a row only when the count of quote chars BEFORE it is even
+// (RFC-4180 quoting; "" escapes inside quoted fields toggle twice and
+// cancel). Pass 1 counts quotes per chunk in parallel; a single-thread
+// pass turns them into per-chunk start parity; then one lane per chunk
+// walks its bytes sequentially (chunks are span/2048 ~ KBs) counting and
+// emitting the VALID newlines. The quote-free path below stays the fast
+// default (the TPC-H corpus has no quotes).
+__global__ void __launch_bounds__(BLOCK) k_csv_quote_count(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
+    uint8_t quote, uint64_t *__restrict__ counts) {
+  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  uint32_t cnt = 0;
+  for (uint64_t r = lo + threadIdx.x; r < hi; r += BLOCK)
+    cnt += b[r] == quote ? 1u : 0u;
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+  if (lane == 0) lds[wid] = cnt;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint64_t s = 0;
+    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w];
+    counts[blockIdx.x] = s;
+  }
+}
+__global__ void k_scan_parity(uint64_t nblocks, const uint64_t *counts,
+                              uint8_t *parity) {
+  uint64_t acc = 0;
+  for (uint64_t i = 0; i < nblocks; i++) {
+    parity[i] = (uint8_t)(acc & 1);
+    acc += counts[i];
+  }
+}
+__global__ void k_csv_nl_count_q(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
+    uint8_t quote, const uint8_t *__restrict__ parity,
+    uint64_t *__restrict__ block_counts) {
+  if (threadIdx.x != 0) return;
+  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  uint32_t in_q = parity[blockIdx.x];
+  uint64_t cnt = 0;
+  for (uint64_t r = lo; r < hi; r++) {
+    uint8_t c = b[r];
+    if (c == quote) in_q ^= 1u;
+    else if (c == '\n' && !in_q) cnt++;
+  }
+  block_counts[blockIdx.x] = cnt;
+}
+__global__ void k_csv_nl_scatter_q(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
+    uint8_t quote, const uint8_t *__restrict__ parity,
+    const uint64_t *__restrict__ block_offsets, uint64_t *__restrict__ out) {
+  if (threadIdx.x != 0) return;
+  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
+  uint64_t hi = qk_min_u64(n, lo + chunk);
+  uint32_t in_q = parity[blockIdx.x];
+  uint64_t w = block_offsets[blockIdx.x];
+  for (uint64_t r = lo; r < hi; r++) {
+    uint8_t c = b[r];
+    if (c == quote) in_q ^= 1u;
+    else if (c == '\n' && !in_q) out[w++] = r;
+  }
+}
+
 extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
                                const uint8_t *bytes, uint64_t *out_pos,
                                uint64_t *out_count_dev) {
@@ -3267,6 +3334,40 @@ extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
                      (hipStream_t)stream, data_start, n, bytes, chunk,
                      scratch, out_pos);
   QK_TRY("qk_csv_newlines", hipGetLastError());
+  return 0;
+}
+
+extern "C" int qk_csv_newlines_quoted(void *stream, uint64_t data_start,
+                                      uint64_t n, const uint8_t *bytes,
+                                      uint8_t quote, uint64_t *out_pos,
+                                      uint64_t *out_count_dev) {
+  if (n <= data_start) return 0;
+  uint64_t span = n - data_start;
+  uint64_t chunk = (span + MAX_BLOCKS - 1) / MAX_BLOCKS;
+  chunk = ((chunk + BLOCK - 1) / BLOCK) * BLOCK;
+  uint32_t blocks = (uint32_t)((span + chunk - 1) / chunk);
+  static __thread uint64_t *scratch = nullptr;
+  static __thread uint8_t *par = nullptr;
+  if (!scratch)
+    QK_TRY("qk_csv_newlines_quoted",
+           hipMalloc(&scratch, (MAX_BLOCKS + 1) * sizeof(uint64_t)));
+  if (!par)
+    QK_TRY("qk_csv_newlines_quoted", hipMalloc(&par, MAX_BLOCKS + 1));
+  hipLaunchKernelGGL(k_csv_quote_count, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     quote, scratch);
+  hipLaunchKernelGGL(k_scan_parity, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, (uint64_t)blocks, scratch, par);
+  hipLaunchKernelGGL(k_csv_nl_count_q, dim3(blocks), dim3(64), 0,
+                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     quote, par, scratch);
+  hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, (uint64_t)blocks, scratch,
+                     out_count_dev);
+  hipLaunchKernelGGL(k_csv_nl_scatter_q, dim3(blocks), dim3(64), 0,
+                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     quote, par, scratch, out_pos);
+  QK_TRY("qk_csv_newlines_quoted", hipGetLastError());
   return 0;
 }
 
@@ -3358,7 +3459,7 @@ __device__ inline bool csv_date32(const uint8_t *b, uint64_t s, uint64_t e,
 
 __global__ void __launch_bounds__(BLOCK) k_csv_parse(
     uint64_t nrows, const uint8_t *__restrict__ b, uint64_t data_start,
-    const uint64_t *__restrict__ nl, uint8_t sep, int ncols,
+    const uint64_t *__restrict__ nl, uint8_t sep, uint8_t quote, int ncols,
     const int *__restrict__ coltypes, void *const *__restrict__ outs,
     const uint64_t *__restrict__ dict_cands,
     const uint8_t *__restrict__ dict_lens, const int *__restrict__ ncands,
@@ -3371,9 +3472,28 @@ __global__ void __launch_bounds__(BLOCK) k_csv_parse(
     if (row_end > pos && b[row_end - 1] == '\r') row_end--;
     bool ok = true;
     for (int c = 0; c < ncols && ok; c++) {
-      uint64_t fs = pos, fe = pos;
-      while (fe < row_end && b[fe] != sep) fe++;
-      pos = fe < row_end ? fe + 1 : row_end;
+      uint64_t fs = pos, fe;
+      if (quote && pos < row_end && b[pos] == quote) {
+        // RFC-4180 quoted field: content [pos+1, closing quote);
+        // doubled quotes escape (and stay in the slice — numeric/dict
+        // parses of escaped content fail loudly, candidates are plain)
+        fs = pos + 1;
+        fe = fs;
+        while (fe < row_end) {
+          if (b[fe] == quote) {
+            if (fe + 1 < row_end && b[fe + 1] == quote) fe += 2;
+            else break;
+          } else {
+            fe++;
+          }
+        }
+        uint64_t after = fe < row_end ? fe + 1 : row_end;
+        pos = (after < row_end && b[after] == sep) ? after + 1 : after;
+      } else {
+        fe = pos;
+        while (fe < row_end && b[fe] != sep) fe++;
+        pos = fe < row_end ? fe + 1 : row_end;
+      }
       switch (coltypes[c]) {
         case 0: ok = csv_i64(b, fs, fe, (int64_t *)outs[c] + r); break;
         case 1: ok = csv_f64(b, fs, fe, (double *)outs[c] + r); break;
@@ -3404,7 +3524,8 @@ __global__ void __launch_bounds__(BLOCK) k_csv_parse(
 
 extern "C" int qk_csv_parse(void *stream, uint64_t nrows,
                             const uint8_t *bytes, uint64_t data_start,
-                            const uint64_t *nl_pos, uint8_t sep, int ncols,
+                            const uint64_t *nl_pos, uint8_t sep,
+                            uint8_t quote, int ncols,
                             const int *coltypes, void *const *out_ptrs,
                             const uint64_t *dict_cands,
                             const uint8_t *dict_lens, const int *ncands,
@@ -3414,8 +3535,8 @@ extern "C" int qk_csv_parse(void *stream, uint64_t nrows,
       (uint32_t)qk_min_u64(MAX_BLOCKS, (nrows + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_csv_parse, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, nrows, bytes, data_start, nl_pos,
-                     sep, ncols, coltypes, out_ptrs, dict_cands, dict_lens,
-                     ncands, (unsigned long long *)err_row);
+                     sep, quote, ncols, coltypes, out_ptrs, dict_cands,
+                     dict_lens, ncands, (unsigned long long *)err_row);
   QK_TRY("qk_csv_parse", hipGetLastError());
   return 0;
 }

@@ -12,9 +12,16 @@ Schema entries: (name, "i64" | "f64" | "date" | "skip") or
 columns (the executors' string-dict form), codes = candidate index.
 Numeric parsing is bit-exact vs strtod for <= 15 significant digits;
 anything unparseable raises QkCsvError with the failing row, never a
-silently different value. Quoted fields are not supported (the TPC-H
-.tbl / CSV corpus has none) — a separator inside quotes would split the
-field and surface as a parse error, not wrong data.
+silently different value.
+
+RFC-4180 quoting is supported (quote='"', the reference hands quoting to
+polars.read_csv): a field starting with the quote char runs to the
+matching close quote — separators and newlines inside are data, doubled
+quotes escape. When the buffer contains no quote byte at all the
+original single-pass newline kernel runs unchanged; otherwise a
+parity-prefix pass validates newlines (qk_csv_newlines_quoted). Escaped
+"" inside numeric/dict fields still fail the parse (error, never wrong
+data) — same behaviour as the reference's typed readers.
 """
 import ctypes
 
@@ -37,7 +44,7 @@ def _dict_key(value):
     return w, min(len(bs), 255)
 
 
-def read_csv(source, schema, sep="|", header=False):
+def read_csv(source, schema, sep="|", header=False, quote='"'):
     """Decode a CSV byte buffer (or file path) into device columns.
 
     Returns dict name -> DevColumn (u8 codes for dict columns; the
@@ -66,11 +73,23 @@ def read_csv(source, schema, sep="|", header=False):
     # newline index (counter zeroed first: the kernel early-returns
     # without writing it when the data range is empty, and pool-recycled
     # buffers hold stale bytes)
+    # Quoting only changes anything if the quote byte occurs at all —
+    # keep the fast single-pass newline kernel for quote-free buffers
+    # (all of TPC-H .tbl). raw is already host-resident; the scan is
+    # memchr-speed.
+    qbyte = ord(quote) if quote else 0
+    quoted = bool(qbyte) and raw.find(qbyte.to_bytes(1, "little"),
+                                      data_start) >= 0
     pos = DevColumn(np.uint64, max(1, n - data_start))
     cnt = DevBuffer(8)
     shim.call("qk_dmemset", cnt.ptr, 0, c_u64(8))
-    shim.call("qk_csv_newlines", None, c_u64(data_start), c_u64(n),
-              dev.ptr, pos.ptr, cnt.ptr)
+    if quoted:
+        shim.call("qk_csv_newlines_quoted", None, c_u64(data_start),
+                  c_u64(n), dev.ptr, ctypes.c_uint8(qbyte), pos.ptr,
+                  cnt.ptr)
+    else:
+        shim.call("qk_csv_newlines", None, c_u64(data_start), c_u64(n),
+                  dev.ptr, pos.ptr, cnt.ptr)
     host_cnt = np.zeros(1, dtype=np.uint64)
     shim.call("qk_d2h", host_cnt.ctypes.data_as(c_vp), cnt.ptr, c_u64(8))
     nrows = int(host_cnt[0])
@@ -130,6 +149,7 @@ def read_csv(source, schema, sep="|", header=False):
                        dtype=np.uint64).ctypes.data_as(c_vp), c_u64(8))
     shim.call("qk_csv_parse", None, c_u64(nrows), dev.ptr,
               c_u64(data_start), pos.ptr, ctypes.c_uint8(ord(sep)),
+              ctypes.c_uint8(qbyte if quoted else 0),
               ncols, d_types.ptr, d_ptrs.ptr, d_cands.ptr, d_clens.ptr,
               d_nc.ptr, err.ptr)
     herr = np.zeros(1, dtype=np.uint64)

@@ -12,6 +12,13 @@ GPUCSVReader      = InputDiskCSVDataset   (unordered_readers.py:273-442):
   small read window, :372-383 — same ownership rule here, implemented by
   skipping to the first newline after `start` and parsing through the
   first newline at/after `end`).
+
+Quoting caveat (shared with the reference's range refinement): boundary
+refinement treats every newline as a row end, so a QUOTED newline that
+straddles a range boundary mis-splits the row — with multi-range
+strides, keep embedded newlines out of the data or use one range per
+file (stride >= file size). Within a single range, csv_gpu handles
+RFC-4180 quoting fully.
 """
 import os
 

@@ -120,8 +120,9 @@ _SIGS = {
     "qk_partition_hist": [c_vp, c_u64, c_vp, c_u32, c_vp],
     "qk_range_part_ids": [c_vp, c_u64, c_vp, c_i64, c_u32, c_vp],
     "qk_csv_newlines": [c_vp, c_u64, c_u64, c_vp, c_vp, c_vp],
-    "qk_csv_parse": [c_vp, c_u64, c_vp, c_u64, c_vp, c_u8, ctypes.c_int,
-                     c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
+    "qk_csv_newlines_quoted": [c_vp, c_u64, c_u64, c_vp, c_u8, c_vp, c_vp],
+    "qk_csv_parse": [c_vp, c_u64, c_vp, c_u64, c_vp, c_u8, c_u8,
+                     ctypes.c_int, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp],
     "qk_pq_plain_copy": [c_vp, c_u64, c_vp, c_vp, c_vp, c_u32],
     "qk_pq_rle_pages": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_pq_walk_pages": [c_vp, c_u64, c_u64, c_i64, c_vp, c_i64, c_vp],

@@ -201,3 +201,119 @@ def test_empty_and_header_only_inputs(gpu):
         for c in cols.values():
             assert c.n == 0
             c.free()
+
+
+# ---- RFC-4180 quoting (the reference delegates quoting to
+# ---- polars.read_csv; here csv_gpu handles it on device) --------------
+
+def test_quoted_separators_and_newlines(gpu):
+    """Separators and newlines INSIDE quotes are data: the quoted
+    newline kernel must count 3 rows, not 5, and the parse must see the
+    unquoted content."""
+    from quokka_amd import csv_gpu
+    raw = (b'1,"2.5","A,B"\n'
+           b'"-7",3.25,"C\nD"\n'
+           b'12,"-0.5","A,B"\n')
+    cols = csv_gpu.read_csv(
+        raw, [("a", "i64"), ("b", "f64"),
+              ("s", "dict", ["A,B", "C\nD"])], sep=",")
+    try:
+        assert cols["a"].to_numpy(3).tolist() == [1, -7, 12]
+        assert cols["b"].to_numpy(3).tolist() == [2.5, 3.25, -0.5]
+        assert cols["s"].to_numpy(3).tolist() == [0, 1, 0]
+    finally:
+        for c in cols.values():
+            c.free()
+
+
+def test_quoted_escapes_and_errors(gpu):
+    from quokka_amd import csv_gpu
+    # doubled "" in a skip column is fine; quoted newline in skip too
+    cols = csv_gpu.read_csv(b'"say ""hi""\nok",5\n"x",6\n',
+                            [("junk", "skip"), ("v", "i64")], sep=",")
+    try:
+        assert cols["v"].to_numpy(2).tolist() == [5, 6]
+    finally:
+        for c in cols.values():
+            c.free()
+    # escaped quote inside a NUMERIC field: content still holds the ""
+    # bytes -> loud parse error, never a silently wrong number
+    with pytest.raises(csv_gpu.QkCsvError):
+        csv_gpu.read_csv(b'"12""3"\n', [("a", "i64")], sep=",")
+    # quote=None disables quoting: '"7"' is not an int
+    with pytest.raises(csv_gpu.QkCsvError):
+        csv_gpu.read_csv(b'"7"\n', [("a", "i64")], sep=",", quote=None)
+
+
+def test_quoted_vs_pyarrow_csv(gpu):
+    """Anchor against pyarrow.csv (independent C++ parser): pyarrow
+    WRITES with quoting-as-needed; the GPU parse of that file must
+    recover the exact source values and row count."""
+    import io
+    import pyarrow as pa
+    import pyarrow.csv as pacsv
+    from quokka_amd import csv_gpu
+    rng = np.random.default_rng(77)
+    n = 4000
+    ints = rng.integers(-1 << 40, 1 << 40, n)
+    floats = np.round(rng.uniform(-1e5, 1e5, n), 3)
+    svals = ["plain", "with,comma", "multi\nline", 'has"quote',
+             "semi;colon"]
+    codes = rng.integers(0, len(svals), n)
+    t = pa.table({"i": ints, "f": floats,
+                  "s": np.array(svals, dtype=object)[codes]})
+    buf = io.BytesIO()
+    pacsv.write_csv(t, buf)
+    raw = buf.getvalue()
+    assert raw.count(b'"') > 0          # quoting actually exercised
+    # 'has"quote' is written as "has""quote" -> its dict candidate can't
+    # match the escaped bytes; restrict the dict to the escape-free
+    # values and check those rows, skipping is exercised separately
+    cols = csv_gpu.read_csv(raw, [("i", "i64"), ("f", "f64"),
+                                  ("s", "skip")], sep=",", header=True)
+    try:
+        assert cols["i"].n == n
+        assert np.array_equal(cols["i"].to_numpy(n), ints)
+        # pyarrow writes shortest-roundtrip floats; float() reparse of
+        # the same text is bit-exact
+        want = pacsv.read_csv(io.BytesIO(raw)).column("f").to_numpy()
+        assert np.array_equal(cols["f"].to_numpy(n), want)
+    finally:
+        for c in cols.values():
+            c.free()
+    # dict path on the escape-free subset
+    keep = np.array([('"' not in s) for s in
+                     np.array(svals, dtype=object)[codes]])
+    t2 = pa.table({"s": np.array(svals, dtype=object)[codes][keep],
+                   "i": ints[keep]})
+    buf2 = io.BytesIO()
+    pacsv.write_csv(t2, buf2)
+    cands = [s for s in svals if '"' not in s]
+    cols2 = csv_gpu.read_csv(buf2.getvalue(),
+                             [("s", "dict", cands), ("i", "i64")],
+                             sep=",", header=True)
+    try:
+        m = int(keep.sum())
+        got = np.array(cands, dtype=object)[cols2["s"].to_numpy(m)]
+        assert np.array_equal(
+            got, np.array(svals, dtype=object)[codes][keep])
+        assert np.array_equal(cols2["i"].to_numpy(m), ints[keep])
+    finally:
+        for c in cols2.values():
+            c.free()
+
+
+def test_quote_free_fast_path_unchanged(gpu):
+    """A quote-free buffer must take the original single-pass kernel and
+    give identical results with quoting on or off."""
+    from quokka_amd import csv_gpu
+    raw = b"1|2.5|\n-3|0.25|\n"
+    for q in ('"', None):
+        cols = csv_gpu.read_csv(raw, [("a", "i64"), ("b", "f64")],
+                                quote=q)
+        try:
+            assert cols["a"].to_numpy(2).tolist() == [1, -3]
+            assert cols["b"].to_numpy(2).tolist() == [2.5, 0.25]
+        finally:
+            for c in cols.values():
+                c.free()
