@@ -3245,73 +3245,6 @@ __global__ void __launch_bounds__(BLOCK) k_csv_nl_scatter(
   }
 }
 
-// ---- quote-aware newline indexing --------------------------------------
-// A '\n' ends a row only when the count of quote chars BEFORE it is even
-// (RFC-4180 quoting; "" escapes inside quoted fields toggle twice and
-// cancel). Pass 1 counts quotes per chunk in parallel; a single-thread
-// pass turns them into per-chunk start parity; then one lane per chunk
-// walks its bytes sequentially (chunks are span/2048 ~ KBs) counting and
-// emitting the VALID newlines. The quote-free path below stays the fast
-// default (the TPC-H corpus has no quotes).
-__global__ void __launch_bounds__(BLOCK) k_csv_quote_count(
-    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
-    uint8_t quote, uint64_t *__restrict__ counts) {
-  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
-  uint64_t hi = qk_min_u64(n, lo + chunk);
-  uint32_t cnt = 0;
-  for (uint64_t r = lo + threadIdx.x; r < hi; r += BLOCK)
-    cnt += b[r] == quote ? 1u : 0u;
-  __shared__ uint32_t lds[BLOCK / WAVE];
-  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
-  int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
-  if (lane == 0) lds[wid] = cnt;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    uint64_t s = 0;
-    for (int w = 0; w < BLOCK / WAVE; w++) s += lds[w];
-    counts[blockIdx.x] = s;
-  }
-}
-__global__ void k_scan_parity(uint64_t nblocks, const uint64_t *counts,
-                              uint8_t *parity) {
-  uint64_t acc = 0;
-  for (uint64_t i = 0; i < nblocks; i++) {
-    parity[i] = (uint8_t)(acc & 1);
-    acc += counts[i];
-  }
-}
-__global__ void k_csv_nl_count_q(
-    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
-    uint8_t quote, const uint8_t *__restrict__ parity,
-    uint64_t *__restrict__ block_counts) {
-  if (threadIdx.x != 0) return;
-  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
-  uint64_t hi = qk_min_u64(n, lo + chunk);
-  uint32_t in_q = parity[blockIdx.x];
-  uint64_t cnt = 0;
-  for (uint64_t r = lo; r < hi; r++) {
-    uint8_t c = b[r];
-    if (c == quote) in_q ^= 1u;
-    else if (c == '\n' && !in_q) cnt++;
-  }
-  block_counts[blockIdx.x] = cnt;
-}
-__global__ void k_csv_nl_scatter_q(
-    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t chunk,
-    uint8_t quote, const uint8_t *__restrict__ parity,
-    const uint64_t *__restrict__ block_offsets, uint64_t *__restrict__ out) {
-  if (threadIdx.x != 0) return;
-  uint64_t lo = lo0 + (uint64_t)blockIdx.x * chunk;
-  uint64_t hi = qk_min_u64(n, lo + chunk);
-  uint32_t in_q = parity[blockIdx.x];
-  uint64_t w = block_offsets[blockIdx.x];
-  for (uint64_t r = lo; r < hi; r++) {
-    uint8_t c = b[r];
-    if (c == quote) in_q ^= 1u;
-    else if (c == '\n' && !in_q) out[w++] = r;
-  }
-}
-
 extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
                                const uint8_t *bytes, uint64_t *out_pos,
                                uint64_t *out_count_dev) {
@@ -3337,35 +3270,96 @@ extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
   return 0;
 }
 
+// Quote-aware newline indexing (RFC-4180): a newline is a row end only
+// when the number of quote chars before it is even. Three passes at
+// per-LANE unit granularity (~KBs each, every unit walked by its own
+// lane so hundreds of thousands run in parallel; the only sequential
+// step is the tiny prefix-parity scan over unit quote counts):
+//   1. quotes per unit  2. prefix parity  3. valid-\n count per unit
+//   4. k_scan_blocks exclusive offsets     5. scatter positions
+__global__ void k_csv_quote_count_u(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
+    uint8_t quote, uint64_t *__restrict__ counts) {
+  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t lo = lo0 + u * unit;
+  if (lo >= n) return;
+  uint64_t hi = qk_min_u64(n, lo + unit);
+  uint64_t cnt = 0;
+  for (uint64_t r = lo; r < hi; r++) cnt += b[r] == quote ? 1u : 0u;
+  counts[u] = cnt;
+}
+__global__ void k_scan_parity(uint64_t nblocks, const uint64_t *counts,
+                              uint8_t *parity) {
+  uint64_t acc = 0;
+  for (uint64_t i = 0; i < nblocks; i++) {
+    parity[i] = (uint8_t)(acc & 1);
+    acc += counts[i];
+  }
+}
+__global__ void k_csv_nl_count_qu(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
+    uint8_t quote, const uint8_t *__restrict__ parity,
+    uint64_t *__restrict__ unit_counts) {
+  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t lo = lo0 + u * unit;
+  if (lo >= n) return;
+  uint64_t hi = qk_min_u64(n, lo + unit);
+  uint32_t in_q = parity[u];
+  uint64_t cnt = 0;
+  for (uint64_t r = lo; r < hi; r++) {
+    uint8_t c = b[r];
+    if (c == quote) in_q ^= 1u;
+    else if (c == '\n' && !in_q) cnt++;
+  }
+  unit_counts[u] = cnt;
+}
+__global__ void k_csv_nl_scatter_qu(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
+    uint8_t quote, const uint8_t *__restrict__ parity,
+    const uint64_t *__restrict__ unit_offsets, uint64_t *__restrict__ out) {
+  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t lo = lo0 + u * unit;
+  if (lo >= n) return;
+  uint64_t hi = qk_min_u64(n, lo + unit);
+  uint32_t in_q = parity[u];
+  uint64_t w = unit_offsets[u];
+  for (uint64_t r = lo; r < hi; r++) {
+    uint8_t c = b[r];
+    if (c == quote) in_q ^= 1u;
+    else if (c == '\n' && !in_q) out[w++] = r;
+  }
+}
 extern "C" int qk_csv_newlines_quoted(void *stream, uint64_t data_start,
                                       uint64_t n, const uint8_t *bytes,
                                       uint8_t quote, uint64_t *out_pos,
                                       uint64_t *out_count_dev) {
   if (n <= data_start) return 0;
   uint64_t span = n - data_start;
-  uint64_t chunk = (span + MAX_BLOCKS - 1) / MAX_BLOCKS;
-  chunk = ((chunk + BLOCK - 1) / BLOCK) * BLOCK;
-  uint32_t blocks = (uint32_t)((span + chunk - 1) / chunk);
+  const uint64_t MAX_UNITS = (uint64_t)MAX_BLOCKS * BLOCK;  // 2048*256
+  uint64_t unit = (span + MAX_UNITS - 1) / MAX_UNITS;
+  unit = ((unit + 63) / 64) * 64;
+  if (unit == 0) unit = 64;
+  uint64_t units = (span + unit - 1) / unit;
+  uint32_t blocks = (uint32_t)((units + BLOCK - 1) / BLOCK);
   static __thread uint64_t *scratch = nullptr;
   static __thread uint8_t *par = nullptr;
   if (!scratch)
     QK_TRY("qk_csv_newlines_quoted",
-           hipMalloc(&scratch, (MAX_BLOCKS + 1) * sizeof(uint64_t)));
+           hipMalloc(&scratch, (MAX_UNITS + 1) * sizeof(uint64_t)));
   if (!par)
-    QK_TRY("qk_csv_newlines_quoted", hipMalloc(&par, MAX_BLOCKS + 1));
-  hipLaunchKernelGGL(k_csv_quote_count, dim3(blocks), dim3(BLOCK), 0,
-                     (hipStream_t)stream, data_start, n, bytes, chunk,
+    QK_TRY("qk_csv_newlines_quoted", hipMalloc(&par, MAX_UNITS + 1));
+  hipLaunchKernelGGL(k_csv_quote_count_u, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, unit,
                      quote, scratch);
   hipLaunchKernelGGL(k_scan_parity, dim3(1), dim3(1), 0,
-                     (hipStream_t)stream, (uint64_t)blocks, scratch, par);
-  hipLaunchKernelGGL(k_csv_nl_count_q, dim3(blocks), dim3(64), 0,
-                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     (hipStream_t)stream, units, scratch, par);
+  hipLaunchKernelGGL(k_csv_nl_count_qu, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, unit,
                      quote, par, scratch);
   hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0,
-                     (hipStream_t)stream, (uint64_t)blocks, scratch,
-                     out_count_dev);
-  hipLaunchKernelGGL(k_csv_nl_scatter_q, dim3(blocks), dim3(64), 0,
-                     (hipStream_t)stream, data_start, n, bytes, chunk,
+                     (hipStream_t)stream, units, scratch, out_count_dev);
+  hipLaunchKernelGGL(k_csv_nl_scatter_qu, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, data_start, n, bytes, unit,
                      quote, par, scratch, out_pos);
   QK_TRY("qk_csv_newlines_quoted", hipGetLastError());
   return 0;
