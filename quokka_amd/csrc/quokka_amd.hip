@@ -3270,24 +3270,15 @@ extern "C" int qk_csv_newlines(void *stream, uint64_t data_start, uint64_t n,
   return 0;
 }
 
-// Quote-aware newline indexing (RFC-4180): a newline is a row end only
-// when the number of quote chars before it is even. Three passes at
-// per-LANE unit granularity (~KBs each, every unit walked by its own
-// lane so hundreds of thousands run in parallel; the only sequential
-// step is the tiny prefix-parity scan over unit quote counts):
-//   1. quotes per unit  2. prefix parity  3. valid-\n count per unit
-//   4. k_scan_blocks exclusive offsets     5. scatter positions
-__global__ void k_csv_quote_count_u(
-    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
-    uint8_t quote, uint64_t *__restrict__ counts) {
-  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t lo = lo0 + u * unit;
-  if (lo >= n) return;
-  uint64_t hi = qk_min_u64(n, lo + unit);
-  uint64_t cnt = 0;
-  for (uint64_t r = lo; r < hi; r++) cnt += b[r] == quote ? 1u : 0u;
-  counts[u] = cnt;
-}
+// Quote-aware newline indexing (RFC-4180): a '\n' ends a row only when
+// the count of quote chars before it is even ("" escapes toggle twice
+// and cancel). One WAVE per ~24 KB unit, walked 64 bytes per step with
+// coalesced lane loads; quote parity propagates across the wave through
+// __ballot + prefix-popcount (quotes below my lane this step) and a
+// running per-step parity, so the only truly sequential work is the
+// tiny prefix scan over per-unit quote counts between passes:
+//   1. quotes per unit   2. prefix parity   3. valid-\n count per unit
+//   4. k_scan_blocks exclusive offsets      5. scatter positions
 __global__ void k_scan_parity(uint64_t nblocks, const uint64_t *counts,
                               uint8_t *parity) {
   uint64_t acc = 0;
@@ -3296,37 +3287,63 @@ __global__ void k_scan_parity(uint64_t nblocks, const uint64_t *counts,
     acc += counts[i];
   }
 }
-__global__ void k_csv_nl_count_qu(
+__global__ void __launch_bounds__(BLOCK) k_csv_quote_count_w(
     uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
-    uint8_t quote, const uint8_t *__restrict__ parity,
-    uint64_t *__restrict__ unit_counts) {
-  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t lo = lo0 + u * unit;
-  if (lo >= n) return;
-  uint64_t hi = qk_min_u64(n, lo + unit);
-  uint32_t in_q = parity[u];
-  uint64_t cnt = 0;
-  for (uint64_t r = lo; r < hi; r++) {
-    uint8_t c = b[r];
-    if (c == quote) in_q ^= 1u;
-    else if (c == '\n' && !in_q) cnt++;
-  }
-  unit_counts[u] = cnt;
+    uint64_t nunits, uint8_t quote, uint64_t *__restrict__ counts) {
+  uint64_t u = ((uint64_t)blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+  if (u >= nunits) return;
+  int lane = threadIdx.x & (WAVE - 1);
+  uint64_t lo = lo0 + u * unit, hi = qk_min_u64(n, lo + unit);
+  uint32_t cnt = 0;
+  for (uint64_t r = lo + lane; r < hi; r += WAVE)
+    cnt += b[r] == quote ? 1u : 0u;
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  if (lane == 0) counts[u] = cnt;
 }
-__global__ void k_csv_nl_scatter_qu(
+__global__ void __launch_bounds__(BLOCK) k_csv_nl_count_qw(
     uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
-    uint8_t quote, const uint8_t *__restrict__ parity,
+    uint64_t nunits, uint8_t quote, const uint8_t *__restrict__ parity,
+    uint64_t *__restrict__ unit_counts) {
+  uint64_t u = ((uint64_t)blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+  if (u >= nunits) return;
+  int lane = threadIdx.x & (WAVE - 1);
+  uint64_t lo = lo0 + u * unit, hi = qk_min_u64(n, lo + unit);
+  uint32_t in_q = parity[u];
+  uint32_t cnt = 0;
+  uint64_t below = (1ULL << lane) - 1;
+  for (uint64_t r0 = lo; r0 < hi; r0 += WAVE) {
+    uint64_t r = r0 + lane;
+    uint8_t c = r < hi ? b[r] : (uint8_t)0;
+    uint64_t qmask = __ballot(c == quote);
+    bool valid = c == '\n' &&
+                 (((in_q + __popcll(qmask & below)) & 1u) == 0u);
+    cnt += valid ? 1u : 0u;
+    in_q ^= (uint32_t)(__popcll(qmask) & 1u);
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) cnt += __shfl_down(cnt, off);
+  if (lane == 0) unit_counts[u] = cnt;
+}
+__global__ void __launch_bounds__(BLOCK) k_csv_nl_scatter_qw(
+    uint64_t lo0, uint64_t n, const uint8_t *__restrict__ b, uint64_t unit,
+    uint64_t nunits, uint8_t quote, const uint8_t *__restrict__ parity,
     const uint64_t *__restrict__ unit_offsets, uint64_t *__restrict__ out) {
-  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t lo = lo0 + u * unit;
-  if (lo >= n) return;
-  uint64_t hi = qk_min_u64(n, lo + unit);
+  uint64_t u = ((uint64_t)blockIdx.x * BLOCK + threadIdx.x) / WAVE;
+  if (u >= nunits) return;
+  int lane = threadIdx.x & (WAVE - 1);
+  uint64_t lo = lo0 + u * unit, hi = qk_min_u64(n, lo + unit);
   uint32_t in_q = parity[u];
   uint64_t w = unit_offsets[u];
-  for (uint64_t r = lo; r < hi; r++) {
-    uint8_t c = b[r];
-    if (c == quote) in_q ^= 1u;
-    else if (c == '\n' && !in_q) out[w++] = r;
+  uint64_t below = (1ULL << lane) - 1;
+  for (uint64_t r0 = lo; r0 < hi; r0 += WAVE) {
+    uint64_t r = r0 + lane;
+    uint8_t c = r < hi ? b[r] : (uint8_t)0;
+    uint64_t qmask = __ballot(c == quote);
+    bool valid = c == '\n' &&
+                 (((in_q + __popcll(qmask & below)) & 1u) == 0u);
+    uint64_t vmask = __ballot(valid);
+    if (valid) out[w + __popcll(vmask & below)] = r;
+    w += __popcll(vmask);
+    in_q ^= (uint32_t)(__popcll(qmask) & 1u);
   }
 }
 extern "C" int qk_csv_newlines_quoted(void *stream, uint64_t data_start,
@@ -3335,12 +3352,13 @@ extern "C" int qk_csv_newlines_quoted(void *stream, uint64_t data_start,
                                       uint64_t *out_count_dev) {
   if (n <= data_start) return 0;
   uint64_t span = n - data_start;
-  const uint64_t MAX_UNITS = (uint64_t)MAX_BLOCKS * BLOCK;  // 2048*256
+  const uint64_t MAX_UNITS = 16384;     // waves; 16K x 64 lanes ~ 8x CU fill
   uint64_t unit = (span + MAX_UNITS - 1) / MAX_UNITS;
-  unit = ((unit + 63) / 64) * 64;
-  if (unit == 0) unit = 64;
+  unit = ((unit + (uint64_t)WAVE - 1) / WAVE) * WAVE;
+  if (unit == 0) unit = WAVE;
   uint64_t units = (span + unit - 1) / unit;
-  uint32_t blocks = (uint32_t)((units + BLOCK - 1) / BLOCK);
+  uint32_t blocks =
+      (uint32_t)((units * WAVE + BLOCK - 1) / BLOCK);
   static __thread uint64_t *scratch = nullptr;
   static __thread uint8_t *par = nullptr;
   if (!scratch)
@@ -3348,19 +3366,19 @@ extern "C" int qk_csv_newlines_quoted(void *stream, uint64_t data_start,
            hipMalloc(&scratch, (MAX_UNITS + 1) * sizeof(uint64_t)));
   if (!par)
     QK_TRY("qk_csv_newlines_quoted", hipMalloc(&par, MAX_UNITS + 1));
-  hipLaunchKernelGGL(k_csv_quote_count_u, dim3(blocks), dim3(BLOCK), 0,
+  hipLaunchKernelGGL(k_csv_quote_count_w, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, data_start, n, bytes, unit,
-                     quote, scratch);
+                     units, quote, scratch);
   hipLaunchKernelGGL(k_scan_parity, dim3(1), dim3(1), 0,
                      (hipStream_t)stream, units, scratch, par);
-  hipLaunchKernelGGL(k_csv_nl_count_qu, dim3(blocks), dim3(BLOCK), 0,
+  hipLaunchKernelGGL(k_csv_nl_count_qw, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, data_start, n, bytes, unit,
-                     quote, par, scratch);
+                     units, quote, par, scratch);
   hipLaunchKernelGGL(k_scan_blocks, dim3(1), dim3(1), 0,
                      (hipStream_t)stream, units, scratch, out_count_dev);
-  hipLaunchKernelGGL(k_csv_nl_scatter_qu, dim3(blocks), dim3(BLOCK), 0,
+  hipLaunchKernelGGL(k_csv_nl_scatter_qw, dim3(blocks), dim3(BLOCK), 0,
                      (hipStream_t)stream, data_start, n, bytes, unit,
-                     quote, par, scratch, out_pos);
+                     units, quote, par, scratch, out_pos);
   QK_TRY("qk_csv_newlines_quoted", hipGetLastError());
   return 0;
 }
