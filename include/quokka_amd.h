@@ -111,6 +111,14 @@ int qk_gen_customer(void *stream, uint64_t n, uint64_t row_offset,
 /* Supplier columns: s_suppkey dense row+1, s_nationkey uniform 0..24. */
 int qk_gen_supplier(void *stream, uint64_t n, uint64_t row_offset,
                     uint64_t seed, int64_t *s_suppkey, int32_t *s_nationkey);
+/* Aux independent-draw column generator (device mirror of
+ * oracle/tpch_gen.py's independent columns). mode 0: out_u8 = uniform
+ * code in [0,a) (l_shipmode, tpch_gen:231). mode 1: out_u8 = bernoulli
+ * flag, P = a/1e6 (o_comment_special, tpch_gen:165). mode 2: out_f64 =
+ * uniform cents in [a,b] / 100 (c_acctbal, tpch_gen:250). */
+int qk_gen_aux(void *stream, uint64_t n, uint64_t row_offset, uint64_t seed,
+               uint64_t salt, int mode, int64_t a, int64_t b,
+               uint8_t *out_u8, double *out_f64);
 
 /* ---- TPC-H Q1: fused filter + group-by partial aggregate ------------- *
  * Replaces the map-side partial agg the reference folds into partition_fn

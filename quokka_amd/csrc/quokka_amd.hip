@@ -1742,6 +1742,41 @@ __global__ void k_gen_supplier(uint64_t n, uint64_t row_offset, uint64_t seed,
     if (s_nationkey) s_nationkey[i] = (int32_t)(h % 25);
   }
 }
+// Aux column generators for at-scale runs of the wider query shapes
+// (device mirrors of oracle/tpch_gen.py's independent draws; separate
+// hash streams per salt so they can be added to any table without
+// disturbing the existing columns' streams):
+//   uniform u8 code   — l_shipmode: rng.integers(0, 7)   (tpch_gen:231)
+//   bernoulli u8 flag — o_comment_special: p=0.019       (tpch_gen:165)
+//   uniform f64 cents — c_acctbal: U[-999.99, 9999.99]   (tpch_gen:250)
+__global__ void k_gen_aux(uint64_t n, uint64_t row_offset, uint64_t seed,
+                          uint64_t salt, int mode, int64_t a, int64_t b,
+                          uint8_t *out_u8, double *out_f64) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t row = row_offset + i;
+    uint64_t h = splitmix64((seed ^ salt) + row * 0x9E3779B97F4A7C15ULL);
+    if (mode == 0)        // uniform code in [0, a)
+      out_u8[i] = (uint8_t)(h % (uint64_t)a);
+    else if (mode == 1)   // bernoulli, P = a / 1e6
+      out_u8[i] = (h % 1000000ULL) < (uint64_t)a ? 1 : 0;
+    else                  // uniform cents in [a, b] -> dollars
+      out_f64[i] = (double)(a + (int64_t)(h % (uint64_t)(b - a + 1))) / 100.0;
+  }
+}
+extern "C" int qk_gen_aux(void *stream, uint64_t n, uint64_t row_offset,
+                          uint64_t seed, uint64_t salt, int mode, int64_t a,
+                          int64_t b, uint8_t *out_u8, double *out_f64) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_gen_aux, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, row_offset, seed, salt, mode,
+                     a, b, out_u8, out_f64);
+  QK_TRY("qk_gen_aux", hipGetLastError());
+  return 0;
+}
+
 extern "C" int qk_gen_supplier(void *stream, uint64_t n, uint64_t row_offset,
                                uint64_t seed, int64_t *s_suppkey,
                                int32_t *s_nationkey) {

@@ -76,6 +76,8 @@ _SIGS = {
                       ctypes.c_int, c_vp, c_vp, c_u64, c_vp],
     "qk_gen_orders": [c_vp, c_u64, c_u64, c_u64, c_i64, c_vp, c_vp, c_vp,
                       c_vp, c_vp, c_vp, c_i64],
+    "qk_gen_aux": [c_vp, c_u64, c_u64, c_u64, c_u64, ctypes.c_int,
+                   c_i64, c_i64, c_vp, c_vp],
     "qk_gen_customer": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp],
     "qk_gen_supplier": [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp],
     "qk_build_keyval_i32": [c_vp, c_u64, c_vp, c_vp, c_u32, c_vp, c_vp,

@@ -1526,3 +1526,41 @@ def test_groupby_extract_where_gt(gpu):
     np.testing.assert_allclose(got_s[1][og], want_s[1][m][ow], rtol=0)
     kc.free(); vc.free()
     gb.free()
+
+
+def test_gen_aux_distributions(gpu):
+    """qk_gen_aux device mirrors of the oracle's independent draws:
+    uniform shipmode codes, bernoulli comment flag, uniform acctbal
+    (tpch_gen.py:231/:165/:250) — distributional match at n=2M."""
+    from quokka_amd.shim import DevColumn, c_u64, c_i64
+    shim = gpu
+    n = 2_000_000
+    mode = DevColumn(np.uint8, n)
+    shim.call("qk_gen_aux", None, c_u64(n), c_u64(0), c_u64(42),
+              c_u64(0x5A1D), 0, c_i64(7), c_i64(0), mode.ptr, None)
+    counts = np.bincount(mode.to_numpy(n), minlength=7)
+    assert len(counts) == 7
+    assert abs(counts / n - 1 / 7).max() < 0.002       # uniform 0..6
+    flag = DevColumn(np.uint8, n)
+    shim.call("qk_gen_aux", None, c_u64(n), c_u64(0), c_u64(42),
+              c_u64(0xC033), 1, c_i64(19000), c_i64(0), flag.ptr, None)
+    fl = flag.to_numpy(n)
+    assert set(np.unique(fl)) <= {0, 1}
+    assert abs(fl.mean() - 0.019) < 0.001              # bernoulli p=.019
+    bal = DevColumn(np.float64, n)
+    shim.call("qk_gen_aux", None, c_u64(n), c_u64(0), c_u64(42),
+              c_u64(0xACC7), 2, c_i64(-99999), c_i64(999999), None,
+              bal.ptr)
+    b = bal.to_numpy(n)
+    assert b.min() >= -999.99 and b.max() <= 9999.99
+    assert abs(b.mean() - 4500.0) < 25                 # U[-999.99,9999.99]
+    assert np.allclose(b * 100, np.round(b * 100))     # exact cents
+    # determinism + row_offset consistency: second half regenerated at
+    # an offset equals the tail of the full draw
+    half = DevColumn(np.uint8, n // 2)
+    shim.call("qk_gen_aux", None, c_u64(n // 2), c_u64(n // 2), c_u64(42),
+              c_u64(0x5A1D), 0, c_i64(7), c_i64(0), half.ptr, None)
+    assert np.array_equal(half.to_numpy(n // 2),
+                          mode.to_numpy(n)[n // 2:])
+    for c in (mode, flag, bal, half):
+        c.free()
