@@ -1752,15 +1752,23 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     acnt_dev.free()
     cand_orders = cand_orders[keep]
     cand_supp = cand_supp[keep]
-    # F-status filter via the orders table (host lookup over candidates)
-    ok_h = ord_cols["o_orderkey"].to_numpy(ord_cols["o_orderkey"].n)
-    os_h = ord_cols["o_orderstatus"].to_numpy(
-        ord_cols["o_orderstatus"].n)
-    nkey = int(ok_h.max()) + 2
-    status_f = np.zeros(nkey, dtype=bool)
-    status_f[ok_h[os_h == 0]] = True
-    m = status_f[np.minimum(cand_orders, nkey - 1)]
-    wait_supp = cand_supp[m]
+    # F-status filter: device SEMI probe of the candidates against the
+    # F-status orderkeys (was a host scatter table over the SPARSE
+    # orderkey range — 180M bools + two full-order-table d2h pulls, the
+    # 4 s host tail that dominated all-22 at SF30)
+    fidx, nf = ops.filter_col(ord_cols["o_orderstatus"], ops.EQ, 0, st)
+    fkeys = ord_cols["o_orderkey"].gather(fidx, nf, st)
+    ftab = ops.JoinTable(max(16, nf), st)
+    if nf:
+        ftab.build(fkeys)
+    ckk = DevColumn.from_numpy(cand_orders)
+    spx, _, nsm = ftab.probe(ckk, mode=1, n=len(cand_orders))
+    if st:
+        st.sync()
+    wait_supp = cand_supp[spx.to_numpy(nsm)]
+    for c in (fidx, fkeys, ckk, spx):
+        c.free()
+    ftab.free()
     saudi = list(nation_names).index("SAUDI ARABIA")
     snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
     wait_supp = wait_supp[snk[wait_supp - 1] == saudi]
