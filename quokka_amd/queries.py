@@ -1277,31 +1277,40 @@ def q2(part_cols, supp_cols, ps_cols, nation_region, nation_names,
     if ns:
         stab.build(eu_keys)
     spx, _, nps = stab.probe(ps_cols["ps_suppkey"], mode=1)
-    pk = ps_cols["ps_partkey"].gather(spx, nps, st)
-    cost = ps_cols["ps_supplycost"].gather(spx, nps, st)
-    sk = ps_cols["ps_suppkey"].gather(spx, nps, st)
-    # per-part MIN cost (the correlated subquery)
-    gb = ops.GroupByI64(expected_groups=max(1024, nps), nvals=1,
+    # part filter FIRST (size == 15 and type LIKE '%BRASS' == syl3 code
+    # 0): the correlated min only matters for qualifying parts
+    # (~1/250), so a device SEMI probe against their keys shrinks the
+    # min group-by and every host pull from millions of ps rows to
+    # thousands (was an argsort-join over the full EU min set)
+    psz = part_cols["p_size"].to_numpy(part_cols["p_size"].n)
+    pty = part_cols["p_type"].to_numpy(part_cols["p_type"].n)
+    pkeys_h = part_cols["p_partkey"].to_numpy(part_cols["p_partkey"].n)
+    qual = pkeys_h[(psz == 15) & (pty % 5 == 0)]
+    ptab = ops.JoinTable(max(16, len(qual)), st)
+    qk_dev = DevColumn.from_numpy(qual.astype(np.int64))
+    if len(qual):
+        ptab.build(qk_dev)
+    pk_all = ps_cols["ps_partkey"].gather(spx, nps, st)
+    qpx, _, nq = ptab.probe(pk_all, mode=1, n=nps)
+    idx2 = spx.gather(qpx, nq, st)          # ps rows: EU AND qual part
+    pk = pk_all.gather(qpx, nq, st)
+    cost = ps_cols["ps_supplycost"].gather(idx2, nq, st)
+    sk = ps_cols["ps_suppkey"].gather(idx2, nq, st)
+    # per-part MIN cost (the correlated subquery) over the small set
+    gb = ops.GroupByI64(expected_groups=max(1024, nq), nvals=1,
                         stream=st, agg_ops=[1])
-    gb.update(pk, [cost], nps)
+    gb.update(pk, [cost], nq)
     gkeys, gmins = gb.extract()
     gb.free()
     if st:
         st.sync()
-    pk_h = pk.to_numpy(nps)
-    cost_h = cost.to_numpy(nps)
-    sk_h = sk.to_numpy(nps)
+    pk_h = pk.to_numpy(nq)
+    cost_h = cost.to_numpy(nq)
+    sk_h = sk.to_numpy(nq)
     order = np.argsort(gkeys)
     pos = np.searchsorted(gkeys, pk_h, sorter=order)
     mins = gmins[0][order[pos]]
-    # part filter: size == 15 and type LIKE '%BRASS' (syl3 code 0)
-    psz = part_cols["p_size"].to_numpy(part_cols["p_size"].n)
-    pty = part_cols["p_type"].to_numpy(part_cols["p_type"].n)
-    part_ok = np.zeros(int(part_cols["p_partkey"].n) + 2, dtype=bool)
-    pkeys_h = part_cols["p_partkey"].to_numpy(part_cols["p_partkey"].n)
-    part_ok[pkeys_h[(psz == 15) & (pty % 5 == 0)]] = True
-    win = part_ok[np.minimum(pk_h, len(part_ok) - 1)] & \
-        (cost_h == mins)
+    win = cost_h == mins
     wk, ws, wc = pk_h[win], sk_h[win], cost_h[win]
     sab = supp_cols["s_acctbal"].to_numpy(supp_cols["s_acctbal"].n)
     snk = supp_cols["s_nationkey"].to_numpy(supp_cols["s_nationkey"].n)
@@ -1316,9 +1325,11 @@ def q2(part_cols, supp_cols, ps_cols, nation_region, nation_names,
     ordr = np.lexsort((out["p_partkey"], out["s_suppkey"], nrank,
                        -out["s_acctbal"]))
     top = ordr[:limit]
-    for c in (sidx, eu_keys, spx, pk, cost, sk):
+    for c in (sidx, eu_keys, spx, qk_dev, pk_all, qpx, idx2, pk, cost,
+              sk):
         c.free()
     stab.free()
+    ptab.free()
     return {k: v[top] for k, v in out.items()}
 
 
