@@ -1734,50 +1734,51 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     gb_late = per_order(late_keys, kl, with_min_supp=True)
     all_keys.free()
     late_keys.free()
-    # candidates: late-supplier count in (0.5, 1.5) == exactly one
-    cand_k, cand_s = gb_late.extract_where_gt(0, 0.5)
-    one = cand_s[0] < 1.5
-    cand_orders = cand_k[one]
-    cand_supp = cand_s[1][one].astype(np.int64)
+    # candidate filtering stays ON DEVICE end-to-end (was: extract of
+    # ~20M-row host arrays, host masks, re-uploads): exactly-one-late
+    # == late count < 1.5 (a group exists only with count >= 1); the
+    # >= 2 all-suppliers check and the F-status check chain as device
+    # probes; only the final waitlisted suppkeys (answer-scale) come
+    # to the host for the SAUDI ARABIA cut + the bincount.
+    from .shim import DevColumnView
+    lkeys, lsums, km, lcap = gb_late.extract_device()
+    j1 = _cached_jit("f", lambda: jit.JitFilter(
+        "c < 1.5", {"c": np.dtype(np.float64)}), "q21_one")
+    i1, n1 = j1.run({"c": DevColumnView(lsums, 0, km)}, st)
+    cand_ok = lkeys.gather(i1, n1, st)              # orderkeys, i64
+    cand_sk = DevColumnView(lsums, lcap, km).gather(i1, n1, st)  # f64
     gb_late.free()
+    lkeys.free()
+    lsums.free()
+    i1.free()
     # all-pairs count >= 2 for those orders: probe the count table
     akeys, asums, kn, acap = gb_all.extract_device()
     atab = ops.JoinTable(max(16, kn), st)
     atab.build(akeys, kn)
-    cko = DevColumn.from_numpy(cand_orders)
-    px, bx, nmm = atab.probe(cko, mode=0, n=len(cand_orders))
-    acnt_dev = shim_gather_f64(asums, bx, nmm, st)
-    if st:
-        st.sync()
-    sel_rows = px.to_numpy(nmm)
-    acnt = acnt_dev.to_numpy(nmm)
-    keep = np.zeros(len(cand_orders), dtype=bool)
-    keep[sel_rows[acnt >= 2.0]] = True
+    px, bx, nmm = atab.probe(cand_ok, mode=0, n=n1)
+    acnt = shim_gather_f64(asums, bx, nmm, st)
+    j2 = _cached_jit("f", lambda: jit.JitFilter(
+        "c > 1.5", {"c": np.dtype(np.float64)}), "q21_ge2")
+    i2, n2 = j2.run({"c": acnt}, st)
+    px2 = px.gather(i2, n2, st)                     # rows into cand_*
+    ok2 = cand_ok.gather(px2, n2, st)
+    sk2 = cand_sk.gather(px2, n2, st)
     gb_all.free()
-    akeys.free()
-    asums.free()
+    for c in (akeys, asums, px, bx, acnt, i2, px2, cand_ok, cand_sk):
+        c.free()
     atab.free()
-    cko.free()
-    px.free()
-    bx.free()
-    acnt_dev.free()
-    cand_orders = cand_orders[keep]
-    cand_supp = cand_supp[keep]
-    # F-status filter: device SEMI probe of the candidates against the
-    # F-status orderkeys (was a host scatter table over the SPARSE
-    # orderkey range — 180M bools + two full-order-table d2h pulls, the
-    # 4 s host tail that dominated all-22 at SF30)
+    # F-status filter: device SEMI probe against the F-status orderkeys
     fidx, nf = ops.filter_col(ord_cols["o_orderstatus"], ops.EQ, 0, st)
     fkeys = ord_cols["o_orderkey"].gather(fidx, nf, st)
     ftab = ops.JoinTable(max(16, nf), st)
     if nf:
         ftab.build(fkeys)
-    ckk = DevColumn.from_numpy(cand_orders)
-    spx, _, nsm = ftab.probe(ckk, mode=1, n=len(cand_orders))
+    spx, _, nsm = ftab.probe(ok2, mode=1, n=n2)
+    wait_dev = sk2.gather(spx, nsm, st)
     if st:
         st.sync()
-    wait_supp = cand_supp[spx.to_numpy(nsm)]
-    for c in (fidx, fkeys, ckk, spx):
+    wait_supp = wait_dev.to_numpy(nsm).astype(np.int64)
+    for c in (fidx, fkeys, spx, ok2, sk2, wait_dev):
         c.free()
     ftab.free()
     saudi = list(nation_names).index("SAUDI ARABIA")
