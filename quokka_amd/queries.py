@@ -813,13 +813,14 @@ def q10(li_cols, ord_cols, cust_cols, cust_host, nation_names,
     # sort over millions of customers: argpartition candidates, then an
     # exact sort over every row at/above the candidate cutoff (fp64
     # ties kept, same scheme as _topk)
-    rev = out["revenue"]
-    if len(rev) > 4 * limit + 64:
-        cand = np.argpartition(-rev, 2 * limit)[: 2 * limit]
-        cand = np.nonzero(rev >= rev[cand].min())[0]
+    revh = out["revenue"]
+    if len(revh) > 4 * limit + 64:
+        cand = np.argpartition(-revh, 2 * limit)[: 2 * limit]
+        cand = np.nonzero(revh >= revh[cand].min())[0]
     else:
-        cand = np.arange(len(rev))
-    top = cand[np.lexsort((out["c_custkey"][cand], -rev[cand]))][:limit]
+        cand = np.arange(len(revh))
+    top = cand[np.lexsort((out["c_custkey"][cand],
+                           -revh[cand]))][:limit]
     out = {k: v[top] for k, v in out.items()}
     row = out["c_custkey"].astype(np.int64) - 1       # dense custkey
     # attach numeric attrs from device customer columns, strings host-side
