@@ -159,6 +159,9 @@ int qk_filter_i32(void *stream, uint64_t n, const int32_t *col, int op,
                   int32_t value, uint32_t *out_idx, uint64_t *out_count_dev);
 int qk_filter_u8(void *stream, uint64_t n, const uint8_t *col, int op,
                  uint8_t value, uint32_t *out_idx, uint64_t *out_count_dev);
+/* f64 variant; threshold is a kernel argument (no JIT literal). */
+int qk_filter_f64(void *stream, uint64_t n, const double *col, int op,
+                  double value, uint32_t *out_idx, uint64_t *out_count_dev);
 
 /* elementwise revenue: out[i] = a[i] * (1 - b[i]) (the per-row product the
  * reference computes before summing, apps/tpc-h/tpch.py:151) */

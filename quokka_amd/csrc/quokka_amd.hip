@@ -686,6 +686,15 @@ extern "C" int qk_filter_u8(void *stream, uint64_t n, const uint8_t *col,
   return qk_filter_impl("qk_filter_u8", stream, n, col, op, value, out_idx,
                         out_count_dev);
 }
+// f64 variant: the threshold is a KERNEL ARGUMENT, so data-dependent
+// cuts (e.g. Q22's "c_acctbal > avg" with avg computed on device) do
+// not force a per-value hiprtc recompile the way a JIT literal does.
+extern "C" int qk_filter_f64(void *stream, uint64_t n, const double *col,
+                             int op, double value, uint32_t *out_idx,
+                             uint64_t *out_count_dev) {
+  return qk_filter_impl("qk_filter_f64", stream, n, col, op, value, out_idx,
+                        out_count_dev);
+}
 
 // ---- elementwise revenue ----------------------------------------------
 __global__ void k_mul_1md(uint64_t n, const double *__restrict__ a,

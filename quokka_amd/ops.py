@@ -41,6 +41,9 @@ def filter_col(col, op, value, stream=None):
     elif col.dtype == np.dtype(np.uint8):
         shim.call("qk_filter_u8", sh, c_u64(col.n), col.ptr, op,
                   ctypes.c_uint8(int(value)), idx.ptr, cnt.ptr)
+    elif col.dtype == np.dtype(np.float64):
+        shim.call("qk_filter_f64", sh, c_u64(col.n), col.ptr, op,
+                  ctypes.c_double(float(value)), idx.ptr, cnt.ptr)
     else:
         raise TypeError("filter_col: unsupported dtype %s" % col.dtype)
     if stream:

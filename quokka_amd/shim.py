@@ -66,6 +66,8 @@ _SIGS = {
                   c_f64, c_f64, c_f64, c_vp],
     "qk_filter_i32": [c_vp, c_u64, c_vp, ctypes.c_int, c_i32, c_vp, c_vp],
     "qk_filter_u8": [c_vp, c_u64, c_vp, ctypes.c_int, c_u8, c_vp, c_vp],
+    "qk_filter_f64": [c_vp, c_u64, c_vp, ctypes.c_int, c_f64, c_vp,
+                      c_vp],
     "qk_mul_1md": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_gather_i64": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_gather_f64": [c_vp, c_u64, c_vp, c_vp, c_vp],

@@ -1564,3 +1564,24 @@ def test_gen_aux_distributions(gpu):
                           mode.to_numpy(n)[n // 2:])
     for c in (mode, flag, bal, half):
         c.free()
+
+
+def test_filter_f64_threshold(gpu):
+    """qk_filter_f64: runtime-threshold f64 compaction (Q22's '> avg'
+    cut) — row set and order vs numpy on all six ops."""
+    from quokka_amd import ops
+    from quokka_amd.shim import DevColumn
+    rng = np.random.default_rng(31)
+    a = np.round(rng.uniform(-100, 100, 100_000), 2)
+    a[rng.integers(0, len(a), 500)] = 3.5          # exact-equality hits
+    col = DevColumn.from_numpy(a)
+    npop = {ops.LT: np.less, ops.LE: np.less_equal, ops.GT: np.greater,
+            ops.GE: np.greater_equal, ops.EQ: np.equal,
+            ops.NE: np.not_equal}
+    for op, f in npop.items():
+        idx, n = ops.filter_col(col, op, 3.5)
+        want = np.nonzero(f(a, 3.5))[0]
+        assert n == len(want)
+        assert np.array_equal(idx.to_numpy(n), want)
+        idx.free()
+    col.free()
