@@ -1126,45 +1126,47 @@ def q17(li_cols, part_cols, brand_code=12, container_code=17,
     part_cols: p_partkey i64, p_brand u8, p_container u8."""
     from . import jit, ops
     st = stream
-    n = li_cols["l_partkey"].n
-    # per-part avg over the WHOLE lineitem: sum + count on the device
-    # group-by (the correlated subquery's aggregate)
-    ones = DevColumn(np.float64, max(1, n))
-    call("qk_fill_f64", st.handle if st else None, ones.ptr,
-         ctypes.c_double(1.0), c_u64(n))
-    gb = ops.GroupByI64(expected_groups=max(1024, n // 4), nvals=2,
-                        stream=st)
-    gb.update(li_cols["l_partkey"], [li_cols["l_quantity"], ones], n)
-    pk_all, sums = gb.extract()
-    gb.free()
-    ones.free()
-    thr_all = 0.2 * sums[0] / np.maximum(sums[1], 1.0)
-    # qualifying parts (brand AND container) -> join table + aligned
-    # threshold column
+    # qualifying parts FIRST (brand AND container, ~1/1000 of parts):
+    # the correlated per-part avg only matters for them, so the
+    # group-by runs over their lines alone (was: a 6M-group extract +
+    # host argsort-join over every part, the query's dominant cost)
     bidx0, nb = ops.filter_col(part_cols["p_brand"], ops.EQ, brand_code,
                                st)
     bkeys = part_cols["p_partkey"].gather(bidx0, nb, st)
     bcont = part_cols["p_container"].gather(bidx0, nb, st)
     cidx0, ncp = ops.filter_col(bcont, ops.EQ, container_code, st)
     qkeys_col = bkeys.gather(cidx0, ncp, st)
+    ptab = ops.JoinTable(max(16, ncp), st)
+    if ncp:
+        ptab.build(qkeys_col, ncp)
+    pidx, bidx, nm = ptab.probe(li_cols["l_partkey"], mode=0)
+    m_qty = li_cols["l_quantity"].gather(pidx, nm, st)
+    m_pr = li_cols["l_extendedprice"].gather(pidx, nm, st)
+    m_pk = li_cols["l_partkey"].gather(pidx, nm, st)
+    # per-part avg over those lines (the subquery has no other filter,
+    # so the probe's line set IS the avg's domain): sum + count
+    ones = DevColumn(np.float64, max(1, nm))
+    call("qk_fill_f64", st.handle if st else None, ones.ptr,
+         ctypes.c_double(1.0), c_u64(nm))
+    gb = ops.GroupByI64(expected_groups=max(1024, ncp), nvals=2,
+                        stream=st)
+    gb.update(m_pk, [m_qty, ones], nm)
+    pk_all, sums = gb.extract()
+    gb.free()
+    ones.free()
+    thr_all = 0.2 * sums[0] / np.maximum(sums[1], 1.0)
     if st:
         st.sync()
+    # align thresholds to the BUILD rows (parts with no lines keep 0 —
+    # they also have no probe matches)
     qkeys = qkeys_col.to_numpy(ncp)
-    # threshold per qualifying part via searchsorted over the group keys
     order = np.argsort(pk_all)
     pos = np.searchsorted(pk_all, qkeys, sorter=order)
     have = (pos < len(pk_all)) & \
         (pk_all[order[np.minimum(pos, len(pk_all) - 1)]] == qkeys)
-    qkeys = qkeys[have]
-    qthr = thr_all[order[pos[have]]]
-    ptab = ops.JoinTable(max(16, len(qkeys)), st)
-    dthr = DevColumn.from_numpy(qthr)
-    dkeys = DevColumn.from_numpy(qkeys)
-    if len(qkeys):
-        ptab.build(dkeys)
-    pidx, bidx, nm = ptab.probe(li_cols["l_partkey"], mode=0)
-    m_qty = li_cols["l_quantity"].gather(pidx, nm, st)
-    m_pr = li_cols["l_extendedprice"].gather(pidx, nm, st)
+    thr_aligned = np.zeros(max(1, ncp))
+    thr_aligned[have] = thr_all[order[pos[have]]]
+    dthr = DevColumn.from_numpy(thr_aligned)
     m_thr = dthr.gather(bidx, nm, st)
     fin = {"l_quantity": m_qty, "thr": m_thr, "l_extendedprice": m_pr}
     fsch = {k: np.dtype(np.float64) for k in fin}
@@ -1177,8 +1179,8 @@ def q17(li_cols, part_cols, brand_code=12, container_code=17,
     if st:
         st.sync()
     out = float(agg.read(acc)[0, 0] / 7.0)
-    for c in (bidx0, bkeys, bcont, cidx0, qkeys_col, dthr, dkeys, pidx,
-              bidx, m_qty, m_pr, m_thr):
+    for c in (bidx0, bkeys, bcont, cidx0, qkeys_col, dthr, pidx,
+              bidx, m_qty, m_pr, m_pk, m_thr):
         c.free()
     acc.free()
     ptab.free()
