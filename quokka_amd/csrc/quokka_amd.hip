@@ -1918,16 +1918,31 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_extract(
     int nvals, uint64_t cap, int64_t *__restrict__ out_keys,
     double *__restrict__ out_sums, uint64_t out_cap,
     uint64_t *__restrict__ cursor) {
+  // wave-aggregated output cursor: one atomicAdd per wave per stride
+  // step instead of one per non-empty slot (extracting ~180M groups
+  // serialized ~90 ms on same-address L2 atomics before this)
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  int lane = (int)(threadIdx.x & (WAVE - 1));
   for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
        s += stride) {
     int64_t key = slot_keys[s];
-    if (key == QK_JOIN_EMPTY) continue;
-    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
-    if (pos < out_cap) {
-      out_keys[pos] = key;
-      for (int c = 0; c < nvals; c++)
-        out_sums[(uint64_t)c * out_cap + pos] = slot_sums[(uint64_t)c * cap + s];
+    bool has = key != QK_JOIN_EMPTY;
+    uint64_t mask = __ballot(has);
+    if (!mask) continue;
+    int src = __ffsll((unsigned long long)mask) - 1;
+    unsigned long long base = 0;
+    if (lane == src)
+      base = atomicAdd((unsigned long long *)cursor,
+                       (unsigned long long)__popcll(mask));
+    base = __shfl(base, src);
+    if (has) {
+      uint64_t pos = base + __popcll(mask & ((1ULL << lane) - 1));
+      if (pos < out_cap) {
+        out_keys[pos] = key;
+        for (int c = 0; c < nvals; c++)
+          out_sums[(uint64_t)c * out_cap + pos] =
+              slot_sums[(uint64_t)c * cap + s];
+      }
     }
   }
 }
@@ -1953,17 +1968,28 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_extract_gt(
     int64_t *__restrict__ out_keys, double *__restrict__ out_sums,
     uint64_t out_cap, uint64_t *__restrict__ cursor) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  int lane = (int)(threadIdx.x & (WAVE - 1));
   for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
        s += stride) {
     int64_t key = slot_keys[s];
-    if (key == QK_JOIN_EMPTY) continue;
-    if (!(slot_sums[(uint64_t)col * cap + s] > threshold)) continue;
-    uint64_t pos = atomicAdd((unsigned long long *)cursor, 1ULL);
-    if (pos < out_cap) {
-      out_keys[pos] = key;
-      for (int c = 0; c < nvals; c++)
-        out_sums[(uint64_t)c * out_cap + pos] =
-            slot_sums[(uint64_t)c * cap + s];
+    bool has = key != QK_JOIN_EMPTY &&
+               slot_sums[(uint64_t)col * cap + s] > threshold;
+    uint64_t mask = __ballot(has);
+    if (!mask) continue;
+    int src = __ffsll((unsigned long long)mask) - 1;
+    unsigned long long base = 0;
+    if (lane == src)
+      base = atomicAdd((unsigned long long *)cursor,
+                       (unsigned long long)__popcll(mask));
+    base = __shfl(base, src);
+    if (has) {
+      uint64_t pos = base + __popcll(mask & ((1ULL << lane) - 1));
+      if (pos < out_cap) {
+        out_keys[pos] = key;
+        for (int c = 0; c < nvals; c++)
+          out_sums[(uint64_t)c * out_cap + pos] =
+              slot_sums[(uint64_t)c * cap + s];
+      }
     }
   }
 }
