@@ -165,6 +165,8 @@ int qk_filter_f64(void *stream, uint64_t n, const double *col, int op,
 
 /* elementwise revenue: out[i] = a[i] * (1 - b[i]) (the per-row product the
  * reference computes before summing, apps/tpc-h/tpch.py:151) */
+int qk_flag_gt_i32(void *stream, uint64_t n, const int32_t *a,
+                   const int32_t *b, double *out); /* out=a>b?1:0 (Q21) */
 int qk_mul_1md(void *stream, uint64_t n, const double *a, const double *b,
                double *out);
 

@@ -696,6 +696,26 @@ extern "C" int qk_filter_f64(void *stream, uint64_t n, const double *col,
                         out_count_dev);
 }
 
+// elementwise comparison flag (a > b -> 1.0) — feeds MAX-aggregated
+// "any late line for this pair" dedup in Q21 (exists-subquery shape)
+__global__ void k_flag_gt_i32(uint64_t n, const int32_t *__restrict__ a,
+                              const int32_t *__restrict__ b,
+                              double *__restrict__ out) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = a[i] > b[i] ? 1.0 : 0.0;
+}
+extern "C" int qk_flag_gt_i32(void *stream, uint64_t n, const int32_t *a,
+                              const int32_t *b, double *out) {
+  if (!n) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_flag_gt_i32, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, n, a, b, out);
+  QK_TRY("qk_flag_gt_i32", hipGetLastError());
+  return 0;
+}
+
 // ---- elementwise revenue ----------------------------------------------
 __global__ void k_mul_1md(uint64_t n, const double *__restrict__ a,
                           const double *__restrict__ b,

@@ -1668,128 +1668,75 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
         stream=None):
     """Device Q21: the exists/not-exists pair reduces to per-order
     DISTINCT supplier counts. Entirely device-side until the final
-    (small) candidate set: (1) dedupe (orderkey*S + suppkey) pairs on
-    the group-by table, (2) shift the EXTRACTED device keys back to
-    orderkeys (qk_i64_shr) and run a SECOND group-by — count for all
-    lines; count + MIN(suppkey) for LATE lines, (3) qualifying orders
-    (F status, >=2 suppliers, exactly 1 late) resolve by probing the
-    late-order candidates against the all-pairs count table and the
-    orders table; only the candidates' few columns come to the host.
-    Returns dict s_suppkey -> numwait."""
+    (answer-scale) candidate set: (1) ONE dedup group-by over the
+    (orderkey*S + suppkey) pair keys carrying MAX(late flag) — a pair
+    is late if ANY of its lines has receipt > commit; (2) shift the
+    EXTRACTED device keys back to orderkeys (qk_i64_shr) and run ONE
+    second-level group-by with three values: COUNT(pairs),
+    SUM(late flag) and MIN(late ? suppkey : BIG); (3) qualifying
+    orders (>= 2 suppliers, exactly 1 late) fall out of a single JIT
+    filter over the extract's value views, and the F-status check is a
+    device SEMI probe. Only the waitlisted suppkeys reach the host for
+    the SAUDI ARABIA cut + bincount. Returns dict s_suppkey ->
+    numwait."""
     from . import jit, ops
-    from .shim import c_i64
+    from .shim import c_i64, DevColumnView
     st = stream
     sh = st.handle if st else None
     S = 1 << (int(supp_cols["s_suppkey"].n) + 1).bit_length()
     sbits = S.bit_length() - 1          # pow2 scale: decompose = shift/mask
     n = li_cols["l_orderkey"].n
 
-    def pair_keys_device(idx=None, nn=None):
-        """Dedupe pairs; return (device keys DevColumn, count)."""
-        if idx is None:
-            kx, ky, cnt = li_cols["l_orderkey"], li_cols["l_suppkey"], n
-        else:
-            kx = li_cols["l_orderkey"].gather(idx, nn, st)
-            ky = li_cols["l_suppkey"].gather(idx, nn, st)
-            cnt = nn
-        ck = DevColumn(np.int64, max(1, cnt))
-        call("qk_i64_combine", sh, c_u64(cnt), kx.ptr, ky.ptr, c_i64(S),
-             ck.ptr)
-        ck.n = cnt
-        ones = DevColumn(np.float64, max(1, cnt))
-        call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0),
-             c_u64(cnt))
-        gb = ops.GroupByI64(expected_groups=max(1024, cnt), nvals=1,
-                            stream=st)
-        gb.update(ck, [ones], cnt)
-        keys_dev, sums_dev, k, _ = gb.extract_device()
-        sums_dev.free()
-        gb.free()
-        ck.free()
-        ones.free()
-        if idx is not None:
-            kx.free()
-            ky.free()
-        return keys_dev, k
-
-    def per_order(keys_dev, k, with_min_supp):
-        """Second-level device group-by keyed by orderkey."""
-        ok = DevColumn(np.int64, max(1, k))
-        call("qk_i64_shr", sh, c_u64(k), keys_dev.ptr, sbits, ok.ptr)
-        ok.n = k
-        ones = DevColumn(np.float64, max(1, k))
-        call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0), c_u64(k))
-        vals = [ones]
-        aggs = [0]
-        sk_f = None
-        if with_min_supp:
-            # suppkey = pairkey - (orderkey << sbits), as f64 for the
-            # MIN aggregate (exact: suppkey < 2^53)
-            sk_i = DevColumn(np.int64, max(1, k))
-            call("qk_i64_combine", sh, c_u64(k), ok.ptr, keys_dev.ptr,
-                 c_i64(-S), sk_i.ptr)
-            sk_i.n = k
-            m = _cached_jit("m", lambda: jit.JitMap(
-                "sk", {"sk": np.dtype(np.int64)}), "q21_sk_f64")
-            sk_f = m.run({"sk": sk_i}, st)
-            sk_i.free()
-            vals.append(sk_f)
-            aggs.append(1)
-        gb = ops.GroupByI64(expected_groups=max(1024, k),
-                            nvals=len(vals), stream=st, agg_ops=aggs)
-        gb.update(ok, vals, k)
-        ok.free()
-        ones.free()
-        if sk_f is not None:
-            sk_f.free()
-        return gb
-
-    all_keys, ka = pair_keys_device()
-    lsch = {k_: v.dtype for k_, v in li_cols.items()}
-    lf = _cached_jit("f", lambda: jit.JitFilter(
-        "l_receiptdate > l_commitdate", lsch),
-        "q21_late", _schema_key(lsch))
-    lidx, nl = lf.run(li_cols, st)
-    late_keys, kl = pair_keys_device(lidx, nl)
-    lidx.free()
-
-    gb_all = per_order(all_keys, ka, with_min_supp=False)
-    gb_late = per_order(late_keys, kl, with_min_supp=True)
-    all_keys.free()
-    late_keys.free()
-    # candidate filtering stays ON DEVICE end-to-end (was: extract of
-    # ~20M-row host arrays, host masks, re-uploads): exactly-one-late
-    # == late count < 1.5 (a group exists only with count >= 1); the
-    # >= 2 all-suppliers check and the F-status check chain as device
-    # probes; only the final waitlisted suppkeys (answer-scale) come
-    # to the host for the SAUDI ARABIA cut + the bincount.
-    from .shim import DevColumnView
-    lkeys, lsums, km, lcap = gb_late.extract_device()
-    j1 = _cached_jit("f", lambda: jit.JitFilter(
-        "c < 1.5", {"c": np.dtype(np.float64)}), "q21_one")
-    i1, n1 = j1.run({"c": DevColumnView(lsums, 0, km)}, st)
-    cand_ok = lkeys.gather(i1, n1, st)              # orderkeys, i64
-    cand_sk = DevColumnView(lsums, lcap, km).gather(i1, n1, st)  # f64
-    gb_late.free()
-    lkeys.free()
-    lsums.free()
-    i1.free()
-    # all-pairs count >= 2 for those orders: probe the count table
-    akeys, asums, kn, acap = gb_all.extract_device()
-    atab = ops.JoinTable(max(16, kn), st)
-    atab.build(akeys, kn)
-    px, bx, nmm = atab.probe(cand_ok, mode=0, n=n1)
-    acnt = shim_gather_f64(asums, bx, nmm, st)
-    j2 = _cached_jit("f", lambda: jit.JitFilter(
-        "c > 1.5", {"c": np.dtype(np.float64)}), "q21_ge2")
-    i2, n2 = j2.run({"c": acnt}, st)
-    px2 = px.gather(i2, n2, st)                     # rows into cand_*
-    ok2 = cand_ok.gather(px2, n2, st)
-    sk2 = cand_sk.gather(px2, n2, st)
-    gb_all.free()
-    for c in (akeys, asums, px, bx, acnt, i2, px2, cand_ok, cand_sk):
+    # pass 1: dedupe pairs, keeping MAX(late flag) per pair
+    flag = DevColumn(np.float64, max(1, n))
+    call("qk_flag_gt_i32", sh, c_u64(n), li_cols["l_receiptdate"].ptr,
+         li_cols["l_commitdate"].ptr, flag.ptr)
+    ck = DevColumn(np.int64, max(1, n))
+    call("qk_i64_combine", sh, c_u64(n), li_cols["l_orderkey"].ptr,
+         li_cols["l_suppkey"].ptr, c_i64(S), ck.ptr)
+    ck.n = n
+    gb1 = ops.GroupByI64(expected_groups=max(1024, n), nvals=1,
+                         stream=st, agg_ops=[2])
+    gb1.update(ck, [flag], n)
+    ck.free()
+    flag.free()
+    pkeys, psums, kp, _pcap = gb1.extract_device()
+    gb1.free()
+    # pass 2: per order — count, late count, MIN(late ? suppkey : BIG)
+    ok1 = DevColumn(np.int64, max(1, kp))
+    call("qk_i64_shr", sh, c_u64(kp), pkeys.ptr, sbits, ok1.ptr)
+    ok1.n = kp
+    sk_i = DevColumn(np.int64, max(1, kp))
+    call("qk_i64_combine", sh, c_u64(kp), ok1.ptr, pkeys.ptr, c_i64(-S),
+         sk_i.ptr)
+    sk_i.n = kp
+    plate = DevColumnView(psums, 0, kp)
+    msel = _cached_jit("m", lambda: jit.JitMap(
+        "f * sk + (1 - f) * 1000000000",
+        {"f": np.dtype(np.float64), "sk": np.dtype(np.int64)}),
+        "q21_minsel").run({"f": plate, "sk": sk_i}, st)
+    ones = DevColumn(np.float64, max(1, kp))
+    call("qk_fill_f64", sh, ones.ptr, ctypes.c_double(1.0), c_u64(kp))
+    gb2 = ops.GroupByI64(expected_groups=max(1024, kp), nvals=3,
+                         stream=st, agg_ops=[0, 0, 1])
+    gb2.update(ok1, [ones, plate, msel], kp)
+    for c in (pkeys, psums, ok1, sk_i, msel, ones):
         c.free()
-    atab.free()
+    okeys, osums, ko, ocap = gb2.extract_device()
+    gb2.free()
+    # qualify: >= 2 distinct suppliers AND exactly one late (counts are
+    # integral f64, so > 1.5 / the (0.5, 1.5) window are exact)
+    jq = _cached_jit("f", lambda: jit.JitFilter(
+        "ac > 1.5 and lc > 0.5 and lc < 1.5",
+        {"ac": np.dtype(np.float64), "lc": np.dtype(np.float64)}),
+        "q21_qual")
+    i1, n1 = jq.run({"ac": DevColumnView(osums, 0, ko),
+                     "lc": DevColumnView(osums, ocap, ko)}, st)
+    cand_ok = okeys.gather(i1, n1, st)              # orderkeys, i64
+    cand_sk = DevColumnView(osums, 2 * ocap, ko).gather(i1, n1, st)
+    ok2, sk2, n2 = cand_ok, cand_sk, n1
+    for c in (okeys, osums, i1):
+        c.free()
     # F-status filter: device SEMI probe against the F-status orderkeys
     fidx, nf = ops.filter_col(ord_cols["o_orderstatus"], ops.EQ, 0, st)
     fkeys = ord_cols["o_orderkey"].gather(fidx, nf, st)

@@ -68,6 +68,7 @@ _SIGS = {
     "qk_filter_u8": [c_vp, c_u64, c_vp, ctypes.c_int, c_u8, c_vp, c_vp],
     "qk_filter_f64": [c_vp, c_u64, c_vp, ctypes.c_int, c_f64, c_vp,
                       c_vp],
+    "qk_flag_gt_i32": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_mul_1md": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_gather_i64": [c_vp, c_u64, c_vp, c_vp, c_vp],
     "qk_gather_f64": [c_vp, c_u64, c_vp, c_vp, c_vp],
