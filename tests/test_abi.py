@@ -188,3 +188,27 @@ def test_jit_filter_col_cap_clean_error():
     pred = " and ".join("c%d > 0" % i for i in range(10))
     with _p.raises(shim.QkError, match="ncols"):
         jit.JitFilter(pred, schema)
+
+
+def test_jit_translate_r02_tail_expressions():
+    """The late-r02 query-tail expressions translate as written in
+    queries.py (pure-Python translator, no GPU): the Q21 arithmetic
+    MIN-select over mixed f64/i64 columns and the multi-bound f64
+    qualification filter."""
+    import numpy as np
+    from quokka_amd import jit
+    e, cols = jit.translate_arith(
+        "f * sk + (1 - f) * 1000000000",
+        {"f": np.dtype(np.float64), "sk": np.dtype(np.int64)})
+    assert set(cols) == {"f", "sk"}
+    # C expression references both columns and keeps the select shape
+    assert "*" in e and "+" in e
+    e2, cols2 = jit.translate(
+        "ac > 1.5 and lc > 0.5 and lc < 1.5",
+        {"ac": np.dtype(np.float64), "lc": np.dtype(np.float64)})
+    assert set(cols2) == {"ac", "lc"}
+    assert "&&" in e2
+    # comparisons are rejected by the ARITH entry (JitMap contract)
+    import pytest as _pytest
+    with _pytest.raises(Exception):
+        jit.translate_arith("a > 1", {"a": np.dtype(np.float64)})
