@@ -13,9 +13,6 @@ from oracle import tpch_gen as G, queries as OQ
 
 pytestmark = pytest.mark.gpu
 
-SF = 0.05
-
-
 @pytest.fixture(scope="module")
 def gpu():
     from quokka_amd import shim
@@ -23,10 +20,15 @@ def gpu():
     return shim
 
 
-@pytest.fixture(scope="module")
-def data():
-    d = G.gen_all(SF, 42)
-    d["customer_s"] = G.gen_customer(SF, 42, strings=True)
+# Two (SF, seed) points, mirroring tests/test_oracle_acero.py: the
+# device pipelines must agree with the oracle at a second seed and
+# size too, not just the one the suite was developed against.
+@pytest.fixture(scope="module", params=[(0.05, 42), (0.08, 7)],
+                ids=["sf.05-seed42", "sf.08-seed7"])
+def data(request):
+    sf, seed = request.param
+    d = G.gen_all(sf, seed)
+    d["customer_s"] = G.gen_customer(sf, seed, strings=True)
     return d
 
 
