@@ -1760,16 +1760,6 @@ def q21(li_cols, ord_cols, supp_cols, nation_names, limit=100,
     return {int(sk[i]): int(cnt[sk[i]]) for i in order}
 
 
-def shim_gather_f64(col_f64, idx_col, n_idx, stream=None):
-    """Gather over a raw f64 DevColumn region (first n entries used)."""
-    out = DevColumn(np.float64, max(1, n_idx))
-    sh = stream.handle if stream else None
-    call("qk_gather_f64", sh, c_u64(n_idx), idx_col.ptr, col_f64.ptr,
-         out.ptr)
-    out.n = n_idx
-    return out
-
-
 def q22(cust_cols, ord_cols, stream=None):
     """Device Q22: the country-code membership + positive-balance
     average run as one JIT grand aggregate; customers with no orders
