@@ -323,25 +323,30 @@ int qk_q5_probe_agg_nt(void *stream, uint64_t n, const int64_t *l_orderkey,
  * before cumulative distinct keys approach capacity (the find-or-insert
  * probe loop would otherwise spin forever on a full table). */
 int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
-                       const double *const *vals_dev,
-                       const int32_t *agg_ops_dev, int nvals,
-                       int64_t *slot_keys, double *slot_sums,
-                       uint64_t capacity, uint64_t *n_inserted_dev);
+                       const void *vals_dev /* dev double*[nvals] */,
+                       const int32_t *agg_ops_dev /* NULL = all SUM */,
+                       int nvals, int rstride /* pow2 words/slot >= 1+nvals */,
+                       int64_t *table, uint64_t cap, uint64_t *n_inserted);
+/* Initialize an INTERLEAVED group-by table: every slot record =
+ * [key=EMPTY | v0..v(nvals-1) = inits[c] | pad] of rstride 8-byte words
+ * (pow2 so a record never straddles a 128 B HBM line: find-or-insert
+ * touches ONE random line per row, not one per array). */
+int qk_groupby_init(void *stream, int64_t *table, uint64_t cap, int rstride,
+                    int nvals, const double *inits_dev);
 int qk_fill_f64(void *stream, double *dst_dev, double value, uint64_t n);
 /* Compact occupied slots to out_keys/out_sums (unordered); out_cursor_dev
  * (u64, zeroed) = number of groups. out capacity must be >= group count. */
-int qk_groupby_extract(void *stream, const int64_t *slot_keys,
-                       const double *slot_sums, int nvals, uint64_t capacity,
-                       int64_t *out_keys, double *out_sums,
-                       uint64_t out_capacity, uint64_t *out_cursor_dev);
+int qk_groupby_extract(void *stream, const int64_t *table, int rstride,
+                       int nvals, uint64_t cap, int64_t *out_keys,
+                       double *out_sums /* column-major, stride out_cap */,
+                       uint64_t out_cap, uint64_t *cursor);
 /* Thresholded extract: only groups whose sums[col] > threshold are
  * compacted (HAVING clauses, e.g. Q18's sum(l_quantity) > 300 over
  * ~n_orders groups — d2h of only the qualifying handful). */
-int qk_groupby_extract_gt(void *stream, const int64_t *slot_keys,
-                          const double *slot_sums, int nvals,
-                          uint64_t capacity, int col, double threshold,
+int qk_groupby_extract_gt(void *stream, const int64_t *table, int rstride,
+                          int nvals, uint64_t cap, int col, double threshold,
                           int64_t *out_keys, double *out_sums,
-                          uint64_t out_capacity, uint64_t *out_cursor_dev);
+                          uint64_t out_cap, uint64_t *cursor);
 
 /* ---- hash partition (shuffle map side) -------------------------------- *
  * Replaces partition_key_str (quokka_runtime.py:217-231). Int key semantics

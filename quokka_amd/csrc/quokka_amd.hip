@@ -1844,30 +1844,32 @@ __device__ inline void atomic_max_f64(double *addr, double v) {
 // agg_ops[c]: 0 = SUM (slot init 0), 1 = MIN (init +inf), 2 = MAX (-inf)
 __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
     uint64_t n, const int64_t *__restrict__ keys, const double *const *vals,
-    const int32_t *__restrict__ agg_ops, int nvals,
-    int64_t *__restrict__ slot_keys, double *__restrict__ slot_sums,
-    uint64_t cap, uint64_t *__restrict__ n_inserted) {
+    const int32_t *__restrict__ agg_ops, int nvals, int rstride,
+    int64_t *__restrict__ table, uint64_t cap,
+    uint64_t *__restrict__ n_inserted) {
+  // INTERLEAVED slot records: [key | v0 .. v(nvals-1) | pad] of rstride
+  // 8-byte words (pow2, so a record never straddles a 128 B line for
+  // rstride <= 16): find-or-insert touches ONE random HBM line per row
+  // instead of one line in the key array plus one per value array —
+  // the dominant cost of high-cardinality dedups (Q13/Q16/Q18/Q21).
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   uint32_t my_inserts = 0;
   for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
     int64_t key = keys[i];
     uint64_t s = slot_of(key, cap);
+    int64_t *rec;
     for (;;) {
-      int64_t cur = slot_keys[s];
+      rec = table + s * (uint64_t)rstride;
+      int64_t cur = rec[0];
       if (cur == key) break;
       if (cur == QK_JOIN_EMPTY) {
-        int64_t prev = (int64_t)atomicCAS((unsigned long long *)&slot_keys[s],
+        int64_t prev = (int64_t)atomicCAS((unsigned long long *)rec,
                                           (unsigned long long)QK_JOIN_EMPTY,
                                           (unsigned long long)key);
         if (prev == QK_JOIN_EMPTY) {
-          // this lane claimed a fresh slot: count the new group so the
-          // host can grow the table before cumulative distinct keys
-          // approach capacity (a full table would spin this loop
-          // forever). Counted per-thread and block-reduced at the end —
-          // a single cursor word takes ~88 atomics/us, and a
-          // high-cardinality batch (Q18: 150M groups) would serialize
-          // ~1.7 s on per-insert atomics.
+          // fresh slot claimed: counted per-thread, block-reduced below
+          // (a single cursor word serializes at ~88 atomics/us)
           my_inserts++;
           break;
         }
@@ -1875,16 +1877,16 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
       }
       s = (s + 1) & (cap - 1);
     }
+    double *vslot = (double *)(rec + 1);
     for (int c = 0; c < nvals; c++) {
-      double *dst = &slot_sums[(uint64_t)c * cap + s];
       double v = vals[c][i];
       int op = agg_ops ? agg_ops[c] : 0;
       if (op == 1)
-        atomic_min_f64(dst, v);
+        atomic_min_f64(&vslot[c], v);
       else if (op == 2)
-        atomic_max_f64(dst, v);
+        atomic_max_f64(&vslot[c], v);
       else
-        atomicAdd(dst, v);
+        atomicAdd(&vslot[c], v);
     }
   }
   if (n_inserted) {
@@ -1902,18 +1904,44 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_sum(
     }
   }
 }
+__global__ void k_groupby_init(uint64_t cap, int rstride, int nvals,
+                               const double *__restrict__ inits,
+                               int64_t *__restrict__ table) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += stride) {
+    int64_t *rec = table + s * (uint64_t)rstride;
+    rec[0] = QK_JOIN_EMPTY;
+    double *v = (double *)(rec + 1);
+    for (int c = 0; c < nvals; c++) v[c] = inits[c];
+  }
+}
+extern "C" int qk_groupby_init(void *stream, int64_t *table, uint64_t cap,
+                               int rstride, int nvals,
+                               const double *inits_dev) {
+  if (!cap) return 0;
+  uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
+  hipLaunchKernelGGL(k_groupby_init, dim3(blocks), dim3(BLOCK), 0,
+                     (hipStream_t)stream, cap, rstride, nvals, inits_dev,
+                     table);
+  QK_TRY("qk_groupby_init", hipGetLastError());
+  return 0;
+}
 extern "C" int qk_groupby_i64_sum(void *stream, uint64_t n, const int64_t *keys,
-                                  const double *const *vals_dev,
+                                  const void *vals_dev,
                                   const int32_t *agg_ops_dev, int nvals,
-                                  int64_t *slot_keys, double *slot_sums,
-                                  uint64_t cap, uint64_t *n_inserted_dev) {
+                                  int rstride, int64_t *table, uint64_t cap,
+                                  uint64_t *n_inserted) {
   if (!n) return 0;
   if (cap & (cap - 1))
     return qk_fail("qk_groupby_i64_sum.cap_pow2", hipErrorInvalidValue);
+  if (rstride < 1 + nvals || (rstride & (rstride - 1)))
+    return qk_fail("qk_groupby_i64_sum.rstride", hipErrorInvalidValue);
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (n + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_groupby_sum, dim3(blocks), dim3(BLOCK), 0,
-                     (hipStream_t)stream, n, keys, vals_dev, agg_ops_dev,
-                     nvals, slot_keys, slot_sums, cap, n_inserted_dev);
+                     (hipStream_t)stream, n, keys,
+                     (const double *const *)vals_dev, agg_ops_dev, nvals,
+                     rstride, table, cap, n_inserted);
   QK_TRY("qk_groupby_i64_sum", hipGetLastError());
   return 0;
 }
@@ -1934,18 +1962,18 @@ extern "C" int qk_fill_f64(void *stream, double *dst, double v, uint64_t n) {
 }
 
 __global__ void __launch_bounds__(BLOCK) k_groupby_extract(
-    const int64_t *__restrict__ slot_keys, const double *__restrict__ slot_sums,
-    int nvals, uint64_t cap, int64_t *__restrict__ out_keys,
-    double *__restrict__ out_sums, uint64_t out_cap,
-    uint64_t *__restrict__ cursor) {
-  // wave-aggregated output cursor: one atomicAdd per wave per stride
-  // step instead of one per non-empty slot (extracting ~180M groups
-  // serialized ~90 ms on same-address L2 atomics before this)
+    const int64_t *__restrict__ table, int rstride, int nvals, uint64_t cap,
+    int64_t *__restrict__ out_keys, double *__restrict__ out_sums,
+    uint64_t out_cap, uint64_t *__restrict__ cursor) {
+  // wave-aggregated output cursor (one atomicAdd per wave per step);
+  // reads INTERLEAVED slot records, writes COLUMN-MAJOR outputs
+  // (consumers index out_sums[c * out_cap + pos])
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   int lane = (int)(threadIdx.x & (WAVE - 1));
   for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
        s += stride) {
-    int64_t key = slot_keys[s];
+    const int64_t *rec = table + s * (uint64_t)rstride;
+    int64_t key = rec[0];
     bool has = key != QK_JOIN_EMPTY;
     uint64_t mask = __ballot(has);
     if (!mask) continue;
@@ -1959,21 +1987,20 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_extract(
       uint64_t pos = base + __popcll(mask & ((1ULL << lane) - 1));
       if (pos < out_cap) {
         out_keys[pos] = key;
+        const double *v = (const double *)(rec + 1);
         for (int c = 0; c < nvals; c++)
-          out_sums[(uint64_t)c * out_cap + pos] =
-              slot_sums[(uint64_t)c * cap + s];
+          out_sums[(uint64_t)c * out_cap + pos] = v[c];
       }
     }
   }
 }
-extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
-                                  const double *slot_sums, int nvals,
-                                  uint64_t cap, int64_t *out_keys,
-                                  double *out_sums, uint64_t out_cap,
-                                  uint64_t *cursor) {
+extern "C" int qk_groupby_extract(void *stream, const int64_t *table,
+                                  int rstride, int nvals, uint64_t cap,
+                                  int64_t *out_keys, double *out_sums,
+                                  uint64_t out_cap, uint64_t *cursor) {
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_groupby_extract, dim3(blocks), dim3(BLOCK), 0,
-                     (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
+                     (hipStream_t)stream, table, rstride, nvals, cap,
                      out_keys, out_sums, out_cap, cursor);
   QK_TRY("qk_groupby_extract", hipGetLastError());
   return 0;
@@ -1983,17 +2010,18 @@ extern "C" int qk_groupby_extract(void *stream, const int64_t *slot_keys,
 // sums[col] > threshold are compacted — Q18 qualifies a handful of its
 // ~n_orders groups, so the d2h stays tiny instead of GBs.
 __global__ void __launch_bounds__(BLOCK) k_groupby_extract_gt(
-    const int64_t *__restrict__ slot_keys, const double *__restrict__ slot_sums,
-    int nvals, uint64_t cap, int col, double threshold,
-    int64_t *__restrict__ out_keys, double *__restrict__ out_sums,
-    uint64_t out_cap, uint64_t *__restrict__ cursor) {
+    const int64_t *__restrict__ table, int rstride, int nvals, uint64_t cap,
+    int col, double threshold, int64_t *__restrict__ out_keys,
+    double *__restrict__ out_sums, uint64_t out_cap,
+    uint64_t *__restrict__ cursor) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   int lane = (int)(threadIdx.x & (WAVE - 1));
   for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
        s += stride) {
-    int64_t key = slot_keys[s];
-    bool has = key != QK_JOIN_EMPTY &&
-               slot_sums[(uint64_t)col * cap + s] > threshold;
+    const int64_t *rec = table + s * (uint64_t)rstride;
+    int64_t key = rec[0];
+    const double *v = (const double *)(rec + 1);
+    bool has = key != QK_JOIN_EMPTY && v[col] > threshold;
     uint64_t mask = __ballot(has);
     if (!mask) continue;
     int src = __ffsll((unsigned long long)mask) - 1;
@@ -2007,21 +2035,20 @@ __global__ void __launch_bounds__(BLOCK) k_groupby_extract_gt(
       if (pos < out_cap) {
         out_keys[pos] = key;
         for (int c = 0; c < nvals; c++)
-          out_sums[(uint64_t)c * out_cap + pos] =
-              slot_sums[(uint64_t)c * cap + s];
+          out_sums[(uint64_t)c * out_cap + pos] = v[c];
       }
     }
   }
 }
-extern "C" int qk_groupby_extract_gt(void *stream, const int64_t *slot_keys,
-                                     const double *slot_sums, int nvals,
-                                     uint64_t cap, int col, double threshold,
+extern "C" int qk_groupby_extract_gt(void *stream, const int64_t *table,
+                                     int rstride, int nvals, uint64_t cap,
+                                     int col, double threshold,
                                      int64_t *out_keys, double *out_sums,
                                      uint64_t out_cap, uint64_t *cursor) {
   uint32_t blocks = (uint32_t)qk_min_u64(MAX_BLOCKS, (cap + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_groupby_extract_gt, dim3(blocks), dim3(BLOCK), 0,
-                     (hipStream_t)stream, slot_keys, slot_sums, nvals, cap,
-                     col, threshold, out_keys, out_sums, out_cap, cursor);
+                     (hipStream_t)stream, table, rstride, nvals, cap, col,
+                     threshold, out_keys, out_sums, out_cap, cursor);
   QK_TRY("qk_groupby_extract_gt", hipGetLastError());
   return 0;
 }

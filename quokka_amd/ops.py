@@ -172,16 +172,21 @@ class GroupByI64:
         self.agg_ops = list(agg_ops) if agg_ops else [0] * nvals
         assert len(self.agg_ops) == nvals
         self.stream = stream
-        self.slot_keys = DevColumn(np.int64, self.cap)
-        self.slot_sums = DevColumn(np.float64, self.cap * nvals)
+        # INTERLEAVED slot records [key | v0..v(nvals-1) | pad], rstride
+        # pow2 8-byte words: find-or-insert touches ONE random HBM line
+        # per row (key and values share the line) instead of one line in
+        # the key array plus one per value array — the bound the
+        # high-cardinality dedups (Q13/Q16/Q18/Q21) sit on
+        self.rstride = 1
+        while self.rstride < 1 + nvals:
+            self.rstride <<= 1
+        self.table = DevColumn(np.int64, self.cap * self.rstride)
+        self._inits_dev = DevColumn.from_numpy(
+            np.asarray([self.AGG_INIT[op] for op in self.agg_ops],
+                       dtype=np.float64))
         sh = stream.handle if stream else None
-        shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
-                  c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
-        import ctypes as _ct
-        for c, op in enumerate(self.agg_ops):
-            shim.call("qk_fill_f64", sh,
-                      shim.c_vp(self.slot_sums.ptr.value + c * self.cap * 8),
-                      _ct.c_double(self.AGG_INIT[op]), c_u64(self.cap))
+        shim.call("qk_groupby_init", sh, self.table.ptr, c_u64(self.cap),
+                  self.rstride, self.nvals, self._inits_dev.ptr)
         self._ops_dev = None
         if any(self.agg_ops):
             self._ops_dev = DevColumn.from_numpy(
@@ -212,8 +217,8 @@ class GroupByI64:
                   c_u64(ptrs.nbytes))
         shim.call("qk_groupby_i64_sum", sh, c_u64(n), keys_col.ptr,
                   dptrs.ptr, self._ops_dev.ptr if self._ops_dev else None,
-                  self.nvals, self.slot_keys.ptr,
-                  self.slot_sums.ptr, c_u64(self.cap), self._nins.ptr)
+                  self.nvals, self.rstride, self.table.ptr,
+                  c_u64(self.cap), self._nins.ptr)
         if self.stream:
             self.stream.sync()
         self.n_groups = _read_u64(self._nins)
@@ -226,19 +231,12 @@ class GroupByI64:
         the op identity) alike. Mirrors GPUDistinctExecutor._grow."""
         keys, sums = self.extract()
         old_groups = len(keys)
-        for c in (self.slot_keys, self.slot_sums):
-            c.free()
+        self.table.free()
         self.cap = _pow2_at_least(max(16, 4 * int(need_groups)))
-        self.slot_keys = DevColumn(np.int64, self.cap)
-        self.slot_sums = DevColumn(np.float64, self.cap * self.nvals)
+        self.table = DevColumn(np.int64, self.cap * self.rstride)
         sh = self.stream.handle if self.stream else None
-        shim.call("qk_fill_i64", sh, self.slot_keys.ptr,
-                  c_i64(int(shim.JOIN_EMPTY)), c_u64(self.cap))
-        import ctypes as _ct
-        for c, op in enumerate(self.agg_ops):
-            shim.call("qk_fill_f64", sh,
-                      shim.c_vp(self.slot_sums.ptr.value + c * self.cap * 8),
-                      _ct.c_double(self.AGG_INIT[op]), c_u64(self.cap))
+        shim.call("qk_groupby_init", sh, self.table.ptr, c_u64(self.cap),
+                  self.rstride, self.nvals, self._inits_dev.ptr)
         shim.call("qk_dmemset", self._nins.ptr, 0, c_u64(8))
         self.n_groups = 0
         if old_groups:
@@ -257,8 +255,8 @@ class GroupByI64:
         out_keys = DevColumn(np.int64, out_cap)
         out_sums = DevColumn(np.float64, out_cap * self.nvals)
         cur = _count_buf()
-        shim.call("qk_groupby_extract", sh, self.slot_keys.ptr,
-                  self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+        shim.call("qk_groupby_extract", sh, self.table.ptr,
+                  self.rstride, self.nvals, c_u64(self.cap),
                   out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
         if self.stream:
             self.stream.sync()
@@ -289,8 +287,8 @@ class GroupByI64:
         out_keys = DevColumn(np.int64, out_cap)
         out_sums = DevColumn(np.float64, out_cap * self.nvals)
         cur = _count_buf()
-        shim.call("qk_groupby_extract", sh, self.slot_keys.ptr,
-                  self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+        shim.call("qk_groupby_extract", sh, self.table.ptr,
+                  self.rstride, self.nvals, c_u64(self.cap),
                   out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
         if self.stream:
             self.stream.sync()
@@ -312,8 +310,8 @@ class GroupByI64:
             out_keys = DevColumn(np.int64, out_cap)
             out_sums = DevColumn(np.float64, out_cap * self.nvals)
             cur = _count_buf()
-            shim.call("qk_groupby_extract_gt", sh, self.slot_keys.ptr,
-                      self.slot_sums.ptr, self.nvals, c_u64(self.cap),
+            shim.call("qk_groupby_extract_gt", sh, self.table.ptr,
+                      self.rstride, self.nvals, c_u64(self.cap),
                       int(col), _ct.c_double(float(threshold)),
                       out_keys.ptr, out_sums.ptr, c_u64(out_cap), cur.ptr)
             if self.stream:
@@ -332,8 +330,8 @@ class GroupByI64:
             out_cap = int(k)
 
     def free(self):
-        self.slot_keys.free()
-        self.slot_sums.free()
+        self.table.free()
+        self._inits_dev.free()
         self._nins.free()
         if self._ops_dev is not None:
             self._ops_dev.free()

@@ -109,12 +109,15 @@ _SIGS = {
     "qk_q3_extract": [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
                       c_u64, c_vp],
     "qk_groupby_i64_sum": [c_vp, c_u64, c_vp, c_vp, c_vp, ctypes.c_int,
-                           c_vp, c_vp, c_u64, c_vp],
+                           ctypes.c_int, c_vp, c_u64, c_vp],
+    "qk_groupby_init": [c_vp, c_vp, c_u64, ctypes.c_int, ctypes.c_int,
+                        c_vp],
     "qk_fill_f64": [c_vp, c_vp, c_f64, c_u64],
-    "qk_groupby_extract": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64, c_vp, c_vp,
-                           c_u64, c_vp],
-    "qk_groupby_extract_gt": [c_vp, c_vp, c_vp, ctypes.c_int, c_u64,
-                              ctypes.c_int, c_f64, c_vp, c_vp, c_u64, c_vp],
+    "qk_groupby_extract": [c_vp, c_vp, ctypes.c_int, ctypes.c_int, c_u64,
+                           c_vp, c_vp, c_u64, c_vp],
+    "qk_groupby_extract_gt": [c_vp, c_vp, ctypes.c_int, ctypes.c_int,
+                              c_u64, ctypes.c_int, c_f64, c_vp, c_vp,
+                              c_u64, c_vp],
     "qk_bloom_count": [c_vp, c_u64, c_vp, c_vp, c_u64, c_vp],
     "qk_sort_pairs_u64": [c_vp, c_u64, c_vp, c_vp, c_vp, c_vp,
                           ctypes.c_int],
