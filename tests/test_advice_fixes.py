@@ -229,3 +229,78 @@ def test_agg_executor_typed_errors():
     # the supported form still parses
     ex = GPUAggExecutor(["k"], [], "sum(s0) / sum(c0) as avg_x")
     assert ex.sum_cols == ["s0", "c0"]
+
+
+def test_groupby_wide_nvals_padded_stride(gpu):
+    """nvals=5 and 7 force the interleaved slot record to pad to
+    rstride=8 words: padding must never leak into values or keys."""
+    from quokka_amd import ops, shim
+    rng = np.random.default_rng(99)
+    n = 20_000
+    keys = rng.integers(0, 500, n).astype(np.int64)
+    for nvals in (5, 7):
+        agg_ops = [int(o) for o in rng.integers(0, 3, nvals)]
+        vals = [np.round(rng.standard_normal(n), 4) for _ in range(nvals)]
+        gb = ops.GroupByI64(500, nvals, agg_ops=agg_ops)
+        assert gb.rstride == 8
+        kcol = shim.DevColumn.from_numpy(keys)
+        vcols = [shim.DevColumn.from_numpy(v) for v in vals]
+        gb.update(kcol, vcols)
+        gk, gs = gb.extract()
+        order = np.argsort(gk)
+        uk = np.unique(keys)
+        assert np.array_equal(gk[order], uk)
+        for c, (op, v) in enumerate(zip(agg_ops, vals)):
+            fn = {0: np.sum, 1: np.min, 2: np.max}[op]
+            want = np.array([fn(v[keys == k]) for k in uk])
+            np.testing.assert_allclose(gs[c][order], want, rtol=1e-9)
+        gb.free()
+        kcol.free()
+        for c in vcols:
+            c.free()
+
+
+def test_groupby_extracts_after_growth(gpu):
+    """extract_where_gt and extract_device must read the REBUILT table
+    after _grow (two batches force growth past the initial sizing)."""
+    from quokka_amd import ops, shim
+    from quokka_amd.shim import DevColumnView
+    rng = np.random.default_rng(17)
+    gb = ops.GroupByI64(expected_groups=8, nvals=2, agg_ops=[0, 1])
+    all_k, all_v0, all_v1 = [], [], []
+    for batch in range(3):
+        n = 4000
+        keys = rng.integers(0, 900, n).astype(np.int64) + 1
+        v0 = np.ones(n)
+        v1 = rng.uniform(0, 100, n)
+        kcol = shim.DevColumn.from_numpy(keys)
+        c0 = shim.DevColumn.from_numpy(v0)
+        c1 = shim.DevColumn.from_numpy(v1)
+        gb.update(kcol, [c0, c1])
+        for c in (kcol, c0, c1):
+            c.free()
+        all_k.append(keys)
+        all_v0.append(v0)
+        all_v1.append(v1)
+    keys = np.concatenate(all_k)
+    counts = {int(k): int((keys == k).sum()) for k in np.unique(keys)}
+    assert gb.cap > 16                      # growth actually happened
+    # HAVING: count > 15
+    gk, gs = gb.extract_where_gt(0, 15.0)
+    want = sorted(k for k, c in counts.items() if c > 15)
+    assert sorted(gk.tolist()) == want
+    # device extract: keys + column-major sums views
+    dkeys, dsums, k, cap = gb.extract_device()
+    assert k == len(counts)
+    got_counts = DevColumnView(dsums, 0, k).to_numpy(k)
+    order = np.argsort(dkeys.to_numpy(k))
+    want_counts = np.array([counts[k_] for k_ in
+                            sorted(counts)], dtype=np.float64)
+    np.testing.assert_allclose(got_counts[order], want_counts)
+    v1all = np.concatenate(all_v1)
+    want_min = np.array([v1all[keys == k_].min() for k_ in sorted(counts)])
+    got_min = DevColumnView(dsums, cap, k).to_numpy(k)
+    np.testing.assert_allclose(got_min[order], want_min, rtol=1e-12)
+    dkeys.free()
+    dsums.free()
+    gb.free()
