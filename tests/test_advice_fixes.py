@@ -231,6 +231,7 @@ def test_agg_executor_typed_errors():
     assert ex.sum_cols == ["s0", "c0"]
 
 
+@pytest.mark.gpu
 def test_groupby_wide_nvals_padded_stride(gpu):
     """nvals=5 and 7 force the interleaved slot record to pad to
     rstride=8 words: padding must never leak into values or keys."""
@@ -260,6 +261,7 @@ def test_groupby_wide_nvals_padded_stride(gpu):
             c.free()
 
 
+@pytest.mark.gpu
 def test_groupby_extracts_after_growth(gpu):
     """extract_where_gt and extract_device must read the REBUILT table
     after _grow (two batches force growth past the initial sizing)."""
