@@ -369,3 +369,46 @@ def test_q22_device_vs_oracle(gpu, data):
     for cs in (ccols, ocols):
         for c in cs.values():
             c.free()
+
+
+def test_pipelines_on_empty_and_single_row(gpu):
+    """Degenerate inputs through the composed pipelines: zero-row
+    tables must produce empty/zero results (not crash in the n=0
+    kernel paths), and a single-row table must round-trip exactly."""
+    from quokka_amd import staging, queries as DQ
+    empty_li = {"l_orderkey": np.empty(0, np.int64),
+                "l_shipdate": np.empty(0, np.int32),
+                "l_commitdate": np.empty(0, np.int32),
+                "l_receiptdate": np.empty(0, np.int32),
+                "l_shipmode": np.empty(0, np.uint8)}
+    empty_od = {"o_orderkey": np.empty(0, np.int64),
+                "o_custkey": np.empty(0, np.int64),
+                "o_orderdate": np.empty(0, np.int32),
+                "o_orderpriority": np.empty(0, np.uint8),
+                "o_comment_special": np.empty(0, np.uint8)}
+    lcols = staging.stage_columns(empty_li)
+    ocols = staging.stage_columns(empty_od)
+    r12 = DQ.q12(lcols, ocols)
+    assert all(v == (0, 0) for v in r12.values())
+    r4 = DQ.q4(lcols, ocols)
+    assert all(v == 0 for v in r4.values())
+    r13 = DQ.q13(ocols, 10)
+    assert r13 == {0: 10}                 # all 10 customers zero-order
+    for cs in (lcols, ocols):
+        for c in cs.values():
+            c.free()
+    # one qualifying row end-to-end through Q12
+    li1 = {"l_orderkey": np.array([33], np.int64),
+           "l_shipdate": np.array([8900], np.int32),
+           "l_commitdate": np.array([8950], np.int32),
+           "l_receiptdate": np.array([8960], np.int32),  # 1994, in window
+           "l_shipmode": np.array([2], np.uint8)}        # MAIL
+    od1 = {"o_orderkey": np.array([33], np.int64),
+           "o_orderpriority": np.array([0], np.uint8)}   # 1-URGENT: high
+    lcols = staging.stage_columns(li1)
+    ocols = staging.stage_columns(od1)
+    r = DQ.q12(lcols, ocols)
+    assert r["MAIL"] == (1, 0) and r["SHIP"] == (0, 0)
+    for cs in (lcols, ocols):
+        for c in cs.values():
+            c.free()
